@@ -1,0 +1,239 @@
+/* trino_gpu.h — C-ABI drop-in boundary for Trino's columnar operator hot path,
+ * MI355X-native implementation (libtrino_gpu.so).
+ *
+ * Each entry point mirrors the io.trino SPI/operator interface a JNI shim would
+ * bind (the reference has no native code; this is the surface LocalExecutionPlanner
+ * -constructed operator factories would delegate to). Reference interfaces mirrored:
+ *   - Page/Block:   core/trino-spi/src/main/java/io/trino/spi/Page.java:31,50-51
+ *                   spi/block/Block.java:21-24 (ValueBlock|DictionaryBlock|RLE),
+ *                   spi/block/LongArrayBlock.java:39-43 (flat values + valid bitmap),
+ *                   spi/block/VariableWidthBlock.java:43-48, DictionaryBlock.java:55-58
+ *   - SelectedPositions: operator/project/SelectedPositions.java:27-58
+ *   - Operator:     operator/Operator.java:18-50
+ *                   (needsInput()/addInput(Page)/getOutput()/finish()/isFinished())
+ *   - OperatorFactory creation sites: sql/planner/LocalExecutionPlanner.java:2121
+ *                   (scan+filter+project), :4080 (hash agg), :2966/:3017 (hash build),
+ *                   OperatorFactories.java:24-47 (lookup join),
+ *                   operator/output/PartitionedOutputOperator.java (partitioned output)
+ * No torch types; plain pointers and sizes only. See INTEGRATION.md for the JNI
+ * binding a Trino maintainer would add.
+ */
+#ifndef TRINO_GPU_H
+#define TRINO_GPU_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- error codes (mirrors unchecked TrinoException at the boundary) ---- */
+typedef enum {
+    TG_OK = 0,
+    TG_ERR_INVALID_ARG = 1,
+    TG_ERR_NO_GPU = 2,        /* HIP device unavailable: the product path FAILS, no CPU fallback */
+    TG_ERR_OOM = 3,
+    TG_ERR_STATE = 4,         /* operator state machine violation */
+    TG_ERR_HIP = 5,           /* underlying HIP error; see tg_last_error() */
+    TG_ERR_UNSUPPORTED = 6
+} tg_status;
+
+const char* tg_last_error(void);
+/* library version / build arch probe */
+const char* tg_version(void);
+
+/* ---- types (spi/type): the subset on the hot path ---- */
+typedef enum {
+    TG_BIGINT = 0,   /* i64  (LongArrayBlock)   */
+    TG_INTEGER = 1,  /* i32  (IntArrayBlock)    */
+    TG_SMALLINT = 2, /* i16  (ShortArrayBlock)  */
+    TG_TINYINT = 3,  /* i8   (ByteArrayBlock)   */
+    TG_DOUBLE = 4,   /* f64  (LongArrayBlock bits) */
+    TG_DATE = 5,     /* i32 epoch days (IntArrayBlock) */
+    TG_BOOLEAN = 6,  /* i8   (ByteArrayBlock)   */
+    TG_VARCHAR = 7   /* VariableWidthBlock      */
+} tg_type;
+
+typedef enum { TG_BK_VALUE = 0, TG_BK_DICTIONARY = 1, TG_BK_RLE = 2 } tg_block_kind;
+
+/* Flat column descriptor. Pointers may be HOST or DEVICE memory; `on_device`
+ * says which. Host blocks are uploaded (flat memcpy of the backing arrays —
+ * zero pivot) on addInput. */
+typedef struct tg_block {
+    int32_t type;               /* tg_type */
+    int32_t kind;               /* tg_block_kind */
+    int64_t position_count;
+    int32_t on_device;          /* 0 = host pointers, 1 = device pointers */
+    int32_t _pad;
+    const void* data;           /* fixed-width values, or utf8 bytes for VARCHAR */
+    const uint64_t* valid;      /* packed bitmap, bit=1 => valid (LongArrayBlock.valueIsValid); NULL => no nulls */
+    const int32_t* offsets;     /* VARCHAR: N+1 offsets into data */
+    const int32_t* ids;         /* DICTIONARY: position -> dictionary id */
+    const struct tg_block* dictionary; /* DICTIONARY/RLE: the value block */
+} tg_block;
+
+typedef struct tg_page {
+    int32_t channel_count;
+    int64_t position_count;
+    const tg_block* blocks;     /* array[channel_count] */
+} tg_page;
+
+/* SelectedPositions (operator/project/SelectedPositions.java:27-58) */
+typedef struct tg_selected {
+    int32_t is_list;            /* 0: range [offset, offset+size); 1: positions list */
+    int32_t offset;
+    int32_t size;
+    const int32_t* positions;   /* when is_list */
+} tg_selected;
+
+/* ---- session = device context ---- */
+typedef struct tg_session tg_session;
+tg_status tg_session_create(int device_ordinal, tg_session** out);
+void      tg_session_close(tg_session*);
+
+/* ---- expression IR (replaces per-query bytecode gen, sql/gen/columnar/) ----
+ * Postfix program over f64/i64 lanes; enough for TPC-H filter/project forms. */
+typedef enum {
+    TG_EXPR_COL = 0,       /* push column[arg0] value */
+    TG_EXPR_CONST_F64 = 1, /* push f64 constant */
+    TG_EXPR_CONST_I64 = 2, /* push i64 constant */
+    TG_EXPR_ADD = 3, TG_EXPR_SUB = 4, TG_EXPR_MUL = 5, TG_EXPR_DIV = 6,
+    /* comparisons produce boolean (filter roots) */
+    TG_EXPR_LE = 7, TG_EXPR_LT = 8, TG_EXPR_GE = 9, TG_EXPR_GT = 10,
+    TG_EXPR_EQ = 11, TG_EXPR_NE = 12,
+    TG_EXPR_AND = 13, TG_EXPR_OR = 14, TG_EXPR_NOT = 15,
+    TG_EXPR_BETWEEN = 16,  /* value, lo, hi on stack */
+    TG_EXPR_IN_I64 = 17    /* arg0 = count, followed by imm list (host side) */
+} tg_expr_op;
+
+typedef struct tg_expr_inst {
+    int32_t op;
+    int32_t arg0;
+    union { double f64; int64_t i64; } imm;
+} tg_expr_inst;
+
+typedef struct tg_expr { const tg_expr_inst* insts; int32_t count; } tg_expr;
+
+/* ---- generic operator handle (operator/Operator.java:18-50) ---- */
+typedef struct tg_operator tg_operator;
+int       tg_operator_needs_input(tg_operator*);
+tg_status tg_operator_add_input(tg_operator*, const tg_page*);
+/* getOutput: fills out_page with DEVICE-resident blocks owned by the operator
+ * until the next call; returns TG_OK with out_page->position_count==0 and
+ * channel_count==0 when no output is ready. *finished set when drained. */
+tg_status tg_operator_get_output(tg_operator*, tg_page* out_page, int* finished);
+tg_status tg_operator_finish(tg_operator*);
+void      tg_operator_close(tg_operator*);
+
+/* ---- FilterAndProject (ScanFilterAndProjectOperator.java:66 /
+ *      PageProcessor.java:102-138; filter = ColumnarFilter.java:42-52) ---- */
+tg_status tg_filter_project_create(tg_session*,
+    const tg_expr* filter,            /* NULL = select all */
+    const tg_expr* projections,       /* array[n_proj] */
+    const int32_t* proj_out_types,    /* tg_type per projection */
+    int32_t n_proj,
+    tg_operator** out);
+
+/* Standalone columnar filter: predicate -> selection vector
+ * (ColumnarFilter.filterPositionsRange/List). Synchronous helper used by tests. */
+tg_status tg_filter_run(tg_session*, const tg_expr* filter, const tg_page* page,
+                        const tg_selected* input_sel,
+                        int32_t* out_positions, int32_t* out_count);
+
+/* ---- HashAggregation (HashAggregationOperator; builder
+ *      InMemoryHashAggregationBuilder.java:140-158) ---- */
+typedef enum { TG_STEP_PARTIAL = 0, TG_STEP_FINAL = 1, TG_STEP_SINGLE = 2 } tg_agg_step;
+typedef enum {
+    TG_AGG_COUNT_STAR = 0,  /* CountAggregation */
+    TG_AGG_COUNT_COL  = 1,
+    TG_AGG_SUM_F64    = 2,  /* DoubleSumAggregation.java:37-45 */
+    TG_AGG_SUM_I64    = 3,  /* BigintSumAggregation */
+    TG_AGG_AVG_F64    = 4   /* DoubleAverageAggregations.java:38-63 (state: count,sum) */
+} tg_agg_fn;
+typedef struct tg_agg_spec { int32_t fn; int32_t input_channel; } tg_agg_spec;
+
+tg_status tg_hash_aggregation_create(tg_session*,
+    const int32_t* group_channels, int32_t n_group_channels,
+    const int32_t* group_types,               /* tg_type per group channel */
+    const tg_agg_spec* aggs, int32_t n_aggs,
+    int32_t step,                              /* tg_agg_step */
+    tg_operator** out);
+
+/* ---- Hash join (HashBuilderOperator nonspilling + LookupJoinOperator) ---- */
+typedef struct tg_join_bridge tg_join_bridge;   /* JoinBridgeManager analog */
+tg_status tg_join_bridge_create(tg_session*, tg_join_bridge** out);
+void      tg_join_bridge_close(tg_join_bridge*);
+
+tg_status tg_hash_builder_create(tg_session*, tg_join_bridge*,
+    const int32_t* build_types, int32_t n_build_channels,
+    const int32_t* key_channels, int32_t n_key_channels,
+    const int32_t* output_channels, int32_t n_output_channels,
+    tg_operator** out);
+
+tg_status tg_lookup_join_create(tg_session*, tg_join_bridge*,
+    const int32_t* probe_types, int32_t n_probe_channels,
+    const int32_t* key_channels, int32_t n_key_channels,
+    const int32_t* probe_output_channels, int32_t n_probe_output,
+    tg_operator** out);
+
+/* ---- Partitioned output (PagePartitioner.java:134-330) ----
+ * Computes per-row partitions with the canonical row hash
+ * (InterpretedHashGenerator.java:57-110, combine 31*h, bigint xxmix) and
+ * splits the page into per-partition pages. Transport (RCCL all-to-all)
+ * happens above this ABI in the rank runner. */
+tg_status tg_page_partitioner_create(tg_session*,
+    const int32_t* types, int32_t n_channels,
+    const int32_t* partition_channels, int32_t n_partition_channels,
+    int32_t partition_count,
+    tg_operator** out);
+/* after add_input, fetch partition p's accumulated page */
+tg_status tg_page_partitioner_get_partition(tg_operator*, int32_t partition,
+                                            tg_page* out_page);
+
+/* ---- synchronous row-hash helper (tests/parity): canonical row hash ---- */
+tg_status tg_hash_rows(tg_session*, const tg_page* page,
+                       const int32_t* channels, int32_t n_channels,
+                       uint64_t* out_hashes /* host, length position_count */);
+
+/* ---- TPC-H device generator (bench inputs; restates io.trino.tpch v1.4
+ *      column streams — plugin/trino-tpch TpchRecordSet.java call sites) ---- */
+typedef struct tg_tpch_lineitem_cols {
+    /* all DEVICE pointers, length row_count */
+    int64_t row_count;
+    int64_t* orderkey;      /* may be NULL if not requested */
+    int32_t* shipdate;      /* epoch days */
+    double*  quantity;
+    double*  extendedprice;
+    double*  discount;
+    double*  tax;
+    uint8_t* returnflag;    /* dictionary id: 0=A 1=N 2=R */
+    uint8_t* linestatus;    /* dictionary id: 0=F 1=O */
+} tg_tpch_lineitem_cols;
+
+/* Generate lineitem rows for orders [order_start, order_start+order_count)
+ * (1-based dense order index, part of 1,500,000×SF orders). Buffers must be
+ * pre-sized to >= 7*order_count rows; actual count returned. */
+tg_status tg_tpch_gen_lineitem(tg_session*, double scale_factor,
+    int64_t order_start, int64_t order_count,
+    tg_tpch_lineitem_cols* cols /* in: buffers, out: row_count */);
+
+/* ---- fused TPC-H Q1 pipeline (the north-star benchmark kernel):
+ * scan+filter(shipdate<=cutoff)+group(returnflag,linestatus)+7 aggregates,
+ * one pass over HBM (38 B/row algorithmic). Results per (rf,ls) combo. ---- */
+typedef struct tg_q1_result {
+    /* indexed [rf*2+ls], rf in {A,N,R}=0,1,2; ls in {F,O}=0,1 */
+    double sum_qty[6], sum_base[6], sum_disc_price[6], sum_charge[6];
+    double avg_qty[6], avg_price[6], avg_disc[6];
+    double sum_disc[6];
+    int64_t count[6];
+    double elapsed_ms;          /* kernel time, HIP events */
+} tg_q1_result;
+
+tg_status tg_q1_run(tg_session*, const tg_tpch_lineitem_cols* cols,
+                    int32_t shipdate_cutoff, tg_q1_result* out);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* TRINO_GPU_H */

@@ -1,0 +1,98 @@
+/* tpchgen.c — CPU oracle entry points for the TPC-H generator restatement.
+ * ORACLE / TEST INFRASTRUCTURE ONLY (see tpch_core.h header for provenance
+ * and parity pins). Exported for ctypes use by tests/ and bench.py's
+ * cpu_baseline leg only; the product path never links this.
+ */
+#include "tpch_core.h"
+#include <string.h>
+
+#define EXPORT __attribute__((visibility("default")))
+
+/* Count lineitem rows for a dense order range [order_start, order_start+order_count)
+ * (1-based). Only consumes the O_LCNT stream. */
+EXPORT int64_t tpch_lineitem_count(double sf, int64_t order_start, int64_t order_count)
+{
+    (void)sf;
+    tpch_rng lcnt;
+    tpch_rng_init(&lcnt, TPCH_SEED_O_LCNT, 1);
+    tpch_rng_skip(&lcnt, order_start - 1);
+    int64_t total = 0;
+    for (int64_t i = 0; i < order_count; i++)
+        total += tpch_rng_int(&lcnt, 1, TPCH_LINES_PER_ORDER_MAX);
+    return total;
+}
+
+/* Generate lineitem columns for a dense order range. Output arrays must hold
+ * >= 7*order_count entries; any pointer may be NULL to skip that column.
+ * Doubles are exactly the values the reference's cursor surfaces
+ * (TpchRecordSet.java:192-210: DATE as epoch-day long, DOUBLE = cents/100.0).
+ * Returns the number of rows written. */
+EXPORT int64_t tpch_gen_lineitem(double sf, int64_t order_start, int64_t order_count,
+    int64_t* orderkey, int64_t* partkey, int32_t* linenumber,
+    int32_t* shipdate, int32_t* commitdate, int32_t* receiptdate,
+    double* quantity, double* extendedprice, double* discount, double* tax,
+    uint8_t* returnflag, uint8_t* linestatus)
+{
+    tpch_order_streams s;
+    tpch_order_streams_init(&s, sf);
+    tpch_order_streams_seek(&s, order_start);
+    int64_t n = 0;
+    for (int64_t i = 0; i < order_count; i++) {
+        tpch_order_row o;
+        tpch_gen_order(&s, order_start + i, &o);
+        for (int j = 0; j < o.line_count; j++, n++) {
+            tpch_lineitem_row l;
+            tpch_gen_line(&s, &o, j, &l);
+            if (orderkey)      orderkey[n] = l.orderkey;
+            if (partkey)       partkey[n] = l.partkey;
+            if (linenumber)    linenumber[n] = l.linenumber;
+            if (shipdate)      shipdate[n] = l.shipdate;
+            if (commitdate)    commitdate[n] = l.commitdate;
+            if (receiptdate)   receiptdate[n] = l.receiptdate;
+            if (quantity)      quantity[n] = (double)l.qty;
+            if (extendedprice) extendedprice[n] = tpch_cents_to_double(l.extprice_cents);
+            if (discount)      discount[n] = (double)l.discount_pct / 100.0;
+            if (tax)           tax[n] = (double)l.tax_pct / 100.0;
+            if (returnflag)    returnflag[n] = l.returnflag;
+            if (linestatus)    linestatus[n] = l.linestatus;
+        }
+        tpch_order_row_finished(&s);
+    }
+    return n;
+}
+
+/* Generate order headers (Q3 build side): orderkey, custkey, orderdate. */
+EXPORT int64_t tpch_gen_orders(double sf, int64_t order_start, int64_t order_count,
+    int64_t* orderkey, int64_t* custkey, int32_t* orderdate)
+{
+    tpch_order_streams s;
+    tpch_order_streams_init(&s, sf);
+    tpch_order_streams_seek(&s, order_start);
+    for (int64_t i = 0; i < order_count; i++) {
+        tpch_order_row o;
+        tpch_gen_order(&s, order_start + i, &o);
+        if (orderkey)  orderkey[i] = o.orderkey;
+        if (custkey)   custkey[i] = o.custkey;
+        if (orderdate) orderdate[i] = o.orderdate;
+        tpch_order_row_finished(&s);
+    }
+    return order_count;
+}
+
+/* Customer mktsegment ids (Q3): id 0..4 in dists.dss order
+ * AUTOMOBILE,BUILDING,FURNITURE,MACHINERY,HOUSEHOLD. custkey = dense index. */
+EXPORT int64_t tpch_gen_customer(double sf, int64_t cust_start, int64_t cust_count,
+    int64_t* custkey, uint8_t* mktsegment)
+{
+    (void)sf;
+    tpch_rng mseg;
+    tpch_rng_init(&mseg, TPCH_SEED_C_MSEG, 1);
+    tpch_rng_skip(&mseg, cust_start - 1);
+    for (int64_t i = 0; i < cust_count; i++) {
+        if (custkey) custkey[i] = cust_start + i;
+        int64_t pick = tpch_rng_int(&mseg, 1, 5);
+        if (mktsegment) mktsegment[i] = (uint8_t)(pick - 1);
+        tpch_rng_row_finished(&mseg);
+    }
+    return cust_count;
+}
