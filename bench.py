@@ -1,0 +1,223 @@
+#!/usr/bin/env python3
+"""bench.py — TPC-H Q1 scan/filter/agg benchmark (BASELINE.json metric:
+rows/sec + achieved HBM GB/s on TPC-H Q1, SF100 per GPU).
+
+A "step" = one full pass of the fused Q1 path over this rank's HBM-resident
+lineitem columns (device-generated TPC-H data, 38 B/row algorithmic). N>1 =
+one process per GPU over RCCL (weak scaling: SF100 shard per GPU = one part of
+SF 100*N), plus the partial-state exchange leg (BASELINE config 4 shape).
+
+Run:  python bench.py [--gpus N] [--steps K] [--warmup W] [--sf-per-gpu S]
+Driver launches N>1 via torch.distributed.run; reads RANK/WORLD_SIZE/etc.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+HBM_PEAK_GBPS = 8000.0         # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+ALG_BYTES_PER_ROW = 38         # shipdate i32 + 4*f64 + 2*u8
+Q1_CUTOFF = 10471              # 1998-09-02 (shipdate <= date '1998-12-01' - 90 day)
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def merge_raw(raws):
+    """Exact cross-rank merge of raw integer Q1 accumulators (python ints)."""
+    total = [0] * 60
+    for raw in raws:
+        for c in range(6):
+            for f in range(4):  # u128 fields
+                lo, hi = raw[c * 10 + 2 * f], raw[c * 10 + 2 * f + 1]
+                cur = total[c * 10 + 2 * f] | (total[c * 10 + 2 * f + 1] << 64)
+                cur += lo | (hi << 64)
+                total[c * 10 + 2 * f] = cur & ((1 << 64) - 1)
+                total[c * 10 + 2 * f + 1] = cur >> 64
+            total[c * 10 + 8] += raw[c * 10 + 8]
+            total[c * 10 + 9] += raw[c * 10 + 9]
+    return total
+
+
+def combo_owner_rank(combo, world):
+    """Exchange routing for the partial->final aggregation leg: the canonical
+    row hash of the (returnflag, linestatus) group key (31*combine of the
+    bigint xxmix per channel — HashGenerator.java:20, AbstractLongType.java:
+    121-125) reduced by the remote partition function (HashGenerator.java:
+    41-46)."""
+    m = (1 << 64) - 1
+
+    def rotl(x, r):
+        return ((x << r) | (x >> (64 - r))) & m
+
+    def xxmix(v):
+        return (rotl((v * 0xC2B2AE3D27D4EB4F) & m, 31) * 0x9E3779B185EBCA87) & m
+
+    rf, ls = combo // 2, combo % 2
+    h = (31 * xxmix(rf) + xxmix(ls)) & m
+    lh = (h ^ (h >> 32)) & 0xFFFFFFFF
+    return (lh * world) >> 32
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--sf-per-gpu", type=float, default=100.0)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(world, args.gpus)
+
+    import torch
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        torch.cuda.set_device(local_rank)
+        tdist.init_process_group(backend="nccl")
+
+    import trino_amd
+
+    sess = trino_amd.Session(local_rank)
+
+    # ---- workload: this rank's shard of SF (sf_per_gpu * N): contiguous
+    # order parts, exactly TpchSplit(part, totalParts) semantics ----
+    total_sf = args.sf_per_gpu * n_gpus
+    total_orders = int(1_500_000 * args.sf_per_gpu) * n_gpus
+    per = total_orders // n_gpus
+    order_start = 1 + rank * per
+    order_count = per + (total_orders % n_gpus if rank == n_gpus - 1 else 0)
+
+    t0 = time.time()
+    cols = sess.tpch_lineitem(total_sf, order_start, order_count)
+    rows_rank = cols.row_count
+    log(f"generated {rows_rank:,} lineitem rows/rank (SF{total_sf:g} part {rank+1}/{n_gpus}) "
+        f"in {time.time()-t0:.1f}s — {rows_rank*ALG_BYTES_PER_ROW/2**30:.1f} GiB in HBM")
+
+    if dist:
+        rows_t = torch.tensor([rows_rank], dtype=torch.int64, device="cuda")
+        dist.all_reduce(rows_t)
+        rows_total = int(rows_t.item())
+    else:
+        rows_total = rows_rank
+
+    def step():
+        r = sess.q1(cols, Q1_CUTOFF)
+        if dist:
+            # exchange leg (config 4 shape): route each group's partial state
+            # to its owner rank (all_to_all over RCCL/xGMI), final-merge there.
+            signed = [x - (1 << 64) if x >= (1 << 63) else x for x in r.raw]
+            raw = torch.tensor(signed, dtype=torch.int64, device="cuda").view(6, 10)
+            send = [torch.zeros(6, 10, dtype=torch.int64, device="cuda")
+                    for _ in range(world)]
+            for c in range(6):
+                send[combo_owner_rank(c, world)][c] = raw[c]
+            recv = [torch.empty(6, 10, dtype=torch.int64, device="cuda")
+                    for _ in range(world)]
+            dist.all_to_all(recv, send)
+            # final agg for owned groups (exact integer merge on host ints)
+            merged = merge_raw([[x & ((1 << 64) - 1) for x in t.flatten().tolist()]
+                                for t in recv])
+            return r, merged
+        return r, None
+
+    # warmup
+    for _ in range(args.warmup):
+        res, _ = step()
+    if dist:
+        dist.barrier()
+    torch.cuda.synchronize()
+
+    kernel_ms = []
+    t_start = time.time()
+    for _ in range(args.steps):
+        res, merged = step()
+        kernel_ms.append(res.elapsed_ms)
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    elapsed = time.time() - t_start
+    if dist:
+        el = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dist.all_reduce(el, op=dist.ReduceOp.MAX)
+        elapsed = float(el.item())
+
+    ms_per_step = elapsed * 1000.0 / args.steps
+    value = rows_total / (elapsed / args.steps)          # whole-job rows/s
+    k_ms = sum(kernel_ms) / len(kernel_ms)
+    achieved_gbps = rows_rank * ALG_BYTES_PER_ROW / (k_ms / 1000.0) / 1e9
+
+    # ---- CPU baseline (rank 0, N=1 only): the parity-pinned oracle timed on
+    # host cores over a bounded sample (SF1 = 6,001,215 rows, repeated) ----
+    cpu_baseline = None
+    if rank == 0 and n_gpus == 1 and not args.skip_cpu_baseline:
+        import oracle
+        cores = os.cpu_count() or 1
+        li = oracle.gen_lineitem(1.0)
+        n_s = len(li["shipdate"])
+        reps, spent = 0, 0.0
+        tcb = time.time()
+        while spent < 10.0 and reps < 200:
+            oracle.q1_exact(li, Q1_CUTOFF, threads=cores)
+            reps += 1
+            spent = time.time() - tcb
+        cpu_rows_s = n_s * reps / spent
+        cpu_baseline = {
+            "value": cpu_rows_s, "unit": "rows/s", "cores": cores, "kind": "port",
+            "sample": f"TPC-H SF1 lineitem ({n_s:,} rows) x{reps} passes, "
+                      f"{spent:.1f}s, OpenMP {cores} threads; generator excluded",
+        }
+        log(f"cpu_baseline: {cpu_rows_s/1e6:.0f} Mrow/s on {cores} cores")
+
+    if rank == 0:
+        out = {
+            "metric": "tpch_q1_scan_filter_agg_throughput",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,   # reference publishes no numbers
+            "dtype": "f64",
+            "data": "synthetic",
+            "config": {
+                "workload": f"TPC-H Q1 (lineitem scan+filter+agg) SF{args.sf_per_gpu:g}"
+                            f" per GPU, HBM-resident device-generated columns",
+                "scale_factor_total": total_sf,
+                "rows_total": rows_total,
+                "parallelism": f"dp{n_gpus}",
+                "cutoff": "shipdate <= 1998-09-02",
+            },
+            "roofline": {
+                "bound": "hbm",
+                "achieved": achieved_gbps,
+                "peak": HBM_PEAK_GBPS,
+                "unit": "GB/s",
+                "frac": achieved_gbps / HBM_PEAK_GBPS,
+                "traffic": None,   # PMC evidence: profiles/ (separate rocprofv3 --pmc runs)
+            },
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(out), flush=True)
+
+    sess.tpch_lineitem_free(cols)
+    sess.close()
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -228,6 +228,9 @@ typedef struct tg_q1_result {
     double sum_disc[6];
     int64_t count[6];
     double elapsed_ms;          /* kernel time, HIP events */
+    /* raw integer accumulators (per combo: base,dp,ch,disc as u128 lo/hi
+     * pairs then qty,cnt), for exact cross-GPU merging */
+    uint64_t raw[60];
 } tg_q1_result;
 
 tg_status tg_q1_run(tg_session*, const tg_tpch_lineitem_cols* cols,

@@ -1,0 +1,62 @@
+/* abi.cpp — C-ABI session/error surface of libtrino_gpu (include/trino_gpu.h).
+ * Host C++; no CPU compute fallback: session creation fails without a HIP
+ * device and every operator entry point requires a session.
+ */
+#include "common.h"
+
+thread_local std::string tg_error_buf;
+
+extern "C" const char* tg_last_error(void)
+{
+    return tg_error_buf.c_str();
+}
+
+extern "C" const char* tg_version(void)
+{
+    return "trino_amd 0.1 (gfx950; HIP " __DATE__ ")";
+}
+
+extern "C" tg_status tg_session_create(int device_ordinal, tg_session** out)
+{
+    if (!out) { TG_SET_ERR("null out"); return TG_ERR_INVALID_ARG; }
+    int count = 0;
+    hipError_t e = hipGetDeviceCount(&count);
+    if (e != hipSuccess || count == 0) {
+        TG_SET_ERR("no HIP device available (count=%d, %s) — trino_gpu has no CPU "
+                   "fallback by design", count, hipGetErrorName(e));
+        return TG_ERR_NO_GPU;
+    }
+    if (device_ordinal < 0 || device_ordinal >= count) {
+        TG_SET_ERR("device %d out of range [0,%d)", device_ordinal, count);
+        return TG_ERR_INVALID_ARG;
+    }
+    TG_HIP_CHECK(hipSetDevice(device_ordinal));
+    tg_session* s = new tg_session();
+    s->device = device_ordinal;
+    if (hipStreamCreate(&s->stream) != hipSuccess ||
+        hipEventCreate(&s->ev_start) != hipSuccess ||
+        hipEventCreate(&s->ev_stop) != hipSuccess) {
+        TG_SET_ERR("failed to create stream/events");
+        delete s;
+        return TG_ERR_HIP;
+    }
+    *out = s;
+    return TG_OK;
+}
+
+extern "C" void tg_session_close(tg_session* s)
+{
+    if (!s) return;
+    hipEventDestroy(s->ev_start);
+    hipEventDestroy(s->ev_stop);
+    hipStreamDestroy(s->stream);
+    delete s;
+}
+
+/* test helper: copy device memory to host (generator parity tests) */
+extern "C" tg_status tg_copy_dtoh(tg_session* s, void* dst, const void* src, int64_t bytes)
+{
+    TG_HIP_CHECK(hipMemcpyAsync(dst, src, (size_t)bytes, hipMemcpyDeviceToHost, s->stream));
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}
