@@ -1,0 +1,467 @@
+/* ops_filter.hip — columnar filter + projection over selection vectors.
+ *
+ * Mirrors:
+ *  - ColumnarFilter.filterPositionsRange/List (sql/gen/columnar/
+ *    ColumnarFilter.java:42-52): predicate -> int32 selection vector,
+ *    branch-free compaction (CallColumnarFilterGenerator.java:160-198:
+ *    outputPositions[count]=position; count += result) — here wave ballot +
+ *    prefix-sum + stable two-phase scatter (order-preserving);
+ *  - NULL semantics: any NULL argument rejects the row (nullable
+ *    specialization, CallColumnarFilterGenerator.java:160-198);
+ *  - PageProjection.project over SelectedPositions (project/PageProcessor.
+ *    java:274-305); per-query bytecode gen is replaced by a device postfix
+ *    interpreter over the tg_expr IR (stack depth <= 4, checked at compile).
+ */
+#include "operators.h"
+
+struct KCol { const void* data; const uint64_t* valid; int32_t type; int32_t _pad; };
+
+#define MAX_STACK 4
+
+__device__ static inline double load_col(const KCol& c, int64_t i, bool* isnull)
+{
+    if (c.valid && !((c.valid[i >> 6] >> (i & 63)) & 1)) { *isnull = true; return 0.0; }
+    switch (c.type) {
+        case TG_BIGINT: return (double)((const int64_t*)c.data)[i];
+        case TG_INTEGER: case TG_DATE: return (double)((const int32_t*)c.data)[i];
+        case TG_SMALLINT: return (double)((const int16_t*)c.data)[i];
+        case TG_TINYINT: case TG_BOOLEAN: return (double)((const int8_t*)c.data)[i];
+        default: return ((const double*)c.data)[i];
+    }
+}
+
+/* postfix interpreter; returns value, sets *rnull. Bool results are 0.0/1.0.
+ * Null comparison/arith poisons; AND/OR treat null as false (a NULL argument
+ * rejects the row for that term, OR still unions other terms). */
+__device__ static double eval_expr(const tg_expr_inst* prog, int count,
+                                   const KCol* cols, int64_t row, bool* rnull)
+{
+    double s0 = 0, s1 = 0, s2 = 0, s3 = 0;
+    bool n0 = false, n1 = false, n2 = false, n3 = false;
+    int sp = 0;
+#define PUSH(v, nn) do { switch (sp) { \
+        case 0: s0 = (v); n0 = (nn); break; case 1: s1 = (v); n1 = (nn); break; \
+        case 2: s2 = (v); n2 = (nn); break; default: s3 = (v); n3 = (nn); break; } \
+        sp++; } while (0)
+#define TOP_V (sp == 1 ? s0 : sp == 2 ? s1 : sp == 3 ? s2 : s3)
+#define TOP_N (sp == 1 ? n0 : sp == 2 ? n1 : sp == 3 ? n2 : n3)
+#define POP(v, nn) do { v = TOP_V; nn = TOP_N; sp--; } while (0)
+
+    for (int k = 0; k < count; k++) {
+        tg_expr_inst in = prog[k];
+        double a, b, c;
+        bool na, nb, nc;
+        switch (in.op) {
+            case TG_EXPR_COL: {
+                bool nl = false;
+                double v = load_col(cols[in.arg0], row, &nl);
+                PUSH(v, nl);
+                break;
+            }
+            case TG_EXPR_CONST_F64: PUSH(in.imm.f64, false); break;
+            case TG_EXPR_CONST_I64: PUSH((double)in.imm.i64, false); break;
+            case TG_EXPR_ADD: POP(b, nb); POP(a, na); PUSH(a + b, na | nb); break;
+            case TG_EXPR_SUB: POP(b, nb); POP(a, na); PUSH(a - b, na | nb); break;
+            case TG_EXPR_MUL: POP(b, nb); POP(a, na); PUSH(a * b, na | nb); break;
+            case TG_EXPR_DIV: POP(b, nb); POP(a, na); PUSH(a / b, na | nb); break;
+            case TG_EXPR_LE: POP(b, nb); POP(a, na); PUSH((a <= b) ? 1.0 : 0.0, na | nb); break;
+            case TG_EXPR_LT: POP(b, nb); POP(a, na); PUSH((a < b) ? 1.0 : 0.0, na | nb); break;
+            case TG_EXPR_GE: POP(b, nb); POP(a, na); PUSH((a >= b) ? 1.0 : 0.0, na | nb); break;
+            case TG_EXPR_GT: POP(b, nb); POP(a, na); PUSH((a > b) ? 1.0 : 0.0, na | nb); break;
+            case TG_EXPR_EQ: POP(b, nb); POP(a, na); PUSH((a == b) ? 1.0 : 0.0, na | nb); break;
+            case TG_EXPR_NE: POP(b, nb); POP(a, na); PUSH((a != b) ? 1.0 : 0.0, na | nb); break;
+            case TG_EXPR_AND: POP(b, nb); POP(a, na);
+                PUSH(((a != 0.0) & !na & (b != 0.0) & !nb) ? 1.0 : 0.0, false); break;
+            case TG_EXPR_OR: POP(b, nb); POP(a, na);
+                PUSH((((a != 0.0) & !na) | ((b != 0.0) & !nb)) ? 1.0 : 0.0, false); break;
+            case TG_EXPR_NOT: POP(a, na); PUSH((a == 0.0) ? 1.0 : 0.0, na); break;
+            case TG_EXPR_BETWEEN: POP(c, nc); POP(b, nb); POP(a, na);
+                PUSH((a >= b && a <= c) ? 1.0 : 0.0, na | nb | nc); break;
+            default: break;
+        }
+    }
+    *rnull = TOP_N;
+    return TOP_V;
+#undef PUSH
+#undef POP
+#undef TOP_V
+#undef TOP_N
+}
+
+/* row index for the k-th selected input position */
+__device__ static inline int64_t sel_row(int has_list, const int32_t* list,
+                                         int32_t offset, int64_t k)
+{
+    return has_list ? (int64_t)list[k] : offset + k;
+}
+
+__global__ void k_filter_flags(const tg_expr_inst* prog, int count, const KCol* cols,
+                               int has_list, const int32_t* list, int32_t offset,
+                               int64_t n, uint8_t* __restrict__ flags)
+{
+    int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; k < n; k += stride) {
+        bool isnull = false;
+        double v = eval_expr(prog, count, cols, sel_row(has_list, list, offset, k), &isnull);
+        flags[k] = (!isnull && v != 0.0) ? 1 : 0;
+    }
+}
+
+/* stable compaction: per-chunk counts -> scan -> scatter. Chunks are
+ * CHUNK-sized contiguous ranges so ordering is preserved. */
+#define CHUNK 16384
+
+__global__ void k_count_chunk(const uint8_t* __restrict__ flags, int64_t n,
+                              int32_t* __restrict__ chunk_counts, int64_t nchunks)
+{
+    int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= nchunks) return;
+    int64_t lo = c * CHUNK, hi = min(lo + CHUNK, n);
+    int32_t cnt = 0;
+    for (int64_t i = lo; i < hi; i++) cnt += flags[i];
+    chunk_counts[c] = cnt;
+}
+
+__global__ void k_scan_serial(int32_t* counts, int64_t n, int32_t* total)
+{
+    if (blockIdx.x || threadIdx.x) return;
+    int32_t run = 0;
+    for (int64_t i = 0; i < n; i++) {
+        int32_t c = counts[i];
+        counts[i] = run;
+        run += c;
+    }
+    *total = run;
+}
+
+__global__ void k_compact(const uint8_t* __restrict__ flags, int64_t n,
+                          const int32_t* __restrict__ chunk_offsets,
+                          int has_list, const int32_t* list, int32_t offset,
+                          int32_t* __restrict__ out_positions)
+{
+    /* one block per chunk; sequential within a wave via ballot prefix */
+    int64_t c = blockIdx.x;
+    int64_t lo = c * CHUNK, hi = min(lo + CHUNK, n);
+    __shared__ int32_t base;
+    if (threadIdx.x == 0) base = chunk_offsets[c];
+    __syncthreads();
+    /* waves process 64-element groups of the chunk in order */
+    int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+    int nwaves = blockDim.x / 64;
+    __shared__ int32_t wave_base[32];
+    for (int64_t g = lo + wave * 64; g < hi; g += (int64_t)nwaves * 64) {
+        int64_t i = g + lane;
+        bool f = (i < hi) && flags[i];
+        unsigned long long b = __ballot(f);
+        /* this group's start = base + flags before it: computed by a serial
+         * pass over groups via wave-ordered accumulation in LDS */
+        int my_before = __popcll(b & ((lane == 63) ? ~0ull : ((1ull << (lane + 1)) - 1))) - (f ? 1 : 0);
+        /* group-level offset: group index within chunk determines order */
+        int group_idx = (int)((g - lo) / 64);
+        if (lane == 0) wave_base[wave] = __popcll(b);
+        /* compute exclusive sum of full groups before this one: recompute by
+         * scanning flags of prior groups is O(n^2); instead two-phase below */
+        (void)group_idx;
+        (void)my_before;
+        break; /* replaced by simpler scheme below */
+    }
+    /* Simpler correct scheme: one WAVE per chunk walks it in 64-wide steps,
+     * carrying a running count (wave 0 only). CHUNK/64 = 256 steps. */
+    if (wave != 0) return;
+    int32_t run = base;
+    for (int64_t g = lo; g < hi; g += 64) {
+        int64_t i = g + lane;
+        bool f = (i < hi) && flags[i];
+        unsigned long long b = __ballot(f);
+        int before = __popcll(b & ((1ull << lane) - 1ull));
+        if (f) {
+            int64_t k = i;
+            out_positions[run + before] =
+                has_list ? list[k] : (int32_t)(offset + k);
+        }
+        run += __popcll(b);
+    }
+}
+
+tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page,
+                     const tg_selected* input_sel,
+                     int32_t** d_positions_out, int32_t* count_out)
+{
+    /* device column descriptors */
+    std::vector<KCol> cols(page.blocks.size());
+    for (size_t i = 0; i < page.blocks.size(); i++)
+        cols[i] = {page.blocks[i].data, page.blocks[i].valid, (int32_t)page.blocks[i].type, 0};
+    KCol* d_cols = nullptr;
+    TG_HIP_CHECK(hipMalloc(&d_cols, cols.size() * sizeof(KCol)));
+    TG_HIP_CHECK(hipMemcpyAsync(d_cols, cols.data(), cols.size() * sizeof(KCol),
+                                hipMemcpyHostToDevice, s->stream));
+
+    int has_list = input_sel && input_sel->is_list;
+    int32_t offset = input_sel ? input_sel->offset : 0;
+    int64_t n = input_sel ? input_sel->size : page.n;
+    const int32_t* d_list = nullptr;
+    int32_t* d_list_owned = nullptr;
+    if (has_list) {
+        TG_HIP_CHECK(hipMalloc(&d_list_owned, n * sizeof(int32_t)));
+        TG_HIP_CHECK(hipMemcpyAsync(d_list_owned, input_sel->positions, n * sizeof(int32_t),
+                                    hipMemcpyHostToDevice, s->stream));
+        d_list = d_list_owned;
+    }
+
+    uint8_t* d_flags = nullptr;
+    TG_HIP_CHECK(hipMalloc(&d_flags, n ? n : 1));
+    hipLaunchKernelGGL(k_filter_flags, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
+                       pred.d_insts, pred.count, d_cols, has_list, d_list, offset, n, d_flags);
+    TG_HIP_CHECK(hipGetLastError());
+
+    int64_t nchunks = (n + CHUNK - 1) / CHUNK;
+    if (nchunks < 1) nchunks = 1;
+    int32_t* d_offsets = nullptr;
+    int32_t* d_total = nullptr;
+    TG_HIP_CHECK(hipMalloc(&d_offsets, nchunks * sizeof(int32_t)));
+    TG_HIP_CHECK(hipMalloc(&d_total, sizeof(int32_t)));
+    hipLaunchKernelGGL(k_count_chunk, dim3(tg_grid_for(nchunks)), dim3(TG_BLOCK), 0, s->stream,
+                       d_flags, n, d_offsets, nchunks);
+    TG_HIP_CHECK(hipGetLastError());
+    hipLaunchKernelGGL(k_scan_serial, dim3(1), dim3(1), 0, s->stream, d_offsets, nchunks, d_total);
+    TG_HIP_CHECK(hipGetLastError());
+    int32_t total = 0;
+    TG_HIP_CHECK(hipMemcpyAsync(&total, d_total, 4, hipMemcpyDeviceToHost, s->stream));
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+
+    int32_t* d_pos = nullptr;
+    TG_HIP_CHECK(hipMalloc(&d_pos, (total ? total : 1) * sizeof(int32_t)));
+    hipLaunchKernelGGL(k_compact, dim3((int)nchunks), dim3(64), 0, s->stream,
+                       d_flags, n, d_offsets, has_list, d_list, offset, d_pos);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+
+    TG_HIP_CHECK(hipFree(d_flags));
+    TG_HIP_CHECK(hipFree(d_offsets));
+    TG_HIP_CHECK(hipFree(d_total));
+    TG_HIP_CHECK(hipFree(d_cols));
+    if (d_list_owned) TG_HIP_CHECK(hipFree(d_list_owned));
+    *d_positions_out = d_pos;
+    *count_out = total;
+    return TG_OK;
+}
+
+/* ---- projection ---- */
+__global__ void k_project(const tg_expr_inst* prog, int count, const KCol* cols,
+                          const int32_t* __restrict__ positions, int32_t n,
+                          double* __restrict__ out, uint64_t* __restrict__ out_valid)
+{
+    int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; k < n; k += stride) {
+        bool isnull = false;
+        double v = eval_expr(prog, count, cols, positions ? positions[k] : k, &isnull);
+        out[k] = isnull ? 0.0 : v;
+        if (out_valid && isnull)
+            atomicAnd((unsigned long long*)&out_valid[k >> 6], ~(1ull << (k & 63)));
+    }
+}
+
+template <typename T>
+__global__ void k_gather(const T* __restrict__ src, const uint64_t* __restrict__ src_valid,
+                         const int32_t* __restrict__ positions, int32_t n,
+                         T* __restrict__ out, uint64_t* __restrict__ out_valid)
+{
+    int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; k < n; k += stride) {
+        int64_t i = positions ? positions[k] : k;
+        out[k] = src[i];
+        if (out_valid && src_valid && !((src_valid[i >> 6] >> (i & 63)) & 1))
+            atomicAnd((unsigned long long*)&out_valid[k >> 6], ~(1ull << (k & 63)));
+    }
+}
+
+tg_status run_gather(tg_session* s, const DevBlock& src, const int32_t* d_positions,
+                     int32_t count, DevBlock* out)
+{
+    out->type = src.type;
+    out->n = count;
+    TG_HIP_CHECK(hipMalloc(&out->data, (int64_t)(count ? count : 1) * src.elem_size()));
+    if (src.valid) {
+        int64_t words = (count + 63) / 64;
+        TG_HIP_CHECK(hipMalloc(&out->valid, (words ? words : 1) * 8));
+        TG_HIP_CHECK(hipMemsetAsync(out->valid, 0xFF, words * 8, s->stream));
+    }
+    int grid = tg_grid_for(count);
+    switch (src.elem_size()) {
+        case 8: hipLaunchKernelGGL(k_gather<int64_t>, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                    (const int64_t*)src.data, src.valid, d_positions, count,
+                    (int64_t*)out->data, out->valid); break;
+        case 4: hipLaunchKernelGGL(k_gather<int32_t>, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                    (const int32_t*)src.data, src.valid, d_positions, count,
+                    (int32_t*)out->data, out->valid); break;
+        case 2: hipLaunchKernelGGL(k_gather<int16_t>, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                    (const int16_t*)src.data, src.valid, d_positions, count,
+                    (int16_t*)out->data, out->valid); break;
+        default: hipLaunchKernelGGL(k_gather<int8_t>, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                    (const int8_t*)src.data, src.valid, d_positions, count,
+                    (int8_t*)out->data, out->valid); break;
+    }
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}
+
+tg_status run_project(tg_session* s, const ExprProgram& proj, tg_type out_type,
+                      const DevPage& page, const int32_t* d_positions, int32_t count,
+                      DevBlock* out)
+{
+    /* identity projection = typed gather */
+    if (proj.count == 1 && proj.insts[0].op == TG_EXPR_COL)
+        return run_gather(s, page.blocks[proj.insts[0].arg0], d_positions, count, out);
+
+    std::vector<KCol> cols(page.blocks.size());
+    bool any_null = false;
+    for (size_t i = 0; i < page.blocks.size(); i++) {
+        cols[i] = {page.blocks[i].data, page.blocks[i].valid, (int32_t)page.blocks[i].type, 0};
+        any_null |= page.blocks[i].valid != nullptr;
+    }
+    KCol* d_cols = nullptr;
+    TG_HIP_CHECK(hipMalloc(&d_cols, cols.size() * sizeof(KCol)));
+    TG_HIP_CHECK(hipMemcpyAsync(d_cols, cols.data(), cols.size() * sizeof(KCol),
+                                hipMemcpyHostToDevice, s->stream));
+    out->type = TG_DOUBLE;
+    out->n = count;
+    TG_HIP_CHECK(hipMalloc(&out->data, (int64_t)(count ? count : 1) * 8));
+    if (any_null) {
+        int64_t words = (count + 63) / 64;
+        TG_HIP_CHECK(hipMalloc(&out->valid, (words ? words : 1) * 8));
+        TG_HIP_CHECK(hipMemsetAsync(out->valid, 0xFF, words * 8, s->stream));
+    }
+    hipLaunchKernelGGL(k_project, dim3(tg_grid_for(count)), dim3(TG_BLOCK), 0, s->stream,
+                       proj.d_insts, proj.count, d_cols, d_positions, count,
+                       (double*)out->data, out->valid);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    TG_HIP_CHECK(hipFree(d_cols));
+    (void)out_type;
+    return TG_OK;
+}
+
+/* ---- FilterAndProject operator ---- */
+struct FilterProjectOp : tg_operator {
+    ExprProgram filter{};
+    bool has_filter = false;
+    std::vector<ExprProgram> projections;
+    std::vector<tg_type> out_types;
+
+    tg_status add_input(const tg_page* page) override
+    {
+        DevPage in;
+        tg_status st = tg_upload_page(s, page, &in);
+        if (st != TG_OK) return st;
+        int32_t* d_pos = nullptr;
+        int32_t count = (int32_t)in.n;
+        if (has_filter) {
+            st = run_filter(s, filter, in, nullptr, &d_pos, &count);
+            if (st != TG_OK) { tg_free_page(s, &in); return st; }
+        }
+        DevPage outp;
+        outp.n = count;
+        outp.blocks.resize(projections.size());
+        for (size_t p = 0; p < projections.size(); p++) {
+            st = run_project(s, projections[p], out_types[p], in, d_pos, count,
+                             &outp.blocks[p]);
+            if (st != TG_OK) break;
+        }
+        if (d_pos) (void)hipFree(d_pos);
+        tg_free_page(s, &in);
+        if (st != TG_OK) return st;
+        stage_output(std::move(outp));
+        return TG_OK;
+    }
+
+    tg_status get_output(tg_page* out, int* finished) override
+    {
+        emit_staged(out, finished);
+        return TG_OK;
+    }
+
+    ~FilterProjectOp() override
+    {
+        tg_free_expr(&filter);
+        for (auto& p : projections) tg_free_expr(&p);
+        for (auto& p : out_pages_) tg_free_page(s, &p);
+    }
+};
+
+static tg_status check_expr_depth(const tg_expr* e)
+{
+    int sp = 0, maxsp = 0;
+    for (int i = 0; i < e->count; i++) {
+        switch (e->insts[i].op) {
+            case TG_EXPR_COL: case TG_EXPR_CONST_F64: case TG_EXPR_CONST_I64: sp++; break;
+            case TG_EXPR_NOT: break;
+            case TG_EXPR_BETWEEN: sp -= 2; break;
+            default: sp -= 1; break;
+        }
+        if (sp > maxsp) maxsp = sp;
+        if (sp < 1) { TG_SET_ERR("expr stack underflow at inst %d", i); return TG_ERR_INVALID_ARG; }
+    }
+    if (maxsp > MAX_STACK) { TG_SET_ERR("expr stack depth %d > %d", maxsp, MAX_STACK); return TG_ERR_UNSUPPORTED; }
+    if (sp != 1) { TG_SET_ERR("expr does not reduce to one value"); return TG_ERR_INVALID_ARG; }
+    return TG_OK;
+}
+
+extern "C" tg_status tg_filter_project_create(tg_session* s, const tg_expr* filter,
+    const tg_expr* projections, const int32_t* proj_out_types, int32_t n_proj,
+    tg_operator** out)
+{
+    if (!s || !out || (n_proj > 0 && !projections)) { TG_SET_ERR("null arg"); return TG_ERR_INVALID_ARG; }
+    auto* op = new FilterProjectOp();
+    op->s = s;
+    tg_status st = TG_OK;
+    if (filter) {
+        st = check_expr_depth(filter);
+        if (st == TG_OK) st = tg_compile_expr(s, filter, &op->filter);
+        op->has_filter = true;
+    }
+    for (int p = 0; st == TG_OK && p < n_proj; p++) {
+        tg_expr e{projections[p].insts, projections[p].count};
+        st = check_expr_depth(&e);
+        if (st != TG_OK) break;
+        op->projections.emplace_back();
+        st = tg_compile_expr(s, &e, &op->projections.back());
+        op->out_types.push_back(proj_out_types ? (tg_type)proj_out_types[p] : TG_DOUBLE);
+    }
+    if (st != TG_OK) { delete op; return st; }
+    *out = op;
+    return TG_OK;
+}
+
+extern "C" tg_status tg_filter_run(tg_session* s, const tg_expr* filter, const tg_page* page,
+                                   const tg_selected* input_sel,
+                                   int32_t* out_positions, int32_t* out_count)
+{
+    if (!s || !filter || !page || !out_positions || !out_count) {
+        TG_SET_ERR("null arg"); return TG_ERR_INVALID_ARG;
+    }
+    tg_status st = check_expr_depth(filter);
+    if (st != TG_OK) return st;
+    ExprProgram prog;
+    st = tg_compile_expr(s, filter, &prog);
+    if (st != TG_OK) return st;
+    DevPage in;
+    st = tg_upload_page(s, page, &in);
+    if (st != TG_OK) { tg_free_expr(&prog); return st; }
+    int32_t* d_pos = nullptr;
+    int32_t count = 0;
+    st = run_filter(s, prog, in, input_sel, &d_pos, &count);
+    if (st == TG_OK) {
+        TG_HIP_CHECK(hipMemcpyAsync(out_positions, d_pos, count * sizeof(int32_t),
+                                    hipMemcpyDeviceToHost, s->stream));
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        *out_count = count;
+    }
+    if (d_pos) (void)hipFree(d_pos);
+    tg_free_page(s, &in);
+    tg_free_expr(&prog);
+    return st;
+}
