@@ -212,11 +212,29 @@ typedef struct tg_tpch_lineitem_cols {
 } tg_tpch_lineitem_cols;
 
 /* Generate lineitem rows for orders [order_start, order_start+order_count)
- * (1-based dense order index, part of 1,500,000×SF orders). Buffers must be
- * pre-sized to >= 7*order_count rows; actual count returned. */
+ * (1-based dense order index, part of 1,500,000×SF orders). Buffers must hold
+ * the actual row count (tg_tpch_lineitem_rows); row_count is set on return. */
 tg_status tg_tpch_gen_lineitem(tg_session*, double scale_factor,
     int64_t order_start, int64_t order_count,
     tg_tpch_lineitem_cols* cols /* in: buffers, out: row_count */);
+tg_status tg_tpch_lineitem_rows(tg_session*, double scale_factor,
+    int64_t order_start, int64_t order_count, int64_t* row_count_out,
+    int64_t** dev_offsets_out /* optional; hipFree */);
+/* allocate device buffers of the exact size and generate (bench/tests) */
+tg_status tg_tpch_lineitem_alloc(tg_session*, double scale_factor,
+    int64_t order_start, int64_t order_count, int with_orderkey,
+    tg_tpch_lineitem_cols* out);
+tg_status tg_tpch_lineitem_free(tg_session*, tg_tpch_lineitem_cols*);
+tg_status tg_tpch_gen_orders(tg_session*, double scale_factor,
+    int64_t order_start, int64_t order_count,
+    int64_t* dev_orderkey, int64_t* dev_custkey, int32_t* dev_orderdate);
+tg_status tg_tpch_gen_customer(tg_session*, double scale_factor,
+    int64_t cust_start, int64_t cust_count,
+    int64_t* dev_custkey, uint8_t* dev_mktsegment);
+tg_status tg_q1_run_naive(tg_session*, const tg_tpch_lineitem_cols*,
+    int32_t shipdate_cutoff, tg_q1_result* out);  /* parity mode */
+/* test helper */
+tg_status tg_copy_dtoh(tg_session*, void* dst, const void* src, int64_t bytes);
 
 /* ---- fused TPC-H Q1 pipeline (the north-star benchmark kernel):
  * scan+filter(shipdate<=cutoff)+group(returnflag,linestatus)+7 aggregates,

@@ -38,9 +38,11 @@ EXPORT uint64_t o_bigint_hash(int64_t v)
 /* DoubleType hashCodeOperator */
 EXPORT uint64_t o_double_hash(double v)
 {
-    if (v == 0) v = 0;          /* normalizes -0.0 to +0.0; NaN bits pass through */
+    if (v == 0) v = 0;          /* normalizes -0.0 to +0.0 */
     uint64_t bits;
     memcpy(&bits, &v, 8);
+    /* Java doubleToLongBits canonicalizes every NaN to 0x7FF8000000000000 */
+    if (v != v) bits = 0x7FF8000000000000ULL;
     return o_bigint_hash((int64_t)bits);
 }
 

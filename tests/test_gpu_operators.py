@@ -1,0 +1,357 @@
+"""GPU parity tests for the operator-level C-ABI (filter/project, group-by
+aggregation, hash join, page partitioner, row hash) against the oracle.
+
+Tolerances (stated per test):
+ - hashes, selection vectors, group keys/ids, counts, integer sums, partition
+   assignment, join match sets: bit-exact;
+ - f64 SUM/AVG through the generic aggregation operator: relative 1e-12
+   (device atomicAdd order, like the reference's own cross-driver combine);
+   the flagship Q1 path is exact (test_gpu_q1.py).
+"""
+import numpy as np
+import pytest
+
+import oracle
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def sess():
+    import trino_amd
+    s = trino_amd.Session(0)
+    yield s
+    s.close()
+
+
+@pytest.fixture(scope="module")
+def ops():
+    from trino_amd import ops
+    return ops
+
+
+def rng(seed=0):
+    return np.random.default_rng(seed)
+
+
+class TestHashRows:
+    def test_bit_exact_multi_type(self, sess, ops):
+        r = rng(1)
+        n = 10000
+        c1 = r.integers(-2**60, 2**60, n).astype(np.int64)
+        c2 = r.standard_normal(n)
+        c2[::100] = -0.0
+        c3 = r.integers(-2**30, 2**30, n).astype(np.int32)
+        page = ops.page_from_numpy([c1, c2, c3])
+        got = ops.hash_rows(sess, page, [0, 1, 2])
+        exp = oracle.hash_rows([c1, c2, c3], [oracle.TG_BIGINT, oracle.TG_DOUBLE, oracle.TG_INTEGER])
+        assert np.array_equal(got, exp)
+
+
+class TestFilter:
+    def test_range_filter_selection_vector(self, sess, ops):
+        r = rng(2)
+        n = 50000
+        sd = r.integers(8000, 11000, n).astype(np.int32)
+        page = ops.page_from_numpy([sd])
+        e = ops.expr(("col", 0), ("i64", 10471), "le")
+        got = ops.filter_run(sess, e, page)
+        exp = np.nonzero(sd <= 10471)[0].astype(np.int32)
+        assert np.array_equal(got, exp)
+
+    def test_and_or_between(self, sess, ops):
+        r = rng(3)
+        n = 20000
+        a = r.uniform(0, 1, n)
+        b = r.integers(1, 51, n).astype(np.int64)
+        page = ops.page_from_numpy([a, b])
+        # Q6 shape: disc between .05 and .07 AND qty < 24
+        e = ops.expr(("col", 0), ("f64", 0.05), ("f64", 0.07), "between",
+                     ("col", 1), ("i64", 24), "lt", "and")
+        got = ops.filter_run(sess, e, page)
+        exp = np.nonzero((a >= 0.05) & (a <= 0.07) & (b < 24))[0].astype(np.int32)
+        assert np.array_equal(got, exp)
+
+    def test_null_rejects_row(self, sess, ops):
+        """any NULL argument rejects the row (nullable ColumnarFilter spec)"""
+        n = 256
+        a = np.arange(n, dtype=np.int64)
+        valid = np.full((n + 63) // 64, ~np.uint64(0), np.uint64)
+        valid[0] = np.uint64(0xFFFFFFFFFFFFFFF0)  # rows 0..3 null
+        page = ops.page_from_numpy([a], valids=[valid])
+        e = ops.expr(("col", 0), ("i64", 1000), "le")
+        got = ops.filter_run(sess, e, page)
+        assert np.array_equal(got, np.arange(4, n, dtype=np.int32))
+
+    def test_input_selection_list(self, sess, ops):
+        n = 1000
+        a = np.arange(n, dtype=np.int64)
+        page = ops.page_from_numpy([a])
+        sel_pos = np.arange(0, n, 3, dtype=np.int32)
+        sel = ops.TgSelected()
+        sel.is_list = 1
+        sel.offset = 0
+        sel.size = len(sel_pos)
+        sel.positions = sel_pos.ctypes.data
+        e = ops.expr(("col", 0), ("i64", 500), "lt")
+        got = ops.filter_run(sess, e, page, input_sel=sel)
+        exp = sel_pos[a[sel_pos] < 500]
+        assert np.array_equal(got, exp)
+
+    def test_filter_project_operator(self, sess, ops):
+        r = rng(4)
+        n = 30000
+        sd = r.integers(8000, 11000, n).astype(np.int32)
+        ep = (r.integers(90100, 209900, n) / 100.0)
+        di = r.integers(0, 11, n) / 100.0
+        page = ops.page_from_numpy([sd, ep, di])
+        f = ops.expr(("col", 0), ("i64", 10471), "le")
+        projs = [ops.expr(("col", 1)),                                  # identity
+                 ops.expr(("col", 1), ("f64", 1.0), ("col", 2), "sub", "mul")]  # ep*(1-di)
+        op = ops.filter_project(sess, f, projs)
+        op.add_input(page)
+        pages = op.drain()
+        op.close()
+        assert len(pages) == 1
+        mask = sd <= 10471
+        assert np.array_equal(pages[0][0]["values"], ep[mask])
+        exp = ep[mask] * (1.0 - di[mask])
+        assert np.array_equal(pages[0][1]["values"], exp)  # same IEEE ops -> bit-exact
+
+
+class TestHashAggregation:
+    def test_bigint_groupby_counts_exact(self, sess, ops):
+        r = rng(5)
+        n = 200000
+        keys = r.integers(0, 5000, n).astype(np.int64)
+        vals = r.integers(0, 1000, n).astype(np.int64)
+        page = ops.page_from_numpy([keys, vals])
+        op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                  [(ops.AGG_COUNT_STAR, -1), (ops.AGG_SUM_I64, 1)])
+        op.add_input(page)
+        out = op.drain()[0]
+        op.close()
+        gids, ng, by_gid, _ = oracle.bigint_groupby(keys)
+        # output groups in first-occurrence (reference id) order
+        assert np.array_equal(out[0]["values"], by_gid)
+        assert np.array_equal(out[1]["values"], oracle.grouped_count(gids, ng))
+        exp_sum = np.zeros(ng, np.int64)
+        np.add.at(exp_sum, gids, vals)
+        assert np.array_equal(out[2]["values"], exp_sum)
+
+    def test_f64_sum_avg_tolerance(self, sess, ops):
+        r = rng(6)
+        n = 100000
+        keys = r.integers(0, 37, n).astype(np.int64)
+        vals = r.uniform(900, 105000, n)
+        page = ops.page_from_numpy([keys, vals])
+        op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                  [(ops.AGG_SUM_F64, 1), (ops.AGG_AVG_F64, 1)])
+        op.add_input(page)
+        out = op.drain()[0]
+        op.close()
+        gids, ng, by_gid, _ = oracle.bigint_groupby(keys)
+        exp_sum = oracle.grouped_sum_f64(gids, vals, ng)
+        cnt = oracle.grouped_count(gids, ng)
+        np.testing.assert_allclose(out[1]["values"], exp_sum, rtol=1e-12)
+        np.testing.assert_allclose(out[2]["values"], exp_sum / cnt, rtol=1e-12)
+
+    def test_multi_channel_keys_and_multipage(self, sess, ops):
+        r = rng(7)
+        pages_np = []
+        for k in range(3):
+            n = 4096
+            k1 = r.integers(0, 9, n).astype(np.int8)
+            k2 = r.integers(0, 7, n).astype(np.int32)
+            v = r.integers(0, 100, n).astype(np.int64)
+            pages_np.append((k1, k2, v))
+        op = ops.hash_aggregation(sess, [0, 1], [ops.TG_TINYINT, ops.TG_INTEGER],
+                                  [(ops.AGG_COUNT_STAR, -1), (ops.AGG_SUM_I64, 2)])
+        for k1, k2, v in pages_np:
+            op.add_input(ops.page_from_numpy([k1, k2, v]))
+        out = op.drain()[0]
+        op.close()
+        allk1 = np.concatenate([p[0] for p in pages_np])
+        allk2 = np.concatenate([p[1] for p in pages_np])
+        allv = np.concatenate([p[2] for p in pages_np])
+        gids, ng, first = oracle.flat_groupby([allk1, allk2],
+                                              [oracle.TG_TINYINT, oracle.TG_INTEGER])
+        assert np.array_equal(out[0]["values"], allk1[first])
+        assert np.array_equal(out[1]["values"], allk2[first])
+        assert np.array_equal(out[2]["values"], oracle.grouped_count(gids, ng))
+        exp_sum = np.zeros(ng, np.int64)
+        np.add.at(exp_sum, gids, allv)
+        assert np.array_equal(out[3]["values"], exp_sum)
+
+    def test_null_key_group(self, sess, ops):
+        n = 128
+        keys = np.arange(n, dtype=np.int64) % 5
+        valid = np.full((n + 63) // 64, ~np.uint64(0), np.uint64)
+        valid[0] &= ~np.uint64(1 << 7)  # row 7 (key 2) null
+        valid[1] &= ~np.uint64(1 << 1)  # row 65 (key 0) null
+        page = ops.page_from_numpy([keys], valids=[valid])
+        op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                  [(ops.AGG_COUNT_STAR, -1)])
+        op.add_input(page)
+        out = op.drain()[0]
+        op.close()
+        vb = np.array([(valid[i // 64] >> np.uint64(i % 64)) & np.uint64(1) for i in range(n)], bool)
+        gids, ng, by_gid, nullg = oracle.bigint_groupby(keys, valid)
+        assert len(out[0]["values"]) == ng
+        # null group key emits NULL
+        keyvalid = out[0]["valid"]
+        assert keyvalid is not None
+        nullbit = (keyvalid[nullg // 64] >> np.uint64(nullg % 64)) & np.uint64(1)
+        assert nullbit == 0
+        assert np.array_equal(out[1]["values"], oracle.grouped_count(gids, ng))
+
+    def test_partial_final_roundtrip(self, sess, ops):
+        """PARTIAL avg state (count,sum) pair -> FINAL combine == SINGLE"""
+        r = rng(8)
+        n = 50000
+        keys = r.integers(0, 91, n).astype(np.int64)
+        vals = r.uniform(0, 100, n)
+        half = n // 2
+        partial_pages = []
+        for sl in (slice(0, half), slice(half, n)):
+            op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                      [(ops.AGG_COUNT_STAR, -1), (ops.AGG_AVG_F64, 1)],
+                                      step=ops.STEP_PARTIAL)
+            op.add_input(ops.page_from_numpy([keys[sl], vals[sl]]))
+            partial_pages.append(op.drain()[0])
+            op.close()
+        fin = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                   [(ops.AGG_COUNT_STAR, 1), (ops.AGG_AVG_F64, 2)],
+                                   step=ops.STEP_FINAL)
+        for p in partial_pages:
+            cols = [p[0]["values"], p[1]["values"], p[2]["values"], p[3]["values"]]
+            fin.add_input(ops.page_from_numpy(cols))
+        out = fin.drain()[0]
+        fin.close()
+        gids, ng, by_gid, _ = oracle.bigint_groupby(keys)
+        # key order: first occurrence across partials arrival order
+        cnt_exp = oracle.grouped_count(gids, ng)
+        got = dict(zip(out[0]["values"].tolist(), out[1]["values"].tolist()))
+        exp = dict(zip(by_gid.tolist(), cnt_exp.tolist()))
+        assert got == exp
+        sums = oracle.grouped_sum_f64(gids, vals, ng)
+        got_avg = dict(zip(out[0]["values"].tolist(), out[2]["values"].tolist()))
+        for k, g in zip(by_gid.tolist(), range(ng)):
+            assert abs(got_avg[k] - sums[g] / cnt_exp[g]) <= 1e-12 * abs(got_avg[k])
+
+
+class TestJoin:
+    def _run_join(self, sess, ops, bk, bvals, pk, pvals):
+        bridge = ops.JoinBridge(sess)
+        build = ops.hash_builder(sess, bridge,
+                                 [ops.TG_BIGINT, ops.TG_BIGINT], [0], [1])
+        build.add_input(ops.page_from_numpy([bk, bvals]))
+        build.drain()
+        build.close()
+        probe = ops.lookup_join(sess, bridge, [ops.TG_BIGINT, ops.TG_BIGINT],
+                                [0], [0, 1])
+        probe.add_input(ops.page_from_numpy([pk, pvals]))
+        pages = probe.drain()
+        probe.close()
+        bridge.close()
+        return pages
+
+    def test_unique_keys_exact(self, sess, ops):
+        """unique build keys: single match per probe row, reference order"""
+        r = rng(9)
+        nb, np_ = 5000, 20000
+        bk = np.arange(nb, dtype=np.int64) * 8 + 1   # sparse like orderkeys
+        bv = r.integers(0, 10**6, nb).astype(np.int64)
+        pk = r.choice(np.concatenate([bk, np.arange(10**7, 10**7 + 5000)]), np_).astype(np.int64)
+        pv = np.arange(np_, dtype=np.int64)
+        pages = self._run_join(sess, ops, bk, bv, pk, pv)
+        out = pages[0]
+        t = oracle.JoinTable(bk)
+        op_, ob_ = t.probe(pk)
+        assert np.array_equal(out[0]["values"], pk[op_])
+        assert np.array_equal(out[1]["values"], pv[op_])
+        assert np.array_equal(out[2]["values"], bv[ob_])
+
+    def test_duplicate_keys_match_set(self, sess, ops):
+        """duplicate build keys: match SET equals oracle (chain order within a
+        key is not deterministic on the wide build — DESIGN.md §6)"""
+        r = rng(10)
+        bk = r.integers(0, 50, 2000).astype(np.int64)
+        bv = np.arange(2000, dtype=np.int64)
+        pk = r.integers(0, 60, 500).astype(np.int64)
+        pv = np.arange(500, dtype=np.int64)
+        pages = self._run_join(sess, ops, bk, bv, pk, pv)
+        out = pages[0]
+        t = oracle.JoinTable(bk)
+        op_, ob_ = t.probe(pk, cap=10**6)
+        got = sorted(zip(out[1]["values"].tolist(), out[2]["values"].tolist()))
+        exp = sorted(zip(pv[op_].tolist(), bv[ob_].tolist()))
+        assert got == exp
+
+    def test_null_keys_never_match(self, sess, ops):
+        bk = np.array([1, 2, 3], np.int64)
+        bv = np.array([10, 20, 30], np.int64)
+        bvalid = np.array([~np.uint64(0)], np.uint64)
+        bvalid[0] &= ~np.uint64(2)  # build row 1 null
+        pk = np.array([2, 1], np.int64)
+        pv = np.array([0, 1], np.int64)
+        bridge = ops.JoinBridge(sess)
+        build = ops.hash_builder(sess, bridge, [ops.TG_BIGINT, ops.TG_BIGINT], [0], [1])
+        build.add_input(ops.page_from_numpy([bk, bv], valids=[bvalid, None]))
+        build.drain()
+        build.close()
+        probe = ops.lookup_join(sess, bridge, [ops.TG_BIGINT, ops.TG_BIGINT], [0], [0, 1])
+        probe.add_input(ops.page_from_numpy([pk, pv]))
+        out = probe.drain()[0]
+        probe.close()
+        bridge.close()
+        assert out[0]["values"].tolist() == [1]   # only key 1 matches
+        assert out[2]["values"].tolist() == [10]
+
+    def test_q3_join_sf001(self, sess, ops):
+        """orders x lineitem on orderkey at sf0.01 vs oracle probe"""
+        o = oracle.gen_orders(0.01, 1, 15000)
+        li = oracle.gen_lineitem(0.01)
+        bridge = ops.JoinBridge(sess)
+        build = ops.hash_builder(sess, bridge,
+                                 [ops.TG_BIGINT, ops.TG_INTEGER], [0], [1])
+        build.add_input(ops.page_from_numpy([o["orderkey"], o["orderdate"]]))
+        build.drain()
+        build.close()
+        probe = ops.lookup_join(sess, bridge,
+                                [ops.TG_BIGINT, ops.TG_DOUBLE], [0], [0, 1])
+        probe.add_input(ops.page_from_numpy([li["orderkey"], li["extendedprice"]]))
+        out = probe.drain()[0]
+        probe.close()
+        bridge.close()
+        t = oracle.JoinTable(o["orderkey"])
+        op_, ob_ = t.probe(li["orderkey"], cap=7_000_000)
+        assert len(out[0]["values"]) == len(op_) == len(li["orderkey"])  # every line matches
+        assert np.array_equal(out[0]["values"], li["orderkey"][op_])
+        assert np.array_equal(out[1]["values"], li["extendedprice"][op_])
+        assert np.array_equal(out[2]["values"], o["orderdate"][ob_])
+
+
+class TestPartitioner:
+    def test_partition_assignment_and_order(self, sess, ops):
+        r = rng(11)
+        n = 100000
+        keys = r.integers(-2**62, 2**62, n).astype(np.int64)
+        vals = r.standard_normal(n)
+        page = ops.page_from_numpy([keys, vals])
+        nparts = 8
+        op = ops.page_partitioner(sess, [ops.TG_BIGINT, ops.TG_DOUBLE], [0], nparts)
+        op.add_input(page)
+        op.finish()
+        hashes = oracle.hash_rows([keys], [oracle.TG_BIGINT])
+        exp_pid = np.array([oracle.partition_remote(int(np.int64(h)), nparts) for h in hashes],
+                           np.int32)
+        for p in range(nparts):
+            cols = ops.get_partition(sess, op, p)
+            rows = np.nonzero(exp_pid == p)[0]
+            if cols is None:
+                assert len(rows) == 0
+                continue
+            assert np.array_equal(cols[0]["values"], keys[rows])   # order preserved
+            assert np.array_equal(cols[1]["values"], vals[rows])
+        op.close()

@@ -1,0 +1,557 @@
+/* ops_groupby.hip — group-by hash + hash aggregation operator.
+ *
+ * Mirrors:
+ *  - BigintGroupByHash.java:191-300 / FlatHash.java:276-330 semantics at the
+ *    operator boundary: group ids dense, assigned by first occurrence in row
+ *    order (GroupByHash.java:121-128), null keys group together; output pages
+ *    emit groups in consecutive group-id order (InMemoryHashAggregationBuilder
+ *    buildResult). The device table assigns provisional ids with atomics and
+ *    a finish-time remap by first-occurrence row restores the reference's
+ *    exact id order deterministically.
+ *  - HashAggregationOperator / GroupedAggregator.processPage
+ *    (GroupedAggregator.java:77-101): COUNT, SUM(bigint/double), AVG(double)
+ *    with PARTIAL state (count,sum) and FINAL combine
+ *    (DoubleSumAggregation.java:37-45, DoubleAverageAggregations.java:38-63).
+ *
+ * Determinism/parity: group keys, ids, counts and integer
+ * sums are bit-exact. f64 sums use device atomicAdd (the reference's own
+ * cross-driver combine order is nondeterministic too); operator parity tests
+ * state their tolerance; the flagship Q1 path (q1.hip) is exact fixed-point.
+ *
+ * Table layout: open addressing, power-of-2, per slot an int32 state
+ * (-1 empty / -2 claimed / else gid) and a key store of canonical 64-bit
+ * words per group (+ null mask word). Key-store writes are device-scope
+ * atomic stores published by a release store of the gid; probes use relaxed
+ * atomic loads (values land at the coherence point before the gid does).
+ */
+#include "dev_hash.h"
+
+struct GTable {
+    int64_t capacity = 0, mask = 0;
+    int32_t* state = nullptr;        /* per slot: -1/-2/gid */
+    int64_t max_groups = 0;
+    uint64_t* keystore = nullptr;    /* [max_groups][n_words] */
+    int64_t* first_row = nullptr;    /* [max_groups] atomicMin of global row idx */
+    int32_t* counter = nullptr;      /* n_groups */
+    int32_t n_words = 0;             /* n_key_channels + 1 (null mask) */
+};
+
+__global__ void k_gt_init(int32_t* state, int64_t cap, int64_t* first_row, int64_t ngroups_cap)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t k = i; k < cap; k += stride) state[k] = -1;
+    for (int64_t k = i; k < ngroups_cap; k += stride) first_row[k] = INT64_MAX;
+}
+
+/* assign group ids for one page; row_base = global row index of row 0 */
+__global__ void k_gt_assign(GTable t, const KColH* cols, int n_ch,
+                            int64_t n, int64_t row_base, int32_t* __restrict__ gids)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        uint64_t h = row_hash(cols, n_ch, i);
+        /* canonical words + null mask for this row */
+        uint64_t w[8];
+        uint64_t nullmask = 0;
+        for (int c = 0; c < n_ch; c++) {
+            bool nl = kcol_is_null(cols[c], i);
+            w[c] = nl ? 0 : kcol_word(cols[c], i);
+            nullmask |= (uint64_t)nl << c;
+        }
+        w[n_ch] = nullmask;
+        int64_t slot = (int64_t)(d_murmur3_mix(h) & (uint64_t)t.mask);
+        int32_t gid = -1;
+        while (true) {
+            int32_t st = __hip_atomic_load(&t.state[slot], __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_AGENT);
+            if (st == -1) {
+                int32_t old = -1;
+                if (__hip_atomic_compare_exchange_strong(&t.state[slot], &old, -2,
+                        __ATOMIC_ACQ_REL, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
+                    gid = atomicAdd(t.counter, 1);
+                    for (int c = 0; c <= n_ch; c++)
+                        __hip_atomic_store(&t.keystore[(int64_t)gid * t.n_words + c], w[c],
+                                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                    __hip_atomic_store(&t.state[slot], gid, __ATOMIC_RELEASE,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+                    break;
+                }
+                st = old;
+            }
+            while (st == -2) {
+                __builtin_amdgcn_s_sleep(1);
+                st = __hip_atomic_load(&t.state[slot], __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_AGENT);
+            }
+            /* st >= 0: compare keys */
+            bool eq = true;
+            for (int c = 0; c <= n_ch && eq; c++) {
+                uint64_t kv = __hip_atomic_load(&t.keystore[(int64_t)st * t.n_words + c],
+                                                __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+                eq = (kv == w[c]);
+            }
+            if (eq) { gid = st; break; }
+            slot = (slot + 1) & t.mask;
+        }
+        gids[i] = gid;
+        atomicMin((unsigned long long*)&t.first_row[gid],
+                  (unsigned long long)(row_base + i));
+    }
+}
+
+/* rehash: reinsert existing groups into a larger table (keystore is stable;
+ * only the slot index array is rebuilt) */
+__global__ void k_gt_rehash(GTable t, int32_t n_groups, const KColH* meta, int n_ch)
+{
+    (void)meta;
+    int32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n_groups) return;
+    /* recompute the row hash from canonical words (types recorded in meta) */
+    int64_t h = 0;
+    uint64_t nullmask = t.keystore[(int64_t)g * t.n_words + n_ch];
+    for (int c = 0; c < n_ch; c++) {
+        uint64_t w = t.keystore[(int64_t)g * t.n_words + c];
+        uint64_t ch;
+        if ((nullmask >> c) & 1) ch = 0;
+        else ch = d_bigint_hash((int64_t)w);   /* canonical words hash as longs:
+                                                  f64 words are normalized bits,
+                                                  matching d_double_hash */
+        h = 31 * h + (int64_t)ch;
+    }
+    int64_t slot = (int64_t)(d_murmur3_mix((uint64_t)h) & (uint64_t)t.mask);
+    while (true) {
+        int32_t old = -1;
+        if (__hip_atomic_compare_exchange_strong(&t.state[slot], &old, g,
+                __ATOMIC_RELAXED, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT))
+            break;
+        slot = (slot + 1) & t.mask;
+    }
+}
+
+/* ---- aggregation update ---- */
+struct KAgg {
+    int32_t fn;           /* tg_agg_fn */
+    int32_t in_ch;        /* -1 for COUNT_STAR; AVG FINAL: count ch, sum = ch+1 */
+    double* sum;          /* f64 state (or null) */
+    long long* cnt;       /* i64 state (count / int sum) */
+};
+#define MAX_AGGS 12
+
+__global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
+                             const KColH* cols, const KAgg* aggs, int n_aggs,
+                             int step /* 0 partial/single input rows, 1 final combine */)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        int32_t g = gids[i];
+        for (int a = 0; a < n_aggs; a++) {
+            KAgg ag = aggs[a];
+            switch (ag.fn) {
+                case TG_AGG_COUNT_STAR:
+                    if (step == 0) atomicAdd((unsigned long long*)&ag.cnt[g], 1ull);
+                    else {
+                        long long v = ((const int64_t*)cols[ag.in_ch].data)[i];
+                        atomicAdd((unsigned long long*)&ag.cnt[g], (unsigned long long)v);
+                    }
+                    break;
+                case TG_AGG_COUNT_COL:
+                    if (step == 0) {
+                        if (!kcol_is_null(cols[ag.in_ch], i))
+                            atomicAdd((unsigned long long*)&ag.cnt[g], 1ull);
+                    }
+                    else {
+                        long long v = ((const int64_t*)cols[ag.in_ch].data)[i];
+                        atomicAdd((unsigned long long*)&ag.cnt[g], (unsigned long long)v);
+                    }
+                    break;
+                case TG_AGG_SUM_I64:
+                    if (!kcol_is_null(cols[ag.in_ch], i)) {
+                        long long v = ((const int64_t*)cols[ag.in_ch].data)[i];
+                        atomicAdd((unsigned long long*)&ag.cnt[g], (unsigned long long)v);
+                    }
+                    break;
+                case TG_AGG_SUM_F64:
+                    if (!kcol_is_null(cols[ag.in_ch], i)) {
+                        double v = ((const double*)cols[ag.in_ch].data)[i];
+                        atomicAdd(&ag.sum[g], v);
+                    }
+                    break;
+                case TG_AGG_AVG_F64:
+                    if (step == 0) {
+                        if (!kcol_is_null(cols[ag.in_ch], i)) {
+                            double v = ((const double*)cols[ag.in_ch].data)[i];
+                            atomicAdd((unsigned long long*)&ag.cnt[g], 1ull);
+                            atomicAdd(&ag.sum[g], v);
+                        }
+                    }
+                    else {   /* FINAL: channels (count, sum) */
+                        long long cv = ((const int64_t*)cols[ag.in_ch].data)[i];
+                        double sv = ((const double*)cols[ag.in_ch + 1].data)[i];
+                        atomicAdd((unsigned long long*)&ag.cnt[g], (unsigned long long)cv);
+                        atomicAdd(&ag.sum[g], sv);
+                    }
+                    break;
+            }
+        }
+    }
+}
+
+/* ---- output materialization (after remap) ---- */
+__global__ void k_emit_keys(const uint64_t* __restrict__ keystore, int n_words,
+                            const int32_t* __restrict__ old_by_new, int32_t n_groups,
+                            int ch, int32_t type, void* __restrict__ out,
+                            uint64_t* __restrict__ out_valid)
+{
+    int32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n_groups) return;
+    int32_t og = old_by_new[g];
+    uint64_t w = keystore[(int64_t)og * n_words + ch];
+    uint64_t nullmask = keystore[(int64_t)og * n_words + (n_words - 1)];
+    bool isnull = (nullmask >> ch) & 1;
+    switch (type) {
+        case TG_BIGINT: ((int64_t*)out)[g] = (int64_t)w; break;
+        case TG_INTEGER: case TG_DATE: ((int32_t*)out)[g] = (int32_t)w; break;
+        case TG_SMALLINT: ((int16_t*)out)[g] = (int16_t)w; break;
+        case TG_TINYINT: case TG_BOOLEAN: ((int8_t*)out)[g] = (int8_t)w; break;
+        default: ((double*)out)[g] = __longlong_as_double((long long)w); break;
+    }
+    if (out_valid && isnull)
+        atomicAnd((unsigned long long*)&out_valid[g >> 6], ~(1ull << (g & 63)));
+}
+
+__global__ void k_emit_f64(const double* __restrict__ state, const int32_t* old_by_new,
+                           int32_t n, double* __restrict__ out)
+{
+    int32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g < n) out[g] = state[old_by_new[g]];
+}
+
+__global__ void k_emit_i64(const long long* __restrict__ state, const int32_t* old_by_new,
+                           int32_t n, int64_t* __restrict__ out)
+{
+    int32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g < n) out[g] = state[old_by_new[g]];
+}
+
+__global__ void k_emit_avg(const double* __restrict__ sum, const long long* __restrict__ cnt,
+                           const int32_t* old_by_new, int32_t n, double* __restrict__ out,
+                           uint64_t* __restrict__ out_valid)
+{
+    int32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n) return;
+    int32_t og = old_by_new[g];
+    if (cnt[og] == 0) {   /* AVG of no rows = NULL */
+        out[g] = 0.0;
+        atomicAnd((unsigned long long*)&out_valid[g >> 6], ~(1ull << (g & 63)));
+    }
+    else {
+        out[g] = sum[og] / (double)cnt[og];
+    }
+}
+
+/* ---- operator ---- */
+#include <algorithm>
+#include <numeric>
+
+struct HashAggOp : tg_operator {
+    std::vector<int32_t> group_channels;
+    std::vector<tg_type> group_types;
+    std::vector<tg_agg_spec> aggs;
+    tg_agg_step step = TG_STEP_SINGLE;
+
+    GTable t;
+    std::vector<KAgg> agg_state;       /* host mirror with device ptrs */
+    int64_t rows_seen = 0;
+    bool emitted = false;
+
+    tg_status init_table(int64_t cap, int64_t max_groups)
+    {
+        t.capacity = cap; t.mask = cap - 1;
+        t.max_groups = max_groups;
+        t.n_words = (int32_t)group_channels.size() + 1;
+        TG_HIP_CHECK(hipMalloc(&t.state, cap * 4));
+        TG_HIP_CHECK(hipMalloc(&t.keystore, max_groups * t.n_words * 8));
+        TG_HIP_CHECK(hipMalloc(&t.first_row, max_groups * 8));
+        TG_HIP_CHECK(hipMalloc(&t.counter, 4));
+        TG_HIP_CHECK(hipMemsetAsync(t.counter, 0, 4, s->stream));
+        hipLaunchKernelGGL(k_gt_init, dim3(tg_grid_for(cap)), dim3(TG_BLOCK), 0, s->stream,
+                           t.state, cap, t.first_row, max_groups);
+        TG_HIP_CHECK(hipGetLastError());
+        for (auto& a : agg_state) {
+            if (a.sum) TG_HIP_CHECK(hipMemsetAsync(a.sum, 0, max_groups * 8, s->stream));
+            if (a.cnt) TG_HIP_CHECK(hipMemsetAsync(a.cnt, 0, max_groups * 8, s->stream));
+        }
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        return TG_OK;
+    }
+
+    int32_t n_groups_host()
+    {
+        int32_t n = 0;
+        (void)hipMemcpy(&n, t.counter, 4, hipMemcpyDeviceToHost);
+        return n;
+    }
+
+    tg_status grow_if_needed(int64_t incoming)
+    {
+        int32_t ng = n_groups_host();
+        if (ng + incoming <= (int64_t)(t.capacity * 0.7) &&
+            ng + incoming <= t.max_groups) return TG_OK;
+        int64_t need_groups = ng + incoming;
+        int64_t new_groups = t.max_groups;
+        while (new_groups < need_groups) new_groups *= 2;
+        int64_t new_cap = t.capacity;
+        while ((int64_t)(new_cap * 0.7) < need_groups) new_cap *= 2;
+
+        /* grow keystore/first_row/states preserving contents */
+        if (new_groups != t.max_groups) {
+            uint64_t* nk; int64_t* nf;
+            TG_HIP_CHECK(hipMalloc(&nk, new_groups * t.n_words * 8));
+            TG_HIP_CHECK(hipMalloc(&nf, new_groups * 8));
+            TG_HIP_CHECK(hipMemcpyAsync(nk, t.keystore, t.max_groups * t.n_words * 8,
+                                        hipMemcpyDeviceToDevice, s->stream));
+            TG_HIP_CHECK(hipMemcpyAsync(nf, t.first_row, t.max_groups * 8,
+                                        hipMemcpyDeviceToDevice, s->stream));
+            hipLaunchKernelGGL(k_gt_init, dim3(tg_grid_for(new_groups)), dim3(TG_BLOCK),
+                               0, s->stream, (int32_t*)nf /*unused dummy*/, 0,
+                               nf + t.max_groups, new_groups - t.max_groups);
+            TG_HIP_CHECK(hipGetLastError());
+            TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+            TG_HIP_CHECK(hipFree(t.keystore)); TG_HIP_CHECK(hipFree(t.first_row));
+            t.keystore = nk; t.first_row = nf;
+            for (auto& a : agg_state) {
+                if (a.sum) {
+                    double* ns;
+                    TG_HIP_CHECK(hipMalloc(&ns, new_groups * 8));
+                    TG_HIP_CHECK(hipMemsetAsync(ns, 0, new_groups * 8, s->stream));
+                    TG_HIP_CHECK(hipMemcpyAsync(ns, a.sum, t.max_groups * 8,
+                                                hipMemcpyDeviceToDevice, s->stream));
+                    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+                    TG_HIP_CHECK(hipFree(a.sum));
+                    a.sum = ns;
+                }
+                if (a.cnt) {
+                    long long* nc;
+                    TG_HIP_CHECK(hipMalloc(&nc, new_groups * 8));
+                    TG_HIP_CHECK(hipMemsetAsync(nc, 0, new_groups * 8, s->stream));
+                    TG_HIP_CHECK(hipMemcpyAsync(nc, a.cnt, t.max_groups * 8,
+                                                hipMemcpyDeviceToDevice, s->stream));
+                    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+                    TG_HIP_CHECK(hipFree(a.cnt));
+                    a.cnt = nc;
+                }
+            }
+            t.max_groups = new_groups;
+        }
+        if (new_cap != t.capacity) {
+            TG_HIP_CHECK(hipFree(t.state));
+            TG_HIP_CHECK(hipMalloc(&t.state, new_cap * 4));
+            t.capacity = new_cap; t.mask = new_cap - 1;
+            hipLaunchKernelGGL(k_gt_init, dim3(tg_grid_for(new_cap)), dim3(TG_BLOCK),
+                               0, s->stream, t.state, new_cap, t.first_row, 0);
+            TG_HIP_CHECK(hipGetLastError());
+            hipLaunchKernelGGL(k_gt_rehash, dim3((ng + TG_BLOCK - 1) / TG_BLOCK),
+                               dim3(TG_BLOCK), 0, s->stream, t, ng, nullptr,
+                               (int)group_channels.size());
+            TG_HIP_CHECK(hipGetLastError());
+            TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        }
+        return TG_OK;
+    }
+
+    tg_status add_input(const tg_page* page) override
+    {
+        DevPage in;
+        tg_status st = tg_upload_page(s, page, &in);
+        if (st != TG_OK) return st;
+        st = grow_if_needed(in.n);
+        if (st != TG_OK) { tg_free_page(s, &in); return st; }
+
+        KColH* d_keys = nullptr;
+        st = make_kcols(s, in, group_channels.data(), (int)group_channels.size(), &d_keys);
+        if (st != TG_OK) { tg_free_page(s, &in); return st; }
+        int32_t* d_gids = nullptr;
+        TG_HIP_CHECK(hipMalloc(&d_gids, in.n * 4));
+        hipLaunchKernelGGL(k_gt_assign, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK), 0, s->stream,
+                           t, d_keys, (int)group_channels.size(), in.n, rows_seen, d_gids);
+        TG_HIP_CHECK(hipGetLastError());
+
+        /* aggregation inputs: all channels */
+        KColH* d_all = nullptr;
+        st = make_kcols(s, in, nullptr, (int)in.blocks.size(), &d_all);
+        if (st == TG_OK && !aggs.empty()) {
+            KAgg* d_aggs = nullptr;
+            TG_HIP_CHECK(hipMalloc(&d_aggs, agg_state.size() * sizeof(KAgg)));
+            TG_HIP_CHECK(hipMemcpyAsync(d_aggs, agg_state.data(),
+                                        agg_state.size() * sizeof(KAgg),
+                                        hipMemcpyHostToDevice, s->stream));
+            hipLaunchKernelGGL(k_agg_update, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
+                               0, s->stream, d_gids, in.n, d_all, d_aggs,
+                               (int)agg_state.size(), step == TG_STEP_FINAL ? 1 : 0);
+            TG_HIP_CHECK(hipGetLastError());
+            TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+            TG_HIP_CHECK(hipFree(d_aggs));
+        }
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        rows_seen += in.n;
+        TG_HIP_CHECK(hipFree(d_gids));
+        TG_HIP_CHECK(hipFree(d_keys));
+        if (d_all) TG_HIP_CHECK(hipFree(d_all));
+        tg_free_page(s, &in);
+        return st;
+    }
+
+    tg_status emit() /* build the output page: groups in remapped id order */
+    {
+        int32_t ng = n_groups_host();
+        /* remap: order groups by first-occurrence row (== the reference's
+         * insertion-order ids, deterministically) */
+        std::vector<int64_t> first(ng);
+        if (ng) {
+            TG_HIP_CHECK(hipMemcpy(first.data(), t.first_row, (size_t)ng * 8,
+                                   hipMemcpyDeviceToHost));
+        }
+        std::vector<int32_t> old_by_new(ng);
+        std::iota(old_by_new.begin(), old_by_new.end(), 0);
+        std::sort(old_by_new.begin(), old_by_new.end(),
+                  [&](int32_t a, int32_t b) { return first[a] < first[b]; });
+        int32_t* d_obn = nullptr;
+        TG_HIP_CHECK(hipMalloc(&d_obn, (ng ? ng : 1) * 4));
+        if (ng) {
+            TG_HIP_CHECK(hipMemcpyAsync(d_obn, old_by_new.data(), (size_t)ng * 4,
+                                        hipMemcpyHostToDevice, s->stream));
+        }
+
+        DevPage outp;
+        outp.n = ng;
+        int grid = (ng + TG_BLOCK - 1) / TG_BLOCK;
+        if (grid < 1) grid = 1;
+        /* key channels */
+        for (size_t c = 0; c < group_channels.size(); c++) {
+            DevBlock b;
+            b.type = group_types[c];
+            b.n = ng;
+            TG_HIP_CHECK(hipMalloc(&b.data, (int64_t)(ng ? ng : 1) * b.elem_size()));
+            int64_t words = (ng + 63) / 64;
+            TG_HIP_CHECK(hipMalloc(&b.valid, (words ? words : 1) * 8));
+            TG_HIP_CHECK(hipMemsetAsync(b.valid, 0xFF, words * 8, s->stream));
+            hipLaunchKernelGGL(k_emit_keys, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                               t.keystore, t.n_words, d_obn, ng, (int)c,
+                               (int32_t)b.type, b.data, b.valid);
+            TG_HIP_CHECK(hipGetLastError());
+            outp.blocks.push_back(b);
+        }
+        /* aggregate outputs */
+        for (size_t a = 0; a < agg_state.size(); a++) {
+            const KAgg& ag = agg_state[a];
+            bool final_out = (step != TG_STEP_PARTIAL);
+            if (ag.fn == TG_AGG_AVG_F64 && !final_out) {
+                /* PARTIAL avg state: (count BIGINT, sum DOUBLE) channel pair */
+                DevBlock bc; bc.type = TG_BIGINT; bc.n = ng;
+                TG_HIP_CHECK(hipMalloc(&bc.data, (int64_t)(ng ? ng : 1) * 8));
+                hipLaunchKernelGGL(k_emit_i64, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                                   ag.cnt, d_obn, ng, (int64_t*)bc.data);
+                outp.blocks.push_back(bc);
+                DevBlock bs; bs.type = TG_DOUBLE; bs.n = ng;
+                TG_HIP_CHECK(hipMalloc(&bs.data, (int64_t)(ng ? ng : 1) * 8));
+                hipLaunchKernelGGL(k_emit_f64, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                                   ag.sum, d_obn, ng, (double*)bs.data);
+                outp.blocks.push_back(bs);
+            }
+            else if (ag.fn == TG_AGG_AVG_F64) {
+                DevBlock b; b.type = TG_DOUBLE; b.n = ng;
+                TG_HIP_CHECK(hipMalloc(&b.data, (int64_t)(ng ? ng : 1) * 8));
+                int64_t words = (ng + 63) / 64;
+                TG_HIP_CHECK(hipMalloc(&b.valid, (words ? words : 1) * 8));
+                TG_HIP_CHECK(hipMemsetAsync(b.valid, 0xFF, words * 8, s->stream));
+                hipLaunchKernelGGL(k_emit_avg, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                                   ag.sum, ag.cnt, d_obn, ng, (double*)b.data, b.valid);
+                outp.blocks.push_back(b);
+            }
+            else if (ag.fn == TG_AGG_SUM_F64) {
+                DevBlock b; b.type = TG_DOUBLE; b.n = ng;
+                TG_HIP_CHECK(hipMalloc(&b.data, (int64_t)(ng ? ng : 1) * 8));
+                hipLaunchKernelGGL(k_emit_f64, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                                   ag.sum, d_obn, ng, (double*)b.data);
+                outp.blocks.push_back(b);
+            }
+            else {
+                DevBlock b; b.type = TG_BIGINT; b.n = ng;
+                TG_HIP_CHECK(hipMalloc(&b.data, (int64_t)(ng ? ng : 1) * 8));
+                hipLaunchKernelGGL(k_emit_i64, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                                   ag.cnt, d_obn, ng, (int64_t*)b.data);
+                outp.blocks.push_back(b);
+            }
+            TG_HIP_CHECK(hipGetLastError());
+        }
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        TG_HIP_CHECK(hipFree(d_obn));
+        stage_output(std::move(outp));
+        return TG_OK;
+    }
+
+    tg_status get_output(tg_page* out, int* finished) override
+    {
+        if (input_finished && !emitted) {
+            emitted = true;
+            tg_status st = emit();
+            if (st != TG_OK) return st;
+        }
+        emit_staged(out, finished);
+        return TG_OK;
+    }
+
+    ~HashAggOp() override
+    {
+        if (t.state) (void)hipFree(t.state);
+        if (t.keystore) (void)hipFree(t.keystore);
+        if (t.first_row) (void)hipFree(t.first_row);
+        if (t.counter) (void)hipFree(t.counter);
+        for (auto& a : agg_state) {
+            if (a.sum) (void)hipFree(a.sum);
+            if (a.cnt) (void)hipFree(a.cnt);
+        }
+        for (auto& p : out_pages_) tg_free_page(s, &p);
+    }
+};
+
+extern "C" tg_status tg_hash_aggregation_create(tg_session* s,
+    const int32_t* group_channels, int32_t n_group_channels,
+    const int32_t* group_types, const tg_agg_spec* aggs, int32_t n_aggs,
+    int32_t step, tg_operator** out)
+{
+    if (!s || !out || !group_channels || !group_types || n_group_channels < 1 ||
+        n_group_channels > 7 || n_aggs > MAX_AGGS) {
+        TG_SET_ERR("invalid hash aggregation spec (1..7 group channels, <=%d aggs)", MAX_AGGS);
+        return TG_ERR_INVALID_ARG;
+    }
+    auto* op = new HashAggOp();
+    op->s = s;
+    op->group_channels.assign(group_channels, group_channels + n_group_channels);
+    for (int i = 0; i < n_group_channels; i++)
+        op->group_types.push_back((tg_type)group_types[i]);
+    op->step = (tg_agg_step)step;
+    for (int a = 0; a < n_aggs; a++) {
+        op->aggs.push_back(aggs[a]);
+        KAgg k{};
+        k.fn = aggs[a].fn;
+        k.in_ch = aggs[a].input_channel;
+        bool needs_sum = (k.fn == TG_AGG_SUM_F64 || k.fn == TG_AGG_AVG_F64);
+        bool needs_cnt = (k.fn != TG_AGG_SUM_F64);
+        int64_t mg = 1 << 16;
+        if (needs_sum) {
+            if (hipMalloc(&k.sum, mg * 8) != hipSuccess) { delete op; return TG_ERR_OOM; }
+        }
+        if (needs_cnt) {
+            if (hipMalloc(&k.cnt, mg * 8) != hipSuccess) { delete op; return TG_ERR_OOM; }
+        }
+        op->agg_state.push_back(k);
+    }
+    tg_status st = op->init_table(1 << 17, 1 << 16);
+    if (st != TG_OK) { delete op; return st; }
+    *out = op;
+    return TG_OK;
+}

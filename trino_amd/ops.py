@@ -1,0 +1,335 @@
+"""trino_amd.ops — ctypes bindings for the operator-level C-ABI
+(include/trino_gpu.h: Operator/Page/Block/SelectedPositions surfaces).
+Product plumbing: builds descriptors from numpy arrays, drives the
+addInput/getOutput state machine, downloads device output pages.
+"""
+import ctypes
+
+import numpy as np
+
+from . import _lib, _check, Session
+
+TG_BIGINT, TG_INTEGER, TG_SMALLINT, TG_TINYINT, TG_DOUBLE, TG_DATE, TG_BOOLEAN, TG_VARCHAR = range(8)
+STEP_PARTIAL, STEP_FINAL, STEP_SINGLE = 0, 1, 2
+AGG_COUNT_STAR, AGG_COUNT_COL, AGG_SUM_F64, AGG_SUM_I64, AGG_AVG_F64 = range(5)
+
+_NP2TG = {np.dtype(np.int64): TG_BIGINT, np.dtype(np.int32): TG_INTEGER,
+          np.dtype(np.int16): TG_SMALLINT, np.dtype(np.int8): TG_TINYINT,
+          np.dtype(np.uint8): TG_TINYINT, np.dtype(np.float64): TG_DOUBLE}
+_TG2NP = {TG_BIGINT: np.int64, TG_INTEGER: np.int32, TG_SMALLINT: np.int16,
+          TG_TINYINT: np.int8, TG_DOUBLE: np.float64, TG_DATE: np.int32,
+          TG_BOOLEAN: np.int8}
+
+
+class TgBlock(ctypes.Structure):
+    pass
+
+
+TgBlock._fields_ = [
+    ("type", ctypes.c_int32), ("kind", ctypes.c_int32),
+    ("position_count", ctypes.c_int64),
+    ("on_device", ctypes.c_int32), ("_pad", ctypes.c_int32),
+    ("data", ctypes.c_void_p), ("valid", ctypes.c_void_p),
+    ("offsets", ctypes.c_void_p), ("ids", ctypes.c_void_p),
+    ("dictionary", ctypes.POINTER(TgBlock)),
+]
+
+
+class TgPage(ctypes.Structure):
+    _fields_ = [("channel_count", ctypes.c_int32),
+                ("position_count", ctypes.c_int64),
+                ("blocks", ctypes.POINTER(TgBlock))]
+
+
+class TgSelected(ctypes.Structure):
+    _fields_ = [("is_list", ctypes.c_int32), ("offset", ctypes.c_int32),
+                ("size", ctypes.c_int32), ("positions", ctypes.c_void_p)]
+
+
+class TgExprInst(ctypes.Structure):
+    class _Imm(ctypes.Union):
+        _fields_ = [("f64", ctypes.c_double), ("i64", ctypes.c_int64)]
+    _fields_ = [("op", ctypes.c_int32), ("arg0", ctypes.c_int32), ("imm", _Imm)]
+
+
+class TgExpr(ctypes.Structure):
+    _fields_ = [("insts", ctypes.POINTER(TgExprInst)), ("count", ctypes.c_int32)]
+
+
+class TgAggSpec(ctypes.Structure):
+    _fields_ = [("fn", ctypes.c_int32), ("input_channel", ctypes.c_int32)]
+
+
+(OP_COL, OP_CONST_F64, OP_CONST_I64, OP_ADD, OP_SUB, OP_MUL, OP_DIV,
+ OP_LE, OP_LT, OP_GE, OP_GT, OP_EQ, OP_NE, OP_AND, OP_OR, OP_NOT,
+ OP_BETWEEN, OP_IN) = range(18)
+
+
+def expr(*postfix):
+    """Build a TgExpr from postfix tuples: ('col', i) ('f64', v) ('i64', v)
+    or op names 'le','mul','and',..."""
+    ops = {"add": OP_ADD, "sub": OP_SUB, "mul": OP_MUL, "div": OP_DIV,
+           "le": OP_LE, "lt": OP_LT, "ge": OP_GE, "gt": OP_GT,
+           "eq": OP_EQ, "ne": OP_NE, "and": OP_AND, "or": OP_OR,
+           "not": OP_NOT, "between": OP_BETWEEN}
+    insts = (TgExprInst * len(postfix))()
+    for i, it in enumerate(postfix):
+        if isinstance(it, tuple):
+            kind, v = it
+            if kind == "col":
+                insts[i].op = OP_COL
+                insts[i].arg0 = v
+            elif kind == "f64":
+                insts[i].op = OP_CONST_F64
+                insts[i].imm.f64 = float(v)
+            elif kind == "i64":
+                insts[i].op = OP_CONST_I64
+                insts[i].imm.i64 = int(v)
+            else:
+                raise ValueError(kind)
+        else:
+            insts[i].op = ops[it]
+    e = TgExpr()
+    e.insts = insts
+    e.count = len(postfix)
+    e._keepalive = insts
+    return e
+
+
+def page_from_numpy(columns, valids=None):
+    """columns: list of 1-D numpy arrays (host). valids: optional list of
+    packed uint64 bitmaps (bit=1 valid) or None per column."""
+    n = len(columns[0])
+    blocks = (TgBlock * len(columns))()
+    keep = []
+    for i, col in enumerate(columns):
+        assert col.flags["C_CONTIGUOUS"]
+        blocks[i].type = _NP2TG[col.dtype]
+        blocks[i].kind = 0
+        blocks[i].position_count = n
+        blocks[i].on_device = 0
+        blocks[i].data = col.ctypes.data
+        v = valids[i] if valids else None
+        blocks[i].valid = v.ctypes.data if v is not None else None
+        keep.append((col, v))
+    p = TgPage()
+    p.channel_count = len(columns)
+    p.position_count = n
+    p.blocks = blocks
+    p._keepalive = (blocks, keep)
+    return p
+
+
+_lib.tg_operator_needs_input.restype = ctypes.c_int
+_lib.tg_operator_needs_input.argtypes = [ctypes.c_void_p]
+_lib.tg_operator_add_input.restype = ctypes.c_int
+_lib.tg_operator_add_input.argtypes = [ctypes.c_void_p, ctypes.POINTER(TgPage)]
+_lib.tg_operator_get_output.restype = ctypes.c_int
+_lib.tg_operator_get_output.argtypes = [ctypes.c_void_p, ctypes.POINTER(TgPage),
+                                        ctypes.POINTER(ctypes.c_int)]
+_lib.tg_operator_finish.restype = ctypes.c_int
+_lib.tg_operator_finish.argtypes = [ctypes.c_void_p]
+_lib.tg_operator_close.restype = None
+_lib.tg_operator_close.argtypes = [ctypes.c_void_p]
+_lib.tg_filter_project_create.restype = ctypes.c_int
+_lib.tg_filter_project_create.argtypes = [ctypes.c_void_p, ctypes.POINTER(TgExpr),
+                                          ctypes.POINTER(TgExpr), ctypes.c_void_p,
+                                          ctypes.c_int32, ctypes.c_void_p]
+_lib.tg_filter_run.restype = ctypes.c_int
+_lib.tg_filter_run.argtypes = [ctypes.c_void_p, ctypes.POINTER(TgExpr),
+                               ctypes.POINTER(TgPage), ctypes.POINTER(TgSelected),
+                               ctypes.c_void_p, ctypes.c_void_p]
+_lib.tg_hash_aggregation_create.restype = ctypes.c_int
+_lib.tg_hash_aggregation_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                            ctypes.c_int32, ctypes.c_void_p,
+                                            ctypes.c_void_p, ctypes.c_int32,
+                                            ctypes.c_int32, ctypes.c_void_p]
+_lib.tg_join_bridge_create.restype = ctypes.c_int
+_lib.tg_join_bridge_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+_lib.tg_join_bridge_close.restype = None
+_lib.tg_join_bridge_close.argtypes = [ctypes.c_void_p]
+_lib.tg_hash_builder_create.restype = ctypes.c_int
+_lib.tg_hash_builder_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p] + [ctypes.c_void_p, ctypes.c_int32] * 3 + [ctypes.c_void_p]
+_lib.tg_lookup_join_create.restype = ctypes.c_int
+_lib.tg_lookup_join_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p] + [ctypes.c_void_p, ctypes.c_int32] * 3 + [ctypes.c_void_p]
+_lib.tg_page_partitioner_create.restype = ctypes.c_int
+_lib.tg_page_partitioner_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                            ctypes.c_int32, ctypes.c_void_p,
+                                            ctypes.c_int32, ctypes.c_int32, ctypes.c_void_p]
+_lib.tg_page_partitioner_get_partition.restype = ctypes.c_int
+_lib.tg_page_partitioner_get_partition.argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                                   ctypes.POINTER(TgPage)]
+_lib.tg_hash_rows.restype = ctypes.c_int
+_lib.tg_hash_rows.argtypes = [ctypes.c_void_p, ctypes.POINTER(TgPage),
+                              ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p]
+
+
+def _i32arr(lst):
+    return np.array(lst, np.int32)
+
+
+class Operator:
+    """operator/Operator.java:18-50 state machine over the C ABI."""
+
+    def __init__(self, session, handle):
+        self.s = session
+        self._h = handle
+
+    def add_input(self, page):
+        _check(_lib.tg_operator_add_input(self._h, ctypes.byref(page)))
+
+    def finish(self):
+        _check(_lib.tg_operator_finish(self._h))
+
+    def get_output(self):
+        """Returns (columns_as_numpy_or_None, finished). Device blocks are
+        downloaded to host numpy for inspection."""
+        out = TgPage()
+        fin = ctypes.c_int(0)
+        _check(_lib.tg_operator_get_output(self._h, ctypes.byref(out), ctypes.byref(fin)))
+        if out.channel_count == 0:
+            return None, bool(fin.value)
+        return _download_page(self.s, out), bool(fin.value)
+
+    def drain(self):
+        """finish + collect all output pages (list of column dicts)."""
+        self.finish()
+        pages = []
+        while True:
+            cols, fin = self.get_output()
+            if cols is not None:
+                pages.append(cols)
+            else:
+                if fin:
+                    return pages
+
+    def close(self):
+        if self._h:
+            _lib.tg_operator_close(self._h)
+            self._h = None
+
+
+def _download_page(session, page):
+    from . import copy_dtoh
+    cols = []
+    for c in range(page.channel_count):
+        b = page.blocks[c]
+        n = b.position_count
+        arr = np.empty(n, _TG2NP[b.type])
+        if n:
+            copy_dtoh(session, arr, b.data)
+        valid = None
+        if b.valid:
+            valid = np.empty((n + 63) // 64, np.uint64)
+            if n:
+                copy_dtoh(session, valid, b.valid)
+        cols.append({"values": arr, "valid": valid, "type": b.type})
+    return cols
+
+
+def filter_project(session, filter_expr, projections, out_types=None):
+    h = ctypes.c_void_p()
+    n = len(projections)
+    proj_arr = (TgExpr * n)()
+    for i, p in enumerate(projections):
+        proj_arr[i] = p
+    ot = _i32arr(out_types or [TG_DOUBLE] * n)
+    _check(_lib.tg_filter_project_create(
+        session._h, ctypes.byref(filter_expr) if filter_expr else None,
+        proj_arr, ot.ctypes.data, n, ctypes.byref(h)))
+    op = Operator(session, h)
+    op._keep = (proj_arr, ot, filter_expr, projections)
+    return op
+
+
+def filter_run(session, filter_expr, page, input_sel=None):
+    cap = int(page.position_count if input_sel is None else input_sel.size)
+    out = np.empty(max(cap, 1), np.int32)
+    cnt = ctypes.c_int32(0)
+    _check(_lib.tg_filter_run(session._h, ctypes.byref(filter_expr), ctypes.byref(page),
+                              ctypes.byref(input_sel) if input_sel else None,
+                              out.ctypes.data, ctypes.byref(cnt)))
+    return out[:cnt.value]
+
+
+def hash_aggregation(session, group_channels, group_types, aggs, step=STEP_SINGLE):
+    """aggs: list of (fn, input_channel)."""
+    h = ctypes.c_void_p()
+    gc = _i32arr(group_channels)
+    gt = _i32arr(group_types)
+    sp = (TgAggSpec * len(aggs))()
+    for i, (fn, ch) in enumerate(aggs):
+        sp[i].fn = fn
+        sp[i].input_channel = ch
+    _check(_lib.tg_hash_aggregation_create(session._h, gc.ctypes.data, len(gc),
+                                           gt.ctypes.data, sp, len(aggs), step,
+                                           ctypes.byref(h)))
+    op = Operator(session, h)
+    op._keep = (gc, gt, sp)
+    return op
+
+
+class JoinBridge:
+    def __init__(self, session):
+        self.s = session
+        self._h = ctypes.c_void_p()
+        _check(_lib.tg_join_bridge_create(session._h, ctypes.byref(self._h)))
+
+    def close(self):
+        if self._h:
+            _lib.tg_join_bridge_close(self._h)
+            self._h = None
+
+
+def hash_builder(session, bridge, build_types, key_channels, output_channels):
+    h = ctypes.c_void_p()
+    bt = _i32arr(build_types)
+    kc = _i32arr(key_channels)
+    oc = _i32arr(output_channels)
+    _check(_lib.tg_hash_builder_create(session._h, bridge._h, bt.ctypes.data, len(bt),
+                                       kc.ctypes.data, len(kc), oc.ctypes.data, len(oc),
+                                       ctypes.byref(h)))
+    op = Operator(session, h)
+    op._keep = (bt, kc, oc)
+    return op
+
+
+def lookup_join(session, bridge, probe_types, key_channels, probe_output_channels):
+    h = ctypes.c_void_p()
+    pt = _i32arr(probe_types)
+    kc = _i32arr(key_channels)
+    oc = _i32arr(probe_output_channels)
+    _check(_lib.tg_lookup_join_create(session._h, bridge._h, pt.ctypes.data, len(pt),
+                                      kc.ctypes.data, len(kc), oc.ctypes.data, len(oc),
+                                      ctypes.byref(h)))
+    op = Operator(session, h)
+    op._keep = (pt, kc, oc)
+    return op
+
+
+def page_partitioner(session, types, partition_channels, partition_count):
+    h = ctypes.c_void_p()
+    ty = _i32arr(types)
+    pc = _i32arr(partition_channels)
+    _check(_lib.tg_page_partitioner_create(session._h, ty.ctypes.data, len(ty),
+                                           pc.ctypes.data, len(pc), partition_count,
+                                           ctypes.byref(h)))
+    op = Operator(session, h)
+    op._keep = (ty, pc)
+    return op
+
+
+def get_partition(session, partitioner_op, p):
+    out = TgPage()
+    _check(_lib.tg_page_partitioner_get_partition(partitioner_op._h, p, ctypes.byref(out)))
+    if out.channel_count == 0:
+        return None
+    return _download_page(session, out)
+
+
+def hash_rows(session, page, channels):
+    out = np.empty(max(int(page.position_count), 1), np.uint64)
+    ch = _i32arr(channels)
+    _check(_lib.tg_hash_rows(session._h, ctypes.byref(page), ch.ctypes.data,
+                             len(ch), out.ctypes.data))
+    return out[:page.position_count]
