@@ -44,6 +44,30 @@ def merge_raw(raws):
     return total
 
 
+def exchange_partials(tdist, world, rank, raw_u64, device):
+    """Exchange leg of BASELINE config 4: route each group's partial state to
+    its owner rank (PagePartitioner-equivalent assignment over the group key,
+    see combo_owner_rank) with an all_to_all over RCCL; on gloo (CPU tests,
+    which lacks alltoall) the same routing runs over all_gather + local
+    selection — identical results, different transport. Returns the exact
+    integer merge of the states this rank owns (python big-ints)."""
+    import torch
+    signed = [x - (1 << 64) if x >= (1 << 63) else x for x in raw_u64]
+    raw = torch.tensor(signed, dtype=torch.int64, device=device).view(6, 10)
+    send = [torch.zeros(6, 10, dtype=torch.int64, device=device) for _ in range(world)]
+    for c in range(6):
+        send[combo_owner_rank(c, world)][c] = raw[c]
+    if tdist.get_backend() == "nccl":
+        recv = [torch.empty(6, 10, dtype=torch.int64, device=device) for _ in range(world)]
+        tdist.all_to_all(recv, send)
+    else:
+        stacked = torch.stack(send)                       # [world][6][10]
+        allv = [torch.empty_like(stacked) for _ in range(world)]
+        tdist.all_gather(allv, stacked)
+        recv = [allv[r][rank] for r in range(world)]
+    return merge_raw([[x & ((1 << 64) - 1) for x in t.flatten().tolist()] for t in recv])
+
+
 def combo_owner_rank(combo, world):
     """Exchange routing for the partial->final aggregation leg: the canonical
     row hash of the (returnflag, linestatus) group key (31*combine of the
@@ -114,20 +138,8 @@ def main():
     def step():
         r = sess.q1(cols, Q1_CUTOFF)
         if dist:
-            # exchange leg (config 4 shape): route each group's partial state
-            # to its owner rank (all_to_all over RCCL/xGMI), final-merge there.
-            signed = [x - (1 << 64) if x >= (1 << 63) else x for x in r.raw]
-            raw = torch.tensor(signed, dtype=torch.int64, device="cuda").view(6, 10)
-            send = [torch.zeros(6, 10, dtype=torch.int64, device="cuda")
-                    for _ in range(world)]
-            for c in range(6):
-                send[combo_owner_rank(c, world)][c] = raw[c]
-            recv = [torch.empty(6, 10, dtype=torch.int64, device="cuda")
-                    for _ in range(world)]
-            dist.all_to_all(recv, send)
-            # final agg for owned groups (exact integer merge on host ints)
-            merged = merge_raw([[x & ((1 << 64) - 1) for x in t.flatten().tolist()]
-                                for t in recv])
+            # exchange leg (config 4 shape): partial->exchange->final agg
+            merged = exchange_partials(dist, world, rank, list(r.raw), "cuda")
             return r, merged
         return r, None
 

@@ -231,8 +231,6 @@ tg_status tg_tpch_gen_orders(tg_session*, double scale_factor,
 tg_status tg_tpch_gen_customer(tg_session*, double scale_factor,
     int64_t cust_start, int64_t cust_count,
     int64_t* dev_custkey, uint8_t* dev_mktsegment);
-tg_status tg_q1_run_naive(tg_session*, const tg_tpch_lineitem_cols*,
-    int32_t shipdate_cutoff, tg_q1_result* out);  /* parity mode */
 /* test helper */
 tg_status tg_copy_dtoh(tg_session*, void* dst, const void* src, int64_t bytes);
 
@@ -253,6 +251,9 @@ typedef struct tg_q1_result {
 
 tg_status tg_q1_run(tg_session*, const tg_tpch_lineitem_cols* cols,
                     int32_t shipdate_cutoff, tg_q1_result* out);
+/* parity mode: reference sequential accumulation order (page-sized inputs) */
+tg_status tg_q1_run_naive(tg_session*, const tg_tpch_lineitem_cols* cols,
+                          int32_t shipdate_cutoff, tg_q1_result* out);
 
 #ifdef __cplusplus
 }

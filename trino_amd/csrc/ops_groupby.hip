@@ -353,10 +353,12 @@ struct HashAggOp : tg_operator {
             hipLaunchKernelGGL(k_gt_init, dim3(tg_grid_for(new_cap)), dim3(TG_BLOCK),
                                0, s->stream, t.state, new_cap, t.first_row, 0);
             TG_HIP_CHECK(hipGetLastError());
-            hipLaunchKernelGGL(k_gt_rehash, dim3((ng + TG_BLOCK - 1) / TG_BLOCK),
-                               dim3(TG_BLOCK), 0, s->stream, t, ng, nullptr,
-                               (int)group_channels.size());
-            TG_HIP_CHECK(hipGetLastError());
+            if (ng > 0) {
+                hipLaunchKernelGGL(k_gt_rehash, dim3((ng + TG_BLOCK - 1) / TG_BLOCK),
+                                   dim3(TG_BLOCK), 0, s->stream, t, ng, nullptr,
+                                   (int)group_channels.size());
+                TG_HIP_CHECK(hipGetLastError());
+            }
             TG_HIP_CHECK(hipStreamSynchronize(s->stream));
         }
         return TG_OK;
