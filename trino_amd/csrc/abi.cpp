@@ -60,3 +60,18 @@ extern "C" tg_status tg_copy_dtoh(tg_session* s, void* dst, const void* src, int
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     return TG_OK;
 }
+
+/* device buffer helpers for host pipeline drivers (tpch_queries) */
+extern "C" tg_status tg_device_malloc(tg_session* s, void** out, int64_t bytes)
+{
+    (void)s;
+    TG_HIP_CHECK(hipMalloc(out, (size_t)bytes));
+    return TG_OK;
+}
+
+extern "C" tg_status tg_device_free(tg_session* s, void* p)
+{
+    (void)s;
+    if (p) TG_HIP_CHECK(hipFree(p));
+    return TG_OK;
+}

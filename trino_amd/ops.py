@@ -333,3 +333,21 @@ def hash_rows(session, page, channels):
     _check(_lib.tg_hash_rows(session._h, ctypes.byref(page), ch.ctypes.data,
                              len(ch), out.ctypes.data))
     return out[:page.position_count]
+
+
+def page_from_device(session, ptrs_types_n):
+    """Build a device-resident page: list of (device_ptr, tg_type), n rows."""
+    ptrs_types, n = ptrs_types_n
+    blocks = (TgBlock * len(ptrs_types))()
+    for i, (ptr, ty) in enumerate(ptrs_types):
+        blocks[i].type = ty
+        blocks[i].kind = 0
+        blocks[i].position_count = n
+        blocks[i].on_device = 1
+        blocks[i].data = ptr
+    p = TgPage()
+    p.channel_count = len(ptrs_types)
+    p.position_count = n
+    p.blocks = blocks
+    p._keepalive = blocks
+    return p

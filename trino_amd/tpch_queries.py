@@ -1,0 +1,176 @@
+"""trino_amd.tpch_queries — host-side drivers for the covered TPC-H plans,
+composed from the C-ABI operators exactly as LocalExecutionPlanner chains the
+reference's operator factories. Product code: device data
+stays in HBM across the pipeline (pages flow between operators on_device=1);
+only final grouped results come to the host.
+
+Q3 (testing/trino-benchmark-queries/.../tpch/q03.sql):
+  SELECT l_orderkey, sum(l_extendedprice*(1-l_discount)) AS revenue,
+         o_orderdate, o_shippriority
+  FROM customer, orders, lineitem
+  WHERE c_mktsegment='BUILDING' AND c_custkey=o_custkey
+    AND l_orderkey=o_orderkey AND o_orderdate < DATE '1995-03-15'
+    AND l_shipdate > DATE '1995-03-15'
+  GROUP BY l_orderkey, o_orderdate, o_shippriority
+  ORDER BY revenue DESC, o_orderdate LIMIT 10
+Plan shape (reference: broadcast build customer -> join orders -> build ->
+probe lineitem -> hash agg -> TopN). TopN is §8(f) next-row: the final
+ORDER BY ... LIMIT 10 here runs on the grouped output (tiny) on the host.
+"""
+import ctypes
+import time
+
+import numpy as np
+
+from . import _lib, LineitemCols, Session, copy_dtoh
+from . import ops
+
+DATE_1995_03_15 = 9204
+
+
+def q3_gpu(session, sf, order_start=1, order_count=None, cust_start=1, cust_count=None,
+           lineitem_order_start=None, lineitem_order_count=None):
+    """Run Q3 on device; returns (orderkey, orderdate, revenue) arrays for all
+    groups (host) plus timing, and the top-10 rows in the query's output
+    order. Data generation is excluded from the timed region."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    if cust_count is None:
+        cust_count = int(150_000 * sf)
+    if lineitem_order_start is None:
+        lineitem_order_start = order_start
+        lineitem_order_count = order_count
+
+    # ---- inputs (device-resident; generation NOT timed) ----
+    # customer: custkey, mktsegment
+    cust_ck = _device_buffer(session, cust_count * 8)
+    cust_ms = _device_buffer(session, cust_count)
+    _check_lib(_lib.tg_tpch_gen_customer(session._h, sf,
+                                         cust_start, cust_count, cust_ck, cust_ms))
+    # orders: orderkey, custkey, orderdate
+    o_ok = _device_buffer(session, order_count * 8)
+    o_ck = _device_buffer(session, order_count * 8)
+    o_od = _device_buffer(session, order_count * 4)
+    _check_lib(_lib.tg_tpch_gen_orders(session._h, sf,
+                                       order_start, order_count, o_ok, o_ck, o_od))
+    # lineitem: orderkey + q3 columns
+    li = session.tpch_lineitem(sf, lineitem_order_start, lineitem_order_count,
+                               with_orderkey=True)
+
+    t0 = time.time()
+    # ---- stage 1: customer filter (mktsegment = 'BUILDING' = id 1) ----
+    cpage = ops.page_from_device(session, ([(cust_ck.value, ops.TG_BIGINT),
+                                            (cust_ms.value, ops.TG_TINYINT)], cust_count))
+    f1 = ops.filter_project(session, ops.expr(("col", 1), ("i64", 1), "eq"),
+                            [ops.expr(("col", 0))], [ops.TG_BIGINT])
+    f1.add_input(cpage)
+    f1.finish()
+    cust_sel = _take_device_page(session, f1)
+
+    bridge1 = ops.JoinBridge(session)
+    b1 = ops.hash_builder(session, bridge1, [ops.TG_BIGINT], [0], [])
+    b1.add_input(cust_sel)
+    b1.drain()
+
+    # ---- stage 2: orders filter + join customers ----
+    opage = ops.page_from_device(session, ([(o_ok.value, ops.TG_BIGINT),
+                                            (o_ck.value, ops.TG_BIGINT),
+                                            (o_od.value, ops.TG_INTEGER)], order_count))
+    f2 = ops.filter_project(session, ops.expr(("col", 2), ("i64", DATE_1995_03_15), "lt"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1)), ops.expr(("col", 2))],
+                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_INTEGER])
+    f2.add_input(opage)
+    f2.finish()
+    orders_sel = _take_device_page(session, f2)
+
+    j1 = ops.lookup_join(session, bridge1, [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_INTEGER],
+                         [1], [0, 2])      # probe key custkey; emit orderkey, orderdate
+    j1.add_input(orders_sel)
+    j1.finish()
+    orders_building = _take_device_page(session, j1)
+
+    bridge2 = ops.JoinBridge(session)
+    b2 = ops.hash_builder(session, bridge2, [ops.TG_BIGINT, ops.TG_INTEGER],
+                          [0], [1])        # key orderkey; build output orderdate
+    b2.add_input(orders_building)
+    b2.drain()
+
+    # ---- stage 3: lineitem filter+project, probe, aggregate ----
+    lpage = ops.page_from_device(session, ([(li.orderkey, ops.TG_BIGINT),
+                                            (li.shipdate, ops.TG_INTEGER),
+                                            (li.extendedprice, ops.TG_DOUBLE),
+                                            (li.discount, ops.TG_DOUBLE)],
+                                           li.row_count))
+    f3 = ops.filter_project(session,
+                            ops.expr(("col", 1), ("i64", DATE_1995_03_15), "gt"),
+                            [ops.expr(("col", 0)),
+                             ops.expr(("col", 2), ("f64", 1.0), ("col", 3), "sub", "mul")],
+                            [ops.TG_BIGINT, ops.TG_DOUBLE])
+    f3.add_input(lpage)
+    f3.finish()
+    li_sel = _take_device_page(session, f3)
+
+    j2 = ops.lookup_join(session, bridge2, [ops.TG_BIGINT, ops.TG_DOUBLE],
+                         [0], [0, 1])      # emit orderkey, discprice (+ build orderdate)
+    j2.add_input(li_sel)
+    j2.finish()
+    joined = _take_device_page(session, j2)
+
+    agg = ops.hash_aggregation(session, [0, 2], [ops.TG_BIGINT, ops.TG_INTEGER],
+                               [(ops.AGG_SUM_F64, 1)])
+    agg.add_input(joined)
+    pages = agg.drain()
+    elapsed = time.time() - t0
+
+    # cleanup
+    for op in (f1, b1, f2, j1, b2, f3, j2, agg):
+        op.close()
+    bridge1.close()
+    bridge2.close()
+    session.tpch_lineitem_free(li)
+    for p in (cust_ck, cust_ms, o_ok, o_ck, o_od):
+        _device_free(session, p)
+
+    out = pages[0] if pages else None
+    if out is None:
+        return dict(orderkey=np.empty(0, np.int64), orderdate=np.empty(0, np.int32),
+                    revenue=np.empty(0), top10=[], elapsed=elapsed)
+    orderkey = out[0]["values"]
+    orderdate = out[1]["values"]
+    revenue = out[2]["values"]
+    order = np.lexsort((orderdate, -revenue))[:10]
+    top10 = [(int(orderkey[i]), float(revenue[i]), int(orderdate[i]), 0) for i in order]
+    return dict(orderkey=orderkey, orderdate=orderdate, revenue=revenue,
+                top10=top10, elapsed=elapsed)
+
+
+# ---- small device-buffer helpers over the C ABI ----
+_lib.tg_device_malloc.restype = ctypes.c_int
+_lib.tg_device_malloc.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64]
+_lib.tg_device_free.restype = ctypes.c_int
+_lib.tg_device_free.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+
+
+def _check_lib(status):
+    from . import _check
+    _check(status)
+
+
+def _device_buffer(session, nbytes):
+    p = ctypes.c_void_p()
+    _check_lib(_lib.tg_device_malloc(session._h, ctypes.byref(p), ctypes.c_int64(nbytes)))
+    return p
+
+
+def _device_free(session, p):
+    _check_lib(_lib.tg_device_free(session._h, p))
+
+
+def _take_device_page(session, op):
+    """Fetch the operator's staged output page WITHOUT downloading: returns a
+    device tg_page usable as the next operator's input (pointers remain owned
+    by the producing operator; keep it open until the consumer is done)."""
+    out = ops.TgPage()
+    fin = ctypes.c_int(0)
+    _check_lib(_lib.tg_operator_get_output(op._h, ctypes.byref(out), ctypes.byref(fin)))
+    return out
