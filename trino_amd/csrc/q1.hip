@@ -22,6 +22,7 @@
  * order — bit-equal to oracle o_q1_naive for page-sized operator tests.
  */
 #include "common.h"
+#include <cstdlib>
 
 #define NCOMBO 6
 #define S43 8796093022208.0        /* 2^43 */
@@ -56,7 +57,8 @@ __device__ static inline void wave_reduce_u64(unsigned long long& v)
         v += __shfl_xor(v, off, 64);
 }
 
-__global__ __launch_bounds__(TG_BLOCK)
+template <int R>   /* rows per lane per iteration (2 or 4) */
+__global__ __launch_bounds__(TG_BLOCK, R == 2 ? 3 : 2)   /* 2nd arg: waves/SIMD */
 void k_q1_fused(int64_t n, const int32_t* __restrict__ shipdate,
                 const double* __restrict__ qty, const double* __restrict__ extprice,
                 const double* __restrict__ disc, const double* __restrict__ tax,
@@ -69,56 +71,76 @@ void k_q1_fused(int64_t n, const int32_t* __restrict__ shipdate,
 
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;   /* lanes total */
     const int64_t gid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    const int64_t npair = n / 2;
+    const int64_t ngrp = n / R;
 
-    for (int64_t i = gid; i < npair; i += stride) {
-        /* two rows per lane: 16 B/lane vector loads on the f64 columns */
-        const int64_t r = 2 * i;
-        int2    sd = *reinterpret_cast<const int2*>(shipdate + r);
-        double2 vq = *reinterpret_cast<const double2*>(qty + r);
-        double2 ve = *reinterpret_cast<const double2*>(extprice + r);
-        double2 vd = *reinterpret_cast<const double2*>(disc + r);
-        double2 vt = *reinterpret_cast<const double2*>(tax + r);
-        uint8_t rf0 = rflag[r], rf1 = rflag[r + 1];
-        uint8_t ls0 = lstatus[r], ls1 = lstatus[r + 1];
-
-        bool s0 = sd.x <= cutoff, s1 = sd.y <= cutoff;
-        int c0 = rf0 * 2 + ls0, c1 = rf1 * 2 + ls1;
-
-        double dp0 = ve.x * (1.0 - vd.x), dp1 = ve.y * (1.0 - vd.y);
-        double ch0 = dp0 * (1.0 + vt.x), ch1 = dp1 * (1.0 + vt.y);
-        unsigned long long yb0 = (unsigned long long)(ve.x * S43);
-        unsigned long long yb1 = (unsigned long long)(ve.y * S43);
-        unsigned long long yp0 = (unsigned long long)(dp0 * S43);
-        unsigned long long yp1 = (unsigned long long)(dp1 * S43);
-        unsigned long long yc0 = (unsigned long long)(ch0 * S43);
-        unsigned long long yc1 = (unsigned long long)(ch1 * S43);
-        unsigned long long yd0 = (unsigned long long)(vd.x * S59);
-        unsigned long long yd1 = (unsigned long long)(vd.y * S59);
-        unsigned long long yq0 = (unsigned long long)vq.x;
-        unsigned long long yq1 = (unsigned long long)vq.y;
-
+    for (int64_t i = gid; i < ngrp; i += stride) {
+        const int64_t r = R * i;
+        double vq[R], ve[R], vd[R], vt[R];
+        int32_t sd[R];
+        uint8_t rf[R], ls[R];
+        #pragma unroll
+        for (int j = 0; j < R; j += 2) {   /* 16 B/lane vector loads */
+            *reinterpret_cast<double2*>(vq + j) = *reinterpret_cast<const double2*>(qty + r + j);
+            *reinterpret_cast<double2*>(ve + j) = *reinterpret_cast<const double2*>(extprice + r + j);
+            *reinterpret_cast<double2*>(vd + j) = *reinterpret_cast<const double2*>(disc + r + j);
+            *reinterpret_cast<double2*>(vt + j) = *reinterpret_cast<const double2*>(tax + r + j);
+        }
+        if (R == 4) {
+            *reinterpret_cast<int4*>(sd) = *reinterpret_cast<const int4*>(shipdate + r);
+            *reinterpret_cast<uchar4*>(rf) = *reinterpret_cast<const uchar4*>(rflag + r);
+            *reinterpret_cast<uchar4*>(ls) = *reinterpret_cast<const uchar4*>(lstatus + r);
+        }
+        else {
+            *reinterpret_cast<int2*>(sd) = *reinterpret_cast<const int2*>(shipdate + r);
+            *reinterpret_cast<uchar2*>(rf) = *reinterpret_cast<const uchar2*>(rflag + r);
+            *reinterpret_cast<uchar2*>(ls) = *reinterpret_cast<const uchar2*>(lstatus + r);
+        }
+        bool sel[R];
+        int cb[R];
+        unsigned long long yb[R], yp[R], yc[R], yd[R], yq[R];
+        #pragma unroll
+        for (int j = 0; j < R; j++) {
+            sel[j] = sd[j] <= cutoff;
+            cb[j] = rf[j] * 2 + ls[j];
+            double dp = ve[j] * (1.0 - vd[j]);
+            double ch = dp * (1.0 + vt[j]);
+            yb[j] = (unsigned long long)(ve[j] * S43);
+            yp[j] = (unsigned long long)(dp * S43);
+            yc[j] = (unsigned long long)(ch * S43);
+            yd[j] = (unsigned long long)(vd[j] * S59);
+            yq[j] = (unsigned long long)vq[j];
+        }
         #pragma unroll
         for (int c = 0; c < NCOMBO; c++) {
-            bool m0 = s0 & (c0 == c), m1 = s1 & (c1 == c);
-            if (__any(m0 | m1)) {
-                unsigned long long k0 = m0 ? ~0ull : 0ull;
-                unsigned long long k1 = m1 ? ~0ull : 0ull;
-                acc[c].base.add((yb0 & k0)); acc[c].base.add((yb1 & k1));
-                acc[c].dp.add((yp0 & k0));   acc[c].dp.add((yp1 & k1));
-                acc[c].ch.add((yc0 & k0));   acc[c].ch.add((yc1 & k1));
-                acc[c].disc.add((yd0 & k0)); acc[c].disc.add((yd1 & k1));
-                acc[c].qty += (yq0 & k0) + (yq1 & k1);
-                acc[c].cnt += m0 + m1;
+            bool any = false;
+            #pragma unroll
+            for (int j = 0; j < R; j++) any |= (sel[j] & (cb[j] == c));
+            if (__any(any)) {
+                /* combine R masked rows into ONE u64 per sum (< 2^63), then a
+                 * single u128 add per sum per combo */
+                unsigned long long tb = 0, tp = 0, tc = 0, td = 0, tq = 0;
+                int tn = 0;
+                #pragma unroll
+                for (int j = 0; j < R; j++) {
+                    unsigned long long k = (sel[j] & (cb[j] == c)) ? ~0ull : 0ull;
+                    tb += yb[j] & k; tp += yp[j] & k; tc += yc[j] & k;
+                    td += yd[j] & k; tq += yq[j] & k;
+                    tn += (int)(k & 1);
+                }
+                acc[c].base.add(tb);
+                acc[c].dp.add(tp);
+                acc[c].ch.add(tc);
+                acc[c].disc.add(td);
+                acc[c].qty += tq;
+                acc[c].cnt += tn;
             }
         }
     }
-    /* tail row (odd n) handled by the first lane. NOTE: combo index must be
-     * compile-time constant per unrolled iteration — a runtime acc[c] index
-     * demotes the whole accumulator array to scratch (measured: 496 B/lane,
-     * 18% roofline). */
-    if (gid == 0 && (n & 1)) {
-        int64_t r = n - 1;
+    /* leftover rows [ngrp*R, n) handled by the first lane. NOTE: combo index
+     * must be compile-time constant per unrolled iteration — a runtime acc[c]
+     * index demotes the whole accumulator array to scratch (measured:
+     * 496 B/lane, 18% roofline). */
+    for (int64_t r = ngrp * R; gid == 0 && r < n; r++) {
         if (shipdate[r] <= cutoff) {
             int ct = rflag[r] * 2 + lstatus[r];
             double dpv = extprice[r] * (1.0 - disc[r]);
@@ -297,11 +319,21 @@ extern "C" tg_status tg_q1_run(tg_session* s, const tg_tpch_lineitem_cols* cols,
     if (st != TG_OK) return st;
     int64_t n = cols->row_count;
     int grid = tg_grid_for(n, 2);
+    /* variant/grid sweep hooks (measured on hardware; defaults = best) */
+    static int R = [] { const char* e = getenv("TG_Q1_R"); return e ? atoi(e) : 2; }();
+    static int blocks_env = [] { const char* e = getenv("TG_Q1_BLOCKS"); return e ? atoi(e) : 0; }();
+    if (blocks_env > 0) grid = blocks_env;
     TG_HIP_CHECK(hipEventRecord(s->ev_start, s->stream));
-    hipLaunchKernelGGL(k_q1_fused, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
-                       n, cols->shipdate, cols->quantity, cols->extendedprice,
-                       cols->discount, cols->tax, cols->returnflag, cols->linestatus,
-                       cutoff, g_scratch.partials);
+    if (R == 4)
+        hipLaunchKernelGGL(k_q1_fused<4>, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                           n, cols->shipdate, cols->quantity, cols->extendedprice,
+                           cols->discount, cols->tax, cols->returnflag, cols->linestatus,
+                           cutoff, g_scratch.partials);
+    else
+        hipLaunchKernelGGL(k_q1_fused<2>, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                           n, cols->shipdate, cols->quantity, cols->extendedprice,
+                           cols->discount, cols->tax, cols->returnflag, cols->linestatus,
+                           cutoff, g_scratch.partials);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipEventRecord(s->ev_stop, s->stream));
     hipLaunchKernelGGL(k_q1_reduce, dim3(1), dim3(256), 0, s->stream,
