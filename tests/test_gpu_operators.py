@@ -355,3 +355,42 @@ class TestPartitioner:
             assert np.array_equal(cols[0]["values"], keys[rows])   # order preserved
             assert np.array_equal(cols[1]["values"], vals[rows])
         op.close()
+
+
+class TestTopN:
+    def test_topn_matches_reference_order(self, sess, ops):
+        r = rng(12)
+        n = 30000
+        a = r.integers(0, 10**6, n).astype(np.int64)
+        b = r.standard_normal(n)
+        op = ops.topn(sess, [ops.TG_BIGINT, ops.TG_DOUBLE], [1, 0], [1, 0], 25)
+        # multiple pages
+        for sl in (slice(0, 10000), slice(10000, n)):
+            op.add_input(ops.page_from_numpy([a[sl], b[sl]]))
+        out = op.drain()[0]
+        op.close()
+        order = np.lexsort((a, -b))[:25]
+        assert np.array_equal(out[0]["values"], a[order])
+        assert np.array_equal(out[1]["values"], b[order])
+
+    def test_topn_limit_exceeds_rows(self, sess, ops):
+        a = np.array([3, 1, 2], np.int64)
+        op = ops.topn(sess, [ops.TG_BIGINT], [0], [0], 10)
+        op.add_input(ops.page_from_numpy([a]))
+        out = op.drain()[0]
+        op.close()
+        assert out[0]["values"].tolist() == [1, 2, 3]
+
+
+class TestDynamicFilter:
+    def test_join_key_range(self, sess, ops):
+        bk = np.array([42, 7, 99, 13], np.int64)
+        bv = np.zeros(4, np.int64)
+        bridge = ops.JoinBridge(sess)
+        b = ops.hash_builder(sess, bridge, [ops.TG_BIGINT, ops.TG_BIGINT], [0], [1])
+        b.add_input(ops.page_from_numpy([bk, bv]))
+        b.drain()
+        b.close()
+        mn, mx, nr = ops.join_key_range(bridge)
+        assert (mn, mx, nr) == (7, 99, 4)
+        bridge.close()

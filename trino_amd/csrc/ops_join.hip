@@ -34,6 +34,12 @@ struct JoinTable {
 struct tg_join_bridge {
     tg_session* s = nullptr;
     JoinTable t;
+    /* dynamic filter source (DynamicFilterSourceOperator /
+     * sql/gen/columnar/DynamicPageFilter.java analog): min/max of the
+     * non-null build keys, collected at build finish and pushed into the
+     * probe-side scan filter by the pipeline driver */
+    int64_t key_min = 0, key_max = 0;
+    int64_t key_rows = 0;
     /* concatenated build output channels (contiguous by global build row) */
     std::vector<DevBlock> build_channels;
     std::vector<tg_type> build_types;
@@ -58,6 +64,25 @@ __global__ void k_join_init(int32_t* slots, int64_t cap, int32_t* links, int64_t
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t k = i; k < cap; k += stride) slots[k] = -1;
     for (int64_t k = i; k < n; k += stride) links[k] = -1;
+}
+
+__global__ void k_key_minmax(const int64_t* __restrict__ keys,
+                             const uint64_t* __restrict__ valid, int64_t n,
+                             long long* mn, long long* mx, unsigned long long* cnt)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    long long lmn = INT64_MAX, lmx = INT64_MIN;
+    unsigned long long lc = 0;
+    for (; i < n; i += stride) {
+        if (valid && !((valid[i >> 6] >> (i & 63)) & 1)) continue;
+        long long v = keys[i];
+        lmn = min(lmn, v); lmx = max(lmx, v); lc++;
+    }
+    /* block-free: atomics are fine here (once per lane per launch) */
+    atomicMin(mn, lmn);
+    atomicMax(mx, lmx);
+    atomicAdd(cnt, lc);
 }
 
 __global__ void k_join_build(JoinTable t)
@@ -248,6 +273,24 @@ struct HashBuilderOp : tg_operator {
         hipLaunchKernelGGL(k_join_build, dim3(tg_grid_for(total_rows)), dim3(TG_BLOCK),
                            0, s->stream, t);
         TG_HIP_CHECK(hipGetLastError());
+        /* dynamic filter source: min/max over non-null build keys */
+        {
+            long long* d_mm = nullptr;
+            TG_HIP_CHECK(hipMalloc(&d_mm, 3 * 8));
+            long long init[3] = {INT64_MAX, INT64_MIN, 0};
+            TG_HIP_CHECK(hipMemcpyAsync(d_mm, init, 24, hipMemcpyHostToDevice, s->stream));
+            if (total_rows > 0) {
+                hipLaunchKernelGGL(k_key_minmax, dim3(tg_grid_for(total_rows)), dim3(TG_BLOCK),
+                                   0, s->stream, t.keys, t.key_valid, total_rows,
+                                   d_mm, d_mm + 1, (unsigned long long*)(d_mm + 2));
+                TG_HIP_CHECK(hipGetLastError());
+            }
+            long long mm[3];
+            TG_HIP_CHECK(hipMemcpyAsync(mm, d_mm, 24, hipMemcpyDeviceToHost, s->stream));
+            TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+            TG_HIP_CHECK(hipFree(d_mm));
+            bridge->key_min = mm[0]; bridge->key_max = mm[1]; bridge->key_rows = mm[2];
+        }
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
         for (auto& p : pages) tg_free_page(s, &p);
         pages.clear();
@@ -414,5 +457,17 @@ extern "C" tg_status tg_lookup_join_create(tg_session* s, tg_join_bridge* bridge
         op->probe_types.push_back((tg_type)probe_types[i]);
     op->probe_output.assign(probe_output_channels, probe_output_channels + n_probe_output);
     *out = op;
+    return TG_OK;
+}
+
+/* dynamic filter: the build side's non-null key range (valid after the
+ * builder's finish); has_rows=0 means the probe side can skip entirely */
+extern "C" tg_status tg_join_bridge_key_range(tg_join_bridge* b, int64_t* key_min,
+                                              int64_t* key_max, int64_t* key_rows)
+{
+    if (!b || !b->built) { TG_SET_ERR("lookup source not built"); return TG_ERR_STATE; }
+    if (key_min) *key_min = b->key_min;
+    if (key_max) *key_max = b->key_max;
+    if (key_rows) *key_rows = b->key_rows;
     return TG_OK;
 }

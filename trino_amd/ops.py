@@ -351,3 +351,34 @@ def page_from_device(session, ptrs_types_n):
     p.blocks = blocks
     p._keepalive = blocks
     return p
+
+
+_lib.tg_topn_create.restype = ctypes.c_int
+_lib.tg_topn_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32,
+                                ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int32,
+                                ctypes.c_int32, ctypes.c_void_p]
+_lib.tg_join_bridge_key_range.restype = ctypes.c_int
+_lib.tg_join_bridge_key_range.argtypes = [ctypes.c_void_p] + [ctypes.c_void_p] * 3
+
+
+def topn(session, types, sort_channels, sort_desc, limit):
+    """TopN operator (operator/TopNOperator analog): ORDER BY ... LIMIT n."""
+    h = ctypes.c_void_p()
+    ty = _i32arr(types)
+    sc = _i32arr(sort_channels)
+    sd = _i32arr(sort_desc)
+    _check(_lib.tg_topn_create(session._h, ty.ctypes.data, len(ty), sc.ctypes.data,
+                               sd.ctypes.data, len(sc), limit, ctypes.byref(h)))
+    op = Operator(session, h)
+    op._keep = (ty, sc, sd)
+    return op
+
+
+def join_key_range(bridge):
+    """Dynamic filter source: (min, max, non_null_rows) of the build keys."""
+    mn = ctypes.c_int64()
+    mx = ctypes.c_int64()
+    nr = ctypes.c_int64()
+    _check(_lib.tg_join_bridge_key_range(bridge._h, ctypes.byref(mn),
+                                         ctypes.byref(mx), ctypes.byref(nr)))
+    return mn.value, mx.value, nr.value

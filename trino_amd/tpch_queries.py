@@ -119,11 +119,19 @@ def q3_gpu(session, sf, order_start=1, order_count=None, cust_start=1, cust_coun
     agg = ops.hash_aggregation(session, [0, 2], [ops.TG_BIGINT, ops.TG_INTEGER],
                                [(ops.AGG_SUM_F64, 1)])
     agg.add_input(joined)
-    pages = agg.drain()
+    agg.finish()
+    agg_page = _take_device_page(session, agg)
+
+    # TopN: ORDER BY revenue DESC, o_orderdate ASC LIMIT 10 (TopNOperator)
+    top = ops.topn(session, [ops.TG_BIGINT, ops.TG_INTEGER, ops.TG_DOUBLE],
+                   [2, 1], [1, 0], 10)
+    top.add_input(agg_page)
+    top_pages = top.drain()
+    pages_full = ops._download_page(session, agg_page)
     elapsed = time.time() - t0
 
     # cleanup
-    for op in (f1, b1, f2, j1, b2, f3, j2, agg):
+    for op in (f1, b1, f2, j1, b2, f3, j2, agg, top):
         op.close()
     bridge1.close()
     bridge2.close()
@@ -131,15 +139,18 @@ def q3_gpu(session, sf, order_start=1, order_count=None, cust_start=1, cust_coun
     for p in (cust_ck, cust_ms, o_ok, o_ck, o_od):
         _device_free(session, p)
 
-    out = pages[0] if pages else None
+    out = pages_full
     if out is None:
         return dict(orderkey=np.empty(0, np.int64), orderdate=np.empty(0, np.int32),
                     revenue=np.empty(0), top10=[], elapsed=elapsed)
     orderkey = out[0]["values"]
     orderdate = out[1]["values"]
     revenue = out[2]["values"]
-    order = np.lexsort((orderdate, -revenue))[:10]
-    top10 = [(int(orderkey[i]), float(revenue[i]), int(orderdate[i]), 0) for i in order]
+    top10 = []
+    if top_pages:
+        tp = top_pages[0]
+        top10 = [(int(tp[0]["values"][i]), float(tp[2]["values"][i]),
+                  int(tp[1]["values"][i]), 0) for i in range(len(tp[0]["values"]))]
     return dict(orderkey=orderkey, orderdate=orderdate, revenue=revenue,
                 top10=top10, elapsed=elapsed)
 
