@@ -318,11 +318,14 @@ extern "C" tg_status tg_q1_run(tg_session* s, const tg_tpch_lineitem_cols* cols,
     tg_status st = ensure_scratch();
     if (st != TG_OK) return st;
     int64_t n = cols->row_count;
-    int grid = tg_grid_for(n, 2);
-    /* variant/grid sweep hooks (measured on hardware; defaults = best) */
-    static int R = [] { const char* e = getenv("TG_Q1_R"); return e ? atoi(e) : 2; }();
+    /* variant/grid sweep hooks (defaults measured on MI355X: R=4, 1536
+     * blocks -> 6.0 TB/s = 75% of the 8 TB/s roofline; profiles/r01) */
+    static int R = [] { const char* e = getenv("TG_Q1_R"); return e ? atoi(e) : 4; }();
     static int blocks_env = [] { const char* e = getenv("TG_Q1_BLOCKS"); return e ? atoi(e) : 0; }();
+    int grid = tg_grid_for(n, R);
+    if (grid > 1536) grid = 1536;
     if (blocks_env > 0) grid = blocks_env;
+    if (grid > TG_MAX_BLOCKS) grid = TG_MAX_BLOCKS;   /* partials buffer bound */
     TG_HIP_CHECK(hipEventRecord(s->ev_start, s->stream));
     if (R == 4)
         hipLaunchKernelGGL(k_q1_fused<4>, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
