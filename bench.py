@@ -94,6 +94,7 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--sf-per-gpu", type=float, default=100.0)
+    ap.add_argument("--workload", choices=["q1", "q3"], default="q1")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -113,6 +114,13 @@ def main():
     import trino_amd
 
     sess = trino_amd.Session(local_rank)
+
+    if args.workload == "q3":
+        run_q3(args, sess, n_gpus, rank)
+        sess.close()
+        if dist:
+            dist.destroy_process_group()
+        return
 
     # ---- workload: this rank's shard of SF (sf_per_gpu * N): contiguous
     # order parts, exactly TpchSplit(part, totalParts) semantics ----
@@ -229,6 +237,47 @@ def main():
     sess.close()
     if dist:
         dist.destroy_process_group()
+
+
+def run_q3(args, sess, n_gpus, rank):
+    """BASELINE config 3: TPC-H Q3 3-way join build/probe, single GPU.
+    Timed region = the full operator pipeline (2 joins + agg + topn) over
+    HBM-resident inputs; metric = probe-side lineitem rows per second."""
+    import torch
+    from trino_amd import tpch_queries
+    sf = args.sf_per_gpu
+    t0 = time.time()
+    inp = tpch_queries.q3_prepare(sess, sf)
+    rows = inp["li"].row_count
+    log(f"q3 inputs ready: {rows:,} lineitem rows, SF{sf:g}, {time.time()-t0:.1f}s")
+    for _ in range(args.warmup):
+        tpch_queries.q3_execute(sess, inp, download_groups=False)
+    torch.cuda.synchronize()
+    t_start = time.time()
+    for _ in range(args.steps):
+        r = tpch_queries.q3_execute(sess, inp, download_groups=False)
+    torch.cuda.synchronize()
+    elapsed = time.time() - t_start
+    tpch_queries.q3_release(sess, inp)
+    ms = elapsed * 1000 / args.steps
+    value = rows / (elapsed / args.steps)
+    out = {
+        "metric": "tpch_q3_join_probe_throughput",
+        "value": value, "unit": "rows/s", "n_gpus": n_gpus,
+        "steps": args.steps, "warmup": args.warmup, "ms_per_step": ms,
+        "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+        "dtype": "f64", "data": "synthetic",
+        "config": {"workload": f"TPC-H Q3 (3-way hash join build/probe) SF{sf:g} on 1xMI355X",
+                   "rows_lineitem": rows, "parallelism": f"dp{n_gpus}",
+                   "top10_first": r["top10"][0] if r["top10"] else None},
+        "roofline": {"bound": "hbm", "achieved": rows * 28 / (ms / 1000) / 1e9,
+                     "peak": HBM_PEAK_GBPS, "unit": "GB/s",
+                     "frac": rows * 28 / (ms / 1000) / 1e9 / HBM_PEAK_GBPS,
+                     "traffic": None},
+        "cpu_baseline": None,
+    }
+    if rank == 0:
+        print(json.dumps(out), flush=True)
 
 
 if __name__ == "__main__":

@@ -173,16 +173,78 @@ __global__ void k_probe_fill(JoinTable t, const int64_t* __restrict__ pkeys,
     }
 }
 
-__global__ void k_scan_i64(const int32_t* __restrict__ counts, int64_t n,
-                           int64_t* __restrict__ offsets, int64_t* __restrict__ total)
+/* two-level exclusive scan of per-row counts (n can be 10^8+):
+ * chunk sums -> serial chunk scan -> per-row offsets within chunk */
+#define JSCAN_CHUNK 8192
+__global__ void k_scan_chunk_sums(const int32_t* __restrict__ counts, int64_t n,
+                                  int64_t* __restrict__ chunk_sums, int64_t nchunks)
+{
+    int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= nchunks) return;
+    int64_t lo = c * JSCAN_CHUNK, hi = min(lo + JSCAN_CHUNK, n);
+    int64_t s = 0;
+    for (int64_t i = lo; i < hi; i++) s += counts[i];
+    chunk_sums[c] = s;
+}
+
+__global__ void k_scan_chunks_serial(int64_t* chunk_sums, int64_t nchunks, int64_t* total)
 {
     if (blockIdx.x || threadIdx.x) return;
     int64_t run = 0;
-    for (int64_t i = 0; i < n; i++) {
-        offsets[i] = run;
-        run += counts[i];
+    for (int64_t i = 0; i < nchunks; i++) {
+        int64_t v = chunk_sums[i];
+        chunk_sums[i] = run;
+        run += v;
     }
     *total = run;
+}
+
+__global__ void k_scan_row_offsets(const int32_t* __restrict__ counts, int64_t n,
+                                   const int64_t* __restrict__ chunk_sums,
+                                   int64_t* __restrict__ offsets, int64_t nchunks)
+{
+    /* one wave per chunk walks it carrying a running sum */
+    int64_t c = (int64_t)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+    if (c >= nchunks) return;
+    int lane = threadIdx.x % 64;
+    int64_t lo = c * JSCAN_CHUNK, hi = min(lo + JSCAN_CHUNK, n);
+    int64_t run = chunk_sums[c];
+    for (int64_t g = lo; g < hi; g += 64) {
+        int64_t i = g + lane;
+        long long v = (i < hi) ? counts[i] : 0;
+        /* exclusive wave prefix via shuffles */
+        long long pre = v;
+        #pragma unroll
+        for (int off = 1; off < 64; off <<= 1) {
+            long long o = __shfl_up(pre, off, 64);
+            if (lane >= off) pre += o;
+        }
+        long long wave_total = __shfl(pre, 63, 64);
+        if (i < hi) offsets[i] = run + pre - v;
+        run += wave_total;
+    }
+}
+
+static tg_status run_scan_counts(tg_session* s, const int32_t* d_counts, int64_t n,
+                                 int64_t* d_offsets, int64_t* d_total)
+{
+    int64_t nchunks = (n + JSCAN_CHUNK - 1) / JSCAN_CHUNK;
+    if (nchunks < 1) nchunks = 1;
+    int64_t* d_cs = nullptr;
+    TG_HIP_CHECK(hipMalloc(&d_cs, nchunks * 8));
+    hipLaunchKernelGGL(k_scan_chunk_sums, dim3(tg_grid_for(nchunks)), dim3(TG_BLOCK),
+                       0, s->stream, d_counts, n, d_cs, nchunks);
+    TG_HIP_CHECK(hipGetLastError());
+    hipLaunchKernelGGL(k_scan_chunks_serial, dim3(1), dim3(1), 0, s->stream,
+                       d_cs, nchunks, d_total);
+    TG_HIP_CHECK(hipGetLastError());
+    int wpb = TG_BLOCK / 64;
+    hipLaunchKernelGGL(k_scan_row_offsets, dim3((uint32_t)((nchunks + wpb - 1) / wpb)),
+                       dim3(TG_BLOCK), 0, s->stream, d_counts, n, d_cs, d_offsets, nchunks);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    TG_HIP_CHECK(hipFree(d_cs));
+    return TG_OK;
 }
 
 /* ---- build operator ---- */
@@ -339,9 +401,8 @@ struct LookupJoinOp : tg_operator {
         hipLaunchKernelGGL(k_probe_count, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
                            0, s->stream, t, (const int64_t*)kb.data, kb.valid, in.n, d_counts);
         TG_HIP_CHECK(hipGetLastError());
-        hipLaunchKernelGGL(k_scan_i64, dim3(1), dim3(1), 0, s->stream,
-                           d_counts, in.n, d_offsets, d_total);
-        TG_HIP_CHECK(hipGetLastError());
+        tg_status sst = run_scan_counts(s, d_counts, in.n, d_offsets, d_total);
+        if (sst != TG_OK) return sst;
         int64_t total = 0;
         TG_HIP_CHECK(hipMemcpyAsync(&total, d_total, 8, hipMemcpyDeviceToHost, s->stream));
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));

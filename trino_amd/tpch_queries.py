@@ -28,11 +28,9 @@ from . import ops
 DATE_1995_03_15 = 9204
 
 
-def q3_gpu(session, sf, order_start=1, order_count=None, cust_start=1, cust_count=None,
-           lineitem_order_start=None, lineitem_order_count=None):
-    """Run Q3 on device; returns (orderkey, orderdate, revenue) arrays for all
-    groups (host) plus timing, and the top-10 rows in the query's output
-    order. Data generation is excluded from the timed region."""
+def q3_prepare(session, sf, order_start=1, order_count=None, cust_start=1, cust_count=None,
+               lineitem_order_start=None, lineitem_order_count=None):
+    """Generate Q3's device-resident inputs (NOT part of the timed query)."""
     if order_count is None:
         order_count = int(1_500_000 * sf)
     if cust_count is None:
@@ -40,22 +38,32 @@ def q3_gpu(session, sf, order_start=1, order_count=None, cust_start=1, cust_coun
     if lineitem_order_start is None:
         lineitem_order_start = order_start
         lineitem_order_count = order_count
-
-    # ---- inputs (device-resident; generation NOT timed) ----
-    # customer: custkey, mktsegment
     cust_ck = _device_buffer(session, cust_count * 8)
     cust_ms = _device_buffer(session, cust_count)
     _check_lib(_lib.tg_tpch_gen_customer(session._h, sf,
                                          cust_start, cust_count, cust_ck, cust_ms))
-    # orders: orderkey, custkey, orderdate
     o_ok = _device_buffer(session, order_count * 8)
     o_ck = _device_buffer(session, order_count * 8)
     o_od = _device_buffer(session, order_count * 4)
     _check_lib(_lib.tg_tpch_gen_orders(session._h, sf,
                                        order_start, order_count, o_ok, o_ck, o_od))
-    # lineitem: orderkey + q3 columns
     li = session.tpch_lineitem(sf, lineitem_order_start, lineitem_order_count,
                                with_orderkey=True)
+    return dict(cust_ck=cust_ck, cust_ms=cust_ms, cust_count=cust_count,
+                o_ok=o_ok, o_ck=o_ck, o_od=o_od, order_count=order_count, li=li)
+
+
+def q3_release(session, inp):
+    session.tpch_lineitem_free(inp["li"])
+    for k in ("cust_ck", "cust_ms", "o_ok", "o_ck", "o_od"):
+        _device_free(session, inp[k])
+
+
+def q3_execute(session, inp, download_groups=True):
+    """The timed Q3 pipeline over prepared device inputs."""
+    cust_ck, cust_ms, cust_count = inp["cust_ck"], inp["cust_ms"], inp["cust_count"]
+    o_ok, o_ck, o_od, order_count = inp["o_ok"], inp["o_ck"], inp["o_od"], inp["order_count"]
+    li = inp["li"]
 
     t0 = time.time()
     # ---- stage 1: customer filter (mktsegment = 'BUILDING' = id 1) ----
@@ -127,30 +135,27 @@ def q3_gpu(session, sf, order_start=1, order_count=None, cust_start=1, cust_coun
                    [2, 1], [1, 0], 10)
     top.add_input(agg_page)
     top_pages = top.drain()
-    pages_full = ops._download_page(session, agg_page)
     elapsed = time.time() - t0
+    pages_full = ops._download_page(session, agg_page) if download_groups else None
 
-    # cleanup
+    # cleanup (inputs stay alive for repeated execution)
     for op in (f1, b1, f2, j1, b2, f3, j2, agg, top):
         op.close()
     bridge1.close()
     bridge2.close()
-    session.tpch_lineitem_free(li)
-    for p in (cust_ck, cust_ms, o_ok, o_ck, o_od):
-        _device_free(session, p)
 
-    out = pages_full
-    if out is None:
-        return dict(orderkey=np.empty(0, np.int64), orderdate=np.empty(0, np.int32),
-                    revenue=np.empty(0), top10=[], elapsed=elapsed)
-    orderkey = out[0]["values"]
-    orderdate = out[1]["values"]
-    revenue = out[2]["values"]
     top10 = []
     if top_pages:
         tp = top_pages[0]
         top10 = [(int(tp[0]["values"][i]), float(tp[2]["values"][i]),
                   int(tp[1]["values"][i]), 0) for i in range(len(tp[0]["values"]))]
+    out = pages_full
+    if out is None:
+        return dict(orderkey=None, orderdate=None, revenue=None,
+                    top10=top10, elapsed=elapsed)
+    orderkey = out[0]["values"]
+    orderdate = out[1]["values"]
+    revenue = out[2]["values"]
     return dict(orderkey=orderkey, orderdate=orderdate, revenue=revenue,
                 top10=top10, elapsed=elapsed)
 
@@ -185,3 +190,12 @@ def _take_device_page(session, op):
     fin = ctypes.c_int(0)
     _check_lib(_lib.tg_operator_get_output(op._h, ctypes.byref(out), ctypes.byref(fin)))
     return out
+
+
+def q3_gpu(session, sf, **kw):
+    """prepare + execute + release (tests / one-shot use)."""
+    inp = q3_prepare(session, sf, **kw)
+    try:
+        return q3_execute(session, inp)
+    finally:
+        q3_release(session, inp)
