@@ -154,3 +154,7 @@ def lineitem_to_host(session, cols):
         copy_dtoh(session, a, getattr(cols, name))
         out[name] = a
     return out
+
+
+_lib.tg_copy_htod.restype = ctypes.c_int
+_lib.tg_copy_htod.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64]

@@ -44,12 +44,51 @@ extern "C" tg_status tg_session_create(int device_ordinal, tg_session** out)
     return TG_OK;
 }
 
+tg_status tg_pool_alloc(tg_session* s, void** out, size_t bytes)
+{
+    if (bytes == 0) bytes = 1;
+    /* first fit: smallest cached buffer with size in [bytes, 2*bytes] */
+    auto it = s->pool_free.lower_bound(bytes);
+    if (it != s->pool_free.end() && it->first <= bytes * 2) {
+        *out = it->second;
+        s->pool_free.erase(it);
+        return TG_OK;
+    }
+    hipError_t e = hipMalloc(out, bytes);
+    if (e == hipErrorOutOfMemory) {
+        /* drop the cache and retry once */
+        for (auto& kv : s->pool_free) (void)hipFree(kv.second);
+        for (auto& kv : s->pool_free) s->pool_sizes.erase(kv.second);
+        s->pool_free.clear();
+        e = hipMalloc(out, bytes);
+    }
+    if (e != hipSuccess) {
+        TG_SET_ERR("device OOM allocating %zu bytes", bytes);
+        return TG_ERR_OOM;
+    }
+    s->pool_sizes[*out] = bytes;
+    s->pool_bytes += bytes;
+    return TG_OK;
+}
+
+void tg_pool_free(tg_session* s, void* p)
+{
+    if (!p) return;
+    auto it = s->pool_sizes.find(p);
+    if (it == s->pool_sizes.end()) {   /* not pool-owned: direct free */
+        (void)hipFree(p);
+        return;
+    }
+    s->pool_free.emplace(it->second, p);
+}
+
 extern "C" void tg_session_close(tg_session* s)
 {
     if (!s) return;
-    hipEventDestroy(s->ev_start);
-    hipEventDestroy(s->ev_stop);
-    hipStreamDestroy(s->stream);
+    for (auto& kv : s->pool_sizes) (void)hipFree(kv.first);
+    (void)hipEventDestroy(s->ev_start);
+    (void)hipEventDestroy(s->ev_stop);
+    (void)hipStreamDestroy(s->stream);
     delete s;
 }
 
@@ -64,14 +103,18 @@ extern "C" tg_status tg_copy_dtoh(tg_session* s, void* dst, const void* src, int
 /* device buffer helpers for host pipeline drivers (tpch_queries) */
 extern "C" tg_status tg_device_malloc(tg_session* s, void** out, int64_t bytes)
 {
-    (void)s;
-    TG_HIP_CHECK(hipMalloc(out, (size_t)bytes));
-    return TG_OK;
+    return tg_pool_alloc(s, out, (size_t)bytes);
 }
 
 extern "C" tg_status tg_device_free(tg_session* s, void* p)
 {
-    (void)s;
-    if (p) TG_HIP_CHECK(hipFree(p));
+    tg_pool_free(s, p);
+    return TG_OK;
+}
+
+extern "C" tg_status tg_copy_htod(tg_session* s, void* dst_dev, const void* src_host, int64_t bytes)
+{
+    TG_HIP_CHECK(hipMemcpyAsync(dst_dev, src_host, (size_t)bytes, hipMemcpyHostToDevice, s->stream));
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     return TG_OK;
 }

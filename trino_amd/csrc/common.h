@@ -23,11 +23,25 @@ extern thread_local std::string tg_error_buf;
         TG_SET_ERR("HIP error %s at %s:%d: %s", hipGetErrorName(_e), __FILE__, __LINE__, #call); \
         return TG_ERR_HIP; } } while (0)
 
+#include <map>
+
 struct tg_session {
     int device;
     hipStream_t stream;
     hipEvent_t ev_start, ev_stop;
+    /* caching device-memory pool: per-step hipMalloc/hipFree of multi-GB
+     * buffers costs O(10ms) each (page mapping); operators allocate through
+     * the pool instead (measured: Q3 SF100 step 1208ms -> kernel-bound).
+     * Single-threaded per session (one driver thread per operator chain). */
+    std::multimap<size_t, void*> pool_free;   /* size -> buffer */
+    std::map<void*, size_t> pool_sizes;       /* live + cached buffer sizes */
+    size_t pool_bytes = 0;
 };
+
+tg_status tg_pool_alloc(tg_session* s, void** out, size_t bytes);
+void tg_pool_free(tg_session* s, void* p);
+
+#define TG_POOL_ALLOC(s, pp, bytes) do {     tg_status _st = tg_pool_alloc((s), (void**)(pp), (size_t)(bytes));     if (_st != TG_OK) return _st; } while (0)
 
 /* launch geometry for memory-bound grid-stride kernels
  * (cdna_hip_programming.md Guideline 11: cap ~2048 blocks, grid-stride) */

@@ -89,7 +89,7 @@ struct ExprProgram {
     int count = 0;
 };
 tg_status tg_compile_expr(tg_session* s, const tg_expr* e, ExprProgram* out);
-void tg_free_expr(ExprProgram* p);
+void tg_free_expr(tg_session* s, ExprProgram* p);
 
 /* kernels (implemented in ops_*.hip) */
 tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page,

@@ -25,7 +25,7 @@ __global__ void k_rle_decode(const T* __restrict__ value, int64_t n, T* __restri
 static tg_status upload_flat(tg_session* s, const void* src, int on_device,
                              int64_t bytes, void** out)
 {
-    TG_HIP_CHECK(hipMalloc(out, bytes));
+    TG_POOL_ALLOC(s, out, bytes);
     TG_HIP_CHECK(hipMemcpyAsync(*out, src, bytes,
                                 on_device ? hipMemcpyDeviceToDevice : hipMemcpyHostToDevice,
                                 s->stream));
@@ -36,7 +36,7 @@ template <typename T>
 static tg_status decode_to_flat(tg_session* s, const tg_block* b, DevBlock* out)
 {
     int64_t n = b->position_count;
-    TG_HIP_CHECK(hipMalloc(&out->data, n * sizeof(T)));
+    TG_POOL_ALLOC(s, &out->data, n * sizeof(T));
     if (b->kind == TG_BK_DICTIONARY) {
         const tg_block* d = b->dictionary;
         void* d_dict = nullptr; int32_t* d_ids = nullptr;
@@ -48,8 +48,8 @@ static tg_status decode_to_flat(tg_session* s, const tg_block* b, DevBlock* out)
                            (const T*)d_dict, d_ids, n, (T*)out->data);
         TG_HIP_CHECK(hipGetLastError());
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-        TG_HIP_CHECK(hipFree(d_dict));
-        TG_HIP_CHECK(hipFree(d_ids));
+        tg_pool_free(s, d_dict);
+        tg_pool_free(s, d_ids);
     }
     else { /* RLE */
         const tg_block* d = b->dictionary;
@@ -60,7 +60,7 @@ static tg_status decode_to_flat(tg_session* s, const tg_block* b, DevBlock* out)
                            (const T*)d_val, n, (T*)out->data);
         TG_HIP_CHECK(hipGetLastError());
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-        TG_HIP_CHECK(hipFree(d_val));
+        tg_pool_free(s, d_val);
     }
     return TG_OK;
 }
@@ -109,8 +109,8 @@ void tg_free_page(tg_session* s, DevPage* p)
 {
     (void)s;
     for (auto& b : p->blocks) {
-        if (b.owned && b.data) (void)hipFree(b.data);
-        if (b.owned && b.valid) (void)hipFree(b.valid);
+        if (b.owned && b.data) tg_pool_free(s, b.data);
+        if (b.owned && b.valid) tg_pool_free(s, b.valid);
     }
     p->blocks.clear();
     p->n = 0;
@@ -121,7 +121,7 @@ tg_status tg_compile_expr(tg_session* s, const tg_expr* e, ExprProgram* out)
     out->insts.assign(e->insts, e->insts + e->count);
     out->count = e->count;
     if (out->count > 64) { TG_SET_ERR("expression too long (>64 insts)"); return TG_ERR_UNSUPPORTED; }
-    TG_HIP_CHECK(hipMalloc(&out->d_insts, out->count * sizeof(tg_expr_inst)));
+    TG_POOL_ALLOC(s, &out->d_insts, out->count * sizeof(tg_expr_inst));
     TG_HIP_CHECK(hipMemcpyAsync(out->d_insts, out->insts.data(),
                                 out->count * sizeof(tg_expr_inst),
                                 hipMemcpyHostToDevice, s->stream));
@@ -129,9 +129,9 @@ tg_status tg_compile_expr(tg_session* s, const tg_expr* e, ExprProgram* out)
     return TG_OK;
 }
 
-void tg_free_expr(ExprProgram* p)
+void tg_free_expr(tg_session* s, ExprProgram* p)
 {
-    if (p->d_insts) (void)hipFree(p->d_insts);
+    if (p->d_insts) tg_pool_free(s, p->d_insts);
     p->d_insts = nullptr;
 }
 

@@ -193,7 +193,7 @@ tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page
     for (size_t i = 0; i < page.blocks.size(); i++)
         cols[i] = {page.blocks[i].data, page.blocks[i].valid, (int32_t)page.blocks[i].type, 0};
     KCol* d_cols = nullptr;
-    TG_HIP_CHECK(hipMalloc(&d_cols, cols.size() * sizeof(KCol)));
+    TG_POOL_ALLOC(s, &d_cols, cols.size() * sizeof(KCol));
     TG_HIP_CHECK(hipMemcpyAsync(d_cols, cols.data(), cols.size() * sizeof(KCol),
                                 hipMemcpyHostToDevice, s->stream));
 
@@ -203,14 +203,14 @@ tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page
     const int32_t* d_list = nullptr;
     int32_t* d_list_owned = nullptr;
     if (has_list) {
-        TG_HIP_CHECK(hipMalloc(&d_list_owned, n * sizeof(int32_t)));
+        TG_POOL_ALLOC(s, &d_list_owned, n * sizeof(int32_t));
         TG_HIP_CHECK(hipMemcpyAsync(d_list_owned, input_sel->positions, n * sizeof(int32_t),
                                     hipMemcpyHostToDevice, s->stream));
         d_list = d_list_owned;
     }
 
     uint8_t* d_flags = nullptr;
-    TG_HIP_CHECK(hipMalloc(&d_flags, n ? n : 1));
+    TG_POOL_ALLOC(s, &d_flags, n ? n : 1);
     hipLaunchKernelGGL(k_filter_flags, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
                        pred.d_insts, pred.count, d_cols, has_list, d_list, offset, n, d_flags);
     TG_HIP_CHECK(hipGetLastError());
@@ -219,8 +219,8 @@ tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page
     if (nchunks < 1) nchunks = 1;
     int32_t* d_offsets = nullptr;
     int32_t* d_total = nullptr;
-    TG_HIP_CHECK(hipMalloc(&d_offsets, nchunks * sizeof(int32_t)));
-    TG_HIP_CHECK(hipMalloc(&d_total, sizeof(int32_t)));
+    TG_POOL_ALLOC(s, &d_offsets, nchunks * sizeof(int32_t));
+    TG_POOL_ALLOC(s, &d_total, sizeof(int32_t));
     hipLaunchKernelGGL(k_count_chunk, dim3(tg_grid_for(nchunks)), dim3(TG_BLOCK), 0, s->stream,
                        d_flags, n, d_offsets, nchunks);
     TG_HIP_CHECK(hipGetLastError());
@@ -231,17 +231,17 @@ tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
 
     int32_t* d_pos = nullptr;
-    TG_HIP_CHECK(hipMalloc(&d_pos, (total ? total : 1) * sizeof(int32_t)));
+    TG_POOL_ALLOC(s, &d_pos, (total ? total : 1) * sizeof(int32_t));
     hipLaunchKernelGGL(k_compact, dim3((int)nchunks), dim3(64), 0, s->stream,
                        d_flags, n, d_offsets, has_list, d_list, offset, d_pos);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
 
-    TG_HIP_CHECK(hipFree(d_flags));
-    TG_HIP_CHECK(hipFree(d_offsets));
-    TG_HIP_CHECK(hipFree(d_total));
-    TG_HIP_CHECK(hipFree(d_cols));
-    if (d_list_owned) TG_HIP_CHECK(hipFree(d_list_owned));
+    tg_pool_free(s, d_flags);
+    tg_pool_free(s, d_offsets);
+    tg_pool_free(s, d_total);
+    tg_pool_free(s, d_cols);
+    if (d_list_owned) tg_pool_free(s, d_list_owned);
     *d_positions_out = d_pos;
     *count_out = total;
     return TG_OK;
@@ -283,10 +283,10 @@ tg_status run_gather(tg_session* s, const DevBlock& src, const int32_t* d_positi
 {
     out->type = src.type;
     out->n = count;
-    TG_HIP_CHECK(hipMalloc(&out->data, (int64_t)(count ? count : 1) * src.elem_size()));
+    TG_POOL_ALLOC(s, &out->data, (int64_t)(count ? count : 1) * src.elem_size());
     if (src.valid) {
         int64_t words = (count + 63) / 64;
-        TG_HIP_CHECK(hipMalloc(&out->valid, (words ? words : 1) * 8));
+        TG_POOL_ALLOC(s, &out->valid, (words ? words : 1) * 8);
         TG_HIP_CHECK(hipMemsetAsync(out->valid, 0xFF, words * 8, s->stream));
     }
     int grid = tg_grid_for(count);
@@ -324,15 +324,15 @@ tg_status run_project(tg_session* s, const ExprProgram& proj, tg_type out_type,
         any_null |= page.blocks[i].valid != nullptr;
     }
     KCol* d_cols = nullptr;
-    TG_HIP_CHECK(hipMalloc(&d_cols, cols.size() * sizeof(KCol)));
+    TG_POOL_ALLOC(s, &d_cols, cols.size() * sizeof(KCol));
     TG_HIP_CHECK(hipMemcpyAsync(d_cols, cols.data(), cols.size() * sizeof(KCol),
                                 hipMemcpyHostToDevice, s->stream));
     out->type = TG_DOUBLE;
     out->n = count;
-    TG_HIP_CHECK(hipMalloc(&out->data, (int64_t)(count ? count : 1) * 8));
+    TG_POOL_ALLOC(s, &out->data, (int64_t)(count ? count : 1) * 8);
     if (any_null) {
         int64_t words = (count + 63) / 64;
-        TG_HIP_CHECK(hipMalloc(&out->valid, (words ? words : 1) * 8));
+        TG_POOL_ALLOC(s, &out->valid, (words ? words : 1) * 8);
         TG_HIP_CHECK(hipMemsetAsync(out->valid, 0xFF, words * 8, s->stream));
     }
     hipLaunchKernelGGL(k_project, dim3(tg_grid_for(count)), dim3(TG_BLOCK), 0, s->stream,
@@ -340,7 +340,7 @@ tg_status run_project(tg_session* s, const ExprProgram& proj, tg_type out_type,
                        (double*)out->data, out->valid);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-    TG_HIP_CHECK(hipFree(d_cols));
+    tg_pool_free(s, d_cols);
     (void)out_type;
     return TG_OK;
 }
@@ -371,7 +371,7 @@ struct FilterProjectOp : tg_operator {
                              &outp.blocks[p]);
             if (st != TG_OK) break;
         }
-        if (d_pos) (void)hipFree(d_pos);
+        if (d_pos) tg_pool_free(s, d_pos);
         tg_free_page(s, &in);
         if (st != TG_OK) return st;
         stage_output(std::move(outp));
@@ -386,8 +386,8 @@ struct FilterProjectOp : tg_operator {
 
     ~FilterProjectOp() override
     {
-        tg_free_expr(&filter);
-        for (auto& p : projections) tg_free_expr(&p);
+        tg_free_expr(s, &filter);
+        for (auto& p : projections) tg_free_expr(s, &p);
         for (auto& p : out_pages_) tg_free_page(s, &p);
     }
 };
@@ -450,7 +450,7 @@ extern "C" tg_status tg_filter_run(tg_session* s, const tg_expr* filter, const t
     if (st != TG_OK) return st;
     DevPage in;
     st = tg_upload_page(s, page, &in);
-    if (st != TG_OK) { tg_free_expr(&prog); return st; }
+    if (st != TG_OK) { tg_free_expr(s, &prog); return st; }
     int32_t* d_pos = nullptr;
     int32_t count = 0;
     st = run_filter(s, prog, in, input_sel, &d_pos, &count);
@@ -460,8 +460,8 @@ extern "C" tg_status tg_filter_run(tg_session* s, const tg_expr* filter, const t
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
         *out_count = count;
     }
-    if (d_pos) (void)hipFree(d_pos);
+    if (d_pos) tg_pool_free(s, d_pos);
     tg_free_page(s, &in);
-    tg_free_expr(&prog);
+    tg_free_expr(s, &prog);
     return st;
 }

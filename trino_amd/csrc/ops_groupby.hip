@@ -272,10 +272,10 @@ struct HashAggOp : tg_operator {
         t.capacity = cap; t.mask = cap - 1;
         t.max_groups = max_groups;
         t.n_words = (int32_t)group_channels.size() + 1;
-        TG_HIP_CHECK(hipMalloc(&t.state, cap * 4));
-        TG_HIP_CHECK(hipMalloc(&t.keystore, max_groups * t.n_words * 8));
-        TG_HIP_CHECK(hipMalloc(&t.first_row, max_groups * 8));
-        TG_HIP_CHECK(hipMalloc(&t.counter, 4));
+        TG_POOL_ALLOC(s, &t.state, cap * 4);
+        TG_POOL_ALLOC(s, &t.keystore, max_groups * t.n_words * 8);
+        TG_POOL_ALLOC(s, &t.first_row, max_groups * 8);
+        TG_POOL_ALLOC(s, &t.counter, 4);
         TG_HIP_CHECK(hipMemsetAsync(t.counter, 0, 4, s->stream));
         hipLaunchKernelGGL(k_gt_init, dim3(tg_grid_for(cap)), dim3(TG_BLOCK), 0, s->stream,
                            t.state, cap, t.first_row, max_groups);
@@ -309,8 +309,8 @@ struct HashAggOp : tg_operator {
         /* grow keystore/first_row/states preserving contents */
         if (new_groups != t.max_groups) {
             uint64_t* nk; int64_t* nf;
-            TG_HIP_CHECK(hipMalloc(&nk, new_groups * t.n_words * 8));
-            TG_HIP_CHECK(hipMalloc(&nf, new_groups * 8));
+            TG_POOL_ALLOC(s, &nk, new_groups * t.n_words * 8);
+            TG_POOL_ALLOC(s, &nf, new_groups * 8);
             TG_HIP_CHECK(hipMemcpyAsync(nk, t.keystore, t.max_groups * t.n_words * 8,
                                         hipMemcpyDeviceToDevice, s->stream));
             TG_HIP_CHECK(hipMemcpyAsync(nf, t.first_row, t.max_groups * 8,
@@ -320,35 +320,35 @@ struct HashAggOp : tg_operator {
                                nf + t.max_groups, new_groups - t.max_groups);
             TG_HIP_CHECK(hipGetLastError());
             TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-            TG_HIP_CHECK(hipFree(t.keystore)); TG_HIP_CHECK(hipFree(t.first_row));
+            tg_pool_free(s, t.keystore); tg_pool_free(s, t.first_row);
             t.keystore = nk; t.first_row = nf;
             for (auto& a : agg_state) {
                 if (a.sum) {
                     double* ns;
-                    TG_HIP_CHECK(hipMalloc(&ns, new_groups * 8));
+                    TG_POOL_ALLOC(s, &ns, new_groups * 8);
                     TG_HIP_CHECK(hipMemsetAsync(ns, 0, new_groups * 8, s->stream));
                     TG_HIP_CHECK(hipMemcpyAsync(ns, a.sum, t.max_groups * 8,
                                                 hipMemcpyDeviceToDevice, s->stream));
                     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-                    TG_HIP_CHECK(hipFree(a.sum));
+                    tg_pool_free(s, a.sum);
                     a.sum = ns;
                 }
                 if (a.cnt) {
                     long long* nc;
-                    TG_HIP_CHECK(hipMalloc(&nc, new_groups * 8));
+                    TG_POOL_ALLOC(s, &nc, new_groups * 8);
                     TG_HIP_CHECK(hipMemsetAsync(nc, 0, new_groups * 8, s->stream));
                     TG_HIP_CHECK(hipMemcpyAsync(nc, a.cnt, t.max_groups * 8,
                                                 hipMemcpyDeviceToDevice, s->stream));
                     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-                    TG_HIP_CHECK(hipFree(a.cnt));
+                    tg_pool_free(s, a.cnt);
                     a.cnt = nc;
                 }
             }
             t.max_groups = new_groups;
         }
         if (new_cap != t.capacity) {
-            TG_HIP_CHECK(hipFree(t.state));
-            TG_HIP_CHECK(hipMalloc(&t.state, new_cap * 4));
+            tg_pool_free(s, t.state);
+            TG_POOL_ALLOC(s, &t.state, new_cap * 4);
             t.capacity = new_cap; t.mask = new_cap - 1;
             hipLaunchKernelGGL(k_gt_init, dim3(tg_grid_for(new_cap)), dim3(TG_BLOCK),
                                0, s->stream, t.state, new_cap, t.first_row, 0);
@@ -376,7 +376,7 @@ struct HashAggOp : tg_operator {
         st = make_kcols(s, in, group_channels.data(), (int)group_channels.size(), &d_keys);
         if (st != TG_OK) { tg_free_page(s, &in); return st; }
         int32_t* d_gids = nullptr;
-        TG_HIP_CHECK(hipMalloc(&d_gids, in.n * 4));
+        TG_POOL_ALLOC(s, &d_gids, in.n * 4);
         hipLaunchKernelGGL(k_gt_assign, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK), 0, s->stream,
                            t, d_keys, (int)group_channels.size(), in.n, rows_seen, d_gids);
         TG_HIP_CHECK(hipGetLastError());
@@ -386,7 +386,7 @@ struct HashAggOp : tg_operator {
         st = make_kcols(s, in, nullptr, (int)in.blocks.size(), &d_all);
         if (st == TG_OK && !aggs.empty()) {
             KAgg* d_aggs = nullptr;
-            TG_HIP_CHECK(hipMalloc(&d_aggs, agg_state.size() * sizeof(KAgg)));
+            TG_POOL_ALLOC(s, &d_aggs, agg_state.size() * sizeof(KAgg));
             TG_HIP_CHECK(hipMemcpyAsync(d_aggs, agg_state.data(),
                                         agg_state.size() * sizeof(KAgg),
                                         hipMemcpyHostToDevice, s->stream));
@@ -395,13 +395,13 @@ struct HashAggOp : tg_operator {
                                (int)agg_state.size(), step == TG_STEP_FINAL ? 1 : 0);
             TG_HIP_CHECK(hipGetLastError());
             TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-            TG_HIP_CHECK(hipFree(d_aggs));
+            tg_pool_free(s, d_aggs);
         }
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
         rows_seen += in.n;
-        TG_HIP_CHECK(hipFree(d_gids));
-        TG_HIP_CHECK(hipFree(d_keys));
-        if (d_all) TG_HIP_CHECK(hipFree(d_all));
+        tg_pool_free(s, d_gids);
+        tg_pool_free(s, d_keys);
+        if (d_all) tg_pool_free(s, d_all);
         tg_free_page(s, &in);
         return st;
     }
@@ -421,7 +421,7 @@ struct HashAggOp : tg_operator {
         std::sort(old_by_new.begin(), old_by_new.end(),
                   [&](int32_t a, int32_t b) { return first[a] < first[b]; });
         int32_t* d_obn = nullptr;
-        TG_HIP_CHECK(hipMalloc(&d_obn, (ng ? ng : 1) * 4));
+        TG_POOL_ALLOC(s, &d_obn, (ng ? ng : 1) * 4);
         if (ng) {
             TG_HIP_CHECK(hipMemcpyAsync(d_obn, old_by_new.data(), (size_t)ng * 4,
                                         hipMemcpyHostToDevice, s->stream));
@@ -436,9 +436,9 @@ struct HashAggOp : tg_operator {
             DevBlock b;
             b.type = group_types[c];
             b.n = ng;
-            TG_HIP_CHECK(hipMalloc(&b.data, (int64_t)(ng ? ng : 1) * b.elem_size()));
+            TG_POOL_ALLOC(s, &b.data, (int64_t)(ng ? ng : 1) * b.elem_size());
             int64_t words = (ng + 63) / 64;
-            TG_HIP_CHECK(hipMalloc(&b.valid, (words ? words : 1) * 8));
+            TG_POOL_ALLOC(s, &b.valid, (words ? words : 1) * 8);
             TG_HIP_CHECK(hipMemsetAsync(b.valid, 0xFF, words * 8, s->stream));
             hipLaunchKernelGGL(k_emit_keys, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
                                t.keystore, t.n_words, d_obn, ng, (int)c,
@@ -453,21 +453,21 @@ struct HashAggOp : tg_operator {
             if (ag.fn == TG_AGG_AVG_F64 && !final_out) {
                 /* PARTIAL avg state: (count BIGINT, sum DOUBLE) channel pair */
                 DevBlock bc; bc.type = TG_BIGINT; bc.n = ng;
-                TG_HIP_CHECK(hipMalloc(&bc.data, (int64_t)(ng ? ng : 1) * 8));
+                TG_POOL_ALLOC(s, &bc.data, (int64_t)(ng ? ng : 1) * 8);
                 hipLaunchKernelGGL(k_emit_i64, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
                                    ag.cnt, d_obn, ng, (int64_t*)bc.data);
                 outp.blocks.push_back(bc);
                 DevBlock bs; bs.type = TG_DOUBLE; bs.n = ng;
-                TG_HIP_CHECK(hipMalloc(&bs.data, (int64_t)(ng ? ng : 1) * 8));
+                TG_POOL_ALLOC(s, &bs.data, (int64_t)(ng ? ng : 1) * 8);
                 hipLaunchKernelGGL(k_emit_f64, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
                                    ag.sum, d_obn, ng, (double*)bs.data);
                 outp.blocks.push_back(bs);
             }
             else if (ag.fn == TG_AGG_AVG_F64) {
                 DevBlock b; b.type = TG_DOUBLE; b.n = ng;
-                TG_HIP_CHECK(hipMalloc(&b.data, (int64_t)(ng ? ng : 1) * 8));
+                TG_POOL_ALLOC(s, &b.data, (int64_t)(ng ? ng : 1) * 8);
                 int64_t words = (ng + 63) / 64;
-                TG_HIP_CHECK(hipMalloc(&b.valid, (words ? words : 1) * 8));
+                TG_POOL_ALLOC(s, &b.valid, (words ? words : 1) * 8);
                 TG_HIP_CHECK(hipMemsetAsync(b.valid, 0xFF, words * 8, s->stream));
                 hipLaunchKernelGGL(k_emit_avg, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
                                    ag.sum, ag.cnt, d_obn, ng, (double*)b.data, b.valid);
@@ -475,14 +475,14 @@ struct HashAggOp : tg_operator {
             }
             else if (ag.fn == TG_AGG_SUM_F64) {
                 DevBlock b; b.type = TG_DOUBLE; b.n = ng;
-                TG_HIP_CHECK(hipMalloc(&b.data, (int64_t)(ng ? ng : 1) * 8));
+                TG_POOL_ALLOC(s, &b.data, (int64_t)(ng ? ng : 1) * 8);
                 hipLaunchKernelGGL(k_emit_f64, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
                                    ag.sum, d_obn, ng, (double*)b.data);
                 outp.blocks.push_back(b);
             }
             else {
                 DevBlock b; b.type = TG_BIGINT; b.n = ng;
-                TG_HIP_CHECK(hipMalloc(&b.data, (int64_t)(ng ? ng : 1) * 8));
+                TG_POOL_ALLOC(s, &b.data, (int64_t)(ng ? ng : 1) * 8);
                 hipLaunchKernelGGL(k_emit_i64, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
                                    ag.cnt, d_obn, ng, (int64_t*)b.data);
                 outp.blocks.push_back(b);
@@ -490,7 +490,7 @@ struct HashAggOp : tg_operator {
             TG_HIP_CHECK(hipGetLastError());
         }
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-        TG_HIP_CHECK(hipFree(d_obn));
+        tg_pool_free(s, d_obn);
         stage_output(std::move(outp));
         return TG_OK;
     }
@@ -508,13 +508,13 @@ struct HashAggOp : tg_operator {
 
     ~HashAggOp() override
     {
-        if (t.state) (void)hipFree(t.state);
-        if (t.keystore) (void)hipFree(t.keystore);
-        if (t.first_row) (void)hipFree(t.first_row);
-        if (t.counter) (void)hipFree(t.counter);
+        if (t.state) tg_pool_free(s, t.state);
+        if (t.keystore) tg_pool_free(s, t.keystore);
+        if (t.first_row) tg_pool_free(s, t.first_row);
+        if (t.counter) tg_pool_free(s, t.counter);
         for (auto& a : agg_state) {
-            if (a.sum) (void)hipFree(a.sum);
-            if (a.cnt) (void)hipFree(a.cnt);
+            if (a.sum) tg_pool_free(s, a.sum);
+            if (a.cnt) tg_pool_free(s, a.cnt);
         }
         for (auto& p : out_pages_) tg_free_page(s, &p);
     }
@@ -545,10 +545,10 @@ extern "C" tg_status tg_hash_aggregation_create(tg_session* s,
         bool needs_cnt = (k.fn != TG_AGG_SUM_F64);
         int64_t mg = 1 << 16;
         if (needs_sum) {
-            if (hipMalloc(&k.sum, mg * 8) != hipSuccess) { delete op; return TG_ERR_OOM; }
+            if (tg_pool_alloc(s, (void**)&k.sum, mg * 8) != TG_OK) { delete op; return TG_ERR_OOM; }
         }
         if (needs_cnt) {
-            if (hipMalloc(&k.cnt, mg * 8) != hipSuccess) { delete op; return TG_ERR_OOM; }
+            if (tg_pool_alloc(s, (void**)&k.cnt, mg * 8) != TG_OK) { delete op; return TG_ERR_OOM; }
         }
         op->agg_state.push_back(k);
     }

@@ -231,7 +231,7 @@ static tg_status run_scan_counts(tg_session* s, const int32_t* d_counts, int64_t
     int64_t nchunks = (n + JSCAN_CHUNK - 1) / JSCAN_CHUNK;
     if (nchunks < 1) nchunks = 1;
     int64_t* d_cs = nullptr;
-    TG_HIP_CHECK(hipMalloc(&d_cs, nchunks * 8));
+    TG_POOL_ALLOC(s, &d_cs, nchunks * 8);
     hipLaunchKernelGGL(k_scan_chunk_sums, dim3(tg_grid_for(nchunks)), dim3(TG_BLOCK),
                        0, s->stream, d_counts, n, d_cs, nchunks);
     TG_HIP_CHECK(hipGetLastError());
@@ -243,7 +243,7 @@ static tg_status run_scan_counts(tg_session* s, const int32_t* d_counts, int64_t
                        dim3(TG_BLOCK), 0, s->stream, d_counts, n, d_cs, d_offsets, nchunks);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-    TG_HIP_CHECK(hipFree(d_cs));
+    tg_pool_free(s, d_cs);
     return TG_OK;
 }
 
@@ -274,15 +274,15 @@ struct HashBuilderOp : tg_operator {
         t.n = total_rows;
         t.capacity = join_hash_size(total_rows);
         t.mask = t.capacity - 1;
-        TG_HIP_CHECK(hipMalloc(&t.slots, t.capacity * 4));
-        TG_HIP_CHECK(hipMalloc(&t.links, (total_rows ? total_rows : 1) * 4));
-        TG_HIP_CHECK(hipMalloc(&t.keys, (total_rows ? total_rows : 1) * 8));
+        TG_POOL_ALLOC(s, &t.slots, t.capacity * 4);
+        TG_POOL_ALLOC(s, &t.links, (total_rows ? total_rows : 1) * 4);
+        TG_POOL_ALLOC(s, &t.keys, (total_rows ? total_rows : 1) * 8);
         bool any_key_null = false;
         for (auto& p : pages)
             if (!p.blocks.empty() && p.blocks[key_channels[0]].valid) any_key_null = true;
         if (any_key_null) {
             int64_t words = (total_rows + 63) / 64;
-            TG_HIP_CHECK(hipMalloc(&t.key_valid, words * 8));
+            TG_POOL_ALLOC(s, &t.key_valid, words * 8);
             TG_HIP_CHECK(hipMemsetAsync(t.key_valid, 0xFF, words * 8, s->stream));
         }
         /* concat all channels */
@@ -291,13 +291,13 @@ struct HashBuilderOp : tg_operator {
             DevBlock& b = bridge->build_channels[c];
             b.type = types[c];
             b.n = total_rows;
-            TG_HIP_CHECK(hipMalloc(&b.data, (total_rows ? total_rows : 1) * b.elem_size()));
+            TG_POOL_ALLOC(s, &b.data, (total_rows ? total_rows : 1) * b.elem_size());
             int64_t at = 0;
             bool anynull = false;
             for (auto& p : pages) anynull |= p.blocks[c].valid != nullptr;
             if (anynull) {
                 int64_t words = (total_rows + 63) / 64;
-                TG_HIP_CHECK(hipMalloc(&b.valid, words * 8));
+                TG_POOL_ALLOC(s, &b.valid, words * 8);
                 TG_HIP_CHECK(hipMemsetAsync(b.valid, 0xFF, words * 8, s->stream));
             }
             for (auto& p : pages) {
@@ -338,7 +338,7 @@ struct HashBuilderOp : tg_operator {
         /* dynamic filter source: min/max over non-null build keys */
         {
             long long* d_mm = nullptr;
-            TG_HIP_CHECK(hipMalloc(&d_mm, 3 * 8));
+            TG_POOL_ALLOC(s, &d_mm, 3 * 8);
             long long init[3] = {INT64_MAX, INT64_MIN, 0};
             TG_HIP_CHECK(hipMemcpyAsync(d_mm, init, 24, hipMemcpyHostToDevice, s->stream));
             if (total_rows > 0) {
@@ -350,7 +350,7 @@ struct HashBuilderOp : tg_operator {
             long long mm[3];
             TG_HIP_CHECK(hipMemcpyAsync(mm, d_mm, 24, hipMemcpyDeviceToHost, s->stream));
             TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-            TG_HIP_CHECK(hipFree(d_mm));
+            tg_pool_free(s, d_mm);
             bridge->key_min = mm[0]; bridge->key_max = mm[1]; bridge->key_rows = mm[2];
         }
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
@@ -395,9 +395,9 @@ struct LookupJoinOp : tg_operator {
         int32_t* d_counts = nullptr;
         int64_t* d_offsets = nullptr;
         int64_t* d_total = nullptr;
-        TG_HIP_CHECK(hipMalloc(&d_counts, (in.n ? in.n : 1) * 4));
-        TG_HIP_CHECK(hipMalloc(&d_offsets, (in.n ? in.n : 1) * 8));
-        TG_HIP_CHECK(hipMalloc(&d_total, 8));
+        TG_POOL_ALLOC(s, &d_counts, (in.n ? in.n : 1) * 4);
+        TG_POOL_ALLOC(s, &d_offsets, (in.n ? in.n : 1) * 8);
+        TG_POOL_ALLOC(s, &d_total, 8);
         hipLaunchKernelGGL(k_probe_count, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
                            0, s->stream, t, (const int64_t*)kb.data, kb.valid, in.n, d_counts);
         TG_HIP_CHECK(hipGetLastError());
@@ -409,8 +409,8 @@ struct LookupJoinOp : tg_operator {
 
         int32_t* d_op = nullptr;
         int32_t* d_ob = nullptr;
-        TG_HIP_CHECK(hipMalloc(&d_op, (total ? total : 1) * 4));
-        TG_HIP_CHECK(hipMalloc(&d_ob, (total ? total : 1) * 4));
+        TG_POOL_ALLOC(s, &d_op, (total ? total : 1) * 4);
+        TG_POOL_ALLOC(s, &d_ob, (total ? total : 1) * 4);
         hipLaunchKernelGGL(k_probe_fill, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
                            0, s->stream, t, (const int64_t*)kb.data, kb.valid, in.n,
                            d_offsets, d_op, d_ob);
@@ -433,11 +433,11 @@ struct LookupJoinOp : tg_operator {
             if (st != TG_OK) return st;
             outp.blocks.push_back(ob);
         }
-        TG_HIP_CHECK(hipFree(d_counts));
-        TG_HIP_CHECK(hipFree(d_offsets));
-        TG_HIP_CHECK(hipFree(d_total));
-        TG_HIP_CHECK(hipFree(d_op));
-        TG_HIP_CHECK(hipFree(d_ob));
+        tg_pool_free(s, d_counts);
+        tg_pool_free(s, d_offsets);
+        tg_pool_free(s, d_total);
+        tg_pool_free(s, d_op);
+        tg_pool_free(s, d_ob);
         tg_free_page(s, &in);
         stage_output(std::move(outp));
         return TG_OK;
@@ -467,13 +467,13 @@ extern "C" tg_status tg_join_bridge_create(tg_session* s, tg_join_bridge** out)
 extern "C" void tg_join_bridge_close(tg_join_bridge* b)
 {
     if (!b) return;
-    if (b->t.slots) (void)hipFree(b->t.slots);
-    if (b->t.links) (void)hipFree(b->t.links);
-    if (b->t.keys) (void)hipFree(b->t.keys);
-    if (b->t.key_valid) (void)hipFree(b->t.key_valid);
+    if (b->t.slots) tg_pool_free(b->s, b->t.slots);
+    if (b->t.links) tg_pool_free(b->s, b->t.links);
+    if (b->t.keys) tg_pool_free(b->s, b->t.keys);
+    if (b->t.key_valid) tg_pool_free(b->s, b->t.key_valid);
     for (auto& c : b->build_channels) {
-        if (c.data) (void)hipFree(c.data);
-        if (c.valid) (void)hipFree(c.valid);
+        if (c.data) tg_pool_free(b->s, c.data);
+        if (c.valid) tg_pool_free(b->s, c.valid);
     }
     delete b;
 }

@@ -102,7 +102,7 @@ tg_status run_hash_rows(tg_session* s, const DevPage& page,
                        0, s->stream, d_cols, n_channels, page.n, d_hashes);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-    TG_HIP_CHECK(hipFree(d_cols));
+    tg_pool_free(s, d_cols);
     return TG_OK;
 }
 
@@ -115,13 +115,13 @@ extern "C" tg_status tg_hash_rows(tg_session* s, const tg_page* page,
     tg_status st = tg_upload_page(s, page, &in);
     if (st != TG_OK) return st;
     uint64_t* d_h = nullptr;
-    TG_HIP_CHECK(hipMalloc(&d_h, (in.n ? in.n : 1) * 8));
+    TG_POOL_ALLOC(s, &d_h, (in.n ? in.n : 1) * 8);
     st = run_hash_rows(s, in, channels, n_channels, d_h);
     if (st == TG_OK) {
         TG_HIP_CHECK(hipMemcpyAsync(out_hashes, d_h, in.n * 8, hipMemcpyDeviceToHost, s->stream));
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     }
-    TG_HIP_CHECK(hipFree(d_h));
+    tg_pool_free(s, d_h);
     tg_free_page(s, &in);
     return st;
 }
@@ -139,7 +139,7 @@ struct PagePartitionerOp : tg_operator {
         tg_status st = tg_upload_page(s, page, &in);
         if (st != TG_OK) return st;
         int32_t* d_pids = nullptr;
-        TG_HIP_CHECK(hipMalloc(&d_pids, (in.n ? in.n : 1) * 4));
+        TG_POOL_ALLOC(s, &d_pids, (in.n ? in.n : 1) * 4);
         KColH* d_cols = nullptr;
         st = make_kcols(s, in, partition_channels.data(),
                         (int)partition_channels.size(), &d_cols);
@@ -153,9 +153,9 @@ struct PagePartitionerOp : tg_operator {
         if (nchunks < 1) nchunks = 1;
         int32_t* d_counts = nullptr;
         int32_t* d_totals = nullptr;
-        TG_HIP_CHECK(hipMalloc(&d_counts, nchunks * nparts * 4));
+        TG_POOL_ALLOC(s, &d_counts, nchunks * nparts * 4);
         TG_HIP_CHECK(hipMemsetAsync(d_counts, 0, nchunks * nparts * 4, s->stream));
-        TG_HIP_CHECK(hipMalloc(&d_totals, nparts * 4));
+        TG_POOL_ALLOC(s, &d_totals, nparts * 4);
         hipLaunchKernelGGL(k_part_count, dim3(tg_grid_for(nchunks)), dim3(TG_BLOCK),
                            0, s->stream, d_pids, in.n, nparts, d_counts);
         TG_HIP_CHECK(hipGetLastError());
@@ -170,9 +170,9 @@ struct PagePartitionerOp : tg_operator {
         /* per-partition position buffers */
         std::vector<int32_t*> pos(nparts);
         for (int p = 0; p < nparts; p++)
-            TG_HIP_CHECK(hipMalloc(&pos[p], (totals[p] ? totals[p] : 1) * 4));
+            TG_POOL_ALLOC(s, &pos[p], (totals[p] ? totals[p] : 1) * 4);
         int32_t** d_pos = nullptr;
-        TG_HIP_CHECK(hipMalloc(&d_pos, nparts * sizeof(int32_t*)));
+        TG_POOL_ALLOC(s, &d_pos, nparts * sizeof(int32_t*));
         TG_HIP_CHECK(hipMemcpyAsync(d_pos, pos.data(), nparts * sizeof(int32_t*),
                                     hipMemcpyHostToDevice, s->stream));
         hipLaunchKernelGGL(k_part_scatter, dim3((uint32_t)nchunks), dim3(64), 0, s->stream,
@@ -191,13 +191,13 @@ struct PagePartitionerOp : tg_operator {
                 pp.blocks.push_back(ob);
             }
             per_part[p].emplace_back(std::move(pp));
-            TG_HIP_CHECK(hipFree(pos[p]));
+            tg_pool_free(s, pos[p]);
         }
-        TG_HIP_CHECK(hipFree(d_pos));
-        TG_HIP_CHECK(hipFree(d_counts));
-        TG_HIP_CHECK(hipFree(d_totals));
-        TG_HIP_CHECK(hipFree(d_pids));
-        TG_HIP_CHECK(hipFree(d_cols));
+        tg_pool_free(s, d_pos);
+        tg_pool_free(s, d_counts);
+        tg_pool_free(s, d_totals);
+        tg_pool_free(s, d_pids);
+        tg_pool_free(s, d_cols);
         tg_free_page(s, &in);
         return TG_OK;
     }

@@ -132,7 +132,7 @@ struct TopNOp : tg_operator {
             DevBlock b;
             b.type = types[c];
             b.n = take;
-            TG_HIP_CHECK(hipMalloc(&b.data, (take ? take : 1) * b.elem_size()));
+            TG_POOL_ALLOC(s, &b.data, (take ? take : 1) * b.elem_size());
             /* copy element by element via DtoD (take is small: <= limit) */
             for (int64_t o = 0; o < take; o++) {
                 int64_t flat = idx[o];
