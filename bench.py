@@ -185,19 +185,30 @@ def main():
         cores = os.cpu_count() or 1
         li = oracle.gen_lineitem(1.0)
         n_s = len(li["shipdate"])
+        # calibrate the OpenMP thread count (oversubscription on many-core
+        # hosts collapses throughput on this cache-resident sample)
+        best_t, best_rate = 1, 0.0
+        for t in sorted({1, 8, 16, 32, min(64, cores), cores}):
+            t0 = time.time()
+            oracle.q1_exact(li, Q1_CUTOFF, threads=t)
+            oracle.q1_exact(li, Q1_CUTOFF, threads=t)
+            rate = 2 * n_s / (time.time() - t0)
+            if rate > best_rate:
+                best_t, best_rate = t, rate
         reps, spent = 0, 0.0
         tcb = time.time()
-        while spent < 10.0 and reps < 200:
-            oracle.q1_exact(li, Q1_CUTOFF, threads=cores)
+        while spent < 10.0 and reps < 500:
+            oracle.q1_exact(li, Q1_CUTOFF, threads=best_t)
             reps += 1
             spent = time.time() - tcb
         cpu_rows_s = n_s * reps / spent
         cpu_baseline = {
-            "value": cpu_rows_s, "unit": "rows/s", "cores": cores, "kind": "port",
+            "value": cpu_rows_s, "unit": "rows/s", "cores": best_t, "kind": "port",
             "sample": f"TPC-H SF1 lineitem ({n_s:,} rows) x{reps} passes, "
-                      f"{spent:.1f}s, OpenMP {cores} threads; generator excluded",
+                      f"{spent:.1f}s, OpenMP {best_t} threads (calibrated over "
+                      f"{{1..{cores}}}); generator excluded",
         }
-        log(f"cpu_baseline: {cpu_rows_s/1e6:.0f} Mrow/s on {cores} cores")
+        log(f"cpu_baseline: {cpu_rows_s/1e6:.0f} Mrow/s on {best_t} threads")
 
     if rank == 0:
         out = {
