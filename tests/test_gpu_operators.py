@@ -394,3 +394,25 @@ class TestDynamicFilter:
         mn, mx, nr = ops.join_key_range(bridge)
         assert (mn, mx, nr) == (7, 99, 4)
         bridge.close()
+
+
+class TestSemiJoin:
+    def test_matched_channel(self, sess, ops):
+        bk = np.array([5, 9, 11], np.int64)
+        bridge = ops.JoinBridge(sess)
+        b = ops.hash_builder(sess, bridge, [ops.TG_BIGINT], [0], [])
+        b.add_input(ops.page_from_numpy([bk]))
+        b.drain()
+        b.close()
+        pk = np.array([9, 4, 11, 5, 6], np.int64)
+        pvalid = np.array([~np.uint64(0)], np.uint64)
+        pvalid[0] &= ~np.uint64(1 << 1)   # probe row 1 null
+        op = ops.semi_join(sess, bridge, 0)
+        op.add_input(ops.page_from_numpy([pk], valids=[pvalid]))
+        out = op.drain()[0]
+        op.close()
+        bridge.close()
+        assert out[0]["values"].tolist() == pk.tolist()
+        assert out[1]["values"].tolist() == [1, 0, 1, 1, 0]
+        mv = out[1]["valid"]
+        assert mv is not None and not (mv[0] >> np.uint64(1)) & np.uint64(1)

@@ -119,21 +119,26 @@ __global__ void k_join_build(JoinTable t)
     }
 }
 
-/* probe phase 1: count matches per probe row */
+/* probe phase 1 (DefaultPagesHash.getAddressIndex hash-all/gather-all/verify
+ * shape): count matches per probe row, caching the matched head so the fill
+ * pass never re-probes the table */
 __global__ void k_probe_count(JoinTable t, const int64_t* __restrict__ pkeys,
                               const uint64_t* __restrict__ pvalid, int64_t m,
-                              int32_t* __restrict__ counts)
+                              int32_t* __restrict__ counts,
+                              int32_t* __restrict__ heads)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < m; i += stride) {
         int32_t cnt = 0;
+        int32_t head = -1;
         if (!pvalid || ((pvalid[i >> 6] >> (i & 63)) & 1)) {
             int64_t key = pkeys[i];
             int64_t slot = (int64_t)(d_murmur3_mix((uint64_t)key) & (uint64_t)t.mask);
             int32_t cur;
             while ((cur = t.slots[slot]) != -1) {
                 if (t.keys[cur] == key) {
+                    head = cur;
                     for (int32_t p = cur; p != -1; p = t.links[p]) cnt++;
                     break;
                 }
@@ -141,12 +146,13 @@ __global__ void k_probe_count(JoinTable t, const int64_t* __restrict__ pkeys,
             }
         }
         counts[i] = cnt;
+        heads[i] = head;
     }
 }
 
-/* probe phase 2: fill (probe_row, build_row) pairs at scanned offsets */
-__global__ void k_probe_fill(JoinTable t, const int64_t* __restrict__ pkeys,
-                             const uint64_t* __restrict__ pvalid, int64_t m,
+/* probe phase 2: emit (probe_row, build_row) pairs from the cached heads */
+__global__ void k_probe_fill(JoinTable t, int64_t m,
+                             const int32_t* __restrict__ heads,
                              const int64_t* __restrict__ offsets,
                              int32_t* __restrict__ out_probe,
                              int32_t* __restrict__ out_build)
@@ -154,21 +160,11 @@ __global__ void k_probe_fill(JoinTable t, const int64_t* __restrict__ pkeys,
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < m; i += stride) {
-        if (pvalid && !((pvalid[i >> 6] >> (i & 63)) & 1)) continue;
-        int64_t key = pkeys[i];
-        int64_t slot = (int64_t)(d_murmur3_mix((uint64_t)key) & (uint64_t)t.mask);
-        int32_t cur;
         int64_t at = offsets[i];
-        while ((cur = t.slots[slot]) != -1) {
-            if (t.keys[cur] == key) {
-                for (int32_t p = cur; p != -1; p = t.links[p]) {
-                    out_probe[at] = (int32_t)i;
-                    out_build[at] = p;
-                    at++;
-                }
-                break;
-            }
-            slot = (slot + 1) & t.mask;
+        for (int32_t p = heads[i]; p != -1; p = t.links[p]) {
+            out_probe[at] = (int32_t)i;
+            out_build[at] = p;
+            at++;
         }
     }
 }
@@ -393,13 +389,16 @@ struct LookupJoinOp : tg_operator {
 
         JoinTable& t = bridge->t;
         int32_t* d_counts = nullptr;
+        int32_t* d_heads = nullptr;
         int64_t* d_offsets = nullptr;
         int64_t* d_total = nullptr;
         TG_POOL_ALLOC(s, &d_counts, (in.n ? in.n : 1) * 4);
+        TG_POOL_ALLOC(s, &d_heads, (in.n ? in.n : 1) * 4);
         TG_POOL_ALLOC(s, &d_offsets, (in.n ? in.n : 1) * 8);
         TG_POOL_ALLOC(s, &d_total, 8);
         hipLaunchKernelGGL(k_probe_count, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
-                           0, s->stream, t, (const int64_t*)kb.data, kb.valid, in.n, d_counts);
+                           0, s->stream, t, (const int64_t*)kb.data, kb.valid, in.n,
+                           d_counts, d_heads);
         TG_HIP_CHECK(hipGetLastError());
         tg_status sst = run_scan_counts(s, d_counts, in.n, d_offsets, d_total);
         if (sst != TG_OK) return sst;
@@ -412,8 +411,7 @@ struct LookupJoinOp : tg_operator {
         TG_POOL_ALLOC(s, &d_op, (total ? total : 1) * 4);
         TG_POOL_ALLOC(s, &d_ob, (total ? total : 1) * 4);
         hipLaunchKernelGGL(k_probe_fill, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
-                           0, s->stream, t, (const int64_t*)kb.data, kb.valid, in.n,
-                           d_offsets, d_op, d_ob);
+                           0, s->stream, t, in.n, d_heads, d_offsets, d_op, d_ob);
         TG_HIP_CHECK(hipGetLastError());
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
 
@@ -434,6 +432,7 @@ struct LookupJoinOp : tg_operator {
             outp.blocks.push_back(ob);
         }
         tg_pool_free(s, d_counts);
+        tg_pool_free(s, d_heads);
         tg_pool_free(s, d_offsets);
         tg_pool_free(s, d_total);
         tg_pool_free(s, d_op);
@@ -496,6 +495,89 @@ extern "C" tg_status tg_hash_builder_create(tg_session* s, tg_join_bridge* bridg
         op->types.push_back((tg_type)build_types[i]);
     bridge->build_types = op->types;
     bridge->build_output_channels.assign(output_channels, output_channels + n_output_channels);
+    *out = op;
+    return TG_OK;
+}
+
+/* ---- semi join (operator/HashSemiJoinOperator.java: appends a BOOLEAN
+ * "matched" channel to the probe page; null probe keys yield NULL) ---- */
+__global__ void k_semi_probe(JoinTable t, const int64_t* __restrict__ pkeys,
+                             const uint64_t* __restrict__ pvalid, int64_t m,
+                             int8_t* __restrict__ match, uint64_t* __restrict__ mvalid)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < m; i += stride) {
+        if (pvalid && !((pvalid[i >> 6] >> (i & 63)) & 1)) {
+            match[i] = 0;
+            atomicAnd((unsigned long long*)&mvalid[i >> 6], ~(1ull << (i & 63)));
+            continue;
+        }
+        int64_t key = pkeys[i];
+        int64_t slot = (int64_t)(d_murmur3_mix((uint64_t)key) & (uint64_t)t.mask);
+        int8_t hit = 0;
+        int32_t cur;
+        while ((cur = t.slots[slot]) != -1) {
+            if (t.keys[cur] == key) { hit = 1; break; }
+            slot = (slot + 1) & t.mask;
+        }
+        match[i] = hit;
+    }
+}
+
+struct SemiJoinOp : tg_operator {
+    tg_join_bridge* bridge = nullptr;
+    int32_t key_channel = 0;
+
+    tg_status add_input(const tg_page* page) override
+    {
+        if (!bridge->built) { TG_SET_ERR("lookup source not built"); return TG_ERR_STATE; }
+        DevPage in;
+        tg_status st = tg_upload_page(s, page, &in);
+        if (st != TG_OK) return st;
+        const DevBlock& kb = in.blocks[key_channel];
+        if (kb.type != TG_BIGINT) { tg_free_page(s, &in); TG_SET_ERR("semi join keys are BIGINT"); return TG_ERR_UNSUPPORTED; }
+        DevBlock mb;
+        mb.type = TG_BOOLEAN;
+        mb.n = in.n;
+        TG_POOL_ALLOC(s, &mb.data, in.n ? in.n : 1);
+        int64_t words = (in.n + 63) / 64;
+        TG_POOL_ALLOC(s, &mb.valid, (words ? words : 1) * 8);
+        TG_HIP_CHECK(hipMemsetAsync(mb.valid, 0xFF, words * 8, s->stream));
+        hipLaunchKernelGGL(k_semi_probe, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
+                           0, s->stream, bridge->t, (const int64_t*)kb.data, kb.valid,
+                           in.n, (int8_t*)mb.data, mb.valid);
+        TG_HIP_CHECK(hipGetLastError());
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        DevPage outp;
+        outp.n = in.n;
+        for (auto& b : in.blocks) { outp.blocks.push_back(b); }
+        in.blocks.clear();                /* ownership moved */
+        outp.blocks.push_back(mb);
+        stage_output(std::move(outp));
+        return TG_OK;
+    }
+
+    tg_status get_output(tg_page* out, int* finished) override
+    {
+        emit_staged(out, finished);
+        return TG_OK;
+    }
+
+    ~SemiJoinOp() override
+    {
+        for (auto& p : out_pages_) tg_free_page(s, &p);
+    }
+};
+
+extern "C" tg_status tg_semi_join_create(tg_session* s, tg_join_bridge* bridge,
+                                         int32_t key_channel, tg_operator** out)
+{
+    if (!s || !bridge || !out) { TG_SET_ERR("null arg"); return TG_ERR_INVALID_ARG; }
+    auto* op = new SemiJoinOp();
+    op->s = s;
+    op->bridge = bridge;
+    op->key_channel = key_channel;
     *out = op;
     return TG_OK;
 }

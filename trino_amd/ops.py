@@ -382,3 +382,16 @@ def join_key_range(bridge):
     _check(_lib.tg_join_bridge_key_range(bridge._h, ctypes.byref(mn),
                                          ctypes.byref(mx), ctypes.byref(nr)))
     return mn.value, mx.value, nr.value
+
+
+_lib.tg_semi_join_create.restype = ctypes.c_int
+_lib.tg_semi_join_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                     ctypes.c_int32, ctypes.c_void_p]
+
+
+def semi_join(session, bridge, key_channel):
+    """HashSemiJoinOperator analog: probe page + BOOLEAN matched channel
+    (NULL for null probe keys)."""
+    h = ctypes.c_void_p()
+    _check(_lib.tg_semi_join_create(session._h, bridge._h, key_channel, ctypes.byref(h)))
+    return Operator(session, h)
