@@ -369,17 +369,28 @@ struct HashAggOp : tg_operator {
         DevPage in;
         tg_status st = tg_upload_page(s, page, &in);
         if (st != TG_OK) return st;
+        KColH* d_keys = nullptr;
+        int32_t* d_gids = nullptr;
+        if (group_channels.empty()) {
+            /* scalar aggregation (AggregationOperator): one implicit group 0 */
+            TG_POOL_ALLOC(s, &d_gids, (in.n ? in.n : 1) * 4);
+            TG_HIP_CHECK(hipMemsetAsync(d_gids, 0, in.n * 4, s->stream));
+            int32_t one = 1;
+            TG_HIP_CHECK(hipMemcpyAsync(t.counter, &one, 4, hipMemcpyHostToDevice, s->stream));
+            int64_t zero = 0;
+            TG_HIP_CHECK(hipMemcpyAsync(t.first_row, &zero, 8, hipMemcpyHostToDevice, s->stream));
+            goto have_gids;
+        }
         st = grow_if_needed(in.n);
         if (st != TG_OK) { tg_free_page(s, &in); return st; }
 
-        KColH* d_keys = nullptr;
         st = make_kcols(s, in, group_channels.data(), (int)group_channels.size(), &d_keys);
         if (st != TG_OK) { tg_free_page(s, &in); return st; }
-        int32_t* d_gids = nullptr;
         TG_POOL_ALLOC(s, &d_gids, in.n * 4);
         hipLaunchKernelGGL(k_gt_assign, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK), 0, s->stream,
                            t, d_keys, (int)group_channels.size(), in.n, rows_seen, d_gids);
         TG_HIP_CHECK(hipGetLastError());
+have_gids:;
 
         /* aggregation inputs: all channels */
         KColH* d_all = nullptr;
@@ -400,7 +411,7 @@ struct HashAggOp : tg_operator {
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
         rows_seen += in.n;
         tg_pool_free(s, d_gids);
-        tg_pool_free(s, d_keys);
+        if (d_keys) tg_pool_free(s, d_keys);
         if (d_all) tg_pool_free(s, d_all);
         tg_free_page(s, &in);
         return st;
@@ -525,16 +536,18 @@ extern "C" tg_status tg_hash_aggregation_create(tg_session* s,
     const int32_t* group_types, const tg_agg_spec* aggs, int32_t n_aggs,
     int32_t step, tg_operator** out)
 {
-    if (!s || !out || !group_channels || !group_types || n_group_channels < 1 ||
-        n_group_channels > 7 || n_aggs > MAX_AGGS) {
-        TG_SET_ERR("invalid hash aggregation spec (1..7 group channels, <=%d aggs)", MAX_AGGS);
+    if (!s || !out || n_group_channels < 0 || n_group_channels > 7 ||
+        n_aggs > MAX_AGGS || (n_group_channels > 0 && (!group_channels || !group_types))) {
+        TG_SET_ERR("invalid hash aggregation spec (0..7 group channels, <=%d aggs)", MAX_AGGS);
         return TG_ERR_INVALID_ARG;
     }
     auto* op = new HashAggOp();
     op->s = s;
-    op->group_channels.assign(group_channels, group_channels + n_group_channels);
-    for (int i = 0; i < n_group_channels; i++)
-        op->group_types.push_back((tg_type)group_types[i]);
+    if (n_group_channels > 0) {
+        op->group_channels.assign(group_channels, group_channels + n_group_channels);
+        for (int i = 0; i < n_group_channels; i++)
+            op->group_types.push_back((tg_type)group_types[i]);
+    }
     op->step = (tg_agg_step)step;
     for (int a = 0; a < n_aggs; a++) {
         op->aggs.push_back(aggs[a]);

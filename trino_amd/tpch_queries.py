@@ -199,3 +199,44 @@ def q3_gpu(session, sf, **kw):
         return q3_execute(session, inp)
     finally:
         q3_release(session, inp)
+
+
+DATE_1994_01_01 = 8766
+DATE_1995_01_01 = 9131
+
+
+def q6_gpu(session, sf, order_start=1, order_count=None):
+    """TPC-H Q6 (testing/trino-benchmark-queries/.../tpch/q06.sql):
+    SELECT sum(l_extendedprice*l_discount) FROM lineitem
+    WHERE l_shipdate >= '1994-01-01' AND l_shipdate < '1995-01-01'
+      AND l_discount BETWEEN .06-0.01 AND .06+0.01 AND l_quantity < 24
+    Pipeline: filter+project -> scalar aggregation (AggregationOperator)."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    li = session.tpch_lineitem(sf, order_start, order_count)
+    t0 = time.time()
+    lpage = ops.page_from_device(session, ([(li.shipdate, ops.TG_INTEGER),
+                                            (li.quantity, ops.TG_DOUBLE),
+                                            (li.extendedprice, ops.TG_DOUBLE),
+                                            (li.discount, ops.TG_DOUBLE)],
+                                           li.row_count))
+    filt = ops.expr(("col", 0), ("i64", DATE_1994_01_01), "ge",
+                    ("col", 0), ("i64", DATE_1995_01_01), "lt", "and",
+                    ("col", 3), ("f64", 0.05), ("f64", 0.07), "between", "and",
+                    ("col", 1), ("i64", 24), "lt", "and")
+    fp = ops.filter_project(session, filt,
+                            [ops.expr(("col", 2), ("col", 3), "mul")], [ops.TG_DOUBLE])
+    fp.add_input(lpage)
+    fp.finish()
+    sel = _take_device_page(session, fp)
+    agg = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_F64, 0),
+                                                (ops.AGG_COUNT_STAR, -1)])
+    agg.add_input(sel)
+    pages = agg.drain()
+    elapsed = time.time() - t0
+    fp.close()
+    agg.close()
+    session.tpch_lineitem_free(li)
+    out = pages[0]
+    return dict(revenue=float(out[0]["values"][0]), rows=int(out[1]["values"][0]),
+                elapsed=elapsed)
