@@ -416,3 +416,65 @@ class TestSemiJoin:
         assert out[1]["values"].tolist() == [1, 0, 1, 1, 0]
         mv = out[1]["valid"]
         assert mv is not None and not (mv[0] >> np.uint64(1)) & np.uint64(1)
+
+
+class TestEncodedInputBlocks:
+    """DictionaryBlock / RunLengthEncodedBlock inputs decode to flat values on
+    upload (DictionaryBlock.java:55-58; SURVEY §8 micro-semantics: aggregation
+    loops need only the flat forms)."""
+
+    def test_dictionary_block_input(self, sess, ops):
+        import ctypes
+        r = rng(20)
+        n = 10000
+        dict_vals = np.array([100, 200, 300, 400], np.int64)
+        ids = r.integers(0, 4, n).astype(np.int32)
+        dict_block = ops.TgBlock()
+        dict_block.type = ops.TG_BIGINT
+        dict_block.kind = 0
+        dict_block.position_count = 4
+        dict_block.on_device = 0
+        dict_block.data = dict_vals.ctypes.data
+        blocks = (ops.TgBlock * 1)()
+        blocks[0].type = ops.TG_BIGINT
+        blocks[0].kind = 1  # DICTIONARY
+        blocks[0].position_count = n
+        blocks[0].on_device = 0
+        blocks[0].ids = ids.ctypes.data
+        blocks[0].dictionary = ctypes.pointer(dict_block)
+        page = ops.TgPage()
+        page.channel_count = 1
+        page.position_count = n
+        page.blocks = blocks
+        op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT], [(ops.AGG_COUNT_STAR, -1)])
+        op.add_input(page)
+        out = op.drain()[0]
+        op.close()
+        flat = dict_vals[ids]
+        gids, ng, by_gid, _ = oracle.bigint_groupby(flat)
+        assert np.array_equal(out[0]["values"], by_gid)
+        assert np.array_equal(out[1]["values"], oracle.grouped_count(gids, ng))
+
+    def test_rle_block_input(self, sess, ops):
+        import ctypes
+        n = 5000
+        val = np.array([7.5], np.float64)
+        vblock = ops.TgBlock()
+        vblock.type = ops.TG_DOUBLE
+        vblock.kind = 0
+        vblock.position_count = 1
+        vblock.on_device = 0
+        vblock.data = val.ctypes.data
+        blocks = (ops.TgBlock * 1)()
+        blocks[0].type = ops.TG_DOUBLE
+        blocks[0].kind = 2  # RLE
+        blocks[0].position_count = n
+        blocks[0].on_device = 0
+        blocks[0].dictionary = ctypes.pointer(vblock)
+        page = ops.TgPage()
+        page.channel_count = 1
+        page.position_count = n
+        page.blocks = blocks
+        e = ops.expr(("col", 0), ("f64", 7.0), "gt")
+        got = ops.filter_run(sess, e, page)
+        assert np.array_equal(got, np.arange(n, dtype=np.int32))

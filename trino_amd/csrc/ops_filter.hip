@@ -140,36 +140,14 @@ __global__ void k_compact(const uint8_t* __restrict__ flags, int64_t n,
                           int has_list, const int32_t* list, int32_t offset,
                           int32_t* __restrict__ out_positions)
 {
-    /* one block per chunk; sequential within a wave via ballot prefix */
+    /* one WAVE per chunk walks it in 64-wide steps carrying a running
+     * count — order within the chunk (and via the scanned chunk offsets,
+     * globally) is preserved */
     int64_t c = blockIdx.x;
     int64_t lo = c * CHUNK, hi = min(lo + CHUNK, n);
-    __shared__ int32_t base;
-    if (threadIdx.x == 0) base = chunk_offsets[c];
-    __syncthreads();
-    /* waves process 64-element groups of the chunk in order */
-    int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
-    int nwaves = blockDim.x / 64;
-    __shared__ int32_t wave_base[32];
-    for (int64_t g = lo + wave * 64; g < hi; g += (int64_t)nwaves * 64) {
-        int64_t i = g + lane;
-        bool f = (i < hi) && flags[i];
-        unsigned long long b = __ballot(f);
-        /* this group's start = base + flags before it: computed by a serial
-         * pass over groups via wave-ordered accumulation in LDS */
-        int my_before = __popcll(b & ((lane == 63) ? ~0ull : ((1ull << (lane + 1)) - 1))) - (f ? 1 : 0);
-        /* group-level offset: group index within chunk determines order */
-        int group_idx = (int)((g - lo) / 64);
-        if (lane == 0) wave_base[wave] = __popcll(b);
-        /* compute exclusive sum of full groups before this one: recompute by
-         * scanning flags of prior groups is O(n^2); instead two-phase below */
-        (void)group_idx;
-        (void)my_before;
-        break; /* replaced by simpler scheme below */
-    }
-    /* Simpler correct scheme: one WAVE per chunk walks it in 64-wide steps,
-     * carrying a running count (wave 0 only). CHUNK/64 = 256 steps. */
-    if (wave != 0) return;
-    int32_t run = base;
+    int lane = threadIdx.x % 64;
+    if (threadIdx.x >= 64) return;
+    int32_t run = chunk_offsets[c];
     for (int64_t g = lo; g < hi; g += 64) {
         int64_t i = g + lane;
         bool f = (i < hi) && flags[i];
