@@ -15,6 +15,7 @@ struct DevBlock {
     void* data = nullptr;          /* owned device buffer */
     uint64_t* valid = nullptr;     /* packed bitmap, bit=1 valid; null = no nulls */
     bool owned = true;
+    bool valid_owned_override = false;   /* data owned but bitmap borrowed */
     int64_t elem_size() const
     {
         switch (type) {

@@ -80,9 +80,18 @@ tg_status tg_upload_page(tg_session* s, const tg_page* in, DevPage* out)
             return TG_ERR_UNSUPPORTED;
         }
         if (b->kind == TG_BK_VALUE) {
-            tg_status st = upload_flat(s, b->data, b->on_device,
-                                       db->n * db->elem_size(), &db->data);
-            if (st != TG_OK) return st;
+            if (b->on_device) {
+                /* zero-copy borrow: device-resident value blocks flow between
+                 * operators without a DtoD pass (the producer keeps them
+                 * alive until its close — Operator output contract). */
+                db->data = const_cast<void*>(b->data);
+                db->owned = false;
+            }
+            else {
+                tg_status st = upload_flat(s, b->data, b->on_device,
+                                           db->n * db->elem_size(), &db->data);
+                if (st != TG_OK) return st;
+            }
         }
         else {
             tg_status st;
@@ -95,10 +104,18 @@ tg_status tg_upload_page(tg_session* s, const tg_page* in, DevPage* out)
             if (st != TG_OK) return st;
         }
         if (b->valid) {
-            int64_t words = (db->n + 63) / 64;
-            tg_status st = upload_flat(s, b->valid, b->on_device, words * 8,
-                                       (void**)&db->valid);
-            if (st != TG_OK) return st;
+            if (b->on_device) {
+                db->valid = const_cast<uint64_t*>(b->valid);
+                /* owned already false for borrowed VALUE data; for decoded
+                 * dictionary/RLE blocks the bitmap is borrowed separately */
+                if (db->owned) db->valid_owned_override = true;
+            }
+            else {
+                int64_t words = (db->n + 63) / 64;
+                tg_status st = upload_flat(s, b->valid, b->on_device, words * 8,
+                                           (void**)&db->valid);
+                if (st != TG_OK) return st;
+            }
         }
     }
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
@@ -110,7 +127,7 @@ void tg_free_page(tg_session* s, DevPage* p)
     (void)s;
     for (auto& b : p->blocks) {
         if (b.owned && b.data) tg_pool_free(s, b.data);
-        if (b.owned && b.valid) tg_pool_free(s, b.valid);
+        if (b.owned && !b.valid_owned_override && b.valid) tg_pool_free(s, b.valid);
     }
     p->blocks.clear();
     p->n = 0;
