@@ -231,8 +231,31 @@ tg_status tg_tpch_gen_orders(tg_session*, double scale_factor,
 tg_status tg_tpch_gen_customer(tg_session*, double scale_factor,
     int64_t cust_start, int64_t cust_count,
     int64_t* dev_custkey, uint8_t* dev_mktsegment);
-/* test helper */
+/* device buffer management for host pipeline drivers (pool-backed) */
+tg_status tg_device_malloc(tg_session*, void** out, int64_t bytes);
+tg_status tg_device_free(tg_session*, void* p);
+
+/* ---- TopN (operator/TopNOperator.java): ORDER BY ... LIMIT ----
+ * sort_desc[i]: 0 = ASC (nulls last), 1 = DESC (nulls first) */
+tg_status tg_topn_create(tg_session*,
+    const int32_t* types, int32_t n_channels,
+    const int32_t* sort_channels, const int32_t* sort_desc, int32_t n_sort,
+    int32_t limit, tg_operator** out);
+
+/* ---- semi join (operator/HashSemiJoinOperator.java): appends a BOOLEAN
+ * matched channel to the probe page (NULL for null probe keys) ---- */
+tg_status tg_semi_join_create(tg_session*, tg_join_bridge*,
+    int32_t key_channel, tg_operator** out);
+
+/* ---- dynamic filter source (DynamicFilterSourceOperator /
+ * sql/gen/columnar/DynamicPageFilter.java analog): min/max over the build
+ * side's non-null keys, valid after the builder's finish ---- */
+tg_status tg_join_bridge_key_range(tg_join_bridge*, int64_t* key_min,
+                                   int64_t* key_max, int64_t* key_rows);
+
+/* test helpers */
 tg_status tg_copy_dtoh(tg_session*, void* dst, const void* src, int64_t bytes);
+tg_status tg_copy_htod(tg_session*, void* dst_dev, const void* src_host, int64_t bytes);
 
 /* ---- fused TPC-H Q1 pipeline (the north-star benchmark kernel):
  * scan+filter(shipdate<=cutoff)+group(returnflag,linestatus)+7 aggregates,

@@ -1,0 +1,285 @@
+/* trino_gpu_cli — native C++ host driver over the C ABI (include/trino_gpu.h).
+ *
+ * The host side of this engine is C++ (the operator state machines live in
+ * libtrino_gpu); this binary is the standalone driver proving the path needs
+ * no Python: it chains the same operator calls LocalExecutionPlanner-built
+ * factories would make for the covered TPC-H plans.
+ *
+ *   trino_gpu_cli q1 [sf]      fused Q1 (scan/filter/agg), prints the 4 rows
+ *   trino_gpu_cli q3 [sf]      3-way join pipeline, prints the top 10
+ *   trino_gpu_cli q6 [sf]      scan/filter/scalar agg
+ * Requires a HIP device (no CPU fallback by design).
+ */
+#include "../include/trino_gpu.h"
+#include <cstdio>
+#include <cstdlib>
+#include <cstring>
+#include <string>
+#include <vector>
+
+static void die(tg_status st, const char* what)
+{
+    if (st != TG_OK) {
+        fprintf(stderr, "%s failed (%d): %s\n", what, st, tg_last_error());
+        exit(1);
+    }
+}
+
+static const char* RF = "ANR";
+static const char* LS = "FO";
+
+static int run_q1(tg_session* s, double sf)
+{
+    tg_tpch_lineitem_cols cols;
+    die(tg_tpch_lineitem_alloc(s, sf, 1, (int64_t)(1500000 * sf), 0, &cols), "gen");
+    tg_q1_result r;
+    die(tg_q1_run(s, &cols, 10471, &r), "q1");
+    printf("l_returnflag|l_linestatus|sum_qty|sum_base_price|sum_disc_price|sum_charge|avg_qty|avg_price|avg_disc|count_order\n");
+    for (int c = 0; c < 6; c++) {
+        if (!r.count[c]) continue;
+        printf("%c|%c|%.2f|%.2f|%.4f|%.6f|%.6f|%.6f|%.8f|%lld\n",
+               RF[c / 2], LS[c % 2], r.sum_qty[c], r.sum_base[c], r.sum_disc_price[c],
+               r.sum_charge[c], r.avg_qty[c], r.avg_price[c], r.avg_disc[c],
+               (long long)r.count[c]);
+    }
+    fprintf(stderr, "[%lld rows scanned, kernel %.3f ms, %.1f Grow/s, %.0f GB/s]\n",
+            (long long)cols.row_count, r.elapsed_ms,
+            cols.row_count / r.elapsed_ms / 1e6,
+            cols.row_count * 38.0 / (r.elapsed_ms / 1e3) / 1e9);
+    die(tg_tpch_lineitem_free(s, &cols), "free");
+    return 0;
+}
+
+static tg_expr_inst I_col(int c) { tg_expr_inst i{}; i.op = TG_EXPR_COL; i.arg0 = c; return i; }
+static tg_expr_inst I_i64(int64_t v) { tg_expr_inst i{}; i.op = TG_EXPR_CONST_I64; i.imm.i64 = v; return i; }
+static tg_expr_inst I_f64(double v) { tg_expr_inst i{}; i.op = TG_EXPR_CONST_F64; i.imm.f64 = v; return i; }
+static tg_expr_inst I_op(tg_expr_op o) { tg_expr_inst i{}; i.op = o; return i; }
+
+static tg_page dev_page(std::vector<tg_block>& blocks, int64_t n)
+{
+    tg_page p{};
+    p.channel_count = (int32_t)blocks.size();
+    p.position_count = n;
+    p.blocks = blocks.data();
+    return p;
+}
+
+static tg_block dev_block(tg_type t, const void* ptr, int64_t n)
+{
+    tg_block b{};
+    b.type = t;
+    b.kind = TG_BK_VALUE;
+    b.position_count = n;
+    b.on_device = 1;
+    b.data = ptr;
+    return b;
+}
+
+static int run_q6(tg_session* s, double sf)
+{
+    tg_tpch_lineitem_cols li;
+    die(tg_tpch_lineitem_alloc(s, sf, 1, (int64_t)(1500000 * sf), 0, &li), "gen");
+    std::vector<tg_block> blocks = {
+        dev_block(TG_INTEGER, li.shipdate, li.row_count),
+        dev_block(TG_DOUBLE, li.quantity, li.row_count),
+        dev_block(TG_DOUBLE, li.extendedprice, li.row_count),
+        dev_block(TG_DOUBLE, li.discount, li.row_count),
+    };
+    tg_page page = dev_page(blocks, li.row_count);
+    std::vector<tg_expr_inst> f = {
+        I_col(0), I_i64(8766), I_op(TG_EXPR_GE),
+        I_col(0), I_i64(9131), I_op(TG_EXPR_LT), I_op(TG_EXPR_AND),
+        I_col(3), I_f64(0.05), I_f64(0.07), I_op(TG_EXPR_BETWEEN), I_op(TG_EXPR_AND),
+        I_col(1), I_i64(24), I_op(TG_EXPR_LT), I_op(TG_EXPR_AND)};
+    std::vector<tg_expr_inst> proj = {I_col(2), I_col(3), I_op(TG_EXPR_MUL)};
+    tg_expr fe{f.data(), (int32_t)f.size()};
+    tg_expr pe{proj.data(), (int32_t)proj.size()};
+    int32_t ot = TG_DOUBLE;
+    tg_operator* fp = nullptr;
+    die(tg_filter_project_create(s, &fe, &pe, &ot, 1, &fp), "fp create");
+    die(tg_operator_add_input(fp, &page), "fp add");
+    die(tg_operator_finish(fp), "fp finish");
+    tg_page sel{};
+    int fin = 0;
+    die(tg_operator_get_output(fp, &sel, &fin), "fp out");
+
+    tg_agg_spec aggs[2] = {{TG_AGG_SUM_F64, 0}, {TG_AGG_COUNT_STAR, -1}};
+    tg_operator* agg = nullptr;
+    die(tg_hash_aggregation_create(s, nullptr, 0, nullptr, aggs, 2, TG_STEP_SINGLE, &agg), "agg create");
+    die(tg_operator_add_input(agg, &sel), "agg add");
+    die(tg_operator_finish(agg), "agg finish");
+    tg_page out{};
+    die(tg_operator_get_output(agg, &out, &fin), "agg out");
+    double revenue = 0;
+    int64_t rows = 0;
+    die(tg_copy_dtoh(s, &revenue, out.blocks[0].data, 8), "dtoh");
+    die(tg_copy_dtoh(s, &rows, out.blocks[1].data, 8), "dtoh");
+    printf("revenue\n%.4f\n", revenue);
+    fprintf(stderr, "[%lld rows matched]\n", (long long)rows);
+    tg_operator_close(agg);
+    tg_operator_close(fp);
+    die(tg_tpch_lineitem_free(s, &li), "free");
+    return 0;
+}
+
+static int run_q3(tg_session* s, double sf)
+{
+    int64_t n_orders = (int64_t)(1500000 * sf), n_cust = (int64_t)(150000 * sf);
+    /* inputs */
+    void *c_ck, *c_ms, *o_ok, *o_ck, *o_od;
+    die(tg_device_malloc(s, &c_ck, n_cust * 8), "malloc");
+    die(tg_device_malloc(s, &c_ms, n_cust), "malloc");
+    die(tg_device_malloc(s, &o_ok, n_orders * 8), "malloc");
+    die(tg_device_malloc(s, &o_ck, n_orders * 8), "malloc");
+    die(tg_device_malloc(s, &o_od, n_orders * 4), "malloc");
+    die(tg_tpch_gen_customer(s, sf, 1, n_cust, (int64_t*)c_ck, (uint8_t*)c_ms), "gen cust");
+    die(tg_tpch_gen_orders(s, sf, 1, n_orders, (int64_t*)o_ok, (int64_t*)o_ck, (int32_t*)o_od), "gen ord");
+    tg_tpch_lineitem_cols li;
+    die(tg_tpch_lineitem_alloc(s, sf, 1, n_orders, 1, &li), "gen li");
+
+    int fin = 0;
+    /* customer WHERE mktsegment='BUILDING'(id 1) -> build1(custkey) */
+    std::vector<tg_block> cb = {dev_block(TG_BIGINT, c_ck, n_cust),
+                                dev_block(TG_TINYINT, c_ms, n_cust)};
+    tg_page cpage = dev_page(cb, n_cust);
+    std::vector<tg_expr_inst> cf = {I_col(1), I_i64(1), I_op(TG_EXPR_EQ)};
+    std::vector<tg_expr_inst> cp = {I_col(0)};
+    tg_expr cfe{cf.data(), (int32_t)cf.size()};
+    tg_expr cpe{cp.data(), (int32_t)cp.size()};
+    int32_t cot = TG_BIGINT;
+    tg_operator* f1 = nullptr;
+    die(tg_filter_project_create(s, &cfe, &cpe, &cot, 1, &f1), "f1");
+    die(tg_operator_add_input(f1, &cpage), "f1 add");
+    die(tg_operator_finish(f1), "f1 fin");
+    tg_page cust_sel{};
+    die(tg_operator_get_output(f1, &cust_sel, &fin), "f1 out");
+
+    tg_join_bridge* br1 = nullptr;
+    die(tg_join_bridge_create(s, &br1), "br1");
+    int32_t bt1 = TG_BIGINT, kc0 = 0;
+    tg_operator* b1 = nullptr;
+    die(tg_hash_builder_create(s, br1, &bt1, 1, &kc0, 1, nullptr, 0, &b1), "b1");
+    die(tg_operator_add_input(b1, &cust_sel), "b1 add");
+    die(tg_operator_finish(b1), "b1 fin");
+
+    /* orders WHERE orderdate < 1995-03-15 -> join customers -> build2 */
+    std::vector<tg_block> ob = {dev_block(TG_BIGINT, o_ok, n_orders),
+                                dev_block(TG_BIGINT, o_ck, n_orders),
+                                dev_block(TG_INTEGER, o_od, n_orders)};
+    tg_page opage = dev_page(ob, n_orders);
+    std::vector<tg_expr_inst> of = {I_col(2), I_i64(9204), I_op(TG_EXPR_LT)};
+    std::vector<tg_expr_inst> op0 = {I_col(0)}, op1 = {I_col(1)}, op2 = {I_col(2)};
+    tg_expr ofe{of.data(), (int32_t)of.size()};
+    tg_expr opr[3] = {{op0.data(), 1}, {op1.data(), 1}, {op2.data(), 1}};
+    int32_t oot[3] = {TG_BIGINT, TG_BIGINT, TG_INTEGER};
+    tg_operator* f2 = nullptr;
+    die(tg_filter_project_create(s, &ofe, opr, oot, 3, &f2), "f2");
+    die(tg_operator_add_input(f2, &opage), "f2 add");
+    die(tg_operator_finish(f2), "f2 fin");
+    tg_page ord_sel{};
+    die(tg_operator_get_output(f2, &ord_sel, &fin), "f2 out");
+
+    int32_t ptypes[3] = {TG_BIGINT, TG_BIGINT, TG_INTEGER};
+    int32_t j1key = 1, j1out[2] = {0, 2};
+    tg_operator* j1 = nullptr;
+    die(tg_lookup_join_create(s, br1, ptypes, 3, &j1key, 1, j1out, 2, &j1), "j1");
+    die(tg_operator_add_input(j1, &ord_sel), "j1 add");
+    die(tg_operator_finish(j1), "j1 fin");
+    tg_page ob2{};
+    die(tg_operator_get_output(j1, &ob2, &fin), "j1 out");
+
+    tg_join_bridge* br2 = nullptr;
+    die(tg_join_bridge_create(s, &br2), "br2");
+    int32_t bt2[2] = {TG_BIGINT, TG_INTEGER}, b2out = 1;
+    tg_operator* b2 = nullptr;
+    die(tg_hash_builder_create(s, br2, bt2, 2, &kc0, 1, &b2out, 1, &b2), "b2");
+    die(tg_operator_add_input(b2, &ob2), "b2 add");
+    die(tg_operator_finish(b2), "b2 fin");
+
+    /* lineitem WHERE shipdate > 1995-03-15, project (okey, discprice) */
+    std::vector<tg_block> lb = {dev_block(TG_BIGINT, li.orderkey, li.row_count),
+                                dev_block(TG_INTEGER, li.shipdate, li.row_count),
+                                dev_block(TG_DOUBLE, li.extendedprice, li.row_count),
+                                dev_block(TG_DOUBLE, li.discount, li.row_count)};
+    tg_page lpage = dev_page(lb, li.row_count);
+    std::vector<tg_expr_inst> lf = {I_col(1), I_i64(9204), I_op(TG_EXPR_GT)};
+    std::vector<tg_expr_inst> lp0 = {I_col(0)};
+    std::vector<tg_expr_inst> lp1 = {I_col(2), I_f64(1.0), I_col(3), I_op(TG_EXPR_SUB),
+                                     I_op(TG_EXPR_MUL)};
+    tg_expr lfe{lf.data(), (int32_t)lf.size()};
+    tg_expr lpr[2] = {{lp0.data(), 1}, {lp1.data(), (int32_t)lp1.size()}};
+    int32_t lot[2] = {TG_BIGINT, TG_DOUBLE};
+    tg_operator* f3 = nullptr;
+    die(tg_filter_project_create(s, &lfe, lpr, lot, 2, &f3), "f3");
+    die(tg_operator_add_input(f3, &lpage), "f3 add");
+    die(tg_operator_finish(f3), "f3 fin");
+    tg_page li_sel{};
+    die(tg_operator_get_output(f3, &li_sel, &fin), "f3 out");
+
+    int32_t p2types[2] = {TG_BIGINT, TG_DOUBLE}, j2out[2] = {0, 1};
+    tg_operator* j2 = nullptr;
+    die(tg_lookup_join_create(s, br2, p2types, 2, &kc0, 1, j2out, 2, &j2), "j2");
+    die(tg_operator_add_input(j2, &li_sel), "j2 add");
+    die(tg_operator_finish(j2), "j2 fin");
+    tg_page joined{};
+    die(tg_operator_get_output(j2, &joined, &fin), "j2 out");
+
+    /* GROUP BY (orderkey, orderdate) SUM(discprice) -> TopN 10 */
+    int32_t gch[2] = {0, 2}, gty[2] = {TG_BIGINT, TG_INTEGER};
+    tg_agg_spec ag{TG_AGG_SUM_F64, 1};
+    tg_operator* agg = nullptr;
+    die(tg_hash_aggregation_create(s, gch, 2, gty, &ag, 1, TG_STEP_SINGLE, &agg), "agg");
+    die(tg_operator_add_input(agg, &joined), "agg add");
+    die(tg_operator_finish(agg), "agg fin");
+    tg_page groups{};
+    die(tg_operator_get_output(agg, &groups, &fin), "agg out");
+
+    int32_t tty[3] = {TG_BIGINT, TG_INTEGER, TG_DOUBLE};
+    int32_t sch[2] = {2, 1}, sdsc[2] = {1, 0};
+    tg_operator* top = nullptr;
+    die(tg_topn_create(s, tty, 3, sch, sdsc, 2, 10, &top), "topn");
+    die(tg_operator_add_input(top, &groups), "topn add");
+    die(tg_operator_finish(top), "topn fin");
+    tg_page t10{};
+    die(tg_operator_get_output(top, &t10, &fin), "topn out");
+
+    printf("l_orderkey|revenue|o_orderdate|o_shippriority\n");
+    for (int64_t i = 0; i < t10.position_count; i++) {
+        int64_t ok;
+        int32_t od;
+        double rev;
+        die(tg_copy_dtoh(s, &ok, (const char*)t10.blocks[0].data + i * 8, 8), "dtoh");
+        die(tg_copy_dtoh(s, &od, (const char*)t10.blocks[1].data + i * 4, 4), "dtoh");
+        die(tg_copy_dtoh(s, &rev, (const char*)t10.blocks[2].data + i * 8, 8), "dtoh");
+        printf("%lld|%.4f|%d|0\n", (long long)ok, rev, od);
+    }
+    for (tg_operator* op : {f1, b1, f2, j1, b2, f3, j2, agg, top}) tg_operator_close(op);
+    tg_join_bridge_close(br1);
+    tg_join_bridge_close(br2);
+    die(tg_tpch_lineitem_free(s, &li), "free");
+    for (void* p : {c_ck, c_ms, o_ok, o_ck, o_od}) die(tg_device_free(s, p), "free");
+    return 0;
+}
+
+int main(int argc, char** argv)
+{
+    if (argc < 2 || !strcmp(argv[1], "--help")) {
+        printf("usage: %s q1|q3|q6 [scale_factor]  (version: %s)\n",
+               argv[0], tg_version());
+        return argc < 2 ? 1 : 0;
+    }
+    double sf = argc > 2 ? atof(argv[2]) : 1.0;
+    tg_session* s = nullptr;
+    tg_status st = tg_session_create(0, &s);
+    if (st != TG_OK) {
+        fprintf(stderr, "no GPU: %s\n", tg_last_error());
+        return 2;
+    }
+    int rc = 1;
+    if (!strcmp(argv[1], "q1")) rc = run_q1(s, sf);
+    else if (!strcmp(argv[1], "q3")) rc = run_q3(s, sf);
+    else if (!strcmp(argv[1], "q6")) rc = run_q6(s, sf);
+    else fprintf(stderr, "unknown query %s\n", argv[1]);
+    tg_session_close(s);
+    return rc;
+}
