@@ -18,7 +18,7 @@
  *   - sparse order keys, customer mortality 3, part retail price formula,
  *     date arithmetic vs CURRENTDATE 1995-06-17            [dbgen build.c]
  * Parity pins (all inside the reference repo / public answer set):
- *   - plugin/trino-tpch/src/main/resources/tpch/statistics/sf{0.01,1.0}/*.json
+ *   - plugin/trino-tpch/src/main/resources/tpch/statistics/sf{0.01,1.0}/ JSON fixtures
  *     (exact rowCount / min / max / distinct per column, committed fixtures)
  *   - TPC-H official Q1 answer @SF1 (group counts + integral sum(quantity))
  * Above SF1 parity is spec-conformance: "parity pinned at

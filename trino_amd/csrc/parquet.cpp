@@ -2,7 +2,7 @@
  *
  * Re-designed MI355X-first replacement for the reference's Java Parquet
  * column reader on this path (lib/trino-parquet/src/main/java/io/trino/
- * parquet/reader/{ParquetReader,PageReader}.java and reader/flat/* leaf
+ * parquet/reader/{ParquetReader,PageReader}.java and reader/flat/<leaf> leaf
  * decoders): columnar decode straight into flat column buffers that upload
  * to HBM unchanged (no row pivot), serving the "SF100 Parquet" scan config.
  *
