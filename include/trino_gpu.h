@@ -209,6 +209,8 @@ typedef struct tg_tpch_lineitem_cols {
     double*  tax;
     uint8_t* returnflag;    /* dictionary id: 0=A 1=N 2=R */
     uint8_t* linestatus;    /* dictionary id: 0=F 1=O */
+    int32_t* commitdate;    /* optional (flags bit 1) */
+    int32_t* receiptdate;   /* optional (flags bit 1) */
 } tg_tpch_lineitem_cols;
 
 /* Generate lineitem rows for orders [order_start, order_start+order_count)
@@ -220,14 +222,16 @@ tg_status tg_tpch_gen_lineitem(tg_session*, double scale_factor,
 tg_status tg_tpch_lineitem_rows(tg_session*, double scale_factor,
     int64_t order_start, int64_t order_count, int64_t* row_count_out,
     int64_t** dev_offsets_out /* optional; hipFree */);
-/* allocate device buffers of the exact size and generate (bench/tests) */
+/* allocate device buffers of the exact size and generate (bench/tests);
+ * flags: bit0 = orderkey, bit1 = commitdate+receiptdate */
 tg_status tg_tpch_lineitem_alloc(tg_session*, double scale_factor,
-    int64_t order_start, int64_t order_count, int with_orderkey,
+    int64_t order_start, int64_t order_count, int flags,
     tg_tpch_lineitem_cols* out);
 tg_status tg_tpch_lineitem_free(tg_session*, tg_tpch_lineitem_cols*);
 tg_status tg_tpch_gen_orders(tg_session*, double scale_factor,
     int64_t order_start, int64_t order_count,
-    int64_t* dev_orderkey, int64_t* dev_custkey, int32_t* dev_orderdate);
+    int64_t* dev_orderkey, int64_t* dev_custkey, int32_t* dev_orderdate,
+    uint8_t* dev_priority /* nullable; ids 0..4 = 1-URGENT..5-LOW */);
 tg_status tg_tpch_gen_customer(tg_session*, double scale_factor,
     int64_t cust_start, int64_t cust_count,
     int64_t* dev_custkey, uint8_t* dev_mktsegment);

@@ -31,6 +31,8 @@ _lib.tpch_gen_lineitem.restype = ctypes.c_int64
 _lib.tpch_gen_lineitem.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 12
 _lib.tpch_gen_orders.restype = ctypes.c_int64
 _lib.tpch_gen_orders.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 3
+_lib.tpch_gen_orders2.restype = ctypes.c_int64
+_lib.tpch_gen_orders2.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
 _lib.tpch_gen_customer.restype = ctypes.c_int64
 _lib.tpch_gen_customer.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 2
 
@@ -132,8 +134,9 @@ def gen_orders(sf, order_start=1, order_count=None):
     ok = np.empty(order_count, np.int64)
     ck = np.empty(order_count, np.int64)
     od = np.empty(order_count, np.int32)
-    _lib.tpch_gen_orders(sf, order_start, order_count, _ptr(ok), _ptr(ck), _ptr(od))
-    return {"orderkey": ok, "custkey": ck, "orderdate": od}
+    pri = np.empty(order_count, np.uint8)
+    _lib.tpch_gen_orders2(sf, order_start, order_count, _ptr(ok), _ptr(ck), _ptr(od), _ptr(pri))
+    return {"orderkey": ok, "custkey": ck, "orderdate": od, "orderpriority": pri}
 
 
 def gen_customer(sf, cust_start=1, cust_count=None):

@@ -64,6 +64,7 @@
 #define TPCH_SEED_L_RDATE   373135028LL
 #define TPCH_SEED_L_RFLG    717419739LL
 #define TPCH_SEED_C_MSEG   1140279430LL
+#define TPCH_SEED_O_PRIO    591449447LL
 
 /* usage per order row for line-level streams = max lines per order */
 #define TPCH_LINES_PER_ORDER_MAX 7
@@ -151,7 +152,7 @@ TPCH_HD static inline int64_t tpch_order_custkey(tpch_rng* ckey_rng, int64_t max
  * comment, suppkey, shipinstruct, shipmode, text pool) never interact with
  * these streams, so skipping them is exact. */
 typedef struct {
-    tpch_rng odate, lcnt, ckey;
+    tpch_rng odate, lcnt, ckey, opri;
     tpch_rng qty, dcnt, tax, pkey, ship, cdate, rdate, rflg;
     int64_t max_custkey;
     int64_t max_partkey;
@@ -162,6 +163,7 @@ TPCH_HD static inline void tpch_order_streams_init(tpch_order_streams* s, double
     tpch_rng_init(&s->odate, TPCH_SEED_O_ODATE, 1);
     tpch_rng_init(&s->lcnt,  TPCH_SEED_O_LCNT,  1);
     tpch_rng_init(&s->ckey,  TPCH_SEED_O_CKEY,  1);
+    tpch_rng_init(&s->opri,  TPCH_SEED_O_PRIO,  1);
     tpch_rng_init(&s->qty,   TPCH_SEED_L_QTY,   TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_init(&s->dcnt,  TPCH_SEED_L_DCNT,  TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_init(&s->tax,   TPCH_SEED_L_TAX,   TPCH_LINES_PER_ORDER_MAX);
@@ -181,6 +183,7 @@ TPCH_HD static inline void tpch_order_streams_seek(tpch_order_streams* s, int64_
     tpch_rng_skip(&s->odate, n);
     tpch_rng_skip(&s->lcnt,  n);
     tpch_rng_skip(&s->ckey,  n);
+    tpch_rng_skip(&s->opri,  n);
     tpch_rng_skip(&s->qty,   n * TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_skip(&s->dcnt,  n * TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_skip(&s->tax,   n * TPCH_LINES_PER_ORDER_MAX);
@@ -211,6 +214,7 @@ typedef struct {
     int64_t custkey;
     int32_t orderdate;      /* epoch days */
     int32_t line_count;
+    uint8_t priority;       /* 0..4 = 1-URGENT,2-HIGH,3-MEDIUM,4-NOT SPECIFIED,5-LOW */
 } tpch_order_row;
 
 /* Generate one order's header; must be called in sequence (or after seek). */
@@ -221,6 +225,8 @@ TPCH_HD static inline void tpch_gen_order(tpch_order_streams* s, int64_t order_i
     o->custkey   = tpch_order_custkey(&s->ckey, s->max_custkey);
     o->orderdate = TPCH_STARTDATE_EPOCH +
                    (int32_t)tpch_rng_int(&s->odate, 0, TPCH_ORDER_DATE_SPAN - 1);
+    /* dists.dss order priority: 5 uniform entries (pick_str) */
+    o->priority = (uint8_t)(tpch_rng_int(&s->opri, 1, 5) - 1);
     o->line_count = (int32_t)tpch_rng_int(&s->lcnt, 1, TPCH_LINES_PER_ORDER_MAX);
 }
 
@@ -257,6 +263,7 @@ TPCH_HD static inline void tpch_order_row_finished(tpch_order_streams* s)
     tpch_rng_row_finished(&s->odate);
     tpch_rng_row_finished(&s->lcnt);
     tpch_rng_row_finished(&s->ckey);
+    tpch_rng_row_finished(&s->opri);
     tpch_rng_row_finished(&s->qty);
     tpch_rng_row_finished(&s->dcnt);
     tpch_rng_row_finished(&s->tax);

@@ -6,6 +6,9 @@
 #include "tpch_core.h"
 #include <string.h>
 
+int64_t tpch_gen_orders2(double sf, int64_t order_start, int64_t order_count,
+    int64_t* orderkey, int64_t* custkey, int32_t* orderdate, uint8_t* priority);
+
 #define EXPORT __attribute__((visibility("default")))
 
 /* Count lineitem rows for a dense order range [order_start, order_start+order_count)
@@ -65,6 +68,13 @@ EXPORT int64_t tpch_gen_lineitem(double sf, int64_t order_start, int64_t order_c
 EXPORT int64_t tpch_gen_orders(double sf, int64_t order_start, int64_t order_count,
     int64_t* orderkey, int64_t* custkey, int32_t* orderdate)
 {
+    return tpch_gen_orders2(sf, order_start, order_count, orderkey, custkey,
+                            orderdate, 0);
+}
+
+EXPORT int64_t tpch_gen_orders2(double sf, int64_t order_start, int64_t order_count,
+    int64_t* orderkey, int64_t* custkey, int32_t* orderdate, uint8_t* priority)
+{
     tpch_order_streams s;
     tpch_order_streams_init(&s, sf);
     tpch_order_streams_seek(&s, order_start);
@@ -74,6 +84,7 @@ EXPORT int64_t tpch_gen_orders(double sf, int64_t order_start, int64_t order_cou
         if (orderkey)  orderkey[i] = o.orderkey;
         if (custkey)   custkey[i] = o.custkey;
         if (orderdate) orderdate[i] = o.orderdate;
+        if (priority)  priority[i] = o.priority;
         tpch_order_row_finished(&s);
     }
     return order_count;

@@ -46,6 +46,8 @@ class LineitemCols(ctypes.Structure):
         ("tax", ctypes.c_void_p),
         ("returnflag", ctypes.c_void_p),
         ("linestatus", ctypes.c_void_p),
+        ("commitdate", ctypes.c_void_p),
+        ("receiptdate", ctypes.c_void_p),
     ]
 
 
@@ -75,7 +77,7 @@ _lib.tg_q1_run_naive.argtypes = [ctypes.c_void_p, ctypes.POINTER(LineitemCols),
                                  ctypes.c_int32, ctypes.POINTER(Q1Result)]
 _lib.tg_tpch_gen_orders.restype = ctypes.c_int
 _lib.tg_tpch_gen_orders.argtypes = [ctypes.c_void_p, ctypes.c_double, ctypes.c_int64,
-                                    ctypes.c_int64] + [ctypes.c_void_p] * 3
+                                    ctypes.c_int64] + [ctypes.c_void_p] * 4
 _lib.tg_tpch_gen_customer.restype = ctypes.c_int
 _lib.tg_tpch_gen_customer.argtypes = [ctypes.c_void_p, ctypes.c_double, ctypes.c_int64,
                                       ctypes.c_int64] + [ctypes.c_void_p] * 2
@@ -104,12 +106,14 @@ class Session:
             self._h = None
 
     # --- TPC-H device generator (bench/test input infrastructure) ---
-    def tpch_lineitem(self, sf, order_start=1, order_count=None, with_orderkey=False):
+    def tpch_lineitem(self, sf, order_start=1, order_count=None, with_orderkey=False,
+                      with_dates=False):
         if order_count is None:
             order_count = int(1_500_000 * sf)
         cols = LineitemCols()
+        flags = (1 if with_orderkey else 0) | (2 if with_dates else 0)
         _check(_lib.tg_tpch_lineitem_alloc(self._h, sf, order_start, order_count,
-                                           1 if with_orderkey else 0, ctypes.byref(cols)))
+                                           flags, ctypes.byref(cols)))
         return cols
 
     def tpch_lineitem_free(self, cols):

@@ -37,7 +37,8 @@ __global__ void k_gen_lineitem(double sf, int64_t order_start, int64_t order_cou
                                int64_t* orderkey, int32_t* shipdate,
                                double* quantity, double* extendedprice,
                                double* discount, double* tax,
-                               uint8_t* returnflag, uint8_t* linestatus)
+                               uint8_t* returnflag, uint8_t* linestatus,
+                               int32_t* commitdate, int32_t* receiptdate)
 {
     int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (g >= n_groups) return;
@@ -61,13 +62,16 @@ __global__ void k_gen_lineitem(double sf, int64_t order_start, int64_t order_cou
             if (tax)           tax[n] = (double)l.tax_pct / 100.0;
             if (returnflag)    returnflag[n] = l.returnflag;
             if (linestatus)    linestatus[n] = l.linestatus;
+            if (commitdate)    commitdate[n] = l.commitdate;
+            if (receiptdate)   receiptdate[n] = l.receiptdate;
         }
         tpch_order_row_finished(&s);
     }
 }
 
 __global__ void k_gen_orders(double sf, int64_t order_start, int64_t order_count,
-                             int64_t* orderkey, int64_t* custkey, int32_t* orderdate)
+                             int64_t* orderkey, int64_t* custkey, int32_t* orderdate,
+                             uint8_t* priority)
 {
     int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t n_groups = (order_count + GOT - 1) / GOT;
@@ -84,6 +88,7 @@ __global__ void k_gen_orders(double sf, int64_t order_start, int64_t order_count
         if (orderkey)  orderkey[at] = o.orderkey;
         if (custkey)   custkey[at] = o.custkey;
         if (orderdate) orderdate[at] = o.orderdate;
+        if (priority)  priority[at] = o.priority;
         tpch_order_row_finished(&s);
     }
 }
@@ -158,7 +163,8 @@ extern "C" tg_status tg_tpch_gen_lineitem(tg_session* s, double sf,
                        sf, order_start, order_count, d_offsets, n_groups,
                        cols->orderkey, cols->shipdate, cols->quantity,
                        cols->extendedprice, cols->discount, cols->tax,
-                       cols->returnflag, cols->linestatus);
+                       cols->returnflag, cols->linestatus,
+                       cols->commitdate, cols->receiptdate);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     TG_HIP_CHECK(hipFree(d_offsets));
@@ -168,14 +174,18 @@ extern "C" tg_status tg_tpch_gen_lineitem(tg_session* s, double sf,
 
 /* allocate-and-generate convenience (exact sizing; used by bench/tests) */
 extern "C" tg_status tg_tpch_lineitem_alloc(tg_session* s, double sf,
-    int64_t order_start, int64_t order_count, int with_orderkey,
+    int64_t order_start, int64_t order_count, int flags,
     tg_tpch_lineitem_cols* cols)
 {
     memset(cols, 0, sizeof(*cols));
     int64_t rows = 0;
     tg_status st = tg_tpch_lineitem_rows(s, sf, order_start, order_count, &rows, nullptr);
     if (st != TG_OK) return st;
-    if (with_orderkey) TG_HIP_CHECK(hipMalloc(&cols->orderkey, rows * 8));
+    if (flags & 1) TG_HIP_CHECK(hipMalloc(&cols->orderkey, rows * 8));
+    if (flags & 2) {
+        TG_HIP_CHECK(hipMalloc(&cols->commitdate, rows * 4));
+        TG_HIP_CHECK(hipMalloc(&cols->receiptdate, rows * 4));
+    }
     TG_HIP_CHECK(hipMalloc(&cols->shipdate, rows * 4));
     TG_HIP_CHECK(hipMalloc(&cols->quantity, rows * 8));
     TG_HIP_CHECK(hipMalloc(&cols->extendedprice, rows * 8));
@@ -190,6 +200,8 @@ extern "C" tg_status tg_tpch_lineitem_free(tg_session* s, tg_tpch_lineitem_cols*
 {
     (void)s;
     if (cols->orderkey) TG_HIP_CHECK(hipFree(cols->orderkey));
+    if (cols->commitdate) TG_HIP_CHECK(hipFree(cols->commitdate));
+    if (cols->receiptdate) TG_HIP_CHECK(hipFree(cols->receiptdate));
     TG_HIP_CHECK(hipFree(cols->shipdate));
     TG_HIP_CHECK(hipFree(cols->quantity));
     TG_HIP_CHECK(hipFree(cols->extendedprice));
@@ -203,12 +215,14 @@ extern "C" tg_status tg_tpch_lineitem_free(tg_session* s, tg_tpch_lineitem_cols*
 
 extern "C" tg_status tg_tpch_gen_orders(tg_session* s, double sf,
     int64_t order_start, int64_t order_count,
-    int64_t* d_orderkey, int64_t* d_custkey, int32_t* d_orderdate)
+    int64_t* d_orderkey, int64_t* d_custkey, int32_t* d_orderdate,
+    uint8_t* d_priority)
 {
     int64_t n_groups = (order_count + GOT - 1) / GOT;
     int grid = (int)((n_groups + TG_BLOCK - 1) / TG_BLOCK);
     hipLaunchKernelGGL(k_gen_orders, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
-                       sf, order_start, order_count, d_orderkey, d_custkey, d_orderdate);
+                       sf, order_start, order_count, d_orderkey, d_custkey, d_orderdate,
+                       d_priority);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     return TG_OK;

@@ -46,7 +46,7 @@ def q3_prepare(session, sf, order_start=1, order_count=None, cust_start=1, cust_
     o_ck = _device_buffer(session, order_count * 8)
     o_od = _device_buffer(session, order_count * 4)
     _check_lib(_lib.tg_tpch_gen_orders(session._h, sf,
-                                       order_start, order_count, o_ok, o_ck, o_od))
+                                       order_start, order_count, o_ok, o_ck, o_od, None))
     li = session.tpch_lineitem(sf, lineitem_order_start, lineitem_order_count,
                                with_orderkey=True)
     return dict(cust_ck=cust_ck, cust_ms=cust_ms, cust_count=cust_count,
@@ -240,3 +240,77 @@ def q6_gpu(session, sf, order_start=1, order_count=None):
     out = pages[0]
     return dict(revenue=float(out[0]["values"][0]), rows=int(out[1]["values"][0]),
                 elapsed=elapsed)
+
+
+DATE_1993_07_01 = 8582
+DATE_1993_10_01 = 8674
+
+
+def q4_gpu(session, sf, order_start=1, order_count=None):
+    """TPC-H Q4: SELECT o_orderpriority, count(*) FROM orders
+    WHERE o_orderdate in [1993-07-01, 1993-10-01) AND EXISTS (lineitem with
+    l_commitdate < l_receiptdate) GROUP BY o_orderpriority ORDER BY it.
+    Pipeline: lineitem filter -> semi-join source build; orders filter ->
+    semi join -> filter matched -> group-by count."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    o_ok = _device_buffer(session, order_count * 8)
+    o_od = _device_buffer(session, order_count * 4)
+    o_pri = _device_buffer(session, order_count)
+    _check_lib(_lib.tg_tpch_gen_orders(session._h, sf, order_start, order_count,
+                                       o_ok, None, o_od, o_pri))
+    li = session.tpch_lineitem(sf, order_start, order_count,
+                               with_orderkey=True, with_dates=True)
+    t0 = time.time()
+    # build side: lineitem orderkeys where commitdate < receiptdate
+    lpage = ops.page_from_device(session, ([(li.orderkey, ops.TG_BIGINT),
+                                            (li.commitdate, ops.TG_INTEGER),
+                                            (li.receiptdate, ops.TG_INTEGER)],
+                                           li.row_count))
+    f1 = ops.filter_project(session, ops.expr(("col", 1), ("col", 2), "lt"),
+                            [ops.expr(("col", 0))], [ops.TG_BIGINT])
+    f1.add_input(lpage)
+    f1.finish()
+    late = _take_device_page(session, f1)
+    bridge = ops.JoinBridge(session)
+    b = ops.hash_builder(session, bridge, [ops.TG_BIGINT], [0], [])
+    b.add_input(late)
+    b.drain()
+    # probe side: orders in the date window
+    opage = ops.page_from_device(session, ([(o_ok.value, ops.TG_BIGINT),
+                                            (o_od.value, ops.TG_INTEGER),
+                                            (o_pri.value, ops.TG_TINYINT)],
+                                           order_count))
+    f2 = ops.filter_project(session,
+                            ops.expr(("col", 1), ("i64", DATE_1993_07_01), "ge",
+                                     ("col", 1), ("i64", DATE_1993_10_01), "lt", "and"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 2))],
+                            [ops.TG_BIGINT, ops.TG_TINYINT])
+    f2.add_input(opage)
+    f2.finish()
+    owin = _take_device_page(session, f2)
+    sj = ops.semi_join(session, bridge, 0)
+    sj.add_input(owin)
+    sj.finish()
+    marked = _take_device_page(session, sj)        # (orderkey, priority, matched)
+    f3 = ops.filter_project(session, ops.expr(("col", 2), ("i64", 1), "eq"),
+                            [ops.expr(("col", 1))], [ops.TG_TINYINT])
+    f3.add_input(marked)
+    f3.finish()
+    exists = _take_device_page(session, f3)
+    agg = ops.hash_aggregation(session, [0], [ops.TG_TINYINT],
+                               [(ops.AGG_COUNT_STAR, -1)])
+    agg.add_input(exists)
+    pages = agg.drain()
+    elapsed = time.time() - t0
+    for op in (f1, b, f2, sj, f3, agg):
+        op.close()
+    bridge.close()
+    session.tpch_lineitem_free(li)
+    for p in (o_ok, o_od, o_pri):
+        _device_free(session, ctypes.c_void_p(p.value) if isinstance(p, ctypes.c_void_p) else p)
+    out = pages[0]
+    pri = out[0]["values"]
+    cnt = out[1]["values"]
+    order = np.argsort(pri)
+    return dict(priority=pri[order], count=cnt[order], elapsed=elapsed)
