@@ -211,6 +211,7 @@ typedef struct tg_tpch_lineitem_cols {
     uint8_t* linestatus;    /* dictionary id: 0=F 1=O */
     int32_t* commitdate;    /* optional (flags bit 1) */
     int32_t* receiptdate;   /* optional (flags bit 1) */
+    int64_t* partkey;       /* optional (flags bit 2) */
 } tg_tpch_lineitem_cols;
 
 /* Generate lineitem rows for orders [order_start, order_start+order_count)
@@ -223,7 +224,7 @@ tg_status tg_tpch_lineitem_rows(tg_session*, double scale_factor,
     int64_t order_start, int64_t order_count, int64_t* row_count_out,
     int64_t** dev_offsets_out /* optional; hipFree */);
 /* allocate device buffers of the exact size and generate (bench/tests);
- * flags: bit0 = orderkey, bit1 = commitdate+receiptdate */
+ * flags: bit0 = orderkey, bit1 = commitdate+receiptdate, bit2 = partkey */
 tg_status tg_tpch_lineitem_alloc(tg_session*, double scale_factor,
     int64_t order_start, int64_t order_count, int flags,
     tg_tpch_lineitem_cols* out);
@@ -235,6 +236,9 @@ tg_status tg_tpch_gen_orders(tg_session*, double scale_factor,
 tg_status tg_tpch_gen_customer(tg_session*, double scale_factor,
     int64_t cust_start, int64_t cust_count,
     int64_t* dev_custkey, uint8_t* dev_mktsegment);
+tg_status tg_tpch_gen_part(tg_session*, double scale_factor,
+    int64_t part_start, int64_t part_count,
+    int64_t* dev_partkey, uint8_t* dev_type_id /* 0..149; PROMO = >=125 */);
 /* device buffer management for host pipeline drivers (pool-backed) */
 tg_status tg_device_malloc(tg_session*, void** out, int64_t bytes);
 tg_status tg_device_free(tg_session*, void* p);

@@ -65,6 +65,7 @@
 #define TPCH_SEED_L_RFLG    717419739LL
 #define TPCH_SEED_C_MSEG   1140279430LL
 #define TPCH_SEED_O_PRIO    591449447LL
+#define TPCH_SEED_P_TYPE   1841581359LL
 
 /* usage per order row for line-level streams = max lines per order */
 #define TPCH_LINES_PER_ORDER_MAX 7
@@ -120,6 +121,23 @@ TPCH_HD static inline int64_t tpch_make_order_key(int64_t order_index)
     ok <<= 3;
     ok += low_bits;
     return ok;
+}
+
+/* part p_type: pick_str over the 150 'types' strings (6 x 5 x 5 syllables,
+ * id = (s1-1)*25 + (s2-1)*5 + s3, 1-based; PROMO = s1 6 => ids 126..150).
+ * Seed pinned by canonical SF1 parts 1-4 (types 135, 98, 18, 38). */
+TPCH_HD static inline void tpch_gen_part(int64_t part_start, int64_t part_count,
+                                         int64_t* partkey, uint8_t* type_id /* 0..149 */)
+{
+    tpch_rng ty;
+    tpch_rng_init(&ty, TPCH_SEED_P_TYPE, 1);
+    tpch_rng_skip(&ty, part_start - 1);
+    for (int64_t i = 0; i < part_count; i++) {
+        if (partkey) partkey[i] = part_start + i;
+        int64_t d = tpch_rng_int(&ty, 1, 150);
+        if (type_id) type_id[i] = (uint8_t)(d - 1);
+        tpch_rng_row_finished(&ty);
+    }
 }
 
 /* dbgen rpb_routine: part retail price in cents */

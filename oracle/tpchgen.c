@@ -107,3 +107,11 @@ EXPORT int64_t tpch_gen_customer(double sf, int64_t cust_start, int64_t cust_cou
     }
     return cust_count;
 }
+
+EXPORT int64_t tpch_gen_part_cols(double sf, int64_t part_start, int64_t part_count,
+    int64_t* partkey, uint8_t* type_id)
+{
+    (void)sf;
+    tpch_gen_part(part_start, part_count, partkey, type_id);
+    return part_count;
+}

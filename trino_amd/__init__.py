@@ -48,6 +48,7 @@ class LineitemCols(ctypes.Structure):
         ("linestatus", ctypes.c_void_p),
         ("commitdate", ctypes.c_void_p),
         ("receiptdate", ctypes.c_void_p),
+        ("partkey", ctypes.c_void_p),
     ]
 
 
@@ -107,11 +108,12 @@ class Session:
 
     # --- TPC-H device generator (bench/test input infrastructure) ---
     def tpch_lineitem(self, sf, order_start=1, order_count=None, with_orderkey=False,
-                      with_dates=False):
+                      with_dates=False, with_partkey=False):
         if order_count is None:
             order_count = int(1_500_000 * sf)
         cols = LineitemCols()
-        flags = (1 if with_orderkey else 0) | (2 if with_dates else 0)
+        flags = ((1 if with_orderkey else 0) | (2 if with_dates else 0) |
+                 (4 if with_partkey else 0))
         _check(_lib.tg_tpch_lineitem_alloc(self._h, sf, order_start, order_count,
                                            flags, ctypes.byref(cols)))
         return cols
@@ -162,3 +164,8 @@ def lineitem_to_host(session, cols):
 
 _lib.tg_copy_htod.restype = ctypes.c_int
 _lib.tg_copy_htod.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64]
+
+
+_lib.tg_tpch_gen_part.restype = ctypes.c_int
+_lib.tg_tpch_gen_part.argtypes = [ctypes.c_void_p, ctypes.c_double, ctypes.c_int64,
+                                  ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p]

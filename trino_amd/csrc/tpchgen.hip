@@ -38,7 +38,8 @@ __global__ void k_gen_lineitem(double sf, int64_t order_start, int64_t order_cou
                                double* quantity, double* extendedprice,
                                double* discount, double* tax,
                                uint8_t* returnflag, uint8_t* linestatus,
-                               int32_t* commitdate, int32_t* receiptdate)
+                               int32_t* commitdate, int32_t* receiptdate,
+                               int64_t* partkey)
 {
     int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (g >= n_groups) return;
@@ -64,6 +65,7 @@ __global__ void k_gen_lineitem(double sf, int64_t order_start, int64_t order_cou
             if (linestatus)    linestatus[n] = l.linestatus;
             if (commitdate)    commitdate[n] = l.commitdate;
             if (receiptdate)   receiptdate[n] = l.receiptdate;
+            if (partkey)       partkey[n] = l.partkey;
         }
         tpch_order_row_finished(&s);
     }
@@ -91,6 +93,18 @@ __global__ void k_gen_orders(double sf, int64_t order_start, int64_t order_count
         if (priority)  priority[at] = o.priority;
         tpch_order_row_finished(&s);
     }
+}
+
+__global__ void k_gen_part(int64_t part_start, int64_t part_count,
+                           int64_t* partkey, uint8_t* type_id)
+{
+    int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t n_groups = (part_count + GOT - 1) / GOT;
+    if (g >= n_groups) return;
+    int64_t first = part_start + g * GOT;
+    int64_t cnt = min((int64_t)GOT, part_start + part_count - first);
+    tpch_gen_part(first, cnt, partkey ? partkey + (first - part_start) : nullptr,
+                  type_id ? type_id + (first - part_start) : nullptr);
 }
 
 __global__ void k_gen_customer(double sf, int64_t cust_start, int64_t cust_count,
@@ -164,7 +178,7 @@ extern "C" tg_status tg_tpch_gen_lineitem(tg_session* s, double sf,
                        cols->orderkey, cols->shipdate, cols->quantity,
                        cols->extendedprice, cols->discount, cols->tax,
                        cols->returnflag, cols->linestatus,
-                       cols->commitdate, cols->receiptdate);
+                       cols->commitdate, cols->receiptdate, cols->partkey);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     TG_HIP_CHECK(hipFree(d_offsets));
@@ -186,6 +200,7 @@ extern "C" tg_status tg_tpch_lineitem_alloc(tg_session* s, double sf,
         TG_HIP_CHECK(hipMalloc(&cols->commitdate, rows * 4));
         TG_HIP_CHECK(hipMalloc(&cols->receiptdate, rows * 4));
     }
+    if (flags & 4) TG_HIP_CHECK(hipMalloc(&cols->partkey, rows * 8));
     TG_HIP_CHECK(hipMalloc(&cols->shipdate, rows * 4));
     TG_HIP_CHECK(hipMalloc(&cols->quantity, rows * 8));
     TG_HIP_CHECK(hipMalloc(&cols->extendedprice, rows * 8));
@@ -202,6 +217,7 @@ extern "C" tg_status tg_tpch_lineitem_free(tg_session* s, tg_tpch_lineitem_cols*
     if (cols->orderkey) TG_HIP_CHECK(hipFree(cols->orderkey));
     if (cols->commitdate) TG_HIP_CHECK(hipFree(cols->commitdate));
     if (cols->receiptdate) TG_HIP_CHECK(hipFree(cols->receiptdate));
+    if (cols->partkey) TG_HIP_CHECK(hipFree(cols->partkey));
     TG_HIP_CHECK(hipFree(cols->shipdate));
     TG_HIP_CHECK(hipFree(cols->quantity));
     TG_HIP_CHECK(hipFree(cols->extendedprice));
@@ -235,6 +251,19 @@ extern "C" tg_status tg_tpch_gen_customer(tg_session* s, double sf,
     int grid = (int)((n_groups + TG_BLOCK - 1) / TG_BLOCK);
     hipLaunchKernelGGL(k_gen_customer, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
                        sf, cust_start, cust_count, d_custkey, d_mktsegment);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}
+
+extern "C" tg_status tg_tpch_gen_part(tg_session* s, double sf,
+    int64_t part_start, int64_t part_count, int64_t* d_partkey, uint8_t* d_type)
+{
+    (void)sf;
+    int64_t n_groups = (part_count + GOT - 1) / GOT;
+    int grid = (int)((n_groups + TG_BLOCK - 1) / TG_BLOCK);
+    hipLaunchKernelGGL(k_gen_part, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                       part_start, part_count, d_partkey, d_type);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     return TG_OK;

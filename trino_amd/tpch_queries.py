@@ -314,3 +314,66 @@ def q4_gpu(session, sf, order_start=1, order_count=None):
     cnt = out[1]["values"]
     order = np.argsort(pri)
     return dict(priority=pri[order], count=cnt[order], elapsed=elapsed)
+
+
+DATE_1995_09_01 = 9374
+DATE_1995_10_01 = 9404
+
+
+def q14_gpu(session, sf, order_start=1, order_count=None):
+    """TPC-H Q14: promo revenue percent over the Sep-1995 shipdate window.
+    Pipeline: part build (key partkey -> p_type) ; lineitem filter+project ->
+    join -> scalar sums (promo CASE via a type-range filter, and total)."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    n_parts = int(200_000 * sf)
+    p_pk = _device_buffer(session, n_parts * 8)
+    p_ty = _device_buffer(session, n_parts)
+    _check_lib(_lib.tg_tpch_gen_part(session._h, sf, 1, n_parts, p_pk, p_ty))
+    li = session.tpch_lineitem(sf, order_start, order_count, with_partkey=True)
+    t0 = time.time()
+    ppage = ops.page_from_device(session, ([(p_pk.value, ops.TG_BIGINT),
+                                            (p_ty.value, ops.TG_TINYINT)], n_parts))
+    bridge = ops.JoinBridge(session)
+    b = ops.hash_builder(session, bridge, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b.add_input(ppage)
+    b.drain()
+    lpage = ops.page_from_device(session, ([(li.partkey, ops.TG_BIGINT),
+                                            (li.shipdate, ops.TG_INTEGER),
+                                            (li.extendedprice, ops.TG_DOUBLE),
+                                            (li.discount, ops.TG_DOUBLE)],
+                                           li.row_count))
+    f = ops.filter_project(session,
+                           ops.expr(("col", 1), ("i64", DATE_1995_09_01), "ge",
+                                    ("col", 1), ("i64", DATE_1995_10_01), "lt", "and"),
+                           [ops.expr(("col", 0)),
+                            ops.expr(("col", 2), ("f64", 1.0), ("col", 3), "sub", "mul")],
+                           [ops.TG_BIGINT, ops.TG_DOUBLE])
+    f.add_input(lpage)
+    f.finish()
+    sel = _take_device_page(session, f)
+    j = ops.lookup_join(session, bridge, [ops.TG_BIGINT, ops.TG_DOUBLE], [0], [1])
+    j.add_input(sel)
+    j.finish()
+    joined = _take_device_page(session, j)     # (discprice, p_type)
+    # total revenue
+    a1 = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_F64, 0)])
+    a1.add_input(joined)
+    total = a1.drain()[0][0]["values"][0]
+    # promo (CASE WHEN p_type LIKE 'PROMO%'): type ids 125..149
+    f2 = ops.filter_project(session, ops.expr(("col", 1), ("i64", 125), "ge"),
+                            [ops.expr(("col", 0))], [ops.TG_DOUBLE])
+    f2.add_input(joined)
+    f2.finish()
+    promo_page = _take_device_page(session, f2)
+    a2 = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_F64, 0)])
+    a2.add_input(promo_page)
+    promo = a2.drain()[0][0]["values"][0]
+    elapsed = time.time() - t0
+    for op in (b, f, j, a1, f2, a2):
+        op.close()
+    bridge.close()
+    session.tpch_lineitem_free(li)
+    for p in (p_pk, p_ty):
+        _device_free(session, p)
+    return dict(promo_revenue=100.0 * promo / total, elapsed=elapsed)
