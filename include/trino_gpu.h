@@ -238,7 +238,8 @@ tg_status tg_tpch_gen_customer(tg_session*, double scale_factor,
     int64_t* dev_custkey, uint8_t* dev_mktsegment);
 tg_status tg_tpch_gen_part(tg_session*, double scale_factor,
     int64_t part_start, int64_t part_count,
-    int64_t* dev_partkey, uint8_t* dev_type_id /* 0..149; PROMO = >=125 */);
+    int64_t* dev_partkey, int16_t* dev_type_id /* 0..149 (SMALLINT: 150 ids
+    overflow signed TINYINT); PROMO = >=125 */);
 /* device buffer management for host pipeline drivers (pool-backed) */
 tg_status tg_device_malloc(tg_session*, void** out, int64_t bytes);
 tg_status tg_device_free(tg_session*, void* p);

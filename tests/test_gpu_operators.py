@@ -478,3 +478,50 @@ class TestEncodedInputBlocks:
         e = ops.expr(("col", 0), ("f64", 7.0), "gt")
         got = ops.filter_run(sess, e, page)
         assert np.array_equal(got, np.arange(n, dtype=np.int32))
+
+
+class TestOperatorStateMachine:
+    """Operator.java:18-50 contract violations surface as errors, not UB."""
+
+    def test_add_input_after_finish(self, sess, ops):
+        import trino_amd
+        op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT], [(ops.AGG_COUNT_STAR, -1)])
+        page = ops.page_from_numpy([np.arange(4, dtype=np.int64)])
+        op.add_input(page)
+        op.finish()
+        with pytest.raises(trino_amd.TrinoGpuError):
+            op.add_input(page)
+        op.close()
+
+    def test_probe_before_build_errors(self, sess, ops):
+        import trino_amd
+        bridge = ops.JoinBridge(sess)
+        probe = ops.lookup_join(sess, bridge, [ops.TG_BIGINT], [0], [0])
+        with pytest.raises(trino_amd.TrinoGpuError):
+            probe.add_input(ops.page_from_numpy([np.arange(4, dtype=np.int64)]))
+        probe.close()
+        bridge.close()
+
+    def test_varchar_value_block_rejected(self, sess, ops):
+        import trino_amd
+        blocks = (ops.TgBlock * 1)()
+        blocks[0].type = ops.TG_VARCHAR
+        blocks[0].kind = 0
+        blocks[0].position_count = 1
+        page = ops.TgPage()
+        page.channel_count = 1
+        page.position_count = 1
+        page.blocks = blocks
+        op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT], [(ops.AGG_COUNT_STAR, -1)])
+        with pytest.raises(trino_amd.TrinoGpuError) as ei:
+            op.add_input(page)
+        assert "dictionary-encoded" in str(ei.value)
+        op.close()
+
+    def test_empty_page_through_pipeline(self, sess, ops):
+        op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                  [(ops.AGG_COUNT_STAR, -1)])
+        op.add_input(ops.page_from_numpy([np.empty(0, np.int64)]))
+        out = op.drain()[0]
+        op.close()
+        assert len(out[0]["values"]) == 0

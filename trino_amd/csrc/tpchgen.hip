@@ -96,15 +96,18 @@ __global__ void k_gen_orders(double sf, int64_t order_start, int64_t order_count
 }
 
 __global__ void k_gen_part(int64_t part_start, int64_t part_count,
-                           int64_t* partkey, uint8_t* type_id)
+                           int64_t* partkey, int16_t* type_id)
 {
     int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t n_groups = (part_count + GOT - 1) / GOT;
     if (g >= n_groups) return;
     int64_t first = part_start + g * GOT;
     int64_t cnt = min((int64_t)GOT, part_start + part_count - first);
+    uint8_t tmp[GOT];
     tpch_gen_part(first, cnt, partkey ? partkey + (first - part_start) : nullptr,
-                  type_id ? type_id + (first - part_start) : nullptr);
+                  type_id ? tmp : nullptr);
+    if (type_id)
+        for (int64_t i = 0; i < cnt; i++) type_id[first - part_start + i] = tmp[i];
 }
 
 __global__ void k_gen_customer(double sf, int64_t cust_start, int64_t cust_count,
@@ -257,7 +260,7 @@ extern "C" tg_status tg_tpch_gen_customer(tg_session* s, double sf,
 }
 
 extern "C" tg_status tg_tpch_gen_part(tg_session* s, double sf,
-    int64_t part_start, int64_t part_count, int64_t* d_partkey, uint8_t* d_type)
+    int64_t part_start, int64_t part_count, int64_t* d_partkey, int16_t* d_type)
 {
     (void)sf;
     int64_t n_groups = (part_count + GOT - 1) / GOT;

@@ -328,14 +328,14 @@ def q14_gpu(session, sf, order_start=1, order_count=None):
         order_count = int(1_500_000 * sf)
     n_parts = int(200_000 * sf)
     p_pk = _device_buffer(session, n_parts * 8)
-    p_ty = _device_buffer(session, n_parts)
+    p_ty = _device_buffer(session, n_parts * 2)   # SMALLINT: ids 0..149
     _check_lib(_lib.tg_tpch_gen_part(session._h, sf, 1, n_parts, p_pk, p_ty))
     li = session.tpch_lineitem(sf, order_start, order_count, with_partkey=True)
     t0 = time.time()
     ppage = ops.page_from_device(session, ([(p_pk.value, ops.TG_BIGINT),
-                                            (p_ty.value, ops.TG_TINYINT)], n_parts))
+                                            (p_ty.value, ops.TG_SMALLINT)], n_parts))
     bridge = ops.JoinBridge(session)
-    b = ops.hash_builder(session, bridge, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b = ops.hash_builder(session, bridge, [ops.TG_BIGINT, ops.TG_SMALLINT], [0], [1])
     b.add_input(ppage)
     b.drain()
     lpage = ops.page_from_device(session, ([(li.partkey, ops.TG_BIGINT),
