@@ -240,6 +240,10 @@ tg_status tg_tpch_gen_part(tg_session*, double scale_factor,
     int64_t part_start, int64_t part_count,
     int64_t* dev_partkey, int16_t* dev_type_id /* 0..149 (SMALLINT: 150 ids
     overflow signed TINYINT); PROMO = >=125 */);
+/* coarse device-memory accounting (LocalMemoryContext analog): bytes ever
+ * pooled and bytes currently cached; live = total - cached */
+tg_status tg_session_memory(tg_session*, int64_t* total_bytes, int64_t* cached_bytes);
+
 /* device buffer management for host pipeline drivers (pool-backed) */
 tg_status tg_device_malloc(tg_session*, void** out, int64_t bytes);
 tg_status tg_device_free(tg_session*, void* p);

@@ -169,3 +169,15 @@ _lib.tg_copy_htod.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
 _lib.tg_tpch_gen_part.restype = ctypes.c_int
 _lib.tg_tpch_gen_part.argtypes = [ctypes.c_void_p, ctypes.c_double, ctypes.c_int64,
                                   ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p]
+
+
+_lib.tg_session_memory.restype = ctypes.c_int
+_lib.tg_session_memory.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p]
+
+
+def session_memory(session):
+    """(total_pooled_bytes, cached_bytes) — live device use = total - cached."""
+    tot = ctypes.c_int64()
+    cach = ctypes.c_int64()
+    _check(_lib.tg_session_memory(session._h, ctypes.byref(tot), ctypes.byref(cach)))
+    return tot.value, cach.value

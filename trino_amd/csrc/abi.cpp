@@ -118,3 +118,17 @@ extern "C" tg_status tg_copy_htod(tg_session* s, void* dst_dev, const void* src_
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     return TG_OK;
 }
+
+/* coarse device-memory accounting (LocalMemoryContext.setBytes analog,
+ * lib/trino-memory-context): total bytes ever pooled, and bytes currently
+ * cached (free) in the pool. live = total - cached. */
+extern "C" tg_status tg_session_memory(tg_session* s, int64_t* total_bytes,
+                                       int64_t* cached_bytes)
+{
+    if (!s) { TG_SET_ERR("null session"); return TG_ERR_INVALID_ARG; }
+    int64_t cached = 0;
+    for (auto& kv : s->pool_free) cached += (int64_t)kv.first;
+    if (total_bytes) *total_bytes = (int64_t)s->pool_bytes;
+    if (cached_bytes) *cached_bytes = cached;
+    return TG_OK;
+}
