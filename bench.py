@@ -284,7 +284,11 @@ def run_q3(args, sess, n_gpus, rank):
         "roofline": {"bound": "hbm", "achieved": rows * 28 / (ms / 1000) / 1e9,
                      "peak": HBM_PEAK_GBPS, "unit": "GB/s",
                      "frac": rows * 28 / (ms / 1000) / 1e9 / HBM_PEAK_GBPS,
-                     "traffic": None},
+                     "traffic": None,
+                     "note": "whole-pipeline rate against the single-pass 28 B/row "
+                             "probe-side bound (SURVEY §8d); the pipeline makes "
+                             "multiple passes and its builds/probes are random-access "
+                             "bound — per-kernel evidence: profiles/r01_q3_kernels_v2.txt"},
         "cpu_baseline": None,
     }
     if rank == 0:

@@ -308,7 +308,7 @@ def q4_gpu(session, sf, order_start=1, order_count=None):
     bridge.close()
     session.tpch_lineitem_free(li)
     for p in (o_ok, o_od, o_pri):
-        _device_free(session, ctypes.c_void_p(p.value) if isinstance(p, ctypes.c_void_p) else p)
+        _device_free(session, p)
     out = pages[0]
     pri = out[0]["values"]
     cnt = out[1]["values"]
