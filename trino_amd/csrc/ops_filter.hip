@@ -162,6 +162,90 @@ __global__ void k_compact(const uint8_t* __restrict__ flags, int64_t n,
     }
 }
 
+/* ---- fused filter+project (the ScanFilterAndProjectOperator shape:
+ * PageProcessor evaluates filter then projections over the selected
+ * positions in one operator; here two passes with no materialized selection
+ * vector: pass 1 counts survivors per chunk, pass 2 re-evaluates the
+ * predicate and writes every projection at the scanned offsets) ---- */
+struct KProj {
+    const tg_expr_inst* insts;
+    int32_t count;
+    int32_t identity_col;   /* >=0: typed copy of that column */
+    int32_t out_type;
+    void* out;
+    uint64_t* out_valid;
+};
+#define MAX_PROJ 8
+
+__global__ void k_fp_count(const tg_expr_inst* prog, int count, const KCol* cols,
+                           int64_t n, int32_t* __restrict__ chunk_counts, int64_t nchunks)
+{
+    /* one wave per chunk */
+    int64_t c = (int64_t)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+    if (c >= nchunks) return;
+    int lane = threadIdx.x % 64;
+    int64_t lo = c * CHUNK, hi = min(lo + CHUNK, n);
+    int32_t cnt = 0;
+    for (int64_t i = lo + lane; i < hi; i += 64) {
+        bool isnull = false;
+        double v = eval_expr(prog, count, cols, i, &isnull);
+        cnt += (!isnull && v != 0.0) ? 1 : 0;
+    }
+    #pragma unroll
+    for (int off = 32; off >= 1; off >>= 1) cnt += __shfl_xor(cnt, off, 64);
+    if (lane == 0) chunk_counts[c] = cnt;
+}
+
+__global__ void k_fp_write(const tg_expr_inst* prog, int count, const KCol* cols,
+                           int64_t n, const int32_t* __restrict__ chunk_offsets,
+                           const KProj* projs, int nproj, int64_t nchunks)
+{
+    int64_t c = (int64_t)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+    if (c >= nchunks) return;
+    int lane = threadIdx.x % 64;
+    int64_t lo = c * CHUNK, hi = min(lo + CHUNK, n);
+    int32_t run = chunk_offsets[c];
+    for (int64_t g = lo; g < hi; g += 64) {
+        int64_t i = g + lane;
+        bool sel = false;
+        if (i < hi) {
+            bool isnull = false;
+            double v = eval_expr(prog, count, cols, i, &isnull);
+            sel = !isnull && v != 0.0;
+        }
+        unsigned long long b = __ballot(sel);
+        int before = __popcll(b & ((1ull << lane) - 1ull));
+        if (sel) {
+            int64_t at = run + before;
+            for (int p = 0; p < nproj; p++) {
+                const KProj& pr = projs[p];
+                if (pr.identity_col >= 0) {
+                    const KCol& sc = cols[pr.identity_col];
+                    switch (pr.out_type) {
+                        case TG_BIGINT: ((int64_t*)pr.out)[at] = ((const int64_t*)sc.data)[i]; break;
+                        case TG_INTEGER: case TG_DATE: ((int32_t*)pr.out)[at] = ((const int32_t*)sc.data)[i]; break;
+                        case TG_SMALLINT: ((int16_t*)pr.out)[at] = ((const int16_t*)sc.data)[i]; break;
+                        case TG_TINYINT: case TG_BOOLEAN: ((int8_t*)pr.out)[at] = ((const int8_t*)sc.data)[i]; break;
+                        default: ((double*)pr.out)[at] = ((const double*)sc.data)[i]; break;
+                    }
+                    if (pr.out_valid && sc.valid && !((sc.valid[i >> 6] >> (i & 63)) & 1))
+                        atomicAnd((unsigned long long*)&pr.out_valid[at >> 6],
+                                  ~(1ull << (at & 63)));
+                }
+                else {
+                    bool pnull = false;
+                    double pv = eval_expr(pr.insts, pr.count, cols, i, &pnull);
+                    ((double*)pr.out)[at] = pnull ? 0.0 : pv;
+                    if (pr.out_valid && pnull)
+                        atomicAnd((unsigned long long*)&pr.out_valid[at >> 6],
+                                  ~(1ull << (at & 63)));
+                }
+            }
+        }
+        run += __popcll(b);
+    }
+}
+
 tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page,
                      const tg_selected* input_sel,
                      int32_t** d_positions_out, int32_t* count_out)
@@ -335,23 +419,85 @@ struct FilterProjectOp : tg_operator {
         DevPage in;
         tg_status st = tg_upload_page(s, page, &in);
         if (st != TG_OK) return st;
-        int32_t* d_pos = nullptr;
-        int32_t count = (int32_t)in.n;
-        if (has_filter) {
-            st = run_filter(s, filter, in, nullptr, &d_pos, &count);
-            if (st != TG_OK) { tg_free_page(s, &in); return st; }
+        if (!has_filter) {
+            /* pure projection: evaluate every position */
+            DevPage outp;
+            outp.n = in.n;
+            outp.blocks.resize(projections.size());
+            for (size_t p = 0; p < projections.size(); p++) {
+                st = run_project(s, projections[p], out_types[p], in, nullptr,
+                                 (int32_t)in.n, &outp.blocks[p]);
+                if (st != TG_OK) { tg_free_page(s, &in); return st; }
+            }
+            tg_free_page(s, &in);
+            stage_output(std::move(outp));
+            return TG_OK;
         }
+        if (projections.size() > MAX_PROJ) { TG_SET_ERR("too many projections"); return TG_ERR_UNSUPPORTED; }
+
+        std::vector<KCol> cols(in.blocks.size());
+        for (size_t i = 0; i < in.blocks.size(); i++)
+            cols[i] = {in.blocks[i].data, in.blocks[i].valid, (int32_t)in.blocks[i].type, 0};
+        KCol* d_cols = nullptr;
+        TG_POOL_ALLOC(s, &d_cols, cols.size() * sizeof(KCol));
+        TG_HIP_CHECK(hipMemcpyAsync(d_cols, cols.data(), cols.size() * sizeof(KCol),
+                                    hipMemcpyHostToDevice, s->stream));
+
+        int64_t nchunks = (in.n + CHUNK - 1) / CHUNK;
+        if (nchunks < 1) nchunks = 1;
+        int32_t* d_counts = nullptr;
+        int32_t* d_total = nullptr;
+        TG_POOL_ALLOC(s, &d_counts, nchunks * 4);
+        TG_POOL_ALLOC(s, &d_total, 4);
+        int wpb = TG_BLOCK / 64;
+        hipLaunchKernelGGL(k_fp_count, dim3((uint32_t)((nchunks + wpb - 1) / wpb)),
+                           dim3(TG_BLOCK), 0, s->stream,
+                           filter.d_insts, filter.count, d_cols, in.n, d_counts, nchunks);
+        TG_HIP_CHECK(hipGetLastError());
+        hipLaunchKernelGGL(k_scan_serial, dim3(1), dim3(1), 0, s->stream,
+                           d_counts, nchunks, d_total);
+        TG_HIP_CHECK(hipGetLastError());
+        int32_t total = 0;
+        TG_HIP_CHECK(hipMemcpyAsync(&total, d_total, 4, hipMemcpyDeviceToHost, s->stream));
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+
         DevPage outp;
-        outp.n = count;
+        outp.n = total;
         outp.blocks.resize(projections.size());
+        std::vector<KProj> kp(projections.size());
+        bool any_null = false;
+        for (auto& b : in.blocks) any_null |= b.valid != nullptr;
         for (size_t p = 0; p < projections.size(); p++) {
-            st = run_project(s, projections[p], out_types[p], in, d_pos, count,
-                             &outp.blocks[p]);
-            if (st != TG_OK) break;
+            DevBlock& ob = outp.blocks[p];
+            bool ident = projections[p].count == 1 &&
+                         projections[p].insts[0].op == TG_EXPR_COL;
+            int src = ident ? projections[p].insts[0].arg0 : -1;
+            ob.type = ident ? in.blocks[src].type : TG_DOUBLE;
+            ob.n = total;
+            TG_POOL_ALLOC(s, &ob.data, (int64_t)(total ? total : 1) * ob.elem_size());
+            if (any_null) {
+                int64_t words = (total + 63) / 64;
+                TG_POOL_ALLOC(s, &ob.valid, (words ? words : 1) * 8);
+                TG_HIP_CHECK(hipMemsetAsync(ob.valid, 0xFF, words * 8, s->stream));
+            }
+            kp[p] = {projections[p].d_insts, projections[p].count, src,
+                     (int32_t)ob.type, ob.data, ob.valid};
         }
-        if (d_pos) tg_pool_free(s, d_pos);
+        KProj* d_kp = nullptr;
+        TG_POOL_ALLOC(s, &d_kp, kp.size() * sizeof(KProj));
+        TG_HIP_CHECK(hipMemcpyAsync(d_kp, kp.data(), kp.size() * sizeof(KProj),
+                                    hipMemcpyHostToDevice, s->stream));
+        hipLaunchKernelGGL(k_fp_write, dim3((uint32_t)((nchunks + wpb - 1) / wpb)),
+                           dim3(TG_BLOCK), 0, s->stream,
+                           filter.d_insts, filter.count, d_cols, in.n, d_counts,
+                           d_kp, (int)kp.size(), nchunks);
+        TG_HIP_CHECK(hipGetLastError());
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        tg_pool_free(s, d_cols);
+        tg_pool_free(s, d_counts);
+        tg_pool_free(s, d_total);
+        tg_pool_free(s, d_kp);
         tg_free_page(s, &in);
-        if (st != TG_OK) return st;
         stage_output(std::move(outp));
         return TG_OK;
     }

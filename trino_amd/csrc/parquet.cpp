@@ -404,11 +404,14 @@ static tg_status decompress(int codec, const uint8_t* in, size_t clen, size_t ul
                             std::vector<uint8_t>* out)
 {
     out->resize(ulen);
-    if (codec == 0 || clen == ulen) {          /* UNCOMPRESSED (or stored) */
+    if (codec == 0) {                          /* UNCOMPRESSED */
         if (clen != ulen) { TG_SET_ERR("parquet: size mismatch on uncompressed page"); return TG_ERR_INVALID_ARG; }
         memcpy(out->data(), in, ulen);
         return TG_OK;
     }
+    /* NOTE: v1 pages are always codec-compressed; compressed size equal to
+     * uncompressed size is possible and is NOT a stored page (a real snappy
+     * page with clen==ulen==3127 broke the former shortcut). */
     if (codec == 1) {                           /* SNAPPY */
         size_t got = 0;
         if (!snappy_uncompress(in, clen, out->data(), ulen, &got) || got != ulen) {
@@ -511,6 +514,10 @@ extern "C" tg_status tg_parquet_read_column(tg_session* s, tg_parquet_file* f, i
             PageHdr h;
             if (!parse_page_header(in, &h)) { TG_SET_ERR("parquet: bad page header"); return TG_ERR_INVALID_ARG; }
             const uint8_t* body = in.p;
+            if (getenv("TG_PQ_DEBUG"))
+                fprintf(stderr, "[pq] off=%lld type=%d nv=%d enc=%d comp=%d uncomp=%d remaining=%lld row=%lld\n",
+                        (long long)off, h.type, h.num_values, h.encoding,
+                        h.compressed, h.uncompressed, (long long)remaining, (long long)row);
             off = (body - f->bytes.data()) + h.compressed;
 
             if (h.type == 2) {                  /* dictionary page */
