@@ -1,0 +1,136 @@
+"""Randomized operator-pipeline property tests: for seeded random inputs and
+random plan shapes (filter -> [join] -> group-by), the operator chain must
+match an independent numpy composition. Integer outputs bit-exact; f64 sums
+rtol 1e-12 (generic-agg tolerance).
+"""
+import numpy as np
+import pytest
+
+import oracle
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def sess():
+    import trino_amd
+    s = trino_amd.Session(0)
+    yield s
+    s.close()
+
+
+@pytest.fixture(scope="module")
+def ops():
+    from trino_amd import ops
+    return ops
+
+
+@pytest.mark.parametrize("seed", range(8))
+def test_random_filter_groupby(sess, ops, seed):
+    r = np.random.default_rng(seed)
+    n = int(r.integers(1, 60_000))
+    nkeys = int(r.integers(1, 200))
+    keys = r.integers(0, nkeys, n).astype(np.int64)
+    vals_f = r.uniform(-1000, 1000, n)
+    vals_i = r.integers(-10**6, 10**6, n).astype(np.int64)
+    thresh = int(r.integers(0, nkeys))
+    page = ops.page_from_numpy([keys, vals_f, vals_i])
+
+    f = ops.expr(("col", 0), ("i64", thresh), "ge")
+    fp = ops.filter_project(sess, f,
+                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                             ops.expr(("col", 2))],
+                            [ops.TG_BIGINT, ops.TG_DOUBLE, ops.TG_BIGINT])
+    fp.add_input(page)
+    fp.finish()
+    sel_page, fin = fp.get_output()
+    # re-upload host copy into the agg (also exercises download path)
+    agg = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                               [(ops.AGG_COUNT_STAR, -1), (ops.AGG_SUM_F64, 1),
+                                (ops.AGG_SUM_I64, 2)])
+    if sel_page is not None:
+        agg.add_input(ops.page_from_numpy([sel_page[0]["values"],
+                                           sel_page[1]["values"],
+                                           sel_page[2]["values"]]))
+    out = agg.drain()
+    fp.close()
+    agg.close()
+
+    mask = keys >= thresh
+    ek, ev, ei = keys[mask], vals_f[mask], vals_i[mask]
+    if len(ek) == 0:
+        assert not out or len(out[0][0]["values"]) == 0
+        return
+    gids, ng, by_gid, _ = oracle.bigint_groupby(ek)
+    o = out[0]
+    assert np.array_equal(o[0]["values"], by_gid)
+    assert np.array_equal(o[1]["values"], oracle.grouped_count(gids, ng))
+    es = np.zeros(ng)
+    np.add.at(es, gids, ev)
+    np.testing.assert_allclose(o[2]["values"], es, rtol=1e-12, atol=1e-9)
+    eis = np.zeros(ng, np.int64)
+    np.add.at(eis, gids, ei)
+    assert np.array_equal(o[3]["values"], eis)
+
+
+@pytest.mark.parametrize("seed", range(6))
+def test_random_join_agg(sess, ops, seed):
+    r = np.random.default_rng(100 + seed)
+    nb = int(r.integers(1, 20_000))
+    m = int(r.integers(1, 80_000))
+    dup = int(r.integers(1, 4))
+    bk = np.repeat(r.choice(np.arange(10**6), nb // dup + 1, replace=False), dup)[:nb].astype(np.int64)
+    r.shuffle(bk)
+    bv = r.integers(0, 10**6, nb).astype(np.int64)
+    pk = r.choice(np.concatenate([bk, r.integers(10**7, 10**8, max(m // 3, 1))]), m).astype(np.int64)
+    pv = r.uniform(0, 100, m)
+
+    bridge = ops.JoinBridge(sess)
+    b = ops.hash_builder(sess, bridge, [ops.TG_BIGINT, ops.TG_BIGINT], [0], [1])
+    b.add_input(ops.page_from_numpy([bk, bv]))
+    b.drain()
+    j = ops.lookup_join(sess, bridge, [ops.TG_BIGINT, ops.TG_DOUBLE], [0], [0, 1])
+    j.add_input(ops.page_from_numpy([pk, pv]))
+    jp, _ = j.get_output()
+    b.close()
+
+    t = oracle.JoinTable(bk)
+    op_, ob_ = t.probe(pk, cap=8 * m + 64)
+    if jp is None:
+        assert len(op_) == 0
+        j.close()
+        bridge.close()
+        return
+    got = sorted(zip(jp[0]["values"].tolist(), jp[1]["values"].tolist(),
+                     jp[2]["values"].tolist()))
+    exp = sorted(zip(pk[op_].tolist(), pv[op_].tolist(), bv[ob_].tolist()))
+    assert got == exp
+    j.close()
+    bridge.close()
+
+
+@pytest.mark.parametrize("seed", range(4))
+def test_random_partition_roundtrip(sess, ops, seed):
+    r = np.random.default_rng(200 + seed)
+    n = int(r.integers(1, 120_000))
+    nparts = int(r.choice([2, 3, 8, 17]))
+    keys = r.integers(-2**62, 2**62, n).astype(np.int64)
+    vals = r.standard_normal(n)
+    op = ops.page_partitioner(sess, [ops.TG_BIGINT, ops.TG_DOUBLE], [0], nparts)
+    op.add_input(ops.page_from_numpy([keys, vals]))
+    op.finish()
+    hashes = oracle.hash_rows([keys], [oracle.TG_BIGINT])
+    exp_pid = np.array([oracle.partition_remote(int(np.int64(h)), nparts) for h in hashes],
+                       np.int32)
+    seen = 0
+    for p in range(nparts):
+        cols = ops.get_partition(sess, op, p)
+        rows = np.nonzero(exp_pid == p)[0]
+        if cols is None:
+            assert len(rows) == 0
+            continue
+        assert np.array_equal(cols[0]["values"], keys[rows])
+        assert np.array_equal(cols[1]["values"], vals[rows])
+        seen += len(rows)
+    assert seen == n
+    op.close()

@@ -13,6 +13,7 @@
  *    interpreter over the tg_expr IR (stack depth <= 4, checked at compile).
  */
 #include "operators.h"
+#include <cstdlib>
 
 struct KCol { const void* data; const uint64_t* valid; int32_t type; int32_t _pad; };
 
@@ -416,9 +417,30 @@ struct FilterProjectOp : tg_operator {
 
     tg_status add_input(const tg_page* page) override
     {
+        static int fused = [] { const char* e = getenv("TG_FP_FUSED"); return e ? atoi(e) : 1; }();
         DevPage in;
         tg_status st = tg_upload_page(s, page, &in);
         if (st != TG_OK) return st;
+        if (has_filter && !fused) {
+            /* unfused reference path: selection vector + per-projection gather */
+            int32_t* d_pos = nullptr;
+            int32_t count = 0;
+            st = run_filter(s, filter, in, nullptr, &d_pos, &count);
+            if (st != TG_OK) { tg_free_page(s, &in); return st; }
+            DevPage outp;
+            outp.n = count;
+            outp.blocks.resize(projections.size());
+            for (size_t p = 0; p < projections.size(); p++) {
+                st = run_project(s, projections[p], out_types[p], in, d_pos, count,
+                                 &outp.blocks[p]);
+                if (st != TG_OK) break;
+            }
+            if (d_pos) tg_pool_free(s, d_pos);
+            tg_free_page(s, &in);
+            if (st != TG_OK) return st;
+            stage_output(std::move(outp));
+            return TG_OK;
+        }
         if (!has_filter) {
             /* pure projection: evaluate every position */
             DevPage outp;
