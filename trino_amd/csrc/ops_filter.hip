@@ -417,7 +417,13 @@ struct FilterProjectOp : tg_operator {
 
     tg_status add_input(const tg_page* page) override
     {
-        static int fused = [] { const char* e = getenv("TG_FP_FUSED"); return e ? atoi(e) : 1; }();
+        /* A/B on MI355X (Q3 SF100, same box, 2 runs each): fused two-pass
+         * 128.4 ms/step vs selection-vector path 110.5 ms — the wave-per-chunk
+         * write pass starves parallelism (37k waves doing serial 64-wide
+         * evaluation) against grid-stride flag+gather kernels. Default stays
+         * unfused; the fused path needs a block-parallel write design
+         *  before it can win. */
+        static int fused = [] { const char* e = getenv("TG_FP_FUSED"); return e ? atoi(e) : 0; }();
         DevPage in;
         tg_status st = tg_upload_page(s, page, &in);
         if (st != TG_OK) return st;
