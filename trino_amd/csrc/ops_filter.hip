@@ -181,33 +181,45 @@ struct KProj {
 __global__ void k_fp_count(const tg_expr_inst* prog, int count, const KCol* cols,
                            int64_t n, int32_t* __restrict__ chunk_counts, int64_t nchunks)
 {
-    /* one wave per chunk */
-    int64_t c = (int64_t)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+    /* one BLOCK per chunk; thread-per-row evaluation, block-reduced count */
+    int64_t c = blockIdx.x;
     if (c >= nchunks) return;
-    int lane = threadIdx.x % 64;
     int64_t lo = c * CHUNK, hi = min(lo + CHUNK, n);
     int32_t cnt = 0;
-    for (int64_t i = lo + lane; i < hi; i += 64) {
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
         bool isnull = false;
         double v = eval_expr(prog, count, cols, i, &isnull);
         cnt += (!isnull && v != 0.0) ? 1 : 0;
     }
     #pragma unroll
     for (int off = 32; off >= 1; off >>= 1) cnt += __shfl_xor(cnt, off, 64);
-    if (lane == 0) chunk_counts[c] = cnt;
+    __shared__ int32_t wsum[TG_BLOCK / 64];
+    if (threadIdx.x % 64 == 0) wsum[threadIdx.x / 64] = cnt;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int32_t t = 0;
+        for (int w = 0; w < TG_BLOCK / 64; w++) t += wsum[w];
+        chunk_counts[c] = t;
+    }
 }
 
 __global__ void k_fp_write(const tg_expr_inst* prog, int count, const KCol* cols,
                            int64_t n, const int32_t* __restrict__ chunk_offsets,
                            const KProj* projs, int nproj, int64_t nchunks)
 {
-    int64_t c = (int64_t)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+    /* one BLOCK per chunk: flags staged as an LDS bitmask, per-wave
+     * sub-range offsets, then 64-wide selected-only projection evaluation */
+    int64_t c = blockIdx.x;
     if (c >= nchunks) return;
-    int lane = threadIdx.x % 64;
+    constexpr int WAVES = TG_BLOCK / 64;
+    constexpr int GROUPS = CHUNK / 64;          /* 64-row groups per chunk */
+    __shared__ unsigned long long mask[GROUPS];
+    __shared__ int32_t wave_base[WAVES];
     int64_t lo = c * CHUNK, hi = min(lo + CHUNK, n);
-    int32_t run = chunk_offsets[c];
-    for (int64_t g = lo; g < hi; g += 64) {
-        int64_t i = g + lane;
+    int wave = threadIdx.x / 64, lane = threadIdx.x % 64;
+    /* phase 1: evaluate predicate, ballot into the LDS mask */
+    for (int g = wave; g < GROUPS; g += WAVES) {
+        int64_t i = lo + (int64_t)g * 64 + lane;
         bool sel = false;
         if (i < hi) {
             bool isnull = false;
@@ -215,8 +227,37 @@ __global__ void k_fp_write(const tg_expr_inst* prog, int count, const KCol* cols
             sel = !isnull && v != 0.0;
         }
         unsigned long long b = __ballot(sel);
-        int before = __popcll(b & ((1ull << lane) - 1ull));
+        if (lane == 0) mask[g] = b;
+    }
+    __syncthreads();
+    /* phase 2: per-wave quarter counts -> exclusive wave bases */
+    constexpr int GPW = GROUPS / WAVES;         /* groups per wave */
+    {
+        int32_t cnt = 0;
+        for (int g = wave * GPW + lane; g < (wave + 1) * GPW; g += 64)
+            cnt += __popcll(mask[g]);
+        #pragma unroll
+        for (int off = 32; off >= 1; off >>= 1) cnt += __shfl_xor(cnt, off, 64);
+        if (lane == 0) wave_base[wave] = cnt;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        int32_t run = 0;
+        for (int w = 0; w < WAVES; w++) {
+            int32_t v = wave_base[w];
+            wave_base[w] = run;
+            run += v;
+        }
+    }
+    __syncthreads();
+    /* phase 3: selected lanes evaluate projections and write compacted */
+    int32_t run = chunk_offsets[c] + wave_base[wave];
+    for (int g = wave * GPW; g < (wave + 1) * GPW; g++) {
+        unsigned long long b = mask[g];
+        int64_t i = lo + (int64_t)g * 64 + lane;
+        bool sel = (b >> lane) & 1;
         if (sel) {
+            int before = __popcll(b & ((1ull << lane) - 1ull));
             int64_t at = run + before;
             for (int p = 0; p < nproj; p++) {
                 const KProj& pr = projs[p];
@@ -477,9 +518,7 @@ struct FilterProjectOp : tg_operator {
         int32_t* d_total = nullptr;
         TG_POOL_ALLOC(s, &d_counts, nchunks * 4);
         TG_POOL_ALLOC(s, &d_total, 4);
-        int wpb = TG_BLOCK / 64;
-        hipLaunchKernelGGL(k_fp_count, dim3((uint32_t)((nchunks + wpb - 1) / wpb)),
-                           dim3(TG_BLOCK), 0, s->stream,
+        hipLaunchKernelGGL(k_fp_count, dim3((uint32_t)nchunks), dim3(TG_BLOCK), 0, s->stream,
                            filter.d_insts, filter.count, d_cols, in.n, d_counts, nchunks);
         TG_HIP_CHECK(hipGetLastError());
         hipLaunchKernelGGL(k_scan_serial, dim3(1), dim3(1), 0, s->stream,
@@ -515,8 +554,7 @@ struct FilterProjectOp : tg_operator {
         TG_POOL_ALLOC(s, &d_kp, kp.size() * sizeof(KProj));
         TG_HIP_CHECK(hipMemcpyAsync(d_kp, kp.data(), kp.size() * sizeof(KProj),
                                     hipMemcpyHostToDevice, s->stream));
-        hipLaunchKernelGGL(k_fp_write, dim3((uint32_t)((nchunks + wpb - 1) / wpb)),
-                           dim3(TG_BLOCK), 0, s->stream,
+        hipLaunchKernelGGL(k_fp_write, dim3((uint32_t)nchunks), dim3(TG_BLOCK), 0, s->stream,
                            filter.d_insts, filter.count, d_cols, in.n, d_counts,
                            d_kp, (int)kp.size(), nchunks);
         TG_HIP_CHECK(hipGetLastError());
