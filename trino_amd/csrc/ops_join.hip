@@ -95,7 +95,7 @@ __global__ void k_key_minmax(const int64_t* __restrict__ keys,
  * then per-region LDS histogram + scan + scatter. Replaces the global
  * atomic-CAS insert (measured ~1 G CAS/s = 15 ms per 15M-row build). ---- */
 #define JREG_SLOTS 32768              /* slots per region (LDS u32 counts) */
-#define JPART_CHUNK 8192
+#define JPART_CHUNK 1024
 
 __global__ void k_jc_slots(JoinTable t, uint32_t* __restrict__ slot_of,
                            int32_t* __restrict__ region_hist /* [nchunks][nreg] */,
