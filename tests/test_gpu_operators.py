@@ -525,3 +525,62 @@ class TestOperatorStateMachine:
         out = op.drain()[0]
         op.close()
         assert len(out[0]["values"]) == 0
+
+
+class TestMultiKeyJoin:
+    """Generic multi-channel join keys (DefaultPagesHash analog): compound
+    (BIGINT, INTEGER) key, duplicates, vs a numpy composition. Pairs compared
+    as sets; probe rows ascending."""
+
+    def test_compound_key(self, sess, ops):
+        r = np.random.default_rng(33)
+        nb, m = 4000, 20000
+        b1 = r.integers(0, 500, nb).astype(np.int64)
+        b2 = r.integers(0, 7, nb).astype(np.int32)
+        bv = np.arange(nb, dtype=np.int64)
+        p1 = r.integers(0, 600, m).astype(np.int64)
+        p2 = r.integers(0, 8, m).astype(np.int32)
+        pv = np.arange(m, dtype=np.int64)
+        bridge = ops.JoinBridge(sess)
+        b = ops.hash_builder(sess, bridge,
+                             [ops.TG_BIGINT, ops.TG_INTEGER, ops.TG_BIGINT],
+                             [0, 1], [2])
+        b.add_input(ops.page_from_numpy([b1, b2, bv]))
+        b.drain()
+        b.close()
+        probe = ops.lookup_join(sess, bridge,
+                                [ops.TG_BIGINT, ops.TG_INTEGER, ops.TG_BIGINT],
+                                [0, 1], [2])
+        probe.add_input(ops.page_from_numpy([p1, p2, pv]))
+        out = probe.drain()[0]
+        probe.close()
+        bridge.close()
+        # numpy reference join
+        bmap = {}
+        for i in range(nb):
+            bmap.setdefault((int(b1[i]), int(b2[i])), []).append(int(bv[i]))
+        exp = []
+        for i in range(m):
+            for v in bmap.get((int(p1[i]), int(p2[i])), []):
+                exp.append((int(pv[i]), v))
+        got = sorted(zip(out[0]["values"].tolist(), out[1]["values"].tolist()))
+        assert got == sorted(exp)
+
+    def test_double_key_with_nulls(self, sess, ops):
+        bk = np.array([1.5, 2.5, -0.0], np.float64)
+        bv = np.array([10, 20, 30], np.int64)
+        bridge = ops.JoinBridge(sess)
+        b = ops.hash_builder(sess, bridge, [ops.TG_DOUBLE, ops.TG_BIGINT], [0], [1])
+        b.add_input(ops.page_from_numpy([bk, bv]))
+        b.drain()
+        b.close()
+        pk = np.array([2.5, 0.0, 7.0, 1.5], np.float64)   # +0.0 matches -0.0
+        pvalid = np.array([~np.uint64(0)], np.uint64)
+        pvalid[0] &= ~np.uint64(1 << 3)   # last probe row null
+        probe = ops.lookup_join(sess, bridge, [ops.TG_DOUBLE], [0], [0])
+        probe.add_input(ops.page_from_numpy([pk], valids=[pvalid]))
+        out = probe.drain()[0]
+        probe.close()
+        bridge.close()
+        assert out[0]["values"].tolist() == [2.5, 0.0]
+        assert out[1]["values"].tolist() == [20, 30]

@@ -35,7 +35,64 @@ struct JoinTable {
     int32_t* csr_rows = nullptr;    /* [n] */
     int64_t* keys = nullptr;     /* [n] flat copy of build keys */
     uint64_t* key_valid = nullptr; /* packed bitmap or null */
+    /* generic multi-channel keys (DefaultPagesHash analog; the fields above
+     * are the BigintPagesHash specialization). bkeys points at the bridge's
+     * concatenated build channels. */
+    int32_t generic = 0;
+    int32_t n_key_ch = 1;
+    const KColH* bkeys = nullptr;   /* device array[n_key_ch] */
 };
+
+/* slot of a build row; generic keys hash with the canonical row hash
+ * (InterpretedHashGenerator semantics) finalized by murmur3, like
+ * DefaultPagesHash.getHashPosition over strategy raw hashes */
+__device__ static inline uint32_t jt_build_slot(const JoinTable& t, int64_t row)
+{
+    uint64_t h = t.generic ? row_hash(t.bkeys, t.n_key_ch, row)
+                           : (uint64_t)t.keys[row];
+    return (uint32_t)(d_murmur3_mix(h) & (uint64_t)t.mask);
+}
+
+__device__ static inline bool jt_build_null(const JoinTable& t, int64_t row)
+{
+    if (!t.generic)
+        return t.key_valid && !((t.key_valid[row >> 6] >> (row & 63)) & 1);
+    for (int c = 0; c < t.n_key_ch; c++)
+        if (kcol_is_null(t.bkeys[c], row)) return true;
+    return false;
+}
+
+/* probe-side context: fast single-bigint pointer OR generic channel array */
+struct ProbeKeys {
+    const int64_t* pk = nullptr;
+    const uint64_t* pvalid = nullptr;
+    const KColH* pkg = nullptr;   /* device array[n_key_ch] when generic */
+};
+
+__device__ static inline bool probe_null(const JoinTable& t, const ProbeKeys& p, int64_t i)
+{
+    if (!t.generic)
+        return p.pvalid && !((p.pvalid[i >> 6] >> (i & 63)) & 1);
+    for (int c = 0; c < t.n_key_ch; c++)
+        if (kcol_is_null(p.pkg[c], i)) return true;
+    return false;
+}
+
+__device__ static inline uint32_t probe_slot(const JoinTable& t, const ProbeKeys& p, int64_t i)
+{
+    uint64_t h = t.generic ? row_hash(p.pkg, t.n_key_ch, i) : (uint64_t)p.pk[i];
+    return (uint32_t)(d_murmur3_mix(h) & (uint64_t)t.mask);
+}
+
+__device__ static inline bool probe_matches(const JoinTable& t, const ProbeKeys& p,
+                                            int64_t probe_row, int32_t build_row)
+{
+    if (!t.generic) return t.keys[build_row] == p.pk[probe_row];
+    for (int c = 0; c < t.n_key_ch; c++)
+        if (kcol_word(t.bkeys[c], build_row) != kcol_word(p.pkg[c], probe_row))
+            return false;
+    return true;
+}
 
 struct tg_join_bridge {
     tg_session* s = nullptr;
@@ -50,6 +107,8 @@ struct tg_join_bridge {
     std::vector<DevBlock> build_channels;
     std::vector<tg_type> build_types;
     std::vector<int32_t> build_output_channels;
+    std::vector<int32_t> key_channels;
+    KColH* d_bkeys = nullptr;        /* device array over build key channels */
     bool built = false;
 };
 
@@ -108,9 +167,8 @@ __global__ void k_jc_slots(JoinTable t, uint32_t* __restrict__ slot_of,
     for (int64_t r = threadIdx.x; r < nreg; r += blockDim.x) lhist[r] = 0;
     __syncthreads();
     for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-        bool null = t.key_valid && !((t.key_valid[i >> 6] >> (i & 63)) & 1);
-        uint32_t slot = null ? 0xFFFFFFFFu
-                             : (uint32_t)(d_murmur3_mix((uint64_t)t.keys[i]) & (uint64_t)t.mask);
+        bool null = jt_build_null(t, i);
+        uint32_t slot = null ? 0xFFFFFFFFu : jt_build_slot(t, i);
         slot_of[i] = slot;
         if (!null) atomicAdd(&lhist[slot / region_slots], 1);
     }
@@ -265,8 +323,7 @@ __global__ void k_join_build(JoinTable t)
 /* probe phase 1 (DefaultPagesHash.getAddressIndex hash-all/gather-all/verify
  * shape): count matches per probe row, caching the matched head so the fill
  * pass never re-probes the table */
-__global__ void k_probe_count(JoinTable t, const int64_t* __restrict__ pkeys,
-                              const uint64_t* __restrict__ pvalid, int64_t m,
+__global__ void k_probe_count(JoinTable t, ProbeKeys p, int64_t m,
                               int32_t* __restrict__ counts,
                               int32_t* __restrict__ heads)
 {
@@ -275,20 +332,20 @@ __global__ void k_probe_count(JoinTable t, const int64_t* __restrict__ pkeys,
     for (; i < m; i += stride) {
         int32_t cnt = 0;
         int32_t head = -1;
-        if (!pvalid || ((pvalid[i >> 6] >> (i & 63)) & 1)) {
-            int64_t key = pkeys[i];
-            int64_t slot = (int64_t)(d_murmur3_mix((uint64_t)key) & (uint64_t)t.mask);
+        if (!probe_null(t, p, i)) {
+            int64_t slot = (int64_t)probe_slot(t, p, i);
             if (t.csr) {
                 for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1]; x++)
-                    cnt += (t.keys[t.csr_rows[x]] == key);
+                    cnt += probe_matches(t, p, i, t.csr_rows[x]);
                 head = (int32_t)slot;
             }
-            else {
+            else {   /* legacy open addressing: single-BIGINT only */
+                int64_t key = p.pk[i];
                 int32_t cur;
                 while ((cur = t.slots[slot]) != -1) {
                     if (t.keys[cur] == key) {
                         head = cur;
-                        for (int32_t p = cur; p != -1; p = t.links[p]) cnt++;
+                        for (int32_t q = cur; q != -1; q = t.links[q]) cnt++;
                         break;
                     }
                     slot = (slot + 1) & t.mask;
@@ -304,8 +361,7 @@ __global__ void k_probe_count(JoinTable t, const int64_t* __restrict__ pkeys,
  * CSR: walk the bucket DESCENDING — scatter order is global row order, so
  * this reproduces ArrayPositionLinks' reverse-insertion emission exactly
  * (and deterministically). */
-__global__ void k_probe_fill(JoinTable t, const int64_t* __restrict__ pkeys,
-                             int64_t m,
+__global__ void k_probe_fill(JoinTable t, ProbeKeys p, int64_t m,
                              const int32_t* __restrict__ heads,
                              const int64_t* __restrict__ offsets,
                              int32_t* __restrict__ out_probe,
@@ -318,10 +374,9 @@ __global__ void k_probe_fill(JoinTable t, const int64_t* __restrict__ pkeys,
         int32_t h = heads[i];
         if (h < 0) continue;
         if (t.csr) {
-            int64_t key = pkeys[i];
             for (int32_t x = t.bucket_off[h + 1] - 1; x >= t.bucket_off[h]; x--) {
                 int32_t row = t.csr_rows[x];
-                if (t.keys[row] == key) {
+                if (probe_matches(t, p, i, row)) {
                     out_probe[at] = (int32_t)i;
                     out_build[at] = row;
                     at++;
@@ -329,9 +384,9 @@ __global__ void k_probe_fill(JoinTable t, const int64_t* __restrict__ pkeys,
             }
         }
         else {
-            for (int32_t p = h; p != -1; p = t.links[p]) {
+            for (int32_t q = h; q != -1; q = t.links[q]) {
                 out_probe[at] = (int32_t)i;
-                out_build[at] = p;
+                out_build[at] = q;
                 at++;
             }
         }
@@ -490,10 +545,26 @@ struct HashBuilderOp : tg_operator {
                 at += p.n;
             }
         }
-        /* key column = widened copy of the key channel (BIGINT path) */
-        {
+        /* single BIGINT key -> BigintPagesHash-style flat key copy;
+         * otherwise the DefaultPagesHash-style generic channel compare
+         * (requires the CSR table; the legacy open-addressing path stays
+         * single-key). */
+        t.generic = !(key_channels.size() == 1 &&
+                      bridge->build_channels[key_channels[0]].type == TG_BIGINT);
+        t.n_key_ch = (int32_t)key_channels.size();
+        if (t.generic) {
+            DevPage view;   /* non-owning view over the concatenated channels */
+            view.n = total_rows;
+            view.blocks = bridge->build_channels;
+            for (auto& b : view.blocks) b.owned = false;
+            tg_status kst = make_kcols(s, view, key_channels.data(),
+                                       (int)key_channels.size(), &bridge->d_bkeys);
+            view.blocks.clear();
+            if (kst != TG_OK) return kst;
+            t.bkeys = bridge->d_bkeys;
+        }
+        else {
             const DevBlock& kb = bridge->build_channels[key_channels[0]];
-            if (kb.type != TG_BIGINT) { TG_SET_ERR("round 1 join keys are BIGINT"); return TG_ERR_UNSUPPORTED; }
             TG_HIP_CHECK(hipMemcpyAsync(t.keys, kb.data, total_rows * 8,
                                         hipMemcpyDeviceToDevice, s->stream));
             if (t.key_valid && kb.valid) {
@@ -505,6 +576,7 @@ struct HashBuilderOp : tg_operator {
         /* A/B on MI355X (Q3 SF100): CSR 98.0 ms/step vs open addressing
          * 110.1 — CSR default; TG_JOIN_CSR=0 selects the legacy path */
         static int use_csr = [] { const char* e = getenv("TG_JOIN_CSR"); return e ? atoi(e) : 1; }();
+        if (t.generic && !use_csr) { TG_SET_ERR("generic join keys require the CSR table"); return TG_ERR_UNSUPPORTED; }
         if (use_csr && total_rows > 0) {
             t.csr = 1;
             int64_t region_slots = t.capacity < JREG_SLOTS ? t.capacity : JREG_SLOTS;
@@ -569,8 +641,9 @@ struct HashBuilderOp : tg_operator {
                 TG_HIP_CHECK(hipGetLastError());
             }
         }
-        /* dynamic filter source: min/max over non-null build keys */
-        {
+        /* dynamic filter source: min/max over non-null build keys
+         * (single-BIGINT-key builds; generic keys report no range) */
+        if (!t.generic) {
             long long* d_mm = nullptr;
             TG_POOL_ALLOC(s, &d_mm, 3 * 8);
             long long init[3] = {INT64_MAX, INT64_MIN, 0};
@@ -622,10 +695,20 @@ struct LookupJoinOp : tg_operator {
         DevPage in;
         tg_status st = tg_upload_page(s, page, &in);
         if (st != TG_OK) return st;
-        const DevBlock& kb = in.blocks[key_channels[0]];
-        if (kb.type != TG_BIGINT) { tg_free_page(s, &in); TG_SET_ERR("round 1 join keys are BIGINT"); return TG_ERR_UNSUPPORTED; }
-
         JoinTable& t = bridge->t;
+        ProbeKeys pkc{};
+        KColH* d_pkg = nullptr;
+        if (t.generic) {
+            st = make_kcols(s, in, key_channels.data(), (int)key_channels.size(), &d_pkg);
+            if (st != TG_OK) { tg_free_page(s, &in); return st; }
+            pkc.pkg = d_pkg;
+        }
+        else {
+            const DevBlock& kb = in.blocks[key_channels[0]];
+            if (kb.type != TG_BIGINT) { tg_free_page(s, &in); TG_SET_ERR("single join key must be BIGINT (use multi-channel keys otherwise)"); return TG_ERR_UNSUPPORTED; }
+            pkc.pk = (const int64_t*)kb.data;
+            pkc.pvalid = kb.valid;
+        }
         int32_t* d_counts = nullptr;
         int32_t* d_heads = nullptr;
         int64_t* d_offsets = nullptr;
@@ -635,8 +718,7 @@ struct LookupJoinOp : tg_operator {
         TG_POOL_ALLOC(s, &d_offsets, (in.n ? in.n : 1) * 8);
         TG_POOL_ALLOC(s, &d_total, 8);
         hipLaunchKernelGGL(k_probe_count, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
-                           0, s->stream, t, (const int64_t*)kb.data, kb.valid, in.n,
-                           d_counts, d_heads);
+                           0, s->stream, t, pkc, in.n, d_counts, d_heads);
         TG_HIP_CHECK(hipGetLastError());
         tg_status sst = run_scan_counts(s, d_counts, in.n, d_offsets, d_total);
         if (sst != TG_OK) return sst;
@@ -649,8 +731,7 @@ struct LookupJoinOp : tg_operator {
         TG_POOL_ALLOC(s, &d_op, (total ? total : 1) * 4);
         TG_POOL_ALLOC(s, &d_ob, (total ? total : 1) * 4);
         hipLaunchKernelGGL(k_probe_fill, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
-                           0, s->stream, t, (const int64_t*)kb.data, in.n,
-                           d_heads, d_offsets, d_op, d_ob);
+                           0, s->stream, t, pkc, in.n, d_heads, d_offsets, d_op, d_ob);
         TG_HIP_CHECK(hipGetLastError());
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
 
@@ -674,6 +755,7 @@ struct LookupJoinOp : tg_operator {
         tg_pool_free(s, d_heads);
         tg_pool_free(s, d_offsets);
         tg_pool_free(s, d_total);
+        if (d_pkg) tg_pool_free(s, d_pkg);
         tg_pool_free(s, d_op);
         tg_pool_free(s, d_ob);
         tg_free_page(s, &in);
@@ -711,6 +793,7 @@ extern "C" void tg_join_bridge_close(tg_join_bridge* b)
     if (b->t.key_valid) tg_pool_free(b->s, b->t.key_valid);
     if (b->t.bucket_off) tg_pool_free(b->s, b->t.bucket_off);
     if (b->t.csr_rows) tg_pool_free(b->s, b->t.csr_rows);
+    if (b->d_bkeys) tg_pool_free(b->s, b->d_bkeys);
     for (auto& c : b->build_channels) {
         if (c.data) tg_pool_free(b->s, c.data);
         if (c.valid) tg_pool_free(b->s, c.valid);
@@ -724,8 +807,9 @@ extern "C" tg_status tg_hash_builder_create(tg_session* s, tg_join_bridge* bridg
     const int32_t* output_channels, int32_t n_output_channels,
     tg_operator** out)
 {
-    if (!s || !bridge || !build_types || !key_channels || n_key_channels != 1) {
-        TG_SET_ERR("round 1 supports exactly one BIGINT key channel");
+    if (!s || !bridge || !build_types || !key_channels || n_key_channels < 1 ||
+        n_key_channels > 7) {
+        TG_SET_ERR("1..7 join key channels");
         return TG_ERR_INVALID_ARG;
     }
     auto* op = new HashBuilderOp();
@@ -742,26 +826,25 @@ extern "C" tg_status tg_hash_builder_create(tg_session* s, tg_join_bridge* bridg
 
 /* ---- semi join (operator/HashSemiJoinOperator.java: appends a BOOLEAN
  * "matched" channel to the probe page; null probe keys yield NULL) ---- */
-__global__ void k_semi_probe(JoinTable t, const int64_t* __restrict__ pkeys,
-                             const uint64_t* __restrict__ pvalid, int64_t m,
+__global__ void k_semi_probe(JoinTable t, ProbeKeys p, int64_t m,
                              int8_t* __restrict__ match, uint64_t* __restrict__ mvalid)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < m; i += stride) {
-        if (pvalid && !((pvalid[i >> 6] >> (i & 63)) & 1)) {
+        if (probe_null(t, p, i)) {
             match[i] = 0;
             atomicAnd((unsigned long long*)&mvalid[i >> 6], ~(1ull << (i & 63)));
             continue;
         }
-        int64_t key = pkeys[i];
-        int64_t slot = (int64_t)(d_murmur3_mix((uint64_t)key) & (uint64_t)t.mask);
+        int64_t slot = (int64_t)probe_slot(t, p, i);
         int8_t hit = 0;
         if (t.csr) {
             for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1] && !hit; x++)
-                hit = (t.keys[t.csr_rows[x]] == key);
+                hit = probe_matches(t, p, i, t.csr_rows[x]);
         }
         else {
+            int64_t key = p.pk[i];
             int32_t cur;
             while ((cur = t.slots[slot]) != -1) {
                 if (t.keys[cur] == key) { hit = 1; break; }
@@ -782,8 +865,21 @@ struct SemiJoinOp : tg_operator {
         DevPage in;
         tg_status st = tg_upload_page(s, page, &in);
         if (st != TG_OK) return st;
-        const DevBlock& kb = in.blocks[key_channel];
-        if (kb.type != TG_BIGINT) { tg_free_page(s, &in); TG_SET_ERR("semi join keys are BIGINT"); return TG_ERR_UNSUPPORTED; }
+        JoinTable& jt = bridge->t;
+        ProbeKeys pkc{};
+        KColH* d_pkg = nullptr;
+        if (jt.generic) {
+            int32_t kcs[1] = {key_channel};
+            st = make_kcols(s, in, kcs, 1, &d_pkg);
+            if (st != TG_OK) { tg_free_page(s, &in); return st; }
+            pkc.pkg = d_pkg;
+        }
+        else {
+            const DevBlock& kb0 = in.blocks[key_channel];
+            if (kb0.type != TG_BIGINT) { tg_free_page(s, &in); TG_SET_ERR("semi join single key must be BIGINT"); return TG_ERR_UNSUPPORTED; }
+            pkc.pk = (const int64_t*)kb0.data;
+            pkc.pvalid = kb0.valid;
+        }
         DevBlock mb;
         mb.type = TG_BOOLEAN;
         mb.n = in.n;
@@ -792,10 +888,10 @@ struct SemiJoinOp : tg_operator {
         TG_POOL_ALLOC(s, &mb.valid, (words ? words : 1) * 8);
         TG_HIP_CHECK(hipMemsetAsync(mb.valid, 0xFF, words * 8, s->stream));
         hipLaunchKernelGGL(k_semi_probe, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
-                           0, s->stream, bridge->t, (const int64_t*)kb.data, kb.valid,
-                           in.n, (int8_t*)mb.data, mb.valid);
+                           0, s->stream, bridge->t, pkc, in.n, (int8_t*)mb.data, mb.valid);
         TG_HIP_CHECK(hipGetLastError());
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        if (d_pkg) tg_pool_free(s, d_pkg);
         DevPage outp;
         outp.n = in.n;
         for (auto& b : in.blocks) { outp.blocks.push_back(b); }
@@ -835,8 +931,8 @@ extern "C" tg_status tg_lookup_join_create(tg_session* s, tg_join_bridge* bridge
     const int32_t* probe_output_channels, int32_t n_probe_output,
     tg_operator** out)
 {
-    if (!s || !bridge || !key_channels || n_key_channels != 1) {
-        TG_SET_ERR("round 1 supports exactly one BIGINT key channel");
+    if (!s || !bridge || !key_channels || n_key_channels < 1 || n_key_channels > 7) {
+        TG_SET_ERR("1..7 join key channels");
         return TG_ERR_INVALID_ARG;
     }
     auto* op = new LookupJoinOp();
@@ -856,6 +952,7 @@ extern "C" tg_status tg_join_bridge_key_range(tg_join_bridge* b, int64_t* key_mi
                                               int64_t* key_max, int64_t* key_rows)
 {
     if (!b || !b->built) { TG_SET_ERR("lookup source not built"); return TG_ERR_STATE; }
+    if (b->t.generic) { TG_SET_ERR("key range is tracked for single BIGINT keys"); return TG_ERR_UNSUPPORTED; }
     if (key_min) *key_min = b->key_min;
     if (key_max) *key_max = b->key_max;
     if (key_rows) *key_rows = b->key_rows;
