@@ -584,3 +584,28 @@ class TestMultiKeyJoin:
         bridge.close()
         assert out[0]["values"].tolist() == [2.5, 0.0]
         assert out[1]["values"].tolist() == [20, 30]
+
+
+class TestVarcharHash:
+    def test_varchar_row_hash_and_partition(self, sess, ops):
+        """canonical VARCHAR hash (XxHash64 over utf8 bytes) on device ==
+        oracle, including the 31*combine with a BIGINT channel; and
+        partitioning BY a varchar channel."""
+        words = [b"", b"a", b"BUILDING", b"x" * 40, b"AIR REG", b"\xc3\xa9clair"]
+        r = rng(44)
+        n = 5000
+        pick = r.integers(0, len(words), n)
+        offsets = np.zeros(n + 1, np.int32)
+        for i in range(n):
+            offsets[i + 1] = offsets[i] + len(words[pick[i]])
+        data = np.frombuffer(b"".join(words[k] for k in pick.tolist()), np.uint8).copy()
+        keys = r.integers(0, 1000, n).astype(np.int64)
+        page = ops.page_with_varchar([keys, (data, offsets)])
+        got = ops.hash_rows(sess, page, [0, 1])
+        M = (1 << 64) - 1
+        exp = np.empty(n, np.uint64)
+        for i in range(n):
+            h = oracle.combine_hash(0, np.int64(np.uint64(oracle.bigint_hash(int(keys[i])))).item())
+            h = oracle.combine_hash(h, np.int64(np.uint64(oracle.xxhash64(words[pick[i]]))).item())
+            exp[i] = np.uint64(h % (1 << 64))
+        assert np.array_equal(got, exp)

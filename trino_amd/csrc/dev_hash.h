@@ -44,7 +44,12 @@ __device__ static inline uint64_t d_murmur3_mix(uint64_t h)
 }
 
 /* canonical typed channel value widened to 64 bits (for key stores) */
-struct KColH { const void* data; const uint64_t* valid; int32_t type; int32_t _pad; };
+struct KColH { const void* data; const uint64_t* valid; int32_t type; int32_t _pad;
+               const int32_t* offsets; /* VARCHAR */ };
+
+/* standard XxHash64 over bytes, seed 0 (airlift XxHash64; the VARCHAR
+ * HASH_CODE operator, AbstractVariableWidthType) */
+__device__ static inline uint64_t d_xxhash64_bytes(const uint8_t* p, int64_t len);
 
 __device__ static inline bool kcol_is_null(const KColH& c, int64_t i)
 {
@@ -68,6 +73,9 @@ __device__ static inline uint64_t kcol_hash(const KColH& c, int64_t i)
 {
     if (kcol_is_null(c, i)) return 0;            /* NULL_HASH_CODE */
     if (c.type == TG_DOUBLE) return d_double_hash(((const double*)c.data)[i]);
+    if (c.type == TG_VARCHAR)
+        return d_xxhash64_bytes((const uint8_t*)c.data + c.offsets[i],
+                                c.offsets[i + 1] - c.offsets[i]);
     return d_bigint_hash((int64_t)kcol_word(c, i));
 }
 
@@ -99,6 +107,56 @@ __device__ static inline uint64_t d_xxhash64_long(int64_t v)
     return h;
 }
 
+__device__ inline uint64_t d_xxhash64_bytes(const uint8_t* p, int64_t len)
+{
+    const uint64_t P1 = 0x9E3779B185EBCA87ULL, P2 = 0xC2B2AE3D27D4EB4FULL,
+                   P3 = 0x165667B19E3779F9ULL, P4 = 0x85EBCA77C2B2AE63ULL,
+                   P5 = 0x27D4EB2F165667C5ULL;
+    const uint8_t* end = p + len;
+    uint64_t h;
+    auto rd64 = [](const uint8_t* q) {
+        uint64_t v;
+        memcpy(&v, q, 8);
+        return v;
+    };
+    auto mix = [&](uint64_t cur, uint64_t v) { return d_rotl64(cur + v * P2, 31) * P1; };
+    if (len >= 32) {
+        uint64_t v1 = P1 + P2, v2 = P2, v3 = 0, v4 = (uint64_t)0 - P1;
+        do {
+            v1 = mix(v1, rd64(p));
+            v2 = mix(v2, rd64(p + 8));
+            v3 = mix(v3, rd64(p + 16));
+            v4 = mix(v4, rd64(p + 24));
+            p += 32;
+        } while (p <= end - 32);
+        h = d_rotl64(v1, 1) + d_rotl64(v2, 7) + d_rotl64(v3, 12) + d_rotl64(v4, 18);
+        h = (h ^ mix(0, v1)) * P1 + P4;
+        h = (h ^ mix(0, v2)) * P1 + P4;
+        h = (h ^ mix(0, v3)) * P1 + P4;
+        h = (h ^ mix(0, v4)) * P1 + P4;
+    }
+    else {
+        h = P5;
+    }
+    h += (uint64_t)len;
+    while (p + 8 <= end) {
+        h = d_rotl64(h ^ mix(0, rd64(p)), 27) * P1 + P4;
+        p += 8;
+    }
+    if (p + 4 <= end) {
+        uint32_t k;
+        memcpy(&k, p, 4);
+        h = d_rotl64(h ^ ((uint64_t)k * P1), 23) * P2 + P3;
+        p += 4;
+    }
+    while (p < end) {
+        h = d_rotl64(h ^ (*p * P5), 11) * P1;
+        p++;
+    }
+    h ^= h >> 33; h *= P2; h ^= h >> 29; h *= P3; h ^= h >> 32;
+    return h;
+}
+
 __device__ static inline int32_t d_partition_local(uint64_t raw, int32_t n_pow2)
 {
     return (int32_t)(d_xxhash64_long((int64_t)__brevll(raw))) & (n_pow2 - 1);
@@ -111,7 +169,7 @@ static inline tg_status make_kcols(tg_session* s, const DevPage& page,
     std::vector<KColH> h(n);
     for (int i = 0; i < n; i++) {
         const DevBlock& b = page.blocks[channels ? channels[i] : i];
-        h[i] = {b.data, b.valid, (int32_t)b.type, 0};
+        h[i] = {b.data, b.valid, (int32_t)b.type, 0, b.offsets};
     }
     TG_HIP_CHECK(hipMalloc(out, n * sizeof(KColH)));
     TG_HIP_CHECK(hipMemcpyAsync(*out, h.data(), n * sizeof(KColH),

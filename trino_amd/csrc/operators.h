@@ -14,6 +14,7 @@ struct DevBlock {
     int64_t n = 0;
     void* data = nullptr;          /* owned device buffer */
     uint64_t* valid = nullptr;     /* packed bitmap, bit=1 valid; null = no nulls */
+    int32_t* offsets = nullptr;    /* VARCHAR: n+1 offsets into data */
     bool owned = true;
     bool valid_owned_override = false;   /* data owned but bitmap borrowed */
     int64_t elem_size() const

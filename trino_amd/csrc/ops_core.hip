@@ -75,9 +75,33 @@ tg_status tg_upload_page(tg_session* s, const tg_page* in, DevPage* out)
         db->type = (tg_type)b->type;
         db->n = b->position_count;
         if (b->type == TG_VARCHAR) {
-            TG_SET_ERR("VARCHAR channels cross this boundary dictionary-encoded "
-                       "(TINYINT/INTEGER ids) in round 1");
-            return TG_ERR_UNSUPPORTED;
+            if (b->kind != TG_BK_VALUE || !b->offsets) {
+                TG_SET_ERR("VARCHAR blocks cross as flat offsets+bytes or "
+                           "dictionary-encoded ids");
+                return TG_ERR_UNSUPPORTED;
+            }
+            /* VariableWidthBlock: int32 offsets[n+1] + utf8 bytes */
+            int32_t total_bytes = 0;
+            if (b->on_device) {
+                TG_HIP_CHECK(hipMemcpy(&total_bytes, b->offsets + db->n, 4,
+                                       hipMemcpyDeviceToHost));
+            }
+            else {
+                total_bytes = b->offsets[db->n];
+            }
+            tg_status st = upload_flat(s, b->offsets, b->on_device,
+                                       (db->n + 1) * 4, (void**)&db->offsets);
+            if (st != TG_OK) return st;
+            st = upload_flat(s, b->data, b->on_device,
+                             total_bytes ? total_bytes : 1, &db->data);
+            if (st != TG_OK) return st;
+            if (b->valid) {
+                int64_t words = (db->n + 63) / 64;
+                st = upload_flat(s, b->valid, b->on_device, words * 8,
+                                 (void**)&db->valid);
+                if (st != TG_OK) return st;
+            }
+            continue;
         }
         if (b->kind == TG_BK_VALUE) {
             if (b->on_device) {
@@ -128,6 +152,7 @@ void tg_free_page(tg_session* s, DevPage* p)
     for (auto& b : p->blocks) {
         if (b.owned && b.data) tg_pool_free(s, b.data);
         if (b.owned && !b.valid_owned_override && b.valid) tg_pool_free(s, b.valid);
+        if (b.owned && b.offsets) tg_pool_free(s, b.offsets);
     }
     p->blocks.clear();
     p->n = 0;

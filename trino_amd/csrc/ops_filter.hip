@@ -434,6 +434,11 @@ __global__ void k_gather(const T* __restrict__ src, const uint64_t* __restrict__
 tg_status run_gather(tg_session* s, const DevBlock& src, const int32_t* d_positions,
                      int32_t count, DevBlock* out)
 {
+    if (src.type == TG_VARCHAR) {
+        TG_SET_ERR("VARCHAR payload gather is round-2 (hashing/partition-by-"
+                   "varchar works; materialize via dictionary ids for now)");
+        return TG_ERR_UNSUPPORTED;
+    }
     out->type = src.type;
     out->n = count;
     TG_POOL_ALLOC(s, &out->data, (int64_t)(count ? count : 1) * src.elem_size());

@@ -541,6 +541,13 @@ extern "C" tg_status tg_hash_aggregation_create(tg_session* s,
         TG_SET_ERR("invalid hash aggregation spec (0..7 group channels, <=%d aggs)", MAX_AGGS);
         return TG_ERR_INVALID_ARG;
     }
+    for (int i = 0; i < n_group_channels; i++) {
+        if (group_types[i] == TG_VARCHAR) {
+            TG_SET_ERR("VARCHAR group keys cross dictionary-encoded in round 1 "
+                       "(variable-width key store is round-2; DESIGN.md §7b)");
+            return TG_ERR_UNSUPPORTED;
+        }
+    }
     auto* op = new HashAggOp();
     op->s = s;
     if (n_group_channels > 0) {

@@ -812,6 +812,12 @@ extern "C" tg_status tg_hash_builder_create(tg_session* s, tg_join_bridge* bridg
         TG_SET_ERR("1..7 join key channels");
         return TG_ERR_INVALID_ARG;
     }
+    for (int i = 0; i < n_key_channels; i++) {
+        if (build_types[key_channels[i]] == TG_VARCHAR) {
+            TG_SET_ERR("VARCHAR join keys cross dictionary-encoded in round 1");
+            return TG_ERR_UNSUPPORTED;
+        }
+    }
     auto* op = new HashBuilderOp();
     op->s = s;
     op->bridge = bridge;

@@ -395,3 +395,37 @@ def semi_join(session, bridge, key_channel):
     h = ctypes.c_void_p()
     _check(_lib.tg_semi_join_create(session._h, bridge._h, key_channel, ctypes.byref(h)))
     return Operator(session, h)
+
+
+def page_with_varchar(columns):
+    """Build a page where entries may be numpy arrays (fixed width) or
+    (bytes_array_uint8, offsets_int32) tuples for VARCHAR channels."""
+    n = None
+    for c in columns:
+        n = len(c[1]) - 1 if isinstance(c, tuple) else len(c)
+        break
+    blocks = (TgBlock * len(columns))()
+    keep = []
+    for i, col in enumerate(columns):
+        if isinstance(col, tuple):
+            data, offsets = col
+            blocks[i].type = TG_VARCHAR
+            blocks[i].kind = 0
+            blocks[i].position_count = len(offsets) - 1
+            blocks[i].on_device = 0
+            blocks[i].data = data.ctypes.data
+            blocks[i].offsets = offsets.ctypes.data
+            keep.append((data, offsets))
+        else:
+            blocks[i].type = _NP2TG[col.dtype]
+            blocks[i].kind = 0
+            blocks[i].position_count = len(col)
+            blocks[i].on_device = 0
+            blocks[i].data = col.ctypes.data
+            keep.append(col)
+    p = TgPage()
+    p.channel_count = len(columns)
+    p.position_count = blocks[0].position_count
+    p.blocks = blocks
+    p._keepalive = (blocks, keep)
+    return p
