@@ -60,6 +60,10 @@ _lib.o_bigint_groupby.argtypes = [ctypes.c_void_p] * 2 + [ctypes.c_int64] + [cty
 _lib.o_flat_groupby.restype = ctypes.c_int32
 _lib.o_flat_groupby.argtypes = [ctypes.c_int32, ctypes.c_void_p, ctypes.c_void_p,
                                 ctypes.c_int64, ctypes.c_void_p, ctypes.c_void_p]
+_lib.o_flat_groupby_v.restype = ctypes.c_int32
+_lib.o_flat_groupby_v.argtypes = [ctypes.c_int32, ctypes.c_void_p, ctypes.c_void_p,
+                                  ctypes.c_void_p, ctypes.c_int64, ctypes.c_void_p,
+                                  ctypes.c_void_p]
 _lib.o_grouped_sum_f64.restype = None
 _lib.o_grouped_sum_f64.argtypes = [ctypes.c_void_p] * 3 + [ctypes.c_int64, ctypes.c_void_p]
 _lib.o_grouped_count.restype = None
@@ -206,14 +210,26 @@ def bigint_groupby(keys, valid=None):
     return gids, ng, vals[:ng], int(nullg[0])
 
 
-def flat_groupby(columns, types):
-    n = len(columns[0])
+def flat_groupby(columns, types, offsets=None):
+    """columns: np arrays; VARCHAR channels pass bytes array + matching entry
+    in `offsets` (list parallel to columns; None for fixed-width)."""
+    n = None
+    for i, c in enumerate(columns):
+        n = (len(offsets[i]) - 1) if (offsets and offsets[i] is not None) else len(c)
+        break
     gids = np.empty(n, np.int32)
     first = np.empty(max(n, 1), np.int64)
     tarr = np.array(types, np.int32)
     ptrs = (ctypes.c_void_p * len(columns))(*[c.ctypes.data for c in columns])
-    ng = _lib.o_flat_groupby(len(columns), tarr.ctypes.data, ctypes.addressof(ptrs),
-                             n, gids.ctypes.data, first.ctypes.data)
+    if offsets is None:
+        ng = _lib.o_flat_groupby(len(columns), tarr.ctypes.data, ctypes.addressof(ptrs),
+                                 n, gids.ctypes.data, first.ctypes.data)
+    else:
+        optrs = (ctypes.c_void_p * len(columns))(
+            *[(o.ctypes.data if o is not None else None) for o in offsets])
+        ng = _lib.o_flat_groupby_v(len(columns), tarr.ctypes.data, ctypes.addressof(ptrs),
+                                   ctypes.addressof(optrs), n, gids.ctypes.data,
+                                   first.ctypes.data)
     return gids, ng, first[:ng]
 
 

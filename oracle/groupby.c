@@ -25,6 +25,9 @@
 
 uint64_t o_murmur3_mix(uint64_t h);
 uint64_t o_bigint_hash(int64_t v);
+int32_t o_flat_groupby_v(int32_t n_channels, const int32_t* types,
+                         const void* const* datas, const int32_t* const* offsets,
+                         int64_t n, int32_t* out_group_ids, int64_t* out_first_row_by_gid);
 void o_hash_rows(int32_t, const int32_t*, const void* const*, const int32_t* const*,
                  int64_t, uint64_t*);
 
@@ -97,10 +100,19 @@ assigned:
 typedef struct { uint64_t hash; int64_t first_row; int32_t gid; int32_t used; } flat_slot;
 
 static int rows_equal(int32_t n_channels, const int32_t* types,
-                      const void* const* datas, int64_t a, int64_t b)
+                      const void* const* datas, const int32_t* const* offsets,
+                      int64_t a, int64_t b)
 {
     for (int32_t c = 0; c < n_channels; c++) {
         switch (types[c]) {
+            case 7: { /* VARCHAR: byte equality (VarcharType EQUAL) */
+                const int32_t* off = offsets[c];
+                int32_t la = off[a + 1] - off[a], lb = off[b + 1] - off[b];
+                if (la != lb) return 0;
+                if (memcmp((const uint8_t*)datas[c] + off[a],
+                           (const uint8_t*)datas[c] + off[b], (size_t)la) != 0) return 0;
+                break;
+            }
             case 0: if (((const int64_t*)datas[c])[a] != ((const int64_t*)datas[c])[b]) return 0; break;
             case 1: case 5: if (((const int32_t*)datas[c])[a] != ((const int32_t*)datas[c])[b]) return 0; break;
             case 2: if (((const int16_t*)datas[c])[a] != ((const int16_t*)datas[c])[b]) return 0; break;
@@ -121,8 +133,17 @@ EXPORT int32_t o_flat_groupby(int32_t n_channels, const int32_t* types,
                               const void* const* datas, int64_t n,
                               int32_t* out_group_ids, int64_t* out_first_row_by_gid)
 {
+    return o_flat_groupby_v(n_channels, types, datas, NULL, n,
+                            out_group_ids, out_first_row_by_gid);
+}
+
+EXPORT int32_t o_flat_groupby_v(int32_t n_channels, const int32_t* types,
+                                const void* const* datas,
+                                const int32_t* const* offsets, int64_t n,
+                                int32_t* out_group_ids, int64_t* out_first_row_by_gid)
+{
     uint64_t* hashes = malloc(sizeof(uint64_t) * (size_t)n);
-    o_hash_rows(n_channels, types, datas, NULL, n, hashes);
+    o_hash_rows(n_channels, types, datas, offsets, n, hashes);
     int64_t capacity = 16;
     while (capacity * 15 / 16 < 64) capacity *= 2;
     flat_slot* slots = calloc((size_t)capacity, sizeof(flat_slot));
@@ -135,7 +156,7 @@ EXPORT int32_t o_flat_groupby(int32_t n_channels, const int32_t* types,
         while (1) {
             if (!slots[pos].used) break;
             if (slots[pos].hash == h &&
-                rows_equal(n_channels, types, datas, slots[pos].first_row, i)) {
+                rows_equal(n_channels, types, datas, offsets, slots[pos].first_row, i)) {
                 gid = slots[pos].gid; break;
             }
             pos = (pos + 1) & mask;

@@ -609,3 +609,64 @@ class TestVarcharHash:
             h = oracle.combine_hash(h, np.int64(np.uint64(oracle.xxhash64(words[pick[i]]))).item())
             exp[i] = np.uint64(h % (1 << 64))
         assert np.array_equal(got, exp)
+
+
+class TestVarcharGroupBy:
+    """True VARCHAR group keys (variable-width key store; survey hard part
+    (c)): group ids in row order, key bytes round-trip, counts exact."""
+
+    def _varchar(self, words, pick):
+        offsets = np.zeros(len(pick) + 1, np.int32)
+        for i, k in enumerate(pick):
+            offsets[i + 1] = offsets[i] + len(words[k])
+        data = np.frombuffer(b"".join(words[k] for k in pick), np.uint8).copy()
+        return data, offsets
+
+    def test_varchar_keys(self, sess, ops):
+        words = [b"AUTOMOBILE", b"BUILDING", b"", b"x" * 50, b"FURNITURE"]
+        r = rng(55)
+        n = 20000
+        pick = r.integers(0, len(words), n).tolist()
+        data, offsets = self._varchar(words, pick)
+        vals = r.integers(0, 100, n).astype(np.int64)
+        page = ops.page_with_varchar([(data, offsets), vals])
+        op = ops.hash_aggregation(sess, [0], [ops.TG_VARCHAR],
+                                  [(ops.AGG_COUNT_STAR, -1), (ops.AGG_SUM_I64, 1)])
+        op.add_input(page)
+        out = op.drain()[0]
+        op.close()
+        gids, ng, first = oracle.flat_groupby([data], [oracle.TG_VARCHAR],
+                                              offsets=[offsets])
+        assert len(out[0]["values"]) == ng
+        exp_keys = [words[pick[i]] for i in first.tolist()]
+        assert out[0]["values"] == exp_keys
+        assert np.array_equal(out[1]["values"], oracle.grouped_count(gids, ng))
+        es = np.zeros(ng, np.int64)
+        np.add.at(es, gids, vals)
+        assert np.array_equal(out[2]["values"], es)
+
+    def test_varchar_plus_bigint_multipage(self, sess, ops):
+        words = [b"F", b"O", b"NO", b"OF"]   # prefixes that must not collide
+        r = rng(56)
+        allp, allk = [], []
+        op = ops.hash_aggregation(sess, [0, 1], [ops.TG_VARCHAR, ops.TG_BIGINT],
+                                  [(ops.AGG_COUNT_STAR, -1)])
+        for pg in range(3):
+            n = 3000
+            pick = r.integers(0, len(words), n).tolist()
+            keys = r.integers(0, 5, n).astype(np.int64)
+            data, offsets = self._varchar(words, pick)
+            op.add_input(ops.page_with_varchar([(data, offsets), keys]))
+            allp += pick
+            allk.append(keys)
+        out = op.drain()[0]
+        op.close()
+        keys = np.concatenate(allk)
+        data, offsets = self._varchar(words, allp)
+        gids, ng, first = oracle.flat_groupby([data, keys],
+                                              [oracle.TG_VARCHAR, oracle.TG_BIGINT],
+                                              offsets=[offsets, None])
+        assert len(out[0]["values"]) == ng
+        assert out[0]["values"] == [words[allp[i]] for i in first.tolist()]
+        assert np.array_equal(out[1]["values"], keys[first])
+        assert np.array_equal(out[2]["values"], oracle.grouped_count(gids, ng))

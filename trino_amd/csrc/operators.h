@@ -68,6 +68,7 @@ struct tg_operator {
                 tb.on_device = 1;
                 tb.data = b.data;
                 tb.valid = b.valid;
+                tb.offsets = b.offsets;
                 out_blocks_.push_back(tb);
             }
             out->channel_count = (int32_t)p.blocks.size();

@@ -34,7 +34,21 @@ struct GTable {
     int64_t* first_row = nullptr;    /* [max_groups] atomicMin of global row idx */
     int32_t* counter = nullptr;      /* n_groups */
     int32_t n_words = 0;             /* n_key_channels + 1 (null mask) */
+    /* variable-width key store (AppendOnlyVariableWidthData analog): VARCHAR
+     * key channels store word = (len << 40) | byte_offset into varstore */
+    uint8_t* varstore = nullptr;
+    unsigned long long* var_cursor = nullptr;
+    int64_t var_capacity = 0;
 };
+
+#define GT_VLEN(w)  ((int64_t)((w) >> 40))
+#define GT_VOFF(w)  ((int64_t)((w) & ((1ULL << 40) - 1)))
+
+__device__ static inline uint8_t gt_var_byte(const GTable& t, int64_t off)
+{
+    return __hip_atomic_load(&t.varstore[off], __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+}
 
 __global__ void k_gt_init(int32_t* state, int64_t cap, int64_t* first_row, int64_t ngroups_cap)
 {
@@ -52,12 +66,16 @@ __global__ void k_gt_assign(GTable t, const KColH* cols, int n_ch,
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < n; i += stride) {
         uint64_t h = row_hash(cols, n_ch, i);
-        /* canonical words + null mask for this row */
+        /* canonical words + null mask for this row (VARCHAR words are
+         * assigned at insert, after the bytes are appended) */
         uint64_t w[8];
         uint64_t nullmask = 0;
+        bool has_var = false;
         for (int c = 0; c < n_ch; c++) {
             bool nl = kcol_is_null(cols[c], i);
-            w[c] = nl ? 0 : kcol_word(cols[c], i);
+            bool isvar = cols[c].type == TG_VARCHAR;
+            has_var |= isvar;
+            w[c] = (nl || isvar) ? 0 : kcol_word(cols[c], i);
             nullmask |= (uint64_t)nl << c;
         }
         w[n_ch] = nullmask;
@@ -71,6 +89,20 @@ __global__ void k_gt_assign(GTable t, const KColH* cols, int n_ch,
                 if (__hip_atomic_compare_exchange_strong(&t.state[slot], &old, -2,
                         __ATOMIC_ACQ_REL, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT)) {
                     gid = atomicAdd(t.counter, 1);
+                    if (has_var) {
+                        for (int c = 0; c < n_ch; c++) {
+                            if (cols[c].type != TG_VARCHAR) continue;
+                            if ((nullmask >> c) & 1) { w[c] = 0; continue; }
+                            int64_t po = cols[c].offsets[i];
+                            int64_t len = cols[c].offsets[i + 1] - po;
+                            unsigned long long off =
+                                atomicAdd(t.var_cursor, (unsigned long long)len);
+                            const uint8_t* src = (const uint8_t*)cols[c].data + po;
+                            for (int64_t b = 0; b < len; b++)
+                                t.varstore[off + b] = src[b];
+                            w[c] = ((uint64_t)len << 40) | off;
+                        }
+                    }
                     for (int c = 0; c <= n_ch; c++)
                         __hip_atomic_store(&t.keystore[(int64_t)gid * t.n_words + c], w[c],
                                            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
@@ -90,7 +122,19 @@ __global__ void k_gt_assign(GTable t, const KColH* cols, int n_ch,
             for (int c = 0; c <= n_ch && eq; c++) {
                 uint64_t kv = __hip_atomic_load(&t.keystore[(int64_t)st * t.n_words + c],
                                                 __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-                eq = (kv == w[c]);
+                if (c < n_ch && cols[c].type == TG_VARCHAR) {
+                    if ((nullmask >> c) & 1) continue;   /* settled by mask word */
+                    int64_t po = cols[c].offsets[i];
+                    int64_t plen = cols[c].offsets[i + 1] - po;
+                    if (GT_VLEN(kv) != plen) { eq = false; break; }
+                    int64_t so = GT_VOFF(kv);
+                    const uint8_t* src = (const uint8_t*)cols[c].data + po;
+                    for (int64_t b = 0; b < plen && eq; b++)
+                        eq = (gt_var_byte(t, so + b) == src[b]);
+                }
+                else {
+                    eq = (kv == w[c]);
+                }
             }
             if (eq) { gid = st; break; }
             slot = (slot + 1) & t.mask;
@@ -103,18 +147,20 @@ __global__ void k_gt_assign(GTable t, const KColH* cols, int n_ch,
 
 /* rehash: reinsert existing groups into a larger table (keystore is stable;
  * only the slot index array is rebuilt) */
-__global__ void k_gt_rehash(GTable t, int32_t n_groups, const KColH* meta, int n_ch)
+__global__ void k_gt_rehash(GTable t, int32_t n_groups, const int32_t* types, int n_ch)
 {
-    (void)meta;
     int32_t g = blockIdx.x * blockDim.x + threadIdx.x;
     if (g >= n_groups) return;
-    /* recompute the row hash from canonical words (types recorded in meta) */
+    /* recompute the row hash from the stored canonical words (runs in a
+     * later launch: plain varstore reads are safely visible) */
     int64_t h = 0;
     uint64_t nullmask = t.keystore[(int64_t)g * t.n_words + n_ch];
     for (int c = 0; c < n_ch; c++) {
         uint64_t w = t.keystore[(int64_t)g * t.n_words + c];
         uint64_t ch;
         if ((nullmask >> c) & 1) ch = 0;
+        else if (types && types[c] == TG_VARCHAR)
+            ch = d_xxhash64_bytes(t.varstore + GT_VOFF(w), GT_VLEN(w));
         else ch = d_bigint_hash((int64_t)w);   /* canonical words hash as longs:
                                                   f64 words are normalized bits,
                                                   matching d_double_hash */
@@ -222,6 +268,38 @@ __global__ void k_emit_keys(const uint64_t* __restrict__ keystore, int n_words,
         atomicAnd((unsigned long long*)&out_valid[g >> 6], ~(1ull << (g & 63)));
 }
 
+__global__ void k_emit_var_lens(const uint64_t* __restrict__ keystore, int n_words,
+                                const int32_t* __restrict__ old_by_new, int32_t n_groups,
+                                int ch, int32_t* __restrict__ lens)
+{
+    int32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n_groups) return;
+    uint64_t w = keystore[(int64_t)old_by_new[g] * n_words + ch];
+    uint64_t nullmask = keystore[(int64_t)old_by_new[g] * n_words + (n_words - 1)];
+    lens[g] = ((nullmask >> ch) & 1) ? 0 : (int32_t)GT_VLEN(w);
+}
+
+__global__ void k_emit_var_bytes(GTable t, const int32_t* __restrict__ old_by_new,
+                                 int32_t n_groups, int ch,
+                                 const int32_t* __restrict__ out_offsets,
+                                 uint8_t* __restrict__ out_bytes,
+                                 uint64_t* __restrict__ out_valid)
+{
+    int32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n_groups) return;
+    int32_t og = old_by_new[g];
+    uint64_t w = t.keystore[(int64_t)og * t.n_words + ch];
+    uint64_t nullmask = t.keystore[(int64_t)og * t.n_words + (t.n_words - 1)];
+    if ((nullmask >> ch) & 1) {
+        if (out_valid)
+            atomicAnd((unsigned long long*)&out_valid[g >> 6], ~(1ull << (g & 63)));
+        return;
+    }
+    int64_t off = GT_VOFF(w), len = GT_VLEN(w);
+    for (int64_t b = 0; b < len; b++)
+        out_bytes[out_offsets[g] + b] = t.varstore[off + b];
+}
+
 __global__ void k_emit_f64(const double* __restrict__ state, const int32_t* old_by_new,
                            int32_t n, double* __restrict__ out)
 {
@@ -264,6 +342,7 @@ struct HashAggOp : tg_operator {
 
     GTable t;
     std::vector<KAgg> agg_state;       /* host mirror with device ptrs */
+    int32_t* d_group_types = nullptr;  /* device copy for rehash/emit */
     int64_t rows_seen = 0;
     bool emitted = false;
 
@@ -293,6 +372,24 @@ struct HashAggOp : tg_operator {
         int32_t n = 0;
         (void)hipMemcpy(&n, t.counter, 4, hipMemcpyDeviceToHost);
         return n;
+    }
+
+    tg_status ensure_varstore(int64_t incoming_bytes)
+    {
+        if (!t.varstore || incoming_bytes == 0) return TG_OK;
+        unsigned long long cur = 0;
+        TG_HIP_CHECK(hipMemcpy(&cur, t.var_cursor, 8, hipMemcpyDeviceToHost));
+        if ((int64_t)cur + incoming_bytes <= t.var_capacity) return TG_OK;
+        int64_t ncap = t.var_capacity;
+        while (ncap < (int64_t)cur + incoming_bytes) ncap *= 2;
+        uint8_t* nv = nullptr;
+        TG_HIP_CHECK(hipMalloc(&nv, ncap));
+        TG_HIP_CHECK(hipMemcpyAsync(nv, t.varstore, cur, hipMemcpyDeviceToDevice, s->stream));
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        TG_HIP_CHECK(hipFree(t.varstore));
+        t.varstore = nv;
+        t.var_capacity = ncap;
+        return TG_OK;
     }
 
     tg_status grow_if_needed(int64_t incoming)
@@ -355,7 +452,7 @@ struct HashAggOp : tg_operator {
             TG_HIP_CHECK(hipGetLastError());
             if (ng > 0) {
                 hipLaunchKernelGGL(k_gt_rehash, dim3((ng + TG_BLOCK - 1) / TG_BLOCK),
-                                   dim3(TG_BLOCK), 0, s->stream, t, ng, nullptr,
+                                   dim3(TG_BLOCK), 0, s->stream, t, ng, d_group_types,
                                    (int)group_channels.size());
                 TG_HIP_CHECK(hipGetLastError());
             }
@@ -383,6 +480,18 @@ struct HashAggOp : tg_operator {
         }
         st = grow_if_needed(in.n);
         if (st != TG_OK) { tg_free_page(s, &in); return st; }
+        if (t.varstore) {
+            int64_t vb = 0;
+            for (size_t c = 0; c < group_channels.size(); c++) {
+                if (group_types[c] != TG_VARCHAR) continue;
+                const DevBlock& b = in.blocks[group_channels[c]];
+                int32_t total = 0;
+                TG_HIP_CHECK(hipMemcpy(&total, b.offsets + b.n, 4, hipMemcpyDeviceToHost));
+                vb += total;
+            }
+            st = ensure_varstore(vb);
+            if (st != TG_OK) { tg_free_page(s, &in); return st; }
+        }
 
         st = make_kcols(s, in, group_channels.data(), (int)group_channels.size(), &d_keys);
         if (st != TG_OK) { tg_free_page(s, &in); return st; }
@@ -417,6 +526,38 @@ have_gids:;
         return st;
     }
 
+    tg_status emit_varchar_key(int32_t ng, int ch, const int32_t* d_obn, int grid,
+                               DevBlock* out)
+    {
+        out->type = TG_VARCHAR;
+        out->n = ng;
+        int32_t* d_lens = nullptr;
+        TG_POOL_ALLOC(s, &d_lens, (ng ? ng : 1) * 4);
+        hipLaunchKernelGGL(k_emit_var_lens, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                           t.keystore, t.n_words, d_obn, ng, ch, d_lens);
+        TG_HIP_CHECK(hipGetLastError());
+        std::vector<int32_t> lens(ng);
+        if (ng) {
+            TG_HIP_CHECK(hipMemcpy(lens.data(), d_lens, (size_t)ng * 4,
+                                   hipMemcpyDeviceToHost));
+        }
+        std::vector<int32_t> offs(ng + 1, 0);
+        for (int32_t i = 0; i < ng; i++) offs[i + 1] = offs[i] + lens[i];
+        TG_POOL_ALLOC(s, &out->offsets, (ng + 1) * 4);
+        TG_HIP_CHECK(hipMemcpyAsync(out->offsets, offs.data(), (ng + 1) * 4,
+                                    hipMemcpyHostToDevice, s->stream));
+        TG_POOL_ALLOC(s, &out->data, offs[ng] ? offs[ng] : 1);
+        int64_t words = (ng + 63) / 64;
+        TG_POOL_ALLOC(s, &out->valid, (words ? words : 1) * 8);
+        TG_HIP_CHECK(hipMemsetAsync(out->valid, 0xFF, words * 8, s->stream));
+        hipLaunchKernelGGL(k_emit_var_bytes, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                           t, d_obn, ng, ch, out->offsets, (uint8_t*)out->data, out->valid);
+        TG_HIP_CHECK(hipGetLastError());
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        tg_pool_free(s, d_lens);
+        return TG_OK;
+    }
+
     tg_status emit() /* build the output page: groups in remapped id order */
     {
         int32_t ng = n_groups_host();
@@ -444,6 +585,13 @@ have_gids:;
         if (grid < 1) grid = 1;
         /* key channels */
         for (size_t c = 0; c < group_channels.size(); c++) {
+            if (group_types[c] == TG_VARCHAR) {
+                DevBlock b;
+                tg_status st = emit_varchar_key(ng, (int)c, d_obn, grid, &b);
+                if (st != TG_OK) return st;
+                outp.blocks.push_back(b);
+                continue;
+            }
             DevBlock b;
             b.type = group_types[c];
             b.n = ng;
@@ -520,6 +668,9 @@ have_gids:;
     ~HashAggOp() override
     {
         if (t.state) tg_pool_free(s, t.state);
+        if (t.varstore) (void)hipFree(t.varstore);
+        if (t.var_cursor) (void)hipFree(t.var_cursor);
+        if (d_group_types) (void)hipFree(d_group_types);
         if (t.keystore) tg_pool_free(s, t.keystore);
         if (t.first_row) tg_pool_free(s, t.first_row);
         if (t.counter) tg_pool_free(s, t.counter);
@@ -540,13 +691,6 @@ extern "C" tg_status tg_hash_aggregation_create(tg_session* s,
         n_aggs > MAX_AGGS || (n_group_channels > 0 && (!group_channels || !group_types))) {
         TG_SET_ERR("invalid hash aggregation spec (0..7 group channels, <=%d aggs)", MAX_AGGS);
         return TG_ERR_INVALID_ARG;
-    }
-    for (int i = 0; i < n_group_channels; i++) {
-        if (group_types[i] == TG_VARCHAR) {
-            TG_SET_ERR("VARCHAR group keys cross dictionary-encoded in round 1 "
-                       "(variable-width key store is round-2; DESIGN.md §7b)");
-            return TG_ERR_UNSUPPORTED;
-        }
     }
     auto* op = new HashAggOp();
     op->s = s;
@@ -571,6 +715,15 @@ extern "C" tg_status tg_hash_aggregation_create(tg_session* s,
             if (tg_pool_alloc(s, (void**)&k.cnt, mg * 8) != TG_OK) { delete op; return TG_ERR_OOM; }
         }
         op->agg_state.push_back(k);
+    }
+    if (!op->group_types.empty()) {
+        if (hipMalloc(&op->d_group_types, op->group_types.size() * 4) != hipSuccess ||
+            hipMemcpy(op->d_group_types, group_types, op->group_types.size() * 4,
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            delete op;
+            TG_SET_ERR("alloc group types");
+            return TG_ERR_OOM;
+        }
     }
     tg_status st = op->init_table(1 << 17, 1 << 16);
     if (st != TG_OK) { delete op; return st; }

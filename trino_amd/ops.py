@@ -215,14 +215,25 @@ def _download_page(session, page):
     for c in range(page.channel_count):
         b = page.blocks[c]
         n = b.position_count
+        valid = None
+        if b.valid:
+            valid = np.empty(max((n + 63) // 64, 1), np.uint64)
+            if n:
+                copy_dtoh(session, valid, b.valid)
+        if b.type == TG_VARCHAR:
+            offsets = np.empty(n + 1, np.int32)
+            copy_dtoh(session, offsets, b.offsets)
+            nbytes = int(offsets[n])
+            data = np.empty(max(nbytes, 1), np.uint8)
+            if nbytes:
+                copy_dtoh(session, data, b.data)
+            vals = [bytes(data[offsets[i]:offsets[i + 1]]) for i in range(n)]
+            cols.append({"values": vals, "valid": valid, "type": b.type,
+                         "offsets": offsets, "bytes": data[:nbytes]})
+            continue
         arr = np.empty(n, _TG2NP[b.type])
         if n:
             copy_dtoh(session, arr, b.data)
-        valid = None
-        if b.valid:
-            valid = np.empty((n + 63) // 64, np.uint64)
-            if n:
-                copy_dtoh(session, valid, b.valid)
         cols.append({"values": arr, "valid": valid, "type": b.type})
     return cols
 
