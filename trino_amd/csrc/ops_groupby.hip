@@ -725,6 +725,18 @@ extern "C" tg_status tg_hash_aggregation_create(tg_session* s,
             return TG_ERR_OOM;
         }
     }
+    bool any_var = false;
+    for (tg_type gt : op->group_types) any_var |= (gt == TG_VARCHAR);
+    if (any_var) {
+        op->t.var_capacity = 1 << 20;
+        if (hipMalloc(&op->t.varstore, op->t.var_capacity) != hipSuccess ||
+            hipMalloc(&op->t.var_cursor, 8) != hipSuccess ||
+            hipMemset(op->t.var_cursor, 0, 8) != hipSuccess) {
+            delete op;
+            TG_SET_ERR("alloc varchar key store");
+            return TG_ERR_OOM;
+        }
+    }
     tg_status st = op->init_table(1 << 17, 1 << 16);
     if (st != TG_OK) { delete op; return st; }
     *out = op;
