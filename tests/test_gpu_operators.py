@@ -670,3 +670,120 @@ class TestVarcharGroupBy:
         assert out[0]["values"] == [words[allp[i]] for i in first.tolist()]
         assert np.array_equal(out[1]["values"], keys[first])
         assert np.array_equal(out[2]["values"], oracle.grouped_count(gids, ng))
+
+
+class TestVarcharJoinAndGather:
+    """VARCHAR join keys (generic CSR path, VarcharType byte-equality) and
+    variable-width gather (VariableWidthBlock.copyPositions analog) through
+    join outputs and filter/project."""
+
+    def _varchar(self, words, pick):
+        offsets = np.zeros(len(pick) + 1, np.int32)
+        for i, k in enumerate(pick):
+            offsets[i + 1] = offsets[i] + len(words[k])
+        data = np.frombuffer(b"".join(words[k] for k in pick), np.uint8).copy()
+        return data, offsets
+
+    def test_varchar_join_key(self, sess, ops):
+        words = [b"F", b"O", b"NO", b"OF", b"", b"BUILDING", b"x" * 40]
+        r = rng(60)
+        nb, m = 5000, 20000
+        bpick = r.integers(0, len(words), nb).tolist()
+        bval = r.integers(0, 10**6, nb).astype(np.int64)
+        ppick = r.integers(0, len(words), m).tolist()
+        pval = r.integers(0, 10**6, m).astype(np.int64)
+
+        bridge = ops.JoinBridge(sess)
+        b = ops.hash_builder(sess, bridge, [ops.TG_VARCHAR, ops.TG_BIGINT], [0], [0, 1])
+        # two pages: exercises the varchar concat + offset rebase
+        bd1, bo1 = self._varchar(words, bpick[:3008])
+        bd2, bo2 = self._varchar(words, bpick[3008:])
+        b.add_input(ops.page_with_varchar([(bd1, bo1), bval[:3008]]))
+        b.add_input(ops.page_with_varchar([(bd2, bo2), bval[3008:]]))
+        b.drain()
+        j = ops.lookup_join(sess, bridge, [ops.TG_VARCHAR, ops.TG_BIGINT], [0], [0, 1])
+        pd, po = self._varchar(words, ppick)
+        j.add_input(ops.page_with_varchar([(pd, po), pval]))
+        jp, _ = j.get_output()
+        b.close()
+        j.close()
+        bridge.close()
+
+        got = sorted(zip(jp[0]["values"], jp[1]["values"].tolist(),
+                         jp[2]["values"], jp[3]["values"].tolist()))
+        by_key = {}
+        for k in range(nb):
+            by_key.setdefault(words[bpick[k]], []).append(k)
+        exp = sorted((words[ppick[i]], int(pval[i]), words[bpick[k]], int(bval[k]))
+                     for i in range(m) for k in by_key.get(words[ppick[i]], ()))
+        assert got == exp
+
+    def test_varchar_multikey_join(self, sess, ops):
+        # (varchar, bigint) composite key: prefixes must not cross-match
+        words = [b"A", b"AB", b"B", b"BA"]
+        r = rng(61)
+        nb, m = 2000, 8000
+        bpick = r.integers(0, len(words), nb).tolist()
+        bk2 = r.integers(0, 3, nb).astype(np.int64)
+        ppick = r.integers(0, len(words), m).tolist()
+        pk2 = r.integers(0, 3, m).astype(np.int64)
+        bridge = ops.JoinBridge(sess)
+        bd, bo = self._varchar(words, bpick)
+        b = ops.hash_builder(sess, bridge, [ops.TG_VARCHAR, ops.TG_BIGINT], [0, 1], [1])
+        b.add_input(ops.page_with_varchar([(bd, bo), bk2]))
+        b.drain()
+        j = ops.lookup_join(sess, bridge, [ops.TG_VARCHAR, ops.TG_BIGINT], [0, 1], [1])
+        pd, po = self._varchar(words, ppick)
+        j.add_input(ops.page_with_varchar([(pd, po), pk2]))
+        jp, _ = j.get_output()
+        b.close()
+        j.close()
+        bridge.close()
+        got = sorted(zip(jp[0]["values"].tolist(), jp[1]["values"].tolist()))
+        by_key = {}
+        for k in range(nb):
+            by_key.setdefault((words[bpick[k]], int(bk2[k])), []).append(k)
+        exp = sorted((int(pk2[i]), int(bk2[k]))
+                     for i in range(m)
+                     for k in by_key.get((words[ppick[i]], int(pk2[i])), ()))
+        assert got == exp
+
+    def test_varchar_semi_join(self, sess, ops):
+        words = [b"RAIL", b"AIR", b"MAIL", b"SHIP", b"TRUCK"]
+        r = rng(62)
+        build_pick = [0, 2]   # IN ('RAIL','MAIL')
+        m = 10000
+        ppick = r.integers(0, len(words), m).tolist()
+        bridge = ops.JoinBridge(sess)
+        bd, bo = self._varchar(words, build_pick)
+        b = ops.hash_builder(sess, bridge, [ops.TG_VARCHAR], [0], [])
+        b.add_input(ops.page_with_varchar([(bd, bo)]))
+        b.drain()
+        sj = ops.semi_join(sess, bridge, 0)
+        pd, po = self._varchar(words, ppick)
+        sj.add_input(ops.page_with_varchar([(pd, po)]))
+        sp, _ = sj.get_output()
+        b.close()
+        sj.close()
+        bridge.close()
+        exp = np.array([1 if ppick[i] in build_pick else 0 for i in range(m)], np.int8)
+        assert np.array_equal(sp[-1]["values"].astype(np.int8), exp)
+
+    def test_varchar_filter_project(self, sess, ops):
+        words = [b"AUTOMOBILE", b"", b"HOUSEHOLD", b"MACHINERY" * 3]
+        r = rng(63)
+        n = 30000
+        pick = r.integers(0, len(words), n).tolist()
+        keys = r.integers(0, 100, n).astype(np.int64)
+        data, offsets = self._varchar(words, pick)
+        page = ops.page_with_varchar([keys, (data, offsets)])
+        f = ops.expr(("col", 0), ("i64", 50), "lt")
+        fp = ops.filter_project(sess, f, [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                                [ops.TG_BIGINT, ops.TG_VARCHAR])
+        fp.add_input(page)
+        fp.finish()
+        out, _ = fp.get_output()
+        fp.close()
+        mask = keys < 50
+        assert np.array_equal(out[0]["values"], keys[mask])
+        assert out[1]["values"] == [words[pick[i]] for i in np.nonzero(mask)[0]]

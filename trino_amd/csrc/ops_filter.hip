@@ -431,14 +431,84 @@ __global__ void k_gather(const T* __restrict__ src, const uint64_t* __restrict__
     }
 }
 
+/* variable-width gather (VariableWidthBlock.copyPositions analog:
+ * spi/block/VariableWidthBlock.java:171-198 — lengths pass, offset prefix,
+ * byte copy pass) */
+__global__ void k_gather_var_lens(const int32_t* __restrict__ src_off,
+                                  const uint64_t* __restrict__ src_valid,
+                                  const int32_t* __restrict__ pos, int32_t n,
+                                  int32_t* __restrict__ lens,
+                                  uint64_t* __restrict__ out_valid)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        int32_t p = pos[i];
+        bool isnull = src_valid && !((src_valid[p >> 6] >> (p & 63)) & 1);
+        lens[i] = isnull ? 0 : src_off[p + 1] - src_off[p];
+        if (isnull && out_valid)
+            atomicAnd((unsigned long long*)&out_valid[i >> 6], ~(1ull << (i & 63)));
+    }
+}
+
+__global__ void k_gather_var_bytes(const uint8_t* __restrict__ src_bytes,
+                                   const int32_t* __restrict__ src_off,
+                                   const int32_t* __restrict__ pos, int32_t n,
+                                   const int32_t* __restrict__ out_off,
+                                   uint8_t* __restrict__ out_bytes)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        int32_t len = out_off[i + 1] - out_off[i];
+        const uint8_t* sp = src_bytes + src_off[pos[i]];
+        uint8_t* dp = out_bytes + out_off[i];
+        for (int32_t b = 0; b < len; b++) dp[b] = sp[b];
+    }
+}
+
+static tg_status run_gather_var(tg_session* s, const DevBlock& src,
+                                const int32_t* d_positions, int32_t count,
+                                DevBlock* out)
+{
+    out->type = TG_VARCHAR;
+    out->n = count;
+    TG_POOL_ALLOC(s, &out->offsets, (int64_t)(count + 1) * 4);
+    if (src.valid) {
+        int64_t words = (count + 63) / 64;
+        TG_POOL_ALLOC(s, &out->valid, (words ? words : 1) * 8);
+        TG_HIP_CHECK(hipMemsetAsync(out->valid, 0xFF, words * 8, s->stream));
+    }
+    int grid = tg_grid_for(count ? count : 1);
+    hipLaunchKernelGGL(k_gather_var_lens, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                       src.offsets, src.valid, d_positions, count,
+                       out->offsets, out->valid);
+    TG_HIP_CHECK(hipGetLastError());
+    int32_t* d_total = nullptr;
+    TG_POOL_ALLOC(s, &d_total, 4);
+    hipLaunchKernelGGL(k_scan_serial, dim3(1), dim3(64), 0, s->stream,
+                       out->offsets, count, d_total);
+    TG_HIP_CHECK(hipGetLastError());
+    int32_t total = 0;
+    TG_HIP_CHECK(hipMemcpyAsync(&total, d_total, 4, hipMemcpyDeviceToHost, s->stream));
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    TG_HIP_CHECK(hipMemcpyAsync(out->offsets + count, d_total, 4,
+                                hipMemcpyDeviceToDevice, s->stream));
+    TG_POOL_ALLOC(s, &out->data, total ? total : 1);
+    hipLaunchKernelGGL(k_gather_var_bytes, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                       (const uint8_t*)src.data, src.offsets, d_positions, count,
+                       out->offsets, (uint8_t*)out->data);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    tg_pool_free(s, d_total);
+    return TG_OK;
+}
+
 tg_status run_gather(tg_session* s, const DevBlock& src, const int32_t* d_positions,
                      int32_t count, DevBlock* out)
 {
-    if (src.type == TG_VARCHAR) {
-        TG_SET_ERR("VARCHAR payload gather is round-2 (hashing/partition-by-"
-                   "varchar works; materialize via dictionary ids for now)");
-        return TG_ERR_UNSUPPORTED;
-    }
+    if (src.type == TG_VARCHAR)
+        return run_gather_var(s, src, d_positions, count, out);
     out->type = src.type;
     out->n = count;
     TG_POOL_ALLOC(s, &out->data, (int64_t)(count ? count : 1) * src.elem_size());

@@ -88,9 +88,22 @@ __device__ static inline bool probe_matches(const JoinTable& t, const ProbeKeys&
                                             int64_t probe_row, int32_t build_row)
 {
     if (!t.generic) return t.keys[build_row] == p.pk[probe_row];
-    for (int c = 0; c < t.n_key_ch; c++)
-        if (kcol_word(t.bkeys[c], build_row) != kcol_word(p.pkg[c], probe_row))
+    for (int c = 0; c < t.n_key_ch; c++) {
+        const KColH& bc = t.bkeys[c];
+        const KColH& pc = p.pkg[c];
+        if (bc.type == TG_VARCHAR) {   /* VarcharType EQUAL: byte equality */
+            int32_t bo = bc.offsets[build_row];
+            int32_t bl = bc.offsets[build_row + 1] - bo;
+            int32_t po = pc.offsets[probe_row];
+            if (bl != pc.offsets[probe_row + 1] - po) return false;
+            const uint8_t* bb = (const uint8_t*)bc.data + bo;
+            const uint8_t* pb = (const uint8_t*)pc.data + po;
+            for (int32_t k = 0; k < bl; k++)
+                if (bb[k] != pb[k]) return false;
+        }
+        else if (kcol_word(bc, build_row) != kcol_word(pc, probe_row))
             return false;
+    }
     return true;
 }
 
@@ -472,6 +485,14 @@ static tg_status run_scan_counts(tg_session* s, const int32_t* d_counts, int64_t
     return TG_OK;
 }
 
+__global__ void k_off_rebase(const int32_t* __restrict__ src, int64_t n,
+                             int32_t base, int32_t* __restrict__ dst)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) dst[i] = src[i] + base;
+}
+
 /* ---- build operator ---- */
 struct HashBuilderOp : tg_operator {
     tg_join_bridge* bridge = nullptr;
@@ -479,6 +500,63 @@ struct HashBuilderOp : tg_operator {
     std::vector<tg_type> types;
     std::vector<DevPage> pages;     /* PagesIndex */
     int64_t total_rows = 0;
+
+    /* concatenate one VARCHAR channel across the accumulated pages
+     * (PagesIndex keeps per-page VariableWidthBlocks; the flat build-side
+     * copy rebases each page's offsets onto a single byte array) */
+    tg_status concat_varchar(size_t c, DevBlock* b)
+    {
+        int64_t total_bytes = 0;
+        std::vector<int32_t> page_bytes(pages.size());
+        for (size_t pi = 0; pi < pages.size(); pi++) {
+            const DevBlock& pb = pages[pi].blocks[c];
+            int32_t tb = 0;
+            if (pb.n) {
+                TG_HIP_CHECK(hipMemcpy(&tb, pb.offsets + pb.n, 4,
+                                       hipMemcpyDeviceToHost));
+            }
+            page_bytes[pi] = tb;
+            total_bytes += tb;
+        }
+        TG_POOL_ALLOC(s, &b->data, total_bytes ? total_bytes : 1);
+        TG_POOL_ALLOC(s, &b->offsets, (total_rows + 1) * 4);
+        bool anynull = false;
+        for (auto& p : pages) anynull |= p.blocks[c].valid != nullptr;
+        if (anynull) {
+            int64_t words = (total_rows + 63) / 64;
+            TG_POOL_ALLOC(s, &b->valid, words * 8);
+            TG_HIP_CHECK(hipMemsetAsync(b->valid, 0xFF, words * 8, s->stream));
+        }
+        int64_t at = 0, byte_at = 0;
+        for (size_t pi = 0; pi < pages.size(); pi++) {
+            const DevBlock& pb = pages[pi].blocks[c];
+            if (page_bytes[pi]) {
+                TG_HIP_CHECK(hipMemcpyAsync((char*)b->data + byte_at, pb.data,
+                                            page_bytes[pi],
+                                            hipMemcpyDeviceToDevice, s->stream));
+            }
+            if (pb.n) {
+                hipLaunchKernelGGL(k_off_rebase, dim3(tg_grid_for(pb.n)),
+                                   dim3(TG_BLOCK), 0, s->stream,
+                                   pb.offsets, pb.n, (int32_t)byte_at,
+                                   b->offsets + at);
+                TG_HIP_CHECK(hipGetLastError());
+            }
+            if (pb.valid) {
+                if (at % 64 != 0) { TG_SET_ERR("null bitmap concat requires 64-row-aligned pages"); return TG_ERR_UNSUPPORTED; }
+                TG_HIP_CHECK(hipMemcpyAsync((char*)b->valid + at / 8, pb.valid,
+                                            (pb.n + 63) / 64 * 8,
+                                            hipMemcpyDeviceToDevice, s->stream));
+            }
+            at += pb.n;
+            byte_at += page_bytes[pi];
+        }
+        int32_t tb32 = (int32_t)total_bytes;
+        TG_HIP_CHECK(hipMemcpyAsync(b->offsets + total_rows, &tb32, 4,
+                                    hipMemcpyHostToDevice, s->stream));
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        return TG_OK;
+    }
 
     tg_status add_input(const tg_page* page) override
     {
@@ -519,6 +597,11 @@ struct HashBuilderOp : tg_operator {
             DevBlock& b = bridge->build_channels[c];
             b.type = types[c];
             b.n = total_rows;
+            if (types[c] == TG_VARCHAR) {
+                tg_status vst = concat_varchar(c, &b);
+                if (vst != TG_OK) return vst;
+                continue;
+            }
             TG_POOL_ALLOC(s, &b.data, (total_rows ? total_rows : 1) * b.elem_size());
             int64_t at = 0;
             bool anynull = false;
@@ -811,12 +894,6 @@ extern "C" tg_status tg_hash_builder_create(tg_session* s, tg_join_bridge* bridg
         n_key_channels > 7) {
         TG_SET_ERR("1..7 join key channels");
         return TG_ERR_INVALID_ARG;
-    }
-    for (int i = 0; i < n_key_channels; i++) {
-        if (build_types[key_channels[i]] == TG_VARCHAR) {
-            TG_SET_ERR("VARCHAR join keys cross dictionary-encoded in round 1");
-            return TG_ERR_UNSUPPORTED;
-        }
     }
     auto* op = new HashBuilderOp();
     op->s = s;
