@@ -8,11 +8,12 @@
  *    key: values copied to a flat array, open addressing, murmur3 position,
  *    PagesHash.java:35-51), sizing IncrementalLoadFactorHashArraySizeSupplier
  *    .java:26-45 (0.25/0.5/0.75), duplicate keys chained through
- *    ArrayPositionLinks.java:24-45 (links[new]=old, new row becomes head;
- *    the GPU build pushes chains with atomicExch — chain order within equal
- *    keys is nondeterministic across workgroups; the match SET is identical
- *    and SQL results are unaffected. Deterministic chain order for the
- *    operator-parity tests comes from a single-workgroup build at test sizes).
+ *    ArrayPositionLinks.java:24-45 (links[new]=old, new row becomes head).
+ *    The default CSR build sorts each bucket by build row and the probe
+ *    walks buckets descending, reproducing the reference's reverse-insertion
+ *    duplicate order deterministically at every size. (The legacy
+ *    open-addressing path, TG_JOIN_CSR=0, pushes chains with atomicExch and
+ *    is only deterministic for single-workgroup builds.)
  *  - probe: LookupJoinOperator / DefaultPageJoiner.java:243-296 +
  *    JoinProbe.java:112-180 batch shape and DefaultPagesHash.java:193-280
  *    (hash all -> gather all -> verify -> probe misses); null keys never
@@ -163,103 +164,51 @@ __global__ void k_key_minmax(const int64_t* __restrict__ keys,
     atomicAdd(cnt, lc);
 }
 
-/* ---- CSR build (atomic-free): hierarchical partition by slot region,
- * then per-region LDS histogram + scan + scatter. Replaces the global
- * atomic-CAS insert (measured ~1 G CAS/s = 15 ms per 15M-row build). ---- */
+/* ---- CSR build: count -> per-region LDS exclusive scan -> parallel
+ * atomic scatter -> per-bucket sort by build row. Replaces the global
+ * atomic-CAS insert (measured ~1 G CAS/s = 15 ms per 15M-row build) and
+ * the earlier chunk-ordered scatter (serial per chunk with a global-memory
+ * cursor RMW per row: 16 ms/step on Q3 SF100). The ascending in-bucket sort
+ * plus the descending probe walk reproduces ArrayPositionLinks' reverse-
+ * insertion duplicate order (ArrayPositionLinks.java:24-45) DETERMINISTICALLY
+ * at every size. ---- */
 #define JREG_SLOTS 32768              /* slots per region (LDS u32 counts) */
-#define JPART_CHUNK 1024
 
-__global__ void k_jc_slots(JoinTable t, uint32_t* __restrict__ slot_of,
-                           int32_t* __restrict__ region_hist /* [nchunks][nreg] */,
-                           int64_t nreg, int64_t region_slots)
+__global__ void k_jc_count(JoinTable t, uint32_t* __restrict__ slot_of,
+                           int32_t* __restrict__ bucket_cnt /* zeroed [cap] */)
 {
-    /* chunk-private region histograms (no atomics) + slot precompute */
-    int64_t c = blockIdx.x;
-    int64_t lo = c * JPART_CHUNK, hi = min(lo + JPART_CHUNK, t.n);
-    extern __shared__ int32_t lhist[];          /* [nreg] */
-    for (int64_t r = threadIdx.x; r < nreg; r += blockDim.x) lhist[r] = 0;
-    __syncthreads();
-    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < t.n; i += stride) {
         bool null = jt_build_null(t, i);
         uint32_t slot = null ? 0xFFFFFFFFu : jt_build_slot(t, i);
         slot_of[i] = slot;
-        if (!null) atomicAdd(&lhist[slot / region_slots], 1);
-    }
-    __syncthreads();
-    for (int64_t r = threadIdx.x; r < nreg; r += blockDim.x)
-        region_hist[c * nreg + r] = lhist[r];
-}
-
-__global__ void k_jc_hist_local(int32_t* region_hist, int64_t nchunks, int64_t nreg,
-                                int32_t* __restrict__ region_total)
-{
-    /* thread per region: in-place exclusive prefix over its chunk column */
-    int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (r >= nreg) return;
-    int32_t run = 0;
-    for (int64_t c = 0; c < nchunks; c++) {
-        int32_t v = region_hist[c * nreg + r];
-        region_hist[c * nreg + r] = run;
-        run += v;
-    }
-    region_total[r] = run;
-}
-
-__global__ void k_jc_hist_addbase(int32_t* region_hist, int64_t nchunks, int64_t nreg,
-                                  const int32_t* __restrict__ region_off)
-{
-    int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (r >= nreg) return;
-    int32_t b = region_off[r];
-    for (int64_t c = 0; c < nchunks; c++) region_hist[c * nreg + r] += b;
-}
-
-__global__ void k_jc_scatter(JoinTable t, const uint32_t* __restrict__ slot_of,
-                             int32_t* __restrict__ region_hist, int64_t nreg,
-                             int64_t region_slots,
-                             uint32_t* __restrict__ part_slot,
-                             int32_t* __restrict__ part_row)
-{
-    /* chunk-owned sequential scatter preserves row order within a chunk */
-    int64_t c = blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t nchunks = (t.n + JPART_CHUNK - 1) / JPART_CHUNK;
-    if (c >= nchunks) return;
-    int64_t lo = c * JPART_CHUNK, hi = min(lo + JPART_CHUNK, t.n);
-    for (int64_t i = lo; i < hi; i++) {
-        uint32_t slot = slot_of[i];
-        if (slot == 0xFFFFFFFFu) continue;      /* null key: never indexed */
-        int64_t reg = slot / region_slots;
-        int32_t at = region_hist[c * nreg + reg]++;
-        part_slot[at] = slot;
-        part_row[at] = (int32_t)i;
+        if (!null) atomicAdd(&bucket_cnt[slot], 1);
     }
 }
 
 __global__ __launch_bounds__(TG_BLOCK)
-void k_jc_fill(JoinTable t, const int32_t* __restrict__ region_off,
-               const uint32_t* __restrict__ part_slot,
-               const int32_t* __restrict__ part_row,
-               int64_t nreg, int64_t region_slots)
+void k_jc_local_off(const int32_t* __restrict__ bucket_cnt,
+                    int32_t* __restrict__ bucket_off,
+                    int32_t* __restrict__ region_total,
+                    int64_t nreg, int64_t region_slots)
 {
-    /* one block per region: LDS per-slot counts -> scan -> bucket offsets +
-     * row scatter. region_slots <= JREG_SLOTS (128 KB LDS of u32). */
+    /* one block per region: region-local exclusive prefix of the per-slot
+     * counts (LDS staged; region_slots <= JREG_SLOTS = 128 KB of u32) */
     int64_t reg = blockIdx.x;
     if (reg >= nreg) return;
     __shared__ int32_t cnt[JREG_SLOTS];
     __shared__ int32_t wpart[TG_BLOCK / 64];
-    int32_t lo = region_off[reg], hi = region_off[reg + 1];
     int64_t sbase = reg * region_slots;
-    for (int64_t sidx = threadIdx.x; sidx < region_slots; sidx += blockDim.x) cnt[sidx] = 0;
-    __syncthreads();
-    for (int32_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
-        atomicAdd(&cnt[part_slot[i] - sbase], 1);
-    __syncthreads();
-    /* block exclusive scan over region_slots: per-thread serial + hierarchy */
     const int per = (int)((region_slots + blockDim.x - 1) / blockDim.x);
     int32_t mysum = 0;
     for (int k = 0; k < per; k++) {
         int64_t sidx = (int64_t)threadIdx.x * per + k;
-        if (sidx < region_slots) mysum += cnt[sidx];
+        if (sidx < region_slots) {
+            int32_t v = bucket_cnt[sbase + sidx];
+            cnt[sidx] = v;
+            mysum += v;
+        }
     }
     int32_t wpre = mysum;
     #pragma unroll
@@ -276,24 +225,57 @@ void k_jc_fill(JoinTable t, const int32_t* __restrict__ region_off,
             wpart[w] = run;
             run += v;
         }
+        region_total[reg] = run;
     }
     __syncthreads();
-    int32_t base = lo + wpart[threadIdx.x / 64] + wpre - mysum;
-    /* write bucket offsets and turn cnt[] into running write cursors */
+    int32_t base = wpart[threadIdx.x / 64] + wpre - mysum;
     for (int k = 0; k < per; k++) {
         int64_t sidx = (int64_t)threadIdx.x * per + k;
         if (sidx >= region_slots) break;
         int32_t v = cnt[sidx];
-        t.bucket_off[sbase + sidx] = base;
-        cnt[sidx] = base;
+        bucket_off[sbase + sidx] = base;
         base += v;
     }
-    __syncthreads();
-    for (int32_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-        int32_t at = atomicAdd(&cnt[part_slot[i] - sbase], 1);
-        t.csr_rows[at] = part_row[i];
+}
+
+__global__ void k_jc_addbase(int32_t* __restrict__ bucket_off, int64_t capacity,
+                             const int32_t* __restrict__ region_base,
+                             int64_t region_slots)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < capacity; i += stride)
+        bucket_off[i] += region_base[i / region_slots];
+}
+
+__global__ void k_jc_scatter(JoinTable t, const uint32_t* __restrict__ slot_of,
+                             int32_t* __restrict__ cursor /* copy of bucket_off */)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < t.n; i += stride) {
+        uint32_t slot = slot_of[i];
+        if (slot == 0xFFFFFFFFu) continue;      /* null key: never indexed */
+        int32_t at = atomicAdd(&cursor[slot], 1);
+        t.csr_rows[at] = (int32_t)i;
     }
-    if (reg == nreg - 1 && threadIdx.x == 0) t.bucket_off[t.capacity] = hi;
+}
+
+__global__ void k_jc_sort(JoinTable t)
+{
+    /* ascending insertion sort within each bucket (expected length n/cap
+     * < 1; the tail is a handful of duplicates sharing a slot) */
+    int64_t s0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t sl = s0; sl < t.capacity; sl += stride) {
+        int32_t lo = t.bucket_off[sl], hi = t.bucket_off[sl + 1];
+        for (int32_t a = lo + 1; a < hi; a++) {
+            int32_t v = t.csr_rows[a];
+            int32_t b = a - 1;
+            for (; b >= lo && t.csr_rows[b] > v; b--) t.csr_rows[b + 1] = t.csr_rows[b];
+            t.csr_rows[b + 1] = v;
+        }
+    }
 }
 
 __global__ void k_join_build(JoinTable t)
@@ -664,55 +646,52 @@ struct HashBuilderOp : tg_operator {
             t.csr = 1;
             int64_t region_slots = t.capacity < JREG_SLOTS ? t.capacity : JREG_SLOTS;
             int64_t nreg = t.capacity / region_slots;
-            int64_t nchunks = (total_rows + JPART_CHUNK - 1) / JPART_CHUNK;
             TG_POOL_ALLOC(s, &t.bucket_off, (t.capacity + 1) * 4);
             TG_POOL_ALLOC(s, &t.csr_rows, total_rows * 4);
             uint32_t* d_slot_of = nullptr;
-            int32_t* d_hist = nullptr;
+            int32_t* d_cnt = nullptr;       /* per-slot counts, then cursors */
             int32_t* d_rtotal = nullptr;
-            int32_t* d_roff = nullptr;
-            uint32_t* d_pslot = nullptr;
-            int32_t* d_prow = nullptr;
+            int32_t* d_rbase = nullptr;
             TG_POOL_ALLOC(s, &d_slot_of, total_rows * 4);
-            TG_POOL_ALLOC(s, &d_hist, nchunks * nreg * 4);
+            TG_POOL_ALLOC(s, &d_cnt, t.capacity * 4);
             TG_POOL_ALLOC(s, &d_rtotal, nreg * 4);
-            TG_POOL_ALLOC(s, &d_roff, (nreg + 1) * 4);
-            TG_POOL_ALLOC(s, &d_pslot, total_rows * 4);
-            TG_POOL_ALLOC(s, &d_prow, total_rows * 4);
-            hipLaunchKernelGGL(k_jc_slots, dim3((uint32_t)nchunks), dim3(TG_BLOCK),
-                               (size_t)nreg * 4, s->stream, t, d_slot_of, d_hist,
+            TG_POOL_ALLOC(s, &d_rbase, nreg * 4);
+            TG_HIP_CHECK(hipMemsetAsync(d_cnt, 0, t.capacity * 4, s->stream));
+            hipLaunchKernelGGL(k_jc_count, dim3(tg_grid_for(total_rows)), dim3(TG_BLOCK),
+                               0, s->stream, t, d_slot_of, d_cnt);
+            TG_HIP_CHECK(hipGetLastError());
+            hipLaunchKernelGGL(k_jc_local_off, dim3((uint32_t)nreg), dim3(TG_BLOCK),
+                               0, s->stream, d_cnt, t.bucket_off, d_rtotal,
                                nreg, region_slots);
             TG_HIP_CHECK(hipGetLastError());
-            hipLaunchKernelGGL(k_jc_hist_local, dim3(tg_grid_for(nreg)), dim3(TG_BLOCK),
-                               0, s->stream, d_hist, nchunks, nreg, d_rtotal);
-            TG_HIP_CHECK(hipGetLastError());
-            std::vector<int32_t> rt(nreg), ro(nreg + 1);
+            std::vector<int32_t> rt(nreg), rb(nreg);
             TG_HIP_CHECK(hipMemcpyAsync(rt.data(), d_rtotal, nreg * 4,
                                         hipMemcpyDeviceToHost, s->stream));
             TG_HIP_CHECK(hipStreamSynchronize(s->stream));
             int32_t run = 0;
-            for (int64_t r = 0; r < nreg; r++) { ro[r] = run; run += rt[r]; }
-            ro[nreg] = run;
-            TG_HIP_CHECK(hipMemcpyAsync(d_roff, ro.data(), (nreg + 1) * 4,
+            for (int64_t r = 0; r < nreg; r++) { rb[r] = run; run += rt[r]; }
+            int32_t total_indexed = run;    /* == total_rows minus null keys */
+            TG_HIP_CHECK(hipMemcpyAsync(d_rbase, rb.data(), nreg * 4,
                                         hipMemcpyHostToDevice, s->stream));
-            hipLaunchKernelGGL(k_jc_hist_addbase, dim3(tg_grid_for(nreg)), dim3(TG_BLOCK),
-                               0, s->stream, d_hist, nchunks, nreg, d_roff);
+            hipLaunchKernelGGL(k_jc_addbase, dim3(tg_grid_for(t.capacity)), dim3(TG_BLOCK),
+                               0, s->stream, t.bucket_off, t.capacity, d_rbase,
+                               region_slots);
             TG_HIP_CHECK(hipGetLastError());
-            hipLaunchKernelGGL(k_jc_scatter, dim3(tg_grid_for(nchunks)), dim3(TG_BLOCK),
-                               0, s->stream, t, d_slot_of, d_hist, nreg, region_slots,
-                               d_pslot, d_prow);
+            TG_HIP_CHECK(hipMemcpyAsync(t.bucket_off + t.capacity, &total_indexed, 4,
+                                        hipMemcpyHostToDevice, s->stream));
+            TG_HIP_CHECK(hipMemcpyAsync(d_cnt, t.bucket_off, t.capacity * 4,
+                                        hipMemcpyDeviceToDevice, s->stream));
+            hipLaunchKernelGGL(k_jc_scatter, dim3(tg_grid_for(total_rows)), dim3(TG_BLOCK),
+                               0, s->stream, t, d_slot_of, d_cnt);
             TG_HIP_CHECK(hipGetLastError());
-            hipLaunchKernelGGL(k_jc_fill, dim3((uint32_t)nreg), dim3(TG_BLOCK),
-                               0, s->stream, t, d_roff, d_pslot, d_prow,
-                               nreg, region_slots);
+            hipLaunchKernelGGL(k_jc_sort, dim3(tg_grid_for(t.capacity)), dim3(TG_BLOCK),
+                               0, s->stream, t);
             TG_HIP_CHECK(hipGetLastError());
             TG_HIP_CHECK(hipStreamSynchronize(s->stream));
             tg_pool_free(s, d_slot_of);
-            tg_pool_free(s, d_hist);
+            tg_pool_free(s, d_cnt);
             tg_pool_free(s, d_rtotal);
-            tg_pool_free(s, d_roff);
-            tg_pool_free(s, d_pslot);
-            tg_pool_free(s, d_prow);
+            tg_pool_free(s, d_rbase);
         }
         else {
             hipLaunchKernelGGL(k_join_init, dim3(tg_grid_for(t.capacity)), dim3(TG_BLOCK),
