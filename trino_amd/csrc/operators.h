@@ -106,3 +106,7 @@ tg_status run_gather(tg_session* s, const DevBlock& src, const int32_t* d_positi
 tg_status run_hash_rows(tg_session* s, const DevPage& page,
                         const int32_t* channels, int32_t n_channels,
                         uint64_t* d_hashes);
+
+/* device radix-sort helpers (sort.hip; rocPRIM-backed, stable) */
+tg_status run_sort_pairs(tg_session* s, uint64_t* d_keys, int64_t* d_vals, int64_t n);
+tg_status run_argsort_i64(tg_session* s, int64_t* d_keys, int64_t n, int32_t* d_out_idx);

@@ -563,18 +563,27 @@ have_gids:;
         int32_t ng = n_groups_host();
         /* remap: order groups by first-occurrence row (== the reference's
          * insertion-order ids, deterministically) */
-        std::vector<int64_t> first(ng);
-        if (ng) {
-            TG_HIP_CHECK(hipMemcpy(first.data(), t.first_row, (size_t)ng * 8,
-                                   hipMemcpyDeviceToHost));
-        }
-        std::vector<int32_t> old_by_new(ng);
-        std::iota(old_by_new.begin(), old_by_new.end(), 0);
-        std::sort(old_by_new.begin(), old_by_new.end(),
-                  [&](int32_t a, int32_t b) { return first[a] < first[b]; });
         int32_t* d_obn = nullptr;
         TG_POOL_ALLOC(s, &d_obn, (ng ? ng : 1) * 4);
-        if (ng) {
+        if (ng >= 65536) {
+            /* device argsort (first_row keys are unique non-negative rows);
+             * host std::sort at Q3's ~1.1M groups cost tens of ms per step */
+            int64_t* d_fr = nullptr;
+            TG_POOL_ALLOC(s, &d_fr, (int64_t)ng * 8);
+            TG_HIP_CHECK(hipMemcpyAsync(d_fr, t.first_row, (int64_t)ng * 8,
+                                        hipMemcpyDeviceToDevice, s->stream));
+            tg_status sst = run_argsort_i64(s, d_fr, ng, d_obn);
+            if (sst != TG_OK) return sst;
+            tg_pool_free(s, d_fr);
+        }
+        else if (ng) {
+            std::vector<int64_t> first(ng);
+            TG_HIP_CHECK(hipMemcpy(first.data(), t.first_row, (size_t)ng * 8,
+                                   hipMemcpyDeviceToHost));
+            std::vector<int32_t> old_by_new(ng);
+            std::iota(old_by_new.begin(), old_by_new.end(), 0);
+            std::sort(old_by_new.begin(), old_by_new.end(),
+                      [&](int32_t a, int32_t b) { return first[a] < first[b]; });
             TG_HIP_CHECK(hipMemcpyAsync(d_obn, old_by_new.data(), (size_t)ng * 4,
                                         hipMemcpyHostToDevice, s->stream));
         }

@@ -1,0 +1,74 @@
+/* sort.hip — device radix sort helpers (rocPRIM) shared by the group-by
+ * output remap (HashAggOp::emit) and the TopN operator.
+ *
+ * Both previously sorted on the host (std::sort / std::partial_sort over
+ * ~10^6 entries = tens of ms per Q3 step, dwarfing the kernels they fed).
+ * rocprim::radix_sort_pairs is stable, which the multi-key TopN relies on
+ * (least-significant-key-first passes == lexicographic order).
+ */
+#include "common.h"
+#include "operators.h"
+#include <rocprim/rocprim.hpp>
+
+__global__ void k_iota_i64(int64_t* v, int64_t n)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) v[i] = i;
+}
+
+__global__ void k_i64_to_i32(const int64_t* __restrict__ in, int64_t n,
+                             int32_t* __restrict__ out)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) out[i] = (int32_t)in[i];
+}
+
+/* stable ascending sort of (u64 key, i64 value) pairs; keys/vals updated
+ * in place (double-buffered internally) */
+tg_status run_sort_pairs(tg_session* s, uint64_t* d_keys, int64_t* d_vals, int64_t n)
+{
+    if (n <= 1) return TG_OK;
+    uint64_t* d_keys2 = nullptr;
+    int64_t* d_vals2 = nullptr;
+    TG_POOL_ALLOC(s, &d_keys2, n * 8);
+    TG_POOL_ALLOC(s, &d_vals2, n * 8);
+    size_t temp_bytes = 0;
+    hipError_t e0 = rocprim::radix_sort_pairs(nullptr, temp_bytes, d_keys, d_keys2,
+                                              d_vals, d_vals2, (size_t)n, 0, 64,
+                                              s->stream);
+    if (e0 != hipSuccess) { TG_SET_ERR("rocprim size query: %s", hipGetErrorName(e0)); return TG_ERR_HIP; }
+    void* d_temp = nullptr;
+    TG_POOL_ALLOC(s, &d_temp, (int64_t)temp_bytes);
+    hipError_t e = rocprim::radix_sort_pairs(d_temp, temp_bytes, d_keys, d_keys2,
+                                             d_vals, d_vals2, (size_t)n, 0, 64,
+                                             s->stream);
+    if (e != hipSuccess) { TG_SET_ERR("rocprim radix_sort_pairs: %s", hipGetErrorName(e)); return TG_ERR_HIP; }
+    TG_HIP_CHECK(hipMemcpyAsync(d_keys, d_keys2, n * 8, hipMemcpyDeviceToDevice, s->stream));
+    TG_HIP_CHECK(hipMemcpyAsync(d_vals, d_vals2, n * 8, hipMemcpyDeviceToDevice, s->stream));
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    tg_pool_free(s, d_keys2);
+    tg_pool_free(s, d_vals2);
+    tg_pool_free(s, d_temp);
+    return TG_OK;
+}
+
+/* argsort of non-negative i64 keys (ascending, stable): writes the
+ * permutation as int32 into d_out_idx[n]. d_keys is clobbered. */
+tg_status run_argsort_i64(tg_session* s, int64_t* d_keys, int64_t n, int32_t* d_out_idx)
+{
+    int64_t* d_vals = nullptr;
+    TG_POOL_ALLOC(s, &d_vals, (n ? n : 1) * 8);
+    hipLaunchKernelGGL(k_iota_i64, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
+                       d_vals, n);
+    TG_HIP_CHECK(hipGetLastError());
+    tg_status st = run_sort_pairs(s, (uint64_t*)d_keys, d_vals, n);
+    if (st != TG_OK) return st;
+    hipLaunchKernelGGL(k_i64_to_i32, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
+                       d_vals, n, d_out_idx);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    tg_pool_free(s, d_vals);
+    return TG_OK;
+}
