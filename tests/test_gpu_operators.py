@@ -787,3 +787,34 @@ class TestVarcharJoinAndGather:
         mask = keys < 50
         assert np.array_equal(out[0]["values"], keys[mask])
         assert out[1]["values"] == [words[pick[i]] for i in np.nonzero(mask)[0]]
+
+
+class TestTopNDevicePath:
+    """n >= 65536 routes TopN through the device radix-sort path
+    (ops_topn.hip emit_device); parity vs numpy lexsort."""
+
+    def test_large_topn_composite(self, sess, ops):
+        r = rng(70)
+        n = 200_000
+        rev = r.uniform(0, 10**6, n)           # effectively unique f64
+        date = r.integers(8000, 10000, n).astype(np.int32)
+        okey = np.arange(n, dtype=np.int64)
+        page = ops.page_from_numpy([okey, date, rev])
+        top = ops.topn(sess, [ops.TG_BIGINT, ops.TG_INTEGER, ops.TG_DOUBLE],
+                       [2, 1], [1, 0], 25)    # revenue DESC, date ASC
+        top.add_input(page)
+        out = top.drain()[0]
+        order = np.lexsort((date, -rev))[:25]
+        assert np.array_equal(out[0]["values"], okey[order])
+        assert np.array_equal(out[1]["values"], date[order])
+        assert np.array_equal(out[2]["values"], rev[order])
+
+    def test_large_topn_asc_int(self, sess, ops):
+        r = rng(71)
+        n = 100_000
+        v = r.integers(-10**9, 10**9, n).astype(np.int64)
+        page = ops.page_from_numpy([v])
+        top = ops.topn(sess, [ops.TG_BIGINT], [0], [0], 40)
+        top.add_input(page)
+        out = top.drain()[0]
+        assert np.array_equal(out[0]["values"], np.sort(v)[:40])
