@@ -330,9 +330,14 @@ __global__ void k_probe_count(JoinTable t, ProbeKeys p, int64_t m,
         if (!probe_null(t, p, i)) {
             int64_t slot = (int64_t)probe_slot(t, p, i);
             if (t.csr) {
-                for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1]; x++)
-                    cnt += probe_matches(t, p, i, t.csr_rows[x]);
-                head = (int32_t)slot;
+                int32_t matched = -1;
+                for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1]; x++) {
+                    int32_t row = t.csr_rows[x];
+                    if (probe_matches(t, p, i, row)) { cnt++; matched = row; }
+                }
+                /* single match (the usual case: unique build keys): encode
+                 * the row as -2-row so the fill pass skips the bucket walk */
+                head = (cnt == 1) ? -2 - matched : (cnt == 0 ? -1 : (int32_t)slot);
             }
             else {   /* legacy open addressing: single-BIGINT only */
                 int64_t key = p.pk[i];
@@ -367,7 +372,12 @@ __global__ void k_probe_fill(JoinTable t, ProbeKeys p, int64_t m,
     for (; i < m; i += stride) {
         int64_t at = offsets[i];
         int32_t h = heads[i];
-        if (h < 0) continue;
+        if (h == -1) continue;
+        if (h <= -2) {           /* pre-resolved single match */
+            out_probe[at] = (int32_t)i;
+            out_build[at] = -2 - h;
+            continue;
+        }
         if (t.csr) {
             for (int32_t x = t.bucket_off[h + 1] - 1; x >= t.bucket_off[h]; x--) {
                 int32_t row = t.csr_rows[x];
