@@ -23,6 +23,16 @@
  */
 #include "dev_hash.h"
 
+/* inline single-entry bucket (the common case: unique build keys): one
+ * aligned 16-byte random read resolves most probes instead of three
+ * (bucket_off, csr_rows, keys). row -1 = empty bucket, -2 = bucket with
+ * >1 rows (fall back to the CSR walk). Built for single-BIGINT-key tables. */
+struct SlotKV {
+    int64_t key;
+    int32_t row;
+    int32_t _pad;
+};
+
 struct JoinTable {
     int64_t n = 0;               /* build rows */
     int64_t capacity = 0, mask = 0;
@@ -42,6 +52,7 @@ struct JoinTable {
     int32_t generic = 0;
     int32_t n_key_ch = 1;
     const KColH* bkeys = nullptr;   /* device array[n_key_ch] */
+    const SlotKV* kv = nullptr;     /* single-key fast table or null */
 };
 
 /* slot of a build row; generic keys hash with the canonical row hash
@@ -278,6 +289,20 @@ __global__ void k_jc_sort(JoinTable t)
     }
 }
 
+__global__ void k_build_slotkv(JoinTable t, SlotKV* __restrict__ kv)
+{
+    int64_t s0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t sl = s0; sl < t.capacity; sl += stride) {
+        int32_t lo = t.bucket_off[sl], hi = t.bucket_off[sl + 1];
+        SlotKV e;
+        if (hi - lo == 1) { e.row = t.csr_rows[lo]; e.key = t.keys[e.row]; }
+        else { e.row = (hi == lo) ? -1 : -2; e.key = 0; }
+        e._pad = 0;
+        kv[sl] = e;
+    }
+}
+
 __global__ void k_join_build(JoinTable t)
 {
     /* Probe with PLAIN loads and resolve races with device-scope atomics:
@@ -329,7 +354,23 @@ __global__ void k_probe_count(JoinTable t, ProbeKeys p, int64_t m,
         int32_t head = -1;
         if (!probe_null(t, p, i)) {
             int64_t slot = (int64_t)probe_slot(t, p, i);
-            if (t.csr) {
+            if (t.kv) {          /* one 16B read resolves most probes */
+                SlotKV e = t.kv[slot];
+                if (e.row >= 0) {
+                    head = (e.key == p.pk[i]) ? -2 - e.row : -1;
+                    cnt = (head != -1);
+                }
+                else if (e.row == -1) head = -1;
+                else {               /* multi-row bucket: CSR walk */
+                    int32_t matched = -1;
+                    for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1]; x++) {
+                        int32_t row = t.csr_rows[x];
+                        if (probe_matches(t, p, i, row)) { cnt++; matched = row; }
+                    }
+                    head = (cnt == 1) ? -2 - matched : (cnt == 0 ? -1 : (int32_t)slot);
+                }
+            }
+            else if (t.csr) {
                 int32_t matched = -1;
                 for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1]; x++) {
                     int32_t row = t.csr_rows[x];
@@ -697,6 +738,14 @@ struct HashBuilderOp : tg_operator {
             hipLaunchKernelGGL(k_jc_sort, dim3(tg_grid_for(t.capacity)), dim3(TG_BLOCK),
                                0, s->stream, t);
             TG_HIP_CHECK(hipGetLastError());
+            if (!t.generic) {
+                SlotKV* d_kv = nullptr;
+                TG_POOL_ALLOC(s, &d_kv, t.capacity * (int64_t)sizeof(SlotKV));
+                hipLaunchKernelGGL(k_build_slotkv, dim3(tg_grid_for(t.capacity)),
+                                   dim3(TG_BLOCK), 0, s->stream, t, d_kv);
+                TG_HIP_CHECK(hipGetLastError());
+                t.kv = d_kv;
+            }
             TG_HIP_CHECK(hipStreamSynchronize(s->stream));
             tg_pool_free(s, d_slot_of);
             tg_pool_free(s, d_cnt);
@@ -865,6 +914,7 @@ extern "C" void tg_join_bridge_close(tg_join_bridge* b)
     if (b->t.key_valid) tg_pool_free(b->s, b->t.key_valid);
     if (b->t.bucket_off) tg_pool_free(b->s, b->t.bucket_off);
     if (b->t.csr_rows) tg_pool_free(b->s, b->t.csr_rows);
+    if (b->t.kv) tg_pool_free(b->s, (void*)b->t.kv);
     if (b->d_bkeys) tg_pool_free(b->s, b->d_bkeys);
     for (auto& c : b->build_channels) {
         if (c.data) tg_pool_free(b->s, c.data);
@@ -911,7 +961,15 @@ __global__ void k_semi_probe(JoinTable t, ProbeKeys p, int64_t m,
         }
         int64_t slot = (int64_t)probe_slot(t, p, i);
         int8_t hit = 0;
-        if (t.csr) {
+        if (t.kv) {
+            SlotKV e = t.kv[slot];
+            if (e.row >= 0) hit = (e.key == p.pk[i]);
+            else if (e.row == -2) {
+                for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1] && !hit; x++)
+                    hit = probe_matches(t, p, i, t.csr_rows[x]);
+            }
+        }
+        else if (t.csr) {
             for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1] && !hit; x++)
                 hit = probe_matches(t, p, i, t.csr_rows[x]);
         }
