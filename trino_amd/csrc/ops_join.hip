@@ -738,7 +738,12 @@ struct HashBuilderOp : tg_operator {
             hipLaunchKernelGGL(k_jc_sort, dim3(tg_grid_for(t.capacity)), dim3(TG_BLOCK),
                                0, s->stream, t);
             TG_HIP_CHECK(hipGetLastError());
-            if (!t.generic) {
+            /* A/B on MI355X (Q3 SF100): the 512 MB kv array RAISED probe
+             * time (43.1 vs 39.7 ms per 10 launches) — the CSR arrays it
+             * replaces are smaller and L2-cache better. Kept behind
+             * TG_JOIN_KV=1 for small-table workloads. */
+            static int use_kv = [] { const char* e = getenv("TG_JOIN_KV"); return e ? atoi(e) : 0; }();
+            if (!t.generic && use_kv) {
                 SlotKV* d_kv = nullptr;
                 TG_POOL_ALLOC(s, &d_kv, t.capacity * (int64_t)sizeof(SlotKV));
                 hipLaunchKernelGGL(k_build_slotkv, dim3(tg_grid_for(t.capacity)),
