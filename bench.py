@@ -288,7 +288,7 @@ def run_q3(args, sess, n_gpus, rank):
                      "note": "whole-pipeline rate against the single-pass 28 B/row "
                              "probe-side bound (SURVEY §8d); the pipeline makes "
                              "multiple passes and its builds/probes are random-access "
-                             "bound — per-kernel evidence: profiles/r01_q3_kernels_v2.txt"},
+                             "bound — per-kernel evidence: profiles/r01_q3_kernels_v5.txt"},
         "cpu_baseline": None,
     }
     if rank == 0:
