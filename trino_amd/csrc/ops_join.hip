@@ -44,6 +44,10 @@ struct JoinTable {
      * and are filtered by the key compare (no linear probing). */
     int32_t* bucket_off = nullptr;  /* [capacity+1] */
     int32_t* csr_rows = nullptr;    /* [n] */
+    int64_t* csr_keys = nullptr;    /* [n] keys in bucket order (single-key
+                                       tables): the probe compare reads one
+                                       slot-local line instead of the random
+                                       csr_rows -> keys chain */
     int64_t* keys = nullptr;     /* [n] flat copy of build keys */
     uint64_t* key_valid = nullptr; /* packed bitmap or null */
     /* generic multi-channel keys (DefaultPagesHash analog; the fields above
@@ -289,6 +293,14 @@ __global__ void k_jc_sort(JoinTable t)
     }
 }
 
+__global__ void k_csr_keys(JoinTable t, int64_t* __restrict__ ck)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t tot = t.bucket_off[t.capacity];   /* indexed rows (null keys excluded) */
+    for (; i < tot; i += stride) ck[i] = t.keys[t.csr_rows[i]];
+}
+
 __global__ void k_build_slotkv(JoinTable t, SlotKV* __restrict__ kv)
 {
     int64_t s0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -370,6 +382,15 @@ __global__ void k_probe_count(JoinTable t, ProbeKeys p, int64_t m,
                     head = (cnt == 1) ? -2 - matched : (cnt == 0 ? -1 : (int32_t)slot);
                 }
             }
+            else if (t.csr && t.csr_keys) {   /* single-key: slot-local compare */
+                int64_t key = p.pk[i];
+                int32_t matched = -1;
+                for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1]; x++) {
+                    if (t.csr_keys[x] == key) { cnt++; matched = x; }
+                }
+                head = (cnt == 1) ? -2 - t.csr_rows[matched]
+                                  : (cnt == 0 ? -1 : (int32_t)slot);
+            }
             else if (t.csr) {
                 int32_t matched = -1;
                 for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1]; x++) {
@@ -419,7 +440,17 @@ __global__ void k_probe_fill(JoinTable t, ProbeKeys p, int64_t m,
             out_build[at] = -2 - h;
             continue;
         }
-        if (t.csr) {
+        if (t.csr && t.csr_keys) {
+            int64_t key = p.pk[i];
+            for (int32_t x = t.bucket_off[h + 1] - 1; x >= t.bucket_off[h]; x--) {
+                if (t.csr_keys[x] == key) {
+                    out_probe[at] = (int32_t)i;
+                    out_build[at] = t.csr_rows[x];
+                    at++;
+                }
+            }
+        }
+        else if (t.csr) {
             for (int32_t x = t.bucket_off[h + 1] - 1; x >= t.bucket_off[h]; x--) {
                 int32_t row = t.csr_rows[x];
                 if (probe_matches(t, p, i, row)) {
@@ -738,6 +769,12 @@ struct HashBuilderOp : tg_operator {
             hipLaunchKernelGGL(k_jc_sort, dim3(tg_grid_for(t.capacity)), dim3(TG_BLOCK),
                                0, s->stream, t);
             TG_HIP_CHECK(hipGetLastError());
+            if (!t.generic && total_rows > 0) {
+                TG_POOL_ALLOC(s, &t.csr_keys, total_rows * 8);
+                hipLaunchKernelGGL(k_csr_keys, dim3(tg_grid_for(total_rows)),
+                                   dim3(TG_BLOCK), 0, s->stream, t, t.csr_keys);
+                TG_HIP_CHECK(hipGetLastError());
+            }
             /* A/B on MI355X (Q3 SF100): the 512 MB kv array RAISED probe
              * time (43.1 vs 39.7 ms per 10 launches) — the CSR arrays it
              * replaces are smaller and L2-cache better. Kept behind
@@ -919,6 +956,7 @@ extern "C" void tg_join_bridge_close(tg_join_bridge* b)
     if (b->t.key_valid) tg_pool_free(b->s, b->t.key_valid);
     if (b->t.bucket_off) tg_pool_free(b->s, b->t.bucket_off);
     if (b->t.csr_rows) tg_pool_free(b->s, b->t.csr_rows);
+    if (b->t.csr_keys) tg_pool_free(b->s, b->t.csr_keys);
     if (b->t.kv) tg_pool_free(b->s, (void*)b->t.kv);
     if (b->d_bkeys) tg_pool_free(b->s, b->d_bkeys);
     for (auto& c : b->build_channels) {
@@ -973,6 +1011,11 @@ __global__ void k_semi_probe(JoinTable t, ProbeKeys p, int64_t m,
                 for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1] && !hit; x++)
                     hit = probe_matches(t, p, i, t.csr_rows[x]);
             }
+        }
+        else if (t.csr && t.csr_keys) {
+            int64_t key = p.pk[i];
+            for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1] && !hit; x++)
+                hit = (t.csr_keys[x] == key);
         }
         else if (t.csr) {
             for (int32_t x = t.bucket_off[slot]; x < t.bucket_off[slot + 1] && !hit; x++)
