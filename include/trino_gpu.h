@@ -149,9 +149,22 @@ typedef enum {
     TG_AGG_COUNT_COL  = 1,
     TG_AGG_SUM_F64    = 2,  /* DoubleSumAggregation.java:37-45 */
     TG_AGG_SUM_I64    = 3,  /* BigintSumAggregation */
-    TG_AGG_AVG_F64    = 4   /* DoubleAverageAggregations.java:38-63 (state: count,sum) */
+    TG_AGG_AVG_F64    = 4,  /* DoubleAverageAggregations.java:38-63 (state: count,sum) */
+    /* exact fixed-point double sum: the caller asserts value*2^scale_pow is
+     * an integer for every input (true for TPC-H money/discount columns,
+     * DESIGN.md §4) and |sum*2^scale_pow| < 2^127. State is a 128-bit
+     * integer accumulated with carry-propagating 64-bit atomics:
+     * order-INDEPENDENT, so the result is deterministic and correctly
+     * rounded — stronger than the reference's DoubleSumAggregation, equal
+     * to its decimal SUM semantics. */
+    TG_AGG_SUM_F64_EXACT = 5
 } tg_agg_fn;
-typedef struct tg_agg_spec { int32_t fn; int32_t input_channel; } tg_agg_spec;
+typedef struct tg_agg_spec {
+    int32_t fn;
+    int32_t input_channel;
+    int32_t scale_pow;      /* TG_AGG_SUM_F64_EXACT only; else 0 */
+    int32_t _pad;
+} tg_agg_spec;
 
 tg_status tg_hash_aggregation_create(tg_session*,
     const int32_t* group_channels, int32_t n_group_channels,

@@ -818,3 +818,68 @@ class TestTopNDevicePath:
         top.add_input(page)
         out = top.drain()[0]
         assert np.array_equal(out[0]["values"], np.sort(v)[:40])
+
+
+class TestExactFixedPointSum:
+    """TG_AGG_SUM_F64_EXACT: order-independent 128-bit fixed-point sum —
+    bit-equal to the oracle's exact leg and bit-stable across page splits
+    (plain SUM_F64's atomicAdd order is not)."""
+
+    def test_bit_exact_vs_oracle(self, sess, ops):
+        r = rng(80)
+        n = 300_000
+        keys = r.integers(0, 400, n).astype(np.int64)
+        # money-scaled values: v*2^9 integer (cents * 2^-2 grid like Q1 prices)
+        vals = r.integers(90100, 209900, n) / 100.0
+        op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                  [(ops.AGG_SUM_F64_EXACT, 1, 43)])
+        op.add_input(ops.page_from_numpy([keys, vals]))
+        out = op.drain()[0]
+        op.close()
+        gids, ng, by_gid, _ = oracle.bigint_groupby(keys)
+        exp = oracle.grouped_sum_f64_exact(gids, vals, ng, scale_pow=43)
+        assert np.array_equal(out[1]["values"], exp)   # bit-equal
+
+    def test_split_invariance(self, sess, ops):
+        r = rng(81)
+        n = 200_000
+        keys = r.integers(0, 101, n).astype(np.int64)
+        vals = r.integers(0, 11, n) / 100.0            # discounts: v*2^59 integer
+        results = []
+        for splits in ([n], [64, n - 64], [1024, 77056, n - 78080]):
+            op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                      [(ops.AGG_SUM_F64_EXACT, 1, 59)])
+            at = 0
+            for sz in splits:
+                op.add_input(ops.page_from_numpy([keys[at:at + sz], vals[at:at + sz]]))
+                at += sz
+            out = op.drain()[0]
+            op.close()
+            results.append((np.asarray(out[0]["values"]), np.asarray(out[1]["values"])))
+        for k2, v2 in results[1:]:
+            assert np.array_equal(results[0][0], k2)
+            assert np.array_equal(results[0][1], v2)   # bit-stable
+
+    def test_partial_final_combine(self, sess, ops):
+        r = rng(82)
+        n = 120_000
+        keys = r.integers(0, 37, n).astype(np.int64)
+        vals = r.integers(90100, 209900, n) / 100.0
+        part = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                    [(ops.AGG_SUM_F64_EXACT, 1, 43)],
+                                    step=ops.STEP_PARTIAL)
+        part.add_input(ops.page_from_numpy([keys, vals]))
+        pout = part.drain()[0]
+        part.close()
+        fin = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                   [(ops.AGG_SUM_F64_EXACT, 1, 43)],
+                                   step=ops.STEP_FINAL)
+        fin.add_input(ops.page_from_numpy([np.asarray(pout[0]["values"]),
+                                           np.asarray(pout[1]["values"]),
+                                           np.asarray(pout[2]["values"])]))
+        out = fin.drain()[0]
+        fin.close()
+        gids, ng, by_gid, _ = oracle.bigint_groupby(keys)
+        exp = oracle.grouped_sum_f64_exact(gids, vals, ng, scale_pow=43)
+        assert np.array_equal(out[0]["values"], by_gid)
+        assert np.array_equal(out[1]["values"], exp)

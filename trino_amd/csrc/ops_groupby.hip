@@ -14,9 +14,12 @@
  *    (DoubleSumAggregation.java:37-45, DoubleAverageAggregations.java:38-63).
  *
  * Determinism/parity: group keys, ids, counts and integer
- * sums are bit-exact. f64 sums use device atomicAdd (the reference's own
- * cross-driver combine order is nondeterministic too); operator parity tests
- * state their tolerance; the flagship Q1 path (q1.hip) is exact fixed-point.
+ * sums are bit-exact. Plain f64 SUM uses device atomicAdd (the reference's
+ * own cross-driver combine order is nondeterministic too); operator parity
+ * tests state their tolerance. TG_AGG_SUM_F64_EXACT extends the fused Q1
+ * kernel's fixed-point trick to the generic operator: 128-bit integer state
+ * accumulated with carry-propagating u64 atomics — order-independent, so
+ * bit-stable across page splits and grids, and correctly rounded.
  *
  * Table layout: open addressing, power-of-2, per slot an int32 state
  * (-1 empty / -2 claimed / else gid) and a key store of canonical 64-bit
@@ -179,9 +182,11 @@ __global__ void k_gt_rehash(GTable t, int32_t n_groups, const int32_t* types, in
 /* ---- aggregation update ---- */
 struct KAgg {
     int32_t fn;           /* tg_agg_fn */
-    int32_t in_ch;        /* -1 for COUNT_STAR; AVG FINAL: count ch, sum = ch+1 */
-    double* sum;          /* f64 state (or null) */
-    long long* cnt;       /* i64 state (count / int sum) */
+    int32_t in_ch;        /* -1 for COUNT_STAR; AVG FINAL: count ch, sum = ch+1;
+                             SUM_F64_EXACT FINAL: lo ch, hi = ch+1 */
+    double* sum;          /* f64 state (or null); SUM_F64_EXACT: i128 HI words */
+    long long* cnt;       /* i64 state (count / int sum); SUM_F64_EXACT: LO words */
+    double scale;         /* SUM_F64_EXACT: 2^scale_pow */
 };
 #define MAX_AGGS 12
 
@@ -225,6 +230,29 @@ __global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
                         atomicAdd(&ag.sum[g], v);
                     }
                     break;
+                case TG_AGG_SUM_F64_EXACT: {
+                    /* order-independent exact sum: addend*2^scale is an
+                     * integer (caller contract, trino_gpu.h); accumulate a
+                     * 128-bit integer with carry-propagating u64 atomics —
+                     * every interleaving yields the same bits */
+                    unsigned long long vlo, vhi;
+                    if (step == 0) {
+                        if (kcol_is_null(cols[ag.in_ch], i)) break;
+                        double y = ((const double*)cols[ag.in_ch].data)[i] * ag.scale;
+                        __int128 yi = (__int128)(long long)y;
+                        vlo = (unsigned long long)(unsigned __int128)yi;
+                        vhi = (unsigned long long)((unsigned __int128)yi >> 64);
+                    }
+                    else {   /* FINAL: channels (lo, hi) */
+                        vlo = (unsigned long long)((const int64_t*)cols[ag.in_ch].data)[i];
+                        vhi = (unsigned long long)((const int64_t*)cols[ag.in_ch + 1].data)[i];
+                    }
+                    unsigned long long old =
+                        atomicAdd((unsigned long long*)&ag.cnt[g], vlo);
+                    unsigned long long carry = (old + vlo) < vlo ? 1ull : 0ull;
+                    atomicAdd((unsigned long long*)ag.sum + g, vhi + carry);
+                    break;
+                }
                 case TG_AGG_AVG_F64:
                     if (step == 0) {
                         if (!kcol_is_null(cols[ag.in_ch], i)) {
@@ -312,6 +340,18 @@ __global__ void k_emit_i64(const long long* __restrict__ state, const int32_t* o
 {
     int32_t g = blockIdx.x * blockDim.x + threadIdx.x;
     if (g < n) out[g] = state[old_by_new[g]];
+}
+
+__global__ void k_emit_exact(const unsigned long long* __restrict__ lo,
+                             const unsigned long long* __restrict__ hi,
+                             const int32_t* __restrict__ old_by_new, int32_t n,
+                             double inv_scale, double* __restrict__ out)
+{
+    int32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n) return;
+    int32_t og = old_by_new[g];
+    __int128 v = (__int128)(((unsigned __int128)hi[og] << 64) | lo[og]);
+    out[g] = (double)v * inv_scale;   /* i128->f64 correctly rounded, *2^-k exact */
 }
 
 __global__ void k_emit_avg(const double* __restrict__ sum, const long long* __restrict__ cnt,
@@ -641,6 +681,28 @@ have_gids:;
                                    ag.sum, ag.cnt, d_obn, ng, (double*)b.data, b.valid);
                 outp.blocks.push_back(b);
             }
+            else if (ag.fn == TG_AGG_SUM_F64_EXACT && final_out) {
+                DevBlock b; b.type = TG_DOUBLE; b.n = ng;
+                TG_POOL_ALLOC(s, &b.data, (int64_t)(ng ? ng : 1) * 8);
+                hipLaunchKernelGGL(k_emit_exact, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                                   (const unsigned long long*)ag.cnt,
+                                   (const unsigned long long*)ag.sum, d_obn, ng,
+                                   1.0 / ag.scale, (double*)b.data);
+                outp.blocks.push_back(b);
+            }
+            else if (ag.fn == TG_AGG_SUM_F64_EXACT) {
+                /* PARTIAL state: (lo BIGINT, hi BIGINT) channel pair */
+                DevBlock bl; bl.type = TG_BIGINT; bl.n = ng;
+                TG_POOL_ALLOC(s, &bl.data, (int64_t)(ng ? ng : 1) * 8);
+                hipLaunchKernelGGL(k_emit_i64, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                                   ag.cnt, d_obn, ng, (int64_t*)bl.data);
+                outp.blocks.push_back(bl);
+                DevBlock bh; bh.type = TG_BIGINT; bh.n = ng;
+                TG_POOL_ALLOC(s, &bh.data, (int64_t)(ng ? ng : 1) * 8);
+                hipLaunchKernelGGL(k_emit_i64, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                                   (const long long*)ag.sum, d_obn, ng, (int64_t*)bh.data);
+                outp.blocks.push_back(bh);
+            }
             else if (ag.fn == TG_AGG_SUM_F64) {
                 DevBlock b; b.type = TG_DOUBLE; b.n = ng;
                 TG_POOL_ALLOC(s, &b.data, (int64_t)(ng ? ng : 1) * 8);
@@ -714,8 +776,11 @@ extern "C" tg_status tg_hash_aggregation_create(tg_session* s,
         KAgg k{};
         k.fn = aggs[a].fn;
         k.in_ch = aggs[a].input_channel;
-        bool needs_sum = (k.fn == TG_AGG_SUM_F64 || k.fn == TG_AGG_AVG_F64);
+        bool needs_sum = (k.fn == TG_AGG_SUM_F64 || k.fn == TG_AGG_AVG_F64 ||
+                          k.fn == TG_AGG_SUM_F64_EXACT);
         bool needs_cnt = (k.fn != TG_AGG_SUM_F64);
+        k.scale = 1.0;
+        for (int32_t sp_ = 0; sp_ < aggs[a].scale_pow; sp_++) k.scale *= 2.0;
         int64_t mg = 1 << 16;
         if (needs_sum) {
             if (tg_pool_alloc(s, (void**)&k.sum, mg * 8) != TG_OK) { delete op; return TG_ERR_OOM; }

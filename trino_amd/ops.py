@@ -11,7 +11,8 @@ from . import _lib, _check, Session
 
 TG_BIGINT, TG_INTEGER, TG_SMALLINT, TG_TINYINT, TG_DOUBLE, TG_DATE, TG_BOOLEAN, TG_VARCHAR = range(8)
 STEP_PARTIAL, STEP_FINAL, STEP_SINGLE = 0, 1, 2
-AGG_COUNT_STAR, AGG_COUNT_COL, AGG_SUM_F64, AGG_SUM_I64, AGG_AVG_F64 = range(5)
+(AGG_COUNT_STAR, AGG_COUNT_COL, AGG_SUM_F64, AGG_SUM_I64, AGG_AVG_F64,
+ AGG_SUM_F64_EXACT) = range(6)
 
 _NP2TG = {np.dtype(np.int64): TG_BIGINT, np.dtype(np.int32): TG_INTEGER,
           np.dtype(np.int16): TG_SMALLINT, np.dtype(np.int8): TG_TINYINT,
@@ -57,7 +58,8 @@ class TgExpr(ctypes.Structure):
 
 
 class TgAggSpec(ctypes.Structure):
-    _fields_ = [("fn", ctypes.c_int32), ("input_channel", ctypes.c_int32)]
+    _fields_ = [("fn", ctypes.c_int32), ("input_channel", ctypes.c_int32),
+                ("scale_pow", ctypes.c_int32), ("_pad", ctypes.c_int32)]
 
 
 (OP_COL, OP_CONST_F64, OP_CONST_I64, OP_ADD, OP_SUB, OP_MUL, OP_DIV,
@@ -269,9 +271,10 @@ def hash_aggregation(session, group_channels, group_types, aggs, step=STEP_SINGL
     gc = _i32arr(group_channels)
     gt = _i32arr(group_types)
     sp = (TgAggSpec * len(aggs))()
-    for i, (fn, ch) in enumerate(aggs):
-        sp[i].fn = fn
-        sp[i].input_channel = ch
+    for i, a in enumerate(aggs):
+        sp[i].fn = a[0]
+        sp[i].input_channel = a[1]
+        sp[i].scale_pow = a[2] if len(a) > 2 else 0
     _check(_lib.tg_hash_aggregation_create(session._h, gc.ctypes.data, len(gc),
                                            gt.ctypes.data, sp, len(aggs), step,
                                            ctypes.byref(h)))
