@@ -124,8 +124,11 @@ def q3_execute(session, inp, download_groups=True):
     j2.finish()
     joined = _take_device_page(session, j2)
 
+    # revenue = sum(ep*(1-disc)): products >= ~810 (= 2^9 < v < 2^11) sit on
+    # the 2^-43 grid -> TG_AGG_SUM_F64_EXACT is exact AND order-independent
+    # (DESIGN.md §4's Q1 argument applied to the generic operator)
     agg = ops.hash_aggregation(session, [0, 2], [ops.TG_BIGINT, ops.TG_INTEGER],
-                               [(ops.AGG_SUM_F64, 1)])
+                               [(ops.AGG_SUM_F64_EXACT, 1, 43)])
     agg.add_input(joined)
     agg.finish()
     agg_page = _take_device_page(session, agg)
@@ -229,7 +232,8 @@ def q6_gpu(session, sf, order_start=1, order_count=None):
     fp.add_input(lpage)
     fp.finish()
     sel = _take_device_page(session, fp)
-    agg = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_F64, 0),
+    # ep*disc >= ~45 (2^5): exact on the 2^-47 grid
+    agg = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_F64_EXACT, 0, 47),
                                                 (ops.AGG_COUNT_STAR, -1)])
     agg.add_input(sel)
     pages = agg.drain()
@@ -357,7 +361,7 @@ def q14_gpu(session, sf, order_start=1, order_count=None):
     j.finish()
     joined = _take_device_page(session, j)     # (discprice, p_type)
     # total revenue
-    a1 = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_F64, 0)])
+    a1 = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_F64_EXACT, 0, 43)])
     a1.add_input(joined)
     total = a1.drain()[0][0]["values"][0]
     # promo (CASE WHEN p_type LIKE 'PROMO%'): type ids 125..149
@@ -366,7 +370,7 @@ def q14_gpu(session, sf, order_start=1, order_count=None):
     f2.add_input(joined)
     f2.finish()
     promo_page = _take_device_page(session, f2)
-    a2 = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_F64, 0)])
+    a2 = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_F64_EXACT, 0, 43)])
     a2.add_input(promo_page)
     promo = a2.drain()[0][0]["values"][0]
     elapsed = time.time() - t0
