@@ -108,7 +108,8 @@ def main():
     if world > 1:
         import torch.distributed as tdist
         dist = tdist
-        torch.cuda.set_device(local_rank)
+        # modulo: lets world_size-2 validation runs share a 1-GPU box
+        torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
         tdist.init_process_group(backend="nccl")
 
     import trino_amd
@@ -116,7 +117,7 @@ def main():
     sess = trino_amd.Session(local_rank)
 
     if args.workload == "q3":
-        run_q3(args, sess, n_gpus, rank)
+        run_q3(args, sess, n_gpus, rank, dist)
         sess.close()
         if dist:
             dist.destroy_process_group()
@@ -250,25 +251,55 @@ def main():
         dist.destroy_process_group()
 
 
-def run_q3(args, sess, n_gpus, rank):
-    """BASELINE config 3: TPC-H Q3 3-way join build/probe, single GPU.
+def run_q3(args, sess, n_gpus, rank, tdist=None):
+    """BASELINE config 3: TPC-H Q3 3-way join build/probe.
     Timed region = the full operator pipeline (2 joins + agg + topn) over
-    HBM-resident inputs; metric = probe-side lineitem rows per second."""
+    HBM-resident inputs; metric = probe-side lineitem rows per second
+    (whole-job aggregate over ranks).
+
+    N>1 (weak scaling, SF = sf_per_gpu x N): rank r generates customer/
+    orders/lineitem for ITS contiguous key ranges; orders⨝lineitem is
+    co-partitioned by construction; the small filtered-customer build is
+    BROADCAST (all_gather union — trino_amd/dist.py) inside the timed
+    region, like the reference's replicated join; per-rank TopNs merge at
+    the end (orderkey ranges are rank-disjoint)."""
     import torch
     from trino_amd import tpch_queries
+    from trino_amd import dist as tgdist
+    world = tdist.get_world_size() if tdist else 1
     sf = args.sf_per_gpu
+    order_count = int(1_500_000 * sf)
+    cust_count = int(150_000 * sf)
     t0 = time.time()
-    inp = tpch_queries.q3_prepare(sess, sf)
+    inp = tpch_queries.q3_prepare(
+        sess, sf * world,
+        order_start=rank * order_count + 1, order_count=order_count,
+        cust_start=rank * cust_count + 1, cust_count=cust_count)
     rows = inp["li"].row_count
-    log(f"q3 inputs ready: {rows:,} lineitem rows, SF{sf:g}, {time.time()-t0:.1f}s")
+    log(f"q3 inputs ready: {rows:,} lineitem rows/rank, SF{sf * world:g} "
+        f"({world} ranks), {time.time()-t0:.1f}s")
+    hook = (lambda keys: tgdist.gather_union(tdist, keys)) if world > 1 else None
     for _ in range(args.warmup):
-        tpch_queries.q3_execute(sess, inp, download_groups=False)
+        tpch_queries.q3_execute(sess, inp, download_groups=False,
+                                cust_key_exchange=hook)
     torch.cuda.synchronize()
+    if tdist:
+        tdist.barrier()
     t_start = time.time()
     for _ in range(args.steps):
-        r = tpch_queries.q3_execute(sess, inp, download_groups=False)
+        r = tpch_queries.q3_execute(sess, inp, download_groups=False,
+                                    cust_key_exchange=hook)
     torch.cuda.synchronize()
     elapsed = time.time() - t_start
+    if tdist:
+        t_el = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        tdist.all_reduce(t_el, op=tdist.ReduceOp.MAX)
+        elapsed = float(t_el.item())           # max over ranks
+        t_rows = torch.tensor([rows], dtype=torch.float64, device="cuda")
+        tdist.all_reduce(t_rows, op=tdist.ReduceOp.SUM)
+        rows = int(t_rows.item())              # whole-job rows
+        r["top10"] = tgdist.merge_topn(tdist, r["top10"], 10,
+                                       key=lambda t: (-t[1], t[2]))
     tpch_queries.q3_release(sess, inp)
     ms = elapsed * 1000 / args.steps
     value = rows / (elapsed / args.steps)
@@ -278,7 +309,9 @@ def run_q3(args, sess, n_gpus, rank):
         "steps": args.steps, "warmup": args.warmup, "ms_per_step": ms,
         "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
         "dtype": "f64", "data": "synthetic",
-        "config": {"workload": f"TPC-H Q3 (3-way hash join build/probe) SF{sf:g} on 1xMI355X",
+        "config": {"workload": f"TPC-H Q3 (3-way hash join build/probe) "
+                               f"SF{sf * (tdist.get_world_size() if tdist else 1):g} "
+                               f"on {n_gpus}xMI355X",
                    "rows_lineitem": rows, "parallelism": f"dp{n_gpus}",
                    "top10_first": r["top10"][0] if r["top10"] else None},
         "roofline": {"bound": "hbm", "achieved": rows * 28 / (ms / 1000) / 1e9,

@@ -91,3 +91,44 @@ def test_gloo_exchange_world2():
                 assert m[c * 10 + 9] == 3
             else:
                 assert base == 0 and m[c * 10 + 8] == 0
+
+
+def _union_worker(rank, world, port, out):
+    import numpy as np
+    import torch.distributed as tdist
+    import os
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    tdist.init_process_group("gloo", rank=rank, world_size=world)
+    from trino_amd import dist as tgdist
+    local = np.arange(rank * 3, rank * 3 + 2 + rank, dtype=np.int64)
+    u = tgdist.gather_union(tdist, local)
+    top = tgdist.merge_topn(tdist, [(rank, float(10 - rank), rank)], 2,
+                            key=lambda t: (-t[1], t[2]))
+    out.put((rank, u.tolist(), top))
+    tdist.destroy_process_group()
+
+
+def test_gather_union_and_topn_merge_gloo():
+    """trino_amd.dist broadcast-build union + TopN merge (Q3 N>1 legs) on
+    gloo, world_size 2 (bench runs the same calls over RCCL)."""
+    import multiprocessing as mp
+    import socket
+    ctx = mp.get_context("spawn")
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_union_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in ps:
+        p.start()
+    res = {}
+    for _ in range(2):
+        r, u, top = q.get(timeout=120)
+        res[r] = (u, top)
+    for p in ps:
+        p.join(timeout=60)
+    exp_union = [0, 1, 3, 4, 5]
+    for r in range(2):
+        assert res[r][0] == exp_union
+        assert res[r][1] == [(0, 10.0, 0), (1, 9.0, 1)]

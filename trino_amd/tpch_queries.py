@@ -59,8 +59,13 @@ def q3_release(session, inp):
         _device_free(session, inp[k])
 
 
-def q3_execute(session, inp, download_groups=True):
-    """The timed Q3 pipeline over prepared device inputs."""
+def q3_execute(session, inp, download_groups=True, cust_key_exchange=None):
+    """The timed Q3 pipeline over prepared device inputs.
+
+    cust_key_exchange: distributed hook (trino_amd.dist.gather_union bound to
+    the process group) — receives this rank's filtered BUILDING custkeys and
+    returns the union across ranks (broadcast/replicated join build,
+    DetermineJoinDistributionType BROADCAST). None = single-GPU."""
     cust_ck, cust_ms, cust_count = inp["cust_ck"], inp["cust_ms"], inp["cust_count"]
     o_ok, o_ck, o_od, order_count = inp["o_ok"], inp["o_ck"], inp["o_od"], inp["order_count"]
     li = inp["li"]
@@ -73,11 +78,16 @@ def q3_execute(session, inp, download_groups=True):
                             [ops.expr(("col", 0))], [ops.TG_BIGINT])
     f1.add_input(cpage)
     f1.finish()
-    cust_sel = _take_device_page(session, f1)
-
     bridge1 = ops.JoinBridge(session)
     b1 = ops.hash_builder(session, bridge1, [ops.TG_BIGINT], [0], [])
-    b1.add_input(cust_sel)
+    if cust_key_exchange is None:
+        cust_sel = _take_device_page(session, f1)
+        b1.add_input(cust_sel)
+    else:
+        host_sel, _ = f1.get_output()     # downloads the selected custkeys
+        local_keys = host_sel[0]["values"] if host_sel else np.empty(0, np.int64)
+        all_keys = cust_key_exchange(local_keys)
+        b1.add_input(ops.page_from_numpy([np.ascontiguousarray(all_keys)]))
     b1.drain()
 
     # ---- stage 2: orders filter + join customers ----
