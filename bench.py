@@ -96,6 +96,9 @@ def main():
     ap.add_argument("--sf-per-gpu", type=float, default=100.0)
     ap.add_argument("--workload", choices=["q1", "q3"], default="q1")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--backend", choices=["nccl", "gloo"], default="nccl",
+                    help="gloo only for multi-rank validation on one GPU "
+                         "(RCCL requires distinct devices per rank)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -110,7 +113,7 @@ def main():
         dist = tdist
         # modulo: lets world_size-2 validation runs share a 1-GPU box
         torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
-        tdist.init_process_group(backend="nccl")
+        tdist.init_process_group(backend=args.backend)
 
     import trino_amd
 
@@ -292,10 +295,11 @@ def run_q3(args, sess, n_gpus, rank, tdist=None):
     torch.cuda.synchronize()
     elapsed = time.time() - t_start
     if tdist:
-        t_el = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        dev = "cuda" if tdist.get_backend() == "nccl" else "cpu"
+        t_el = torch.tensor([elapsed], dtype=torch.float64, device=dev)
         tdist.all_reduce(t_el, op=tdist.ReduceOp.MAX)
         elapsed = float(t_el.item())           # max over ranks
-        t_rows = torch.tensor([rows], dtype=torch.float64, device="cuda")
+        t_rows = torch.tensor([rows], dtype=torch.float64, device=dev)
         tdist.all_reduce(t_rows, op=tdist.ReduceOp.SUM)
         rows = int(t_rows.item())              # whole-job rows
         r["top10"] = tgdist.merge_topn(tdist, r["top10"], 10,
