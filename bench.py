@@ -112,7 +112,8 @@ def main():
         import torch.distributed as tdist
         dist = tdist
         # modulo: lets world_size-2 validation runs share a 1-GPU box
-        torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
+        local_rank = local_rank % max(1, torch.cuda.device_count())
+        torch.cuda.set_device(local_rank)
         tdist.init_process_group(backend=args.backend)
 
     import trino_amd
