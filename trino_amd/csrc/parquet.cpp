@@ -490,13 +490,21 @@ extern "C" tg_status tg_parquet_read_column(tg_session* s, tg_parquet_file* f, i
         int64_t words = (f->num_rows + 63) / 64;
         memset(out_valid, 0xFF, (size_t)words * 8);
     }
-    int64_t row = 0;
     /* BYTE_ARRAY dictionary accumulated across row groups: per-row-group
      * dictionaries are remapped into one global dictionary by value */
     std::vector<uint8_t> gdict_bytes;
     std::vector<int32_t> gdict_offsets{0};
 
-    for (auto& rg : f->row_groups) {
+    /* per-row-group start rows: row groups are independent for numeric
+     * columns, so they decode in parallel (OpenMP); BYTE_ARRAY columns stay
+     * sequential (the global dictionary remap is order-dependent) */
+    std::vector<int64_t> rg_base(f->row_groups.size() + 1, 0);
+    for (size_t g = 0; g < f->row_groups.size(); g++)
+        rg_base[g + 1] = rg_base[g] + f->row_groups[g].num_rows;
+
+    auto read_group = [&](size_t gi) -> tg_status {
+        auto& rg = f->row_groups[gi];
+        int64_t row = rg_base[gi];
         const PqColumnMeta& cm = rg.cols[col];
         int64_t off = cm.dict_page_offset >= 0 &&
                       (cm.data_page_offset < 0 || cm.dict_page_offset < cm.data_page_offset)
@@ -657,7 +665,23 @@ extern "C" tg_status tg_parquet_read_column(tg_session* s, tg_parquet_file* f, i
             row += nv;
             remaining -= nv;
         }
+        return TG_OK;
+    };
+
+    tg_status gst = TG_OK;
+    if (is_ba) {
+        for (size_t gi = 0; gi < f->row_groups.size() && gst == TG_OK; gi++)
+            gst = read_group(gi);
     }
+    else {
+        #pragma omp parallel for schedule(dynamic)
+        for (int64_t gi = 0; gi < (int64_t)f->row_groups.size(); gi++) {
+            if (gst != TG_OK) continue;
+            tg_status st = read_group((size_t)gi);
+            if (st != TG_OK) gst = st;   /* benign race: any error wins */
+        }
+    }
+    if (gst != TG_OK) return gst;
     if (is_ba) {
         int32_t dc = (int32_t)gdict_offsets.size() - 1;
         if (out_dict_count) *out_dict_count = dc;
