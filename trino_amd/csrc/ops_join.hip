@@ -355,6 +355,79 @@ __global__ void k_join_build(JoinTable t)
 /* probe phase 1 (DefaultPagesHash.getAddressIndex hash-all/gather-all/verify
  * shape): count matches per probe row, caching the matched head so the fill
  * pass never re-probes the table */
+/* ---- partitioned single-pass probe (large single-BIGINT-key tables) ----
+ * The two-pass count/fill walk reads 2-3 random cache lines per probe over a
+ * table far larger than L2. Instead: bucket-partition the PROBE ROWS by slot
+ * region (so consecutive rows hit a table slice that stays cache-resident),
+ * walk each bucket once appending (probe_row, build_row) matches, then
+ * restore probe-row order with a stable radix sort — the reference emits
+ * matches in probe position order (DefaultPageJoiner), duplicates in
+ * reverse-insertion order (descending walk + stable sort preserves it). */
+__global__ void k_pb_slots(JoinTable t, ProbeKeys p, int64_t m, int64_t nparts,
+                           int shift, uint32_t* __restrict__ slot_of,
+                           int32_t* __restrict__ pcount)
+{
+    extern __shared__ int32_t lh[];
+    for (int64_t k = threadIdx.x; k < nparts; k += blockDim.x) lh[k] = 0;
+    __syncthreads();
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < m; i += stride) {
+        uint32_t slot = probe_null(t, p, i) ? 0xFFFFFFFFu : probe_slot(t, p, i);
+        slot_of[i] = slot;
+        if (slot != 0xFFFFFFFFu) atomicAdd(&lh[slot >> shift], 1);
+    }
+    __syncthreads();
+    for (int64_t k = threadIdx.x; k < nparts; k += blockDim.x)
+        if (lh[k]) atomicAdd(&pcount[k], lh[k]);
+}
+
+__global__ void k_pb_scatter(JoinTable t, ProbeKeys p, int64_t m, int shift,
+                             const uint32_t* __restrict__ slot_of,
+                             int32_t* __restrict__ pcur,
+                             uint32_t* __restrict__ pi, uint32_t* __restrict__ pslot,
+                             int64_t* __restrict__ pkey)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < m; i += stride) {
+        uint32_t slot = slot_of[i];
+        if (slot == 0xFFFFFFFFu) continue;
+        int32_t at = atomicAdd(&pcur[slot >> shift], 1);
+        pi[at] = (uint32_t)i;
+        pslot[at] = slot;
+        pkey[at] = p.pk[i];
+    }
+}
+
+__global__ void k_pb_probe(JoinTable t, int64_t midx,
+                           const uint32_t* __restrict__ pi,
+                           const uint32_t* __restrict__ pslot,
+                           const int64_t* __restrict__ pkey,
+                           unsigned long long* __restrict__ out_cnt,
+                           int64_t cap,
+                           uint32_t* __restrict__ mp, int32_t* __restrict__ mb)
+{
+    /* block-contiguous ranges keep each block inside one slot region.
+     * DESCENDING walk + per-thread ordered appends + the later stable sort
+     * by probe row = the reference's reverse-insertion duplicate order. */
+    int64_t chunk = (midx + gridDim.x - 1) / gridDim.x;
+    int64_t lo = (int64_t)blockIdx.x * chunk, hi = min(lo + chunk, midx);
+    for (int64_t j = lo + threadIdx.x; j < hi; j += blockDim.x) {
+        uint32_t slot = pslot[j];
+        int64_t key = pkey[j];
+        for (int32_t x = t.bucket_off[slot + 1] - 1; x >= t.bucket_off[slot]; x--) {
+            if (t.csr_keys[x] == key) {
+                unsigned long long pos = atomicAdd(out_cnt, 1ull);
+                if ((int64_t)pos < cap) {   /* overflow: count on, write off */
+                    mp[pos] = pi[j];
+                    mb[pos] = t.csr_rows[x];
+                }
+            }
+        }
+    }
+}
+
 __global__ void k_probe_count(JoinTable t, ProbeKeys p, int64_t m,
                               int32_t* __restrict__ counts,
                               int32_t* __restrict__ heads)
@@ -876,27 +949,107 @@ struct LookupJoinOp : tg_operator {
         int32_t* d_heads = nullptr;
         int64_t* d_offsets = nullptr;
         int64_t* d_total = nullptr;
-        TG_POOL_ALLOC(s, &d_counts, (in.n ? in.n : 1) * 4);
-        TG_POOL_ALLOC(s, &d_heads, (in.n ? in.n : 1) * 4);
-        TG_POOL_ALLOC(s, &d_offsets, (in.n ? in.n : 1) * 8);
-        TG_POOL_ALLOC(s, &d_total, 8);
-        hipLaunchKernelGGL(k_probe_count, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
-                           0, s->stream, t, pkc, in.n, d_counts, d_heads);
-        TG_HIP_CHECK(hipGetLastError());
-        tg_status sst = run_scan_counts(s, d_counts, in.n, d_offsets, d_total);
-        if (sst != TG_OK) return sst;
-        int64_t total = 0;
-        TG_HIP_CHECK(hipMemcpyAsync(&total, d_total, 8, hipMemcpyDeviceToHost, s->stream));
-        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
-
         int32_t* d_op = nullptr;
         int32_t* d_ob = nullptr;
-        TG_POOL_ALLOC(s, &d_op, (total ? total : 1) * 4);
-        TG_POOL_ALLOC(s, &d_ob, (total ? total : 1) * 4);
-        hipLaunchKernelGGL(k_probe_fill, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
-                           0, s->stream, t, pkc, in.n, d_heads, d_offsets, d_op, d_ob);
-        TG_HIP_CHECK(hipGetLastError());
-        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        int64_t total = 0;
+        int64_t tbl_bytes = t.capacity * 4 + t.n * 12;
+        static int use_part = [] { const char* e = getenv("TG_JOIN_PART"); return e ? atoi(e) : 1; }();
+        if (use_part && t.csr && !t.generic && t.csr_keys &&
+            in.n >= (1 << 22) && tbl_bytes > (64ll << 20)) {
+            /* partitioned single-pass probe + stable match sort */
+            int64_t nparts = 1;
+            while (tbl_bytes / nparts > (16ll << 20)) nparts <<= 1;
+            if (nparts > t.capacity) nparts = t.capacity;
+            int shift = 0;
+            while (((int64_t)1 << shift) < t.capacity / nparts) shift++;
+            uint32_t* d_slot_of = nullptr;
+            int32_t* d_pc = nullptr;
+            uint32_t* d_pi = nullptr;
+            uint32_t* d_ps = nullptr;
+            int64_t* d_pk = nullptr;
+            unsigned long long* d_cnt = nullptr;
+            TG_POOL_ALLOC(s, &d_slot_of, in.n * 4);
+            TG_POOL_ALLOC(s, &d_pc, (nparts + 1) * 4);
+            TG_POOL_ALLOC(s, &d_pi, in.n * 4);
+            TG_POOL_ALLOC(s, &d_ps, in.n * 4);
+            TG_POOL_ALLOC(s, &d_pk, in.n * 8);
+            TG_POOL_ALLOC(s, &d_cnt, 8);
+            TG_HIP_CHECK(hipMemsetAsync(d_pc, 0, (nparts + 1) * 4, s->stream));
+            TG_HIP_CHECK(hipMemsetAsync(d_cnt, 0, 8, s->stream));
+            hipLaunchKernelGGL(k_pb_slots, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
+                               (size_t)nparts * 4, s->stream, t, pkc, in.n, nparts,
+                               shift, d_slot_of, d_pc);
+            TG_HIP_CHECK(hipGetLastError());
+            std::vector<int32_t> pc(nparts), pb(nparts);
+            TG_HIP_CHECK(hipMemcpyAsync(pc.data(), d_pc, nparts * 4,
+                                        hipMemcpyDeviceToHost, s->stream));
+            TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+            int32_t run = 0;
+            for (int64_t q = 0; q < nparts; q++) { pb[q] = run; run += pc[q]; }
+            int64_t midx = run;               /* non-null probe rows */
+            TG_HIP_CHECK(hipMemcpyAsync(d_pc, pb.data(), nparts * 4,
+                                        hipMemcpyHostToDevice, s->stream));
+            hipLaunchKernelGGL(k_pb_scatter, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
+                               0, s->stream, t, pkc, in.n, shift, d_slot_of, d_pc,
+                               d_pi, d_ps, d_pk);
+            TG_HIP_CHECK(hipGetLastError());
+            /* first-try match cap (exact when build keys are unique, the
+               usual case); k_pb_probe counts past it without writing, so an
+               overflow is detected and retried with the exact size */
+            int64_t cap = midx + t.n + 64;
+            TG_POOL_ALLOC(s, &d_op, cap * 4);
+            TG_POOL_ALLOC(s, &d_ob, cap * 4);
+            if (midx > 0) {
+                hipLaunchKernelGGL(k_pb_probe, dim3(2048), dim3(TG_BLOCK), 0, s->stream,
+                                   t, midx, d_pi, d_ps, d_pk, d_cnt, cap,
+                                   (uint32_t*)d_op, d_ob);
+                TG_HIP_CHECK(hipGetLastError());
+            }
+            unsigned long long cnt = 0;
+            TG_HIP_CHECK(hipMemcpyAsync(&cnt, d_cnt, 8, hipMemcpyDeviceToHost, s->stream));
+            TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+            total = (int64_t)cnt;
+            if (total > cap) {       /* many-to-many overflow: exact retry */
+                tg_pool_free(s, d_op);
+                tg_pool_free(s, d_ob);
+                cap = total;
+                TG_POOL_ALLOC(s, &d_op, cap * 4);
+                TG_POOL_ALLOC(s, &d_ob, cap * 4);
+                TG_HIP_CHECK(hipMemsetAsync(d_cnt, 0, 8, s->stream));
+                hipLaunchKernelGGL(k_pb_probe, dim3(2048), dim3(TG_BLOCK), 0, s->stream,
+                                   t, midx, d_pi, d_ps, d_pk, d_cnt, cap,
+                                   (uint32_t*)d_op, d_ob);
+                TG_HIP_CHECK(hipGetLastError());
+                TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+            }
+            tg_status pst = run_sort_pairs_u32(s, (uint32_t*)d_op, d_ob, total);
+            if (pst != TG_OK) return pst;
+            tg_pool_free(s, d_slot_of);
+            tg_pool_free(s, d_pc);
+            tg_pool_free(s, d_pi);
+            tg_pool_free(s, d_ps);
+            tg_pool_free(s, d_pk);
+            tg_pool_free(s, d_cnt);
+        }
+        else {
+            TG_POOL_ALLOC(s, &d_counts, (in.n ? in.n : 1) * 4);
+            TG_POOL_ALLOC(s, &d_heads, (in.n ? in.n : 1) * 4);
+            TG_POOL_ALLOC(s, &d_offsets, (in.n ? in.n : 1) * 8);
+            TG_POOL_ALLOC(s, &d_total, 8);
+            hipLaunchKernelGGL(k_probe_count, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
+                               0, s->stream, t, pkc, in.n, d_counts, d_heads);
+            TG_HIP_CHECK(hipGetLastError());
+            tg_status sst = run_scan_counts(s, d_counts, in.n, d_offsets, d_total);
+            if (sst != TG_OK) return sst;
+            TG_HIP_CHECK(hipMemcpyAsync(&total, d_total, 8, hipMemcpyDeviceToHost, s->stream));
+            TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+            TG_POOL_ALLOC(s, &d_op, (total ? total : 1) * 4);
+            TG_POOL_ALLOC(s, &d_ob, (total ? total : 1) * 4);
+            hipLaunchKernelGGL(k_probe_fill, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
+                               0, s->stream, t, pkc, in.n, d_heads, d_offsets, d_op, d_ob);
+            TG_HIP_CHECK(hipGetLastError());
+            TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        }
 
         /* output: probe output channels gathered by probe row, then build
          * output channels gathered by build row (LookupJoinPageBuilder) */
@@ -914,10 +1067,10 @@ struct LookupJoinOp : tg_operator {
             if (st != TG_OK) return st;
             outp.blocks.push_back(ob);
         }
-        tg_pool_free(s, d_counts);
-        tg_pool_free(s, d_heads);
-        tg_pool_free(s, d_offsets);
-        tg_pool_free(s, d_total);
+        if (d_counts) tg_pool_free(s, d_counts);
+        if (d_heads) tg_pool_free(s, d_heads);
+        if (d_offsets) tg_pool_free(s, d_offsets);
+        if (d_total) tg_pool_free(s, d_total);
         if (d_pkg) tg_pool_free(s, d_pkg);
         tg_pool_free(s, d_op);
         tg_pool_free(s, d_ob);

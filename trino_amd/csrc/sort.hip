@@ -54,6 +54,35 @@ tg_status run_sort_pairs(tg_session* s, uint64_t* d_keys, int64_t* d_vals, int64
     return TG_OK;
 }
 
+/* stable ascending sort of (u32 key, i32 value) pairs (join match
+ * reordering: sort matches by probe row) */
+tg_status run_sort_pairs_u32(tg_session* s, uint32_t* d_keys, int32_t* d_vals, int64_t n)
+{
+    if (n <= 1) return TG_OK;
+    uint32_t* d_keys2 = nullptr;
+    int32_t* d_vals2 = nullptr;
+    TG_POOL_ALLOC(s, &d_keys2, n * 4);
+    TG_POOL_ALLOC(s, &d_vals2, n * 4);
+    size_t temp_bytes = 0;
+    hipError_t e0 = rocprim::radix_sort_pairs(nullptr, temp_bytes, d_keys, d_keys2,
+                                              d_vals, d_vals2, (size_t)n, 0, 32,
+                                              s->stream);
+    if (e0 != hipSuccess) { TG_SET_ERR("rocprim size query: %s", hipGetErrorName(e0)); return TG_ERR_HIP; }
+    void* d_temp = nullptr;
+    TG_POOL_ALLOC(s, &d_temp, (int64_t)temp_bytes);
+    hipError_t e = rocprim::radix_sort_pairs(d_temp, temp_bytes, d_keys, d_keys2,
+                                             d_vals, d_vals2, (size_t)n, 0, 32,
+                                             s->stream);
+    if (e != hipSuccess) { TG_SET_ERR("rocprim radix_sort_pairs: %s", hipGetErrorName(e)); return TG_ERR_HIP; }
+    TG_HIP_CHECK(hipMemcpyAsync(d_keys, d_keys2, n * 4, hipMemcpyDeviceToDevice, s->stream));
+    TG_HIP_CHECK(hipMemcpyAsync(d_vals, d_vals2, n * 4, hipMemcpyDeviceToDevice, s->stream));
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    tg_pool_free(s, d_keys2);
+    tg_pool_free(s, d_vals2);
+    tg_pool_free(s, d_temp);
+    return TG_OK;
+}
+
 /* argsort of non-negative i64 keys (ascending, stable): writes the
  * permutation as int32 into d_out_idx[n]. d_keys is clobbered. */
 tg_status run_argsort_i64(tg_session* s, int64_t* d_keys, int64_t n, int32_t* d_out_idx)
