@@ -883,3 +883,48 @@ class TestExactFixedPointSum:
         exp = oracle.grouped_sum_f64_exact(gids, vals, ng, scale_pow=43)
         assert np.array_equal(out[0]["values"], by_gid)
         assert np.array_equal(out[1]["values"], exp)
+
+
+class TestPartitionedProbe:
+    """Partitioned single-pass probe + stable match sort (large tables):
+    thresholds forced down via env so the path runs at test sizes. Match
+    LIST order is compared (probe-row order, duplicates reverse-insertion) —
+    exactly the classic two-pass path's contract."""
+
+    def test_matches_and_order_vs_classic(self, sess, ops):
+        import os
+        r = rng(90)
+        nb, m = 50_000, 200_000
+        # ~25% duplicate build keys (1-4 copies)
+        base = r.choice(np.arange(10**7), 30_000, replace=False)
+        bk = np.concatenate([base, r.choice(base, nb - len(base))]).astype(np.int64)
+        r.shuffle(bk)
+        bv = np.arange(nb, dtype=np.int64)
+        pk = r.choice(np.concatenate([base, r.integers(10**8, 10**9, 50_000)]),
+                      m).astype(np.int64)
+
+        def run_join():
+            bridge = ops.JoinBridge(sess)
+            b = ops.hash_builder(sess, bridge, [ops.TG_BIGINT, ops.TG_BIGINT], [0], [1])
+            b.add_input(ops.page_from_numpy([bk, bv]))
+            b.drain()
+            j = ops.lookup_join(sess, bridge, [ops.TG_BIGINT], [0], [0])
+            j.add_input(ops.page_from_numpy([pk]))
+            jp, _ = j.get_output()
+            b.close()
+            j.close()
+            bridge.close()
+            return (jp[0]["values"].copy(), jp[1]["values"].copy())
+
+        os.environ["TG_JOIN_PART_MIN_ROWS"] = "1000"
+        os.environ["TG_JOIN_PART_MIN_BYTES"] = "1000"
+        try:
+            got_p = run_join()
+            os.environ["TG_JOIN_PART"] = "0"
+            got_c = run_join()
+        finally:
+            os.environ.pop("TG_JOIN_PART", None)
+            os.environ.pop("TG_JOIN_PART_MIN_ROWS", None)
+            os.environ.pop("TG_JOIN_PART_MIN_BYTES", None)
+        assert np.array_equal(got_p[0], got_c[0])   # identical list ORDER
+        assert np.array_equal(got_p[1], got_c[1])

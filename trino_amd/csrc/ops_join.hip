@@ -953,9 +953,14 @@ struct LookupJoinOp : tg_operator {
         int32_t* d_ob = nullptr;
         int64_t total = 0;
         int64_t tbl_bytes = t.capacity * 4 + t.n * 12;
-        static int use_part = [] { const char* e = getenv("TG_JOIN_PART"); return e ? atoi(e) : 1; }();
+        const char* ep = getenv("TG_JOIN_PART");
+        int use_part = ep ? atoi(ep) : 1;
+        const char* emr = getenv("TG_JOIN_PART_MIN_ROWS");
+        int64_t min_rows = emr ? atoll(emr) : (1 << 22);
+        const char* emb = getenv("TG_JOIN_PART_MIN_BYTES");
+        int64_t min_bytes = emb ? atoll(emb) : (64ll << 20);
         if (use_part && t.csr && !t.generic && t.csr_keys &&
-            in.n >= (1 << 22) && tbl_bytes > (64ll << 20)) {
+            in.n >= min_rows && tbl_bytes > min_bytes) {
             /* partitioned single-pass probe + stable match sort */
             int64_t nparts = 1;
             while (tbl_bytes / nparts > (16ll << 20)) nparts <<= 1;
