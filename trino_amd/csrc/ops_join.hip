@@ -417,7 +417,7 @@ __global__ void k_pb_scatter(JoinTable t, ProbeKeys p, int64_t m, int shift,
     }
 }
 
-__global__ void k_pb_probe(JoinTable t, int64_t midx,
+__global__ void k_pb_probe(JoinTable t, int64_t jlo, int64_t jhi,
                            const uint32_t* __restrict__ pi,
                            const uint32_t* __restrict__ pslot,
                            const int64_t* __restrict__ pkey,
@@ -425,11 +425,14 @@ __global__ void k_pb_probe(JoinTable t, int64_t midx,
                            int64_t cap,
                            uint32_t* __restrict__ mp, int32_t* __restrict__ mb)
 {
-    /* block-contiguous ranges keep each block inside one slot region.
+    /* ONE LAUNCH PER PARTITION (host loop, stream-ordered): while this
+     * launch runs, every bucket_off/csr_keys/csr_rows read lands in the
+     * partition's ~16 MB table slice, which stays cache-resident.
      * DESCENDING walk + per-thread ordered appends + the later stable sort
      * by probe row = the reference's reverse-insertion duplicate order. */
+    int64_t midx = jhi - jlo;
     int64_t chunk = (midx + gridDim.x - 1) / gridDim.x;
-    int64_t lo = (int64_t)blockIdx.x * chunk, hi = min(lo + chunk, midx);
+    int64_t lo = jlo + (int64_t)blockIdx.x * chunk, hi = min(lo + chunk, jhi);
     int lane = threadIdx.x % 64;
     /* wave-aggregated reservation: one atomicAdd per wave iteration instead
      * of one per match (a single global counter otherwise serializes) */
@@ -1046,9 +1049,11 @@ struct LookupJoinOp : tg_operator {
             int64_t cap = midx + t.n + 64;
             TG_POOL_ALLOC(s, &d_op, cap * 4);
             TG_POOL_ALLOC(s, &d_ob, cap * 4);
-            if (midx > 0) {
-                hipLaunchKernelGGL(k_pb_probe, dim3(2048), dim3(TG_BLOCK), 0, s->stream,
-                                   t, midx, d_pi, d_ps, d_pk, d_cnt, cap,
+            for (int64_t q = 0; q < nparts; q++) {
+                if (!pc[q]) continue;
+                hipLaunchKernelGGL(k_pb_probe, dim3(1024), dim3(TG_BLOCK), 0, s->stream,
+                                   t, (int64_t)pb[q], (int64_t)pb[q] + pc[q],
+                                   d_pi, d_ps, d_pk, d_cnt, cap,
                                    (uint32_t*)d_op, d_ob);
                 TG_HIP_CHECK(hipGetLastError());
             }
@@ -1063,10 +1068,14 @@ struct LookupJoinOp : tg_operator {
                 TG_POOL_ALLOC(s, &d_op, cap * 4);
                 TG_POOL_ALLOC(s, &d_ob, cap * 4);
                 TG_HIP_CHECK(hipMemsetAsync(d_cnt, 0, 8, s->stream));
-                hipLaunchKernelGGL(k_pb_probe, dim3(2048), dim3(TG_BLOCK), 0, s->stream,
-                                   t, midx, d_pi, d_ps, d_pk, d_cnt, cap,
-                                   (uint32_t*)d_op, d_ob);
-                TG_HIP_CHECK(hipGetLastError());
+                for (int64_t q = 0; q < nparts; q++) {
+                    if (!pc[q]) continue;
+                    hipLaunchKernelGGL(k_pb_probe, dim3(1024), dim3(TG_BLOCK), 0, s->stream,
+                                       t, (int64_t)pb[q], (int64_t)pb[q] + pc[q],
+                                       d_pi, d_ps, d_pk, d_cnt, cap,
+                                       (uint32_t*)d_op, d_ob);
+                    TG_HIP_CHECK(hipGetLastError());
+                }
                 TG_HIP_CHECK(hipStreamSynchronize(s->stream));
             }
             tg_status pst = run_sort_pairs_u32(s, (uint32_t*)d_op, d_ob, total);
