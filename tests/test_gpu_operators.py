@@ -916,6 +916,7 @@ class TestPartitionedProbe:
             bridge.close()
             return (jp[0]["values"].copy(), jp[1]["values"].copy())
 
+        os.environ["TG_JOIN_PART"] = "1"
         os.environ["TG_JOIN_PART_MIN_ROWS"] = "1000"
         os.environ["TG_JOIN_PART_MIN_BYTES"] = "1000"
         try:
