@@ -61,10 +61,14 @@ __global__ void k_gt_init(int32_t* state, int64_t cap, int64_t* first_row, int64
     for (int64_t k = i; k < ngroups_cap; k += stride) first_row[k] = INT64_MAX;
 }
 
-/* assign group ids for one page; row_base = global row index of row 0 */
-__global__ void k_gt_assign(GTable t, const KColH* cols, int n_ch,
+/* assign group ids for one page; row_base = global row index of row 0.
+ * NCH > 0 = compile-time channel count (registers + unrolled compares);
+ * NCH == 0 = dynamic fallback for >6 channels. */
+template <int NCH>
+__global__ void k_gt_assign(GTable t, const KColH* cols, int n_ch_dyn,
                             int64_t n, int64_t row_base, int32_t* __restrict__ gids)
 {
+    const int n_ch = NCH > 0 ? NCH : n_ch_dyn;
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < n; i += stride) {
@@ -536,9 +540,15 @@ struct HashAggOp : tg_operator {
         st = make_kcols(s, in, group_channels.data(), (int)group_channels.size(), &d_keys);
         if (st != TG_OK) { tg_free_page(s, &in); return st; }
         TG_POOL_ALLOC(s, &d_gids, in.n * 4);
-        hipLaunchKernelGGL(k_gt_assign, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK), 0, s->stream,
-                           t, d_keys, (int)group_channels.size(), in.n, rows_seen, d_gids);
-        TG_HIP_CHECK(hipGetLastError());
+        {
+            int nch = (int)group_channels.size();
+            auto kfn = nch == 1 ? k_gt_assign<1> : nch == 2 ? k_gt_assign<2> :
+                       nch == 3 ? k_gt_assign<3> : nch == 4 ? k_gt_assign<4> :
+                       k_gt_assign<0>;
+            hipLaunchKernelGGL(kfn, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK), 0, s->stream,
+                               t, d_keys, nch, in.n, rows_seen, d_gids);
+            TG_HIP_CHECK(hipGetLastError());
+        }
 have_gids:;
 
         /* aggregation inputs: all channels */
