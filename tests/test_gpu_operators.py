@@ -929,3 +929,53 @@ class TestPartitionedProbe:
             os.environ.pop("TG_JOIN_PART_MIN_BYTES", None)
         assert np.array_equal(got_p[0], got_c[0])   # identical list ORDER
         assert np.array_equal(got_p[1], got_c[1])
+
+
+class TestSemiJoinNullSemantics:
+    """Three-valued IN semantics (HashSemiJoinOperator.java:180-201):
+    probe NULL -> false when the build set is empty, else NULL;
+    a miss against a set containing NULL -> NULL."""
+
+    def test_null_in_build_and_probe(self, sess, ops):
+        bk = np.array([10, 20, 30, 0], np.int64)
+        bvalid = np.array([0b0111], np.uint64)       # build row 3 is NULL
+        bridge = ops.JoinBridge(sess)
+        b = ops.hash_builder(sess, bridge, [ops.TG_BIGINT], [0], [])
+        b.add_input(ops.page_from_numpy([bk], valids=[bvalid]))
+        b.drain()
+        sj = ops.semi_join(sess, bridge, 0)
+        pk = np.array([10, 99, 0, 20], np.int64)
+        pvalid = np.array([0b1011], np.uint64)       # probe row 2 is NULL
+        sj.add_input(ops.page_from_numpy([pk], valids=[pvalid]))
+        out, _ = sj.get_output()
+        b.close()
+        sj.close()
+        bridge.close()
+        m = out[-1]
+        vals = m["values"].astype(np.int8)
+        valid = np.asarray(m["valid"])
+        def bit(i):
+            return (int(valid[i >> 6]) >> (i & 63)) & 1
+        assert vals[0] == 1 and bit(0) == 1          # 10 IN set -> true
+        assert bit(1) == 0                           # 99 miss, set has NULL -> NULL
+        assert bit(2) == 0                           # NULL probe, set nonempty -> NULL
+        assert vals[3] == 1 and bit(3) == 1          # 20 -> true
+
+    def test_null_probe_empty_build(self, sess, ops):
+        bridge = ops.JoinBridge(sess)
+        b = ops.hash_builder(sess, bridge, [ops.TG_BIGINT], [0], [])
+        b.add_input(ops.page_from_numpy([np.empty(0, np.int64)]))
+        b.drain()
+        sj = ops.semi_join(sess, bridge, 0)
+        pk = np.array([5, 0], np.int64)
+        pvalid = np.array([0b01], np.uint64)
+        sj.add_input(ops.page_from_numpy([pk], valids=[pvalid]))
+        out, _ = sj.get_output()
+        b.close()
+        sj.close()
+        bridge.close()
+        m = out[-1]
+        vals = m["values"].astype(np.int8)
+        valid = np.asarray(m["valid"])
+        assert vals[0] == 0 and ((int(valid[0]) >> 0) & 1) == 1   # miss, no null -> false
+        assert vals[1] == 0 and ((int(valid[0]) >> 1) & 1) == 1   # NULL probe, empty set -> false

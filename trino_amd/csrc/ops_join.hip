@@ -57,6 +57,7 @@ struct JoinTable {
     int32_t n_key_ch = 1;
     const KColH* bkeys = nullptr;   /* device array[n_key_ch] */
     const SlotKV* kv = nullptr;     /* single-key fast table or null */
+    int32_t has_null_key = 0;       /* any build key row null (semi-join 3VL) */
 };
 
 /* slot of a build row; generic keys hash with the canonical row hash
@@ -667,6 +668,21 @@ static tg_status run_scan_counts(tg_session* s, const int32_t* d_counts, int64_t
     return TG_OK;
 }
 
+__global__ void k_bitmap_concat(const uint64_t* __restrict__ src, int64_t n,
+                                int64_t at, uint64_t* __restrict__ dst)
+{
+    /* dst preset to all-valid; clear dst bit (at+i) where src row i is null.
+     * Works at any bit offset (the memcpy path needs 64-row alignment). */
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        if (!((src[i >> 6] >> (i & 63)) & 1)) {
+            int64_t r = at + i;
+            atomicAnd((unsigned long long*)&dst[r >> 6], ~(1ull << (r & 63)));
+        }
+    }
+}
+
 __global__ void k_off_rebase(const int32_t* __restrict__ src, int64_t n,
                              int32_t base, int32_t* __restrict__ dst)
 {
@@ -725,10 +741,10 @@ struct HashBuilderOp : tg_operator {
                 TG_HIP_CHECK(hipGetLastError());
             }
             if (pb.valid) {
-                if (at % 64 != 0) { TG_SET_ERR("null bitmap concat requires 64-row-aligned pages"); return TG_ERR_UNSUPPORTED; }
-                TG_HIP_CHECK(hipMemcpyAsync((char*)b->valid + at / 8, pb.valid,
-                                            (pb.n + 63) / 64 * 8,
-                                            hipMemcpyDeviceToDevice, s->stream));
+                hipLaunchKernelGGL(k_bitmap_concat, dim3(tg_grid_for(pb.n)),
+                                   dim3(TG_BLOCK), 0, s->stream, pb.valid, pb.n,
+                                   at, b->valid);
+                TG_HIP_CHECK(hipGetLastError());
             }
             at += pb.n;
             byte_at += page_bytes[pi];
@@ -802,10 +818,10 @@ struct HashBuilderOp : tg_operator {
                  * kernel would go here — round 1 requires page sizes to be
                  * multiples of 64 when nulls are present */
                 if (p.blocks[c].valid) {
-                    if (at % 64 != 0) { TG_SET_ERR("null bitmap concat requires 64-row-aligned pages"); return TG_ERR_UNSUPPORTED; }
-                    TG_HIP_CHECK(hipMemcpyAsync((char*)b.valid + at / 8, p.blocks[c].valid,
-                                                (p.n + 63) / 64 * 8,
-                                                hipMemcpyDeviceToDevice, s->stream));
+                    hipLaunchKernelGGL(k_bitmap_concat, dim3(tg_grid_for(p.n)),
+                                       dim3(TG_BLOCK), 0, s->stream, p.blocks[c].valid,
+                                       p.n, at, b.valid);
+                    TG_HIP_CHECK(hipGetLastError());
                 }
                 at += p.n;
             }
@@ -871,6 +887,7 @@ struct HashBuilderOp : tg_operator {
             int32_t run = 0;
             for (int64_t r = 0; r < nreg; r++) { rb[r] = run; run += rt[r]; }
             int32_t total_indexed = run;    /* == total_rows minus null keys */
+            t.has_null_key = (int64_t)total_indexed < total_rows ? 1 : 0;
             TG_HIP_CHECK(hipMemcpyAsync(d_rbase, rb.data(), nreg * 4,
                                         hipMemcpyHostToDevice, s->stream));
             hipLaunchKernelGGL(k_jc_addbase, dim3(tg_grid_for(t.capacity)), dim3(TG_BLOCK),
@@ -1210,9 +1227,13 @@ __global__ void k_semi_probe(JoinTable t, ProbeKeys p, int64_t m,
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < m; i += stride) {
+        /* three-valued IN semantics (HashSemiJoinOperator.java:180-201):
+         * probe null -> false if the build set is empty, else NULL;
+         * miss against a set containing NULL -> NULL. */
         if (probe_null(t, p, i)) {
             match[i] = 0;
-            atomicAnd((unsigned long long*)&mvalid[i >> 6], ~(1ull << (i & 63)));
+            if (t.n > 0)
+                atomicAnd((unsigned long long*)&mvalid[i >> 6], ~(1ull << (i & 63)));
             continue;
         }
         int64_t slot = (int64_t)probe_slot(t, p, i);
@@ -1243,6 +1264,8 @@ __global__ void k_semi_probe(JoinTable t, ProbeKeys p, int64_t m,
             }
         }
         match[i] = hit;
+        if (!hit && t.has_null_key)
+            atomicAnd((unsigned long long*)&mvalid[i >> 6], ~(1ull << (i & 63)));
     }
 }
 
