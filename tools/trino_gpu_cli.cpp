@@ -261,10 +261,209 @@ static int run_q3(tg_session* s, double sf)
     return 0;
 }
 
+
+static int run_q4(tg_session* s, double sf)
+{
+    /* Q4: EXISTS semi join + grouped count (python mirror:
+     * trino_amd/tpch_queries.py q4_gpu) */
+    int64_t n_orders = (int64_t)(1500000 * sf);
+    void *o_ok, *o_od, *o_pri;
+    die(tg_device_malloc(s, &o_ok, n_orders * 8), "malloc");
+    die(tg_device_malloc(s, &o_od, n_orders * 4), "malloc");
+    die(tg_device_malloc(s, &o_pri, n_orders), "malloc");
+    die(tg_tpch_gen_orders(s, sf, 1, n_orders, (int64_t*)o_ok, nullptr,
+                           (int32_t*)o_od, (uint8_t*)o_pri), "gen ord");
+    tg_tpch_lineitem_cols li;
+    die(tg_tpch_lineitem_alloc(s, sf, 1, n_orders, 1 | 2, &li), "gen li");
+    int fin = 0;
+
+    std::vector<tg_block> lb = {dev_block(TG_BIGINT, li.orderkey, li.row_count),
+                                dev_block(TG_INTEGER, li.commitdate, li.row_count),
+                                dev_block(TG_INTEGER, li.receiptdate, li.row_count)};
+    tg_page lpage = dev_page(lb, li.row_count);
+    std::vector<tg_expr_inst> lf = {I_col(1), I_col(2), I_op(TG_EXPR_LT)};
+    std::vector<tg_expr_inst> lp = {I_col(0)};
+    tg_expr lfe{lf.data(), (int32_t)lf.size()};
+    tg_expr lpe{lp.data(), 1};
+    int32_t lot = TG_BIGINT;
+    tg_operator* f1 = nullptr;
+    die(tg_filter_project_create(s, &lfe, &lpe, &lot, 1, &f1), "f1");
+    die(tg_operator_add_input(f1, &lpage), "f1 add");
+    die(tg_operator_finish(f1), "f1 fin");
+    tg_page late{};
+    die(tg_operator_get_output(f1, &late, &fin), "f1 out");
+
+    tg_join_bridge* br = nullptr;
+    die(tg_join_bridge_create(s, &br), "br");
+    int32_t bt = TG_BIGINT, kc0 = 0;
+    tg_operator* b = nullptr;
+    die(tg_hash_builder_create(s, br, &bt, 1, &kc0, 1, nullptr, 0, &b), "b");
+    die(tg_operator_add_input(b, &late), "b add");
+    die(tg_operator_finish(b), "b fin");
+
+    std::vector<tg_block> ob = {dev_block(TG_BIGINT, o_ok, n_orders),
+                                dev_block(TG_INTEGER, o_od, n_orders),
+                                dev_block(TG_TINYINT, o_pri, n_orders)};
+    tg_page opage = dev_page(ob, n_orders);
+    std::vector<tg_expr_inst> of = {I_col(1), I_i64(8582), I_op(TG_EXPR_GE),
+                                    I_col(1), I_i64(8674), I_op(TG_EXPR_LT),
+                                    I_op(TG_EXPR_AND)};
+    std::vector<tg_expr_inst> op0 = {I_col(0)}, op1 = {I_col(2)};
+    tg_expr ofe{of.data(), (int32_t)of.size()};
+    tg_expr opr[2] = {{op0.data(), 1}, {op1.data(), 1}};
+    int32_t oot[2] = {TG_BIGINT, TG_TINYINT};
+    tg_operator* f2 = nullptr;
+    die(tg_filter_project_create(s, &ofe, opr, oot, 2, &f2), "f2");
+    die(tg_operator_add_input(f2, &opage), "f2 add");
+    die(tg_operator_finish(f2), "f2 fin");
+    tg_page owin{};
+    die(tg_operator_get_output(f2, &owin, &fin), "f2 out");
+
+    tg_operator* sj = nullptr;
+    die(tg_semi_join_create(s, br, 0, &sj), "sj");
+    die(tg_operator_add_input(sj, &owin), "sj add");
+    die(tg_operator_finish(sj), "sj fin");
+    tg_page marked{};
+    die(tg_operator_get_output(sj, &marked, &fin), "sj out");
+
+    std::vector<tg_expr_inst> mf = {I_col(2), I_i64(1), I_op(TG_EXPR_EQ)};
+    std::vector<tg_expr_inst> mp = {I_col(1)};
+    tg_expr mfe{mf.data(), (int32_t)mf.size()};
+    tg_expr mpe{mp.data(), 1};
+    int32_t mot = TG_TINYINT;
+    tg_operator* f3 = nullptr;
+    die(tg_filter_project_create(s, &mfe, &mpe, &mot, 1, &f3), "f3");
+    die(tg_operator_add_input(f3, &marked), "f3 add");
+    die(tg_operator_finish(f3), "f3 fin");
+    tg_page exists{};
+    die(tg_operator_get_output(f3, &exists, &fin), "f3 out");
+
+    int32_t gch = 0, gty = TG_TINYINT;
+    tg_agg_spec ag{TG_AGG_COUNT_STAR, -1};
+    tg_operator* agg = nullptr;
+    die(tg_hash_aggregation_create(s, &gch, 1, &gty, &ag, 1, TG_STEP_SINGLE, &agg), "agg");
+    die(tg_operator_add_input(agg, &exists), "agg add");
+    die(tg_operator_finish(agg), "agg fin");
+    tg_page out{};
+    die(tg_operator_get_output(agg, &out, &fin), "agg out");
+
+    static const char* PRI[5] = {"1-URGENT", "2-HIGH", "3-MEDIUM",
+                                 "4-NOT SPECIFIED", "5-LOW"};
+    struct Row { int8_t p; int64_t c; };
+    std::vector<Row> rows;
+    for (int64_t i = 0; i < out.position_count; i++) {
+        Row r;
+        die(tg_copy_dtoh(s, &r.p, (const char*)out.blocks[0].data + i, 1), "dtoh");
+        die(tg_copy_dtoh(s, &r.c, (const char*)out.blocks[1].data + i * 8, 8), "dtoh");
+        rows.push_back(r);
+    }
+    printf("o_orderpriority|order_count\n");
+    for (int p = 0; p < 5; p++)
+        for (auto& r : rows)
+            if (r.p == p) printf("%s|%lld\n", PRI[p], (long long)r.c);
+    for (tg_operator* o : {f1, b, f2, sj, f3, agg}) tg_operator_close(o);
+    tg_join_bridge_close(br);
+    die(tg_tpch_lineitem_free(s, &li), "free");
+    for (void* pp : {o_ok, o_od, o_pri}) die(tg_device_free(s, pp), "free");
+    return 0;
+}
+
+static int run_q14(tg_session* s, double sf)
+{
+    /* Q14: part join + conditional aggregation (python mirror: q14_gpu) */
+    int64_t n_orders = (int64_t)(1500000 * sf), n_parts = (int64_t)(200000 * sf);
+    void *p_pk, *p_ty;
+    die(tg_device_malloc(s, &p_pk, n_parts * 8), "malloc");
+    die(tg_device_malloc(s, &p_ty, n_parts * 2), "malloc");
+    die(tg_tpch_gen_part(s, sf, 1, n_parts, (int64_t*)p_pk, (int16_t*)p_ty), "gen part");
+    tg_tpch_lineitem_cols li;
+    die(tg_tpch_lineitem_alloc(s, sf, 1, n_orders, 4, &li), "gen li");
+    int fin = 0;
+
+    std::vector<tg_block> pb = {dev_block(TG_BIGINT, p_pk, n_parts),
+                                dev_block(TG_SMALLINT, p_ty, n_parts)};
+    tg_page ppage = dev_page(pb, n_parts);
+    tg_join_bridge* br = nullptr;
+    die(tg_join_bridge_create(s, &br), "br");
+    int32_t bt[2] = {TG_BIGINT, TG_SMALLINT}, kc0 = 0, bout = 1;
+    tg_operator* b = nullptr;
+    die(tg_hash_builder_create(s, br, bt, 2, &kc0, 1, &bout, 1, &b), "b");
+    die(tg_operator_add_input(b, &ppage), "b add");
+    die(tg_operator_finish(b), "b fin");
+
+    std::vector<tg_block> lb = {dev_block(TG_BIGINT, li.partkey, li.row_count),
+                                dev_block(TG_INTEGER, li.shipdate, li.row_count),
+                                dev_block(TG_DOUBLE, li.extendedprice, li.row_count),
+                                dev_block(TG_DOUBLE, li.discount, li.row_count)};
+    tg_page lpage = dev_page(lb, li.row_count);
+    std::vector<tg_expr_inst> lf = {I_col(1), I_i64(9374), I_op(TG_EXPR_GE),
+                                    I_col(1), I_i64(9404), I_op(TG_EXPR_LT),
+                                    I_op(TG_EXPR_AND)};
+    std::vector<tg_expr_inst> lp0 = {I_col(0)};
+    std::vector<tg_expr_inst> lp1 = {I_col(2), I_f64(1.0), I_col(3),
+                                     I_op(TG_EXPR_SUB), I_op(TG_EXPR_MUL)};
+    tg_expr lfe{lf.data(), (int32_t)lf.size()};
+    tg_expr lpr[2] = {{lp0.data(), 1}, {lp1.data(), (int32_t)lp1.size()}};
+    int32_t lot[2] = {TG_BIGINT, TG_DOUBLE};
+    tg_operator* f = nullptr;
+    die(tg_filter_project_create(s, &lfe, lpr, lot, 2, &f), "f");
+    die(tg_operator_add_input(f, &lpage), "f add");
+    die(tg_operator_finish(f), "f fin");
+    tg_page sel{};
+    die(tg_operator_get_output(f, &sel, &fin), "f out");
+
+    int32_t ptypes[2] = {TG_BIGINT, TG_DOUBLE}, jout = 1;
+    tg_operator* j = nullptr;
+    die(tg_lookup_join_create(s, br, ptypes, 2, &kc0, 1, &jout, 1, &j), "j");
+    die(tg_operator_add_input(j, &sel), "j add");
+    die(tg_operator_finish(j), "j fin");
+    tg_page joined{};
+    die(tg_operator_get_output(j, &joined, &fin), "j out");
+
+    tg_agg_spec a1{TG_AGG_SUM_F64_EXACT, 0, 43};
+    tg_operator* agg1 = nullptr;
+    die(tg_hash_aggregation_create(s, nullptr, 0, nullptr, &a1, 1, TG_STEP_SINGLE, &agg1), "a1");
+    die(tg_operator_add_input(agg1, &joined), "a1 add");
+    die(tg_operator_finish(agg1), "a1 fin");
+    tg_page tot{};
+    die(tg_operator_get_output(agg1, &tot, &fin), "a1 out");
+    double total = 0;
+    die(tg_copy_dtoh(s, &total, tot.blocks[0].data, 8), "dtoh");
+
+    std::vector<tg_expr_inst> pf = {I_col(1), I_i64(125), I_op(TG_EXPR_GE)};
+    std::vector<tg_expr_inst> pp0 = {I_col(0)};
+    tg_expr pfe{pf.data(), (int32_t)pf.size()};
+    tg_expr ppe{pp0.data(), 1};
+    int32_t pot = TG_DOUBLE;
+    tg_operator* f2 = nullptr;
+    die(tg_filter_project_create(s, &pfe, &ppe, &pot, 1, &f2), "f2");
+    die(tg_operator_add_input(f2, &joined), "f2 add");
+    die(tg_operator_finish(f2), "f2 fin");
+    tg_page promo_p{};
+    die(tg_operator_get_output(f2, &promo_p, &fin), "f2 out");
+    tg_agg_spec a2{TG_AGG_SUM_F64_EXACT, 0, 43};
+    tg_operator* agg2 = nullptr;
+    die(tg_hash_aggregation_create(s, nullptr, 0, nullptr, &a2, 1, TG_STEP_SINGLE, &agg2), "a2");
+    die(tg_operator_add_input(agg2, &promo_p), "a2 add");
+    die(tg_operator_finish(agg2), "a2 fin");
+    tg_page pr{};
+    die(tg_operator_get_output(agg2, &pr, &fin), "a2 out");
+    double promo = 0;
+    die(tg_copy_dtoh(s, &promo, pr.blocks[0].data, 8), "dtoh");
+
+    printf("promo_revenue\n%.13g\n", 100.0 * promo / total);
+    for (tg_operator* o : {b, f, j, agg1, f2, agg2}) tg_operator_close(o);
+    tg_join_bridge_close(br);
+    die(tg_tpch_lineitem_free(s, &li), "free");
+    die(tg_device_free(s, p_pk), "free");
+    die(tg_device_free(s, p_ty), "free");
+    return 0;
+}
+
 int main(int argc, char** argv)
 {
     if (argc < 2 || !strcmp(argv[1], "--help")) {
-        printf("usage: %s q1|q3|q6 [scale_factor]  (version: %s)\n",
+        printf("usage: %s q1|q3|q4|q6|q14 [scale_factor]  (version: %s)\n",
                argv[0], tg_version());
         return argc < 2 ? 1 : 0;
     }
@@ -279,6 +478,8 @@ int main(int argc, char** argv)
     if (!strcmp(argv[1], "q1")) rc = run_q1(s, sf);
     else if (!strcmp(argv[1], "q3")) rc = run_q3(s, sf);
     else if (!strcmp(argv[1], "q6")) rc = run_q6(s, sf);
+    else if (!strcmp(argv[1], "q4")) rc = run_q4(s, sf);
+    else if (!strcmp(argv[1], "q14")) rc = run_q14(s, sf);
     else fprintf(stderr, "unknown query %s\n", argv[1]);
     tg_session_close(s);
     return rc;
