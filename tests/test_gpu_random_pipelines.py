@@ -134,3 +134,88 @@ def test_random_partition_roundtrip(sess, ops, seed):
         seen += len(rows)
     assert seen == n
     op.close()
+
+
+@pytest.mark.parametrize("seed", range(4))
+def test_random_semi_join_with_nulls(sess, ops, seed):
+    """Semi join three-valued logic on random data with null keys on both
+    sides (HashSemiJoinOperator.java:180-201)."""
+    r = np.random.default_rng(300 + seed)
+    nb = int(r.integers(1, 5000))
+    m = int(r.integers(1, 40_000))
+    bk = r.integers(0, 3000, nb).astype(np.int64)
+    b_nulls = r.random(nb) < 0.05
+    bvalid = np.full((nb + 63) // 64, ~np.uint64(0), np.uint64)
+    for i in np.nonzero(b_nulls)[0]:
+        bvalid[i >> 6] &= ~np.uint64(1 << (i & 63))
+    pk = r.integers(0, 6000, m).astype(np.int64)
+    p_nulls = r.random(m) < 0.05
+    pvalid = np.full((m + 63) // 64, ~np.uint64(0), np.uint64)
+    for i in np.nonzero(p_nulls)[0]:
+        pvalid[i >> 6] &= ~np.uint64(1 << (i & 63))
+
+    bridge = ops.JoinBridge(sess)
+    b = ops.hash_builder(sess, bridge, [ops.TG_BIGINT], [0], [])
+    b.add_input(ops.page_from_numpy([bk], valids=[bvalid]))
+    b.drain()
+    sj = ops.semi_join(sess, bridge, 0)
+    sj.add_input(ops.page_from_numpy([pk], valids=[pvalid]))
+    out, _ = sj.get_output()
+    b.close()
+    sj.close()
+    bridge.close()
+
+    bset = set(bk[~b_nulls].tolist())
+    build_has_null = bool(b_nulls.any())
+    vals = out[-1]["values"].astype(np.int8)
+    valid = np.asarray(out[-1]["valid"])
+    for i in range(m):
+        bit = (int(valid[i >> 6]) >> (i & 63)) & 1
+        if p_nulls[i]:
+            exp_null = nb > 0            # set nonempty -> NULL
+            assert bit == (0 if exp_null else 1)
+        elif int(pk[i]) in bset:
+            assert vals[i] == 1 and bit == 1
+        elif build_has_null:
+            assert bit == 0              # miss vs set containing NULL -> NULL
+        else:
+            assert vals[i] == 0 and bit == 1
+
+
+@pytest.mark.parametrize("seed", range(4))
+def test_random_exact_sum_pipeline(sess, ops, seed):
+    """filter -> group-by with TG_AGG_SUM_F64_EXACT: bit-equal to the
+    oracle's exact leg on money-grid values, any page split."""
+    r = np.random.default_rng(400 + seed)
+    n = int(r.integers(1000, 150_000))
+    keys = r.integers(0, 500, n).astype(np.int64)
+    vals = r.integers(90100, 209900, n) / 100.0      # >= 2^9: 2^-43 grid
+    thresh = int(r.integers(0, 500))
+    split = int(r.integers(1, n))
+    op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                              [(ops.AGG_SUM_F64_EXACT, 1, 43)])
+    f = ops.expr(("col", 0), ("i64", thresh), "lt")
+    for lo, hi in ((0, split), (split, n)):
+        if hi == lo:
+            continue
+        fp = ops.filter_project(sess, f,
+                                [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                                [ops.TG_BIGINT, ops.TG_DOUBLE])
+        fp.add_input(ops.page_from_numpy([keys[lo:hi], vals[lo:hi]]))
+        fp.finish()
+        sel, _ = fp.get_output()
+        fp.close()
+        if sel is not None:
+            op.add_input(ops.page_from_numpy([sel[0]["values"], sel[1]["values"]]))
+    out = op.drain()
+    op.close()
+    mask = keys < thresh
+    ek, ev = keys[mask], vals[mask]
+    if len(ek) == 0:
+        assert not out or len(out[0][0]["values"]) == 0
+        return
+    gids, ng, by_gid, _ = oracle.bigint_groupby(ek)
+    exp = oracle.grouped_sum_f64_exact(gids, ev, ng, scale_pow=43)
+    o = out[0]
+    assert np.array_equal(o[0]["values"], by_gid)
+    assert np.array_equal(o[1]["values"], exp)
