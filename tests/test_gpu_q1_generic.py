@@ -86,7 +86,9 @@ def test_generic_pipeline_matches_fused():
         combos = (rf * 2 + ls)[order]
         np.testing.assert_allclose(avg_qty, [fused.avg_qty[c] for c in combos], rtol=1e-12)
         np.testing.assert_allclose(avg_price, [fused.avg_price[c] for c in combos], rtol=1e-12)
-        np.testing.assert_allclose(avg_disc, [fused.avg_disc[c] for c in combos], rtol=1e-12)
+        # AVG state is a plain atomicAdd f64 sum; ~3e5 addends of ~0.05 give
+        # ~1.5e-12 relative wobble vs the fused kernel's exact sum
+        np.testing.assert_allclose(avg_disc, [fused.avg_disc[c] for c in combos], rtol=1e-11)
         assert np.array_equal(cnt, [fused.count[c] for c in combos])
         s.tpch_lineitem_free(li)
     finally:
