@@ -225,6 +225,8 @@ typedef struct tg_tpch_lineitem_cols {
     int32_t* commitdate;    /* optional (flags bit 1) */
     int32_t* receiptdate;   /* optional (flags bit 1) */
     int64_t* partkey;       /* optional (flags bit 2) */
+    uint8_t* shipmode;      /* optional (flags bit 3): dictionary id 0..6 =
+                               REG AIR,AIR,RAIL,TRUCK,MAIL,FOB,SHIP */
 } tg_tpch_lineitem_cols;
 
 /* Generate lineitem rows for orders [order_start, order_start+order_count)

@@ -141,3 +141,55 @@ def test_naive_vs_exact_wobble(li_sf1):
         max_ulps = max(max_ulps, abs(a - b) / ulp)
         assert abs(a - b) / abs(b) < 1e-9   # still tiny in relative terms
     assert max_ulps >= 1.0  # naive is NOT within 1 ULP of exact at SF1
+
+
+class TestShipmodeStream:
+    """L_SMODE seed 675466456 pinned by the reference's own fixtures
+    (tests/golden/ref_fixtures.json, extracted by tools/extract_ref_goldens.py
+    from the 785 canonical SF1 lineitem rows + the Q12 answer)."""
+
+    MODES = ["REG AIR", "AIR", "RAIL", "TRUCK", "MAIL", "FOB", "SHIP"]
+
+    @pytest.fixture(scope="class")
+    def fixtures(self):
+        import json, os
+        p = os.path.join(os.path.dirname(__file__), "golden", "ref_fixtures.json")
+        return json.load(open(p))
+
+    def test_canonical_rows_all_columns(self, fixtures):
+        rows = fixtures["lineitem_canonical_sf1"]["rows"]
+        cols = oracle.gen_lineitem(1.0, 1, 300)
+        idx = {(int(o), int(l)): i
+               for i, (o, l) in enumerate(zip(cols["orderkey"], cols["linenumber"]))}
+        import datetime
+        epoch = datetime.date(1970, 1, 1)
+        for r in rows:
+            i = idx[(r["orderkey"], r["linenumber"])]
+            assert self.MODES[cols["shipmode"][i]] == r["shipmode"]
+            assert int(cols["quantity"][i]) == r["quantity"]
+            assert abs(cols["extendedprice"][i] - float(r["extendedprice"])) < 5e-3
+            assert abs(cols["discount"][i] - float(r["discount"])) < 1e-9
+            assert (epoch + datetime.timedelta(days=int(cols["shipdate"][i])
+                                               )).isoformat() == r["shipdate"]
+            assert (epoch + datetime.timedelta(days=int(cols["receiptdate"][i])
+                                               )).isoformat() == r["receiptdate"]
+            assert "ANR"[cols["returnflag"][i]] == r["returnflag"]
+
+    def test_q12_answer_exact(self, fixtures):
+        exp = {r["shipmode"]: (r["high"], r["low"]) for r in fixtures["q12_sf1"]["rows"]}
+        cols = oracle.gen_lineitem(1.0, columns=["orderkey", "shipmode", "shipdate",
+                                                 "commitdate", "receiptdate"])
+        orders = oracle.gen_orders(1.0)
+        pri = dict(zip(orders["orderkey"].tolist(), orders["orderpriority"].tolist()))
+        d94, d95 = 8766, 9131
+        m = ((np.isin(cols["shipmode"], [4, 6])) &
+             (cols["commitdate"] < cols["receiptdate"]) &
+             (cols["shipdate"] < cols["commitdate"]) &
+             (cols["receiptdate"] >= d94) & (cols["receiptdate"] < d95))
+        got = {}
+        for mode_id, name in ((4, "MAIL"), (6, "SHIP")):
+            sel = m & (cols["shipmode"] == mode_id)
+            oks = cols["orderkey"][sel]
+            highs = sum(1 for ok in oks.tolist() if pri[ok] <= 1)
+            got[name] = (highs, int(sel.sum()) - highs)
+        assert got == exp

@@ -1,0 +1,67 @@
+#!/usr/bin/env python3
+"""Extract golden fixtures from the reference's OWN test data into
+tests/golden/ref_fixtures.json (derived values with provenance, never file
+copies). Sources (all under /root/reference, PUBLIC reference content):
+
+ - plugin/trino-example-http/src/test/resources/example-data/lineitem-*.csv:
+   785 canonical dbgen SF1 lineitem rows (the standard first rows of
+   lineitem.tbl) — pins per-line generator streams positionally, including
+   l_shipmode (seed 675466456, found by exhaustive search over these rows);
+ - testing/trino-product-tests/src/test/resources/sql-tests/testcases/
+   hive_tpch/q12.result: the SF1 Q12 answer (MAIL 6202|9324, SHIP 6200|9262).
+
+Run in the build container (where /root/reference exists); the committed
+JSON travels to GPU boxes. Re-run to refresh: python tools/extract_ref_goldens.py
+"""
+import json
+import os
+
+REF = "/root/reference"
+OUT = os.path.join(os.path.dirname(__file__), "..", "tests", "golden",
+                   "ref_fixtures.json")
+
+
+def main():
+    fixtures = {}
+    # canonical lineitem rows: (orderkey, linenumber) -> selected columns
+    rows = []
+    base = f"{REF}/plugin/trino-example-http/src/test/resources/example-data"
+    for fn in ("lineitem-1.csv", "lineitem-2.csv"):
+        for line in open(f"{base}/{fn}"):
+            p = [x.strip() for x in line.split(",")]
+            if len(p) < 16:
+                continue
+            rows.append({
+                "orderkey": int(p[0]), "partkey": int(p[1]),
+                "linenumber": int(p[3]), "quantity": int(p[4]),
+                "extendedprice": p[5], "discount": p[6], "tax": p[7],
+                "returnflag": p[8], "linestatus": p[9],
+                "shipdate": p[10], "commitdate": p[11], "receiptdate": p[12],
+                "shipmode": p[14],
+            })
+    fixtures["lineitem_canonical_sf1"] = {
+        "source": "plugin/trino-example-http/src/test/resources/example-data/"
+                  "lineitem-{1,2}.csv",
+        "rows": rows,
+    }
+    q12 = []
+    path = (f"{REF}/testing/trino-product-tests/src/test/resources/sql-tests/"
+            "testcases/hive_tpch/q12.result")
+    for line in open(path):
+        if line.startswith("--"):
+            continue
+        p = [x for x in line.strip().split("|") if x]
+        if len(p) == 3:
+            q12.append({"shipmode": p[0], "high": int(p[1]), "low": int(p[2])})
+    fixtures["q12_sf1"] = {
+        "source": "testing/trino-product-tests/src/test/resources/sql-tests/"
+                  "testcases/hive_tpch/q12.result",
+        "rows": q12,
+    }
+    with open(OUT, "w") as f:
+        json.dump(fixtures, f, indent=1)
+    print(f"wrote {OUT}: {len(rows)} canonical rows, q12 = {q12}")
+
+
+if __name__ == "__main__":
+    main()

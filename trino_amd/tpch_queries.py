@@ -391,3 +391,80 @@ def q14_gpu(session, sf, order_start=1, order_count=None):
     for p in (p_pk, p_ty):
         _device_free(session, p)
     return dict(promo_revenue=100.0 * promo / total, elapsed=elapsed)
+
+
+MODE_MAIL, MODE_SHIP = 4, 6   # dists.dss order REG AIR,AIR,RAIL,TRUCK,MAIL,FOB,SHIP
+
+
+def q12_gpu(session, sf, order_start=1, order_count=None):
+    """TPC-H Q12 (testing/trino-benchmark-queries/.../tpch/q12.sql):
+    per shipmode in (MAIL, SHIP), count late lines split by order priority
+    (high = 1-URGENT/2-HIGH). Pins the L_SMODE stream: the SF1 answer must
+    equal the reference fixture testing/trino-product-tests/.../hive_tpch/
+    q12.result (MAIL 6202|9324, SHIP 6200|9262).
+    Pipeline: orders build (orderkey -> priority); lineitem filter
+    (mode IN + commit<receipt + ship<commit + receipt in 1994) -> join ->
+    project high-flag -> group by shipmode."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    o_ok = _device_buffer(session, order_count * 8)
+    o_pri = _device_buffer(session, order_count)
+    _check_lib(_lib.tg_tpch_gen_orders(session._h, sf, order_start, order_count,
+                                       o_ok, None, None, o_pri))
+    li = session.tpch_lineitem(sf, order_start, order_count,
+                               with_orderkey=True, with_dates=True,
+                               with_shipmode=True)
+    t0 = time.time()
+    bridge = ops.JoinBridge(session)
+    b = ops.hash_builder(session, bridge, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b.add_input(ops.page_from_device(session, ([(o_ok.value, ops.TG_BIGINT),
+                                                (o_pri.value, ops.TG_TINYINT)],
+                                               order_count)))
+    b.drain()
+    lpage = ops.page_from_device(session, ([(li.orderkey, ops.TG_BIGINT),
+                                            (li.shipmode, ops.TG_TINYINT),
+                                            (li.shipdate, ops.TG_INTEGER),
+                                            (li.commitdate, ops.TG_INTEGER),
+                                            (li.receiptdate, ops.TG_INTEGER)],
+                                           li.row_count))
+    f = ops.filter_project(session,
+                           ops.expr(("col", 1), ("i64", MODE_MAIL), "eq",
+                                    ("col", 1), ("i64", MODE_SHIP), "eq", "or",
+                                    ("col", 3), ("col", 4), "lt", "and",
+                                    ("col", 2), ("col", 3), "lt", "and",
+                                    ("col", 4), ("i64", DATE_1994_01_01), "ge", "and",
+                                    ("col", 4), ("i64", DATE_1994_01_01 + 365), "lt", "and"),
+                           [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                           [ops.TG_BIGINT, ops.TG_TINYINT])
+    f.add_input(lpage)
+    f.finish()
+    sel = _take_device_page(session, f)
+    j = ops.lookup_join(session, bridge, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    j.add_input(sel)
+    j.finish()
+    joined = _take_device_page(session, j)     # (shipmode, priority)
+    fp = ops.filter_project(session, None,
+                            [ops.expr(("col", 0)),
+                             ops.expr(("col", 1), ("i64", 1), "le")],
+                            [ops.TG_TINYINT, ops.TG_DOUBLE])
+    fp.add_input(joined)
+    fp.finish()
+    flagged = _take_device_page(session, fp)   # (shipmode, high 0/1)
+    agg = ops.hash_aggregation(session, [0], [ops.TG_TINYINT],
+                               [(ops.AGG_SUM_F64, 1), (ops.AGG_COUNT_STAR, -1)])
+    agg.add_input(flagged)
+    pages = agg.drain()
+    elapsed = time.time() - t0
+    for op in (b, f, j, fp, agg):
+        op.close()
+    bridge.close()
+    session.tpch_lineitem_free(li)
+    for p in (o_ok, o_pri):
+        _device_free(session, p)
+    out = pages[0]
+    mode = np.asarray(out[0]["values"])
+    high = np.asarray(out[1]["values"]).astype(np.int64)
+    cnt = np.asarray(out[2]["values"])
+    order = np.argsort(mode)                    # MAIL(4) before SHIP(6)
+    return dict(shipmode=mode[order], high=high[order],
+                low=(cnt - high)[order], elapsed=elapsed)
