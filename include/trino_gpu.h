@@ -227,6 +227,9 @@ typedef struct tg_tpch_lineitem_cols {
     int64_t* partkey;       /* optional (flags bit 2) */
     uint8_t* shipmode;      /* optional (flags bit 3): dictionary id 0..6 =
                                REG AIR,AIR,RAIL,TRUCK,MAIL,FOB,SHIP */
+    int64_t* tp_cents;      /* optional (flags bit 4): this line's
+                               o_totalprice contribution in cents (dbgen
+                               mk_order truncation; sums to o_totalprice) */
 } tg_tpch_lineitem_cols;
 
 /* Generate lineitem rows for orders [order_start, order_start+order_count)

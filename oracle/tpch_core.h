@@ -232,6 +232,11 @@ typedef struct {
     int32_t discount_pct;   /* 0..10 */
     int32_t tax_pct;        /* 0..8 */
     int64_t extprice_cents;
+    int64_t tp_cents;       /* this line's o_totalprice contribution:
+                               dbgen build.c mk_order integer truncation
+                               ((ep*(100-disc))/100*(100+tax))/100 — verified
+                               exact on the 200 canonical orders' totalprice
+                               (example-data/orders-*.csv) */
     uint8_t returnflag;     /* 0=A 1=N 2=R */
     uint8_t linestatus;     /* 0=F 1=O */
     uint8_t shipmode;       /* 0..6 = REG AIR,AIR,RAIL,TRUCK,MAIL,FOB,SHIP */
@@ -284,6 +289,8 @@ TPCH_HD static inline void tpch_gen_line(tpch_order_streams* s, const tpch_order
     }
     l->linestatus = (l->shipdate > TPCH_CURRENTDATE_EPOCH) ? 1 /*O*/ : 0 /*F*/;
     l->shipmode = (uint8_t)(tpch_rng_int(&s->smode, 1, 7) - 1);
+    int64_t t = l->extprice_cents * (100 - l->discount_pct) / 100;
+    l->tp_cents = t * (100 + l->tax_pct) / 100;
 }
 
 /* End the current order row: advance every stream to its fixed per-row usage. */

@@ -53,6 +53,22 @@ def main():
         p = [x for x in line.strip().split("|") if x]
         if len(p) == 3:
             q12.append({"shipmode": p[0], "high": int(p[1]), "low": int(p[2])})
+    q18 = []
+    p18 = (f"{REF}/testing/trino-product-tests/src/test/resources/sql-tests/"
+           "testcases/hive_tpch/q18.result")
+    for line in open(p18):
+        if line.startswith("--"):
+            continue
+        p = [x for x in line.strip().split("|") if x != ""]
+        if len(p) >= 6:
+            q18.append({"c_name": p[0], "custkey": int(p[1]),
+                        "orderkey": int(p[2]), "orderdate": p[3],
+                        "totalprice": p[4], "sum_qty": p[5]})
+    fixtures["q18_sf1"] = {
+        "source": "testing/trino-product-tests/src/test/resources/sql-tests/"
+                  "testcases/hive_tpch/q18.result",
+        "rows": q18,
+    }
     fixtures["q12_sf1"] = {
         "source": "testing/trino-product-tests/src/test/resources/sql-tests/"
                   "testcases/hive_tpch/q12.result",

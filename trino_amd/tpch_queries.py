@@ -468,3 +468,75 @@ def q12_gpu(session, sf, order_start=1, order_count=None):
     order = np.argsort(mode)                    # MAIL(4) before SHIP(6)
     return dict(shipmode=mode[order], high=high[order],
                 low=(cnt - high)[order], elapsed=elapsed)
+
+
+def q18_gpu(session, sf, order_start=1, order_count=None, limit=100):
+    """TPC-H Q18 (large-orders query): customers with orders whose total
+    quantity exceeds 300. The SF1 answer must equal the reference fixture
+    (hive_tpch/q18.result; o_totalprice to the cent via the generator's
+    tp_cents column — dbgen mk_order truncation, verified on the canonical
+    orders). Pipeline: lineitem group-by orderkey (sum qty ~1.5M groups, sum
+    tp_cents) -> HAVING filter -> build -> probe orders -> TopN
+    (totalprice DESC, orderdate ASC) -> c_name formatting host-side."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    o_ok = _device_buffer(session, order_count * 8)
+    o_ck = _device_buffer(session, order_count * 8)
+    o_od = _device_buffer(session, order_count * 4)
+    _check_lib(_lib.tg_tpch_gen_orders(session._h, sf, order_start, order_count,
+                                       o_ok, o_ck, o_od, None))
+    li = session.tpch_lineitem(sf, order_start, order_count,
+                               with_orderkey=True, with_totalprice=True)
+    t0 = time.time()
+    lpage = ops.page_from_device(session, ([(li.orderkey, ops.TG_BIGINT),
+                                            (li.quantity, ops.TG_DOUBLE),
+                                            (li.tp_cents, ops.TG_BIGINT)],
+                                           li.row_count))
+    agg1 = ops.hash_aggregation(session, [0], [ops.TG_BIGINT],
+                                [(ops.AGG_SUM_F64_EXACT, 1, 0),
+                                 (ops.AGG_SUM_I64, 2)])
+    agg1.add_input(lpage)
+    agg1.finish()
+    groups = _take_device_page(session, agg1)   # (orderkey, sumqty, totcents)
+    f = ops.filter_project(session, ops.expr(("col", 1), ("f64", 300.0), "gt"),
+                           [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                            ops.expr(("col", 2))],
+                           [ops.TG_BIGINT, ops.TG_DOUBLE, ops.TG_BIGINT])
+    f.add_input(groups)
+    f.finish()
+    big = _take_device_page(session, f)
+    bridge = ops.JoinBridge(session)
+    b = ops.hash_builder(session, bridge,
+                         [ops.TG_BIGINT, ops.TG_DOUBLE, ops.TG_BIGINT], [0], [1, 2])
+    b.add_input(big)
+    b.drain()
+    j = ops.lookup_join(session, bridge,
+                        [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_INTEGER],
+                        [0], [0, 1, 2])
+    j.add_input(ops.page_from_device(session, ([(o_ok.value, ops.TG_BIGINT),
+                                                (o_ck.value, ops.TG_BIGINT),
+                                                (o_od.value, ops.TG_INTEGER)],
+                                               order_count)))
+    j.finish()
+    matched = _take_device_page(session, j)  # (okey, ckey, odate, sumqty, totcents)
+    top = ops.topn(session, [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_INTEGER,
+                             ops.TG_DOUBLE, ops.TG_BIGINT],
+                   [4, 2], [1, 0], limit)    # totalprice DESC, orderdate ASC
+    top.add_input(matched)
+    pages = top.drain()
+    elapsed = time.time() - t0
+    for op in (agg1, f, b, j, top):
+        op.close()
+    bridge.close()
+    session.tpch_lineitem_free(li)
+    for p in (o_ok, o_ck, o_od):
+        _device_free(session, p)
+    tp = pages[0]
+    okey = np.asarray(tp[0]["values"])
+    ckey = np.asarray(tp[1]["values"])
+    odate = np.asarray(tp[2]["values"])
+    sumqty = np.asarray(tp[3]["values"])
+    totc = np.asarray(tp[4]["values"])
+    names = [f"Customer#{int(c):09d}" for c in ckey]
+    return dict(c_name=names, custkey=ckey, orderkey=okey, orderdate=odate,
+                totalprice_cents=totc, sum_qty=sumqty, elapsed=elapsed)
