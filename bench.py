@@ -94,7 +94,7 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--sf-per-gpu", type=float, default=100.0)
-    ap.add_argument("--workload", choices=["q1", "q3"], default="q1")
+    ap.add_argument("--workload", choices=["q1", "q3", "sweep"], default="q1")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--backend", choices=["nccl", "gloo"], default="nccl",
                     help="gloo only for multi-rank validation on one GPU "
@@ -122,6 +122,12 @@ def main():
 
     if args.workload == "q3":
         run_q3(args, sess, n_gpus, rank, dist)
+        sess.close()
+        if dist:
+            dist.destroy_process_group()
+        return
+    if args.workload == "sweep":
+        run_sweep(args, sess, n_gpus, rank)
         sess.close()
         if dist:
             dist.destroy_process_group()
@@ -327,6 +333,61 @@ def run_q3(args, sess, n_gpus, rank, tdist=None):
                              "probe-side bound (SURVEY §8d); the pipeline makes "
                              "multiple passes and its builds/probes are random-access "
                              "bound — per-kernel evidence: profiles/r01_q3_kernels_v5.txt"},
+        "cpu_baseline": None,
+    }
+    if rank == 0:
+        print(json.dumps(out), flush=True)
+
+
+def run_sweep(args, sess, n_gpus, rank):
+    """Implemented-query sweep (BASELINE config 5 shape, the round-1 subset):
+    the seven reference-fixture-exact queries back-to-back on one GPU.
+    Metric = queries per hour (power-run style); per-query ms reported.
+    Inputs are generated per query inside the timed region is NOT the case —
+    generation is excluded per query (HBM-resident inputs, like q1/q3)."""
+    from trino_amd import tpch_queries as q
+    sf = args.sf_per_gpu
+
+    def run_once(collect):
+        for name, fn in (("q1", lambda: _sweep_q1(sess, sf)),
+                         ("q3", lambda: q.q3_gpu(sess, sf)),
+                         ("q4", lambda: q.q4_gpu(sess, sf)),
+                         ("q6", lambda: q.q6_gpu(sess, sf)),
+                         ("q12", lambda: q.q12_gpu(sess, sf)),
+                         ("q14", lambda: q.q14_gpu(sess, sf)),
+                         ("q18", lambda: q.q18_gpu(sess, sf))):
+            r = fn()
+            collect[name] = collect.get(name, 0.0) + r["elapsed"]
+
+    def _sweep_q1(s, sf):
+        import trino_amd
+        li = s.tpch_lineitem(sf)
+        t0 = time.time()
+        s.q1(li)
+        el = time.time() - t0
+        s.tpch_lineitem_free(li)
+        return {"elapsed": el}
+
+    for _ in range(max(args.warmup, 1)):
+        run_once({})
+    per = {}
+    t0 = time.time()
+    for _ in range(args.steps):
+        run_once(per)
+    wall = time.time() - t0
+    out = {
+        "metric": "tpch_sweep_queries_per_hour",
+        "value": 7 * args.steps / wall * 3600, "unit": "queries/h",
+        "n_gpus": n_gpus, "steps": args.steps, "warmup": args.warmup,
+        "ms_per_step": wall * 1000 / args.steps, "higher_is_better": True,
+        "scaling": "weak", "vs_baseline": None, "dtype": "f64",
+        "data": "synthetic",
+        "config": {"workload": f"TPC-H 7-query sweep (Q1,Q3,Q4,Q6,Q12,Q14,Q18) "
+                               f"SF{sf:g} on {n_gpus}xMI355X; all reference-"
+                               f"fixture-exact at SF1",
+                   "per_query_ms": {k: round(v * 1000 / args.steps, 2)
+                                    for k, v in per.items()}},
+        "roofline": None,
         "cpu_baseline": None,
     }
     if rank == 0:
