@@ -342,9 +342,10 @@ def run_q3(args, sess, n_gpus, rank, tdist=None):
 def run_sweep(args, sess, n_gpus, rank):
     """Implemented-query sweep (BASELINE config 5 shape, the round-1 subset):
     the seven reference-fixture-exact queries back-to-back on one GPU.
-    Metric = queries per hour (power-run style); per-query ms reported.
-    Inputs are generated per query inside the timed region is NOT the case —
-    generation is excluded per query (HBM-resident inputs, like q1/q3)."""
+    Metric = queries per hour over WALL time (power-run style, including
+    per-query input generation); per_query_ms reports each query's own
+    pipeline time over HBM-resident inputs (generation excluded, like
+    q1/q3's step timing)."""
     from trino_amd import tpch_queries as q
     sf = args.sf_per_gpu
 
