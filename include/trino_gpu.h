@@ -166,6 +166,11 @@ typedef struct tg_agg_spec {
     int32_t _pad;
 } tg_agg_spec;
 
+/* StreamingAggregationOperator analog: input grouped (clustered) by a
+ * single non-null BIGINT key channel — no hash table; group ids are run
+ * indexes (first-occurrence row order), page-spanning runs continue. */
+tg_status tg_streaming_aggregation_create(tg_session*, int32_t key_channel,
+    const tg_agg_spec* aggs, int32_t n_aggs, int32_t step, tg_operator** out);
 tg_status tg_hash_aggregation_create(tg_session*,
     const int32_t* group_channels, int32_t n_group_channels,
     const int32_t* group_types,               /* tg_type per group channel */

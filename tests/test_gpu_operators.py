@@ -979,3 +979,51 @@ class TestSemiJoinNullSemantics:
         valid = np.asarray(m["valid"])
         assert vals[0] == 0 and ((int(valid[0]) >> 0) & 1) == 1   # miss, no null -> false
         assert vals[1] == 0 and ((int(valid[0]) >> 1) & 1) == 1   # NULL probe, empty set -> false
+
+
+class TestStreamingAggregation:
+    """StreamingAggregationOperator analog (ops_streamagg.hip): clustered
+    single-BIGINT-key input, group ids = run order, page-spanning runs
+    continue. Parity vs the hash aggregation operator on the same input."""
+
+    def test_matches_hash_agg_multi_page(self, sess, ops):
+        r = rng(95)
+        # clustered keys: sorted with duplicates, split at a run boundary AND
+        # mid-run across three pages
+        keys = np.sort(r.integers(0, 40_000, 120_000)).astype(np.int64)
+        vals = r.integers(90100, 209900, len(keys)) / 100.0
+        iv = r.integers(0, 1000, len(keys)).astype(np.int64)
+        cuts = [0, 50_000, 50_001, len(keys)]   # second page = 1 row (mid-run)
+        sa = ops.streaming_aggregation(sess, 0,
+                                       [(ops.AGG_COUNT_STAR, -1),
+                                        (ops.AGG_SUM_F64_EXACT, 1, 43),
+                                        (ops.AGG_SUM_I64, 2),
+                                        (ops.AGG_AVG_F64, 1)])
+        ha = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                  [(ops.AGG_COUNT_STAR, -1),
+                                   (ops.AGG_SUM_F64_EXACT, 1, 43),
+                                   (ops.AGG_SUM_I64, 2),
+                                   (ops.AGG_AVG_F64, 1)])
+        for lo, hi in zip(cuts[:-1], cuts[1:]):
+            page = [keys[lo:hi], vals[lo:hi], iv[lo:hi]]
+            sa.add_input(ops.page_from_numpy(page))
+            ha.add_input(ops.page_from_numpy(page))
+        so = sa.drain()[0]
+        ho = ha.drain()[0]
+        sa.close()
+        ha.close()
+        assert np.array_equal(so[0]["values"], ho[0]["values"])   # keys, run order
+        assert np.array_equal(so[1]["values"], ho[1]["values"])   # counts
+        assert np.array_equal(so[2]["values"], ho[2]["values"])   # exact sums: bit-equal
+        assert np.array_equal(so[3]["values"], ho[3]["values"])   # int sums
+        np.testing.assert_allclose(so[4]["values"], ho[4]["values"], rtol=1e-12)
+
+    def test_single_run_and_single_rows(self, sess, ops):
+        sa = ops.streaming_aggregation(sess, 0, [(ops.AGG_COUNT_STAR, -1)])
+        sa.add_input(ops.page_from_numpy([np.full(100, 7, np.int64)]))
+        sa.add_input(ops.page_from_numpy([np.full(1, 7, np.int64)]))
+        sa.add_input(ops.page_from_numpy([np.array([8], np.int64)]))
+        out = sa.drain()[0]
+        sa.close()
+        assert out[0]["values"].tolist() == [7, 8]
+        assert out[1]["values"].tolist() == [101, 1]

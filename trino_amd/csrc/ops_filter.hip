@@ -236,7 +236,7 @@ __global__ void k_scan_serial(int32_t* counts, int64_t n, int32_t* total)
     if (lane == 0) *total = run;
 }
 
-static tg_status run_scan_i32(tg_session* s, int32_t* d_arr, int64_t n, int32_t* d_total)
+tg_status run_scan_i32(tg_session* s, int32_t* d_arr, int64_t n, int32_t* d_total)
 {
     if (n <= 2 * FSCAN_CHUNK) {   /* small: one launch beats three */
         hipLaunchKernelGGL(k_scan_serial, dim3(1), dim3(64), 0, s->stream,

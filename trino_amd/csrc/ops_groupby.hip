@@ -183,16 +183,8 @@ __global__ void k_gt_rehash(GTable t, int32_t n_groups, const int32_t* types, in
     }
 }
 
-/* ---- aggregation update ---- */
-struct KAgg {
-    int32_t fn;           /* tg_agg_fn */
-    int32_t in_ch;        /* -1 for COUNT_STAR; AVG FINAL: count ch, sum = ch+1;
-                             SUM_F64_EXACT FINAL: lo ch, hi = ch+1 */
-    double* sum;          /* f64 state (or null); SUM_F64_EXACT: i128 HI words */
-    long long* cnt;       /* i64 state (count / int sum); SUM_F64_EXACT: LO words */
-    double scale;         /* SUM_F64_EXACT: 2^scale_pow */
-};
-#define MAX_AGGS 12
+/* ---- aggregation update (KAgg + shared kernel decls: agg_common.h) ---- */
+#include "agg_common.h"
 
 __global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
                              const KColH* cols, const KAgg* aggs, int n_aggs,

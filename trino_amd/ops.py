@@ -141,6 +141,10 @@ _lib.tg_filter_run.restype = ctypes.c_int
 _lib.tg_filter_run.argtypes = [ctypes.c_void_p, ctypes.POINTER(TgExpr),
                                ctypes.POINTER(TgPage), ctypes.POINTER(TgSelected),
                                ctypes.c_void_p, ctypes.c_void_p]
+_lib.tg_streaming_aggregation_create.restype = ctypes.c_int
+_lib.tg_streaming_aggregation_create.argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                                 ctypes.c_void_p, ctypes.c_int32,
+                                                 ctypes.c_int32, ctypes.c_void_p]
 _lib.tg_hash_aggregation_create.restype = ctypes.c_int
 _lib.tg_hash_aggregation_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                             ctypes.c_int32, ctypes.c_void_p,
@@ -263,6 +267,22 @@ def filter_run(session, filter_expr, page, input_sel=None):
                               ctypes.byref(input_sel) if input_sel else None,
                               out.ctypes.data, ctypes.byref(cnt)))
     return out[:cnt.value]
+
+
+def streaming_aggregation(session, key_channel, aggs, step=STEP_SINGLE):
+    """StreamingAggregationOperator analog: input clustered by a single
+    non-null BIGINT key. aggs: list of (fn, input_channel[, scale_pow])."""
+    h = ctypes.c_void_p()
+    sp = (TgAggSpec * len(aggs))()
+    for i, a in enumerate(aggs):
+        sp[i].fn = a[0]
+        sp[i].input_channel = a[1]
+        sp[i].scale_pow = a[2] if len(a) > 2 else 0
+    _check(_lib.tg_streaming_aggregation_create(session._h, key_channel, sp,
+                                                len(aggs), step, ctypes.byref(h)))
+    op = Operator(session, h)
+    op._keep = (sp,)
+    return op
 
 
 def hash_aggregation(session, group_channels, group_types, aggs, step=STEP_SINGLE):
