@@ -121,6 +121,112 @@ __global__ void k_filter_cmp(KCol col, int op, double cval,
     }
 }
 
+/* conjunction fast path: AND-chains of simple terms (the dominant sweep
+ * filter shape — Q4/Q6/Q12's date windows, col-col compares and IN pairs).
+ * Extracted host-side from the postfix program; each term is evaluated
+ * directly, early-out on the first false, no interpreter stack. */
+#define FTERM_MAX 8
+struct FTerm {
+    int32_t kind;     /* 0 = colA cmp const, 1 = colA cmp colB,
+                         2 = colA between [c1,c2], 3 = colA in {c1,c2} */
+    int32_t colA, colB, op;
+    double c1, c2;
+};
+struct FTerms { FTerm t[FTERM_MAX]; int n; };
+
+__device__ static inline bool fterm_cmp(int op, double a, double b)
+{
+    switch (op) {
+        case TG_EXPR_LE: return a <= b;
+        case TG_EXPR_LT: return a < b;
+        case TG_EXPR_GE: return a >= b;
+        case TG_EXPR_GT: return a > b;
+        case TG_EXPR_EQ: return a == b;
+        default: return a != b;
+    }
+}
+
+__global__ void k_filter_terms(FTerms ft, const KCol* __restrict__ cols,
+                               int has_list, const int32_t* __restrict__ list,
+                               int32_t offset, int64_t n,
+                               uint8_t* __restrict__ flags)
+{
+    int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; k < n; k += stride) {
+        int64_t i = has_list ? (int64_t)list[k] : offset + k;
+        bool pass = true;
+        for (int t = 0; t < ft.n && pass; t++) {
+            const FTerm& f = ft.t[t];
+            bool nl = false;
+            double a = load_col(cols[f.colA], i, &nl);
+            switch (f.kind) {
+                case 0: pass = !nl && fterm_cmp(f.op, a, f.c1); break;
+                case 1: {
+                    double b = load_col(cols[f.colB], i, &nl);
+                    pass = !nl && fterm_cmp(f.op, a, b);
+                    break;
+                }
+                case 2: pass = !nl && a >= f.c1 && a <= f.c2; break;
+                default: pass = !nl && (a == f.c1 || a == f.c2); break;
+            }
+        }
+        flags[k] = pass ? 1 : 0;
+    }
+}
+
+/* try to parse the postfix program as T1 (T2 AND)*; returns term count or 0 */
+static int parse_fterms(const ExprProgram& pred, FTerms* out)
+{
+    auto cval = [](const tg_expr_inst& in, double* v) {
+        if (in.op == TG_EXPR_CONST_F64) { *v = in.imm.f64; return true; }
+        if (in.op == TG_EXPR_CONST_I64) { *v = (double)in.imm.i64; return true; }
+        return false;
+    };
+    auto is_cmp = [](int op) { return op >= TG_EXPR_LE && op <= TG_EXPR_NE; };
+    int p = 0, nt = 0;
+    const auto* I = pred.insts.data();
+    int N = pred.count;
+    auto term = [&]() -> bool {
+        if (nt >= FTERM_MAX || p >= N || I[p].op != TG_EXPR_COL) return false;
+        FTerm& f = out->t[nt];
+        /* IN-pair: COL C EQ COL C EQ OR (same column) */
+        if (p + 6 < N && cval(I[p + 1], &f.c1) && I[p + 2].op == TG_EXPR_EQ &&
+            I[p + 3].op == TG_EXPR_COL && I[p + 3].arg0 == I[p].arg0 &&
+            cval(I[p + 4], &f.c2) && I[p + 5].op == TG_EXPR_EQ &&
+            I[p + 6].op == TG_EXPR_OR) {
+            f.kind = 3; f.colA = I[p].arg0; f.colB = -1; f.op = 0;
+            p += 7; nt++; return true;
+        }
+        /* BETWEEN: COL C C BETWEEN */
+        if (p + 3 < N && cval(I[p + 1], &f.c1) && cval(I[p + 2], &f.c2) &&
+            I[p + 3].op == TG_EXPR_BETWEEN) {
+            f.kind = 2; f.colA = I[p].arg0; f.colB = -1; f.op = 0;
+            p += 4; nt++; return true;
+        }
+        /* COL cmp CONST */
+        if (p + 2 < N && cval(I[p + 1], &f.c1) && is_cmp(I[p + 2].op)) {
+            f.kind = 0; f.colA = I[p].arg0; f.colB = -1; f.op = I[p + 2].op;
+            p += 3; nt++; return true;
+        }
+        /* COL cmp COL */
+        if (p + 2 < N && I[p + 1].op == TG_EXPR_COL && is_cmp(I[p + 2].op)) {
+            f.kind = 1; f.colA = I[p].arg0; f.colB = I[p + 1].arg0;
+            f.op = I[p + 2].op; f.c1 = f.c2 = 0;
+            p += 3; nt++; return true;
+        }
+        return false;
+    };
+    if (!term()) return 0;
+    while (p < N) {
+        if (!term()) return 0;
+        if (p >= N || I[p].op != TG_EXPR_AND) return 0;
+        p++;
+    }
+    out->n = nt;
+    return nt;
+}
+
 __global__ void k_filter_flags(const tg_expr_inst* prog, int count, const KCol* cols,
                                int has_list, const int32_t* list, int32_t offset,
                                int64_t n, uint8_t* __restrict__ flags)
@@ -451,6 +557,10 @@ tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page
         hipLaunchKernelGGL(k_filter_cmp, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
                            cols[pred.insts[0].arg0], pred.insts[2].op, cval,
                            has_list, d_list, offset, n, d_flags);
+    }
+    else if (FTerms ft{}; parse_fterms(pred, &ft) > 0) {
+        hipLaunchKernelGGL(k_filter_terms, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
+                           ft, d_cols, has_list, d_list, offset, n, d_flags);
     }
     else {
         hipLaunchKernelGGL(k_filter_flags, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
