@@ -460,10 +460,201 @@ static int run_q14(tg_session* s, double sf)
     return 0;
 }
 
+
+static int run_q12(tg_session* s, double sf)
+{
+    /* Q12 (python mirror: q12_gpu): small filtered-lineitem build, probe
+     * orders; conditional counts by shipmode. MAIL=4 SHIP=6. */
+    int64_t n_orders = (int64_t)(1500000 * sf);
+    void *o_ok, *o_pri;
+    die(tg_device_malloc(s, &o_ok, n_orders * 8), "malloc");
+    die(tg_device_malloc(s, &o_pri, n_orders), "malloc");
+    die(tg_tpch_gen_orders(s, sf, 1, n_orders, (int64_t*)o_ok, nullptr, nullptr,
+                           (uint8_t*)o_pri), "gen ord");
+    tg_tpch_lineitem_cols li;
+    die(tg_tpch_lineitem_alloc(s, sf, 1, n_orders, 1 | 2 | 8, &li), "gen li");
+    int fin = 0;
+
+    std::vector<tg_block> lb = {dev_block(TG_BIGINT, li.orderkey, li.row_count),
+                                dev_block(TG_TINYINT, li.shipmode, li.row_count),
+                                dev_block(TG_INTEGER, li.shipdate, li.row_count),
+                                dev_block(TG_INTEGER, li.commitdate, li.row_count),
+                                dev_block(TG_INTEGER, li.receiptdate, li.row_count)};
+    tg_page lpage = dev_page(lb, li.row_count);
+    std::vector<tg_expr_inst> lf = {
+        I_col(1), I_i64(4), I_op(TG_EXPR_EQ), I_col(1), I_i64(6), I_op(TG_EXPR_EQ),
+        I_op(TG_EXPR_OR),
+        I_col(3), I_col(4), I_op(TG_EXPR_LT), I_op(TG_EXPR_AND),
+        I_col(2), I_col(3), I_op(TG_EXPR_LT), I_op(TG_EXPR_AND),
+        I_col(4), I_i64(8766), I_op(TG_EXPR_GE), I_op(TG_EXPR_AND),
+        I_col(4), I_i64(9131), I_op(TG_EXPR_LT), I_op(TG_EXPR_AND)};
+    std::vector<tg_expr_inst> lp0 = {I_col(0)}, lp1 = {I_col(1)};
+    tg_expr lfe{lf.data(), (int32_t)lf.size()};
+    tg_expr lpr[2] = {{lp0.data(), 1}, {lp1.data(), 1}};
+    int32_t lot[2] = {TG_BIGINT, TG_TINYINT};
+    tg_operator* f = nullptr;
+    die(tg_filter_project_create(s, &lfe, lpr, lot, 2, &f), "f");
+    die(tg_operator_add_input(f, &lpage), "f add");
+    die(tg_operator_finish(f), "f fin");
+    tg_page sel{};
+    die(tg_operator_get_output(f, &sel, &fin), "f out");
+
+    tg_join_bridge* br = nullptr;
+    die(tg_join_bridge_create(s, &br), "br");
+    int32_t bt[2] = {TG_BIGINT, TG_TINYINT}, kc0 = 0, bout = 1;
+    tg_operator* b = nullptr;
+    die(tg_hash_builder_create(s, br, bt, 2, &kc0, 1, &bout, 1, &b), "b");
+    die(tg_operator_add_input(b, &sel), "b add");
+    die(tg_operator_finish(b), "b fin");
+
+    std::vector<tg_block> ob = {dev_block(TG_BIGINT, o_ok, n_orders),
+                                dev_block(TG_TINYINT, o_pri, n_orders)};
+    tg_page opage = dev_page(ob, n_orders);
+    int32_t pt[2] = {TG_BIGINT, TG_TINYINT}, jout = 1;
+    tg_operator* j = nullptr;
+    die(tg_lookup_join_create(s, br, pt, 2, &kc0, 1, &jout, 1, &j), "j");
+    die(tg_operator_add_input(j, &opage), "j add");
+    die(tg_operator_finish(j), "j fin");
+    tg_page joined{};
+    die(tg_operator_get_output(j, &joined, &fin), "j out");  /* (pri, mode) */
+
+    std::vector<tg_expr_inst> pm = {I_col(1)};
+    std::vector<tg_expr_inst> ph = {I_col(0), I_i64(1), I_op(TG_EXPR_LE)};
+    tg_expr pr2[2] = {{pm.data(), 1}, {ph.data(), (int32_t)ph.size()}};
+    int32_t pot[2] = {TG_TINYINT, TG_DOUBLE};
+    tg_operator* fp = nullptr;
+    die(tg_filter_project_create(s, nullptr, pr2, pot, 2, &fp), "fp");
+    die(tg_operator_add_input(fp, &joined), "fp add");
+    die(tg_operator_finish(fp), "fp fin");
+    tg_page flagged{};
+    die(tg_operator_get_output(fp, &flagged, &fin), "fp out");
+
+    int32_t gch = 0, gty = TG_TINYINT;
+    tg_agg_spec ags[2] = {{TG_AGG_SUM_F64, 1}, {TG_AGG_COUNT_STAR, -1}};
+    tg_operator* agg = nullptr;
+    die(tg_hash_aggregation_create(s, &gch, 1, &gty, ags, 2, TG_STEP_SINGLE, &agg), "agg");
+    die(tg_operator_add_input(agg, &flagged), "agg add");
+    die(tg_operator_finish(agg), "agg fin");
+    tg_page out{};
+    die(tg_operator_get_output(agg, &out, &fin), "agg out");
+    printf("l_shipmode|high_line_count|low_line_count\n");
+    struct Row { int8_t m; double high; int64_t cnt; };
+    std::vector<Row> rows;
+    for (int64_t i = 0; i < out.position_count; i++) {
+        Row r;
+        die(tg_copy_dtoh(s, &r.m, (const char*)out.blocks[0].data + i, 1), "dtoh");
+        die(tg_copy_dtoh(s, &r.high, (const char*)out.blocks[1].data + i * 8, 8), "dtoh");
+        die(tg_copy_dtoh(s, &r.cnt, (const char*)out.blocks[2].data + i * 8, 8), "dtoh");
+        rows.push_back(r);
+    }
+    static const char* MODES[7] = {"REG AIR", "AIR", "RAIL", "TRUCK", "MAIL",
+                                   "FOB", "SHIP"};
+    for (int m = 0; m < 7; m++)
+        for (auto& r : rows)
+            if (r.m == m)
+                printf("%s|%lld|%lld\n", MODES[m], (long long)r.high,
+                       (long long)(r.cnt - (long long)r.high));
+    for (tg_operator* o : {f, b, j, fp, agg}) tg_operator_close(o);
+    tg_join_bridge_close(br);
+    die(tg_tpch_lineitem_free(s, &li), "free");
+    for (void* pp : {o_ok, o_pri}) die(tg_device_free(s, pp), "free");
+    return 0;
+}
+
+static int run_q18(tg_session* s, double sf)
+{
+    /* Q18 (python mirror: q18_gpu): streaming aggregation over the
+     * orderkey-clustered lineitem, HAVING, join orders, TopN 100. */
+    int64_t n_orders = (int64_t)(1500000 * sf);
+    void *o_ok, *o_ck, *o_od;
+    die(tg_device_malloc(s, &o_ok, n_orders * 8), "malloc");
+    die(tg_device_malloc(s, &o_ck, n_orders * 8), "malloc");
+    die(tg_device_malloc(s, &o_od, n_orders * 4), "malloc");
+    die(tg_tpch_gen_orders(s, sf, 1, n_orders, (int64_t*)o_ok, (int64_t*)o_ck,
+                           (int32_t*)o_od, nullptr), "gen ord");
+    tg_tpch_lineitem_cols li;
+    die(tg_tpch_lineitem_alloc(s, sf, 1, n_orders, 1 | 16, &li), "gen li");
+    int fin = 0;
+
+    std::vector<tg_block> lb = {dev_block(TG_BIGINT, li.orderkey, li.row_count),
+                                dev_block(TG_DOUBLE, li.quantity, li.row_count),
+                                dev_block(TG_BIGINT, li.tp_cents, li.row_count)};
+    tg_page lpage = dev_page(lb, li.row_count);
+    tg_agg_spec sags[2] = {{TG_AGG_SUM_F64_EXACT, 1, 0}, {TG_AGG_SUM_I64, 2}};
+    tg_operator* sa = nullptr;
+    die(tg_streaming_aggregation_create(s, 0, sags, 2, TG_STEP_SINGLE, &sa), "sa");
+    die(tg_operator_add_input(sa, &lpage), "sa add");
+    die(tg_operator_finish(sa), "sa fin");
+    tg_page groups{};
+    die(tg_operator_get_output(sa, &groups, &fin), "sa out");
+
+    std::vector<tg_expr_inst> hf = {I_col(1), I_f64(300.0), I_op(TG_EXPR_GT)};
+    std::vector<tg_expr_inst> h0 = {I_col(0)}, h1 = {I_col(1)}, h2 = {I_col(2)};
+    tg_expr hfe{hf.data(), (int32_t)hf.size()};
+    tg_expr hpr[3] = {{h0.data(), 1}, {h1.data(), 1}, {h2.data(), 1}};
+    int32_t hot[3] = {TG_BIGINT, TG_DOUBLE, TG_BIGINT};
+    tg_operator* f = nullptr;
+    die(tg_filter_project_create(s, &hfe, hpr, hot, 3, &f), "having");
+    die(tg_operator_add_input(f, &groups), "having add");
+    die(tg_operator_finish(f), "having fin");
+    tg_page big{};
+    die(tg_operator_get_output(f, &big, &fin), "having out");
+
+    tg_join_bridge* br = nullptr;
+    die(tg_join_bridge_create(s, &br), "br");
+    int32_t bt[3] = {TG_BIGINT, TG_DOUBLE, TG_BIGINT}, kc0 = 0, bouts[2] = {1, 2};
+    tg_operator* b = nullptr;
+    die(tg_hash_builder_create(s, br, bt, 3, &kc0, 1, bouts, 2, &b), "b");
+    die(tg_operator_add_input(b, &big), "b add");
+    die(tg_operator_finish(b), "b fin");
+
+    std::vector<tg_block> ob = {dev_block(TG_BIGINT, o_ok, n_orders),
+                                dev_block(TG_BIGINT, o_ck, n_orders),
+                                dev_block(TG_INTEGER, o_od, n_orders)};
+    tg_page opage = dev_page(ob, n_orders);
+    int32_t pt[3] = {TG_BIGINT, TG_BIGINT, TG_INTEGER}, jouts[3] = {0, 1, 2};
+    tg_operator* j = nullptr;
+    die(tg_lookup_join_create(s, br, pt, 3, &kc0, 1, jouts, 3, &j), "j");
+    die(tg_operator_add_input(j, &opage), "j add");
+    die(tg_operator_finish(j), "j fin");
+    tg_page matched{};
+    die(tg_operator_get_output(j, &matched, &fin), "j out");
+    /* (okey, ckey, odate, sumqty, totcents) */
+
+    int32_t tty[5] = {TG_BIGINT, TG_BIGINT, TG_INTEGER, TG_DOUBLE, TG_BIGINT};
+    int32_t sch[2] = {4, 2}, sdsc[2] = {1, 0};
+    tg_operator* top = nullptr;
+    die(tg_topn_create(s, tty, 5, sch, sdsc, 2, 100, &top), "topn");
+    die(tg_operator_add_input(top, &matched), "topn add");
+    die(tg_operator_finish(top), "topn fin");
+    tg_page t100{};
+    die(tg_operator_get_output(top, &t100, &fin), "topn out");
+
+    printf("c_name|c_custkey|o_orderkey|o_orderdate|o_totalprice|sum_qty\n");
+    for (int64_t i = 0; i < t100.position_count; i++) {
+        int64_t ok, ck, tc;
+        int32_t od;
+        double sq;
+        die(tg_copy_dtoh(s, &ok, (const char*)t100.blocks[0].data + i * 8, 8), "dtoh");
+        die(tg_copy_dtoh(s, &ck, (const char*)t100.blocks[1].data + i * 8, 8), "dtoh");
+        die(tg_copy_dtoh(s, &od, (const char*)t100.blocks[2].data + i * 4, 4), "dtoh");
+        die(tg_copy_dtoh(s, &sq, (const char*)t100.blocks[3].data + i * 8, 8), "dtoh");
+        die(tg_copy_dtoh(s, &tc, (const char*)t100.blocks[4].data + i * 8, 8), "dtoh");
+        printf("Customer#%09lld|%lld|%lld|%d|%lld.%02lld|%lld\n",
+               (long long)ck, (long long)ck, (long long)ok, od,
+               (long long)(tc / 100), (long long)(tc % 100), (long long)sq);
+    }
+    for (tg_operator* o : {sa, f, b, j, top}) tg_operator_close(o);
+    tg_join_bridge_close(br);
+    die(tg_tpch_lineitem_free(s, &li), "free");
+    for (void* pp : {o_ok, o_ck, o_od}) die(tg_device_free(s, pp), "free");
+    return 0;
+}
+
 int main(int argc, char** argv)
 {
     if (argc < 2 || !strcmp(argv[1], "--help")) {
-        printf("usage: %s q1|q3|q4|q6|q14 [scale_factor]  (version: %s)\n",
+        printf("usage: %s q1|q3|q4|q6|q12|q14|q18 [scale_factor]  (version: %s)\n",
                argv[0], tg_version());
         return argc < 2 ? 1 : 0;
     }
@@ -480,6 +671,8 @@ int main(int argc, char** argv)
     else if (!strcmp(argv[1], "q6")) rc = run_q6(s, sf);
     else if (!strcmp(argv[1], "q4")) rc = run_q4(s, sf);
     else if (!strcmp(argv[1], "q14")) rc = run_q14(s, sf);
+    else if (!strcmp(argv[1], "q12")) rc = run_q12(s, sf);
+    else if (!strcmp(argv[1], "q18")) rc = run_q18(s, sf);
     else fprintf(stderr, "unknown query %s\n", argv[1]);
     tg_session_close(s);
     return rc;
