@@ -312,6 +312,12 @@ __global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
                     }
                     todo &= ~mask;
                     rounds++;
+                    /* small first cluster => likely high cardinality: the
+                     * match loop would cost more than direct atomics (run-
+                     * structured inputs like streaming agg average ~4 rows
+                     * per group) — bail and let leftovers take the fast
+                     * per-lane path */
+                    if (rounds == 1 && __popcll(mask) < 8) break;
                 }
                 /* leftovers (wave saw >8 distinct groups): direct atomics */
                 bool leftover = active && ((todo >> lane) & 1ull);
