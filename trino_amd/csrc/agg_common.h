@@ -17,6 +17,9 @@ struct KAgg {
 __global__ void k_agg_update(const int32_t* gids, int64_t n,
                              const KColH* cols, const KAgg* aggs, int n_aggs,
                              int step);
+__global__ void k_agg_update_sorted(const int32_t* gids, int64_t n,
+                                    const KColH* cols, const KAgg* aggs,
+                                    int n_aggs, int step);
 __global__ void k_emit_f64(const double* state, const int32_t* old_by_new,
                            int32_t n, double* out);
 __global__ void k_emit_i64(const long long* state, const int32_t* old_by_new,

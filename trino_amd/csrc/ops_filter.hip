@@ -357,9 +357,10 @@ tg_status run_scan_i32(tg_session* s, int32_t* d_arr, int64_t n, int32_t* d_tota
     hipLaunchKernelGGL(k_s32_chunk_sums, dim3((uint32_t)((nchunks + wpb - 1) / wpb)),
                        dim3(TG_BLOCK), 0, s->stream, d_arr, n, d_cs, nchunks);
     TG_HIP_CHECK(hipGetLastError());
-    hipLaunchKernelGGL(k_s32_chunks_serial, dim3(1), dim3(1), 0, s->stream,
-                       d_cs, nchunks, d_total);
-    TG_HIP_CHECK(hipGetLastError());
+    /* the middle scan recurses: at 600M inputs nchunks is ~293k and a
+     * single-thread pass costs ~15 ms */
+    tg_status st = run_scan_i32(s, d_cs, nchunks, d_total);
+    if (st != TG_OK) return st;
     hipLaunchKernelGGL(k_s32_offsets, dim3((uint32_t)((nchunks + wpb - 1) / wpb)),
                        dim3(TG_BLOCK), 0, s->stream, d_arr, n, d_cs, nchunks);
     TG_HIP_CHECK(hipGetLastError());

@@ -162,9 +162,10 @@ struct StreamAggOp : tg_operator {
             TG_HIP_CHECK(hipMemcpyAsync(d_aggs, agg_state.data(),
                                         agg_state.size() * sizeof(KAgg),
                                         hipMemcpyHostToDevice, s->stream));
-            hipLaunchKernelGGL(k_agg_update, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
-                               0, s->stream, d_gids, in.n, d_all, d_aggs,
-                               (int)agg_state.size(), step == TG_STEP_FINAL ? 1 : 0);
+            hipLaunchKernelGGL(k_agg_update_sorted, dim3(tg_grid_for(in.n)),
+                               dim3(TG_BLOCK), 0, s->stream, d_gids, in.n, d_all,
+                               d_aggs, (int)agg_state.size(),
+                               step == TG_STEP_FINAL ? 1 : 0);
             TG_HIP_CHECK(hipGetLastError());
             TG_HIP_CHECK(hipStreamSynchronize(s->stream));
             tg_pool_free(s, d_aggs);
