@@ -280,6 +280,12 @@ tg_status tg_topn_create(tg_session*,
 
 /* ---- semi join (operator/HashSemiJoinOperator.java): appends a BOOLEAN
  * matched channel to the probe page (NULL for null probe keys) ---- */
+/* SetBuilderOperator analog (ChannelSet): semi-join membership source.
+ * Dense single-BIGINT key ranges build a bitmap; sparse fall back to the
+ * positional index. The bridge then serves tg_semi_join_create only. */
+tg_status tg_set_builder_create(tg_session*, tg_join_bridge*,
+    const int32_t* build_types, int32_t n_build_channels,
+    int32_t key_channel, tg_operator** out);
 tg_status tg_semi_join_create(tg_session*, tg_join_bridge*,
     int32_t key_channel, tg_operator** out);
 

@@ -1027,3 +1027,45 @@ class TestStreamingAggregation:
         sa.close()
         assert out[0]["values"].tolist() == [7, 8]
         assert out[1]["values"].tolist() == [101, 1]
+
+
+class TestSetBuilder:
+    """SetBuilderOperator analog (ChannelSet): dense-range bitmap semi-join
+    source. Results must equal the positional-index semi join, including the
+    three-valued null semantics."""
+
+    def test_bitmap_matches_index_semi(self, sess, ops):
+        r = rng(120)
+        nb, m = 40_000, 150_000
+        bk = r.integers(10**6, 10**6 + 500_000, nb).astype(np.int64)
+        b_nulls = r.random(nb) < 0.02
+        bvalid = np.full((nb + 63) // 64, ~np.uint64(0), np.uint64)
+        for i in np.nonzero(b_nulls)[0]:
+            bvalid[i >> 6] &= ~np.uint64(1 << (i & 63))
+        pk = r.integers(10**6 - 1000, 10**6 + 501_000, m).astype(np.int64)
+        p_nulls = r.random(m) < 0.02
+        pvalid = np.full((m + 63) // 64, ~np.uint64(0), np.uint64)
+        for i in np.nonzero(p_nulls)[0]:
+            pvalid[i >> 6] &= ~np.uint64(1 << (i & 63))
+
+        def run(use_set):
+            bridge = ops.JoinBridge(sess)
+            if use_set:
+                b = ops.set_builder(sess, bridge, [ops.TG_BIGINT], 0)
+            else:
+                b = ops.hash_builder(sess, bridge, [ops.TG_BIGINT], [0], [])
+            b.add_input(ops.page_from_numpy([bk], valids=[bvalid]))
+            b.drain()
+            sj = ops.semi_join(sess, bridge, 0)
+            sj.add_input(ops.page_from_numpy([pk], valids=[pvalid]))
+            out, _ = sj.get_output()
+            b.close()
+            sj.close()
+            bridge.close()
+            return (out[-1]["values"].astype(np.int8).copy(),
+                    np.asarray(out[-1]["valid"]).copy())
+
+        vb, valb = run(True)
+        vi, vali = run(False)
+        assert np.array_equal(vb, vi)
+        assert np.array_equal(valb, vali)

@@ -418,9 +418,25 @@ def join_key_range(bridge):
     return mn.value, mx.value, nr.value
 
 
+_lib.tg_set_builder_create.restype = ctypes.c_int
+_lib.tg_set_builder_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                       ctypes.c_void_p, ctypes.c_int32,
+                                       ctypes.c_int32, ctypes.c_void_p]
 _lib.tg_semi_join_create.restype = ctypes.c_int
 _lib.tg_semi_join_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                      ctypes.c_int32, ctypes.c_void_p]
+
+
+def set_builder(session, bridge, build_types, key_channel):
+    """SetBuilderOperator analog: semi-join membership source (bitmap for
+    dense BIGINT ranges)."""
+    h = ctypes.c_void_p()
+    bt = _i32arr(build_types)
+    _check(_lib.tg_set_builder_create(session._h, bridge._h, bt.ctypes.data,
+                                      len(bt), key_channel, ctypes.byref(h)))
+    op = Operator(session, h)
+    op._keep = (bt,)
+    return op
 
 
 def semi_join(session, bridge, key_channel):

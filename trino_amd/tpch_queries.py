@@ -287,7 +287,9 @@ def q4_gpu(session, sf, order_start=1, order_count=None):
     f1.finish()
     late = _take_device_page(session, f1)
     bridge = ops.JoinBridge(session)
-    b = ops.hash_builder(session, bridge, [ops.TG_BIGINT], [0], [])
+    # SetBuilderOperator: semi-join source needs membership only — the dense
+    # orderkey range builds a bitmap instead of a 380M-row positional index
+    b = ops.set_builder(session, bridge, [ops.TG_BIGINT], 0)
     b.add_input(late)
     b.drain()
     # probe side: orders in the date window
