@@ -999,7 +999,7 @@ struct HashBuilderOp : tg_operator {
             tg_pool_free(s, d_rtotal);
             tg_pool_free(s, d_rbase);
         }
-        else {
+        else if (!bitmap_built) {
             hipLaunchKernelGGL(k_join_init, dim3(tg_grid_for(t.capacity)), dim3(TG_BLOCK),
                                0, s->stream, t.slots, t.capacity, t.links, total_rows);
             TG_HIP_CHECK(hipGetLastError());
