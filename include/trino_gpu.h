@@ -189,6 +189,13 @@ tg_status tg_hash_builder_create(tg_session*, tg_join_bridge*,
     const int32_t* output_channels, int32_t n_output_channels,
     tg_operator** out);
 
+/* join_type: 0 = inner, 1 = probe-outer (LEFT: unmatched probe rows emit
+ * one row with a NULL build side — LookupJoinOperators probe-outer) */
+tg_status tg_lookup_join_create_ex(tg_session*, tg_join_bridge*,
+    const int32_t* probe_types, int32_t n_probe_channels,
+    const int32_t* key_channels, int32_t n_key_channels,
+    const int32_t* probe_output_channels, int32_t n_probe_output,
+    int32_t join_type, tg_operator** out);
 tg_status tg_lookup_join_create(tg_session*, tg_join_bridge*,
     const int32_t* probe_types, int32_t n_probe_channels,
     const int32_t* key_channels, int32_t n_key_channels,

@@ -1069,3 +1069,47 @@ class TestSetBuilder:
         vi, vali = run(False)
         assert np.array_equal(vb, vi)
         assert np.array_equal(valb, vali)
+
+
+class TestProbeOuterJoin:
+    """Probe-outer (LEFT) lookup join: unmatched probe rows (including null
+    keys) emit one row with a NULL build side (LookupJoinOperators
+    probe-outer), in probe-row order."""
+
+    def test_left_join_with_nulls(self, sess, ops):
+        r = rng(130)
+        nb, m = 5000, 40_000
+        bk = r.choice(np.arange(10**6), nb, replace=False).astype(np.int64)
+        bv = r.integers(0, 10**6, nb).astype(np.int64)
+        pk = r.choice(np.concatenate([bk, r.integers(10**7, 10**8, 20_000)]),
+                      m).astype(np.int64)
+        p_nulls = r.random(m) < 0.03
+        pvalid = np.full((m + 63) // 64, ~np.uint64(0), np.uint64)
+        for i in np.nonzero(p_nulls)[0]:
+            pvalid[i >> 6] &= ~np.uint64(1 << (i & 63))
+
+        bridge = ops.JoinBridge(sess)
+        b = ops.hash_builder(sess, bridge, [ops.TG_BIGINT, ops.TG_BIGINT], [0], [1])
+        b.add_input(ops.page_from_numpy([bk, bv]))
+        b.drain()
+        j = ops.lookup_join(sess, bridge, [ops.TG_BIGINT], [0], [0], join_type=1)
+        j.add_input(ops.page_from_numpy([pk], valids=[pvalid]))
+        jp, _ = j.get_output()
+        b.close()
+        j.close()
+        bridge.close()
+
+        bmap = dict(zip(bk.tolist(), bv.tolist()))
+        got_k = jp[0]["values"]
+        got_v = jp[1]["values"]
+        vvalid = np.asarray(jp[1]["valid"])
+        assert len(got_k) == m                      # every probe row exactly once
+        for i in range(m):
+            bit = (int(vvalid[i >> 6]) >> (i & 63)) & 1
+            if p_nulls[i] or int(pk[i]) not in bmap:
+                assert bit == 0                     # NULL build side
+            else:
+                assert bit == 1 and int(got_v[i]) == bmap[int(pk[i])]
+        # probe keys pass through in order (null rows carry their slot)
+        keep = ~p_nulls
+        assert np.array_equal(np.asarray(got_k)[keep], pk[keep])

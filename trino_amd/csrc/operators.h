@@ -102,7 +102,7 @@ tg_status run_project(tg_session* s, const ExprProgram& proj, tg_type out_type,
                       const DevPage& page, const int32_t* d_positions, int32_t count,
                       DevBlock* out);
 tg_status run_gather(tg_session* s, const DevBlock& src, const int32_t* d_positions,
-                     int32_t count, DevBlock* out);
+                     int32_t count, DevBlock* out, bool null_positions = false);
 tg_status run_hash_rows(tg_session* s, const DevPage& page,
                         const int32_t* channels, int32_t n_channels,
                         uint64_t* d_hashes);

@@ -628,6 +628,12 @@ __global__ void k_gather(const T* __restrict__ src, const uint64_t* __restrict__
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; k < n; k += stride) {
         int64_t i = positions ? positions[k] : k;
+        if (i < 0) {   /* probe-outer null position */
+            out[k] = T();
+            if (out_valid)
+                atomicAnd((unsigned long long*)&out_valid[k >> 6], ~(1ull << (k & 63)));
+            continue;
+        }
         out[k] = src[i];
         if (out_valid && src_valid && !((src_valid[i >> 6] >> (i & 63)) & 1))
             atomicAnd((unsigned long long*)&out_valid[k >> 6], ~(1ull << (k & 63)));
@@ -647,7 +653,8 @@ __global__ void k_gather_var_lens(const int32_t* __restrict__ src_off,
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < n; i += stride) {
         int32_t p = pos[i];
-        bool isnull = src_valid && !((src_valid[p >> 6] >> (p & 63)) & 1);
+        bool isnull = p < 0 ||
+                      (src_valid && !((src_valid[p >> 6] >> (p & 63)) & 1));
         lens[i] = isnull ? 0 : src_off[p + 1] - src_off[p];
         if (isnull && out_valid)
             atomicAnd((unsigned long long*)&out_valid[i >> 6], ~(1ull << (i & 63)));
@@ -664,6 +671,7 @@ __global__ void k_gather_var_bytes(const uint8_t* __restrict__ src_bytes,
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < n; i += stride) {
         int32_t len = out_off[i + 1] - out_off[i];
+        if (pos[i] < 0) continue;   /* null: zero-length */
         const uint8_t* sp = src_bytes + src_off[pos[i]];
         uint8_t* dp = out_bytes + out_off[i];
         for (int32_t b = 0; b < len; b++) dp[b] = sp[b];
@@ -707,14 +715,14 @@ static tg_status run_gather_var(tg_session* s, const DevBlock& src,
 }
 
 tg_status run_gather(tg_session* s, const DevBlock& src, const int32_t* d_positions,
-                     int32_t count, DevBlock* out)
+                     int32_t count, DevBlock* out, bool null_positions)
 {
     if (src.type == TG_VARCHAR)
         return run_gather_var(s, src, d_positions, count, out);
     out->type = src.type;
     out->n = count;
     TG_POOL_ALLOC(s, &out->data, (int64_t)(count ? count : 1) * src.elem_size());
-    if (src.valid) {
+    if (src.valid || null_positions) {
         int64_t words = (count + 63) / 64;
         TG_POOL_ALLOC(s, &out->valid, (words ? words : 1) * 8);
         TG_HIP_CHECK(hipMemsetAsync(out->valid, 0xFF, words * 8, s->stream));

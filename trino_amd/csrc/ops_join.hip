@@ -494,7 +494,7 @@ __global__ void k_pb_probe(JoinTable t, int64_t jlo, int64_t jhi,
     }
 }
 
-__global__ void k_probe_count(JoinTable t, ProbeKeys p, int64_t m,
+__global__ void k_probe_count(JoinTable t, ProbeKeys p, int64_t m, int outer,
                               int32_t* __restrict__ counts,
                               int32_t* __restrict__ heads)
 {
@@ -553,6 +553,10 @@ __global__ void k_probe_count(JoinTable t, ProbeKeys p, int64_t m,
                 }
             }
         }
+        /* probe-outer (LEFT): unmatched rows (incl. null keys) emit one
+         * row with a null build side (LookupJoinOperator probeOnOuterSide,
+         * LookupJoinOperators.java) */
+        if (outer && cnt == 0) { cnt = 1; head = INT32_MIN; }
         counts[i] = cnt;
         heads[i] = head;
     }
@@ -574,6 +578,11 @@ __global__ void k_probe_fill(JoinTable t, ProbeKeys p, int64_t m,
         int64_t at = offsets[i];
         int32_t h = heads[i];
         if (h == -1) continue;
+        if (h == INT32_MIN) {    /* probe-outer miss: null build side */
+            out_probe[at] = (int32_t)i;
+            out_build[at] = -1;
+            continue;
+        }
         if (h <= -2) {           /* pre-resolved single match */
             out_probe[at] = (int32_t)i;
             out_build[at] = -2 - h;
@@ -1043,6 +1052,7 @@ struct LookupJoinOp : tg_operator {
     std::vector<int32_t> key_channels;
     std::vector<tg_type> probe_types;
     std::vector<int32_t> probe_output;
+    bool outer = false;             /* probe-outer (LEFT) */
 
     tg_status add_input(const tg_page* page) override
     {
@@ -1083,7 +1093,7 @@ struct LookupJoinOp : tg_operator {
         int64_t min_rows = emr ? atoll(emr) : (1 << 22);
         const char* emb = getenv("TG_JOIN_PART_MIN_BYTES");
         int64_t min_bytes = emb ? atoll(emb) : (64ll << 20);
-        if (use_part && t.csr && !t.generic && t.csr_keys &&
+        if (use_part && !outer && t.csr && !t.generic && t.csr_keys &&
             in.n >= min_rows && tbl_bytes > min_bytes) {
             /* partitioned single-pass probe + stable match sort */
             int64_t nparts = 1;
@@ -1172,7 +1182,8 @@ struct LookupJoinOp : tg_operator {
             TG_POOL_ALLOC(s, &d_offsets, (in.n ? in.n : 1) * 8);
             TG_POOL_ALLOC(s, &d_total, 8);
             hipLaunchKernelGGL(k_probe_count, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
-                               0, s->stream, t, pkc, in.n, d_counts, d_heads);
+                               0, s->stream, t, pkc, in.n, outer ? 1 : 0,
+                               d_counts, d_heads);
             TG_HIP_CHECK(hipGetLastError());
             tg_status sst = run_scan_counts(s, d_counts, in.n, d_offsets, d_total);
             if (sst != TG_OK) return sst;
@@ -1198,7 +1209,8 @@ struct LookupJoinOp : tg_operator {
         }
         for (int32_t ch : bridge->build_output_channels) {
             DevBlock ob;
-            st = run_gather(s, bridge->build_channels[ch], d_ob, (int32_t)total, &ob);
+            st = run_gather(s, bridge->build_channels[ch], d_ob, (int32_t)total, &ob,
+                            outer /* -1 positions become NULL */);
             if (st != TG_OK) return st;
             outp.blocks.push_back(ob);
         }
@@ -1429,6 +1441,12 @@ extern "C" tg_status tg_semi_join_create(tg_session* s, tg_join_bridge* bridge,
     return TG_OK;
 }
 
+extern "C" tg_status tg_lookup_join_create_ex(tg_session* s, tg_join_bridge* bridge,
+    const int32_t* probe_types, int32_t n_probe_channels,
+    const int32_t* key_channels, int32_t n_key_channels,
+    const int32_t* probe_output_channels, int32_t n_probe_output,
+    int32_t join_type /* 0 inner, 1 probe-outer */, tg_operator** out);
+
 extern "C" tg_status tg_lookup_join_create(tg_session* s, tg_join_bridge* bridge,
     const int32_t* probe_types, int32_t n_probe_channels,
     const int32_t* key_channels, int32_t n_key_channels,
@@ -1447,6 +1465,20 @@ extern "C" tg_status tg_lookup_join_create(tg_session* s, tg_join_bridge* bridge
         op->probe_types.push_back((tg_type)probe_types[i]);
     op->probe_output.assign(probe_output_channels, probe_output_channels + n_probe_output);
     *out = op;
+    return TG_OK;
+}
+
+extern "C" tg_status tg_lookup_join_create_ex(tg_session* s, tg_join_bridge* bridge,
+    const int32_t* probe_types, int32_t n_probe_channels,
+    const int32_t* key_channels, int32_t n_key_channels,
+    const int32_t* probe_output_channels, int32_t n_probe_output,
+    int32_t join_type, tg_operator** out)
+{
+    tg_status st = tg_lookup_join_create(s, bridge, probe_types, n_probe_channels,
+                                         key_channels, n_key_channels,
+                                         probe_output_channels, n_probe_output, out);
+    if (st != TG_OK) return st;
+    static_cast<LookupJoinOp*>(*out)->outer = (join_type == 1);
     return TG_OK;
 }
 

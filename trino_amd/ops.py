@@ -158,6 +158,8 @@ _lib.tg_hash_builder_create.restype = ctypes.c_int
 _lib.tg_hash_builder_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p] + [ctypes.c_void_p, ctypes.c_int32] * 3 + [ctypes.c_void_p]
 _lib.tg_lookup_join_create.restype = ctypes.c_int
 _lib.tg_lookup_join_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p] + [ctypes.c_void_p, ctypes.c_int32] * 3 + [ctypes.c_void_p]
+_lib.tg_lookup_join_create_ex.restype = ctypes.c_int
+_lib.tg_lookup_join_create_ex.argtypes = [ctypes.c_void_p, ctypes.c_void_p] + [ctypes.c_void_p, ctypes.c_int32] * 3 + [ctypes.c_int32, ctypes.c_void_p]
 _lib.tg_page_partitioner_create.restype = ctypes.c_int
 _lib.tg_page_partitioner_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                             ctypes.c_int32, ctypes.c_void_p,
@@ -328,14 +330,16 @@ def hash_builder(session, bridge, build_types, key_channels, output_channels):
     return op
 
 
-def lookup_join(session, bridge, probe_types, key_channels, probe_output_channels):
+def lookup_join(session, bridge, probe_types, key_channels, probe_output_channels,
+                join_type=0):
+    """join_type: 0 inner, 1 probe-outer (LEFT)."""
     h = ctypes.c_void_p()
     pt = _i32arr(probe_types)
     kc = _i32arr(key_channels)
     oc = _i32arr(probe_output_channels)
-    _check(_lib.tg_lookup_join_create(session._h, bridge._h, pt.ctypes.data, len(pt),
-                                      kc.ctypes.data, len(kc), oc.ctypes.data, len(oc),
-                                      ctypes.byref(h)))
+    _check(_lib.tg_lookup_join_create_ex(session._h, bridge._h, pt.ctypes.data, len(pt),
+                                         kc.ctypes.data, len(kc), oc.ctypes.data,
+                                         len(oc), join_type, ctypes.byref(h)))
     op = Operator(session, h)
     op._keep = (pt, kc, oc)
     return op
