@@ -219,3 +219,89 @@ def test_random_exact_sum_pipeline(sess, ops, seed):
     o = out[0]
     assert np.array_equal(o[0]["values"], by_gid)
     assert np.array_equal(o[1]["values"], exp)
+
+
+@pytest.mark.parametrize("seed", range(3))
+def test_random_outer_join_multipage(sess, ops, seed):
+    """Probe-outer join across multiple probe pages vs a dict composition."""
+    r = np.random.default_rng(500 + seed)
+    nb = int(r.integers(1, 8000))
+    bk = r.choice(np.arange(10**6), nb, replace=False).astype(np.int64)
+    bv = r.standard_normal(nb)
+    bridge = ops.JoinBridge(sess)
+    b = ops.hash_builder(sess, bridge, [ops.TG_BIGINT, ops.TG_DOUBLE], [0], [1])
+    b.add_input(ops.page_from_numpy([bk, bv]))
+    b.drain()
+    j = ops.lookup_join(sess, bridge, [ops.TG_BIGINT], [0], [0], join_type=1)
+    bmap = dict(zip(bk.tolist(), bv.tolist()))
+    outs = []
+    for pg in range(3):
+        m = int(r.integers(1, 20_000))
+        pk = r.choice(np.concatenate([bk, r.integers(10**7, 10**8, 5000)]),
+                      m).astype(np.int64)
+        j.add_input(ops.page_from_numpy([pk]))
+        jp, _ = j.get_output()
+        assert len(jp[0]["values"]) == m
+        vvalid = np.asarray(jp[1]["valid"])
+        for i in range(m):
+            bit = (int(vvalid[i >> 6]) >> (i & 63)) & 1
+            if int(pk[i]) in bmap:
+                assert bit == 1 and jp[1]["values"][i] == bmap[int(pk[i])]
+            else:
+                assert bit == 0
+        outs.append(jp)
+    j.close()
+    b.close()
+    bridge.close()
+
+
+@pytest.mark.parametrize("seed", range(3))
+def test_random_streaming_partial_final(sess, ops, seed):
+    """Streaming PARTIAL -> hash FINAL over clustered keys: the distributed
+    aggregation shape, exact sums bit-equal to the oracle."""
+    r = np.random.default_rng(600 + seed)
+    n = int(r.integers(1000, 80_000))
+    keys = np.sort(r.integers(0, n // 3 + 1, n)).astype(np.int64)
+    vals = r.integers(90100, 209900, n) / 100.0
+    part = ops.streaming_aggregation(sess, 0,
+                                     [(ops.AGG_COUNT_STAR, -1),
+                                      (ops.AGG_SUM_F64_EXACT, 1, 43)],
+                                     step=ops.STEP_PARTIAL)
+    part.add_input(ops.page_from_numpy([keys, vals]))
+    pout = part.drain()[0]
+    part.close()
+    fin = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                               [(ops.AGG_COUNT_STAR, 1),
+                                (ops.AGG_SUM_F64_EXACT, 2, 43)],
+                               step=ops.STEP_FINAL)
+    fin.add_input(ops.page_from_numpy([np.asarray(pout[0]["values"]),
+                                       np.asarray(pout[1]["values"]),
+                                       np.asarray(pout[2]["values"]),
+                                       np.asarray(pout[3]["values"])]))
+    out = fin.drain()[0]
+    fin.close()
+    gids, ng, by_gid, _ = oracle.bigint_groupby(keys)
+    exp = oracle.grouped_sum_f64_exact(gids, vals, ng, scale_pow=43)
+    assert np.array_equal(out[0]["values"], by_gid)
+    assert np.array_equal(out[1]["values"], oracle.grouped_count(gids, ng))
+    assert np.array_equal(out[2]["values"], exp)
+
+
+def test_set_builder_sparse_fallback(sess, ops):
+    """Key range beyond the bitmap bound falls back to the positional index
+    with identical semi results."""
+    r = np.random.default_rng(700)
+    bk = r.choice(np.arange(0, 2**40, 2**20), 5000, replace=False).astype(np.int64)
+    pk = np.concatenate([bk[:700], r.integers(0, 2**40, 9000)]).astype(np.int64)
+    bridge = ops.JoinBridge(sess)
+    b = ops.set_builder(sess, bridge, [ops.TG_BIGINT], 0)
+    b.add_input(ops.page_from_numpy([bk]))
+    b.drain()
+    sj = ops.semi_join(sess, bridge, 0)
+    sj.add_input(ops.page_from_numpy([pk]))
+    out, _ = sj.get_output()
+    b.close()
+    sj.close()
+    bridge.close()
+    exp = np.isin(pk, bk)
+    assert np.array_equal(out[-1]["values"].astype(bool), exp)
