@@ -74,6 +74,24 @@ def main():
                   "testcases/hive_tpch/q12.result",
         "rows": q12,
     }
+    # every hive_tpch answer (round-2 targets: the remaining 15 queries'
+    # streams get pinned against these, like q12/q18 were)
+    import glob
+    allq = {}
+    for fp in sorted(glob.glob(f"{REF}/testing/trino-product-tests/src/test/"
+                               "resources/sql-tests/testcases/hive_tpch/q*.result")):
+        name = fp.rsplit("/", 1)[1].split(".")[0]
+        qrows = []
+        for line in open(fp):
+            if line.startswith("--"):
+                continue
+            qrows.append([x for x in line.rstrip("\n").split("|")][:-1])
+        allq[name] = qrows
+    fixtures["all_answers_sf1"] = {
+        "source": "testing/trino-product-tests/src/test/resources/sql-tests/"
+                  "testcases/hive_tpch/qNN.result",
+        "rows": allq,
+    }
     with open(OUT, "w") as f:
         json.dump(fixtures, f, indent=1)
     print(f"wrote {OUT}: {len(rows)} canonical rows, q12 = {q12}")
