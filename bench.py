@@ -147,8 +147,9 @@ def main():
     log(f"generated {rows_rank:,} lineitem rows/rank (SF{total_sf:g} part {rank+1}/{n_gpus}) "
         f"in {time.time()-t0:.1f}s — {rows_rank*ALG_BYTES_PER_ROW/2**30:.1f} GiB in HBM")
 
+    coll_dev = "cuda" if (dist and dist.get_backend() == "nccl") else "cpu"
     if dist:
-        rows_t = torch.tensor([rows_rank], dtype=torch.int64, device="cuda")
+        rows_t = torch.tensor([rows_rank], dtype=torch.int64, device=coll_dev)
         dist.all_reduce(rows_t)
         rows_total = int(rows_t.item())
     else:
@@ -158,7 +159,7 @@ def main():
         r = sess.q1(cols, Q1_CUTOFF)
         if dist:
             # exchange leg (config 4 shape): partial->exchange->final agg
-            merged = exchange_partials(dist, world, rank, list(r.raw), "cuda")
+            merged = exchange_partials(dist, world, rank, list(r.raw), coll_dev)
             return r, merged
         return r, None
 
@@ -179,7 +180,7 @@ def main():
         dist.barrier()
     elapsed = time.time() - t_start
     if dist:
-        el = torch.tensor([elapsed], dtype=torch.float64, device="cuda")
+        el = torch.tensor([elapsed], dtype=torch.float64, device=coll_dev)
         dist.all_reduce(el, op=dist.ReduceOp.MAX)
         elapsed = float(el.item())
 
