@@ -34,7 +34,7 @@ _lib.tpch_gen_orders.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64
 _lib.tpch_gen_orders2.restype = ctypes.c_int64
 _lib.tpch_gen_orders2.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
 _lib.tpch_gen_customer.restype = ctypes.c_int64
-_lib.tpch_gen_customer.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 2
+_lib.tpch_gen_customer.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
 
 _lib.o_bigint_hash.restype = ctypes.c_uint64
 _lib.o_bigint_hash.argtypes = [ctypes.c_int64]
@@ -149,8 +149,12 @@ def gen_customer(sf, cust_start=1, cust_count=None):
         cust_count = int(150_000 * sf)
     ck = np.empty(cust_count, np.int64)
     ms = np.empty(cust_count, np.uint8)
-    _lib.tpch_gen_customer(sf, cust_start, cust_count, _ptr(ck), _ptr(ms))
-    return {"custkey": ck, "mktsegment": ms}
+    nk = np.empty(cust_count, np.uint8)
+    ab = np.empty(cust_count, np.int64)
+    _lib.tpch_gen_customer(sf, cust_start, cust_count, _ptr(ck), _ptr(ms),
+                           _ptr(nk), _ptr(ab))
+    return {"custkey": ck, "mktsegment": ms, "nationkey": nk,
+            "acctbal_cents": ab}
 
 
 def bigint_hash(v):

@@ -64,6 +64,11 @@
 #define TPCH_SEED_L_RDATE   373135028LL
 #define TPCH_SEED_L_RFLG    717419739LL
 #define TPCH_SEED_C_MSEG   1140279430LL
+/* pinned from the reference's Q10 SF1 answer fixture (20 rows expose
+ * (custkey -> nation, acctbal); the phone country code nationkey+10
+ * cross-checks every row) by exhaustive seed search — unique solutions */
+#define TPCH_SEED_C_NKEY   1489529863LL
+#define TPCH_SEED_C_ABAL    298370230LL
 #define TPCH_SEED_O_PRIO    591449447LL
 #define TPCH_SEED_P_TYPE   1841581359LL
 /* pinned from the reference's own fixtures: 785 canonical SF1 lineitem rows

@@ -96,17 +96,28 @@ EXPORT int64_t tpch_gen_orders2(double sf, int64_t order_start, int64_t order_co
 /* Customer mktsegment ids (Q3): id 0..4 in dists.dss order
  * AUTOMOBILE,BUILDING,FURNITURE,MACHINERY,HOUSEHOLD. custkey = dense index. */
 EXPORT int64_t tpch_gen_customer(double sf, int64_t cust_start, int64_t cust_count,
-    int64_t* custkey, uint8_t* mktsegment)
+    int64_t* custkey, uint8_t* mktsegment, uint8_t* nationkey,
+    int64_t* acctbal_cents)
 {
     (void)sf;
-    tpch_rng mseg;
+    tpch_rng mseg, nk, abal;
     tpch_rng_init(&mseg, TPCH_SEED_C_MSEG, 1);
+    tpch_rng_init(&nk,   TPCH_SEED_C_NKEY, 1);
+    tpch_rng_init(&abal, TPCH_SEED_C_ABAL, 1);
     tpch_rng_skip(&mseg, cust_start - 1);
+    tpch_rng_skip(&nk,   cust_start - 1);
+    tpch_rng_skip(&abal, cust_start - 1);
     for (int64_t i = 0; i < cust_count; i++) {
         if (custkey) custkey[i] = cust_start + i;
         int64_t pick = tpch_rng_int(&mseg, 1, 5);
         if (mktsegment) mktsegment[i] = (uint8_t)(pick - 1);
+        int64_t nkv = tpch_rng_int(&nk, 0, 24);
+        if (nationkey) nationkey[i] = (uint8_t)nkv;
+        int64_t bal = tpch_rng_int(&abal, -99999, 999999);
+        if (acctbal_cents) acctbal_cents[i] = bal;
         tpch_rng_row_finished(&mseg);
+        tpch_rng_row_finished(&nk);
+        tpch_rng_row_finished(&abal);
     }
     return cust_count;
 }

@@ -193,3 +193,37 @@ class TestShipmodeStream:
             highs = sum(1 for ok in oks.tolist() if pri[ok] <= 1)
             got[name] = (highs, int(sel.sum()) - highs)
         assert got == exp
+
+
+class TestCustomerStreams:
+    """C_NKEY (1489529863) and C_ABAL (298370230) pinned by the Q10 SF1
+    answer fixture's 20 (custkey -> nation, acctbal) rows; the phone country
+    code (nationkey+10) cross-checks every row. Q22's full answer verifies
+    both streams end-to-end (tests/test_gpu_q22.py on device; here the
+    oracle composition)."""
+
+    NATIONS = ["ALGERIA", "ARGENTINA", "BRAZIL", "CANADA", "EGYPT", "ETHIOPIA",
+               "FRANCE", "GERMANY", "INDIA", "INDONESIA", "IRAN", "IRAQ",
+               "JAPAN", "JORDAN", "KENYA", "MOROCCO", "MOZAMBIQUE", "PERU",
+               "CHINA", "ROMANIA", "SAUDI ARABIA", "VIETNAM", "RUSSIA",
+               "UNITED KINGDOM", "UNITED STATES"]
+
+    def test_q22_answer_exact(self):
+        import json, os
+        fx = json.load(open(os.path.join(os.path.dirname(__file__), "golden",
+                                         "ref_fixtures.json")))
+        c = oracle.gen_customer(1.0)
+        orders = oracle.gen_orders(1.0)
+        has_order = np.zeros(150_001, bool)
+        has_order[np.unique(orders["custkey"])] = True
+        cc = c["nationkey"].astype(int) + 10
+        codes = [13, 17, 18, 23, 29, 30, 31]
+        sel = np.isin(cc, codes)
+        pos = sel & (c["acctbal_cents"] > 0)
+        total, npos = int(c["acctbal_cents"][pos].sum()), int(pos.sum())
+        qual = sel & (c["acctbal_cents"] * npos > total) & ~has_order[c["custkey"]]
+        exp = fx["all_answers_sf1"]["rows"]["q22"]
+        for i, code in enumerate(codes):
+            m = qual & (cc == code)
+            assert int(m.sum()) == int(exp[i][1])
+            assert int(c["acctbal_cents"][m].sum()) == round(float(exp[i][2]) * 100)

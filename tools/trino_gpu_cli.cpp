@@ -132,7 +132,8 @@ static int run_q3(tg_session* s, double sf)
     die(tg_device_malloc(s, &o_ok, n_orders * 8), "malloc");
     die(tg_device_malloc(s, &o_ck, n_orders * 8), "malloc");
     die(tg_device_malloc(s, &o_od, n_orders * 4), "malloc");
-    die(tg_tpch_gen_customer(s, sf, 1, n_cust, (int64_t*)c_ck, (uint8_t*)c_ms), "gen cust");
+    die(tg_tpch_gen_customer(s, sf, 1, n_cust, (int64_t*)c_ck, (uint8_t*)c_ms,
+                             nullptr, nullptr), "gen cust");
     die(tg_tpch_gen_orders(s, sf, 1, n_orders, (int64_t*)o_ok, (int64_t*)o_ck, (int32_t*)o_od, nullptr), "gen ord");
     tg_tpch_lineitem_cols li;
     die(tg_tpch_lineitem_alloc(s, sf, 1, n_orders, 1, &li), "gen li");

@@ -114,22 +114,33 @@ __global__ void k_gen_part(int64_t part_start, int64_t part_count,
 }
 
 __global__ void k_gen_customer(double sf, int64_t cust_start, int64_t cust_count,
-                               int64_t* custkey, uint8_t* mktsegment)
+                               int64_t* custkey, uint8_t* mktsegment,
+                               uint8_t* nationkey, int64_t* acctbal_cents)
 {
     int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t n_groups = (cust_count + GOT - 1) / GOT;
     if (g >= n_groups) return;
     int64_t first = cust_start + g * GOT;
     int64_t cnt = min((int64_t)GOT, cust_start + cust_count - first);
-    tpch_rng mseg;
+    tpch_rng mseg, nk, abal;
     tpch_rng_init(&mseg, TPCH_SEED_C_MSEG, 1);
+    tpch_rng_init(&nk,   TPCH_SEED_C_NKEY, 1);
+    tpch_rng_init(&abal, TPCH_SEED_C_ABAL, 1);
     tpch_rng_skip(&mseg, first - 1);
+    tpch_rng_skip(&nk,   first - 1);
+    tpch_rng_skip(&abal, first - 1);
     for (int64_t i = 0; i < cnt; i++) {
         int64_t at = first - cust_start + i;
         if (custkey) custkey[at] = first + i;
         int64_t pick = tpch_rng_int(&mseg, 1, 5);
         if (mktsegment) mktsegment[at] = (uint8_t)(pick - 1);
+        int64_t nkv = tpch_rng_int(&nk, 0, 24);
+        if (nationkey) nationkey[at] = (uint8_t)nkv;
+        int64_t bal = tpch_rng_int(&abal, -99999, 999999);
+        if (acctbal_cents) acctbal_cents[at] = bal;
         tpch_rng_row_finished(&mseg);
+        tpch_rng_row_finished(&nk);
+        tpch_rng_row_finished(&abal);
     }
 }
 
@@ -256,12 +267,14 @@ extern "C" tg_status tg_tpch_gen_orders(tg_session* s, double sf,
 }
 
 extern "C" tg_status tg_tpch_gen_customer(tg_session* s, double sf,
-    int64_t cust_start, int64_t cust_count, int64_t* d_custkey, uint8_t* d_mktsegment)
+    int64_t cust_start, int64_t cust_count, int64_t* d_custkey, uint8_t* d_mktsegment,
+    uint8_t* d_nationkey, int64_t* d_acctbal_cents)
 {
     int64_t n_groups = (cust_count + GOT - 1) / GOT;
     int grid = (int)((n_groups + TG_BLOCK - 1) / TG_BLOCK);
     hipLaunchKernelGGL(k_gen_customer, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
-                       sf, cust_start, cust_count, d_custkey, d_mktsegment);
+                       sf, cust_start, cust_count, d_custkey, d_mktsegment,
+                       d_nationkey, d_acctbal_cents);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     return TG_OK;

@@ -41,7 +41,8 @@ def q3_prepare(session, sf, order_start=1, order_count=None, cust_start=1, cust_
     cust_ck = _device_buffer(session, cust_count * 8)
     cust_ms = _device_buffer(session, cust_count)
     _check_lib(_lib.tg_tpch_gen_customer(session._h, sf,
-                                         cust_start, cust_count, cust_ck, cust_ms))
+                                         cust_start, cust_count, cust_ck, cust_ms,
+                                         None, None))
     o_ok = _device_buffer(session, order_count * 8)
     o_ck = _device_buffer(session, order_count * 8)
     o_od = _device_buffer(session, order_count * 4)
@@ -547,3 +548,92 @@ def q18_gpu(session, sf, order_start=1, order_count=None, limit=100):
     names = [f"Customer#{int(c):09d}" for c in ckey]
     return dict(c_name=names, custkey=ckey, orderkey=okey, orderdate=odate,
                 totalprice_cents=totc, sum_qty=sumqty, elapsed=elapsed)
+
+
+Q22_CODES = [13, 17, 18, 23, 29, 30, 31]   # cntrycode = c_nationkey + 10
+
+
+def q22_gpu(session, sf, cust_count=None, order_count=None):
+    """TPC-H Q22 (global sales opportunity): customers in 7 country codes
+    with above-average positive balances and no orders. Country code =
+    substring(c_phone,1,2) = nationkey+10 (the phone generator prefixes the
+    nation; pinned via the Q10 fixture rows). Exact: balances are generated
+    and summed in CENTS (i64), the average threshold compares
+    cents*count > total (integer-exact in f64 below 2^53).
+    Pipeline: customer filter -> scalar sums (avg) -> orders set-builder
+    bitmap -> semi join -> filter (no order AND above avg) -> group by
+    nationkey."""
+    if cust_count is None:
+        cust_count = int(150_000 * sf)
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    c_ck = _device_buffer(session, cust_count * 8)
+    c_nk = _device_buffer(session, cust_count)
+    c_ab = _device_buffer(session, cust_count * 8)
+    _check_lib(_lib.tg_tpch_gen_customer(session._h, sf, 1, cust_count,
+                                         c_ck, None, c_nk, c_ab))
+    o_ck = _device_buffer(session, order_count * 8)
+    _check_lib(_lib.tg_tpch_gen_orders(session._h, sf, 1, order_count,
+                                       None, o_ck, None, None))
+    t0 = time.time()
+    cpage = ops.page_from_device(session, ([(c_ck.value, ops.TG_BIGINT),
+                                            (c_nk.value, ops.TG_TINYINT),
+                                            (c_ab.value, ops.TG_BIGINT)], cust_count))
+    in_chain = []
+    for i, code in enumerate(Q22_CODES):
+        in_chain += [("col", 1), ("i64", code - 10), "eq"]
+        if i:
+            in_chain.append("or")
+    # stage A: avg of positive balances in the code set
+    fa = ops.filter_project(session,
+                            ops.expr(*(in_chain + [("col", 2), ("i64", 0), "gt", "and"])),
+                            [ops.expr(("col", 2))], [ops.TG_BIGINT])
+    fa.add_input(cpage)
+    fa.finish()
+    posbal = _take_device_page(session, fa)
+    aa = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_I64, 0),
+                                               (ops.AGG_COUNT_STAR, -1)])
+    aa.add_input(posbal)
+    apages = aa.drain()
+    total_cents = int(apages[0][0]["values"][0])
+    npos = int(apages[0][1]["values"][0])
+    fa.close()
+    aa.close()
+    # stage B: orders custkey membership set
+    bridge = ops.JoinBridge(session)
+    b = ops.set_builder(session, bridge, [ops.TG_BIGINT], 0)
+    b.add_input(ops.page_from_device(session, ([(o_ck.value, ops.TG_BIGINT)],
+                                               order_count)))
+    b.drain()
+    sj = ops.semi_join(session, bridge, 0)
+    sj.add_input(cpage)
+    sj.finish()
+    marked = _take_device_page(session, sj)   # (ck, nk, cents, matched)
+    # no order AND in code set AND cents*npos > total (integer-exact in f64)
+    fb = ops.filter_project(session,
+                            ops.expr(*(in_chain +
+                                       [("col", 3), ("i64", 0), "eq", "and",
+                                        ("col", 2), ("i64", npos), "mul",
+                                        ("i64", total_cents), "gt", "and"])),
+                            [ops.expr(("col", 1)), ops.expr(("col", 2))],
+                            [ops.TG_TINYINT, ops.TG_BIGINT])
+    fb.add_input(marked)
+    fb.finish()
+    qual = _take_device_page(session, fb)
+    agg = ops.hash_aggregation(session, [0], [ops.TG_TINYINT],
+                               [(ops.AGG_COUNT_STAR, -1), (ops.AGG_SUM_I64, 1)])
+    agg.add_input(qual)
+    pages = agg.drain()
+    elapsed = time.time() - t0
+    for op in (b, sj, fb, agg):
+        op.close()
+    bridge.close()
+    for p in (c_ck, c_nk, c_ab, o_ck):
+        _device_free(session, p)
+    out = pages[0]
+    nk = np.asarray(out[0]["values"]).astype(np.int64)
+    cnt = np.asarray(out[1]["values"])
+    cents = np.asarray(out[2]["values"])
+    order = np.argsort(nk)
+    return dict(cntrycode=(nk + 10)[order], numcust=cnt[order],
+                totacctbal_cents=cents[order], elapsed=elapsed)
