@@ -69,6 +69,11 @@
  * cross-checks every row) by exhaustive seed search — unique solutions */
 #define TPCH_SEED_C_NKEY   1489529863LL
 #define TPCH_SEED_C_ABAL    298370230LL
+/* pinned the same way from the Q10 fixture's 20 phone numbers (3 draws per
+ * customer: RANDOM(100,999) x2 + RANDOM(1000,9999); country code is
+ * nationkey+10, not a draw). Not yet surfaced by a generator — no
+ * implemented query prints phones; recorded for round 2 (Q10/Q13). */
+#define TPCH_SEED_C_PHNE   1521138112LL
 #define TPCH_SEED_O_PRIO    591449447LL
 #define TPCH_SEED_P_TYPE   1841581359LL
 /* pinned from the reference's own fixtures: 785 canonical SF1 lineitem rows
