@@ -74,6 +74,12 @@
  * nationkey+10, not a draw). Not yet surfaced by a generator — no
  * implemented query prints phones; recorded for round 2 (Q10/Q13). */
 #define TPCH_SEED_C_PHNE   1521138112LL
+/* L_SINS (shipinstruct) from the same 785 canonical rows: seed 1371272478
+ * with entry order [DELIVER IN PERSON, COLLECT COD, TAKE BACK RETURN, NONE]
+ * fits all 785 (the mirror solution 776211169/reversed order fits too; a
+ * Q19-class answer disambiguates in round 2 — no implemented query reads
+ * shipinstruct yet). */
+#define TPCH_SEED_L_SINS   1371272478LL
 #define TPCH_SEED_O_PRIO    591449447LL
 #define TPCH_SEED_P_TYPE   1841581359LL
 /* pinned from the reference's own fixtures: 785 canonical SF1 lineitem rows
