@@ -242,6 +242,8 @@ typedef struct tg_tpch_lineitem_cols {
     int64_t* tp_cents;      /* optional (flags bit 4): this line's
                                o_totalprice contribution in cents (dbgen
                                mk_order truncation; sums to o_totalprice) */
+    int64_t* suppkey;       /* optional (flags bit 5): partsupp-bridge
+                               l_suppkey (canonical-row verified) */
 } tg_tpch_lineitem_cols;
 
 /* Generate lineitem rows for orders [order_start, order_start+order_count)

@@ -80,6 +80,10 @@
  * Q19-class answer disambiguates in round 2 — no implemented query reads
  * shipinstruct yet). */
 #define TPCH_SEED_L_SINS   1371272478LL
+/* l_suppkey = partsupp bridge over a 0..3 draw: supplier j of part p is
+ * (p + j*(S/4 + (p-1)/S)) %% S + 1 with S = 10,000*SF (dbgen PART_SUPP
+ * bridge); formula AND seed verified on all 785 canonical rows. */
+#define TPCH_SEED_L_SKEY   2095021727LL
 #define TPCH_SEED_O_PRIO    591449447LL
 #define TPCH_SEED_P_TYPE   1841581359LL
 /* pinned from the reference's own fixtures: 785 canonical SF1 lineitem rows
@@ -194,9 +198,10 @@ TPCH_HD static inline int64_t tpch_order_custkey(tpch_rng* ckey_rng, int64_t max
  * these streams, so skipping them is exact. */
 typedef struct {
     tpch_rng odate, lcnt, ckey, opri;
-    tpch_rng qty, dcnt, tax, pkey, ship, cdate, rdate, rflg, smode;
+    tpch_rng qty, dcnt, tax, pkey, ship, cdate, rdate, rflg, smode, skey;
     int64_t max_custkey;
     int64_t max_partkey;
+    int64_t n_suppliers;
 } tpch_order_streams;
 
 TPCH_HD static inline void tpch_order_streams_init(tpch_order_streams* s, double sf)
@@ -214,7 +219,9 @@ TPCH_HD static inline void tpch_order_streams_init(tpch_order_streams* s, double
     tpch_rng_init(&s->rdate, TPCH_SEED_L_RDATE, TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_init(&s->rflg,  TPCH_SEED_L_RFLG,  TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_init(&s->smode, TPCH_SEED_L_SMODE, TPCH_LINES_PER_ORDER_MAX);
+    tpch_rng_init(&s->skey,  TPCH_SEED_L_SKEY,  TPCH_LINES_PER_ORDER_MAX);
     s->max_custkey = (int64_t)(TPCH_CUSTOMER_BASE * sf);
+    s->n_suppliers = (int64_t)(10000 * sf);
     s->max_partkey = (int64_t)(TPCH_PART_BASE * sf);
 }
 
@@ -235,11 +242,13 @@ TPCH_HD static inline void tpch_order_streams_seek(tpch_order_streams* s, int64_
     tpch_rng_skip(&s->rdate, n * TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_skip(&s->rflg,  n * TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_skip(&s->smode, n * TPCH_LINES_PER_ORDER_MAX);
+    tpch_rng_skip(&s->skey,  n * TPCH_LINES_PER_ORDER_MAX);
 }
 
 typedef struct {
     int64_t orderkey;
     int64_t partkey;
+    int64_t suppkey;
     int32_t linenumber;     /* 1-based */
     int32_t shipdate;       /* epoch days */
     int32_t commitdate;
@@ -305,6 +314,10 @@ TPCH_HD static inline void tpch_gen_line(tpch_order_streams* s, const tpch_order
     }
     l->linestatus = (l->shipdate > TPCH_CURRENTDATE_EPOCH) ? 1 /*O*/ : 0 /*F*/;
     l->shipmode = (uint8_t)(tpch_rng_int(&s->smode, 1, 7) - 1);
+    int64_t sj = tpch_rng_int(&s->skey, 0, 3);
+    l->suppkey = (l->partkey +
+                  sj * (s->n_suppliers / 4 + (l->partkey - 1) / s->n_suppliers))
+                 % s->n_suppliers + 1;
     int64_t t = l->extprice_cents * (100 - l->discount_pct) / 100;
     l->tp_cents = t * (100 + l->tax_pct) / 100;
 }
@@ -325,6 +338,7 @@ TPCH_HD static inline void tpch_order_row_finished(tpch_order_streams* s)
     tpch_rng_row_finished(&s->rdate);
     tpch_rng_row_finished(&s->rflg);
     tpch_rng_row_finished(&s->smode);
+    tpch_rng_row_finished(&s->skey);
 }
 
 /* double views exactly as io.trino.tpch getDouble (cents/100.0 etc.) */

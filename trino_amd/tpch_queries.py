@@ -637,3 +637,51 @@ def q22_gpu(session, sf, cust_count=None, order_count=None):
     order = np.argsort(nk)
     return dict(cntrycode=(nk + 10)[order], numcust=cnt[order],
                 totacctbal_cents=cents[order], elapsed=elapsed)
+
+
+DATE_1996_01_01 = 9496
+DATE_1996_04_01 = 9587
+
+
+def q15_gpu(session, sf, order_start=1, order_count=None):
+    """TPC-H Q15 (top supplier): revenue view over a 3-month shipdate window
+    grouped by l_suppkey (partsupp-bridge stream, canonical-row verified),
+    suppliers at the maximum. Exact fixed-point revenue sums. The SF1 answer
+    (supplier 8449, 1772627.2087) must match the reference fixture; s_name
+    is the formatted key, address/phone are unpinned text columns and are
+    not produced."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    li = session.tpch_lineitem(sf, order_start, order_count, with_suppkey=True)
+    t0 = time.time()
+    lpage = ops.page_from_device(session, ([(li.suppkey, ops.TG_BIGINT),
+                                            (li.shipdate, ops.TG_INTEGER),
+                                            (li.extendedprice, ops.TG_DOUBLE),
+                                            (li.discount, ops.TG_DOUBLE)],
+                                           li.row_count))
+    f = ops.filter_project(session,
+                           ops.expr(("col", 1), ("i64", DATE_1996_01_01), "ge",
+                                    ("col", 1), ("i64", DATE_1996_04_01), "lt", "and"),
+                           [ops.expr(("col", 0)),
+                            ops.expr(("col", 2), ("f64", 1.0), ("col", 3), "sub", "mul")],
+                           [ops.TG_BIGINT, ops.TG_DOUBLE])
+    f.add_input(lpage)
+    f.finish()
+    sel = _take_device_page(session, f)
+    agg = ops.hash_aggregation(session, [0], [ops.TG_BIGINT],
+                               [(ops.AGG_SUM_F64_EXACT, 1, 43)])
+    agg.add_input(sel)
+    pages = agg.drain()
+    elapsed = time.time() - t0
+    f.close()
+    agg.close()
+    session.tpch_lineitem_free(li)
+    out = pages[0]
+    sk = np.asarray(out[0]["values"])
+    rev = np.asarray(out[1]["values"])
+    mx = rev.max()
+    pick = np.nonzero(rev == mx)[0]
+    order = pick[np.argsort(sk[pick])]
+    return dict(suppkey=sk[order],
+                s_name=[f"Supplier#{int(k):09d}" for k in sk[order]],
+                total_revenue=rev[order], elapsed=elapsed)
