@@ -357,6 +357,7 @@ def run_sweep(args, sess, n_gpus, rank):
                          ("q6", lambda: q.q6_gpu(sess, sf)),
                          ("q12", lambda: q.q12_gpu(sess, sf)),
                          ("q14", lambda: q.q14_gpu(sess, sf)),
+                         ("q15", lambda: q.q15_gpu(sess, sf)),
                          ("q18", lambda: q.q18_gpu(sess, sf)),
                          ("q22", lambda: q.q22_gpu(sess, sf))):
             r = fn()
@@ -380,12 +381,12 @@ def run_sweep(args, sess, n_gpus, rank):
     wall = time.time() - t0
     out = {
         "metric": "tpch_sweep_queries_per_hour",
-        "value": 8 * args.steps / wall * 3600, "unit": "queries/h",
+        "value": 9 * args.steps / wall * 3600, "unit": "queries/h",
         "n_gpus": n_gpus, "steps": args.steps, "warmup": args.warmup,
         "ms_per_step": wall * 1000 / args.steps, "higher_is_better": True,
         "scaling": "weak", "vs_baseline": None, "dtype": "f64",
         "data": "synthetic",
-        "config": {"workload": f"TPC-H 8-query sweep (Q1,Q3,Q4,Q6,Q12,Q14,Q18,Q22) "
+        "config": {"workload": f"TPC-H 9-query sweep (Q1,Q3,Q4,Q6,Q12,Q14,Q15,Q18,Q22) "
                                f"SF{sf:g} on {n_gpus}xMI355X; all reference-"
                                f"fixture-exact at SF1",
                    "per_query_ms": {k: round(v * 1000 / args.steps, 2)
