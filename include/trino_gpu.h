@@ -270,6 +270,9 @@ tg_status tg_tpch_gen_customer(tg_session*, double scale_factor,
     int64_t* dev_custkey, uint8_t* dev_mktsegment,
     uint8_t* dev_nationkey /* 0..24 or NULL */,
     int64_t* dev_acctbal_cents /* c_acctbal*100 or NULL */);
+tg_status tg_tpch_gen_supplier(tg_session*, double scale_factor,
+    int64_t supp_start, int64_t supp_count,
+    int64_t* dev_suppkey, uint8_t* dev_nationkey /* 0..24 */);
 tg_status tg_tpch_gen_part(tg_session*, double scale_factor,
     int64_t part_start, int64_t part_count,
     int64_t* dev_partkey, int16_t* dev_type_id /* 0..149 (SMALLINT: 150 ids

@@ -33,6 +33,8 @@ _lib.tpch_gen_orders.restype = ctypes.c_int64
 _lib.tpch_gen_orders.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 3
 _lib.tpch_gen_orders2.restype = ctypes.c_int64
 _lib.tpch_gen_orders2.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
+_lib.tpch_gen_supplier.restype = ctypes.c_int64
+_lib.tpch_gen_supplier.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 2
 _lib.tpch_gen_customer.restype = ctypes.c_int64
 _lib.tpch_gen_customer.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
 
@@ -142,6 +144,15 @@ def gen_orders(sf, order_start=1, order_count=None):
     pri = np.empty(order_count, np.uint8)
     _lib.tpch_gen_orders2(sf, order_start, order_count, _ptr(ok), _ptr(ck), _ptr(od), _ptr(pri))
     return {"orderkey": ok, "custkey": ck, "orderdate": od, "orderpriority": pri}
+
+
+def gen_supplier(sf, supp_start=1, supp_count=None):
+    if supp_count is None:
+        supp_count = int(10_000 * sf)
+    sk = np.empty(supp_count, np.int64)
+    nk = np.empty(supp_count, np.uint8)
+    _lib.tpch_gen_supplier(sf, supp_start, supp_count, _ptr(sk), _ptr(nk))
+    return {"suppkey": sk, "nationkey": nk}
 
 
 def gen_customer(sf, cust_start=1, cust_count=None):

@@ -84,6 +84,10 @@
  * (p + j*(S/4 + (p-1)/S)) %% S + 1 with S = 10,000*SF (dbgen PART_SUPP
  * bridge); formula AND seed verified on all 785 canonical rows. */
 #define TPCH_SEED_L_SKEY   2095021727LL
+/* pinned from 286 (suppkey -> nation) constraints in the reference's Q20
+ * (CANADA suppliers) and Q21 (SAUDI ARABIA suppliers) SF1 answer fixtures —
+ * unique solution; one draw per supplier row like the customer streams */
+#define TPCH_SEED_S_NKEY    110356601LL
 #define TPCH_SEED_O_PRIO    591449447LL
 #define TPCH_SEED_P_TYPE   1841581359LL
 /* pinned from the reference's own fixtures: 785 canonical SF1 lineitem rows

@@ -123,6 +123,22 @@ EXPORT int64_t tpch_gen_customer(double sf, int64_t cust_start, int64_t cust_cou
     return cust_count;
 }
 
+EXPORT int64_t tpch_gen_supplier(double sf, int64_t supp_start, int64_t supp_count,
+    int64_t* suppkey, uint8_t* nationkey)
+{
+    (void)sf;
+    tpch_rng nk;
+    tpch_rng_init(&nk, TPCH_SEED_S_NKEY, 1);
+    tpch_rng_skip(&nk, supp_start - 1);
+    for (int64_t i = 0; i < supp_count; i++) {
+        if (suppkey) suppkey[i] = supp_start + i;
+        int64_t v = tpch_rng_int(&nk, 0, 24);
+        if (nationkey) nationkey[i] = (uint8_t)v;
+        tpch_rng_row_finished(&nk);
+    }
+    return supp_count;
+}
+
 EXPORT int64_t tpch_gen_part_cols(double sf, int64_t part_start, int64_t part_count,
     int64_t* partkey, uint8_t* type_id)
 {

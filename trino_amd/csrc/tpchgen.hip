@@ -99,6 +99,26 @@ __global__ void k_gen_orders(double sf, int64_t order_start, int64_t order_count
     }
 }
 
+__global__ void k_gen_supplier(double sf, int64_t supp_start, int64_t supp_count,
+                               int64_t* suppkey, uint8_t* nationkey)
+{
+    int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t n_groups = (supp_count + GOT - 1) / GOT;
+    if (g >= n_groups) return;
+    int64_t first = supp_start + g * GOT;
+    int64_t cnt = min((int64_t)GOT, supp_start + supp_count - first);
+    tpch_rng nk;
+    tpch_rng_init(&nk, TPCH_SEED_S_NKEY, 1);
+    tpch_rng_skip(&nk, first - 1);
+    for (int64_t i = 0; i < cnt; i++) {
+        int64_t at = first - supp_start + i;
+        if (suppkey) suppkey[at] = first + i;
+        int64_t v = tpch_rng_int(&nk, 0, 24);
+        if (nationkey) nationkey[at] = (uint8_t)v;
+        tpch_rng_row_finished(&nk);
+    }
+}
+
 __global__ void k_gen_part(int64_t part_start, int64_t part_count,
                            int64_t* partkey, int16_t* type_id)
 {
@@ -278,6 +298,18 @@ extern "C" tg_status tg_tpch_gen_customer(tg_session* s, double sf,
     hipLaunchKernelGGL(k_gen_customer, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
                        sf, cust_start, cust_count, d_custkey, d_mktsegment,
                        d_nationkey, d_acctbal_cents);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}
+
+extern "C" tg_status tg_tpch_gen_supplier(tg_session* s, double sf,
+    int64_t supp_start, int64_t supp_count, int64_t* d_suppkey, uint8_t* d_nationkey)
+{
+    int64_t n_groups = (supp_count + GOT - 1) / GOT;
+    int grid = (int)((n_groups + TG_BLOCK - 1) / TG_BLOCK);
+    hipLaunchKernelGGL(k_gen_supplier, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                       sf, supp_start, supp_count, d_suppkey, d_nationkey);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     return TG_OK;

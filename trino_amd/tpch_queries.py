@@ -685,3 +685,128 @@ def q15_gpu(session, sf, order_start=1, order_count=None):
     return dict(suppkey=sk[order],
                 s_name=[f"Supplier#{int(k):09d}" for k in sk[order]],
                 total_revenue=rev[order], elapsed=elapsed)
+
+
+NATION_NAMES = ["ALGERIA", "ARGENTINA", "BRAZIL", "CANADA", "EGYPT", "ETHIOPIA",
+                "FRANCE", "GERMANY", "INDIA", "INDONESIA", "IRAN", "IRAQ",
+                "JAPAN", "JORDAN", "KENYA", "MOROCCO", "MOZAMBIQUE", "PERU",
+                "CHINA", "ROMANIA", "SAUDI ARABIA", "VIETNAM", "RUSSIA",
+                "UNITED KINGDOM", "UNITED STATES"]
+ASIA_NATIONS = [8, 9, 12, 18, 21]   # region ASIA per the fixed nation table
+
+
+def q5_gpu(session, sf, order_start=1, order_count=None):
+    """TPC-H Q5 (local supplier volume): ASIA-region revenue where the
+    customer and supplier share a nation; 1994 order window. Uses the pinned
+    customer/supplier nationkey streams and the partsupp-bridge l_suppkey;
+    the fixed 25-row nation/region tables are spec constants. SF1 must match
+    the reference fixture exactly (5 nations to 4 decimals)."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    cust_count = int(150_000 * sf)
+    supp_count = int(10_000 * sf)
+    c_ck = _device_buffer(session, cust_count * 8)
+    c_nk = _device_buffer(session, cust_count)
+    _check_lib(_lib.tg_tpch_gen_customer(session._h, sf, 1, cust_count,
+                                         c_ck, None, c_nk, None))
+    s_sk = _device_buffer(session, supp_count * 8)
+    s_nk = _device_buffer(session, supp_count)
+    _check_lib(_lib.tg_tpch_gen_supplier(session._h, sf, 1, supp_count, s_sk, s_nk))
+    o_ok = _device_buffer(session, order_count * 8)
+    o_ck = _device_buffer(session, order_count * 8)
+    o_od = _device_buffer(session, order_count * 4)
+    _check_lib(_lib.tg_tpch_gen_orders(session._h, sf, order_start, order_count,
+                                       o_ok, o_ck, o_od, None))
+    li = session.tpch_lineitem(sf, order_start, order_count,
+                               with_orderkey=True, with_suppkey=True)
+    t0 = time.time()
+    # build1: customers (custkey -> nationkey)
+    br1 = ops.JoinBridge(session)
+    b1 = ops.hash_builder(session, br1, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b1.add_input(ops.page_from_device(session, ([(c_ck.value, ops.TG_BIGINT),
+                                                 (c_nk.value, ops.TG_TINYINT)],
+                                                cust_count)))
+    b1.drain()
+    # orders in 1994 -> join customers
+    f1 = ops.filter_project(session,
+                            ops.expr(("col", 2), ("i64", DATE_1994_01_01), "ge",
+                                     ("col", 2), ("i64", DATE_1994_01_01 + 365),
+                                     "lt", "and"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                            [ops.TG_BIGINT, ops.TG_BIGINT])
+    f1.add_input(ops.page_from_device(session, ([(o_ok.value, ops.TG_BIGINT),
+                                                 (o_ck.value, ops.TG_BIGINT),
+                                                 (o_od.value, ops.TG_INTEGER)],
+                                                order_count)))
+    f1.finish()
+    owin = _take_device_page(session, f1)
+    j1 = ops.lookup_join(session, br1, [ops.TG_BIGINT, ops.TG_BIGINT], [1], [0])
+    j1.add_input(owin)
+    j1.finish()
+    ojoined = _take_device_page(session, j1)    # (orderkey, c_nationkey)
+    br2 = ops.JoinBridge(session)
+    b2 = ops.hash_builder(session, br2, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b2.add_input(ojoined)
+    b2.drain()
+    # lineitem -> (orderkey, suppkey, revenue) -> join orders
+    fp = ops.filter_project(session, None,
+                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                             ops.expr(("col", 2), ("f64", 1.0), ("col", 3),
+                                      "sub", "mul")],
+                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE])
+    fp.add_input(ops.page_from_device(session, ([(li.orderkey, ops.TG_BIGINT),
+                                                 (li.suppkey, ops.TG_BIGINT),
+                                                 (li.extendedprice, ops.TG_DOUBLE),
+                                                 (li.discount, ops.TG_DOUBLE)],
+                                                li.row_count)))
+    fp.finish()
+    lsel = _take_device_page(session, fp)
+    j2 = ops.lookup_join(session, br2, [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE],
+                         [0], [1, 2])
+    j2.add_input(lsel)
+    j2.finish()
+    lj = _take_device_page(session, j2)         # (suppkey, rev, c_nk)
+    # join supplier nation
+    br3 = ops.JoinBridge(session)
+    b3 = ops.hash_builder(session, br3, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b3.add_input(ops.page_from_device(session, ([(s_sk.value, ops.TG_BIGINT),
+                                                 (s_nk.value, ops.TG_TINYINT)],
+                                                supp_count)))
+    b3.drain()
+    j3 = ops.lookup_join(session, br3,
+                         [ops.TG_BIGINT, ops.TG_DOUBLE, ops.TG_TINYINT],
+                         [0], [1, 2])
+    j3.add_input(lj)
+    j3.finish()
+    final = _take_device_page(session, j3)      # (rev, c_nk, s_nk)
+    in_chain = []
+    for i, nk in enumerate(ASIA_NATIONS):
+        in_chain += [("col", 1), ("i64", nk), "eq"]
+        if i:
+            in_chain.append("or")
+    f2 = ops.filter_project(session,
+                            ops.expr(*(in_chain +
+                                       [("col", 1), ("col", 2), "eq", "and"])),
+                            [ops.expr(("col", 1)), ops.expr(("col", 0))],
+                            [ops.TG_TINYINT, ops.TG_DOUBLE])
+    f2.add_input(final)
+    f2.finish()
+    qual = _take_device_page(session, f2)
+    agg = ops.hash_aggregation(session, [0], [ops.TG_TINYINT],
+                               [(ops.AGG_SUM_F64_EXACT, 1, 43)])
+    agg.add_input(qual)
+    pages = agg.drain()
+    elapsed = time.time() - t0
+    for op in (b1, f1, j1, b2, fp, j2, b3, j3, f2, agg):
+        op.close()
+    for br in (br1, br2, br3):
+        br.close()
+    session.tpch_lineitem_free(li)
+    for p in (c_ck, c_nk, s_sk, s_nk, o_ok, o_ck, o_od):
+        _device_free(session, p)
+    out = pages[0]
+    nk = np.asarray(out[0]["values"]).astype(np.int64)
+    rev = np.asarray(out[1]["values"])
+    order = np.argsort(-rev)
+    return dict(n_name=[NATION_NAMES[k] for k in nk[order]],
+                revenue=rev[order], elapsed=elapsed)
