@@ -84,6 +84,9 @@ _lib.tg_tpch_gen_orders.argtypes = [ctypes.c_void_p, ctypes.c_double, ctypes.c_i
                                     ctypes.c_int64] + [ctypes.c_void_p] * 4
 _lib.tg_tpch_gen_customer.restype = ctypes.c_int
 _lib.tg_tpch_gen_customer.argtypes = [ctypes.c_void_p, ctypes.c_double, ctypes.c_int64,
+                                      ctypes.c_int64] + [ctypes.c_void_p] * 4
+_lib.tg_tpch_gen_supplier.restype = ctypes.c_int
+_lib.tg_tpch_gen_supplier.argtypes = [ctypes.c_void_p, ctypes.c_double, ctypes.c_int64,
                                       ctypes.c_int64] + [ctypes.c_void_p] * 2
 
 
