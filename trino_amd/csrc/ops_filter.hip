@@ -757,7 +757,7 @@ __global__ void k_proj_mulsub(const double* __restrict__ a, const double* __rest
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < n; i += stride) {
-        int32_t p = pos[i];
+        int64_t p = pos ? (int64_t)pos[i] : i;   /* null = identity (no filter) */
         out[i] = a[p] * (c - b[p]);
     }
 }
