@@ -810,3 +810,114 @@ def q5_gpu(session, sf, order_start=1, order_count=None):
     order = np.argsort(-rev)
     return dict(n_name=[NATION_NAMES[k] for k in nk[order]],
                 revenue=rev[order], elapsed=elapsed)
+
+
+def q7_gpu(session, sf, order_start=1, order_count=None):
+    """TPC-H Q7 (volume shipping): FRANCE<->GERMANY revenue by ship year
+    (1995-1996). Same pinned streams as Q5; l_year derived from shipdate in
+    the projection. SF1 must match the reference fixture exactly."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    cust_count = int(150_000 * sf)
+    supp_count = int(10_000 * sf)
+    c_ck = _device_buffer(session, cust_count * 8)
+    c_nk = _device_buffer(session, cust_count)
+    _check_lib(_lib.tg_tpch_gen_customer(session._h, sf, 1, cust_count,
+                                         c_ck, None, c_nk, None))
+    s_sk = _device_buffer(session, supp_count * 8)
+    s_nk = _device_buffer(session, supp_count)
+    _check_lib(_lib.tg_tpch_gen_supplier(session._h, sf, 1, supp_count, s_sk, s_nk))
+    o_ok = _device_buffer(session, order_count * 8)
+    o_ck = _device_buffer(session, order_count * 8)
+    _check_lib(_lib.tg_tpch_gen_orders(session._h, sf, order_start, order_count,
+                                       o_ok, o_ck, None, None))
+    li = session.tpch_lineitem(sf, order_start, order_count,
+                               with_orderkey=True, with_suppkey=True)
+    t0 = time.time()
+    br1 = ops.JoinBridge(session)
+    b1 = ops.hash_builder(session, br1, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b1.add_input(ops.page_from_device(session, ([(c_ck.value, ops.TG_BIGINT),
+                                                 (c_nk.value, ops.TG_TINYINT)],
+                                                cust_count)))
+    b1.drain()
+    j1 = ops.lookup_join(session, br1, [ops.TG_BIGINT, ops.TG_BIGINT], [1], [0])
+    j1.add_input(ops.page_from_device(session, ([(o_ok.value, ops.TG_BIGINT),
+                                                 (o_ck.value, ops.TG_BIGINT)],
+                                                order_count)))
+    j1.finish()
+    ojoined = _take_device_page(session, j1)
+    br2 = ops.JoinBridge(session)
+    b2 = ops.hash_builder(session, br2, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b2.add_input(ojoined)
+    b2.drain()
+    f = ops.filter_project(session,
+                           ops.expr(("col", 4), ("i64", DATE_1995_01_01), "ge",
+                                    ("col", 4), ("i64", DATE_1995_01_01 + 731),
+                                    "lt", "and"),
+                           [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                            ops.expr(("col", 2), ("f64", 1.0), ("col", 3),
+                                     "sub", "mul"),
+                            ops.expr(("col", 4), ("i64", DATE_1996_01_01), "ge")],
+                           [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE,
+                            ops.TG_DOUBLE])
+    f.add_input(ops.page_from_device(session, ([(li.orderkey, ops.TG_BIGINT),
+                                                (li.suppkey, ops.TG_BIGINT),
+                                                (li.extendedprice, ops.TG_DOUBLE),
+                                                (li.discount, ops.TG_DOUBLE),
+                                                (li.shipdate, ops.TG_INTEGER)],
+                                               li.row_count)))
+    f.finish()
+    lsel = _take_device_page(session, f)   # (ok, sk, rev, yearflag)
+    j2 = ops.lookup_join(session, br2,
+                         [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE, ops.TG_DOUBLE],
+                         [0], [1, 2, 3])
+    j2.add_input(lsel)
+    j2.finish()
+    lj = _take_device_page(session, j2)    # (sk, rev, yf, c_nk)
+    br3 = ops.JoinBridge(session)
+    b3 = ops.hash_builder(session, br3, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b3.add_input(ops.page_from_device(session, ([(s_sk.value, ops.TG_BIGINT),
+                                                 (s_nk.value, ops.TG_TINYINT)],
+                                                supp_count)))
+    b3.drain()
+    j3 = ops.lookup_join(session, br3,
+                         [ops.TG_BIGINT, ops.TG_DOUBLE, ops.TG_DOUBLE, ops.TG_TINYINT],
+                         [0], [1, 2, 3])
+    j3.add_input(lj)
+    j3.finish()
+    final = _take_device_page(session, j3)  # (rev, yf, c_nk, s_nk)
+    f2 = ops.filter_project(session,
+                            ops.expr(("col", 3), ("i64", 6), "eq",
+                                     ("col", 2), ("i64", 7), "eq", "and",
+                                     ("col", 3), ("i64", 7), "eq",
+                                     ("col", 2), ("i64", 6), "eq", "and", "or"),
+                            [ops.expr(("col", 3)), ops.expr(("col", 2)),
+                             ops.expr(("col", 1)), ops.expr(("col", 0))],
+                            [ops.TG_TINYINT, ops.TG_TINYINT, ops.TG_DOUBLE,
+                             ops.TG_DOUBLE])
+    f2.add_input(final)
+    f2.finish()
+    qual = _take_device_page(session, f2)   # (s_nk, c_nk, yf, rev)
+    agg = ops.hash_aggregation(session, [0, 1, 2],
+                               [ops.TG_TINYINT, ops.TG_TINYINT, ops.TG_DOUBLE],
+                               [(ops.AGG_SUM_F64_EXACT, 3, 43)])
+    agg.add_input(qual)
+    pages = agg.drain()
+    elapsed = time.time() - t0
+    for op in (b1, j1, b2, f, j2, b3, j3, f2, agg):
+        op.close()
+    for br in (br1, br2, br3):
+        br.close()
+    session.tpch_lineitem_free(li)
+    for p in (c_ck, c_nk, s_sk, s_nk, o_ok, o_ck):
+        _device_free(session, p)
+    out = pages[0]
+    sn = np.asarray(out[0]["values"]).astype(np.int64)
+    cn = np.asarray(out[1]["values"]).astype(np.int64)
+    yf = np.asarray(out[2]["values"])
+    rev = np.asarray(out[3]["values"])
+    year = np.where(yf > 0.5, 1996, 1995)
+    order = np.lexsort((year, cn, sn))
+    return dict(supp_nation=[NATION_NAMES[k] for k in sn[order]],
+                cust_nation=[NATION_NAMES[k] for k in cn[order]],
+                l_year=year[order], revenue=rev[order], elapsed=elapsed)
