@@ -33,6 +33,8 @@ _lib.tpch_gen_orders.restype = ctypes.c_int64
 _lib.tpch_gen_orders.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 3
 _lib.tpch_gen_orders2.restype = ctypes.c_int64
 _lib.tpch_gen_orders2.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
+_lib.tpch_gen_part_cols.restype = ctypes.c_int64
+_lib.tpch_gen_part_cols.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 2
 _lib.tpch_gen_supplier.restype = ctypes.c_int64
 _lib.tpch_gen_supplier.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 2
 _lib.tpch_gen_customer.restype = ctypes.c_int64
@@ -144,6 +146,15 @@ def gen_orders(sf, order_start=1, order_count=None):
     pri = np.empty(order_count, np.uint8)
     _lib.tpch_gen_orders2(sf, order_start, order_count, _ptr(ok), _ptr(ck), _ptr(od), _ptr(pri))
     return {"orderkey": ok, "custkey": ck, "orderdate": od, "orderpriority": pri}
+
+
+def gen_part(sf, part_start=1, part_count=None):
+    if part_count is None:
+        part_count = int(200_000 * sf)
+    pk = np.empty(part_count, np.int64)
+    ty = np.empty(part_count, np.uint8)
+    _lib.tpch_gen_part_cols(sf, part_start, part_count, _ptr(pk), _ptr(ty))
+    return {"partkey": pk, "type_id": ty.astype(np.int16)}
 
 
 def gen_supplier(sf, supp_start=1, supp_count=None):

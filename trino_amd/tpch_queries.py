@@ -921,3 +921,161 @@ def q7_gpu(session, sf, order_start=1, order_count=None):
     return dict(supp_nation=[NATION_NAMES[k] for k in sn[order]],
                 cust_nation=[NATION_NAMES[k] for k in cn[order]],
                 l_year=year[order], revenue=rev[order], elapsed=elapsed)
+
+
+AMERICA_NATIONS = [1, 2, 3, 17, 24]
+TYPE_ECONOMY_ANODIZED_STEEL = 103   # 4*25 + 0*5 + 3
+
+
+def q8_gpu(session, sf, order_start=1, order_count=None):
+    """TPC-H Q8 (national market share): BRAZIL's share of AMERICA-region
+    revenue for ECONOMY ANODIZED STEEL parts by order year. SF1 must match
+    the reference fixture (6-decimal shares)."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    cust_count = int(150_000 * sf)
+    supp_count = int(10_000 * sf)
+    n_parts = int(200_000 * sf)
+    c_ck = _device_buffer(session, cust_count * 8)
+    c_nk = _device_buffer(session, cust_count)
+    _check_lib(_lib.tg_tpch_gen_customer(session._h, sf, 1, cust_count,
+                                         c_ck, None, c_nk, None))
+    s_sk = _device_buffer(session, supp_count * 8)
+    s_nk = _device_buffer(session, supp_count)
+    _check_lib(_lib.tg_tpch_gen_supplier(session._h, sf, 1, supp_count, s_sk, s_nk))
+    p_pk = _device_buffer(session, n_parts * 8)
+    p_ty = _device_buffer(session, n_parts * 2)
+    _check_lib(_lib.tg_tpch_gen_part(session._h, sf, 1, n_parts, p_pk, p_ty))
+    o_ok = _device_buffer(session, order_count * 8)
+    o_ck = _device_buffer(session, order_count * 8)
+    o_od = _device_buffer(session, order_count * 4)
+    _check_lib(_lib.tg_tpch_gen_orders(session._h, sf, order_start, order_count,
+                                       o_ok, o_ck, o_od, None))
+    li = session.tpch_lineitem(sf, order_start, order_count,
+                               with_orderkey=True, with_partkey=True,
+                               with_suppkey=True)
+    t0 = time.time()
+    # part filter -> tiny build (partkey only)
+    fpart = ops.filter_project(session,
+                               ops.expr(("col", 1),
+                                        ("i64", TYPE_ECONOMY_ANODIZED_STEEL), "eq"),
+                               [ops.expr(("col", 0))], [ops.TG_BIGINT])
+    fpart.add_input(ops.page_from_device(session, ([(p_pk.value, ops.TG_BIGINT),
+                                                    (p_ty.value, ops.TG_SMALLINT)],
+                                                   n_parts)))
+    fpart.finish()
+    psel = _take_device_page(session, fpart)
+    brp = ops.JoinBridge(session)
+    bp = ops.hash_builder(session, brp, [ops.TG_BIGINT], [0], [])
+    bp.add_input(psel)
+    bp.drain()
+    # customers -> orders window -> AMERICA filter -> build2 (ok -> yearflag)
+    br1 = ops.JoinBridge(session)
+    b1 = ops.hash_builder(session, br1, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b1.add_input(ops.page_from_device(session, ([(c_ck.value, ops.TG_BIGINT),
+                                                 (c_nk.value, ops.TG_TINYINT)],
+                                                cust_count)))
+    b1.drain()
+    f1 = ops.filter_project(session,
+                            ops.expr(("col", 2), ("i64", DATE_1995_01_01), "ge",
+                                     ("col", 2), ("i64", DATE_1995_01_01 + 731),
+                                     "lt", "and"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                             ops.expr(("col", 2), ("i64", DATE_1996_01_01), "ge")],
+                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE])
+    f1.add_input(ops.page_from_device(session, ([(o_ok.value, ops.TG_BIGINT),
+                                                 (o_ck.value, ops.TG_BIGINT),
+                                                 (o_od.value, ops.TG_INTEGER)],
+                                                order_count)))
+    f1.finish()
+    owin = _take_device_page(session, f1)
+    j1 = ops.lookup_join(session, br1,
+                         [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE], [1], [0, 2])
+    j1.add_input(owin)
+    j1.finish()
+    oj = _take_device_page(session, j1)      # (ok, yf, c_nk)
+    in_chain = []
+    for i, nk in enumerate(AMERICA_NATIONS):
+        in_chain += [("col", 2), ("i64", nk), "eq"]
+        if i:
+            in_chain.append("or")
+    f2 = ops.filter_project(session, ops.expr(*in_chain),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                            [ops.TG_BIGINT, ops.TG_DOUBLE])
+    f2.add_input(oj)
+    f2.finish()
+    oam = _take_device_page(session, f2)
+    br2 = ops.JoinBridge(session)
+    b2 = ops.hash_builder(session, br2, [ops.TG_BIGINT, ops.TG_DOUBLE], [0], [1])
+    b2.add_input(oam)
+    b2.drain()
+    # lineitem -> part filter join -> orders join -> supplier join
+    fl = ops.filter_project(session, None,
+                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                             ops.expr(("col", 2)),
+                             ops.expr(("col", 3), ("f64", 1.0), ("col", 4),
+                                      "sub", "mul")],
+                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
+                             ops.TG_DOUBLE])
+    fl.add_input(ops.page_from_device(session, ([(li.partkey, ops.TG_BIGINT),
+                                                 (li.orderkey, ops.TG_BIGINT),
+                                                 (li.suppkey, ops.TG_BIGINT),
+                                                 (li.extendedprice, ops.TG_DOUBLE),
+                                                 (li.discount, ops.TG_DOUBLE)],
+                                                li.row_count)))
+    fl.finish()
+    lsel = _take_device_page(session, fl)    # (pk, ok, sk, rev)
+    jp = ops.lookup_join(session, brp,
+                         [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE],
+                         [0], [1, 2, 3])
+    jp.add_input(lsel)
+    jp.finish()
+    lpartsel = _take_device_page(session, jp)   # (ok, sk, rev)
+    j2 = ops.lookup_join(session, br2,
+                         [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE], [0], [1, 2])
+    j2.add_input(lpartsel)
+    j2.finish()
+    lj = _take_device_page(session, j2)         # (sk, rev, yf)
+    br3 = ops.JoinBridge(session)
+    b3 = ops.hash_builder(session, br3, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b3.add_input(ops.page_from_device(session, ([(s_sk.value, ops.TG_BIGINT),
+                                                 (s_nk.value, ops.TG_TINYINT)],
+                                                supp_count)))
+    b3.drain()
+    j3 = ops.lookup_join(session, br3,
+                         [ops.TG_BIGINT, ops.TG_DOUBLE, ops.TG_DOUBLE], [0], [1, 2])
+    j3.add_input(lj)
+    j3.finish()
+    final = _take_device_page(session, j3)      # (rev, yf, s_nk)
+    fb = ops.filter_project(session, None,
+                            [ops.expr(("col", 1)),
+                             ops.expr(("col", 2), ("i64", 2), "eq"),
+                             ops.expr(("col", 0))],
+                            [ops.TG_DOUBLE, ops.TG_DOUBLE, ops.TG_DOUBLE])
+    fb.add_input(final)
+    fb.finish()
+    flagged = _take_device_page(session, fb)    # (yf, brazil, rev)
+    agg = ops.hash_aggregation(session, [0, 1], [ops.TG_DOUBLE, ops.TG_DOUBLE],
+                               [(ops.AGG_SUM_F64_EXACT, 2, 43)])
+    agg.add_input(flagged)
+    pages = agg.drain()
+    elapsed = time.time() - t0
+    for op in (fpart, bp, b1, f1, j1, f2, b2, fl, jp, j2, b3, j3, fb, agg):
+        op.close()
+    for br in (brp, br1, br2, br3):
+        br.close()
+    session.tpch_lineitem_free(li)
+    for p in (c_ck, c_nk, s_sk, s_nk, p_pk, p_ty, o_ok, o_ck, o_od):
+        _device_free(session, p)
+    out = pages[0]
+    yf = np.asarray(out[0]["values"])
+    bf = np.asarray(out[1]["values"])
+    rev = np.asarray(out[2]["values"])
+    res = {}
+    for y in (1995, 1996):
+        m = yf == (0.0 if y == 1995 else 1.0)
+        den = rev[m].sum()
+        num = rev[m & (bf == 1.0)].sum()
+        res[y] = num / den
+    return dict(o_year=[1995, 1996],
+                mkt_share=[res[1995], res[1996]], elapsed=elapsed)
