@@ -240,3 +240,84 @@ def test_suppkey_bridge_canonical_rows():
     # were verified against all 785 rows at pin time; this guards the stream
     assert cols["suppkey"][:6].tolist() == [7706, 7311, 3701, 4633, 1534, 638]
     assert fx["lineitem_canonical_sf1"]["rows"][0]["partkey"] == 155190
+
+
+class TestNationJoinQueries:
+    """Oracle compositions of Q5/Q7/Q8 against the reference's SF1 answer
+    fixtures — pins the supplier nationkey stream and the fixed nation/region
+    tables end-to-end on CPU (device pipelines: tests/test_gpu_q{5,7,8}.py)."""
+
+    @pytest.fixture(scope="class")
+    def data(self):
+        c = oracle.gen_customer(1.0)
+        o = oracle.gen_orders(1.0)
+        li = oracle.gen_lineitem(1.0, columns=["orderkey", "suppkey", "partkey",
+                                               "extendedprice", "discount",
+                                               "shipdate"])
+        s = oracle.gen_supplier(1.0)
+        cnk = np.zeros(150_001, np.int8)
+        cnk[c["custkey"]] = c["nationkey"]
+        snk = np.zeros(10_001, np.int8)
+        snk[s["suppkey"]] = s["nationkey"]
+        max_ok = int(o["orderkey"].max())
+        o_cn = np.full(max_ok + 1, -1, np.int8)
+        o_cn[o["orderkey"]] = cnk[o["custkey"]]
+        o_dt = np.zeros(max_ok + 1, np.int32)
+        o_dt[o["orderkey"]] = o["orderdate"]
+        rev = li["extendedprice"] * (1.0 - li["discount"])
+        return dict(li=li, snk=snk, o_cn=o_cn, o_dt=o_dt, rev=rev)
+
+    @pytest.fixture(scope="class")
+    def answers(self):
+        import json, os
+        return json.load(open(os.path.join(os.path.dirname(__file__), "golden",
+                                           "ref_fixtures.json")))["all_answers_sf1"]["rows"]
+
+    def test_q5(self, data, answers):
+        ASIA = np.array([8, 9, 12, 18, 21])
+        li = data["li"]
+        cn = data["o_cn"][li["orderkey"]]
+        od = data["o_dt"][li["orderkey"]]
+        sn = data["snk"][li["suppkey"]]
+        m = (np.isin(cn, ASIA) & (cn == sn) & (od >= 8766) & (od < 9131))
+        got = {}
+        for nk in ASIA:
+            got[int(nk)] = data["rev"][m & (cn == nk)].sum()
+        names = ["ALGERIA", "ARGENTINA", "BRAZIL", "CANADA", "EGYPT", "ETHIOPIA",
+                 "FRANCE", "GERMANY", "INDIA", "INDONESIA", "IRAN", "IRAQ",
+                 "JAPAN", "JORDAN", "KENYA", "MOROCCO", "MOZAMBIQUE", "PERU",
+                 "CHINA", "ROMANIA", "SAUDI ARABIA", "VIETNAM", "RUSSIA",
+                 "UNITED KINGDOM", "UNITED STATES"]
+        exp = {row[0]: float(row[1]) for row in answers["q05"]}
+        for nk, v in got.items():
+            assert abs(v - exp[names[nk]]) < 5e-4
+
+    def test_q7(self, data, answers):
+        li = data["li"]
+        cn = data["o_cn"][li["orderkey"]]
+        sn = data["snk"][li["suppkey"]]
+        sd = li["shipdate"]
+        win = (sd >= 9131) & (sd < 9862)
+        year = np.where(sd < 9496, 1995, 1996)
+        for i, row in enumerate(answers["q07"]):
+            s_nk = 6 if row[0] == "FRANCE" else 7
+            c_nk = 13 - s_nk
+            m = win & (sn == s_nk) & (cn == c_nk) & (year == int(row[2]))
+            assert abs(data["rev"][m].sum() - float(row[3])) < 5e-4
+
+    def test_q8(self, data, answers):
+        p = oracle.gen_part(1.0)
+        ptype = np.zeros(200_001, np.int16)
+        ptype[p["partkey"]] = p["type_id"]
+        li = data["li"]
+        cn = data["o_cn"][li["orderkey"]]
+        od = data["o_dt"][li["orderkey"]]
+        sn = data["snk"][li["suppkey"]]
+        AMERICA = np.array([1, 2, 3, 17, 24])
+        m = ((ptype[li["partkey"]] == 103) & np.isin(cn, AMERICA) &
+             (od >= 9131) & (od < 9862))
+        year = np.where(od < 9496, 1995, 1996)
+        for i, row in enumerate(answers["q08"]):
+            ym = m & (year == int(row[0]))
+            share = data["rev"][ym & (sn == 2)].sum() / data["rev"][ym].sum()
+            assert abs(share - float(row[1])) < 5e-7
