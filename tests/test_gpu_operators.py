@@ -98,6 +98,89 @@ class TestFilter:
         exp = sel_pos[a[sel_pos] < 500]
         assert np.array_equal(got, exp)
 
+    def test_bigint_exact_above_2p53(self, sess, ops):
+        """typed i64 lane: BIGINT compares are exact above 2^53 (the f64 lane
+        would collapse base and base+1 to the same double). Exercises both the
+        single-cmp fast path and the FTerms AND path."""
+        base = (1 << 60) + 1
+        a = np.array([base, base + 1, base - 1, base, 2, -base], dtype=np.int64)
+        page = ops.page_from_numpy([a])
+        got = ops.filter_run(sess, ops.expr(("col", 0), ("i64", base), "eq"), page)
+        assert np.array_equal(got, np.nonzero(a == base)[0].astype(np.int32))
+        got = ops.filter_run(sess, ops.expr(("col", 0), ("i64", base), "le"), page)
+        assert np.array_equal(got, np.nonzero(a <= base)[0].astype(np.int32))
+        # AND-of-terms path (k_filter_terms): a == base AND a > 0
+        e = ops.expr(("col", 0), ("i64", base), "eq",
+                     ("col", 0), ("i64", 0), "gt", "and")
+        got = ops.filter_run(sess, e, page)
+        assert np.array_equal(got, np.nonzero((a == base) & (a > 0))[0].astype(np.int32))
+        # interpreter path (arith forces generic): (a - 1) == base
+        e = ops.expr(("col", 0), ("i64", 1), "sub", ("i64", base), "eq")
+        got = ops.filter_run(sess, e, page)
+        assert np.array_equal(got, np.nonzero((a - 1) == base)[0].astype(np.int32))
+
+    def test_kleene_three_valued_logic(self, sess, ops):
+        """AND/OR/NOT follow SQL Kleene 3VL (io.trino.sql.ir.Logical):
+        NOT(NULL OR FALSE) is NULL (reject), NOT(NULL AND FALSE) is TRUE
+        (select). The round-1 null-to-false coercion selected the first and
+        rejected the second."""
+        n = 8
+        a = np.array([1, 1, 0, 0, 1, 0, 1, 0], dtype=np.int64)
+        b = np.array([1, 0, 1, 0, 1, 1, 0, 0], dtype=np.int64)
+        # a is NULL on rows 0..3
+        valid = np.array([~np.uint64(0) << np.uint64(4)], dtype=np.uint64)
+        page = ops.page_from_numpy([a, b], valids=[valid, None])
+        av = [None, None, None, None, True, False, True, False]
+        bv = [True, False, True, False, True, True, False, False]
+
+        def k_or(x, y):
+            if x is True or y is True: return True
+            if x is None or y is None: return None
+            return False
+
+        def k_and(x, y):
+            if x is False or y is False: return False
+            if x is None or y is None: return None
+            return True
+
+        def k_not(x):
+            return None if x is None else not x
+
+        # NOT (a OR b)
+        e = ops.expr(("col", 0), ("i64", 0), "ne", ("col", 1), ("i64", 0), "ne",
+                     "or", "not")
+        got = ops.filter_run(sess, e, page)
+        exp = [i for i in range(n) if k_not(k_or(av[i], bv[i])) is True]
+        assert got.tolist() == exp
+        # NOT (a AND b)
+        e = ops.expr(("col", 0), ("i64", 0), "ne", ("col", 1), ("i64", 0), "ne",
+                     "and", "not")
+        got = ops.filter_run(sess, e, page)
+        exp = [i for i in range(n) if k_not(k_and(av[i], bv[i])) is True]
+        assert got.tolist() == exp
+        # plain OR: NULL OR TRUE selects (true dominates null)
+        e = ops.expr(("col", 0), ("i64", 0), "ne", ("col", 1), ("i64", 0), "ne", "or")
+        got = ops.filter_run(sess, e, page)
+        exp = [i for i in range(n) if k_or(av[i], bv[i]) is True]
+        assert got.tolist() == exp
+
+    def test_bigint_projection_lane(self, sess, ops):
+        """non-identity BIGINT projection keeps exact int64 (orderkey+1 shape
+        above 2^53)"""
+        base = (1 << 60) + 7
+        a = np.array([base, base + 2, 5, -base], dtype=np.int64)
+        sd = np.array([0, 1, 0, 1], dtype=np.int32)
+        page = ops.page_from_numpy([a, sd])
+        f = ops.expr(("col", 1), ("i64", 0), "ge")   # select all
+        projs = [ops.expr(("col", 0), ("i64", 1), "add")]
+        op = ops.filter_project(sess, f, projs, out_types=[ops.TG_BIGINT])
+        op.add_input(page)
+        pages = op.drain()
+        op.close()
+        got = pages[0][0]["values"]
+        assert got.dtype == np.int64
+        assert np.array_equal(got, a + 1)
+
     def test_filter_project_operator(self, sess, ops):
         r = rng(4)
         n = 30000

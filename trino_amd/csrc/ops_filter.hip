@@ -19,74 +19,125 @@ struct KCol { const void* data; const uint64_t* valid; int32_t type; int32_t _pa
 
 #define MAX_STACK 4
 
-__device__ static inline double load_col(const KCol& c, int64_t i, bool* isnull)
+/* Typed stack value: BIGINT/INTEGER/DATE/… stay in an exact int64 lane
+ * (mirrors sql/gen/columnar/CallColumnarFilterGenerator.java:160-198 which
+ * compiles the comparison at the column's Java type — long compares are
+ * exact there, so they must be exact here; keys above 2^53 would silently
+ * lose precision in a double lane). `f` mirrors the value for mixed-type
+ * ops (Trino coerces BIGINT to DOUBLE in mixed expressions). */
+struct TVal { double f; int64_t i; bool isint; bool null; };
+
+__device__ static inline TVal load_tcol(const KCol& c, int64_t i)
 {
-    if (c.valid && !((c.valid[i >> 6] >> (i & 63)) & 1)) { *isnull = true; return 0.0; }
+    TVal v;
+    v.null = c.valid && !((c.valid[i >> 6] >> (i & 63)) & 1);
+    if (v.null) { v.f = 0.0; v.i = 0; v.isint = true; return v; }
     switch (c.type) {
-        case TG_BIGINT: return (double)((const int64_t*)c.data)[i];
-        case TG_INTEGER: case TG_DATE: return (double)((const int32_t*)c.data)[i];
-        case TG_SMALLINT: return (double)((const int16_t*)c.data)[i];
-        case TG_TINYINT: case TG_BOOLEAN: return (double)((const int8_t*)c.data)[i];
-        default: return ((const double*)c.data)[i];
+        case TG_BIGINT: v.i = ((const int64_t*)c.data)[i]; v.isint = true; v.f = (double)v.i; break;
+        case TG_INTEGER: case TG_DATE: v.i = ((const int32_t*)c.data)[i]; v.isint = true; v.f = (double)v.i; break;
+        case TG_SMALLINT: v.i = ((const int16_t*)c.data)[i]; v.isint = true; v.f = (double)v.i; break;
+        case TG_TINYINT: case TG_BOOLEAN: v.i = ((const int8_t*)c.data)[i]; v.isint = true; v.f = (double)v.i; break;
+        default: v.f = ((const double*)c.data)[i]; v.i = 0; v.isint = false; break;
     }
+    return v;
 }
 
-/* postfix interpreter; returns value, sets *rnull. Bool results are 0.0/1.0.
- * Null comparison/arith poisons; AND/OR treat null as false (a NULL argument
- * rejects the row for that term, OR still unions other terms). */
-__device__ static double eval_expr(const tg_expr_inst* prog, int count,
-                                   const KCol* cols, int64_t row, bool* rnull)
+__device__ static inline bool tval_truthy(const TVal& v) { return v.isint ? v.i != 0 : v.f != 0.0; }
+
+/* postfix interpreter; returns a typed value. Integer ops are exact int64
+ * (incl. BIGINT/BIGINT truncated division); mixed int/double coerces to
+ * double. Comparisons compare in int64 iff both sides are int. AND/OR/NOT
+ * follow SQL Kleene three-valued logic (io.trino.sql.ir.Logical semantics:
+ * AND — false dominates null; OR — true dominates null; NOT null = null),
+ * so compositions under NOT are correct; a null FILTER result rejects the
+ * row at the call site (CallColumnarFilterGenerator's nullable loop). */
+__device__ static TVal eval_expr(const tg_expr_inst* prog, int count,
+                                 const KCol* cols, int64_t row)
 {
-    double s0 = 0, s1 = 0, s2 = 0, s3 = 0;
-    bool n0 = false, n1 = false, n2 = false, n3 = false;
+    TVal s0{}, s1{}, s2{}, s3{};
     int sp = 0;
-#define PUSH(v, nn) do { switch (sp) { \
-        case 0: s0 = (v); n0 = (nn); break; case 1: s1 = (v); n1 = (nn); break; \
-        case 2: s2 = (v); n2 = (nn); break; default: s3 = (v); n3 = (nn); break; } \
+#define PUSH(v) do { switch (sp) { \
+        case 0: s0 = (v); break; case 1: s1 = (v); break; \
+        case 2: s2 = (v); break; default: s3 = (v); break; } \
         sp++; } while (0)
-#define TOP_V (sp == 1 ? s0 : sp == 2 ? s1 : sp == 3 ? s2 : s3)
-#define TOP_N (sp == 1 ? n0 : sp == 2 ? n1 : sp == 3 ? n2 : n3)
-#define POP(v, nn) do { v = TOP_V; nn = TOP_N; sp--; } while (0)
+#define TOP (sp == 1 ? s0 : sp == 2 ? s1 : sp == 3 ? s2 : s3)
+#define POP(v) do { v = TOP; sp--; } while (0)
+#define ARITH(expr_i, expr_f) do { TVal r; r.null = a.null | b.null; \
+        r.isint = a.isint & b.isint; \
+        if (r.isint) { r.i = (expr_i); r.f = (double)r.i; } \
+        else { r.f = (expr_f); r.i = 0; } PUSH(r); } while (0)
+#define CMP(op_) do { TVal r; r.null = a.null | b.null; r.isint = true; \
+        r.i = (a.isint & b.isint) ? (a.i op_ b.i ? 1 : 0) \
+                                  : (a.f op_ b.f ? 1 : 0); \
+        r.f = (double)r.i; PUSH(r); } while (0)
 
     for (int k = 0; k < count; k++) {
         tg_expr_inst in = prog[k];
-        double a, b, c;
-        bool na, nb, nc;
+        TVal a, b, c;
         switch (in.op) {
-            case TG_EXPR_COL: {
-                bool nl = false;
-                double v = load_col(cols[in.arg0], row, &nl);
-                PUSH(v, nl);
-                break;
+            case TG_EXPR_COL: PUSH(load_tcol(cols[in.arg0], row)); break;
+            case TG_EXPR_CONST_F64: { TVal v; v.f = in.imm.f64; v.i = 0; v.isint = false; v.null = false; PUSH(v); break; }
+            case TG_EXPR_CONST_I64: { TVal v; v.i = in.imm.i64; v.f = (double)v.i; v.isint = true; v.null = false; PUSH(v); break; }
+            case TG_EXPR_ADD: POP(b); POP(a); ARITH(a.i + b.i, a.f + b.f); break;
+            case TG_EXPR_SUB: POP(b); POP(a); ARITH(a.i - b.i, a.f - b.f); break;
+            case TG_EXPR_MUL: POP(b); POP(a); ARITH(a.i * b.i, a.f * b.f); break;
+            case TG_EXPR_DIV: POP(b); POP(a); {
+                TVal r; r.isint = a.isint & b.isint;
+                if (r.isint) {   /* BIGINT/BIGINT truncates; /0 poisons to null
+                                    (the C-ABI has no per-row error channel;
+                                    Trino raises DIVISION_BY_ZERO) */
+                    r.null = a.null | b.null | (b.i == 0);
+                    r.i = (b.i == 0) ? 0 : a.i / b.i;
+                    r.f = (double)r.i;
+                }
+                else { r.null = a.null | b.null; r.f = a.f / b.f; r.i = 0; }
+                PUSH(r); break;
             }
-            case TG_EXPR_CONST_F64: PUSH(in.imm.f64, false); break;
-            case TG_EXPR_CONST_I64: PUSH((double)in.imm.i64, false); break;
-            case TG_EXPR_ADD: POP(b, nb); POP(a, na); PUSH(a + b, na | nb); break;
-            case TG_EXPR_SUB: POP(b, nb); POP(a, na); PUSH(a - b, na | nb); break;
-            case TG_EXPR_MUL: POP(b, nb); POP(a, na); PUSH(a * b, na | nb); break;
-            case TG_EXPR_DIV: POP(b, nb); POP(a, na); PUSH(a / b, na | nb); break;
-            case TG_EXPR_LE: POP(b, nb); POP(a, na); PUSH((a <= b) ? 1.0 : 0.0, na | nb); break;
-            case TG_EXPR_LT: POP(b, nb); POP(a, na); PUSH((a < b) ? 1.0 : 0.0, na | nb); break;
-            case TG_EXPR_GE: POP(b, nb); POP(a, na); PUSH((a >= b) ? 1.0 : 0.0, na | nb); break;
-            case TG_EXPR_GT: POP(b, nb); POP(a, na); PUSH((a > b) ? 1.0 : 0.0, na | nb); break;
-            case TG_EXPR_EQ: POP(b, nb); POP(a, na); PUSH((a == b) ? 1.0 : 0.0, na | nb); break;
-            case TG_EXPR_NE: POP(b, nb); POP(a, na); PUSH((a != b) ? 1.0 : 0.0, na | nb); break;
-            case TG_EXPR_AND: POP(b, nb); POP(a, na);
-                PUSH(((a != 0.0) & !na & (b != 0.0) & !nb) ? 1.0 : 0.0, false); break;
-            case TG_EXPR_OR: POP(b, nb); POP(a, na);
-                PUSH((((a != 0.0) & !na) | ((b != 0.0) & !nb)) ? 1.0 : 0.0, false); break;
-            case TG_EXPR_NOT: POP(a, na); PUSH((a == 0.0) ? 1.0 : 0.0, na); break;
-            case TG_EXPR_BETWEEN: POP(c, nc); POP(b, nb); POP(a, na);
-                PUSH((a >= b && a <= c) ? 1.0 : 0.0, na | nb | nc); break;
+            case TG_EXPR_LE: POP(b); POP(a); CMP(<=); break;
+            case TG_EXPR_LT: POP(b); POP(a); CMP(<); break;
+            case TG_EXPR_GE: POP(b); POP(a); CMP(>=); break;
+            case TG_EXPR_GT: POP(b); POP(a); CMP(>); break;
+            case TG_EXPR_EQ: POP(b); POP(a); CMP(==); break;
+            case TG_EXPR_NE: POP(b); POP(a); CMP(!=); break;
+            case TG_EXPR_AND: { POP(b); POP(a);
+                bool af = !a.null && !tval_truthy(a), bf = !b.null && !tval_truthy(b);
+                TVal r; r.isint = true;
+                r.null = !(af | bf) && (a.null | b.null);
+                r.i = (!r.null && !af && !bf) ? 1 : 0;
+                r.f = (double)r.i; PUSH(r); break;
+            }
+            case TG_EXPR_OR: { POP(b); POP(a);
+                bool at = !a.null && tval_truthy(a), bt = !b.null && tval_truthy(b);
+                TVal r; r.isint = true;
+                r.null = !(at | bt) && (a.null | b.null);
+                r.i = (at | bt) ? 1 : 0;
+                r.f = (double)r.i; PUSH(r); break;
+            }
+            case TG_EXPR_NOT: { POP(a);
+                TVal r; r.isint = true; r.null = a.null;
+                r.i = (!a.null && !tval_truthy(a)) ? 1 : 0;
+                r.f = (double)r.i; PUSH(r); break;
+            }
+            case TG_EXPR_BETWEEN: { POP(c); POP(b); POP(a);
+                /* a >= b AND a <= c under Kleene AND, typed compares */
+                bool ge, le, ge_n = a.null | b.null, le_n = a.null | c.null;
+                ge = (a.isint & b.isint) ? a.i >= b.i : a.f >= b.f;
+                le = (a.isint & c.isint) ? a.i <= c.i : a.f <= c.f;
+                bool af = !ge_n && !ge, bf = !le_n && !le;
+                TVal r; r.isint = true;
+                r.null = !(af | bf) && (ge_n | le_n);
+                r.i = (!r.null && !af && !bf) ? 1 : 0;
+                r.f = (double)r.i; PUSH(r); break;
+            }
             default: break;
         }
     }
-    *rnull = TOP_N;
-    return TOP_V;
+    return TOP;
 #undef PUSH
 #undef POP
-#undef TOP_V
-#undef TOP_N
+#undef TOP
+#undef ARITH
+#undef CMP
 }
 
 /* row index for the k-th selected input position */
@@ -97,8 +148,10 @@ __device__ static inline int64_t sel_row(int has_list, const int32_t* list,
 }
 
 /* specialization for the dominant shape COL <cmp> CONST (ColumnarFilter's
- * compiled single-predicate case): no interpreter loop, vector-friendly */
-__global__ void k_filter_cmp(KCol col, int op, double cval,
+ * compiled single-predicate case): no interpreter loop, vector-friendly.
+ * int_mode: col is an integer type and the constant was CONST_I64 —
+ * compare exactly in int64 (keys above 2^53 must not round). */
+__global__ void k_filter_cmp(KCol col, int op, double cval, int64_t icval, int int_mode,
                              int has_list, const int32_t* list, int32_t offset,
                              int64_t n, uint8_t* __restrict__ flags)
 {
@@ -106,18 +159,29 @@ __global__ void k_filter_cmp(KCol col, int op, double cval,
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; k < n; k += stride) {
         int64_t i = has_list ? (int64_t)list[k] : offset + k;
-        bool isnull = false;
-        double v = load_col(col, i, &isnull);
         bool r;
-        switch (op) {
-            case TG_EXPR_LE: r = v <= cval; break;
-            case TG_EXPR_LT: r = v < cval; break;
-            case TG_EXPR_GE: r = v >= cval; break;
-            case TG_EXPR_GT: r = v > cval; break;
-            case TG_EXPR_EQ: r = v == cval; break;
-            default: r = v != cval; break;
+        TVal v = load_tcol(col, i);
+        if (int_mode) {
+            switch (op) {
+                case TG_EXPR_LE: r = v.i <= icval; break;
+                case TG_EXPR_LT: r = v.i < icval; break;
+                case TG_EXPR_GE: r = v.i >= icval; break;
+                case TG_EXPR_GT: r = v.i > icval; break;
+                case TG_EXPR_EQ: r = v.i == icval; break;
+                default: r = v.i != icval; break;
+            }
         }
-        flags[k] = (!isnull && r) ? 1 : 0;
+        else {
+            switch (op) {
+                case TG_EXPR_LE: r = v.f <= cval; break;
+                case TG_EXPR_LT: r = v.f < cval; break;
+                case TG_EXPR_GE: r = v.f >= cval; break;
+                case TG_EXPR_GT: r = v.f > cval; break;
+                case TG_EXPR_EQ: r = v.f == cval; break;
+                default: r = v.f != cval; break;
+            }
+        }
+        flags[k] = (!v.null && r) ? 1 : 0;
     }
 }
 
@@ -131,10 +195,25 @@ struct FTerm {
                          2 = colA between [c1,c2], 3 = colA in {c1,c2} */
     int32_t colA, colB, op;
     double c1, c2;
+    int64_t i1, i2;   /* exact int64 views of c1/c2 when int_mode */
+    int32_t int_mode; /* colA is an integer type and consts were CONST_I64 */
+    int32_t _pad;
 };
 struct FTerms { FTerm t[FTERM_MAX]; int n; };
 
 __device__ static inline bool fterm_cmp(int op, double a, double b)
+{
+    switch (op) {
+        case TG_EXPR_LE: return a <= b;
+        case TG_EXPR_LT: return a < b;
+        case TG_EXPR_GE: return a >= b;
+        case TG_EXPR_GT: return a > b;
+        case TG_EXPR_EQ: return a == b;
+        default: return a != b;
+    }
+}
+
+__device__ static inline bool fterm_cmpi(int op, int64_t a, int64_t b)
 {
     switch (op) {
         case TG_EXPR_LE: return a <= b;
@@ -158,17 +237,21 @@ __global__ void k_filter_terms(FTerms ft, const KCol* __restrict__ cols,
         bool pass = true;
         for (int t = 0; t < ft.n && pass; t++) {
             const FTerm& f = ft.t[t];
-            bool nl = false;
-            double a = load_col(cols[f.colA], i, &nl);
+            TVal a = load_tcol(cols[f.colA], i);
             switch (f.kind) {
-                case 0: pass = !nl && fterm_cmp(f.op, a, f.c1); break;
+                case 0: pass = !a.null && (f.int_mode ? fterm_cmpi(f.op, a.i, f.i1)
+                                                      : fterm_cmp(f.op, a.f, f.c1)); break;
                 case 1: {
-                    double b = load_col(cols[f.colB], i, &nl);
-                    pass = !nl && fterm_cmp(f.op, a, b);
+                    TVal b = load_tcol(cols[f.colB], i);
+                    pass = !a.null && !b.null &&
+                           ((a.isint & b.isint) ? fterm_cmpi(f.op, a.i, b.i)
+                                                : fterm_cmp(f.op, a.f, b.f));
                     break;
                 }
-                case 2: pass = !nl && a >= f.c1 && a <= f.c2; break;
-                default: pass = !nl && (a == f.c1 || a == f.c2); break;
+                case 2: pass = !a.null && (f.int_mode ? (a.i >= f.i1 && a.i <= f.i2)
+                                                      : (a.f >= f.c1 && a.f <= f.c2)); break;
+                default: pass = !a.null && (f.int_mode ? (a.i == f.i1 || a.i == f.i2)
+                                                       : (a.f == f.c1 || a.f == f.c2)); break;
             }
         }
         flags[k] = pass ? 1 : 0;
@@ -176,11 +259,18 @@ __global__ void k_filter_terms(FTerms ft, const KCol* __restrict__ cols,
 }
 
 /* try to parse the postfix program as T1 (T2 AND)*; returns term count or 0 */
-static int parse_fterms(const ExprProgram& pred, FTerms* out)
+static bool type_is_int(int32_t t)
 {
-    auto cval = [](const tg_expr_inst& in, double* v) {
-        if (in.op == TG_EXPR_CONST_F64) { *v = in.imm.f64; return true; }
-        if (in.op == TG_EXPR_CONST_I64) { *v = (double)in.imm.i64; return true; }
+    return t == TG_BIGINT || t == TG_INTEGER || t == TG_DATE ||
+           t == TG_SMALLINT || t == TG_TINYINT || t == TG_BOOLEAN;
+}
+
+static int parse_fterms(const ExprProgram& pred, const std::vector<KCol>& cols, FTerms* out)
+{
+    /* i64-ness tracked per constant so integer-column terms compare exactly */
+    auto cval = [](const tg_expr_inst& in, double* v, int64_t* iv, bool* is_i64) {
+        if (in.op == TG_EXPR_CONST_F64) { *v = in.imm.f64; *iv = (int64_t)in.imm.f64; *is_i64 = false; return true; }
+        if (in.op == TG_EXPR_CONST_I64) { *v = (double)in.imm.i64; *iv = in.imm.i64; *is_i64 = true; return true; }
         return false;
     };
     auto is_cmp = [](int op) { return op >= TG_EXPR_LE && op <= TG_EXPR_NE; };
@@ -190,29 +280,35 @@ static int parse_fterms(const ExprProgram& pred, FTerms* out)
     auto term = [&]() -> bool {
         if (nt >= FTERM_MAX || p >= N || I[p].op != TG_EXPR_COL) return false;
         FTerm& f = out->t[nt];
+        bool i1 = false, i2 = false;
+        bool col_int = type_is_int(cols[I[p].arg0].type);
         /* IN-pair: COL C EQ COL C EQ OR (same column) */
-        if (p + 6 < N && cval(I[p + 1], &f.c1) && I[p + 2].op == TG_EXPR_EQ &&
+        if (p + 6 < N && cval(I[p + 1], &f.c1, &f.i1, &i1) && I[p + 2].op == TG_EXPR_EQ &&
             I[p + 3].op == TG_EXPR_COL && I[p + 3].arg0 == I[p].arg0 &&
-            cval(I[p + 4], &f.c2) && I[p + 5].op == TG_EXPR_EQ &&
+            cval(I[p + 4], &f.c2, &f.i2, &i2) && I[p + 5].op == TG_EXPR_EQ &&
             I[p + 6].op == TG_EXPR_OR) {
             f.kind = 3; f.colA = I[p].arg0; f.colB = -1; f.op = 0;
+            f.int_mode = col_int && i1 && i2;
             p += 7; nt++; return true;
         }
         /* BETWEEN: COL C C BETWEEN */
-        if (p + 3 < N && cval(I[p + 1], &f.c1) && cval(I[p + 2], &f.c2) &&
+        if (p + 3 < N && cval(I[p + 1], &f.c1, &f.i1, &i1) &&
+            cval(I[p + 2], &f.c2, &f.i2, &i2) &&
             I[p + 3].op == TG_EXPR_BETWEEN) {
             f.kind = 2; f.colA = I[p].arg0; f.colB = -1; f.op = 0;
+            f.int_mode = col_int && i1 && i2;
             p += 4; nt++; return true;
         }
         /* COL cmp CONST */
-        if (p + 2 < N && cval(I[p + 1], &f.c1) && is_cmp(I[p + 2].op)) {
+        if (p + 2 < N && cval(I[p + 1], &f.c1, &f.i1, &i1) && is_cmp(I[p + 2].op)) {
             f.kind = 0; f.colA = I[p].arg0; f.colB = -1; f.op = I[p + 2].op;
+            f.int_mode = col_int && i1;
             p += 3; nt++; return true;
         }
         /* COL cmp COL */
         if (p + 2 < N && I[p + 1].op == TG_EXPR_COL && is_cmp(I[p + 2].op)) {
             f.kind = 1; f.colA = I[p].arg0; f.colB = I[p + 1].arg0;
-            f.op = I[p + 2].op; f.c1 = f.c2 = 0;
+            f.op = I[p + 2].op; f.c1 = f.c2 = 0; f.i1 = f.i2 = 0; f.int_mode = 0;
             p += 3; nt++; return true;
         }
         return false;
@@ -234,9 +330,8 @@ __global__ void k_filter_flags(const tg_expr_inst* prog, int count, const KCol* 
     int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; k < n; k += stride) {
-        bool isnull = false;
-        double v = eval_expr(prog, count, cols, sel_row(has_list, list, offset, k), &isnull);
-        flags[k] = (!isnull && v != 0.0) ? 1 : 0;
+        TVal v = eval_expr(prog, count, cols, sel_row(has_list, list, offset, k));
+        flags[k] = (!v.null && tval_truthy(v)) ? 1 : 0;
     }
 }
 
@@ -420,9 +515,8 @@ __global__ void k_fp_count(const tg_expr_inst* prog, int count, const KCol* cols
     int64_t lo = c * CHUNK, hi = min(lo + CHUNK, n);
     int32_t cnt = 0;
     for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-        bool isnull = false;
-        double v = eval_expr(prog, count, cols, i, &isnull);
-        cnt += (!isnull && v != 0.0) ? 1 : 0;
+        TVal v = eval_expr(prog, count, cols, i);
+        cnt += (!v.null && tval_truthy(v)) ? 1 : 0;
     }
     #pragma unroll
     for (int off = 32; off >= 1; off >>= 1) cnt += __shfl_xor(cnt, off, 64);
@@ -455,9 +549,8 @@ __global__ void k_fp_write(const tg_expr_inst* prog, int count, const KCol* cols
         int64_t i = lo + (int64_t)g * 64 + lane;
         bool sel = false;
         if (i < hi) {
-            bool isnull = false;
-            double v = eval_expr(prog, count, cols, i, &isnull);
-            sel = !isnull && v != 0.0;
+            TVal v = eval_expr(prog, count, cols, i);
+            sel = !v.null && tval_truthy(v);
         }
         unsigned long long b = __ballot(sel);
         if (lane == 0) mask[g] = b;
@@ -508,10 +601,13 @@ __global__ void k_fp_write(const tg_expr_inst* prog, int count, const KCol* cols
                                   ~(1ull << (at & 63)));
                 }
                 else {
-                    bool pnull = false;
-                    double pv = eval_expr(pr.insts, pr.count, cols, i, &pnull);
-                    ((double*)pr.out)[at] = pnull ? 0.0 : pv;
-                    if (pr.out_valid && pnull)
+                    TVal pv = eval_expr(pr.insts, pr.count, cols, i);
+                    if (pr.out_type == TG_BIGINT)
+                        ((int64_t*)pr.out)[at] = pv.null ? 0
+                            : (pv.isint ? pv.i : (int64_t)pv.f);
+                    else
+                        ((double*)pr.out)[at] = pv.null ? 0.0 : pv.f;
+                    if (pr.out_valid && pv.null)
                         atomicAnd((unsigned long long*)&pr.out_valid[at >> 6],
                                   ~(1ull << (at & 63)));
                 }
@@ -553,13 +649,15 @@ tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page
                 (pred.insts[1].op == TG_EXPR_CONST_F64 || pred.insts[1].op == TG_EXPR_CONST_I64) &&
                 pred.insts[2].op >= TG_EXPR_LE && pred.insts[2].op <= TG_EXPR_NE;
     if (fast) {
-        double cval = pred.insts[1].op == TG_EXPR_CONST_F64 ? pred.insts[1].imm.f64
-                                                            : (double)pred.insts[1].imm.i64;
+        bool is_i64 = pred.insts[1].op == TG_EXPR_CONST_I64;
+        double cval = is_i64 ? (double)pred.insts[1].imm.i64 : pred.insts[1].imm.f64;
+        int64_t icval = is_i64 ? pred.insts[1].imm.i64 : (int64_t)pred.insts[1].imm.f64;
+        int int_mode = is_i64 && type_is_int(cols[pred.insts[0].arg0].type);
         hipLaunchKernelGGL(k_filter_cmp, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
-                           cols[pred.insts[0].arg0], pred.insts[2].op, cval,
+                           cols[pred.insts[0].arg0], pred.insts[2].op, cval, icval, int_mode,
                            has_list, d_list, offset, n, d_flags);
     }
-    else if (FTerms ft{}; parse_fterms(pred, &ft) > 0) {
+    else if (FTerms ft{}; parse_fterms(pred, cols, &ft) > 0) {
         hipLaunchKernelGGL(k_filter_terms, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
                            ft, d_cols, has_list, d_list, offset, n, d_flags);
     }
@@ -606,15 +704,18 @@ tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page
 /* ---- projection ---- */
 __global__ void k_project(const tg_expr_inst* prog, int count, const KCol* cols,
                           const int32_t* __restrict__ positions, int32_t n,
-                          double* __restrict__ out, uint64_t* __restrict__ out_valid)
+                          void* __restrict__ out, int out_is_i64,
+                          uint64_t* __restrict__ out_valid)
 {
     int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; k < n; k += stride) {
-        bool isnull = false;
-        double v = eval_expr(prog, count, cols, positions ? positions[k] : k, &isnull);
-        out[k] = isnull ? 0.0 : v;
-        if (out_valid && isnull)
+        TVal v = eval_expr(prog, count, cols, positions ? positions[k] : k);
+        if (out_is_i64)
+            ((int64_t*)out)[k] = v.null ? 0 : (v.isint ? v.i : (int64_t)v.f);
+        else
+            ((double*)out)[k] = v.null ? 0.0 : v.f;
+        if (out_valid && v.null)
             atomicAnd((unsigned long long*)&out_valid[k >> 6], ~(1ull << (k & 63)));
     }
 }
@@ -800,7 +901,8 @@ tg_status run_project(tg_session* s, const ExprProgram& proj, tg_type out_type,
     TG_POOL_ALLOC(s, &d_cols, cols.size() * sizeof(KCol));
     TG_HIP_CHECK(hipMemcpyAsync(d_cols, cols.data(), cols.size() * sizeof(KCol),
                                 hipMemcpyHostToDevice, s->stream));
-    out->type = TG_DOUBLE;
+    int out_is_i64 = out_type == TG_BIGINT;
+    out->type = out_is_i64 ? TG_BIGINT : TG_DOUBLE;
     out->n = count;
     TG_POOL_ALLOC(s, &out->data, (int64_t)(count ? count : 1) * 8);
     if (any_null) {
@@ -810,11 +912,10 @@ tg_status run_project(tg_session* s, const ExprProgram& proj, tg_type out_type,
     }
     hipLaunchKernelGGL(k_project, dim3(tg_grid_for(count)), dim3(TG_BLOCK), 0, s->stream,
                        proj.d_insts, proj.count, d_cols, d_positions, count,
-                       (double*)out->data, out->valid);
+                       out->data, out_is_i64, out->valid);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     tg_pool_free(s, d_cols);
-    (void)out_type;
     return TG_OK;
 }
 
@@ -907,7 +1008,8 @@ struct FilterProjectOp : tg_operator {
             bool ident = projections[p].count == 1 &&
                          projections[p].insts[0].op == TG_EXPR_COL;
             int src = ident ? projections[p].insts[0].arg0 : -1;
-            ob.type = ident ? in.blocks[src].type : TG_DOUBLE;
+            ob.type = ident ? in.blocks[src].type
+                            : (out_types[p] == TG_BIGINT ? TG_BIGINT : TG_DOUBLE);
             ob.n = total;
             TG_POOL_ALLOC(s, &ob.data, (int64_t)(total ? total : 1) * ob.elem_size());
             if (any_null) {

@@ -89,7 +89,11 @@ __global__ void k_gt_assign(GTable t, const KColH* cols, int n_ch_dyn,
         int64_t slot = (int64_t)(d_murmur3_mix(h) & (uint64_t)t.mask);
         int32_t gid = -1;
         while (true) {
-            int32_t st = __hip_atomic_load(&t.state[slot], __ATOMIC_RELAXED,
+            /* ACQUIRE pairs with the release publication of state[slot] below:
+             * a thread that observes gid must also observe the keystore words
+             * (the address dependency works on current hardware, but the
+             * formal HIP memory model needs the acquire) */
+            int32_t st = __hip_atomic_load(&t.state[slot], __ATOMIC_ACQUIRE,
                                            __HIP_MEMORY_SCOPE_AGENT);
             if (st == -1) {
                 int32_t old = -1;
@@ -121,7 +125,7 @@ __global__ void k_gt_assign(GTable t, const KColH* cols, int n_ch_dyn,
             }
             while (st == -2) {
                 __builtin_amdgcn_s_sleep(1);
-                st = __hip_atomic_load(&t.state[slot], __ATOMIC_RELAXED,
+                st = __hip_atomic_load(&t.state[slot], __ATOMIC_ACQUIRE,
                                        __HIP_MEMORY_SCOPE_AGENT);
             }
             /* st >= 0: compare keys */

@@ -828,6 +828,15 @@ struct HashBuilderOp : tg_operator {
         t.n = total_rows;
         t.capacity = join_hash_size(total_rows);
         t.mask = t.capacity - 1;
+        /* probe kernels encode a multi-match bucket slot as (int32_t)slot
+         * (k_probe_count heads); a table beyond INT32_MAX slots would wrap
+         * negative and be read back as a miss — reject at build, mirroring
+         * BigintGroupByHash's own 2^30 capacity cap (BigintGroupByHash.java) */
+        if (t.capacity > (int64_t)INT32_MAX) {
+            TG_SET_ERR("join build of %lld rows needs %lld slots > INT32_MAX",
+                       (long long)total_rows, (long long)t.capacity);
+            return TG_ERR_OOM;
+        }
         static int pre_csr = [] { const char* e = getenv("TG_JOIN_CSR"); return e ? atoi(e) : 1; }();
         if (!(pre_csr && total_rows > 0)) {
             TG_POOL_ALLOC(s, &t.slots, t.capacity * 4);
