@@ -146,3 +146,91 @@ EXPORT int64_t tpch_gen_part_cols(double sf, int64_t part_start, int64_t part_co
     tpch_gen_part(part_start, part_count, partkey, type_id);
     return part_count;
 }
+
+/* ---- round 2: text pool + string streams (tpch_text.h) ---- */
+#include "tpch_text.h"
+#include <stdlib.h>
+
+static char* g_pool = NULL;
+
+EXPORT const char* tpch_text_pool(void)
+{
+    if (!g_pool) {
+        g_pool = (char*)malloc(TPCH_TEXT_POOL_SIZE);
+        tpch_text_pool_build(g_pool, TPCH_TEXT_POOL_SIZE);
+    }
+    return g_pool;
+}
+
+/* comment slices for `count` rows of a stream seeded `seed`, `per_value`
+ * values per row, `usage` stream uses per row, average length `avg`.
+ * offs/lens arrays sized count*per_value. */
+EXPORT void tpch_text_slices(int64_t seed, int64_t row_start /*1-based*/,
+                             int64_t count, int per_value, int usage, int avg,
+                             int64_t* offs, int32_t* lens)
+{
+    tpch_rng r;
+    tpch_rng_init(&r, seed, usage);
+    tpch_rng_skip(&r, (row_start - 1) * usage);
+    for (int64_t i = 0; i < count; i++) {
+        for (int v = 0; v < per_value; v++)
+            tpch_text_slice(&r, avg, &offs[i * per_value + v],
+                            &lens[i * per_value + v]);
+        tpch_rng_row_finished(&r);
+    }
+}
+
+EXPORT void tpch_gen_vstr(int64_t seed, int64_t row_start, int64_t count,
+                          int usage, int avg, char* out, int stride,
+                          int32_t* lens)
+{
+    tpch_rng r;
+    tpch_rng_init(&r, seed, usage);
+    tpch_rng_skip(&r, (row_start - 1) * usage);
+    for (int64_t i = 0; i < count; i++) {
+        lens[i] = tpch_vstr(&r, avg, out + i * stride);
+        tpch_rng_row_finished(&r);
+    }
+}
+
+EXPORT void tpch_gen_part_names(int64_t part_start, int64_t count, uint8_t* ids5)
+{
+    tpch_rng r;
+    tpch_rng_init(&r, TPCH_SEED_P_NAME, TPCH_P_NAME_USAGE);
+    tpch_rng_skip(&r, (part_start - 1) * TPCH_P_NAME_USAGE);
+    for (int64_t i = 0; i < count; i++) {
+        tpch_part_name_ids(&r, ids5 + i * 5);
+        tpch_rng_row_finished(&r);
+    }
+}
+
+/* pool build with a sentence-start index (offset, draw ordinal) for the
+ * table-pinning harness (tools/check_textpool.py) */
+static int64_t* g_sent_off = NULL;
+static int64_t* g_sent_draw = NULL;
+static int64_t g_sent_n = 0;
+
+EXPORT int64_t tpch_text_pool_sentences(int64_t** offs, int64_t** draw_ords)
+{
+    if (!g_sent_off) {
+        int64_t cap = 16 * 1024 * 1024;
+        g_sent_off = (int64_t*)malloc(cap * 8);
+        g_sent_draw = (int64_t*)malloc(cap * 8);
+        char* buf = (char*)malloc(TPCH_TEXT_POOL_SIZE);
+        tpch_sb b = { buf, 0, TPCH_TEXT_POOL_SIZE };
+        tpch_rng r;
+        tpch_rng_init(&r, TPCH_SEED_TEXT, 1 << 30);
+        while (b.len < TPCH_TEXT_POOL_SIZE) {
+            if (g_sent_n < cap) {
+                g_sent_off[g_sent_n] = b.len;
+                g_sent_draw[g_sent_n] = r.used;
+                g_sent_n++;
+            }
+            tpch_text_sentence(&b, &r);
+        }
+        free(buf);
+    }
+    *offs = g_sent_off;
+    *draw_ords = g_sent_draw;
+    return g_sent_n;
+}
