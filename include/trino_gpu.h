@@ -277,6 +277,52 @@ tg_status tg_tpch_gen_part(tg_session*, double scale_factor,
     int64_t part_start, int64_t part_count,
     int64_t* dev_partkey, int16_t* dev_type_id /* 0..149 (SMALLINT: 150 ids
     overflow signed TINYINT); PROMO = >=125 */);
+
+/* round 2: extended tables (streams pinned in oracle/tpch_text.h against
+ * the reference's own sf0.01 dataset + SF1 answer fixtures).
+ * part2: type 0..149, brand 11..55, container 0..39, name_ids = 5 color
+ * ids/row (dists.dss colors, alphabetical). partsupp: 4 rows per part in
+ * dbgen bridge order. orders3 adds derived o_orderstatus (0=F 1=O 2=P),
+ * o_totalprice cents and o_comment pool slices. */
+tg_status tg_tpch_gen_part2(tg_session*, double scale_factor,
+    int64_t part_start, int64_t part_count, int64_t* d_partkey,
+    int16_t* d_type, uint8_t* d_brand, int32_t* d_size, uint8_t* d_container,
+    uint8_t* d_name_ids, int64_t* d_retail_cents);
+tg_status tg_tpch_gen_partsupp(tg_session*, double scale_factor,
+    int64_t part_start, int64_t part_count, int64_t* d_partkey,
+    int64_t* d_suppkey, int32_t* d_availqty, int64_t* d_supplycost_cents);
+tg_status tg_tpch_gen_supplier2(tg_session*, double scale_factor,
+    int64_t supp_start, int64_t supp_count, int64_t* d_suppkey,
+    uint8_t* d_nationkey, int64_t* d_acctbal_cents);
+tg_status tg_tpch_gen_orders3(tg_session*, double scale_factor,
+    int64_t order_start, int64_t order_count, int64_t* d_orderkey,
+    int64_t* d_custkey, int32_t* d_orderdate, uint8_t* d_priority,
+    uint8_t* d_orderstatus, int64_t* d_totalprice_cents,
+    int64_t* d_cmnt_off, int32_t* d_cmnt_len);
+/* 300 MiB dbgen text pool, device-resident (built host-side, cached) */
+tg_status tg_tpch_pool(tg_session*, const uint8_t** d_pool);
+/* LIKE '%a%b%' (no '_') over pool slices / var-width columns -> 0/1 flags */
+tg_status tg_pool_like_flags(tg_session*, const int64_t* d_offs,
+    const int32_t* d_lens, int64_t n, const char* pattern, uint8_t* d_flags);
+tg_status tg_varchar_like_flags(tg_session*, const uint8_t* d_bytes,
+    const int32_t* d_offsets, int64_t n, const char* pattern, uint8_t* d_flags);
+/* s_comment var-width column incl. the BBB "Customer ...Complaints/
+ * Recommends" overlay (10 rows per 10,000 suppliers) */
+tg_status tg_tpch_gen_supplier_comments(tg_session*, double scale_factor,
+    int64_t supp_start, int64_t supp_count, int32_t** d_offsets_out,
+    uint8_t** d_bytes_out);
+/* host-side string materialization for final output assembly (few rows) */
+tg_status tg_tpch_supplier_strings(double sf, const int64_t* keys, int32_t n,
+    int32_t stride, char* name, char* address, char* phone, char* comment,
+    int64_t* acctbal_cents, int32_t* nationkey);
+tg_status tg_tpch_customer_strings(double sf, const int64_t* keys, int32_t n,
+    int32_t stride, char* name, char* address, char* phone, char* comment,
+    int64_t* acctbal_cents, int32_t* nationkey);
+tg_status tg_tpch_part_strings(double sf, const int64_t* keys, int32_t n,
+    int32_t stride, char* name, char* mfgr, char* brand, char* type,
+    char* container);
+tg_status tg_tpch_nation_name(int32_t nationkey, char out[32]);
+int32_t tg_tpch_nation_region(int32_t nationkey);
 /* coarse device-memory accounting (LocalMemoryContext analog): bytes ever
  * pooled and bytes currently cached; live = total - cached */
 tg_status tg_session_memory(tg_session*, int64_t* total_bytes, int64_t* cached_bytes);

@@ -316,7 +316,7 @@ static inline void tpch_text_pool_build(char* buf, int64_t size)
 /* ---- per-row value generators over independent streams ---- */
 
 /* comment slice: 2 draws; off in [0, pool - maxLen], len in [.4avg,1.6avg] */
-static inline void tpch_text_slice(tpch_rng* r, int avg,
+TPCH_HD static inline void tpch_text_slice(tpch_rng* r, int avg,
                                    int64_t* off, int32_t* len)
 {
     int lo = (int)(avg * 0.4), hi = (int)(avg * 1.6);
@@ -325,7 +325,7 @@ static inline void tpch_text_slice(tpch_rng* r, int avg,
 }
 
 /* v_str (addresses): len then one draw per 5 chars; pinned decode */
-static inline int tpch_vstr(tpch_rng* r, int avg, char* out /* >= 1.6*avg */)
+TPCH_HD static inline int tpch_vstr(tpch_rng* r, int avg, char* out /* >= 1.6*avg */)
 {
     int lo = (int)(avg * 0.4), hi = (int)(avg * 1.6);
     int len = (int)tpch_rng_int(r, lo, hi);
@@ -339,7 +339,7 @@ static inline int tpch_vstr(tpch_rng* r, int avg, char* out /* >= 1.6*avg */)
 }
 
 /* phone: 3 draws, "CC-LLL-LLL-LLLL" with CC = nationkey + 10 */
-static inline void tpch_phone(tpch_rng* r, int nationkey, char out[16])
+TPCH_HD static inline void tpch_phone(tpch_rng* r, int nationkey, char out[16])
 {
     int c = 10 + nationkey;
     int l1 = (int)tpch_rng_int(r, 100, 999);

@@ -327,3 +327,465 @@ extern "C" tg_status tg_tpch_gen_part(tg_session* s, double sf,
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     return TG_OK;
 }
+
+/* ================= round 2: extended tables + text columns =============== */
+#include "../../oracle/tpch_text.h"
+
+__global__ void k_gen_part2(int64_t part_start, int64_t part_count,
+                            int64_t* partkey, int16_t* type_id,
+                            uint8_t* brand, int32_t* size, uint8_t* container,
+                            uint8_t* name_ids /* 5 per row */,
+                            int64_t* retail_cents)
+{
+    int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t n_groups = (part_count + GOT - 1) / GOT;
+    if (g >= n_groups) return;
+    int64_t first = part_start + g * GOT;
+    int64_t cnt = min((int64_t)GOT, part_start + part_count - first);
+    tpch_rng ty, mf, br, sz, cn, nm;
+    tpch_rng_init(&ty, TPCH_SEED_P_TYPE, 1); tpch_rng_skip(&ty, first - 1);
+    tpch_rng_init(&mf, TPCH_SEED_P_MFG, 1);  tpch_rng_skip(&mf, first - 1);
+    tpch_rng_init(&br, TPCH_SEED_P_BRND, 1); tpch_rng_skip(&br, first - 1);
+    tpch_rng_init(&sz, TPCH_SEED_P_SIZE, 1); tpch_rng_skip(&sz, first - 1);
+    tpch_rng_init(&cn, TPCH_SEED_P_CNTR, 1); tpch_rng_skip(&cn, first - 1);
+    tpch_rng_init(&nm, TPCH_SEED_P_NAME, TPCH_P_NAME_USAGE);
+    tpch_rng_skip(&nm, (first - 1) * TPCH_P_NAME_USAGE);
+    for (int64_t i = 0; i < cnt; i++) {
+        int64_t at = first - part_start + i;
+        int64_t p = first + i;
+        if (partkey) partkey[at] = p;
+        int64_t tid = tpch_rng_int(&ty, 1, 150) - 1;
+        if (type_id) type_id[at] = (int16_t)tid;
+        int64_t m = tpch_rng_int(&mf, 1, 5);
+        int64_t b = m * 10 + tpch_rng_int(&br, 1, 5);
+        if (brand) brand[at] = (uint8_t)b;
+        if (size) size[at] = (int32_t)tpch_rng_int(&sz, 1, 50);
+        if (container) container[at] = (uint8_t)(tpch_rng_int(&cn, 1, 40) - 1);
+        if (name_ids) tpch_part_name_ids(&nm, name_ids + at * 5);
+        else { tpch_rng_skip(&nm, 5); nm.used += 5; }
+        tpch_rng_row_finished(&nm);
+        if (retail_cents) retail_cents[at] = tpch_part_price_cents(p);
+    }
+}
+
+extern "C" tg_status tg_tpch_gen_part2(tg_session* s, double sf,
+    int64_t part_start, int64_t part_count, int64_t* d_partkey,
+    int16_t* d_type, uint8_t* d_brand, int32_t* d_size, uint8_t* d_container,
+    uint8_t* d_name_ids, int64_t* d_retail_cents)
+{
+    (void)sf;
+    int64_t n_groups = (part_count + GOT - 1) / GOT;
+    int grid = (int)((n_groups + TG_BLOCK - 1) / TG_BLOCK);
+    hipLaunchKernelGGL(k_gen_part2, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                       part_start, part_count, d_partkey, d_type, d_brand,
+                       d_size, d_container, d_name_ids, d_retail_cents);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}
+
+/* partsupp: 4 rows per part in dbgen bridge order */
+__global__ void k_gen_partsupp(double sf, int64_t part_start, int64_t part_count,
+                               int64_t* partkey, int64_t* suppkey,
+                               int32_t* availqty, int64_t* supplycost_cents)
+{
+    int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t n_groups = (part_count + GOT - 1) / GOT;
+    if (g >= n_groups) return;
+    int64_t first = part_start + g * GOT;
+    int64_t cnt = min((int64_t)GOT, part_start + part_count - first);
+    int64_t S = (int64_t)(10000 * sf);
+    tpch_rng qt, sc;
+    tpch_rng_init(&qt, TPCH_SEED_PS_QTY, 4);  tpch_rng_skip(&qt, (first - 1) * 4);
+    tpch_rng_init(&sc, TPCH_SEED_PS_SCST, 4); tpch_rng_skip(&sc, (first - 1) * 4);
+    for (int64_t i = 0; i < cnt; i++) {
+        int64_t p = first + i;
+        for (int j = 0; j < 4; j++) {
+            int64_t at = (p - part_start) * 4 + j;
+            if (partkey) partkey[at] = p;
+            if (suppkey) suppkey[at] = (p + j * (S / 4 + (p - 1) / S)) % S + 1;
+            int64_t q = tpch_rng_int(&qt, 1, 9999);
+            if (availqty) availqty[at] = (int32_t)q;
+            int64_t c = tpch_rng_int(&sc, 100, 100000);
+            if (supplycost_cents) supplycost_cents[at] = c;
+        }
+        tpch_rng_row_finished(&qt);
+        tpch_rng_row_finished(&sc);
+    }
+}
+
+extern "C" tg_status tg_tpch_gen_partsupp(tg_session* s, double sf,
+    int64_t part_start, int64_t part_count, int64_t* d_partkey,
+    int64_t* d_suppkey, int32_t* d_availqty, int64_t* d_supplycost_cents)
+{
+    int64_t n_groups = (part_count + GOT - 1) / GOT;
+    int grid = (int)((n_groups + TG_BLOCK - 1) / TG_BLOCK);
+    hipLaunchKernelGGL(k_gen_partsupp, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                       sf, part_start, part_count, d_partkey, d_suppkey,
+                       d_availqty, d_supplycost_cents);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}
+
+__global__ void k_gen_supplier2(double sf, int64_t supp_start, int64_t supp_count,
+                                int64_t* suppkey, uint8_t* nationkey,
+                                int64_t* acctbal_cents)
+{
+    int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t n_groups = (supp_count + GOT - 1) / GOT;
+    if (g >= n_groups) return;
+    int64_t first = supp_start + g * GOT;
+    int64_t cnt = min((int64_t)GOT, supp_start + supp_count - first);
+    (void)sf;
+    tpch_rng nk, ab;
+    tpch_rng_init(&nk, TPCH_SEED_S_NKEY, 1); tpch_rng_skip(&nk, first - 1);
+    tpch_rng_init(&ab, TPCH_SEED_S_ABAL, 1); tpch_rng_skip(&ab, first - 1);
+    for (int64_t i = 0; i < cnt; i++) {
+        int64_t at = first - supp_start + i;
+        if (suppkey) suppkey[at] = first + i;
+        int64_t v = tpch_rng_int(&nk, 0, 24);
+        if (nationkey) nationkey[at] = (uint8_t)v;
+        int64_t b = tpch_rng_int(&ab, -99999, 999999);
+        if (acctbal_cents) acctbal_cents[at] = b;
+        tpch_rng_row_finished(&nk);
+        tpch_rng_row_finished(&ab);
+    }
+}
+
+extern "C" tg_status tg_tpch_gen_supplier2(tg_session* s, double sf,
+    int64_t supp_start, int64_t supp_count, int64_t* d_suppkey,
+    uint8_t* d_nationkey, int64_t* d_acctbal_cents)
+{
+    int64_t n_groups = (supp_count + GOT - 1) / GOT;
+    int grid = (int)((n_groups + TG_BLOCK - 1) / TG_BLOCK);
+    hipLaunchKernelGGL(k_gen_supplier2, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                       sf, supp_start, supp_count, d_suppkey, d_nationkey,
+                       d_acctbal_cents);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}
+
+/* orders with derived status (0=F 1=O 2=P), totalprice cents, and o_comment
+ * pool slices. Status/totalprice require walking the order's lines
+ * (dbgen mk_order does the same). */
+__global__ void k_gen_orders3(double sf, int64_t order_start, int64_t order_count,
+                              int64_t* orderkey, int64_t* custkey,
+                              int32_t* orderdate, uint8_t* priority,
+                              uint8_t* orderstatus, int64_t* totalprice_cents,
+                              int64_t* cmnt_off, int32_t* cmnt_len)
+{
+    int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t n_groups = (order_count + GOT - 1) / GOT;
+    if (g >= n_groups) return;
+    int64_t first = order_start + g * GOT;
+    int64_t cnt = min((int64_t)GOT, order_start + order_count - first);
+    tpch_order_streams s;
+    tpch_order_streams_init(&s, sf);
+    tpch_order_streams_seek(&s, first);
+    tpch_rng cm;
+    tpch_rng_init(&cm, TPCH_SEED_O_CMNT, 2);
+    tpch_rng_skip(&cm, (first - 1) * 2);
+    for (int64_t i = 0; i < cnt; i++) {
+        tpch_order_row o;
+        tpch_gen_order(&s, first + i, &o);
+        int64_t at = first - order_start + i;
+        if (orderkey)  orderkey[at] = o.orderkey;
+        if (custkey)   custkey[at] = o.custkey;
+        if (orderdate) orderdate[at] = o.orderdate;
+        if (priority)  priority[at] = o.priority;
+        if (orderstatus || totalprice_cents) {
+            int64_t tp = 0;
+            int nF = 0, nO = 0;
+            for (int j = 0; j < o.line_count; j++) {
+                tpch_lineitem_row l;
+                tpch_gen_line(&s, &o, j, &l);
+                tp += l.tp_cents;
+                if (l.linestatus) nO++; else nF++;
+            }
+            if (totalprice_cents) totalprice_cents[at] = tp;
+            if (orderstatus)
+                orderstatus[at] = (nO == 0) ? 0 : (nF == 0 ? 1 : 2);
+        }
+        if (cmnt_off || cmnt_len) {
+            int64_t off; int32_t len;
+            tpch_text_slice(&cm, TPCH_CMNT_AVG_O, &off, &len);
+            if (cmnt_off) cmnt_off[at] = off;
+            if (cmnt_len) cmnt_len[at] = len;
+        }
+        tpch_rng_row_finished(&cm);
+        tpch_order_row_finished(&s);
+    }
+}
+
+extern "C" tg_status tg_tpch_gen_orders3(tg_session* s, double sf,
+    int64_t order_start, int64_t order_count,
+    int64_t* d_orderkey, int64_t* d_custkey, int32_t* d_orderdate,
+    uint8_t* d_priority, uint8_t* d_orderstatus, int64_t* d_totalprice_cents,
+    int64_t* d_cmnt_off, int32_t* d_cmnt_len)
+{
+    int64_t n_groups = (order_count + GOT - 1) / GOT;
+    int grid = (int)((n_groups + TG_BLOCK - 1) / TG_BLOCK);
+    hipLaunchKernelGGL(k_gen_orders3, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                       sf, order_start, order_count, d_orderkey, d_custkey,
+                       d_orderdate, d_priority, d_orderstatus,
+                       d_totalprice_cents, d_cmnt_off, d_cmnt_len);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}
+
+/* ---- text pool on device (built host-side once, cached per-process) ---- */
+static uint8_t* g_d_pool = nullptr;
+static char* g_h_pool = nullptr;
+
+const char* tg_tpch_host_pool(void)
+{
+    if (!g_h_pool) {
+        g_h_pool = (char*)malloc(TPCH_TEXT_POOL_SIZE);
+        tpch_text_pool_build(g_h_pool, TPCH_TEXT_POOL_SIZE);
+    }
+    return g_h_pool;
+}
+
+extern "C" tg_status tg_tpch_pool(tg_session* s, const uint8_t** d_pool)
+{
+    if (!g_d_pool) {
+        const char* h = tg_tpch_host_pool();
+        TG_HIP_CHECK(hipMalloc(&g_d_pool, TPCH_TEXT_POOL_SIZE));
+        TG_HIP_CHECK(hipMemcpyAsync(g_d_pool, h, TPCH_TEXT_POOL_SIZE,
+                                    hipMemcpyHostToDevice, s->stream));
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    }
+    *d_pool = g_d_pool;
+    return TG_OK;
+}
+
+/* LIKE '%seg1%seg2%...%' over pool slices: per row, find each literal
+ * segment in order within pool[off, off+len). anchor_start/end pin the
+ * first/last segment (LIKE without leading/trailing %%).
+ * Mirrors io.trino.type.LikeFunctions / LikePattern matching semantics for
+ * patterns without '_' wildcards. */
+#define LIKE_MAX_SEGS 4
+#define LIKE_MAX_PAT 64
+struct LikePat {
+    char bytes[LIKE_MAX_PAT];
+    int seg_off[LIKE_MAX_SEGS], seg_len[LIKE_MAX_SEGS];
+    int n_segs, anchor_start, anchor_end;
+};
+
+__device__ static bool like_match(const uint8_t* txt, int len, const LikePat& p)
+{
+    int pos = 0;
+    for (int k = 0; k < p.n_segs; k++) {
+        int sl = p.seg_len[k];
+        const char* seg = p.bytes + p.seg_off[k];
+        if (k == 0 && p.anchor_start) {
+            if (len < sl) return false;
+            for (int i = 0; i < sl; i++) if (txt[i] != (uint8_t)seg[i]) return false;
+            pos = sl;
+            continue;
+        }
+        if (k == p.n_segs - 1 && p.anchor_end) {
+            if (len - pos < sl) return false;
+            for (int i = 0; i < sl; i++)
+                if (txt[len - sl + i] != (uint8_t)seg[i]) return false;
+            pos = len;
+            continue;
+        }
+        bool found = false;
+        for (int i = pos; i + sl <= len && !found; i++) {
+            bool eq = true;
+            for (int j = 0; j < sl && eq; j++) eq = txt[i + j] == (uint8_t)seg[j];
+            if (eq) { pos = i + sl; found = true; }
+        }
+        if (!found) return false;
+    }
+    return true;
+}
+
+__global__ void k_pool_like(const uint8_t* __restrict__ pool,
+                            const int64_t* __restrict__ offs,
+                            const int32_t* __restrict__ lens, int64_t n,
+                            LikePat p, uint8_t* __restrict__ flags)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride)
+        flags[i] = like_match(pool + offs[i], lens[i], p) ? 1 : 0;
+}
+
+__global__ void k_var_like(const uint8_t* __restrict__ bytes,
+                           const int32_t* __restrict__ offsets, int64_t n,
+                           LikePat p, uint8_t* __restrict__ flags)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride)
+        flags[i] = like_match(bytes + offsets[i], offsets[i + 1] - offsets[i], p)
+                       ? 1 : 0;
+}
+
+static tg_status parse_like(const char* pattern, LikePat* p)
+{
+    memset(p, 0, sizeof(*p));
+    size_t L = strlen(pattern);
+    if (L >= LIKE_MAX_PAT) { TG_SET_ERR("pattern too long"); return TG_ERR_UNSUPPORTED; }
+    p->anchor_start = L > 0 && pattern[0] != '%';
+    p->anchor_end = L > 0 && pattern[L - 1] != '%';
+    int n = 0, bo = 0;
+    const char* q = pattern;
+    while (*q) {
+        while (*q == '%') q++;
+        if (!*q) break;
+        const char* e = q;
+        while (*e && *e != '%') {
+            if (*e == '_') { TG_SET_ERR("'_' wildcard unsupported"); return TG_ERR_UNSUPPORTED; }
+            e++;
+        }
+        if (n >= LIKE_MAX_SEGS) { TG_SET_ERR("too many %% segments"); return TG_ERR_UNSUPPORTED; }
+        p->seg_off[n] = bo; p->seg_len[n] = (int)(e - q);
+        memcpy(p->bytes + bo, q, e - q);
+        bo += (int)(e - q);
+        n++;
+        q = e;
+    }
+    p->n_segs = n;
+    return TG_OK;
+}
+
+/* flags[i] = (pool slice i LIKE pattern); slices device-resident */
+extern "C" tg_status tg_pool_like_flags(tg_session* s, const int64_t* d_offs,
+    const int32_t* d_lens, int64_t n, const char* pattern, uint8_t* d_flags)
+{
+    const uint8_t* d_pool = nullptr;
+    tg_status st = tg_tpch_pool(s, &d_pool);
+    if (st != TG_OK) return st;
+    LikePat p;
+    st = parse_like(pattern, &p);
+    if (st != TG_OK) return st;
+    hipLaunchKernelGGL(k_pool_like, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0,
+                       s->stream, d_pool, d_offs, d_lens, n, p, d_flags);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}
+
+/* generic VARCHAR LIKE on a device var-width column (bytes + offsets[n+1]) */
+extern "C" tg_status tg_varchar_like_flags(tg_session* s, const uint8_t* d_bytes,
+    const int32_t* d_offsets, int64_t n, const char* pattern, uint8_t* d_flags)
+{
+    LikePat p;
+    tg_status st = parse_like(pattern, &p);
+    if (st != TG_OK) return st;
+    hipLaunchKernelGGL(k_var_like, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0,
+                       s->stream, d_bytes, d_offsets, n, p, d_flags);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}
+
+/* supplier comments as a device var-width column, WITH the dbgen BBB
+ * overlay ("Customer ...Complaints"/"Recommends" written over 10 rows per
+ * 10,000; selection/type/junk/offset each one independent stream draw/row).
+ * Overlay semantics pinned by the SF1 Q16 answer fixture. */
+__global__ void k_s_cmnt_lens(int64_t supp_start, int64_t n, int32_t* lens)
+{
+    int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t n_groups = (n + GOT - 1) / GOT;
+    if (g >= n_groups) return;
+    int64_t first = supp_start + g * GOT;
+    int64_t cnt = min((int64_t)GOT, supp_start + n - first);
+    tpch_rng cm;
+    tpch_rng_init(&cm, TPCH_SEED_S_CMNT, 2);
+    tpch_rng_skip(&cm, (first - 1) * 2);
+    for (int64_t i = 0; i < cnt; i++) {
+        int64_t off; int32_t len;
+        tpch_text_slice(&cm, TPCH_CMNT_AVG_S, &off, &len);
+        lens[first - supp_start + i] = len;
+        tpch_rng_row_finished(&cm);
+    }
+}
+
+__global__ void k_s_cmnt_bytes(const uint8_t* __restrict__ pool,
+                               int64_t supp_start, int64_t n,
+                               const int32_t* __restrict__ offsets,
+                               uint8_t* __restrict__ out)
+{
+    int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t n_groups = (n + GOT - 1) / GOT;
+    if (g >= n_groups) return;
+    int64_t first = supp_start + g * GOT;
+    int64_t cnt = min((int64_t)GOT, supp_start + n - first);
+    tpch_rng cm, bs, bj, bo, bt;
+    tpch_rng_init(&cm, TPCH_SEED_S_CMNT, 2);
+    tpch_rng_skip(&cm, (first - 1) * 2);
+    tpch_rng_init(&bs, TPCH_SEED_BBB_CMNT, 1); tpch_rng_skip(&bs, first - 1);
+    tpch_rng_init(&bj, TPCH_SEED_BBB_JNK, 1);  tpch_rng_skip(&bj, first - 1);
+    tpch_rng_init(&bo, TPCH_SEED_BBB_OFF, 1);  tpch_rng_skip(&bo, first - 1);
+    tpch_rng_init(&bt, TPCH_SEED_BBB_TYPE, 1); tpch_rng_skip(&bt, first - 1);
+    const char* BASE = "Customer ";
+    const char* GOOD = "Recommends";
+    const char* BAD = "Complaints";
+    for (int64_t i = 0; i < cnt; i++) {
+        int64_t off; int32_t len;
+        tpch_text_slice(&cm, TPCH_CMNT_AVG_S, &off, &len);
+        int64_t at = first - supp_start + i;
+        uint8_t* dst = out + offsets[at];
+        for (int32_t b = 0; b < len; b++) dst[b] = pool[off + b];
+        int64_t sel = tpch_rng_int(&bs, 1, 10000);
+        int64_t type = tpch_rng_int(&bt, 0, 100);
+        int64_t junk = tpch_rng_int(&bj, 0, len - 19);
+        int64_t boff = tpch_rng_int(&bo, 0, len - (19 + junk));
+        if (sel <= 10) {
+            const char* tail = (type < 50) ? BAD : GOOD;
+            for (int b = 0; b < 9; b++) dst[boff + b] = (uint8_t)BASE[b];
+            for (int b = 0; b < 10; b++) dst[boff + 9 + junk + b] = (uint8_t)tail[b];
+        }
+        tpch_rng_row_finished(&cm);
+        tpch_rng_row_finished(&bs);
+        tpch_rng_row_finished(&bj);
+        tpch_rng_row_finished(&bo);
+        tpch_rng_row_finished(&bt);
+    }
+}
+
+tg_status run_scan_i32(tg_session* s, int32_t* d_arr, int64_t n, int32_t* d_total);
+
+extern "C" tg_status tg_tpch_gen_supplier_comments(tg_session* s, double sf,
+    int64_t supp_start, int64_t supp_count,
+    int32_t** d_offsets_out /* [n+1], pool-allocated */,
+    uint8_t** d_bytes_out)
+{
+    (void)sf;
+    const uint8_t* d_pool = nullptr;
+    tg_status st = tg_tpch_pool(s, &d_pool);
+    if (st != TG_OK) return st;
+    int32_t* d_off = nullptr;
+    TG_POOL_ALLOC(s, &d_off, (supp_count + 1) * 4);
+    int64_t n_groups = (supp_count + GOT - 1) / GOT;
+    int grid = (int)((n_groups + TG_BLOCK - 1) / TG_BLOCK);
+    hipLaunchKernelGGL(k_s_cmnt_lens, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                       supp_start, supp_count, d_off);
+    TG_HIP_CHECK(hipGetLastError());
+    int32_t* d_total = nullptr;
+    TG_POOL_ALLOC(s, &d_total, 4);
+    st = run_scan_i32(s, d_off, supp_count, d_total);
+    if (st != TG_OK) return st;
+    int32_t total = 0;
+    TG_HIP_CHECK(hipMemcpyAsync(&total, d_total, 4, hipMemcpyDeviceToHost, s->stream));
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    TG_HIP_CHECK(hipMemcpyAsync(d_off + supp_count, d_total, 4,
+                                hipMemcpyDeviceToDevice, s->stream));
+    uint8_t* d_bytes = nullptr;
+    TG_POOL_ALLOC(s, &d_bytes, total ? total : 1);
+    hipLaunchKernelGGL(k_s_cmnt_bytes, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                       d_pool, supp_start, supp_count, d_off, d_bytes);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    tg_pool_free(s, d_total);
+    *d_offsets_out = d_off;
+    *d_bytes_out = d_bytes;
+    return TG_OK;
+}
