@@ -326,3 +326,105 @@ def q1_exact(cols, cutoff=10471, threads=1):
                               cols["returnflag"].ctypes.data, cols["linestatus"].ctypes.data,
                               cutoff, threads, ctypes.byref(r))
     return r, elapsed
+
+
+# ---- round 2: extended tables + text (tpch_text.h restatement) ----
+_lib.tpch_gen_part2.restype = None
+_lib.tpch_gen_part2.argtypes = [ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 7
+_lib.tpch_gen_partsupp2.restype = None
+_lib.tpch_gen_partsupp2.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
+_lib.tpch_gen_supplier2.restype = None
+_lib.tpch_gen_supplier2.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 3
+_lib.tpch_gen_orders3.restype = None
+_lib.tpch_gen_orders3.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 9
+_lib.tpch_text_pool.restype = ctypes.c_void_p
+_lib.tpch_gen_supplier_comments.restype = None
+_lib.tpch_gen_supplier_comments.argtypes = [ctypes.c_int64, ctypes.c_int64,
+                                            ctypes.c_void_p, ctypes.c_int32,
+                                            ctypes.c_void_p]
+
+
+def gen_part2(sf, part_start=1, part_count=None):
+    if part_count is None:
+        part_count = int(200_000 * sf)
+    pk = np.empty(part_count, np.int64)
+    ty = np.empty(part_count, np.int16)
+    br = np.empty(part_count, np.uint8)
+    sz = np.empty(part_count, np.int32)
+    cn = np.empty(part_count, np.uint8)
+    nm = np.empty(part_count * 5, np.uint8)
+    rp = np.empty(part_count, np.int64)
+    _lib.tpch_gen_part2(part_start, part_count, _ptr(pk), _ptr(ty), _ptr(br),
+                        _ptr(sz), _ptr(cn), _ptr(nm), _ptr(rp))
+    return {"partkey": pk, "type_id": ty, "brand": br, "size": sz,
+            "container": cn, "name_ids": nm.reshape(part_count, 5),
+            "retail_cents": rp}
+
+
+def gen_partsupp(sf, part_start=1, part_count=None):
+    if part_count is None:
+        part_count = int(200_000 * sf)
+    n = part_count * 4
+    pk = np.empty(n, np.int64)
+    sk = np.empty(n, np.int64)
+    aq = np.empty(n, np.int32)
+    sc = np.empty(n, np.int64)
+    _lib.tpch_gen_partsupp2(sf, part_start, part_count, _ptr(pk), _ptr(sk),
+                            _ptr(aq), _ptr(sc))
+    return {"partkey": pk, "suppkey": sk, "availqty": aq,
+            "supplycost_cents": sc}
+
+
+def gen_supplier2(sf, supp_start=1, supp_count=None):
+    if supp_count is None:
+        supp_count = int(10_000 * sf)
+    sk = np.empty(supp_count, np.int64)
+    nk = np.empty(supp_count, np.uint8)
+    ab = np.empty(supp_count, np.int64)
+    _lib.tpch_gen_supplier2(sf, supp_start, supp_count, _ptr(sk), _ptr(nk), _ptr(ab))
+    return {"suppkey": sk, "nationkey": nk, "acctbal_cents": ab}
+
+
+def gen_orders3(sf, order_start=1, order_count=None):
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    ok = np.empty(order_count, np.int64)
+    ck = np.empty(order_count, np.int64)
+    od = np.empty(order_count, np.int32)
+    pri = np.empty(order_count, np.uint8)
+    st = np.empty(order_count, np.uint8)
+    tp = np.empty(order_count, np.int64)
+    co = np.empty(order_count, np.int64)
+    cl = np.empty(order_count, np.int32)
+    cln = np.empty(order_count, np.int32)
+    _lib.tpch_gen_orders3(sf, order_start, order_count, _ptr(ok), _ptr(ck),
+                          _ptr(od), _ptr(pri), _ptr(st), _ptr(tp), _ptr(co),
+                          _ptr(cln), _ptr(cl))
+    return {"orderkey": ok, "custkey": ck, "orderdate": od, "orderpriority": pri,
+            "orderstatus": st, "totalprice_cents": tp, "cmnt_off": co,
+            "cmnt_len": cln, "clerk": cl}
+
+
+def text_pool_bytes():
+    """The full 300 MiB pool as a (read-only) numpy uint8 view."""
+    p = _lib.tpch_text_pool()
+    return np.ctypeslib.as_array((ctypes.c_uint8 * (300 * 1024 * 1024)).from_address(p))
+
+
+def gen_supplier_comments(supp_start, count):
+    buf = np.zeros((count, 101), np.uint8)
+    lens = np.zeros(count, np.int32)
+    _lib.tpch_gen_supplier_comments(supp_start, count, _ptr(buf), 101, _ptr(lens))
+    return [bytes(buf[i, :lens[i]]).decode("latin1") for i in range(count)]
+
+
+NATIONS = ["ALGERIA", "ARGENTINA", "BRAZIL", "CANADA", "EGYPT", "ETHIOPIA",
+           "FRANCE", "GERMANY", "INDIA", "INDONESIA", "IRAN", "IRAQ",
+           "JAPAN", "JORDAN", "KENYA", "MOROCCO", "MOZAMBIQUE", "PERU",
+           "CHINA", "ROMANIA", "SAUDI ARABIA", "VIETNAM", "RUSSIA",
+           "UNITED KINGDOM", "UNITED STATES"]
+NATION_REGION = [0, 1, 1, 1, 4, 0, 3, 3, 2, 2, 4, 4, 2, 4, 0, 0, 0, 1, 2, 3,
+                 4, 2, 3, 3, 1]
+REGIONS = ["AFRICA", "AMERICA", "ASIA", "EUROPE", "MIDDLE EAST"]
+COLOR_GREEN = 33
+COLOR_FOREST = 28

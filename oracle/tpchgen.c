@@ -234,3 +234,155 @@ EXPORT int64_t tpch_text_pool_sentences(int64_t** offs, int64_t** draw_ords)
     *draw_ords = g_sent_draw;
     return g_sent_n;
 }
+
+/* extended tables (mirror trino_amd/csrc/tpchgen.hip round-2 kernels) */
+EXPORT void tpch_gen_part2(int64_t part_start, int64_t count,
+    int64_t* partkey, int16_t* type_id, uint8_t* brand, int32_t* size,
+    uint8_t* container, uint8_t* name_ids, int64_t* retail_cents)
+{
+    tpch_rng ty, mf, br, sz, cn, nm;
+    tpch_rng_init(&ty, TPCH_SEED_P_TYPE, 1); tpch_rng_skip(&ty, part_start - 1);
+    tpch_rng_init(&mf, TPCH_SEED_P_MFG, 1);  tpch_rng_skip(&mf, part_start - 1);
+    tpch_rng_init(&br, TPCH_SEED_P_BRND, 1); tpch_rng_skip(&br, part_start - 1);
+    tpch_rng_init(&sz, TPCH_SEED_P_SIZE, 1); tpch_rng_skip(&sz, part_start - 1);
+    tpch_rng_init(&cn, TPCH_SEED_P_CNTR, 1); tpch_rng_skip(&cn, part_start - 1);
+    tpch_rng_init(&nm, TPCH_SEED_P_NAME, TPCH_P_NAME_USAGE);
+    tpch_rng_skip(&nm, (part_start - 1) * TPCH_P_NAME_USAGE);
+    for (int64_t i = 0; i < count; i++) {
+        int64_t p = part_start + i;
+        if (partkey) partkey[i] = p;
+        int64_t tid = tpch_rng_int(&ty, 1, 150) - 1;
+        if (type_id) type_id[i] = (int16_t)tid;
+        int64_t m = tpch_rng_int(&mf, 1, 5);
+        int64_t b = m * 10 + tpch_rng_int(&br, 1, 5);
+        if (brand) brand[i] = (uint8_t)b;
+        if (size) size[i] = (int32_t)tpch_rng_int(&sz, 1, 50);
+        if (container) container[i] = (uint8_t)(tpch_rng_int(&cn, 1, 40) - 1);
+        if (name_ids) tpch_part_name_ids(&nm, name_ids + i * 5);
+        tpch_rng_row_finished(&nm);
+        if (retail_cents) retail_cents[i] = tpch_part_price_cents(p);
+    }
+}
+
+EXPORT void tpch_gen_partsupp2(double sf, int64_t part_start, int64_t count,
+    int64_t* partkey, int64_t* suppkey, int32_t* availqty, int64_t* supplycost_cents)
+{
+    int64_t S = (int64_t)(10000 * sf);
+    tpch_rng qt, sc;
+    tpch_rng_init(&qt, TPCH_SEED_PS_QTY, 4);  tpch_rng_skip(&qt, (part_start - 1) * 4);
+    tpch_rng_init(&sc, TPCH_SEED_PS_SCST, 4); tpch_rng_skip(&sc, (part_start - 1) * 4);
+    for (int64_t i = 0; i < count; i++) {
+        int64_t p = part_start + i;
+        for (int j = 0; j < 4; j++) {
+            int64_t at = i * 4 + j;
+            if (partkey) partkey[at] = p;
+            if (suppkey) suppkey[at] = (p + j * (S / 4 + (p - 1) / S)) % S + 1;
+            int64_t q = tpch_rng_int(&qt, 1, 9999);
+            if (availqty) availqty[at] = (int32_t)q;
+            int64_t c = tpch_rng_int(&sc, 100, 100000);
+            if (supplycost_cents) supplycost_cents[at] = c;
+        }
+        tpch_rng_row_finished(&qt);
+        tpch_rng_row_finished(&sc);
+    }
+}
+
+EXPORT void tpch_gen_supplier2(double sf, int64_t supp_start, int64_t count,
+    int64_t* suppkey, uint8_t* nationkey, int64_t* acctbal_cents)
+{
+    (void)sf;
+    tpch_rng nk, ab;
+    tpch_rng_init(&nk, TPCH_SEED_S_NKEY, 1); tpch_rng_skip(&nk, supp_start - 1);
+    tpch_rng_init(&ab, TPCH_SEED_S_ABAL, 1); tpch_rng_skip(&ab, supp_start - 1);
+    for (int64_t i = 0; i < count; i++) {
+        if (suppkey) suppkey[i] = supp_start + i;
+        int64_t v = tpch_rng_int(&nk, 0, 24);
+        if (nationkey) nationkey[i] = (uint8_t)v;
+        int64_t b = tpch_rng_int(&ab, -99999, 999999);
+        if (acctbal_cents) acctbal_cents[i] = b;
+        tpch_rng_row_finished(&nk);
+        tpch_rng_row_finished(&ab);
+    }
+}
+
+EXPORT void tpch_gen_orders3(double sf, int64_t order_start, int64_t count,
+    int64_t* orderkey, int64_t* custkey, int32_t* orderdate, uint8_t* priority,
+    uint8_t* orderstatus, int64_t* totalprice_cents,
+    int64_t* cmnt_off, int32_t* cmnt_len, int32_t* clerk)
+{
+    tpch_order_streams s;
+    tpch_order_streams_init(&s, sf);
+    tpch_order_streams_seek(&s, order_start);
+    tpch_rng cm, cl;
+    tpch_rng_init(&cm, TPCH_SEED_O_CMNT, 2);
+    tpch_rng_skip(&cm, (order_start - 1) * 2);
+    tpch_rng_init(&cl, TPCH_SEED_O_CLRK, 1);
+    tpch_rng_skip(&cl, order_start - 1);
+    int64_t clerks = (int64_t)(1000 * sf);
+    if (clerks < 1000) clerks = 1000;
+    for (int64_t i = 0; i < count; i++) {
+        tpch_order_row o;
+        tpch_gen_order(&s, order_start + i, &o);
+        if (orderkey)  orderkey[i] = o.orderkey;
+        if (custkey)   custkey[i] = o.custkey;
+        if (orderdate) orderdate[i] = o.orderdate;
+        if (priority)  priority[i] = o.priority;
+        if (orderstatus || totalprice_cents) {
+            int64_t tp = 0;
+            int nF = 0, nO = 0;
+            for (int j = 0; j < o.line_count; j++) {
+                tpch_lineitem_row l;
+                tpch_gen_line(&s, &o, j, &l);
+                tp += l.tp_cents;
+                if (l.linestatus) nO++; else nF++;
+            }
+            if (totalprice_cents) totalprice_cents[i] = tp;
+            if (orderstatus) orderstatus[i] = (nO == 0) ? 0 : (nF == 0 ? 1 : 2);
+        }
+        {
+            int64_t off; int32_t len;
+            tpch_text_slice(&cm, TPCH_CMNT_AVG_O, &off, &len);
+            if (cmnt_off) cmnt_off[i] = off;
+            if (cmnt_len) cmnt_len[i] = len;
+        }
+        if (clerk) clerk[i] = (int32_t)tpch_rng_int(&cl, 1, clerks);
+        tpch_rng_row_finished(&cm);
+        tpch_rng_row_finished(&cl);
+        tpch_order_row_finished(&s);
+    }
+}
+
+/* s_comment with BBB overlay; out strides of `stride` bytes, lens filled */
+EXPORT void tpch_gen_supplier_comments(int64_t supp_start, int64_t count,
+    char* out, int32_t stride, int32_t* lens)
+{
+    const char* pool = tpch_text_pool();
+    tpch_rng cm, bs, bj, bo, bt;
+    tpch_rng_init(&cm, TPCH_SEED_S_CMNT, 2);
+    tpch_rng_skip(&cm, (supp_start - 1) * 2);
+    tpch_rng_init(&bs, TPCH_SEED_BBB_CMNT, 1); tpch_rng_skip(&bs, supp_start - 1);
+    tpch_rng_init(&bj, TPCH_SEED_BBB_JNK, 1);  tpch_rng_skip(&bj, supp_start - 1);
+    tpch_rng_init(&bo, TPCH_SEED_BBB_OFF, 1);  tpch_rng_skip(&bo, supp_start - 1);
+    tpch_rng_init(&bt, TPCH_SEED_BBB_TYPE, 1); tpch_rng_skip(&bt, supp_start - 1);
+    for (int64_t i = 0; i < count; i++) {
+        int64_t off; int32_t len;
+        tpch_text_slice(&cm, TPCH_CMNT_AVG_S, &off, &len);
+        char* dst = out + i * stride;
+        memcpy(dst, pool + off, len);
+        lens[i] = len;
+        int64_t sel = tpch_rng_int(&bs, 1, 10000);
+        int64_t type = tpch_rng_int(&bt, 0, 100);
+        int64_t junk = tpch_rng_int(&bj, 0, len - 19);
+        int64_t boff = tpch_rng_int(&bo, 0, len - (19 + junk));
+        if (sel <= 10) {
+            memcpy(dst + boff, "Customer ", 9);
+            memcpy(dst + boff + 9 + junk,
+                   (type < 50) ? "Complaints" : "Recommends", 10);
+        }
+        tpch_rng_row_finished(&cm);
+        tpch_rng_row_finished(&bs);
+        tpch_rng_row_finished(&bj);
+        tpch_rng_row_finished(&bo);
+        tpch_rng_row_finished(&bt);
+    }
+}
