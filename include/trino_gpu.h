@@ -157,7 +157,11 @@ typedef enum {
      * order-INDEPENDENT, so the result is deterministic and correctly
      * rounded — stronger than the reference's DoubleSumAggregation, equal
      * to its decimal SUM semantics. */
-    TG_AGG_SUM_F64_EXACT = 5
+    TG_AGG_SUM_F64_EXACT = 5,
+    /* min/max over BIGINT (MinAggregationFunction/MaxAggregationFunction for
+     * BIGINT inputs): null-skipping; state = sign-bit-biased u64 extremes */
+    TG_AGG_MIN_I64 = 6,
+    TG_AGG_MAX_I64 = 7
 } tg_agg_fn;
 typedef struct tg_agg_spec {
     int32_t fn;
@@ -244,6 +248,9 @@ typedef struct tg_tpch_lineitem_cols {
                                mk_order truncation; sums to o_totalprice) */
     int64_t* suppkey;       /* optional (flags bit 5): partsupp-bridge
                                l_suppkey (canonical-row verified) */
+    uint8_t* shipinstruct;  /* optional (flags bit 6): dictionary id 0..3 =
+                               DELIVER IN PERSON,COLLECT COD,TAKE BACK
+                               RETURN,NONE (pinned) */
 } tg_tpch_lineitem_cols;
 
 /* Generate lineitem rows for orders [order_start, order_start+order_count)

@@ -28,7 +28,7 @@ _lib = ctypes.CDLL(_SO)
 _lib.tpch_lineitem_count.restype = ctypes.c_int64
 _lib.tpch_lineitem_count.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64]
 _lib.tpch_gen_lineitem.restype = ctypes.c_int64
-_lib.tpch_gen_lineitem.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 15
+_lib.tpch_gen_lineitem.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 16
 _lib.tpch_gen_orders.restype = ctypes.c_int64
 _lib.tpch_gen_orders.argtypes = [ctypes.c_double, ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 3
 _lib.tpch_gen_orders2.restype = ctypes.c_int64
@@ -124,12 +124,14 @@ def gen_lineitem(sf, order_start=1, order_count=None, columns=None):
     cap = 7 * order_count
     all_cols = ["orderkey", "partkey", "linenumber", "shipdate", "commitdate",
                 "receiptdate", "quantity", "extendedprice", "discount", "tax",
-                "returnflag", "linestatus", "shipmode", "tp_cents", "suppkey"]
+                "returnflag", "linestatus", "shipmode", "tp_cents", "suppkey",
+                "shipinstruct"]
     dtypes = dict(orderkey=np.int64, partkey=np.int64, linenumber=np.int32,
                   shipdate=np.int32, commitdate=np.int32, receiptdate=np.int32,
                   quantity=np.float64, extendedprice=np.float64, discount=np.float64,
                   tax=np.float64, returnflag=np.uint8, linestatus=np.uint8,
-                  shipmode=np.uint8, tp_cents=np.int64, suppkey=np.int64)
+                  shipmode=np.uint8, tp_cents=np.int64, suppkey=np.int64,
+                  shipinstruct=np.uint8)
     want = set(columns) if columns else set(all_cols)
     bufs = {c: (np.empty(cap, dtypes[c]) if c in want else None) for c in all_cols}
     n = _lib.tpch_gen_lineitem(sf, order_start, order_count,

@@ -202,7 +202,7 @@ TPCH_HD static inline int64_t tpch_order_custkey(tpch_rng* ckey_rng, int64_t max
  * these streams, so skipping them is exact. */
 typedef struct {
     tpch_rng odate, lcnt, ckey, opri;
-    tpch_rng qty, dcnt, tax, pkey, ship, cdate, rdate, rflg, smode, skey;
+    tpch_rng qty, dcnt, tax, pkey, ship, cdate, rdate, rflg, smode, skey, sins;
     int64_t max_custkey;
     int64_t max_partkey;
     int64_t n_suppliers;
@@ -224,6 +224,7 @@ TPCH_HD static inline void tpch_order_streams_init(tpch_order_streams* s, double
     tpch_rng_init(&s->rflg,  TPCH_SEED_L_RFLG,  TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_init(&s->smode, TPCH_SEED_L_SMODE, TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_init(&s->skey,  TPCH_SEED_L_SKEY,  TPCH_LINES_PER_ORDER_MAX);
+    tpch_rng_init(&s->sins,  TPCH_SEED_L_SINS,  TPCH_LINES_PER_ORDER_MAX);
     s->max_custkey = (int64_t)(TPCH_CUSTOMER_BASE * sf);
     s->n_suppliers = (int64_t)(10000 * sf);
     s->max_partkey = (int64_t)(TPCH_PART_BASE * sf);
@@ -247,6 +248,7 @@ TPCH_HD static inline void tpch_order_streams_seek(tpch_order_streams* s, int64_
     tpch_rng_skip(&s->rflg,  n * TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_skip(&s->smode, n * TPCH_LINES_PER_ORDER_MAX);
     tpch_rng_skip(&s->skey,  n * TPCH_LINES_PER_ORDER_MAX);
+    tpch_rng_skip(&s->sins,  n * TPCH_LINES_PER_ORDER_MAX);
 }
 
 typedef struct {
@@ -269,6 +271,8 @@ typedef struct {
     uint8_t returnflag;     /* 0=A 1=N 2=R */
     uint8_t linestatus;     /* 0=F 1=O */
     uint8_t shipmode;       /* 0..6 = REG AIR,AIR,RAIL,TRUCK,MAIL,FOB,SHIP */
+    uint8_t shipinstruct;   /* 0..3 = DELIVER IN PERSON,COLLECT COD,
+                                      TAKE BACK RETURN,NONE (pinned order) */
 } tpch_lineitem_row;
 
 typedef struct {
@@ -318,6 +322,7 @@ TPCH_HD static inline void tpch_gen_line(tpch_order_streams* s, const tpch_order
     }
     l->linestatus = (l->shipdate > TPCH_CURRENTDATE_EPOCH) ? 1 /*O*/ : 0 /*F*/;
     l->shipmode = (uint8_t)(tpch_rng_int(&s->smode, 1, 7) - 1);
+    l->shipinstruct = (uint8_t)(tpch_rng_int(&s->sins, 1, 4) - 1);
     int64_t sj = tpch_rng_int(&s->skey, 0, 3);
     l->suppkey = (l->partkey +
                   sj * (s->n_suppliers / 4 + (l->partkey - 1) / s->n_suppliers))
@@ -343,6 +348,7 @@ TPCH_HD static inline void tpch_order_row_finished(tpch_order_streams* s)
     tpch_rng_row_finished(&s->rflg);
     tpch_rng_row_finished(&s->smode);
     tpch_rng_row_finished(&s->skey);
+    tpch_rng_row_finished(&s->sins);
 }
 
 /* double views exactly as io.trino.tpch getDouble (cents/100.0 etc.) */

@@ -35,7 +35,7 @@ EXPORT int64_t tpch_gen_lineitem(double sf, int64_t order_start, int64_t order_c
     int32_t* shipdate, int32_t* commitdate, int32_t* receiptdate,
     double* quantity, double* extendedprice, double* discount, double* tax,
     uint8_t* returnflag, uint8_t* linestatus, uint8_t* shipmode,
-    int64_t* tp_cents, int64_t* suppkey)
+    int64_t* tp_cents, int64_t* suppkey, uint8_t* shipinstruct)
 {
     tpch_order_streams s;
     tpch_order_streams_init(&s, sf);
@@ -62,6 +62,7 @@ EXPORT int64_t tpch_gen_lineitem(double sf, int64_t order_start, int64_t order_c
             if (shipmode)      shipmode[n] = l.shipmode;
             if (tp_cents)      tp_cents[n] = l.tp_cents;
             if (suppkey)       suppkey[n] = l.suppkey;
+            if (shipinstruct)  shipinstruct[n] = l.shipinstruct;
         }
         tpch_order_row_finished(&s);
     }

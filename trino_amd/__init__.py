@@ -52,6 +52,7 @@ class LineitemCols(ctypes.Structure):
         ("shipmode", ctypes.c_void_p),
         ("tp_cents", ctypes.c_void_p),
         ("suppkey", ctypes.c_void_p),
+        ("shipinstruct", ctypes.c_void_p),
     ]
 
 
@@ -115,13 +116,15 @@ class Session:
     # --- TPC-H device generator (bench/test input infrastructure) ---
     def tpch_lineitem(self, sf, order_start=1, order_count=None, with_orderkey=False,
                       with_dates=False, with_partkey=False, with_shipmode=False,
-                      with_totalprice=False, with_suppkey=False):
+                      with_totalprice=False, with_suppkey=False,
+                      with_shipinstruct=False):
         if order_count is None:
             order_count = int(1_500_000 * sf)
         cols = LineitemCols()
         flags = ((1 if with_orderkey else 0) | (2 if with_dates else 0) |
                  (4 if with_partkey else 0) | (8 if with_shipmode else 0) |
-                 (16 if with_totalprice else 0) | (32 if with_suppkey else 0))
+                 (16 if with_totalprice else 0) | (32 if with_suppkey else 0) |
+                 (64 if with_shipinstruct else 0))
         _check(_lib.tg_tpch_lineitem_alloc(self._h, sf, order_start, order_count,
                                            flags, ctypes.byref(cols)))
         return cols

@@ -24,6 +24,9 @@ __global__ void k_emit_f64(const double* state, const int32_t* old_by_new,
                            int32_t n, double* out);
 __global__ void k_emit_i64(const long long* state, const int32_t* old_by_new,
                            int32_t n, int64_t* out);
+__global__ void k_emit_i64_biased(const long long* state,
+                                  const int32_t* old_by_new, int32_t n,
+                                  int64_t* out);
 __global__ void k_emit_avg(const double* sum, const long long* cnt,
                            const int32_t* old_by_new, int32_t n, double* out,
                            uint64_t* out_valid);

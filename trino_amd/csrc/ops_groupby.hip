@@ -214,8 +214,11 @@ __global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
         bool uniform = (__ballot(g == g0) == ~0ull) && (__ballot(active) == ~0ull);
         for (int a = 0; a < n_aggs; a++) {
             KAgg ag = aggs[a];
+            int mmode = (ag.fn == TG_AGG_MIN_I64) ? 1
+                      : (ag.fn == TG_AGG_MAX_I64) ? 2 : 0;
             /* per-lane addend for this agg */
-            long long ci = 0;       /* integer/count addend */
+            long long ci = mmode == 1 ? INT64_MAX
+                         : mmode == 2 ? INT64_MIN : 0;   /* integer addend */
             double cf = 0.0;        /* f64 addend */
             unsigned long long lo = 0, hi = 0;   /* exact i128 addend */
             if (active) {
@@ -250,6 +253,12 @@ __global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
                         }
                         break;
                     }
+                    case TG_AGG_MIN_I64: case TG_AGG_MAX_I64:
+                        /* raw value rides in ci (identity preset); FINAL
+                           combine reads the partial's plain int64 column */
+                        if (!kcol_is_null(cols[ag.in_ch], i))
+                            ci = ((const int64_t*)cols[ag.in_ch].data)[i];
+                        break;
                     case TG_AGG_AVG_F64:
                         if (step == 0) {
                             if (!kcol_is_null(cols[ag.in_ch], i)) {
@@ -277,12 +286,15 @@ __global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
                     int32_t gsel = __shfl(g, lead, 64);
                     bool mine = active && g == gsel;
                     unsigned long long mask = __ballot(mine);
-                    long long rci = mine ? ci : 0;
+                    long long rci = mine ? ci
+                        : mmode == 1 ? INT64_MAX : mmode == 2 ? INT64_MIN : 0;
                     double rcf = mine ? cf : 0.0;
                     unsigned long long rlo = mine ? lo : 0, rhi = mine ? hi : 0;
                     #pragma unroll
                     for (int off = 32; off >= 1; off >>= 1) {
-                        rci += __shfl_xor(rci, off, 64);
+                        { long long o = __shfl_xor(rci, off, 64);
+                          rci = mmode == 1 ? (o < rci ? o : rci)
+                              : mmode == 2 ? (o > rci ? o : rci) : rci + o; }
                         rcf += __shfl_xor(rcf, off, 64);
                         unsigned long long olo = __shfl_xor(rlo, off, 64);
                         unsigned long long ohi = __shfl_xor(rhi, off, 64);
@@ -307,6 +319,14 @@ __global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
                                 atomicAdd((unsigned long long*)ag.sum + gsel, rhi + carry);
                                 break;
                             }
+                            case TG_AGG_MIN_I64:
+                                atomicMin((unsigned long long*)&ag.cnt[gsel],
+                                          (unsigned long long)rci ^ 0x8000000000000000ull);
+                                break;
+                            case TG_AGG_MAX_I64:
+                                atomicMax((unsigned long long*)&ag.cnt[gsel],
+                                          (unsigned long long)rci ^ 0x8000000000000000ull);
+                                break;
                             case TG_AGG_AVG_F64:
                                 if (rci) atomicAdd((unsigned long long*)&ag.cnt[gsel],
                                                    (unsigned long long)rci);
@@ -342,6 +362,14 @@ __global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
                             atomicAdd((unsigned long long*)ag.sum + g, hi + carry);
                             break;
                         }
+                        case TG_AGG_MIN_I64:
+                            atomicMin((unsigned long long*)&ag.cnt[g],
+                                      (unsigned long long)ci ^ 0x8000000000000000ull);
+                            break;
+                        case TG_AGG_MAX_I64:
+                            atomicMax((unsigned long long*)&ag.cnt[g],
+                                      (unsigned long long)ci ^ 0x8000000000000000ull);
+                            break;
                         case TG_AGG_AVG_F64:
                             if (ci) atomicAdd((unsigned long long*)&ag.cnt[g],
                                               (unsigned long long)ci);
@@ -354,7 +382,9 @@ __global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
                 /* wave reduce, one atomic from lane 0 */
                 #pragma unroll
                 for (int off = 32; off >= 1; off >>= 1) {
-                    ci += __shfl_xor(ci, off, 64);
+                    { long long o = __shfl_xor(ci, off, 64);
+                      ci = mmode == 1 ? (o < ci ? o : ci)
+                         : mmode == 2 ? (o > ci ? o : ci) : ci + o; }
                     cf += __shfl_xor(cf, off, 64);
                     unsigned long long olo = __shfl_xor(lo, off, 64);
                     unsigned long long ohi = __shfl_xor(hi, off, 64);
@@ -379,6 +409,14 @@ __global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
                             atomicAdd((unsigned long long*)ag.sum + g0, hi + carry);
                             break;
                         }
+                        case TG_AGG_MIN_I64:
+                            atomicMin((unsigned long long*)&ag.cnt[g0],
+                                      (unsigned long long)ci ^ 0x8000000000000000ull);
+                            break;
+                        case TG_AGG_MAX_I64:
+                            atomicMax((unsigned long long*)&ag.cnt[g0],
+                                      (unsigned long long)ci ^ 0x8000000000000000ull);
+                            break;
                         case TG_AGG_AVG_F64:
                             if (ci) atomicAdd((unsigned long long*)&ag.cnt[g0],
                                               (unsigned long long)ci);
@@ -412,7 +450,10 @@ __global__ void k_agg_update_sorted(const int32_t* __restrict__ gids, int64_t n,
         bool is_last = active && (lane == 63 || i + 1 >= n || gnext != g);
         for (int a = 0; a < n_aggs; a++) {
             KAgg ag = aggs[a];
-            long long ci = 0;
+            int mmode = (ag.fn == TG_AGG_MIN_I64) ? 1
+                      : (ag.fn == TG_AGG_MAX_I64) ? 2 : 0;
+            long long ci = mmode == 1 ? INT64_MAX
+                         : mmode == 2 ? INT64_MIN : 0;
             double cf = 0.0;
             unsigned long long lo = 0, hi = 0;
             if (active) {
@@ -447,6 +488,12 @@ __global__ void k_agg_update_sorted(const int32_t* __restrict__ gids, int64_t n,
                         }
                         break;
                     }
+                    case TG_AGG_MIN_I64: case TG_AGG_MAX_I64:
+                        /* raw value rides in ci (identity preset); FINAL
+                           combine reads the partial's plain int64 column */
+                        if (!kcol_is_null(cols[ag.in_ch], i))
+                            ci = ((const int64_t*)cols[ag.in_ch].data)[i];
+                        break;
                     case TG_AGG_AVG_F64:
                         if (step == 0) {
                             if (!kcol_is_null(cols[ag.in_ch], i)) {
@@ -470,7 +517,8 @@ __global__ void k_agg_update_sorted(const int32_t* __restrict__ gids, int64_t n,
                 unsigned long long ohi = __shfl_up(hi, off, 64);
                 int32_t og = __shfl_up(g, off, 64);
                 if (lane >= off && og == g) {
-                    ci += oci;
+                    ci = mmode == 1 ? (oci < ci ? oci : ci)
+                       : mmode == 2 ? (oci > ci ? oci : ci) : ci + oci;
                     cf += ocf;
                     unsigned long long nlo = lo + olo;
                     hi = hi + ohi + (nlo < lo ? 1ull : 0ull);
@@ -494,6 +542,14 @@ __global__ void k_agg_update_sorted(const int32_t* __restrict__ gids, int64_t n,
                         atomicAdd((unsigned long long*)ag.sum + g, hi + carry);
                         break;
                     }
+                    case TG_AGG_MIN_I64:
+                        atomicMin((unsigned long long*)&ag.cnt[g],
+                                  (unsigned long long)ci ^ 0x8000000000000000ull);
+                        break;
+                    case TG_AGG_MAX_I64:
+                        atomicMax((unsigned long long*)&ag.cnt[g],
+                                  (unsigned long long)ci ^ 0x8000000000000000ull);
+                        break;
                     case TG_AGG_AVG_F64:
                         if (ci) atomicAdd((unsigned long long*)&ag.cnt[g],
                                           (unsigned long long)ci);
@@ -567,6 +623,16 @@ __global__ void k_emit_f64(const double* __restrict__ state, const int32_t* old_
     if (g < n) out[g] = state[old_by_new[g]];
 }
 
+__global__ void k_emit_i64_biased(const long long* __restrict__ state,
+                                  const int32_t* __restrict__ old_by_new,
+                                  int32_t n, int64_t* __restrict__ out)
+{
+    int32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n) return;
+    out[g] = (int64_t)((unsigned long long)state[old_by_new ? old_by_new[g] : g]
+                       ^ 0x8000000000000000ull);
+}
+
 __global__ void k_emit_i64(const long long* __restrict__ state, const int32_t* old_by_new,
                            int32_t n, int64_t* __restrict__ out)
 {
@@ -633,7 +699,8 @@ struct HashAggOp : tg_operator {
         TG_HIP_CHECK(hipGetLastError());
         for (auto& a : agg_state) {
             if (a.sum) TG_HIP_CHECK(hipMemsetAsync(a.sum, 0, max_groups * 8, s->stream));
-            if (a.cnt) TG_HIP_CHECK(hipMemsetAsync(a.cnt, 0, max_groups * 8, s->stream));
+            if (a.cnt) TG_HIP_CHECK(hipMemsetAsync(a.cnt,
+                    a.fn == TG_AGG_MIN_I64 ? 0xFF : 0, max_groups * 8, s->stream));
         }
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
         return TG_OK;
@@ -705,7 +772,8 @@ struct HashAggOp : tg_operator {
                 if (a.cnt) {
                     long long* nc;
                     TG_POOL_ALLOC(s, &nc, new_groups * 8);
-                    TG_HIP_CHECK(hipMemsetAsync(nc, 0, new_groups * 8, s->stream));
+                    TG_HIP_CHECK(hipMemsetAsync(nc,
+                            a.fn == TG_AGG_MIN_I64 ? 0xFF : 0, new_groups * 8, s->stream));
                     TG_HIP_CHECK(hipMemcpyAsync(nc, a.cnt, t.max_groups * 8,
                                                 hipMemcpyDeviceToDevice, s->stream));
                     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
@@ -940,6 +1008,15 @@ have_gids:;
                 hipLaunchKernelGGL(k_emit_i64, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
                                    (const long long*)ag.sum, d_obn, ng, (int64_t*)bh.data);
                 outp.blocks.push_back(bh);
+            }
+            else if (ag.fn == TG_AGG_MIN_I64 || ag.fn == TG_AGG_MAX_I64) {
+                /* state is sign-bit-biased u64; emit the plain int64 value
+                   (PARTIAL and FINAL states are identical) */
+                DevBlock b; b.type = TG_BIGINT; b.n = ng;
+                TG_POOL_ALLOC(s, &b.data, (int64_t)(ng ? ng : 1) * 8);
+                hipLaunchKernelGGL(k_emit_i64_biased, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                                   ag.cnt, d_obn, ng, (int64_t*)b.data);
+                outp.blocks.push_back(b);
             }
             else if (ag.fn == TG_AGG_SUM_F64) {
                 DevBlock b; b.type = TG_DOUBLE; b.n = ng;

@@ -96,7 +96,8 @@ struct StreamAggOp : tg_operator {
             if (a.cnt) {
                 long long* nc = nullptr;
                 TG_POOL_ALLOC(s, &nc, ncap * 8);
-                TG_HIP_CHECK(hipMemsetAsync(nc, 0, ncap * 8, s->stream));
+                TG_HIP_CHECK(hipMemsetAsync(nc,
+                        a.fn == TG_AGG_MIN_I64 ? 0xFF : 0, ncap * 8, s->stream));
                 if (n_runs) {
                     TG_HIP_CHECK(hipMemcpyAsync(nc, a.cnt, n_runs * 8,
                                                 hipMemcpyDeviceToDevice, s->stream));
@@ -295,6 +296,13 @@ void StreamAggOp::st_emit_aggs(DevPage* outp, int64_t ng)
                                (const long long*)ag.sum, d_obn, (int32_t)ng,
                                (int64_t*)bh.data);
             outp->blocks.push_back(bh);
+        }
+        else if (ag.fn == TG_AGG_MIN_I64 || ag.fn == TG_AGG_MAX_I64) {
+            DevBlock b; b.type = TG_BIGINT; b.n = ng;
+            (void)tg_pool_alloc(s, (void**)&b.data, (int64_t)(ng ? ng : 1) * 8);
+            hipLaunchKernelGGL(k_emit_i64_biased, dim3(grid), dim3(TG_BLOCK), 0, s->stream,
+                               ag.cnt, d_obn, (int32_t)ng, (int64_t*)b.data);
+            outp->blocks.push_back(b);
         }
         else if (ag.fn == TG_AGG_SUM_F64) {
             DevBlock b; b.type = TG_DOUBLE; b.n = ng;

@@ -40,7 +40,8 @@ __global__ void k_gen_lineitem(double sf, int64_t order_start, int64_t order_cou
                                uint8_t* returnflag, uint8_t* linestatus,
                                int32_t* commitdate, int32_t* receiptdate,
                                int64_t* partkey, uint8_t* shipmode,
-                               int64_t* tp_cents, int64_t* suppkey)
+                               int64_t* tp_cents, int64_t* suppkey,
+                               uint8_t* shipinstruct)
 {
     int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (g >= n_groups) return;
@@ -70,6 +71,7 @@ __global__ void k_gen_lineitem(double sf, int64_t order_start, int64_t order_cou
             if (shipmode)      shipmode[n] = l.shipmode;
             if (tp_cents)      tp_cents[n] = l.tp_cents;
             if (suppkey)       suppkey[n] = l.suppkey;
+            if (shipinstruct)  shipinstruct[n] = l.shipinstruct;
         }
         tpch_order_row_finished(&s);
     }
@@ -217,7 +219,8 @@ extern "C" tg_status tg_tpch_gen_lineitem(tg_session* s, double sf,
                        cols->extendedprice, cols->discount, cols->tax,
                        cols->returnflag, cols->linestatus,
                        cols->commitdate, cols->receiptdate, cols->partkey,
-                       cols->shipmode, cols->tp_cents, cols->suppkey);
+                       cols->shipmode, cols->tp_cents, cols->suppkey,
+                       cols->shipinstruct);
     TG_HIP_CHECK(hipGetLastError());
     TG_HIP_CHECK(hipStreamSynchronize(s->stream));
     TG_HIP_CHECK(hipFree(d_offsets));
@@ -243,6 +246,7 @@ extern "C" tg_status tg_tpch_lineitem_alloc(tg_session* s, double sf,
     if (flags & 8) TG_HIP_CHECK(hipMalloc(&cols->shipmode, rows));
     if (flags & 16) TG_HIP_CHECK(hipMalloc(&cols->tp_cents, rows * 8));
     if (flags & 32) TG_HIP_CHECK(hipMalloc(&cols->suppkey, rows * 8));
+    if (flags & 64) TG_HIP_CHECK(hipMalloc(&cols->shipinstruct, rows));
     TG_HIP_CHECK(hipMalloc(&cols->shipdate, rows * 4));
     TG_HIP_CHECK(hipMalloc(&cols->quantity, rows * 8));
     TG_HIP_CHECK(hipMalloc(&cols->extendedprice, rows * 8));
@@ -263,6 +267,7 @@ extern "C" tg_status tg_tpch_lineitem_free(tg_session* s, tg_tpch_lineitem_cols*
     if (cols->shipmode) TG_HIP_CHECK(hipFree(cols->shipmode));
     if (cols->tp_cents) TG_HIP_CHECK(hipFree(cols->tp_cents));
     if (cols->suppkey) TG_HIP_CHECK(hipFree(cols->suppkey));
+    if (cols->shipinstruct) TG_HIP_CHECK(hipFree(cols->shipinstruct));
     TG_HIP_CHECK(hipFree(cols->shipdate));
     TG_HIP_CHECK(hipFree(cols->quantity));
     TG_HIP_CHECK(hipFree(cols->extendedprice));

@@ -12,7 +12,7 @@ from . import _lib, _check, Session
 TG_BIGINT, TG_INTEGER, TG_SMALLINT, TG_TINYINT, TG_DOUBLE, TG_DATE, TG_BOOLEAN, TG_VARCHAR = range(8)
 STEP_PARTIAL, STEP_FINAL, STEP_SINGLE = 0, 1, 2
 (AGG_COUNT_STAR, AGG_COUNT_COL, AGG_SUM_F64, AGG_SUM_I64, AGG_AVG_F64,
- AGG_SUM_F64_EXACT) = range(6)
+ AGG_SUM_F64_EXACT, AGG_MIN_I64, AGG_MAX_I64) = range(8)
 
 _NP2TG = {np.dtype(np.int64): TG_BIGINT, np.dtype(np.int32): TG_INTEGER,
           np.dtype(np.int16): TG_SMALLINT, np.dtype(np.int8): TG_TINYINT,
