@@ -483,3 +483,40 @@ def page_with_varchar(columns):
     p.blocks = blocks
     p._keepalive = (blocks, keep)
     return p
+
+
+# ---- round 2: extended TPC-H tables + text/LIKE (device) ----
+_lib.tg_tpch_gen_part2.restype = ctypes.c_int
+_lib.tg_tpch_gen_part2.argtypes = [ctypes.c_void_p, ctypes.c_double,
+                                   ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 7
+_lib.tg_tpch_gen_partsupp.restype = ctypes.c_int
+_lib.tg_tpch_gen_partsupp.argtypes = [ctypes.c_void_p, ctypes.c_double,
+                                      ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
+_lib.tg_tpch_gen_supplier2.restype = ctypes.c_int
+_lib.tg_tpch_gen_supplier2.argtypes = [ctypes.c_void_p, ctypes.c_double,
+                                       ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 3
+_lib.tg_tpch_gen_orders3.restype = ctypes.c_int
+_lib.tg_tpch_gen_orders3.argtypes = [ctypes.c_void_p, ctypes.c_double,
+                                     ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 8
+_lib.tg_pool_like_flags.restype = ctypes.c_int
+_lib.tg_pool_like_flags.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                    ctypes.c_void_p, ctypes.c_int64,
+                                    ctypes.c_char_p, ctypes.c_void_p]
+_lib.tg_varchar_like_flags.restype = ctypes.c_int
+_lib.tg_varchar_like_flags.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                       ctypes.c_void_p, ctypes.c_int64,
+                                       ctypes.c_char_p, ctypes.c_void_p]
+_lib.tg_tpch_gen_supplier_comments.restype = ctypes.c_int
+_lib.tg_tpch_gen_supplier_comments.argtypes = [ctypes.c_void_p, ctypes.c_double,
+                                               ctypes.c_int64, ctypes.c_int64,
+                                               ctypes.c_void_p, ctypes.c_void_p]
+
+
+def pool_like_flags(session, d_offs, d_lens, n, pattern, d_flags):
+    _check(_lib.tg_pool_like_flags(session._h, d_offs, d_lens, n,
+                                   pattern.encode(), d_flags))
+
+
+def varchar_like_flags(session, d_bytes, d_offsets, n, pattern, d_flags):
+    _check(_lib.tg_varchar_like_flags(session._h, d_bytes, d_offsets, n,
+                                      pattern.encode(), d_flags))
