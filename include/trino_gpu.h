@@ -330,6 +330,9 @@ tg_status tg_tpch_part_strings(double sf, const int64_t* keys, int32_t n,
     char* container);
 tg_status tg_tpch_nation_name(int32_t nationkey, char out[32]);
 int32_t tg_tpch_nation_region(int32_t nationkey);
+/* p_name predicate flags over the 5 color ids per part (green=33 forest=28) */
+tg_status tg_tpch_part_name_flag(tg_session*, const uint8_t* d_name_ids,
+    int64_t n, int32_t color_id, int32_t first_only, uint8_t* d_flags);
 /* coarse device-memory accounting (LocalMemoryContext analog): bytes ever
  * pooled and bytes currently cached; live = total - cached */
 tg_status tg_session_memory(tg_session*, int64_t* total_bytes, int64_t* cached_bytes);

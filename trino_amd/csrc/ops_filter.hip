@@ -17,7 +17,7 @@
 
 struct KCol { const void* data; const uint64_t* valid; int32_t type; int32_t _pad; };
 
-#define MAX_STACK 4
+#define MAX_STACK 6
 
 /* Typed stack value: BIGINT/INTEGER/DATE/… stay in an exact int64 lane
  * (mirrors sql/gen/columnar/CallColumnarFilterGenerator.java:160-198 which
@@ -54,13 +54,15 @@ __device__ static inline bool tval_truthy(const TVal& v) { return v.isint ? v.i 
 __device__ static TVal eval_expr(const tg_expr_inst* prog, int count,
                                  const KCol* cols, int64_t row)
 {
-    TVal s0{}, s1{}, s2{}, s3{};
+    TVal s0{}, s1{}, s2{}, s3{}, s4{}, s5{};
     int sp = 0;
 #define PUSH(v) do { switch (sp) { \
         case 0: s0 = (v); break; case 1: s1 = (v); break; \
-        case 2: s2 = (v); break; default: s3 = (v); break; } \
+        case 2: s2 = (v); break; case 3: s3 = (v); break; \
+        case 4: s4 = (v); break; default: s5 = (v); break; } \
         sp++; } while (0)
-#define TOP (sp == 1 ? s0 : sp == 2 ? s1 : sp == 3 ? s2 : s3)
+#define TOP (sp == 1 ? s0 : sp == 2 ? s1 : sp == 3 ? s2 : sp == 4 ? s3 \
+             : sp == 5 ? s4 : s5)
 #define POP(v) do { v = TOP; sp--; } while (0)
 #define ARITH(expr_i, expr_f) do { TVal r; r.null = a.null | b.null; \
         r.isint = a.isint & b.isint; \

@@ -794,3 +794,34 @@ extern "C" tg_status tg_tpch_gen_supplier_comments(tg_session* s, double sf,
     *d_bytes_out = d_bytes;
     return TG_OK;
 }
+
+/* p_name predicate helper: flags[i] = name contains color `cid` (any of the
+ * 5 words) or, with first_only, name starts with it ('forest%'-class
+ * prefixes / '%green%'-class contains — color words never occur as
+ * substrings of other colors at word boundaries in the 92-color dist). */
+__global__ void k_part_name_flag(const uint8_t* __restrict__ name_ids,
+                                 int64_t n, int cid, int first_only,
+                                 uint8_t* __restrict__ flags)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        const uint8_t* w = name_ids + i * 5;
+        bool f = first_only ? (w[0] == cid)
+                            : (w[0] == cid || w[1] == cid || w[2] == cid ||
+                               w[3] == cid || w[4] == cid);
+        flags[i] = f ? 1 : 0;
+    }
+}
+
+extern "C" tg_status tg_tpch_part_name_flag(tg_session* s,
+    const uint8_t* d_name_ids, int64_t n, int32_t color_id, int32_t first_only,
+    uint8_t* d_flags)
+{
+    hipLaunchKernelGGL(k_part_name_flag, dim3(tg_grid_for(n)), dim3(TG_BLOCK),
+                       0, s->stream, d_name_ids, n, color_id, first_only,
+                       d_flags);
+    TG_HIP_CHECK(hipGetLastError());
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}

@@ -1096,3 +1096,1182 @@ def q8_gpu(session, sf, order_start=1, order_count=None):
         res[y] = num / den
     return dict(o_year=[1995, 1996],
                 mkt_share=[res[1995], res[1996]], elapsed=elapsed)
+
+
+# ===================== round 2: the remaining ten queries ==================
+# Each mirrors the oracle composition pinned (vs the reference's SF1 answer
+# fixtures) in tests/test_oracle_queries.py; pipelines are operator-ABI
+# chains like the reference's LocalExecutionPlanner operator factories.
+
+import datetime as _dt
+
+
+def _D(y, m, d):
+    return (_dt.date(y, m, d) - _dt.date(1970, 1, 1)).days
+
+
+_lib.tg_tpch_gen_orders3.restype = ctypes.c_int
+_lib.tg_tpch_gen_orders3.argtypes = [ctypes.c_void_p, ctypes.c_double,
+                                     ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 8
+_lib.tg_tpch_gen_part2.restype = ctypes.c_int
+_lib.tg_tpch_gen_part2.argtypes = [ctypes.c_void_p, ctypes.c_double,
+                                   ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 7
+_lib.tg_tpch_gen_partsupp.restype = ctypes.c_int
+_lib.tg_tpch_gen_partsupp.argtypes = [ctypes.c_void_p, ctypes.c_double,
+                                      ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 4
+_lib.tg_tpch_gen_supplier2.restype = ctypes.c_int
+_lib.tg_tpch_gen_supplier2.argtypes = [ctypes.c_void_p, ctypes.c_double,
+                                       ctypes.c_int64, ctypes.c_int64] + [ctypes.c_void_p] * 3
+_lib.tg_tpch_gen_supplier_comments.restype = ctypes.c_int
+_lib.tg_tpch_gen_supplier_comments.argtypes = [ctypes.c_void_p, ctypes.c_double,
+                                               ctypes.c_int64, ctypes.c_int64,
+                                               ctypes.c_void_p, ctypes.c_void_p]
+for _fn in (_lib.tg_tpch_supplier_strings, _lib.tg_tpch_customer_strings):
+    _fn.restype = ctypes.c_int
+    _fn.argtypes = [ctypes.c_double, ctypes.c_void_p, ctypes.c_int32,
+                    ctypes.c_int32] + [ctypes.c_void_p] * 6
+_lib.tg_tpch_part_strings.restype = ctypes.c_int
+_lib.tg_tpch_part_strings.argtypes = [ctypes.c_double, ctypes.c_void_p,
+                                      ctypes.c_int32, ctypes.c_int32] + [ctypes.c_void_p] * 5
+
+NATIONS = ["ALGERIA", "ARGENTINA", "BRAZIL", "CANADA", "EGYPT", "ETHIOPIA",
+           "FRANCE", "GERMANY", "INDIA", "INDONESIA", "IRAN", "IRAQ",
+           "JAPAN", "JORDAN", "KENYA", "MOROCCO", "MOZAMBIQUE", "PERU",
+           "CHINA", "ROMANIA", "SAUDI ARABIA", "VIETNAM", "RUSSIA",
+           "UNITED KINGDOM", "UNITED STATES"]
+NATION_REGION = [0, 1, 1, 1, 4, 0, 3, 3, 2, 2, 4, 4, 2, 4, 0, 0, 0, 1, 2, 3,
+                 4, 2, 3, 3, 1]
+TYPE_S1 = ["STANDARD", "SMALL", "MEDIUM", "LARGE", "ECONOMY", "PROMO"]
+TYPE_S2 = ["ANODIZED", "BURNISHED", "PLATED", "POLISHED", "BRUSHED"]
+TYPE_S3 = ["TIN", "NICKEL", "BRASS", "STEEL", "COPPER"]
+
+
+def _strings(fn, sf, keys, fields):
+    """host materialization of supplier/customer strings by key"""
+    keys = np.ascontiguousarray(np.asarray(keys, np.int64))
+    n = len(keys)
+    stride = 160
+    bufs = {f: ctypes.create_string_buffer(max(n, 1) * stride)
+            for f in fields if f in ("name", "address", "phone", "comment")}
+    ab = np.zeros(max(n, 1), np.int64) if "acctbal" in fields else None
+    nk = np.zeros(max(n, 1), np.int32) if "nationkey" in fields else None
+    _check_lib(fn(sf, keys.ctypes.data, n, stride,
+                  bufs.get("name"), bufs.get("address"), bufs.get("phone"),
+                  bufs.get("comment"),
+                  None if ab is None else ab.ctypes.data,
+                  None if nk is None else nk.ctypes.data))
+    out = {}
+    for f, b in bufs.items():
+        out[f] = [b.raw[i * stride:(i + 1) * stride].split(b"\0")[0].decode()
+                  for i in range(n)]
+    if ab is not None:
+        out["acctbal"] = ab[:n]
+    if nk is not None:
+        out["nationkey"] = nk[:n]
+    return out
+
+
+def _gen_orders3(session, sf, order_count, want):
+    """want: subset of (ok, ck, od, pri, status, tp, coff, clen)"""
+    sizes = dict(ok=8, ck=8, od=4, pri=1, status=1, tp=8, coff=8, clen=4)
+    bufs = {k: (_device_buffer(session, order_count * sizes[k]) if k in want else None)
+            for k in sizes}
+    _check_lib(_lib.tg_tpch_gen_orders3(
+        session._h, sf, 1, order_count, bufs["ok"], bufs["ck"], bufs["od"],
+        bufs["pri"], bufs["status"], bufs["tp"], bufs["coff"], bufs["clen"]))
+    return bufs
+
+
+def _free_bufs(session, bufs):
+    for v in bufs.values():
+        if v is not None:
+            _device_free(session, v)
+
+
+def q13_gpu(session, sf, order_count=None, cust_count=None):
+    """TPC-H Q13: customer order-count distribution over orders whose
+    o_comment is NOT LIKE '%special%requests%' (LEFT JOIN: customers with no
+    qualifying orders count in the 0 bucket). Comment LIKE runs directly over
+    the device text pool slices (no 7 GB materialization)."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    if cust_count is None:
+        cust_count = int(150_000 * sf)
+    bufs = _gen_orders3(session, sf, order_count, ("ck", "coff", "clen"))
+    t0 = time.time()
+    d_flags = _device_buffer(session, order_count)
+    ops.pool_like_flags(session, bufs["coff"], bufs["clen"], order_count,
+                        "%special%requests%", d_flags)
+    opage = ops.page_from_device(session, ([(bufs["ck"].value, ops.TG_BIGINT),
+                                            (d_flags.value, ops.TG_TINYINT)],
+                                           order_count))
+    f = ops.filter_project(session, ops.expr(("col", 1), ("i64", 0), "eq"),
+                           [ops.expr(("col", 0))], [ops.TG_BIGINT])
+    f.add_input(opage)
+    f.finish()
+    kept = _take_device_page(session, f)
+    a1 = ops.hash_aggregation(session, [0], [ops.TG_BIGINT],
+                              [(ops.AGG_COUNT_STAR, -1)])
+    a1.add_input(kept)
+    a1.finish()
+    percust = _take_device_page(session, a1)       # (custkey, c_count)
+    n_with = percust.position_count
+    a2 = ops.hash_aggregation(session, [1], [ops.TG_BIGINT],
+                              [(ops.AGG_COUNT_STAR, -1)])
+    a2.add_input(percust)
+    pages = a2.drain()
+    elapsed = time.time() - t0
+    for op in (f, a1, a2):
+        op.close()
+    _device_free(session, d_flags)
+    _free_bufs(session, bufs)
+    out = pages[0]
+    c_count = np.asarray(out[0]["values"]).astype(np.int64)
+    custdist = np.asarray(out[1]["values"]).astype(np.int64)
+    c_count = np.append(c_count, 0)
+    custdist = np.append(custdist, cust_count - n_with)
+    order = np.lexsort((-c_count, -custdist))
+    return dict(c_count=c_count[order], custdist=custdist[order],
+                elapsed=elapsed)
+
+
+def q16_gpu(session, sf, part_count=None):
+    """TPC-H Q16: supplier counts by (brand, type, size) excluding
+    Brand#45 / MEDIUM POLISHED% / sizes outside the 8-set and suppliers with
+    '%Customer%Complaints%' comments (BBB overlay + generic VARCHAR LIKE).
+    count(DISTINCT ps_suppkey) = dedup group-by then count."""
+    if part_count is None:
+        part_count = int(200_000 * sf)
+    supp_count = int(10_000 * sf)
+    p_pk = _device_buffer(session, part_count * 8)
+    p_ty = _device_buffer(session, part_count * 2)
+    p_br = _device_buffer(session, part_count)
+    p_sz = _device_buffer(session, part_count * 4)
+    _check_lib(_lib.tg_tpch_gen_part2(session._h, sf, 1, part_count,
+                                      p_pk, p_ty, p_br, p_sz, None, None, None))
+    ps_pk = _device_buffer(session, part_count * 4 * 8)
+    ps_sk = _device_buffer(session, part_count * 4 * 8)
+    _check_lib(_lib.tg_tpch_gen_partsupp(session._h, sf, 1, part_count,
+                                         ps_pk, ps_sk, None, None))
+    t0 = time.time()
+    # complaint suppliers (BBB overlay) via the generic VARCHAR LIKE kernel
+    d_soff = ctypes.c_void_p()
+    d_sbytes = ctypes.c_void_p()
+    _check_lib(_lib.tg_tpch_gen_supplier_comments(
+        session._h, sf, 1, supp_count, ctypes.byref(d_soff), ctypes.byref(d_sbytes)))
+    d_cflag = _device_buffer(session, supp_count)
+    ops.varchar_like_flags(session, d_sbytes, d_soff, supp_count,
+                           "%Customer%Complaints%", d_cflag)
+    # suppkey column 1..n on device = partsupp suppkeys of part 1..? simpler:
+    # download tiny flag vector, upload complaint keys as build page
+    flags = np.empty(supp_count, np.uint8)
+    from . import copy_dtoh
+    copy_dtoh(session, flags, d_cflag)
+    bad = np.nonzero(flags)[0].astype(np.int64) + 1
+    bridge_bad = ops.JoinBridge(session)
+    bb = ops.set_builder(session, bridge_bad, [ops.TG_BIGINT], 0)
+    bb.add_input(ops.page_from_numpy([np.ascontiguousarray(bad)]))
+    bb.drain()
+    # part filter
+    ppage = ops.page_from_device(session, ([(p_pk.value, ops.TG_BIGINT),
+                                            (p_br.value, ops.TG_TINYINT),
+                                            (p_ty.value, ops.TG_SMALLINT),
+                                            (p_sz.value, ops.TG_INTEGER)],
+                                           part_count))
+    sizes = [49, 14, 23, 45, 19, 3, 36, 9]
+    in_chain = []
+    for i, v in enumerate(sizes):
+        in_chain += [("col", 3), ("i64", v), "eq"]
+        if i:
+            in_chain.append("or")
+    fexpr = ops.expr(*(in_chain +
+                       [("col", 1), ("i64", 45), "ne", "and",
+                        ("col", 2), ("i64", 65), ("i64", 69), "between",
+                        "not", "and"]))
+    fp = ops.filter_project(session, fexpr,
+                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                             ops.expr(("col", 2)), ops.expr(("col", 3))],
+                            [ops.TG_BIGINT, ops.TG_TINYINT, ops.TG_SMALLINT,
+                             ops.TG_INTEGER])
+    fp.add_input(ppage)
+    fp.finish()
+    part_sel = _take_device_page(session, fp)
+    bridge_p = ops.JoinBridge(session)
+    bp = ops.hash_builder(session, bridge_p, [ops.TG_BIGINT, ops.TG_TINYINT,
+                                              ops.TG_SMALLINT, ops.TG_INTEGER],
+                          [0], [1, 2, 3])
+    bp.add_input(part_sel)
+    bp.drain()
+    # partsupp: drop complaint suppliers, then join part attrs
+    pspage = ops.page_from_device(session, ([(ps_pk.value, ops.TG_BIGINT),
+                                             (ps_sk.value, ops.TG_BIGINT)],
+                                            part_count * 4))
+    sj = ops.semi_join(session, bridge_bad, 1)
+    sj.add_input(pspage)
+    sj.finish()
+    marked = _take_device_page(session, sj)
+    fnb = ops.filter_project(session, ops.expr(("col", 2), ("i64", 0), "eq"),
+                             [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                             [ops.TG_BIGINT, ops.TG_BIGINT])
+    fnb.add_input(marked)
+    fnb.finish()
+    ps_ok = _take_device_page(session, fnb)
+    j = ops.lookup_join(session, bridge_p, [ops.TG_BIGINT, ops.TG_BIGINT],
+                        [0], [1])          # emit suppkey + (brand,type,size)
+    j.add_input(ps_ok)
+    j.finish()
+    joined = _take_device_page(session, j)  # (sk, brand, type, size)
+    # DISTINCT: dedup by the full tuple, then count per (brand,type,size)
+    d1 = ops.hash_aggregation(session, [1, 2, 3, 0],
+                              [ops.TG_TINYINT, ops.TG_SMALLINT, ops.TG_INTEGER,
+                               ops.TG_BIGINT], [])
+    d1.add_input(joined)
+    d1.finish()
+    dedup = _take_device_page(session, d1)
+    d2 = ops.hash_aggregation(session, [0, 1, 2],
+                              [ops.TG_TINYINT, ops.TG_SMALLINT, ops.TG_INTEGER],
+                              [(ops.AGG_COUNT_STAR, -1)])
+    d2.add_input(dedup)
+    pages = d2.drain()
+    elapsed = time.time() - t0
+    for op in (bb, fp, bp, sj, fnb, j, d1, d2):
+        op.close()
+    bridge_bad.close()
+    bridge_p.close()
+    for p in (p_pk, p_ty, p_br, p_sz, ps_pk, ps_sk, d_cflag):
+        _device_free(session, p)
+    out = pages[0]
+    br = np.asarray(out[0]["values"]).astype(np.int64)
+    ty = np.asarray(out[1]["values"]).astype(np.int64)
+    sz = np.asarray(out[2]["values"]).astype(np.int64)
+    cnt = np.asarray(out[3]["values"]).astype(np.int64)
+    rows = [(f"Brand#{br[i]}",
+             f"{TYPE_S1[ty[i]//25]} {TYPE_S2[(ty[i]//5)%5]} {TYPE_S3[ty[i]%5]}",
+             int(sz[i]), int(cnt[i])) for i in range(len(br))]
+    rows.sort(key=lambda r: (-r[3], r[0], r[1], r[2]))
+    return dict(rows=rows, elapsed=elapsed)
+
+
+def q11_gpu(session, sf, part_count=None):
+    """TPC-H Q11 (important stock in GERMANY): value per partkey =
+    sum(ps_supplycost * ps_availqty) over German suppliers, kept where
+    value > 0.0001 * total. Products/sums computed EXACTLY in integer cents
+    (cost_cents * qty fits i64); the threshold compare mirrors Trino's
+    double compare on the exact cents (margins are far above 1 ulp)."""
+    if part_count is None:
+        part_count = int(200_000 * sf)
+    supp_count = int(10_000 * sf)
+    n = part_count * 4
+    ps_pk = _device_buffer(session, n * 8)
+    ps_sk = _device_buffer(session, n * 8)
+    ps_aq = _device_buffer(session, n * 4)
+    ps_sc = _device_buffer(session, n * 8)
+    _check_lib(_lib.tg_tpch_gen_partsupp(session._h, sf, 1, part_count,
+                                         ps_pk, ps_sk, ps_aq, ps_sc))
+    s_sk = _device_buffer(session, supp_count * 8)
+    s_nk = _device_buffer(session, supp_count)
+    _check_lib(_lib.tg_tpch_gen_supplier2(session._h, sf, 1, supp_count,
+                                          s_sk, s_nk, None))
+    t0 = time.time()
+    spage = ops.page_from_device(session, ([(s_sk.value, ops.TG_BIGINT),
+                                            (s_nk.value, ops.TG_TINYINT)],
+                                           supp_count))
+    fs = ops.filter_project(session, ops.expr(("col", 1), ("i64", 7), "eq"),
+                            [ops.expr(("col", 0))], [ops.TG_BIGINT])
+    fs.add_input(spage)
+    fs.finish()
+    de = _take_device_page(session, fs)
+    bridge = ops.JoinBridge(session)
+    b = ops.set_builder(session, bridge, [ops.TG_BIGINT], 0)
+    b.add_input(de)
+    b.drain()
+    pspage = ops.page_from_device(session, ([(ps_pk.value, ops.TG_BIGINT),
+                                             (ps_sk.value, ops.TG_BIGINT),
+                                             (ps_aq.value, ops.TG_INTEGER),
+                                             (ps_sc.value, ops.TG_BIGINT)], n))
+    sj = ops.semi_join(session, bridge, 1)
+    sj.add_input(pspage)
+    sj.finish()
+    marked = _take_device_page(session, sj)
+    # value_cents = cost_cents * availqty (exact i64 via the typed lane)
+    fv = ops.filter_project(session, ops.expr(("col", 4), ("i64", 1), "eq"),
+                            [ops.expr(("col", 0)),
+                             ops.expr(("col", 3), ("col", 2), "mul")],
+                            [ops.TG_BIGINT, ops.TG_BIGINT])
+    fv.add_input(marked)
+    fv.finish()
+    vals = _take_device_page(session, fv)
+    agg = ops.hash_aggregation(session, [0], [ops.TG_BIGINT],
+                               [(ops.AGG_SUM_I64, 1)])
+    agg.add_input(vals)
+    agg.finish()
+    groups = _take_device_page(session, agg)       # (partkey, value_cents)
+    tot = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_I64, 1)])
+    tot.add_input(groups)
+    tpages = tot.drain()
+    total_cents = int(tpages[0][0]["values"][0])
+    # value > 0.0001 * total  <=>  value_cents * 10000 > total_cents
+    ff = ops.filter_project(session,
+                            ops.expr(("col", 1), ("i64", 10000), "mul",
+                                     ("i64", total_cents), "gt"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                            [ops.TG_BIGINT, ops.TG_BIGINT])
+    ff.add_input(groups)
+    pages = ff.drain()
+    elapsed = time.time() - t0
+    for op in (fs, b, sj, fv, agg, tot, ff):
+        op.close()
+    bridge.close()
+    for p in (ps_pk, ps_sk, ps_aq, ps_sc, s_sk, s_nk):
+        _device_free(session, p)
+    out = pages[0]
+    pk = np.asarray(out[0]["values"]).astype(np.int64)
+    cents = np.asarray(out[1]["values"]).astype(np.int64)
+    order = np.lexsort((pk, -cents))
+    return dict(partkey=pk[order], value_cents=cents[order], elapsed=elapsed)
+
+
+def q17_gpu(session, sf, order_count=None, part_count=None):
+    """TPC-H Q17 (small-quantity-order revenue): Brand#23 / MED BOX parts;
+    avg yearly = sum(l_extendedprice where l_quantity < 0.2*avg(qty) per
+    part) / 7. Two probe passes over the part-filtered lineitem join."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    if part_count is None:
+        part_count = int(200_000 * sf)
+    p_pk = _device_buffer(session, part_count * 8)
+    p_br = _device_buffer(session, part_count)
+    p_cn = _device_buffer(session, part_count)
+    _check_lib(_lib.tg_tpch_gen_part2(session._h, sf, 1, part_count,
+                                      p_pk, None, p_br, None, p_cn, None, None))
+    li = session.tpch_lineitem(sf, 1, order_count, with_partkey=True)
+    t0 = time.time()
+    ppage = ops.page_from_device(session, ([(p_pk.value, ops.TG_BIGINT),
+                                            (p_br.value, ops.TG_TINYINT),
+                                            (p_cn.value, ops.TG_TINYINT)],
+                                           part_count))
+    fp = ops.filter_project(session,
+                            ops.expr(("col", 1), ("i64", 23), "eq",
+                                     ("col", 2), ("i64", 17), "eq", "and"),
+                            [ops.expr(("col", 0))], [ops.TG_BIGINT])
+    fp.add_input(ppage)
+    fp.finish()
+    sel_parts = _take_device_page(session, fp)
+    bridge = ops.JoinBridge(session)
+    b = ops.hash_builder(session, bridge, [ops.TG_BIGINT], [0], [])
+    b.add_input(sel_parts)
+    b.drain()
+    lpage = ops.page_from_device(session, ([(li.partkey, ops.TG_BIGINT),
+                                            (li.quantity, ops.TG_DOUBLE),
+                                            (li.extendedprice, ops.TG_DOUBLE)],
+                                           li.row_count))
+    j = ops.lookup_join(session, bridge, [ops.TG_BIGINT, ops.TG_DOUBLE,
+                                          ops.TG_DOUBLE], [0], [0, 1, 2])
+    j.add_input(lpage)
+    j.finish()
+    rows = _take_device_page(session, j)           # (pk, qty, ep)
+    agg = ops.hash_aggregation(session, [0], [ops.TG_BIGINT],
+                               [(ops.AGG_SUM_F64, 1), (ops.AGG_COUNT_STAR, -1)])
+    agg.add_input(rows)
+    agg.finish()
+    peravg = _take_device_page(session, agg)       # (pk, sumq, cnt)
+    bridge2 = ops.JoinBridge(session)
+    b2 = ops.hash_builder(session, bridge2, [ops.TG_BIGINT, ops.TG_DOUBLE,
+                                             ops.TG_BIGINT], [0], [1, 2])
+    b2.add_input(peravg)
+    b2.drain()
+    j2 = ops.lookup_join(session, bridge2, [ops.TG_BIGINT, ops.TG_DOUBLE,
+                                            ops.TG_DOUBLE], [0], [1, 2])
+    j2.add_input(rows)
+    j2.finish()
+    wavg = _take_device_page(session, j2)          # (qty, ep, sumq, cnt)
+    # qty < 0.2 * (sumq / cnt)  — same IEEE ops as DoubleAverageAggregations
+    ffin = ops.filter_project(session,
+                              ops.expr(("col", 0),
+                                       ("f64", 0.2), ("col", 2), ("col", 3),
+                                       "div", "mul", "lt"),
+                              [ops.expr(("col", 1))], [ops.TG_DOUBLE])
+    ffin.add_input(wavg)
+    ffin.finish()
+    kept = _take_device_page(session, ffin)
+    sagg = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_F64_EXACT, 0, 43)])
+    sagg.add_input(kept)
+    pages = sagg.drain()
+    elapsed = time.time() - t0
+    for op in (fp, b, j, agg, b2, j2, ffin, sagg):
+        op.close()
+    bridge.close()
+    bridge2.close()
+    session.tpch_lineitem_free(li)
+    for p in (p_pk, p_br, p_cn):
+        _device_free(session, p)
+    total = float(pages[0][0]["values"][0])
+    return dict(avg_yearly=total / 7.0, elapsed=elapsed)
+
+
+def q19_gpu(session, sf, order_count=None, part_count=None):
+    """TPC-H Q19 (discounted revenue): three brand/container/size/quantity
+    branches, AIR shipmode + DELIVER IN PERSON, summed exactly."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    if part_count is None:
+        part_count = int(200_000 * sf)
+    p_pk = _device_buffer(session, part_count * 8)
+    p_br = _device_buffer(session, part_count)
+    p_cn = _device_buffer(session, part_count)
+    p_sz = _device_buffer(session, part_count * 4)
+    _check_lib(_lib.tg_tpch_gen_part2(session._h, sf, 1, part_count,
+                                      p_pk, None, p_br, p_sz, p_cn, None, None))
+    li = session.tpch_lineitem(sf, 1, order_count, with_partkey=True,
+                               with_shipmode=True, with_shipinstruct=True)
+    t0 = time.time()
+    bridge = ops.JoinBridge(session)
+    b = ops.hash_builder(session, bridge,
+                         [ops.TG_BIGINT, ops.TG_TINYINT, ops.TG_TINYINT,
+                          ops.TG_INTEGER], [0], [1, 2, 3])
+    b.add_input(ops.page_from_device(session, ([(p_pk.value, ops.TG_BIGINT),
+                                                (p_br.value, ops.TG_TINYINT),
+                                                (p_cn.value, ops.TG_TINYINT),
+                                                (p_sz.value, ops.TG_INTEGER)],
+                                               part_count)))
+    b.drain()
+    lpage = ops.page_from_device(session, ([(li.partkey, ops.TG_BIGINT),
+                                            (li.quantity, ops.TG_DOUBLE),
+                                            (li.extendedprice, ops.TG_DOUBLE),
+                                            (li.discount, ops.TG_DOUBLE),
+                                            (li.shipmode, ops.TG_TINYINT),
+                                            (li.shipinstruct, ops.TG_TINYINT)],
+                                           li.row_count))
+    # cheap common predicate first: shipmode AIR(1), DELIVER IN PERSON(0)
+    f1 = ops.filter_project(session,
+                            ops.expr(("col", 4), ("i64", 1), "eq",
+                                     ("col", 5), ("i64", 0), "eq", "and"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                             ops.expr(("col", 2), ("f64", 1.0), ("col", 3),
+                                      "sub", "mul")],
+                            [ops.TG_BIGINT, ops.TG_DOUBLE, ops.TG_DOUBLE])
+    f1.add_input(lpage)
+    f1.finish()
+    lsel = _take_device_page(session, f1)
+    j = ops.lookup_join(session, bridge, [ops.TG_BIGINT, ops.TG_DOUBLE,
+                                          ops.TG_DOUBLE], [0], [1, 2])
+    j.add_input(lsel)
+    j.finish()
+    rows = _take_device_page(session, j)  # (qty, disc_price, brand, cont, size)
+
+    def branch(bid, conts, qlo, qhi, shi):
+        e = []
+        for i, c in enumerate(conts):
+            e += [("col", 3), ("i64", c), "eq"]
+            if i:
+                e.append("or")
+        e += [("col", 2), ("i64", bid), "eq", "and",
+              ("col", 0), ("f64", float(qlo)), ("f64", float(qhi)), "between",
+              "and",
+              ("col", 4), ("i64", 1), ("i64", shi), "between", "and"]
+        return e
+
+    e1 = branch(12, [0, 1, 4, 5], 1, 11, 5)
+    e2 = branch(23, [17, 18, 20, 21], 10, 20, 10)
+    e3 = branch(34, [8, 9, 12, 13], 20, 30, 15)
+    full = e1 + e2 + ["or"] + e3 + ["or"]
+    f2 = ops.filter_project(session, ops.expr(*full),
+                            [ops.expr(("col", 1))], [ops.TG_DOUBLE])
+    f2.add_input(rows)
+    f2.finish()
+    kept = _take_device_page(session, f2)
+    sagg = ops.hash_aggregation(session, [], [], [(ops.AGG_SUM_F64_EXACT, 0, 43)])
+    sagg.add_input(kept)
+    pages = sagg.drain()
+    elapsed = time.time() - t0
+    for op in (b, f1, j, f2, sagg):
+        op.close()
+    bridge.close()
+    session.tpch_lineitem_free(li)
+    for p in (p_pk, p_br, p_cn, p_sz):
+        _device_free(session, p)
+    return dict(revenue=float(pages[0][0]["values"][0]), elapsed=elapsed)
+
+
+_lib.tg_tpch_part_name_flag.restype = ctypes.c_int
+_lib.tg_tpch_part_name_flag.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                        ctypes.c_int64, ctypes.c_int32,
+                                        ctypes.c_int32, ctypes.c_void_p]
+COLOR_GREEN, COLOR_FOREST = 33, 28
+
+
+def _strings(fn, sf, keys, fields):
+    """host materialization of supplier/customer strings by key"""
+    keys = np.ascontiguousarray(np.asarray(keys, np.int64))
+    n = len(keys)
+    stride = 160
+    bufs = {f: ctypes.create_string_buffer(max(n, 1) * stride)
+            for f in fields if f in ("name", "address", "phone", "comment")}
+    ab = np.zeros(max(n, 1), np.int64) if "acctbal" in fields else None
+    nk = np.zeros(max(n, 1), np.int32) if "nationkey" in fields else None
+    _check_lib(fn(sf, keys.ctypes.data, n, stride,
+                  bufs.get("name"), bufs.get("address"), bufs.get("phone"),
+                  bufs.get("comment"),
+                  None if ab is None else ab.ctypes.data,
+                  None if nk is None else nk.ctypes.data))
+    out = {}
+    for f, b in bufs.items():
+        out[f] = [b.raw[i * stride:(i + 1) * stride].split(b"\0")[0].decode()
+                  for i in range(n)]
+    if ab is not None:
+        out["acctbal"] = ab[:n]
+    if nk is not None:
+        out["nationkey"] = nk[:n]
+    return out
+
+
+def _strings_part(sf, keys, fields):
+    keys = np.ascontiguousarray(np.asarray(keys, np.int64))
+    n = len(keys)
+    stride = 160
+    bufs = {f: ctypes.create_string_buffer(max(n, 1) * stride)
+            for f in fields}
+    _check_lib(_lib.tg_tpch_part_strings(
+        sf, keys.ctypes.data, n, stride, bufs.get("name"), bufs.get("mfgr"),
+        bufs.get("brand"), bufs.get("type"), bufs.get("container")))
+    return {f: [b.raw[i * stride:(i + 1) * stride].split(b"\0")[0].decode()
+                for i in range(n)] for f, b in bufs.items()}
+
+
+def q2_gpu(session, sf, part_count=None):
+    """TPC-H Q2 (minimum-cost supplier): size=15, type '%BRASS', EUROPE;
+    rows where ps_supplycost equals the per-part minimum over European
+    suppliers (TG_AGG_MIN_I64 on exact cents); output strings host-side."""
+    if part_count is None:
+        part_count = int(200_000 * sf)
+    supp_count = int(10_000 * sf)
+    p_pk = _device_buffer(session, part_count * 8)
+    p_ty = _device_buffer(session, part_count * 2)
+    p_sz = _device_buffer(session, part_count * 4)
+    _check_lib(_lib.tg_tpch_gen_part2(session._h, sf, 1, part_count,
+                                      p_pk, p_ty, None, p_sz, None, None, None))
+    n = part_count * 4
+    ps_pk = _device_buffer(session, n * 8)
+    ps_sk = _device_buffer(session, n * 8)
+    ps_sc = _device_buffer(session, n * 8)
+    _check_lib(_lib.tg_tpch_gen_partsupp(session._h, sf, 1, part_count,
+                                         ps_pk, ps_sk, None, ps_sc))
+    s_sk = _device_buffer(session, supp_count * 8)
+    s_nk = _device_buffer(session, supp_count)
+    s_ab = _device_buffer(session, supp_count * 8)
+    _check_lib(_lib.tg_tpch_gen_supplier2(session._h, sf, 1, supp_count,
+                                          s_sk, s_nk, s_ab))
+    t0 = time.time()
+    eu = [nk for nk in range(25) if NATION_REGION[nk] == 3]
+    spage = ops.page_from_device(session, ([(s_sk.value, ops.TG_BIGINT),
+                                            (s_nk.value, ops.TG_TINYINT),
+                                            (s_ab.value, ops.TG_BIGINT)],
+                                           supp_count))
+    chain = []
+    for i, nk in enumerate(eu):
+        chain += [("col", 1), ("i64", nk), "eq"]
+        if i:
+            chain.append("or")
+    fs = ops.filter_project(session, ops.expr(*chain),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                             ops.expr(("col", 2))],
+                            [ops.TG_BIGINT, ops.TG_TINYINT, ops.TG_BIGINT])
+    fs.add_input(spage)
+    fs.finish()
+    eu_supp = _take_device_page(session, fs)
+    bridge_s = ops.JoinBridge(session)
+    bs = ops.hash_builder(session, bridge_s,
+                          [ops.TG_BIGINT, ops.TG_TINYINT, ops.TG_BIGINT],
+                          [0], [1, 2])
+    bs.add_input(eu_supp)
+    bs.drain()
+    ppage = ops.page_from_device(session, ([(p_pk.value, ops.TG_BIGINT),
+                                            (p_ty.value, ops.TG_SMALLINT),
+                                            (p_sz.value, ops.TG_INTEGER)],
+                                           part_count))
+    # size == 15 AND type %% 5 == 2 ('%BRASS'): ((ty-2)/5)*5 == ty-2
+    fexpr = ops.expr(("col", 2), ("i64", 15), "eq",
+                     ("col", 1), ("i64", 2), "sub", ("i64", 5), "div",
+                     ("i64", 5), "mul",
+                     ("col", 1), ("i64", 2), "sub", "eq", "and")
+    fp = ops.filter_project(session, fexpr, [ops.expr(("col", 0))],
+                            [ops.TG_BIGINT])
+    fp.add_input(ppage)
+    fp.finish()
+    sel_parts = _take_device_page(session, fp)
+    bridge_p = ops.JoinBridge(session)
+    bp = ops.hash_builder(session, bridge_p, [ops.TG_BIGINT], [0], [])
+    bp.add_input(sel_parts)
+    bp.drain()
+    pspage = ops.page_from_device(session, ([(ps_pk.value, ops.TG_BIGINT),
+                                             (ps_sk.value, ops.TG_BIGINT),
+                                             (ps_sc.value, ops.TG_BIGINT)], n))
+    j1 = ops.lookup_join(session, bridge_p, [ops.TG_BIGINT, ops.TG_BIGINT,
+                                             ops.TG_BIGINT], [0], [0, 1, 2])
+    j1.add_input(pspage)
+    j1.finish()
+    ps_parts = _take_device_page(session, j1)
+    j2 = ops.lookup_join(session, bridge_s, [ops.TG_BIGINT, ops.TG_BIGINT,
+                                             ops.TG_BIGINT], [1], [0, 1, 2])
+    j2.add_input(ps_parts)
+    j2.finish()
+    rows = _take_device_page(session, j2)          # (pk, sk, cost, nk, ab)
+    agg = ops.hash_aggregation(session, [0], [ops.TG_BIGINT],
+                               [(ops.AGG_MIN_I64, 2)])
+    agg.add_input(rows)
+    agg.finish()
+    mins = _take_device_page(session, agg)
+    bridge_m = ops.JoinBridge(session)
+    bm = ops.hash_builder(session, bridge_m, [ops.TG_BIGINT, ops.TG_BIGINT],
+                          [0], [1])
+    bm.add_input(mins)
+    bm.drain()
+    j3 = ops.lookup_join(session, bridge_m,
+                         [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
+                          ops.TG_TINYINT, ops.TG_BIGINT], [0],
+                         [0, 1, 2, 3, 4])
+    j3.add_input(rows)
+    j3.finish()
+    withmin = _take_device_page(session, j3)
+    ff = ops.filter_project(session, ops.expr(("col", 2), ("col", 5), "eq"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                             ops.expr(("col", 3)), ops.expr(("col", 4))],
+                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_TINYINT,
+                             ops.TG_BIGINT])
+    ff.add_input(withmin)
+    pages = ff.drain()
+    elapsed = time.time() - t0
+    for op in (fs, bs, fp, bp, j1, j2, agg, bm, j3, ff):
+        op.close()
+    for br in (bridge_s, bridge_p, bridge_m):
+        br.close()
+    for p in (p_pk, p_ty, p_sz, ps_pk, ps_sk, ps_sc, s_sk, s_nk, s_ab):
+        _device_free(session, p)
+    out = pages[0]
+    pk = np.asarray(out[0]["values"]).astype(np.int64)
+    sk = np.asarray(out[1]["values"]).astype(np.int64)
+    nk = np.asarray(out[2]["values"]).astype(np.int64)
+    ab = np.asarray(out[3]["values"]).astype(np.int64)
+    rows_h = sorted(zip(ab.tolist(), [NATIONS[x] for x in nk], sk.tolist(),
+                        pk.tolist()),
+                    key=lambda r: (-r[0], r[1], r[2], r[3]))[:100]
+    st = _strings(_lib.tg_tpch_supplier_strings, sf, [r[2] for r in rows_h],
+                  ("name", "address", "phone", "comment"))
+    pstr = _strings_part(sf, [r[3] for r in rows_h], ("mfgr",))
+    result = []
+    for i, r in enumerate(rows_h):
+        result.append(dict(acctbal_cents=r[0], s_name=st["name"][i],
+                           n_name=r[1], partkey=r[3], mfgr=pstr["mfgr"][i],
+                           address=st["address"][i], phone=st["phone"][i],
+                           comment=st["comment"][i]))
+    return dict(rows=result, elapsed=elapsed)
+
+
+def q10_gpu(session, sf, order_count=None, limit=20):
+    """TPC-H Q10 (returned items): revenue per customer over returnflag R
+    lines of 1993-10..1994-01 orders; top `limit` by revenue desc."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    bufs = _gen_orders3(session, sf, order_count, ("ok", "ck", "od"))
+    li = session.tpch_lineitem(sf, 1, order_count, with_orderkey=True)
+    t0 = time.time()
+    opage = ops.page_from_device(session, ([(bufs["ok"].value, ops.TG_BIGINT),
+                                            (bufs["ck"].value, ops.TG_BIGINT),
+                                            (bufs["od"].value, ops.TG_INTEGER)],
+                                           order_count))
+    fo = ops.filter_project(session,
+                            ops.expr(("col", 2), ("i64", _D(1993, 10, 1)), "ge",
+                                     ("col", 2), ("i64", _D(1994, 1, 1)), "lt",
+                                     "and"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                            [ops.TG_BIGINT, ops.TG_BIGINT])
+    fo.add_input(opage)
+    fo.finish()
+    owin = _take_device_page(session, fo)
+    bridge = ops.JoinBridge(session)
+    b = ops.hash_builder(session, bridge, [ops.TG_BIGINT, ops.TG_BIGINT],
+                         [0], [1])
+    b.add_input(owin)
+    b.drain()
+    lpage = ops.page_from_device(session, ([(li.orderkey, ops.TG_BIGINT),
+                                            (li.returnflag, ops.TG_TINYINT),
+                                            (li.extendedprice, ops.TG_DOUBLE),
+                                            (li.discount, ops.TG_DOUBLE)],
+                                           li.row_count))
+    fl = ops.filter_project(session, ops.expr(("col", 1), ("i64", 2), "eq"),
+                            [ops.expr(("col", 0)),
+                             ops.expr(("col", 2), ("f64", 1.0), ("col", 3),
+                                      "sub", "mul")],
+                            [ops.TG_BIGINT, ops.TG_DOUBLE])
+    fl.add_input(lpage)
+    fl.finish()
+    lr = _take_device_page(session, fl)
+    j = ops.lookup_join(session, bridge, [ops.TG_BIGINT, ops.TG_DOUBLE],
+                        [0], [1])
+    j.add_input(lr)
+    j.finish()
+    joined = _take_device_page(session, j)     # (discprice, custkey)
+    agg = ops.hash_aggregation(session, [1], [ops.TG_BIGINT],
+                               [(ops.AGG_SUM_F64_EXACT, 0, 43)])
+    agg.add_input(joined)
+    agg.finish()
+    groups = _take_device_page(session, agg)
+    top = ops.topn(session, [ops.TG_BIGINT, ops.TG_DOUBLE], [1, 0], [1, 0],
+                   limit)
+    top.add_input(groups)
+    pages = top.drain()
+    elapsed = time.time() - t0
+    for op in (fo, b, fl, j, agg, top):
+        op.close()
+    bridge.close()
+    session.tpch_lineitem_free(li)
+    _free_bufs(session, bufs)
+    out = pages[0]
+    ck = np.asarray(out[0]["values"]).astype(np.int64)
+    rev = np.asarray(out[1]["values"])
+    st = _strings(_lib.tg_tpch_customer_strings, sf, ck,
+                  ("name", "address", "phone", "comment", "acctbal",
+                   "nationkey"))
+    rows = []
+    for i in range(len(ck)):
+        rows.append(dict(custkey=int(ck[i]), name=st["name"][i],
+                         revenue=float(rev[i]),
+                         acctbal_cents=int(st["acctbal"][i]),
+                         n_name=NATIONS[st["nationkey"][i]],
+                         address=st["address"][i], phone=st["phone"][i],
+                         comment=st["comment"][i]))
+    return dict(rows=rows, elapsed=elapsed)
+
+
+def q20_gpu(session, sf, order_count=None, part_count=None):
+    """TPC-H Q20 (potential part promotion): 'forest%' parts; availqty >
+    0.5 * 1994-shipped qty per (part,supplier) — the correlated sum is NULL
+    with no shipments, so the comparison joins INNER; CANADA suppliers,
+    ordered by s_name."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    if part_count is None:
+        part_count = int(200_000 * sf)
+    supp_count = int(10_000 * sf)
+    p_pk = _device_buffer(session, part_count * 8)
+    p_nm = _device_buffer(session, part_count * 5)
+    _check_lib(_lib.tg_tpch_gen_part2(session._h, sf, 1, part_count,
+                                      p_pk, None, None, None, None, p_nm, None))
+    n = part_count * 4
+    ps_pk = _device_buffer(session, n * 8)
+    ps_sk = _device_buffer(session, n * 8)
+    ps_aq = _device_buffer(session, n * 4)
+    _check_lib(_lib.tg_tpch_gen_partsupp(session._h, sf, 1, part_count,
+                                         ps_pk, ps_sk, ps_aq, None))
+    s_sk = _device_buffer(session, supp_count * 8)
+    s_nk = _device_buffer(session, supp_count)
+    _check_lib(_lib.tg_tpch_gen_supplier2(session._h, sf, 1, supp_count,
+                                          s_sk, s_nk, None))
+    li = session.tpch_lineitem(sf, 1, order_count, with_partkey=True,
+                               with_suppkey=True)
+    t0 = time.time()
+    d_ff = _device_buffer(session, part_count)
+    _check_lib(_lib.tg_tpch_part_name_flag(session._h, p_nm, part_count,
+                                           COLOR_FOREST, 1, d_ff))
+    ppage = ops.page_from_device(session, ([(p_pk.value, ops.TG_BIGINT),
+                                            (d_ff.value, ops.TG_TINYINT)],
+                                           part_count))
+    fp = ops.filter_project(session, ops.expr(("col", 1), ("i64", 1), "eq"),
+                            [ops.expr(("col", 0))], [ops.TG_BIGINT])
+    fp.add_input(ppage)
+    fp.finish()
+    forest = _take_device_page(session, fp)
+    bridge_f = ops.JoinBridge(session)
+    bf = ops.set_builder(session, bridge_f, [ops.TG_BIGINT], 0)
+    bf.add_input(forest)
+    bf.drain()
+    # lineitem: 1994 window + forest part, grouped qty per (pk, sk)
+    lpage = ops.page_from_device(session, ([(li.partkey, ops.TG_BIGINT),
+                                            (li.suppkey, ops.TG_BIGINT),
+                                            (li.quantity, ops.TG_DOUBLE),
+                                            (li.shipdate, ops.TG_INTEGER)],
+                                           li.row_count))
+    fl = ops.filter_project(session,
+                            ops.expr(("col", 3), ("i64", _D(1994, 1, 1)), "ge",
+                                     ("col", 3), ("i64", _D(1995, 1, 1)), "lt",
+                                     "and"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                             ops.expr(("col", 2))],
+                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE])
+    fl.add_input(lpage)
+    fl.finish()
+    lwin = _take_device_page(session, fl)
+    sjf = ops.semi_join(session, bridge_f, 0)
+    sjf.add_input(lwin)
+    sjf.finish()
+    lmark = _take_device_page(session, sjf)
+    flm = ops.filter_project(session, ops.expr(("col", 3), ("i64", 1), "eq"),
+                             [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                              ops.expr(("col", 2))],
+                             [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE])
+    flm.add_input(lmark)
+    flm.finish()
+    lsel = _take_device_page(session, flm)
+    agg = ops.hash_aggregation(session, [0, 1], [ops.TG_BIGINT, ops.TG_BIGINT],
+                               [(ops.AGG_SUM_F64, 2)])
+    agg.add_input(lsel)
+    agg.finish()
+    sums = _take_device_page(session, agg)         # (pk, sk, sumq)
+    bridge_q = ops.JoinBridge(session)
+    bq = ops.hash_builder(session, bridge_q,
+                          [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE],
+                          [0, 1], [2])
+    bq.add_input(sums)
+    bq.drain()
+    # partsupp of forest parts joined (INNER) with the sums
+    pspage = ops.page_from_device(session, ([(ps_pk.value, ops.TG_BIGINT),
+                                             (ps_sk.value, ops.TG_BIGINT),
+                                             (ps_aq.value, ops.TG_INTEGER)], n))
+    sjp = ops.semi_join(session, bridge_f, 0)
+    sjp.add_input(pspage)
+    sjp.finish()
+    pmark = _take_device_page(session, sjp)
+    fpm = ops.filter_project(session, ops.expr(("col", 3), ("i64", 1), "eq"),
+                             [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                              ops.expr(("col", 2))],
+                             [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_INTEGER])
+    fpm.add_input(pmark)
+    fpm.finish()
+    ps_f = _take_device_page(session, fpm)
+    jq = ops.lookup_join(session, bridge_q,
+                         [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_INTEGER],
+                         [0, 1], [1, 2])           # emit (sk, aq) + sumq
+    jq.add_input(ps_f)
+    jq.finish()
+    wsum = _take_device_page(session, jq)          # (sk, aq, sumq)
+    fcmp = ops.filter_project(session,
+                              ops.expr(("col", 1), ("f64", 0.5), ("col", 2),
+                                       "mul", "gt"),
+                              [ops.expr(("col", 0))], [ops.TG_BIGINT])
+    fcmp.add_input(wsum)
+    fcmp.finish()
+    goodsk = _take_device_page(session, fcmp)
+    dd = ops.hash_aggregation(session, [0], [ops.TG_BIGINT], [])
+    dd.add_input(goodsk)
+    pages_sk = dd.drain()
+    elapsed = time.time() - t0
+    for op in (fp, bf, fl, sjf, flm, agg, bq, sjp, fpm, jq, fcmp, dd):
+        op.close()
+    bridge_f.close()
+    bridge_q.close()
+    session.tpch_lineitem_free(li)
+    for p in (p_pk, p_nm, ps_pk, ps_sk, ps_aq, d_ff):
+        _device_free(session, p)
+    # CANADA suppliers among the distinct keys (supplier table host-side:
+    # nationkey via strings helper), sorted by name == suppkey
+    sk_all = np.asarray(pages_sk[0][0]["values"]).astype(np.int64)
+    st = _strings(_lib.tg_tpch_supplier_strings, sf, sk_all,
+                  ("nationkey",)) if len(sk_all) else {"nationkey": np.empty(0, np.int32)}
+    keep = sorted(int(k) for k, nk in zip(sk_all, st["nationkey"]) if nk == 3)
+    out = _strings(_lib.tg_tpch_supplier_strings, sf, keep,
+                   ("name", "address"))
+    for p in (s_sk, s_nk):
+        _device_free(session, p)
+    return dict(names=out["name"], addresses=out["address"],
+                suppkeys=keep, elapsed=elapsed)
+
+
+def q9_gpu(session, sf, order_count=None, part_count=None):
+    """TPC-H Q9 (product type profit): '%green%' parts; profit =
+    l_extendedprice*(1-l_discount) - ps_supplycost*l_quantity grouped by
+    (supplier nation, order year). Composite (pk,sk) join against partsupp;
+    year from o_orderdate by epoch-day bins."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    if part_count is None:
+        part_count = int(200_000 * sf)
+    supp_count = int(10_000 * sf)
+    p_pk = _device_buffer(session, part_count * 8)
+    p_nm = _device_buffer(session, part_count * 5)
+    _check_lib(_lib.tg_tpch_gen_part2(session._h, sf, 1, part_count,
+                                      p_pk, None, None, None, None, p_nm, None))
+    n = part_count * 4
+    ps_pk = _device_buffer(session, n * 8)
+    ps_sk = _device_buffer(session, n * 8)
+    ps_sc = _device_buffer(session, n * 8)
+    _check_lib(_lib.tg_tpch_gen_partsupp(session._h, sf, 1, part_count,
+                                         ps_pk, ps_sk, None, ps_sc))
+    s_sk = _device_buffer(session, supp_count * 8)
+    s_nk = _device_buffer(session, supp_count)
+    _check_lib(_lib.tg_tpch_gen_supplier2(session._h, sf, 1, supp_count,
+                                          s_sk, s_nk, None))
+    obufs = _gen_orders3(session, sf, order_count, ("ok", "od"))
+    li = session.tpch_lineitem(sf, 1, order_count, with_orderkey=True,
+                               with_partkey=True, with_suppkey=True)
+    t0 = time.time()
+    d_gf = _device_buffer(session, part_count)
+    _check_lib(_lib.tg_tpch_part_name_flag(session._h, p_nm, part_count,
+                                           COLOR_GREEN, 0, d_gf))
+    ppage = ops.page_from_device(session, ([(p_pk.value, ops.TG_BIGINT),
+                                            (d_gf.value, ops.TG_TINYINT)],
+                                           part_count))
+    fp = ops.filter_project(session, ops.expr(("col", 1), ("i64", 1), "eq"),
+                            [ops.expr(("col", 0))], [ops.TG_BIGINT])
+    fp.add_input(ppage)
+    fp.finish()
+    green = _take_device_page(session, fp)
+    bridge_g = ops.JoinBridge(session)
+    bg = ops.set_builder(session, bridge_g, [ops.TG_BIGINT], 0)
+    bg.add_input(green)
+    bg.drain()
+    # partsupp (green parts): composite build (pk, sk) -> cost
+    pspage = ops.page_from_device(session, ([(ps_pk.value, ops.TG_BIGINT),
+                                             (ps_sk.value, ops.TG_BIGINT),
+                                             (ps_sc.value, ops.TG_BIGINT)], n))
+    sjp = ops.semi_join(session, bridge_g, 0)
+    sjp.add_input(pspage)
+    sjp.finish()
+    psm = _take_device_page(session, sjp)
+    fps = ops.filter_project(session, ops.expr(("col", 3), ("i64", 1), "eq"),
+                             [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                              ops.expr(("col", 2))],
+                             [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT])
+    fps.add_input(psm)
+    fps.finish()
+    ps_green = _take_device_page(session, fps)
+    bridge_ps = ops.JoinBridge(session)
+    bps = ops.hash_builder(session, bridge_ps,
+                           [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT],
+                           [0, 1], [2])
+    bps.add_input(ps_green)
+    bps.drain()
+    # orders: (ok -> year) build
+    opage = ops.page_from_device(session, ([(obufs["ok"].value, ops.TG_BIGINT),
+                                            (obufs["od"].value, ops.TG_INTEGER)],
+                                           order_count))
+    # year = 1992 + bin(od): bins at Jan 1 epoch days
+    jan = [_D(y, 1, 1) for y in range(1993, 2000)]
+    yexpr = [("col", 1), ("i64", jan[0]), "ge"]
+    for j_ in jan[1:]:
+        yexpr += [("col", 1), ("i64", j_), "ge", "add"]
+    fo = ops.filter_project(session, None, [ops.expr(("col", 0)),
+                                            ops.expr(*yexpr)],
+                            [ops.TG_BIGINT, ops.TG_BIGINT])
+    fo.add_input(opage)
+    fo.finish()
+    oyear = _take_device_page(session, fo)         # (ok, yearbin 0..7)
+    bridge_o = ops.JoinBridge(session)
+    bo = ops.hash_builder(session, bridge_o, [ops.TG_BIGINT, ops.TG_BIGINT],
+                          [0], [1])
+    bo.add_input(oyear)
+    bo.drain()
+    # supplier: (sk -> nationkey) build
+    sn = ops.page_from_device(session, ([(s_sk.value, ops.TG_BIGINT),
+                                         (s_nk.value, ops.TG_TINYINT)],
+                                        supp_count))
+    bridge_n = ops.JoinBridge(session)
+    bn = ops.hash_builder(session, bridge_n, [ops.TG_BIGINT, ops.TG_TINYINT],
+                          [0], [1])
+    bn.add_input(sn)
+    bn.drain()
+    # lineitem: green filter, join partsupp (pk,sk), orders, supplier
+    lpage = ops.page_from_device(session, ([(li.partkey, ops.TG_BIGINT),
+                                            (li.suppkey, ops.TG_BIGINT),
+                                            (li.orderkey, ops.TG_BIGINT),
+                                            (li.quantity, ops.TG_DOUBLE),
+                                            (li.extendedprice, ops.TG_DOUBLE),
+                                            (li.discount, ops.TG_DOUBLE)],
+                                           li.row_count))
+    sjl = ops.semi_join(session, bridge_g, 0)
+    sjl.add_input(lpage)
+    sjl.finish()
+    lm = _take_device_page(session, sjl)
+    fl = ops.filter_project(session, ops.expr(("col", 6), ("i64", 1), "eq"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                             ops.expr(("col", 2)), ops.expr(("col", 3)),
+                             ops.expr(("col", 4), ("f64", 1.0), ("col", 5),
+                                      "sub", "mul")],
+                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
+                             ops.TG_DOUBLE, ops.TG_DOUBLE])
+    fl.add_input(lm)
+    fl.finish()
+    lg = _take_device_page(session, fl)     # (pk, sk, ok, qty, rev)
+    jps = ops.lookup_join(session, bridge_ps,
+                          [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
+                           ops.TG_DOUBLE, ops.TG_DOUBLE], [0, 1],
+                          [1, 2, 3, 4])
+    jps.add_input(lg)
+    jps.finish()
+    wcost = _take_device_page(session, jps)  # (sk, ok, qty, rev, cost_cents)
+    jo = ops.lookup_join(session, bridge_o,
+                         [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE,
+                          ops.TG_DOUBLE, ops.TG_BIGINT], [1],
+                         [0, 2, 3, 4])
+    jo.add_input(wcost)
+    jo.finish()
+    wyear = _take_device_page(session, jo)   # (sk, qty, rev, cost, yearbin)
+    jn = ops.lookup_join(session, bridge_n,
+                         [ops.TG_BIGINT, ops.TG_DOUBLE, ops.TG_DOUBLE,
+                          ops.TG_BIGINT, ops.TG_BIGINT], [0],
+                         [1, 2, 3, 4])
+    jn.add_input(wyear)
+    jn.finish()
+    wn = _take_device_page(session, jn)      # (qty, rev, cost, yearbin, nk)
+    # amount = rev - (cost/100.0) * qty  (same IEEE ops as the reference:
+    # supplycost DOUBLE times quantity DOUBLE)
+    fa = ops.filter_project(session, None,
+                            [ops.expr(("col", 3)), ops.expr(("col", 4)),
+                             ops.expr(("col", 1),
+                                      ("col", 2), ("f64", 100.0), "div",
+                                      ("col", 0), "mul", "sub")],
+                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE])
+    fa.add_input(wn)
+    fa.finish()
+    amounts = _take_device_page(session, fa)   # (yearbin, nk, amount)
+    agg = ops.hash_aggregation(session, [1, 0], [ops.TG_BIGINT, ops.TG_BIGINT],
+                               [(ops.AGG_SUM_F64_EXACT, 2, 40)])
+    agg.add_input(amounts)
+    pages = agg.drain()
+    elapsed = time.time() - t0
+    for op in (fp, bg, sjp, fps, bps, fo, bo, bn, sjl, fl, jps, jo, jn, fa, agg):
+        op.close()
+    for br in (bridge_g, bridge_ps, bridge_o, bridge_n):
+        br.close()
+    session.tpch_lineitem_free(li)
+    _free_bufs(session, obufs)
+    for p in (p_pk, p_nm, ps_pk, ps_sk, ps_sc, s_sk, s_nk, d_gf):
+        _device_free(session, p)
+    out = pages[0]
+    nk = np.asarray(out[0]["values"]).astype(np.int64)
+    yb = np.asarray(out[1]["values"]).astype(np.int64)
+    amt = np.asarray(out[2]["values"])
+    rows = sorted(((NATIONS[nk[i]], int(1992 + yb[i]), float(amt[i]))
+                   for i in range(len(nk))), key=lambda r: (r[0], -r[1]))
+    return dict(rows=rows, elapsed=elapsed)
+
+
+def q21_gpu(session, sf, order_count=None, limit=100):
+    """TPC-H Q21 (suppliers who kept orders waiting): per F-status order,
+    the EXISTS/NOT EXISTS pair reduces to min/max supplier aggregates
+    (>=2 distinct suppliers overall <=> minAll != maxAll; exactly one
+    distinct late supplier <=> minLate == maxLate); numwait counts late
+    LINES of qualifying orders per SAUDI ARABIA supplier. Lineitem is
+    orderkey-clustered, so both aggregations stream (run-segmented)."""
+    if order_count is None:
+        order_count = int(1_500_000 * sf)
+    supp_count = int(10_000 * sf)
+    obufs = _gen_orders3(session, sf, order_count, ("ok", "status"))
+    s_sk = _device_buffer(session, supp_count * 8)
+    s_nk = _device_buffer(session, supp_count)
+    _check_lib(_lib.tg_tpch_gen_supplier2(session._h, sf, 1, supp_count,
+                                          s_sk, s_nk, None))
+    li = session.tpch_lineitem(sf, 1, order_count, with_orderkey=True,
+                               with_suppkey=True, with_dates=True)
+    t0 = time.time()
+    lpage = ops.page_from_device(session, ([(li.orderkey, ops.TG_BIGINT),
+                                            (li.suppkey, ops.TG_BIGINT),
+                                            (li.commitdate, ops.TG_INTEGER),
+                                            (li.receiptdate, ops.TG_INTEGER)],
+                                           li.row_count))
+    # all lines: min/max supplier per order (streaming: orderkey-sorted)
+    a_all = ops.streaming_aggregation(session, 0, [(ops.AGG_MIN_I64, 1),
+                                                   (ops.AGG_MAX_I64, 1)])
+    a_all.add_input(lpage)
+    a_all.finish()
+    allmm = _take_device_page(session, a_all)       # (ok, minA, maxA)
+    # late lines only (filter preserves order -> still clustered)
+    fl = ops.filter_project(session, ops.expr(("col", 3), ("col", 2), "gt"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                            [ops.TG_BIGINT, ops.TG_BIGINT])
+    fl.add_input(lpage)
+    fl.finish()
+    late = _take_device_page(session, fl)
+    a_late = ops.streaming_aggregation(session, 0, [(ops.AGG_MIN_I64, 1),
+                                                    (ops.AGG_MAX_I64, 1),
+                                                    (ops.AGG_COUNT_STAR, -1)])
+    a_late.add_input(late)
+    a_late.finish()
+    latemm = _take_device_page(session, a_late)     # (ok, minL, maxL, nlate)
+    # join: late orders with their all-lines min/max
+    bridge_a = ops.JoinBridge(session)
+    ba = ops.hash_builder(session, bridge_a,
+                          [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT],
+                          [0], [1, 2])
+    ba.add_input(allmm)
+    ba.drain()
+    j1 = ops.lookup_join(session, bridge_a,
+                         [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
+                          ops.TG_BIGINT], [0], [0, 1, 2, 3])
+    j1.add_input(latemm)
+    j1.finish()
+    both = _take_device_page(session, j1)  # (ok, minL, maxL, nlate, minA, maxA)
+    # F-status orders
+    opage = ops.page_from_device(session, ([(obufs["ok"].value, ops.TG_BIGINT),
+                                            (obufs["status"].value, ops.TG_TINYINT)],
+                                           order_count))
+    fo = ops.filter_project(session, ops.expr(("col", 1), ("i64", 0), "eq"),
+                            [ops.expr(("col", 0))], [ops.TG_BIGINT])
+    fo.add_input(opage)
+    fo.finish()
+    fords = _take_device_page(session, fo)
+    bridge_o = ops.JoinBridge(session)
+    bo = ops.set_builder(session, bridge_o, [ops.TG_BIGINT], 0)
+    bo.add_input(fords)
+    bo.drain()
+    sj = ops.semi_join(session, bridge_o, 0)
+    sj.add_input(both)
+    sj.finish()
+    marked = _take_device_page(session, sj)
+    # qualify: F-status AND minA != maxA AND minL == maxL
+    fq = ops.filter_project(session,
+                            ops.expr(("col", 6), ("i64", 1), "eq",
+                                     ("col", 4), ("col", 5), "ne", "and",
+                                     ("col", 1), ("col", 2), "eq", "and"),
+                            [ops.expr(("col", 1)), ops.expr(("col", 3))],
+                            [ops.TG_BIGINT, ops.TG_BIGINT])
+    fq.add_input(marked)
+    fq.finish()
+    qual = _take_device_page(session, fq)           # (suppkey, nlate)
+    agg = ops.hash_aggregation(session, [0], [ops.TG_BIGINT],
+                               [(ops.AGG_SUM_I64, 1)])
+    agg.add_input(qual)
+    agg.finish()
+    counts = _take_device_page(session, agg)        # (sk, numwait)
+    # SAUDI ARABIA suppliers (nation 20)
+    spage = ops.page_from_device(session, ([(s_sk.value, ops.TG_BIGINT),
+                                            (s_nk.value, ops.TG_TINYINT)],
+                                           supp_count))
+    fs = ops.filter_project(session, ops.expr(("col", 1), ("i64", 20), "eq"),
+                            [ops.expr(("col", 0))], [ops.TG_BIGINT])
+    fs.add_input(spage)
+    fs.finish()
+    saudi = _take_device_page(session, fs)
+    bridge_s = ops.JoinBridge(session)
+    bs = ops.set_builder(session, bridge_s, [ops.TG_BIGINT], 0)
+    bs.add_input(saudi)
+    bs.drain()
+    sj2 = ops.semi_join(session, bridge_s, 0)
+    sj2.add_input(counts)
+    sj2.finish()
+    cm = _take_device_page(session, sj2)
+    ff = ops.filter_project(session, ops.expr(("col", 2), ("i64", 1), "eq"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                            [ops.TG_BIGINT, ops.TG_BIGINT])
+    ff.add_input(cm)
+    ff.finish()
+    sc = _take_device_page(session, ff)
+    # ORDER BY numwait DESC, s_name (= suppkey) ASC LIMIT 100
+    top = ops.topn(session, [ops.TG_BIGINT, ops.TG_BIGINT], [1, 0], [1, 0],
+                   limit)
+    top.add_input(sc)
+    pages = top.drain()
+    elapsed = time.time() - t0
+    for op in (a_all, fl, a_late, ba, j1, fo, bo, sj, fq, agg, fs, bs, sj2,
+               ff, top):
+        op.close()
+    for br in (bridge_a, bridge_o, bridge_s):
+        br.close()
+    session.tpch_lineitem_free(li)
+    _free_bufs(session, obufs)
+    for p in (s_sk, s_nk):
+        _device_free(session, p)
+    out = pages[0]
+    sk = np.asarray(out[0]["values"]).astype(np.int64)
+    nw = np.asarray(out[1]["values"]).astype(np.int64)
+    names = [f"Supplier#{k:09d}" for k in sk]
+    return dict(names=names, numwait=nw, suppkeys=sk, elapsed=elapsed)
