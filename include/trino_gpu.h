@@ -333,6 +333,10 @@ int32_t tg_tpch_nation_region(int32_t nationkey);
 /* p_name predicate flags over the 5 color ids per part (green=33 forest=28) */
 tg_status tg_tpch_part_name_flag(tg_session*, const uint8_t* d_name_ids,
     int64_t n, int32_t color_id, int32_t first_only, uint8_t* d_flags);
+/* stream timer (HIP events on the session stream) for bench rooflines */
+tg_status tg_timer_start(tg_session*);
+tg_status tg_timer_stop(tg_session*, double* elapsed_ms);
+
 /* coarse device-memory accounting (LocalMemoryContext analog): bytes ever
  * pooled and bytes currently cached; live = total - cached */
 tg_status tg_session_memory(tg_session*, int64_t* total_bytes, int64_t* cached_bytes);

@@ -132,3 +132,22 @@ extern "C" tg_status tg_session_memory(tg_session* s, int64_t* total_bytes,
     if (cached_bytes) *cached_bytes = cached;
     return TG_OK;
 }
+
+/* generic stream timer over the session's HIP events: bracket operator
+ * calls (all kernels launch on s->stream) for per-phase rooflines in
+ * bench.py. start/stop pairs must not nest. */
+extern "C" tg_status tg_timer_start(tg_session* s)
+{
+    TG_HIP_CHECK(hipEventRecord(s->ev_start, s->stream));
+    return TG_OK;
+}
+
+extern "C" tg_status tg_timer_stop(tg_session* s, double* ms)
+{
+    TG_HIP_CHECK(hipEventRecord(s->ev_stop, s->stream));
+    TG_HIP_CHECK(hipEventSynchronize(s->ev_stop));
+    float f = 0.f;
+    TG_HIP_CHECK(hipEventElapsedTime(&f, s->ev_start, s->ev_stop));
+    *ms = (double)f;
+    return TG_OK;
+}

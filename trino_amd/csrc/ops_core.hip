@@ -162,7 +162,7 @@ tg_status tg_compile_expr(tg_session* s, const tg_expr* e, ExprProgram* out)
 {
     out->insts.assign(e->insts, e->insts + e->count);
     out->count = e->count;
-    if (out->count > 64) { TG_SET_ERR("expression too long (>64 insts)"); return TG_ERR_UNSUPPORTED; }
+    if (out->count > 128) { TG_SET_ERR("expression too long (>128 insts)"); return TG_ERR_UNSUPPORTED; }
     TG_POOL_ALLOC(s, &out->d_insts, out->count * sizeof(tg_expr_inst));
     TG_HIP_CHECK(hipMemcpyAsync(out->d_insts, out->insts.data(),
                                 out->count * sizeof(tg_expr_inst),
