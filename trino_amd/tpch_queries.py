@@ -131,9 +131,11 @@ def q3_execute(session, inp, download_groups=True, cust_key_exchange=None):
 
     j2 = ops.lookup_join(session, bridge2, [ops.TG_BIGINT, ops.TG_DOUBLE],
                          [0], [0, 1])      # emit orderkey, discprice (+ build orderdate)
-    j2.add_input(li_sel)
+    probe_ms = _timed(session, lambda: j2.add_input(li_sel))
     j2.finish()
     joined = _take_device_page(session, j2)
+    probe_rows = li_sel.position_count
+    match_rows = joined.position_count
 
     # revenue = sum(ep*(1-disc)): products >= ~810 (= 2^9 < v < 2^11) sit on
     # the 2^-43 grid -> TG_AGG_SUM_F64_EXACT is exact AND order-independent
@@ -166,15 +168,31 @@ def q3_execute(session, inp, download_groups=True, cust_key_exchange=None):
     out = pages_full
     if out is None:
         return dict(orderkey=None, orderdate=None, revenue=None,
-                    top10=top10, elapsed=elapsed)
+                    top10=top10, elapsed=elapsed, probe_ms=probe_ms,
+                    probe_rows=probe_rows, match_rows=match_rows)
     orderkey = out[0]["values"]
     orderdate = out[1]["values"]
     revenue = out[2]["values"]
     return dict(orderkey=orderkey, orderdate=orderdate, revenue=revenue,
-                top10=top10, elapsed=elapsed)
+                top10=top10, elapsed=elapsed, probe_ms=probe_ms,
+                probe_rows=probe_rows, match_rows=match_rows)
 
 
 # ---- small device-buffer helpers over the C ABI ----
+_lib.tg_timer_start.restype = ctypes.c_int
+_lib.tg_timer_start.argtypes = [ctypes.c_void_p]
+_lib.tg_timer_stop.restype = ctypes.c_int
+_lib.tg_timer_stop.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+
+
+def _timed(session, fn):
+    ms = ctypes.c_double()
+    _check_lib(_lib.tg_timer_start(session._h))
+    fn()
+    _check_lib(_lib.tg_timer_stop(session._h, ctypes.byref(ms)))
+    return ms.value
+
+
 _lib.tg_device_malloc.restype = ctypes.c_int
 _lib.tg_device_malloc.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int64]
 _lib.tg_device_free.restype = ctypes.c_int
