@@ -333,6 +333,18 @@ int32_t tg_tpch_nation_region(int32_t nationkey);
 /* p_name predicate flags over the 5 color ids per part (green=33 forest=28) */
 tg_status tg_tpch_part_name_flag(tg_session*, const uint8_t* d_name_ids,
     int64_t n, int32_t color_id, int32_t first_only, uint8_t* d_flags);
+/* fused dynamic filter (sql/gen/columnar/DynamicPageFilter.java +
+ * operator/DynamicFilterSourceOperator.java analog): request a dense-range
+ * membership bitmap over the build keys BEFORE the build finishes, then
+ * create the probe-side scan with the bridge attached — the membership test
+ * runs inside the scan's filter kernel after the static predicate. Best
+ * effort like the reference: generic/multi-channel keys or ranges beyond
+ * 2^33 skip the bitmap and the scan runs the static filter alone. */
+tg_status tg_join_bridge_request_bitmap(tg_join_bridge*);
+tg_status tg_filter_project_create_df(tg_session*, const tg_expr* filter,
+    const tg_expr* projections, const int32_t* proj_out_types, int32_t n_proj,
+    tg_join_bridge* df_bridge, int32_t df_key_channel, tg_operator** out);
+
 /* stream timer (HIP events on the session stream) for bench rooflines */
 tg_status tg_timer_start(tg_session*);
 tg_status tg_timer_stop(tg_session*, double* elapsed_ms);

@@ -95,9 +95,11 @@ tg_status tg_compile_expr(tg_session* s, const tg_expr* e, ExprProgram* out);
 void tg_free_expr(tg_session* s, ExprProgram* p);
 
 /* kernels (implemented in ops_*.hip) */
+struct DF;   /* fused dynamic filter spec (ops_filter.hip) */
 tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page,
                      const tg_selected* input_sel,
-                     int32_t** d_positions_out, int32_t* count_out);
+                     int32_t** d_positions_out, int32_t* count_out,
+                     const DF* df = nullptr);
 tg_status run_project(tg_session* s, const ExprProgram& proj, tg_type out_type,
                       const DevPage& page, const int32_t* d_positions, int32_t count,
                       DevBlock* out);

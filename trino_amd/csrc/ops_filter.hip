@@ -17,6 +17,23 @@
 
 struct KCol { const void* data; const uint64_t* valid; int32_t type; int32_t _pad; };
 
+/* fused dynamic filter (sql/gen/columnar/DynamicPageFilter.java analog):
+ * membership bitmap over a dense build-key range, tested INSIDE the scan
+ * kernels after the static predicate passes (no extra materialized pass) */
+struct DF { const uint64_t* bm; int64_t mn, mx; int32_t col; int32_t _pad; };
+
+__device__ static inline bool df_test(const DF& df, const KCol* cols, int64_t i)
+{
+    if (!df.bm) return true;
+    const KCol& c = cols[df.col];
+    if (c.valid && !((c.valid[i >> 6] >> (i & 63)) & 1)) return false;
+    int64_t k = (c.type == TG_BIGINT) ? ((const int64_t*)c.data)[i]
+                                      : (int64_t)((const int32_t*)c.data)[i];
+    if (k < df.mn || k > df.mx) return false;
+    int64_t b = k - df.mn;
+    return (df.bm[b >> 6] >> (b & 63)) & 1;
+}
+
 #define MAX_STACK 6
 
 /* Typed stack value: BIGINT/INTEGER/DATE/… stay in an exact int64 lane
@@ -155,7 +172,8 @@ __device__ static inline int64_t sel_row(int has_list, const int32_t* list,
  * compare exactly in int64 (keys above 2^53 must not round). */
 __global__ void k_filter_cmp(KCol col, int op, double cval, int64_t icval, int int_mode,
                              int has_list, const int32_t* list, int32_t offset,
-                             int64_t n, uint8_t* __restrict__ flags)
+                             int64_t n, uint8_t* __restrict__ flags,
+                             DF df, const KCol* __restrict__ all_cols)
 {
     int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -183,7 +201,9 @@ __global__ void k_filter_cmp(KCol col, int op, double cval, int64_t icval, int i
                 default: r = v.f != cval; break;
             }
         }
-        flags[k] = (!v.null && r) ? 1 : 0;
+        bool pass = !v.null && r;
+        if (pass) pass = df_test(df, all_cols, i);
+        flags[k] = pass ? 1 : 0;
     }
 }
 
@@ -230,7 +250,7 @@ __device__ static inline bool fterm_cmpi(int op, int64_t a, int64_t b)
 __global__ void k_filter_terms(FTerms ft, const KCol* __restrict__ cols,
                                int has_list, const int32_t* __restrict__ list,
                                int32_t offset, int64_t n,
-                               uint8_t* __restrict__ flags)
+                               uint8_t* __restrict__ flags, DF df)
 {
     int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -256,6 +276,7 @@ __global__ void k_filter_terms(FTerms ft, const KCol* __restrict__ cols,
                                                        : (a.f == f.c1 || a.f == f.c2)); break;
             }
         }
+        if (pass) pass = df_test(df, cols, i);
         flags[k] = pass ? 1 : 0;
     }
 }
@@ -327,13 +348,22 @@ static int parse_fterms(const ExprProgram& pred, const std::vector<KCol>& cols, 
 
 __global__ void k_filter_flags(const tg_expr_inst* prog, int count, const KCol* cols,
                                int has_list, const int32_t* list, int32_t offset,
-                               int64_t n, uint8_t* __restrict__ flags)
+                               int64_t n, uint8_t* __restrict__ flags, DF df)
 {
     int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; k < n; k += stride) {
-        TVal v = eval_expr(prog, count, cols, sel_row(has_list, list, offset, k));
-        flags[k] = (!v.null && tval_truthy(v)) ? 1 : 0;
+        int64_t i = sel_row(has_list, list, offset, k);
+        bool pass;
+        if (count > 0) {
+            TVal v = eval_expr(prog, count, cols, i);
+            pass = !v.null && tval_truthy(v);
+        }
+        else {
+            pass = true;    /* dynamic-filter-only scan */
+        }
+        if (pass) pass = df_test(df, cols, i);
+        flags[k] = pass ? 1 : 0;
     }
 }
 
@@ -621,8 +651,11 @@ __global__ void k_fp_write(const tg_expr_inst* prog, int count, const KCol* cols
 
 tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page,
                      const tg_selected* input_sel,
-                     int32_t** d_positions_out, int32_t* count_out)
+                     int32_t** d_positions_out, int32_t* count_out,
+                     const DF* df_in)
 {
+    DF df{nullptr, 0, 0, 0, 0};
+    if (df_in) df = *df_in;
     /* device column descriptors */
     std::vector<KCol> cols(page.blocks.size());
     for (size_t i = 0; i < page.blocks.size(); i++)
@@ -657,15 +690,15 @@ tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page
         int int_mode = is_i64 && type_is_int(cols[pred.insts[0].arg0].type);
         hipLaunchKernelGGL(k_filter_cmp, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
                            cols[pred.insts[0].arg0], pred.insts[2].op, cval, icval, int_mode,
-                           has_list, d_list, offset, n, d_flags);
+                           has_list, d_list, offset, n, d_flags, df, d_cols);
     }
-    else if (FTerms ft{}; parse_fterms(pred, cols, &ft) > 0) {
+    else if (FTerms ft{}; pred.count > 0 && parse_fterms(pred, cols, &ft) > 0) {
         hipLaunchKernelGGL(k_filter_terms, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
-                           ft, d_cols, has_list, d_list, offset, n, d_flags);
+                           ft, d_cols, has_list, d_list, offset, n, d_flags, df);
     }
     else {
         hipLaunchKernelGGL(k_filter_flags, dim3(tg_grid_for(n)), dim3(TG_BLOCK), 0, s->stream,
-                           pred.d_insts, pred.count, d_cols, has_list, d_list, offset, n, d_flags);
+                           pred.d_insts, pred.count, d_cols, has_list, d_list, offset, n, d_flags, df);
     }
     TG_HIP_CHECK(hipGetLastError());
 
@@ -922,11 +955,15 @@ tg_status run_project(tg_session* s, const ExprProgram& proj, tg_type out_type,
 }
 
 /* ---- FilterAndProject operator ---- */
+bool tg_bridge_df(tg_join_bridge* b, const uint64_t** bm, int64_t* mn, int64_t* mx);
+
 struct FilterProjectOp : tg_operator {
     ExprProgram filter{};
     bool has_filter = false;
     std::vector<ExprProgram> projections;
     std::vector<tg_type> out_types;
+    tg_join_bridge* df_bridge = nullptr;    /* fused dynamic filter source */
+    int32_t df_channel = -1;
 
     tg_status add_input(const tg_page* page) override
     {
@@ -940,11 +977,19 @@ struct FilterProjectOp : tg_operator {
         DevPage in;
         tg_status st = tg_upload_page(s, page, &in);
         if (st != TG_OK) return st;
-        if (has_filter && !fused) {
+        DF df{nullptr, 0, 0, 0, 0};
+        bool have_df = false;
+        if (df_bridge) {
+            have_df = tg_bridge_df(df_bridge, &df.bm, &df.mn, &df.mx);
+            df.col = df_channel;
+            /* best effort (like the reference): no bitmap -> plain filter */
+        }
+        if ((has_filter || have_df) && !fused) {
             /* unfused reference path: selection vector + per-projection gather */
             int32_t* d_pos = nullptr;
             int32_t count = 0;
-            st = run_filter(s, filter, in, nullptr, &d_pos, &count);
+            st = run_filter(s, filter, in, nullptr, &d_pos, &count,
+                            have_df ? &df : nullptr);
             if (st != TG_OK) { tg_free_page(s, &in); return st; }
             DevPage outp;
             outp.n = count;
@@ -1069,6 +1114,20 @@ static tg_status check_expr_depth(const tg_expr* e)
     }
     if (maxsp > MAX_STACK) { TG_SET_ERR("expr stack depth %d > %d", maxsp, MAX_STACK); return TG_ERR_UNSUPPORTED; }
     if (sp != 1) { TG_SET_ERR("expr does not reduce to one value"); return TG_ERR_INVALID_ARG; }
+    return TG_OK;
+}
+
+extern "C" tg_status tg_filter_project_create_df(tg_session* s,
+    const tg_expr* filter, const tg_expr* projections,
+    const int32_t* proj_out_types, int32_t n_proj, tg_join_bridge* df_bridge,
+    int32_t df_key_channel, tg_operator** out)
+{
+    tg_status st = tg_filter_project_create(s, filter, projections,
+                                            proj_out_types, n_proj, out);
+    if (st != TG_OK) return st;
+    auto* op = static_cast<FilterProjectOp*>(*out);
+    op->df_bridge = df_bridge;
+    op->df_channel = df_key_channel;
     return TG_OK;
 }
 

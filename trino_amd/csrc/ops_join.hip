@@ -144,6 +144,11 @@ struct tg_join_bridge {
     std::vector<int32_t> key_channels;
     KColH* d_bkeys = nullptr;        /* device array over build key channels */
     bool built = false;
+    /* fused dynamic filter (DynamicPageFilter.java analog): when requested
+     * BEFORE build finish, a dense-range key membership bitmap is built
+     * alongside the index and pushed into probe-side scan kernels. Best
+     * effort like the reference: generic keys / huge ranges skip it. */
+    bool want_bitmap = false;
 };
 
 /* IncrementalLoadFactorHashArraySizeSupplier (multiplier 1) */
@@ -946,6 +951,28 @@ struct HashBuilderOp : tg_operator {
                 bitmap_built = true;
             }
         }
+        /* dynamic-filter bitmap alongside the real index (requested via
+         * tg_join_bridge_request_bitmap; single int64 key, dense range) */
+        if (!set_only && bridge->want_bitmap && !t.generic && total_rows > 0) {
+            tg_status mst = compute_key_range();
+            if (mst != TG_OK) return mst;
+            int64_t range = bridge->key_rows > 0
+                          ? bridge->key_max - bridge->key_min + 1 : 0;
+            if (range > 0 && range <= (1ll << 33)) {
+                int64_t words = (range + 63) / 64;
+                uint64_t* bm = nullptr;
+                TG_POOL_ALLOC(s, &bm, words * 8);
+                TG_HIP_CHECK(hipMemsetAsync(bm, 0, words * 8, s->stream));
+                hipLaunchKernelGGL(k_set_bits, dim3(tg_grid_for(total_rows)),
+                                   dim3(TG_BLOCK), 0, s->stream, t.keys,
+                                   t.key_valid, total_rows, bridge->key_min, bm);
+                TG_HIP_CHECK(hipGetLastError());
+                TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+                t.set_bitmap = bm;
+                t.set_min = bridge->key_min;
+                t.set_max = bridge->key_max;
+            }
+        }
         if (!bitmap_built && use_csr && total_rows > 0) {
             t.csr = 1;
             int64_t region_slots = t.capacity < JREG_SLOTS ? t.capacity : JREG_SLOTS;
@@ -1502,4 +1529,23 @@ extern "C" tg_status tg_join_bridge_key_range(tg_join_bridge* b, int64_t* key_mi
     if (key_max) *key_max = b->key_max;
     if (key_rows) *key_rows = b->key_rows;
     return TG_OK;
+}
+
+/* request a probe-side dynamic-filter bitmap from this bridge's build
+ * (must be called before the builder's finish) */
+extern "C" tg_status tg_join_bridge_request_bitmap(tg_join_bridge* b)
+{
+    if (!b) { TG_SET_ERR("null bridge"); return TG_ERR_INVALID_ARG; }
+    b->want_bitmap = true;
+    return TG_OK;
+}
+
+/* internal: fetch the bitmap for the filter operator (ops_filter.hip) */
+bool tg_bridge_df(tg_join_bridge* b, const uint64_t** bm, int64_t* mn, int64_t* mx)
+{
+    if (!b || !b->built || !b->t.set_bitmap) return false;
+    *bm = b->t.set_bitmap;
+    *mn = b->t.set_min;
+    *mx = b->t.set_max;
+    return true;
 }

@@ -520,3 +520,31 @@ def pool_like_flags(session, d_offs, d_lens, n, pattern, d_flags):
 def varchar_like_flags(session, d_bytes, d_offsets, n, pattern, d_flags):
     _check(_lib.tg_varchar_like_flags(session._h, d_bytes, d_offsets, n,
                                       pattern.encode(), d_flags))
+
+
+_lib.tg_join_bridge_request_bitmap.restype = ctypes.c_int
+_lib.tg_join_bridge_request_bitmap.argtypes = [ctypes.c_void_p]
+_lib.tg_filter_project_create_df.restype = ctypes.c_int
+
+
+def filter_project_df(session, filter_expr, projections, out_types,
+                      df_bridge, df_key_channel):
+    """filter_project with a fused dynamic filter from a join bridge
+    (request_bitmap must have been called on the bridge before build)."""
+    h = ctypes.c_void_p()
+    n = len(projections)
+    proj_arr = (TgExpr * n)()
+    for i, p in enumerate(projections):
+        proj_arr[i] = p
+    ot = _i32arr(out_types or [TG_DOUBLE] * n)
+    _check(_lib.tg_filter_project_create_df(
+        session._h, ctypes.byref(filter_expr) if filter_expr else None,
+        proj_arr, ot.ctypes.data, n, df_bridge._h, df_key_channel,
+        ctypes.byref(h)))
+    op = Operator(session, h)
+    op._keep = (proj_arr, ot, filter_expr, projections)
+    return op
+
+
+def request_bitmap(bridge):
+    _check(_lib.tg_join_bridge_request_bitmap(bridge._h))

@@ -109,6 +109,7 @@ def q3_execute(session, inp, download_groups=True, cust_key_exchange=None):
     orders_building = _take_device_page(session, j1)
 
     bridge2 = ops.JoinBridge(session)
+    ops.request_bitmap(bridge2)   # dynamic filter -> pushed into the scan below
     b2 = ops.hash_builder(session, bridge2, [ops.TG_BIGINT, ops.TG_INTEGER],
                           [0], [1])        # key orderkey; build output orderdate
     b2.add_input(orders_building)
@@ -120,11 +121,13 @@ def q3_execute(session, inp, download_groups=True, cust_key_exchange=None):
                                             (li.extendedprice, ops.TG_DOUBLE),
                                             (li.discount, ops.TG_DOUBLE)],
                                            li.row_count))
-    f3 = ops.filter_project(session,
-                            ops.expr(("col", 1), ("i64", DATE_1995_03_15), "gt"),
-                            [ops.expr(("col", 0)),
-                             ops.expr(("col", 2), ("f64", 1.0), ("col", 3), "sub", "mul")],
-                            [ops.TG_BIGINT, ops.TG_DOUBLE])
+    # scan filter + FUSED dynamic filter (orderkey ∈ build keys): cuts the
+    # probe input to BUILDING-customer orders before the join
+    f3 = ops.filter_project_df(session,
+                               ops.expr(("col", 1), ("i64", DATE_1995_03_15), "gt"),
+                               [ops.expr(("col", 0)),
+                                ops.expr(("col", 2), ("f64", 1.0), ("col", 3), "sub", "mul")],
+                               [ops.TG_BIGINT, ops.TG_DOUBLE], bridge2, 0)
     f3.add_input(lpage)
     f3.finish()
     li_sel = _take_device_page(session, f3)
@@ -738,12 +741,25 @@ def q5_gpu(session, sf, order_start=1, order_count=None):
     li = session.tpch_lineitem(sf, order_start, order_count,
                                with_orderkey=True, with_suppkey=True)
     t0 = time.time()
-    # build1: customers (custkey -> nationkey)
-    br1 = ops.JoinBridge(session)
-    b1 = ops.hash_builder(session, br1, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
-    b1.add_input(ops.page_from_device(session, ([(c_ck.value, ops.TG_BIGINT),
+    # build1: ASIA customers only (selective filter FIRST — the same lever
+    # that took Q7 95->38 ms in round 1: 20%% of customers, so the orders
+    # join and everything downstream shrinks 5x)
+    asia_chain = []
+    for _i, _nk in enumerate(ASIA_NATIONS):
+        asia_chain += [("col", 1), ("i64", _nk), "eq"]
+        if _i:
+            asia_chain.append("or")
+    fc = ops.filter_project(session, ops.expr(*asia_chain),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                            [ops.TG_BIGINT, ops.TG_TINYINT])
+    fc.add_input(ops.page_from_device(session, ([(c_ck.value, ops.TG_BIGINT),
                                                  (c_nk.value, ops.TG_TINYINT)],
                                                 cust_count)))
+    fc.finish()
+    asia_cust = _take_device_page(session, fc)
+    br1 = ops.JoinBridge(session)
+    b1 = ops.hash_builder(session, br1, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
+    b1.add_input(asia_cust)
     b1.drain()
     # orders in 1994 -> join customers
     f1 = ops.filter_project(session,
@@ -763,15 +779,18 @@ def q5_gpu(session, sf, order_start=1, order_count=None):
     j1.finish()
     ojoined = _take_device_page(session, j1)    # (orderkey, c_nationkey)
     br2 = ops.JoinBridge(session)
+    ops.request_bitmap(br2)
     b2 = ops.hash_builder(session, br2, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
     b2.add_input(ojoined)
     b2.drain()
-    # lineitem -> (orderkey, suppkey, revenue) -> join orders
-    fp = ops.filter_project(session, None,
-                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
-                             ops.expr(("col", 2), ("f64", 1.0), ("col", 3),
-                                      "sub", "mul")],
-                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE])
+    # lineitem scan with the FUSED dynamic filter (orderkey ∈ 1994-ASIA
+    # orders, ~3%%): revenue is only computed for survivors
+    fp = ops.filter_project_df(session, None,
+                               [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                                ops.expr(("col", 2), ("f64", 1.0), ("col", 3),
+                                         "sub", "mul")],
+                               [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE],
+                               br2, 0)
     fp.add_input(ops.page_from_device(session, ([(li.orderkey, ops.TG_BIGINT),
                                                  (li.suppkey, ops.TG_BIGINT),
                                                  (li.extendedprice, ops.TG_DOUBLE),
@@ -784,12 +803,18 @@ def q5_gpu(session, sf, order_start=1, order_count=None):
     j2.add_input(lsel)
     j2.finish()
     lj = _take_device_page(session, j2)         # (suppkey, rev, c_nk)
-    # join supplier nation
+    # join supplier nation (ASIA suppliers only — inner join drops the rest)
+    fsup = ops.filter_project(session, ops.expr(*asia_chain),
+                              [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                              [ops.TG_BIGINT, ops.TG_TINYINT])
+    fsup.add_input(ops.page_from_device(session, ([(s_sk.value, ops.TG_BIGINT),
+                                                   (s_nk.value, ops.TG_TINYINT)],
+                                                  supp_count)))
+    fsup.finish()
+    asia_supp = _take_device_page(session, fsup)
     br3 = ops.JoinBridge(session)
     b3 = ops.hash_builder(session, br3, [ops.TG_BIGINT, ops.TG_TINYINT], [0], [1])
-    b3.add_input(ops.page_from_device(session, ([(s_sk.value, ops.TG_BIGINT),
-                                                 (s_nk.value, ops.TG_TINYINT)],
-                                                supp_count)))
+    b3.add_input(asia_supp)
     b3.drain()
     j3 = ops.lookup_join(session, br3,
                          [ops.TG_BIGINT, ops.TG_DOUBLE, ops.TG_TINYINT],
@@ -815,7 +840,7 @@ def q5_gpu(session, sf, order_start=1, order_count=None):
     agg.add_input(qual)
     pages = agg.drain()
     elapsed = time.time() - t0
-    for op in (b1, f1, j1, b2, fp, j2, b3, j3, f2, agg):
+    for op in (fc, fsup, b1, f1, j1, b2, fp, j2, b3, j3, f2, agg):
         op.close()
     for br in (br1, br2, br3):
         br.close()
@@ -1001,6 +1026,7 @@ def q8_gpu(session, sf, order_start=1, order_count=None):
     fpart.finish()
     psel = _take_device_page(session, fpart)
     brp = ops.JoinBridge(session)
+    ops.request_bitmap(brp)   # 0.67%% of parts -> 2.5 MB bitmap, L2-resident
     bp = ops.hash_builder(session, brp, [ops.TG_BIGINT], [0], [])
     bp.add_input(psel)
     bp.drain()
@@ -1044,14 +1070,14 @@ def q8_gpu(session, sf, order_start=1, order_count=None):
     b2 = ops.hash_builder(session, br2, [ops.TG_BIGINT, ops.TG_DOUBLE], [0], [1])
     b2.add_input(oam)
     b2.drain()
-    # lineitem -> part filter join -> orders join -> supplier join
-    fl = ops.filter_project(session, None,
-                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
-                             ops.expr(("col", 2)),
-                             ops.expr(("col", 3), ("f64", 1.0), ("col", 4),
-                                      "sub", "mul")],
-                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
-                             ops.TG_DOUBLE])
+    # lineitem scan with FUSED dynamic part filter -> orders join -> supplier
+    fl = ops.filter_project_df(session, None,
+                               [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                                ops.expr(("col", 2)),
+                                ops.expr(("col", 3), ("f64", 1.0), ("col", 4),
+                                         "sub", "mul")],
+                               [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
+                                ops.TG_DOUBLE], brp, 0)
     fl.add_input(ops.page_from_device(session, ([(li.partkey, ops.TG_BIGINT),
                                                  (li.orderkey, ops.TG_BIGINT),
                                                  (li.suppkey, ops.TG_BIGINT),
