@@ -1196,3 +1196,40 @@ class TestProbeOuterJoin:
         # probe keys pass through in order (null rows carry their slot)
         keep = ~p_nulls
         assert np.array_equal(np.asarray(got_k)[keep], pk[keep])
+
+
+class TestDynamicFilter2:
+    def test_fused_bitmap_filter(self, sess, ops):
+        """DynamicPageFilter analog: request_bitmap on the build bridge, scan
+        with filter_project_df — selection equals static predicate AND key
+        membership; join results unchanged vs the unfiltered path."""
+        r = rng(11)
+        n = 200_000
+        build_keys = np.unique(r.integers(0, 50_000, 5_000)).astype(np.int64)
+        probe_keys = r.integers(0, 50_000, n).astype(np.int64)
+        vals = r.standard_normal(n)
+        bridge = ops.JoinBridge(sess)
+        ops.request_bitmap(bridge)
+        b = ops.hash_builder(sess, bridge, [ops.TG_BIGINT], [0], [])
+        b.add_input(ops.page_from_numpy([build_keys]))
+        b.drain()
+        page = ops.page_from_numpy([probe_keys, vals])
+        f = ops.filter_project_df(sess, ops.expr(("col", 1), ("f64", 0.0), "gt"),
+                                  [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                                  [ops.TG_BIGINT, ops.TG_DOUBLE], bridge, 0)
+        f.add_input(page)
+        pages = f.drain()
+        f.close()
+        got_k = pages[0][0]["values"]
+        mask = (vals > 0) & np.isin(probe_keys, build_keys)
+        assert np.array_equal(got_k, probe_keys[mask])
+        # df-only scan (no static predicate)
+        f2 = ops.filter_project_df(sess, None, [ops.expr(("col", 0))],
+                                   [ops.TG_BIGINT], bridge, 0)
+        f2.add_input(page)
+        pages2 = f2.drain()
+        f2.close()
+        assert np.array_equal(pages2[0][0]["values"],
+                              probe_keys[np.isin(probe_keys, build_keys)])
+        bridge.close()
+        b.close()

@@ -1833,6 +1833,7 @@ def q10_gpu(session, sf, order_count=None, limit=20):
     fo.finish()
     owin = _take_device_page(session, fo)
     bridge = ops.JoinBridge(session)
+    ops.request_bitmap(bridge)
     b = ops.hash_builder(session, bridge, [ops.TG_BIGINT, ops.TG_BIGINT],
                          [0], [1])
     b.add_input(owin)
@@ -1842,11 +1843,12 @@ def q10_gpu(session, sf, order_count=None, limit=20):
                                             (li.extendedprice, ops.TG_DOUBLE),
                                             (li.discount, ops.TG_DOUBLE)],
                                            li.row_count))
-    fl = ops.filter_project(session, ops.expr(("col", 1), ("i64", 2), "eq"),
-                            [ops.expr(("col", 0)),
-                             ops.expr(("col", 2), ("f64", 1.0), ("col", 3),
-                                      "sub", "mul")],
-                            [ops.TG_BIGINT, ops.TG_DOUBLE])
+    fl = ops.filter_project_df(session,
+                               ops.expr(("col", 1), ("i64", 2), "eq"),
+                               [ops.expr(("col", 0)),
+                                ops.expr(("col", 2), ("f64", 1.0), ("col", 3),
+                                         "sub", "mul")],
+                               [ops.TG_BIGINT, ops.TG_DOUBLE], bridge, 0)
     fl.add_input(lpage)
     fl.finish()
     lr = _take_device_page(session, fl)
