@@ -1233,3 +1233,30 @@ class TestDynamicFilter2:
                               probe_keys[np.isin(probe_keys, build_keys)])
         bridge.close()
         b.close()
+
+
+class TestMemoryAccounting:
+    def test_oom_clean_and_accounting(self, sess, ops):
+        """lib/trino-memory-context analog: an impossible device allocation
+        fails with a clean error (no crash, session stays usable) and
+        tg_session_memory reports pool growth."""
+        import ctypes
+        from trino_amd import _lib, TrinoGpuError
+        _lib.tg_session_memory.restype = ctypes.c_int
+        _lib.tg_session_memory.argtypes = [ctypes.c_void_p] * 3
+        tot0 = ctypes.c_int64()
+        cach0 = ctypes.c_int64()
+        _lib.tg_session_memory(sess._h, ctypes.byref(tot0), ctypes.byref(cach0))
+        p = ctypes.c_void_p()
+        rc = _lib.tg_device_malloc(sess._h, ctypes.byref(p), 1 << 61)
+        assert rc != 0          # clean failure, not a crash
+        err = _lib.tg_last_error().decode()
+        assert err              # error string populated
+        # session still works
+        q = ctypes.c_void_p()
+        assert _lib.tg_device_malloc(sess._h, ctypes.byref(q), 1 << 20) == 0
+        tot1 = ctypes.c_int64()
+        cach1 = ctypes.c_int64()
+        _lib.tg_session_memory(sess._h, ctypes.byref(tot1), ctypes.byref(cach1))
+        assert tot1.value >= tot0.value
+        _lib.tg_device_free(sess._h, q)
