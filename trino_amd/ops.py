@@ -525,6 +525,12 @@ def varchar_like_flags(session, d_bytes, d_offsets, n, pattern, d_flags):
 _lib.tg_join_bridge_request_bitmap.restype = ctypes.c_int
 _lib.tg_join_bridge_request_bitmap.argtypes = [ctypes.c_void_p]
 _lib.tg_filter_project_create_df.restype = ctypes.c_int
+_lib.tg_filter_project_create_df.argtypes = [ctypes.c_void_p,
+                                             ctypes.POINTER(TgExpr),
+                                             ctypes.POINTER(TgExpr),
+                                             ctypes.c_void_p, ctypes.c_int32,
+                                             ctypes.c_void_p, ctypes.c_int32,
+                                             ctypes.c_void_p]
 
 
 def filter_project_df(session, filter_expr, projections, out_types,
