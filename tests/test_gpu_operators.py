@@ -1244,6 +1244,9 @@ class TestMemoryAccounting:
         from trino_amd import _lib, TrinoGpuError
         _lib.tg_session_memory.restype = ctypes.c_int
         _lib.tg_session_memory.argtypes = [ctypes.c_void_p] * 3
+        _lib.tg_device_malloc.restype = ctypes.c_int
+        _lib.tg_device_malloc.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                          ctypes.c_int64]
         tot0 = ctypes.c_int64()
         cach0 = ctypes.c_int64()
         _lib.tg_session_memory(sess._h, ctypes.byref(tot0), ctypes.byref(cach0))
