@@ -345,6 +345,26 @@ tg_status tg_filter_project_create_df(tg_session*, const tg_expr* filter,
     const tg_expr* projections, const int32_t* proj_out_types, int32_t n_proj,
     tg_join_bridge* df_bridge, int32_t df_key_channel, tg_operator** out);
 
+/* ---- native Parquet reader (csrc/parquet.cpp; mirrors
+ * lib/trino-parquet/reader/ParquetReader.java surface) ---- */
+typedef struct tg_parquet_file tg_parquet_file;
+tg_status tg_parquet_open(tg_session*, const char* path, tg_parquet_file**);
+void tg_parquet_close(tg_parquet_file*);
+int64_t tg_parquet_num_rows(tg_parquet_file*);
+int32_t tg_parquet_num_columns(tg_parquet_file*);
+const char* tg_parquet_column_name(tg_parquet_file*, int32_t);
+int32_t tg_parquet_physical_type(tg_parquet_file*, int32_t);
+tg_status tg_parquet_read_column(tg_session*, tg_parquet_file*, int32_t col,
+    void* out_values, uint64_t* out_valid, int32_t* out_ids,
+    uint8_t* out_dict_bytes, int64_t dict_bytes_cap,
+    int32_t* out_dict_offsets, int32_t* out_dict_count);
+/* multi-column parallel decode (columns x row groups nested) */
+tg_status tg_parquet_read_columns(tg_session*, tg_parquet_file*,
+    const int32_t* cols, int32_t n_cols, void** out_values,
+    uint64_t** out_valid, int32_t** out_ids, uint8_t** out_dict_bytes,
+    const int64_t* dict_caps, int32_t** out_dict_offsets,
+    int32_t** out_dict_counts);
+
 /* stream timer (HIP events on the session stream) for bench rooflines */
 tg_status tg_timer_start(tg_session*);
 tg_status tg_timer_stop(tg_session*, double* elapsed_ms);

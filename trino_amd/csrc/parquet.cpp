@@ -319,16 +319,44 @@ static bool rle_decode(const uint8_t* in, size_t len, int bit_width, int64_t cou
             if (!(b & 0x80)) break;
         }
         if (h & 1) {                       /* bit-packed: (h>>1) groups of 8 */
-            int64_t n = (int64_t)(h >> 1) * 8;
-            uint64_t buf = 0;
-            int bits = 0;
-            for (int64_t k = 0; k < n && o < count + 7; k++) {
-                while (bits < bit_width && ip < len) { buf |= (uint64_t)in[ip++] << bits; bits += 8; }
-                int32_t v = (int32_t)(buf & ((1ull << bit_width) - 1));
-                buf >>= bit_width;
-                bits -= bit_width;
-                if (o < count) out[o++] = v;
-                else { /* padding values beyond count: drop */ }
+            int64_t groups = (int64_t)(h >> 1);
+            if (bit_width <= 16) {
+                /* one group = 8 values = bit_width BYTES: unpack via a
+                 * 128-bit register (scalar byte loop ran ~10x slower) */
+                uint64_t mask = (1ull << bit_width) - 1;
+                for (int64_t g = 0; g < groups; g++) {
+                    unsigned __int128 bits = 0;
+                    size_t take = (size_t)bit_width;
+                    if (ip + take > len) take = len - ip;
+                    memcpy(&bits, in + ip, take);
+                    ip += take;
+                    if (o + 8 <= count) {
+                        #pragma unroll
+                        for (int k = 0; k < 8; k++) {
+                            out[o + k] = (int32_t)((uint64_t)bits & mask);
+                            bits >>= bit_width;
+                        }
+                        o += 8;
+                    }
+                    else {
+                        for (int k = 0; k < 8; k++) {
+                            if (o < count) out[o++] = (int32_t)((uint64_t)bits & mask);
+                            bits >>= bit_width;
+                        }
+                    }
+                }
+            }
+            else {
+                int64_t n = groups * 8;
+                uint64_t buf = 0;
+                int bits = 0;
+                for (int64_t k = 0; k < n && o < count + 7; k++) {
+                    while (bits < bit_width && ip < len) { buf |= (uint64_t)in[ip++] << bits; bits += 8; }
+                    int32_t v = (int32_t)(buf & ((1ull << bit_width) - 1));
+                    buf >>= bit_width;
+                    bits -= bit_width;
+                    if (o < count) out[o++] = v;
+                }
             }
         }
         else {                             /* run: (h>>1) copies */
@@ -339,6 +367,24 @@ static bool rle_decode(const uint8_t* in, size_t len, int bit_width, int64_t cou
         }
     }
     return o == count;
+}
+
+/* def-level stream == one RLE run of 1s covering nv values (the dominant
+ * null-free OPTIONAL-column case): skip materializing levels entirely */
+static bool defs_all_ones(const uint8_t* in, size_t len, int64_t nv)
+{
+    size_t ip = 0;
+    uint64_t h = 0;
+    int sh = 0;
+    while (ip < len) {
+        uint8_t b = in[ip++];
+        h |= (uint64_t)(b & 0x7F) << sh;
+        sh += 7;
+        if (!(b & 0x80)) break;
+    }
+    if (h & 1) return false;               /* bit-packed: inspect normally */
+    if ((int64_t)(h >> 1) < nv) return false;
+    return ip < len && in[ip] == 1;        /* run value (bit width 1 => 1 byte) */
 }
 
 /* ---------- column reading ---------- */
@@ -583,10 +629,16 @@ extern "C" tg_status tg_parquet_read_column(tg_session* s, tg_parquet_file* f, i
                 else { vals = enc_vals; vals_len = un_len; }
             }
             else {
-                tg_status st = decompress(cm.codec, body, h.compressed, h.uncompressed, &page);
-                if (st != TG_OK) return st;
-                vals = page.data();
-                vals_len = page.size();
+                if (cm.codec == 0) {        /* uncompressed: zero-copy */
+                    vals = body;
+                    vals_len = (size_t)h.uncompressed;
+                }
+                else {
+                    tg_status st = decompress(cm.codec, body, h.compressed, h.uncompressed, &page);
+                    if (st != TG_OK) return st;
+                    vals = page.data();
+                    vals_len = page.size();
+                }
                 /* v1: [def level length i32 + RLE] before values (flat schema:
                  * no rep levels; REQUIRED columns have no level section).
                  * Heuristic per spec: optional column => levels present. */
@@ -601,7 +653,11 @@ extern "C" tg_status tg_parquet_read_column(tg_session* s, tg_parquet_file* f, i
             int64_t nv = h.num_values;
             std::vector<int32_t> defs;
             int64_t n_nonnull = nv;
-            if (levels_src && def_len > 0) {
+            if (levels_src && def_len > 0 &&
+                defs_all_ones(levels_src, (size_t)def_len, nv)) {
+                /* null-free page: fall through with empty defs */
+            }
+            else if (levels_src && def_len > 0) {
                 defs.resize(nv);
                 if (!rle_decode(levels_src, (size_t)def_len, 1, nv, defs.data())) {
                     TG_SET_ERR("parquet: def level decode failed");
@@ -630,6 +686,32 @@ extern "C" tg_status tg_parquet_read_column(tg_session* s, tg_parquet_file* f, i
 
             size_t pp = 0;            /* PLAIN cursor */
             int64_t k = 0;            /* non-null cursor */
+            if (defs.empty() && !is_ba) {
+                /* null-free fast paths (TPC-H columns are REQUIRED) */
+                if (!dict_encoded) {
+                    memcpy((uint8_t*)out_values + row * esz, vals, (size_t)nv * esz);
+                }
+                else if (esz == 8) {
+                    const uint64_t* dv = (const uint64_t*)dict_raw.data();
+                    uint64_t* ov = (uint64_t*)out_values + row;
+                    for (int64_t i = 0; i < nv; i++) ov[i] = dv[idx[i]];
+                }
+                else {
+                    const uint32_t* dv = (const uint32_t*)dict_raw.data();
+                    uint32_t* ov = (uint32_t*)out_values + row;
+                    for (int64_t i = 0; i < nv; i++) ov[i] = dv[idx[i]];
+                }
+                row += nv;
+                remaining -= nv;
+                continue;
+            }
+            if (defs.empty() && is_ba && dict_encoded && out_ids) {
+                int32_t* ov = out_ids + row;
+                for (int64_t i = 0; i < nv; i++) ov[i] = ba_dict_remap[idx[i]];
+                row += nv;
+                remaining -= nv;
+                continue;
+            }
             for (int64_t i = 0; i < nv; i++) {
                 int64_t r = row + i;
                 bool isnull = !defs.empty() && defs[i] == 0;
@@ -692,4 +774,37 @@ extern "C" tg_status tg_parquet_read_column(tg_session* s, tg_parquet_file* f, i
         }
     }
     return TG_OK;
+}
+
+#include <omp.h>
+
+/* Decode several columns concurrently: columns in an outer OMP team, row
+ * groups in each column's inner team (nested; ~cores total). The
+ * reference's ParquetReader prefetches/decodes column chunks in parallel
+ * the same way (reader/ParquetReader.java row-group column chunks). */
+extern "C" tg_status tg_parquet_read_columns(tg_session* s, tg_parquet_file* f,
+    const int32_t* cols, int32_t n_cols, void** out_values,
+    uint64_t** out_valid, int32_t** out_ids, uint8_t** out_dict_bytes,
+    const int64_t* dict_caps, int32_t** out_dict_offsets,
+    int32_t** out_dict_counts)
+{
+    tg_status st = TG_OK;
+    int total = omp_get_max_threads();
+    int inner = total / (n_cols > 0 ? n_cols : 1);
+    if (inner < 1) inner = 1;
+    omp_set_max_active_levels(2);
+    #pragma omp parallel for num_threads(n_cols) schedule(dynamic)
+    for (int32_t c = 0; c < n_cols; c++) {
+        omp_set_num_threads(inner);
+        tg_status cst = tg_parquet_read_column(
+            s, f, cols[c], out_values ? out_values[c] : nullptr,
+            out_valid ? out_valid[c] : nullptr,
+            out_ids ? out_ids[c] : nullptr,
+            out_dict_bytes ? out_dict_bytes[c] : nullptr,
+            dict_caps ? dict_caps[c] : 0,
+            out_dict_offsets ? out_dict_offsets[c] : nullptr,
+            out_dict_counts ? out_dict_counts[c] : nullptr);
+        if (cst != TG_OK) st = cst;
+    }
+    return st;
 }

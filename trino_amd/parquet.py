@@ -86,3 +86,56 @@ def _maybe_valid(valid, n):
     else:
         got = valid
     return None if np.array_equal(got, full) else valid
+
+
+_lib.tg_parquet_read_columns.restype = ctypes.c_int
+_lib.tg_parquet_read_columns.argtypes = [ctypes.c_void_p] * 2 + \
+    [ctypes.c_void_p, ctypes.c_int32] + [ctypes.c_void_p] * 7
+
+
+def read_columns(pf, names, session=None):
+    """Parallel multi-column decode (columns x row groups): returns
+    {name: ndarray or (ids, dict_strings)} like ParquetFile.read_column."""
+    import numpy as np
+    sh = session._h if session is not None else None
+    n = pf.num_rows
+    k = len(names)
+    cols = [pf.columns.index(x) for x in names]
+    types = [pf.types[c] for c in cols]
+    vals_ptr = (ctypes.c_void_p * k)()
+    ids_ptr = (ctypes.c_void_p * k)()
+    db_ptr = (ctypes.c_void_p * k)()
+    do_ptr = (ctypes.c_void_p * k)()
+    dc = np.zeros(k, np.int32)
+    dc_ptr = (ctypes.c_void_p * k)()
+    caps = (ctypes.c_int64 * k)()
+    keep = {}
+    for i, (c, t) in enumerate(zip(cols, types)):
+        if t == 6:       # BYTE_ARRAY: dictionary ids + global dict
+            ids = np.empty(n, np.int32)
+            db = np.empty(1 << 24, np.uint8)
+            do = np.empty(1 << 20, np.int32)
+            keep[i] = (ids, db, do)
+            ids_ptr[i] = ids.ctypes.data
+            db_ptr[i] = db.ctypes.data
+            do_ptr[i] = do.ctypes.data
+            caps[i] = db.nbytes
+            dc_ptr[i] = dc.ctypes.data + 4 * i
+        else:
+            dt = {1: np.int32, 2: np.int64, 5: np.float64}[t]
+            a = np.empty(n, dt)
+            keep[i] = a
+            vals_ptr[i] = a.ctypes.data
+    carr = (ctypes.c_int32 * k)(*cols)
+    _check(_lib.tg_parquet_read_columns(sh, pf._h, carr, k, vals_ptr, None,
+                                        ids_ptr, db_ptr, caps, do_ptr, dc_ptr))
+    out = {}
+    for i, (name, t) in enumerate(zip(names, types)):
+        if t == 6:
+            ids, db, do = keep[i]
+            cnt = int(dc[i])
+            strs = [bytes(db[do[j]:do[j + 1]]) for j in range(cnt)]
+            out[name] = (ids, strs)
+        else:
+            out[name] = keep[i]
+    return out
