@@ -1263,3 +1263,53 @@ class TestMemoryAccounting:
         _lib.tg_session_memory(sess._h, ctypes.byref(tot1), ctypes.byref(cach1))
         assert tot1.value >= tot0.value
         _lib.tg_device_free(sess._h, q)
+
+
+class TestDictionaryAwareFilter:
+    def test_dict_filter_verdict_per_entry(self, sess, ops):
+        """DictionaryAwareColumnarFilter analog: a dictionary-encoded channel
+        filters by evaluating the predicate once per dictionary entry and
+        expanding verdict[id] — identical selection to the flat evaluation."""
+        r = rng(13)
+        n = 300_000
+        dvals = np.arange(0, 5000, dtype=np.int64) * 7 + 3
+        ids = r.integers(0, len(dvals), n).astype(np.int32)
+        flat = dvals[ids]
+        e = ops.expr(("col", 0), ("i64", 17_000), "gt")
+        page_d = ops.page_with_dict([(dvals, ids)])
+        got = ops.filter_run(sess, e, page_d)
+        exp = np.nonzero(flat > 17_000)[0].astype(np.int32)
+        assert np.array_equal(got, exp)
+        # arithmetic single-column predicate takes the dict path too
+        e2 = ops.expr(("col", 0), ("i64", 3), "sub", ("i64", 7), "div",
+                      ("i64", 2500), "lt")
+        got2 = ops.filter_run(sess, e2, page_d)
+        exp2 = np.nonzero((flat - 3) // 7 < 2500)[0].astype(np.int32)
+        assert np.array_equal(got2, exp2)
+
+    def test_dict_filter_speed(self, sess, ops):
+        """measured win: dict path reads 4 B/row of ids vs 8 B/row flat (plus
+        skipping the decode pass) — expect the dict-encoded filter to be
+        faster at equal selectivity on a dict-heavy column."""
+        import time
+        r = rng(14)
+        n = 20_000_000
+        dvals = np.arange(0, 64, dtype=np.int64)
+        ids = r.integers(0, 64, n).astype(np.int32)
+        flat = dvals[ids].copy()
+        e = ops.expr(("col", 0), ("i64", 32), "lt")
+        page_d = ops.page_with_dict([(dvals, ids)])
+        page_f = ops.page_from_numpy([flat])
+        # warmup both
+        ops.filter_run(sess, e, page_d)
+        ops.filter_run(sess, e, page_f)
+        t0 = time.time()
+        for _ in range(3):
+            ops.filter_run(sess, e, page_d)
+        t_dict = time.time() - t0
+        t0 = time.time()
+        for _ in range(3):
+            ops.filter_run(sess, e, page_f)
+        t_flat = time.time() - t0
+        # includes HtoD upload either way (ids half the bytes of values)
+        assert t_dict < t_flat, (t_dict, t_flat)

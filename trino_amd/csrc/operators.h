@@ -96,6 +96,8 @@ void tg_free_expr(tg_session* s, ExprProgram* p);
 
 /* kernels (implemented in ops_*.hip) */
 struct DF;   /* fused dynamic filter spec (ops_filter.hip) */
+tg_status upload_flat(tg_session* s, const void* src, int on_device,
+                      int64_t bytes, void** out);
 tg_status run_filter(tg_session* s, const ExprProgram& pred, const DevPage& page,
                      const tg_selected* input_sel,
                      int32_t** d_positions_out, int32_t* count_out,

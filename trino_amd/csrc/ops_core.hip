@@ -22,8 +22,8 @@ __global__ void k_rle_decode(const T* __restrict__ value, int64_t n, T* __restri
     for (; i < n; i += stride) out[i] = v;
 }
 
-static tg_status upload_flat(tg_session* s, const void* src, int on_device,
-                             int64_t bytes, void** out)
+tg_status upload_flat(tg_session* s, const void* src, int on_device,
+                      int64_t bytes, void** out)
 {
     TG_POOL_ALLOC(s, out, bytes);
     TG_HIP_CHECK(hipMemcpyAsync(*out, src, bytes,

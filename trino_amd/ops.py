@@ -554,3 +554,43 @@ def filter_project_df(session, filter_expr, projections, out_types,
 
 def request_bitmap(bridge):
     _check(_lib.tg_join_bridge_request_bitmap(bridge._h))
+
+
+def page_with_dict(columns):
+    """Build a page where entries may be numpy arrays (flat) or
+    (dict_values_array, ids_int32_array) tuples for dictionary-encoded
+    channels (DictionaryBlock analog)."""
+    blocks = (TgBlock * len(columns))()
+    keep = []
+    n = None
+    for i, col in enumerate(columns):
+        if isinstance(col, tuple):
+            dvals, ids = col
+            n = len(ids)
+            d = TgBlock()
+            d.type = _NP2TG[dvals.dtype]
+            d.kind = 0
+            d.position_count = len(dvals)
+            d.on_device = 0
+            d.data = dvals.ctypes.data
+            keep += [dvals, ids, d]
+            blocks[i].type = d.type
+            blocks[i].kind = 1            # TG_BK_DICTIONARY
+            blocks[i].position_count = n
+            blocks[i].on_device = 0
+            blocks[i].ids = ids.ctypes.data
+            blocks[i].dictionary = ctypes.pointer(d)
+        else:
+            n = len(col)
+            blocks[i].type = _NP2TG[col.dtype]
+            blocks[i].kind = 0
+            blocks[i].position_count = n
+            blocks[i].on_device = 0
+            blocks[i].data = col.ctypes.data
+            keep.append(col)
+    page = TgPage()
+    page.channel_count = len(columns)
+    page.position_count = n
+    page.blocks = ctypes.cast(blocks, ctypes.POINTER(TgBlock))
+    page._keepalive = (blocks, keep)
+    return page
