@@ -296,7 +296,23 @@ def run_q3(args, sess, n_gpus, rank, tdist=None):
     rows = inp["li"].row_count
     log(f"q3 inputs ready: {rows:,} lineitem rows/rank, SF{sf * world:g} "
         f"({world} ranks), {time.time()-t0:.1f}s")
-    hook = (lambda keys: tgdist.gather_union(tdist, keys)) if world > 1 else None
+    hook = None
+    if world > 1 and args.backend == "nccl":
+        import ctypes as _ct
+        from trino_amd import _lib as _tglib
+        _tglib.tg_copy_dtod.restype = _ct.c_int
+        _tglib.tg_copy_dtod.argtypes = [_ct.c_void_p] * 3 + [_ct.c_int64]
+
+        def hook(session, dev_page):
+            n = dev_page.position_count
+            t = torch.empty(max(int(n), 1), dtype=torch.int64, device="cuda")
+            if n:
+                _tglib.tg_copy_dtod(session._h, t.data_ptr(),
+                                    dev_page.blocks[0].data, int(n) * 8)
+            return tgdist.gather_union_device(tdist, t[:int(n)])
+        hook.device_resident = True
+    elif world > 1:
+        hook = lambda keys: tgdist.gather_union(tdist, keys)
     for _ in range(args.warmup):
         tpch_queries.q3_execute(sess, inp, download_groups=False,
                                 cust_key_exchange=hook)

@@ -365,6 +365,9 @@ tg_status tg_parquet_read_columns(tg_session*, tg_parquet_file*,
     const int64_t* dict_caps, int32_t** out_dict_offsets,
     int32_t** out_dict_counts);
 
+tg_status tg_copy_dtod(tg_session*, void* dst_dev, const void* src_dev,
+                       int64_t bytes);
+
 /* stream timer (HIP events on the session stream) for bench rooflines */
 tg_status tg_timer_start(tg_session*);
 tg_status tg_timer_stop(tg_session*, double* elapsed_ms);

@@ -103,6 +103,11 @@ def _union_worker(rank, world, port, out):
     from trino_amd import dist as tgdist
     local = np.arange(rank * 3, rank * 3 + 2 + rank, dtype=np.int64)
     u = tgdist.gather_union(tdist, local)
+    # device-resident variant (same logic; cpu tensors under gloo — the
+    # bench's RCCL hook calls this with cuda tensors)
+    import torch
+    ud = tgdist.gather_union_device(tdist, torch.from_numpy(local)).numpy()
+    assert ud.tolist() == u.tolist()
     top = tgdist.merge_topn(tdist, [(rank, float(10 - rank), rank)], 2,
                             key=lambda t: (-t[1], t[2]))
     out.put((rank, u.tolist(), top))

@@ -112,6 +112,14 @@ extern "C" tg_status tg_device_free(tg_session* s, void* p)
     return TG_OK;
 }
 
+extern "C" tg_status tg_copy_dtod(tg_session* s, void* dst_dev, const void* src_dev, int64_t bytes)
+{
+    TG_HIP_CHECK(hipMemcpyAsync(dst_dev, src_dev, (size_t)bytes,
+                                hipMemcpyDeviceToDevice, s->stream));
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    return TG_OK;
+}
+
 extern "C" tg_status tg_copy_htod(tg_session* s, void* dst_dev, const void* src_host, int64_t bytes)
 {
     TG_HIP_CHECK(hipMemcpyAsync(dst_dev, src_host, (size_t)bytes, hipMemcpyHostToDevice, s->stream));

@@ -37,3 +37,23 @@ def merge_topn(tdist, rows, limit, key):
     merged = [t for o in objs for t in o]
     merged.sort(key=key)
     return merged[:limit]
+
+
+def gather_union_device(tdist, t):
+    """Device-resident variable-length all-gather (RCCL over xGMI for cuda
+    tensors; same code path runs on cpu tensors under gloo for the
+    world-size-2 CPU tests): lengths all-gather, pad to max, tensor
+    all-gather, trim + concat. No host round-trip."""
+    import torch
+    world = tdist.get_world_size()
+    n = torch.tensor([t.numel()], dtype=torch.int64, device=t.device)
+    ns = [torch.zeros_like(n) for _ in range(world)]
+    tdist.all_gather(ns, n)
+    counts = [int(x.item()) for x in ns]
+    m = max(counts + [1])
+    pad = torch.zeros(m, dtype=t.dtype, device=t.device)
+    if t.numel():
+        pad[:t.numel()] = t
+    outs = [torch.empty(m, dtype=t.dtype, device=t.device) for _ in range(world)]
+    tdist.all_gather(outs, pad)
+    return torch.cat([o[:c] for o, c in zip(outs, counts)])

@@ -81,9 +81,18 @@ def q3_execute(session, inp, download_groups=True, cust_key_exchange=None):
     f1.finish()
     bridge1 = ops.JoinBridge(session)
     b1 = ops.hash_builder(session, bridge1, [ops.TG_BIGINT], [0], [])
+    keep_exchange = None
     if cust_key_exchange is None:
         cust_sel = _take_device_page(session, f1)
         b1.add_input(cust_sel)
+    elif getattr(cust_key_exchange, "device_resident", False):
+        # BROADCAST build stays in HBM: tg buffer -> torch cuda tensor ->
+        # RCCL all-gather over xGMI -> build input page (no host round-trip)
+        cust_sel = _take_device_page(session, f1)
+        t = cust_key_exchange(session, cust_sel)
+        keep_exchange = t
+        b1.add_input(ops.page_from_device(
+            session, ([(t.data_ptr(), ops.TG_BIGINT)], t.numel())))
     else:
         host_sel, _ = f1.get_output()     # downloads the selected custkeys
         local_keys = host_sel[0]["values"] if host_sel else np.empty(0, np.int64)
