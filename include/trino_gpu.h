@@ -368,6 +368,12 @@ tg_status tg_parquet_read_columns(tg_session*, tg_parquet_file*,
 tg_status tg_copy_dtod(tg_session*, void* dst_dev, const void* src_dev,
                        int64_t bytes);
 
+/* dense-range single-BIGINT-key aggregation (direct array state, one
+ * atomic per row; groups emit in key order). For count-shaped aggregations
+ * over keys with known dense statistics (e.g. generated custkeys). */
+tg_status tg_dense_aggregation_create(tg_session*, int32_t key_channel,
+    int64_t key_min, int64_t key_max, const tg_agg_spec* agg, tg_operator**);
+
 /* stream timer (HIP events on the session stream) for bench rooflines */
 tg_status tg_timer_start(tg_session*);
 tg_status tg_timer_stop(tg_session*, double* elapsed_ms);

@@ -1313,3 +1313,25 @@ class TestDictionaryAwareFilter:
         t_flat = time.time() - t0
         # includes HtoD upload either way (ids half the bytes of values)
         assert t_dict < t_flat, (t_dict, t_flat)
+
+
+class TestDenseAggregation:
+    def test_dense_count_matches_hash(self, sess, ops):
+        r = rng(21)
+        n = 500_000
+        keys = r.integers(1, 40_000, n).astype(np.int64)
+        page = ops.page_from_numpy([keys])
+        d = ops.dense_aggregation(sess, 0, 1, 40_000, (ops.AGG_COUNT_STAR, -1))
+        d.add_input(page)
+        d.finish()
+        pages = d.drain()
+        d.close()
+        gk = pages[0][0]["values"]
+        gc = pages[0][1]["values"]
+        import collections
+        exp = collections.Counter(keys.tolist())
+        assert len(gk) == len(exp)
+        # key-ordered emission
+        assert np.all(np.diff(gk) > 0)
+        for k, c in zip(gk.tolist(), gc.tolist()):
+            assert exp[k] == c

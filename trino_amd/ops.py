@@ -594,3 +594,25 @@ def page_with_dict(columns):
     page.blocks = ctypes.cast(blocks, ctypes.POINTER(TgBlock))
     page._keepalive = (blocks, keep)
     return page
+
+
+_lib.tg_dense_aggregation_create.restype = ctypes.c_int
+_lib.tg_dense_aggregation_create.argtypes = [ctypes.c_void_p, ctypes.c_int32,
+                                             ctypes.c_int64, ctypes.c_int64,
+                                             ctypes.c_void_p, ctypes.c_void_p]
+
+
+def dense_aggregation(session, key_channel, key_min, key_max, agg):
+    """Direct-array aggregation over a dense BIGINT key range; one
+    (fn, input_channel) agg; output (key, value) in key order."""
+    h = ctypes.c_void_p()
+    sp = TgAggSpec()
+    sp.fn = agg[0]
+    sp.input_channel = agg[1]
+    sp.scale_pow = 0
+    _check(_lib.tg_dense_aggregation_create(session._h, key_channel, key_min,
+                                            key_max, ctypes.byref(sp),
+                                            ctypes.byref(h)))
+    op = Operator(session, h)
+    op._keep = (sp,)
+    return op

@@ -1263,8 +1263,10 @@ def q13_gpu(session, sf, order_count=None, cust_count=None):
     f.add_input(opage)
     f.finish()
     kept = _take_device_page(session, f)
-    a1 = ops.hash_aggregation(session, [0], [ops.TG_BIGINT],
-                              [(ops.AGG_COUNT_STAR, -1)])
+    # custkeys are dense 1..150k*SF: direct-array counting (one atomic/row,
+    # no hash probes) instead of a ~10M-group hash table at SF100
+    a1 = ops.dense_aggregation(session, 0, 1, cust_count,
+                               (ops.AGG_COUNT_STAR, -1))
     a1.add_input(kept)
     a1.finish()
     percust = _take_device_page(session, a1)       # (custkey, c_count)
