@@ -443,8 +443,7 @@ def run_sweep(args, sess, n_gpus, rank):
         print(json.dumps(out), flush=True)
 
 
-if __name__ == "__main__":
-    main()
+
 
 
 def run_parquet(args, sess, n_gpus, rank):
@@ -564,3 +563,7 @@ def run_parquet(args, sess, n_gpus, rank):
         tq._device_free(sess, k)
     if rank == 0:
         print(json.dumps(out_rec), flush=True)
+
+
+if __name__ == "__main__":
+    main()
