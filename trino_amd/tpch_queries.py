@@ -2225,46 +2225,38 @@ def q21_gpu(session, sf, order_count=None, limit=100):
                                             (li.commitdate, ops.TG_INTEGER),
                                             (li.receiptdate, ops.TG_INTEGER)],
                                            li.row_count))
-    # single pass: arithmetic masking folds the late-lines aggregates into
-    # the same streaming aggregation as the all-lines min/max (no 150M-row
-    # join): late = (receipt > commit) as exact 0/1 int64;
-    #   skL  = sk*late + (1-late)*2^40  (min identity for non-late rows)
-    #   skL2 = sk*late - (1-late)       (max identity -1)
-    BIG = 1 << 40
-    late_e = [("col", 3), ("col", 2), "gt"]
-    proj = ops.filter_project(
-        session, None,
-        [ops.expr(("col", 0)),
-         ops.expr(("col", 1)),
-         ops.expr(*( [("col", 1)] + late_e + ["mul", ("i64", 1)] + late_e +
-                     ["sub", ("i64", BIG), "mul", "add"])),
-         ops.expr(*( [("col", 1)] + late_e + ["mul", ("i64", 1)] + late_e +
-                     ["sub", "sub"])),
-         ops.expr(*late_e)],
-        [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
-         ops.TG_BIGINT])
-    proj.add_input(lpage)
-    proj.finish()
-    lproj = _take_device_page(session, proj)   # (ok, sk, skL, skL2, late)
-    a_all = ops.streaming_aggregation(session, 0,
-                                      [(ops.AGG_MIN_I64, 1),
-                                       (ops.AGG_MAX_I64, 1),
-                                       (ops.AGG_MIN_I64, 2),
-                                       (ops.AGG_MAX_I64, 3),
-                                       (ops.AGG_SUM_I64, 4)])
-    a_all.add_input(lproj)
+    # all lines: min/max supplier per order (streaming: orderkey-sorted)
+    a_all = ops.streaming_aggregation(session, 0, [(ops.AGG_MIN_I64, 1),
+                                                   (ops.AGG_MAX_I64, 1)])
+    a_all.add_input(lpage)
     a_all.finish()
-    allmm = _take_device_page(session, a_all)  # (ok,minA,maxA,minL,maxL,nlate)
-    # orders with late lines only
-    fl = ops.filter_project(session, ops.expr(("col", 5), ("i64", 0), "gt"),
-                            [ops.expr(("col", 0)), ops.expr(("col", 3)),
-                             ops.expr(("col", 4)), ops.expr(("col", 5)),
-                             ops.expr(("col", 1)), ops.expr(("col", 2))],
-                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
-                             ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT])
-    fl.add_input(allmm)
+    allmm = _take_device_page(session, a_all)       # (ok, minA, maxA)
+    # late lines only (filter preserves order -> still clustered)
+    fl = ops.filter_project(session, ops.expr(("col", 3), ("col", 2), "gt"),
+                            [ops.expr(("col", 0)), ops.expr(("col", 1))],
+                            [ops.TG_BIGINT, ops.TG_BIGINT])
+    fl.add_input(lpage)
     fl.finish()
-    both = _take_device_page(session, fl)  # (ok, minL, maxL, nlate, minA, maxA)
+    late = _take_device_page(session, fl)
+    a_late = ops.streaming_aggregation(session, 0, [(ops.AGG_MIN_I64, 1),
+                                                    (ops.AGG_MAX_I64, 1),
+                                                    (ops.AGG_COUNT_STAR, -1)])
+    a_late.add_input(late)
+    a_late.finish()
+    latemm = _take_device_page(session, a_late)     # (ok, minL, maxL, nlate)
+    # join: late orders with their all-lines min/max
+    bridge_a = ops.JoinBridge(session)
+    ba = ops.hash_builder(session, bridge_a,
+                          [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT],
+                          [0], [1, 2])
+    ba.add_input(allmm)
+    ba.drain()
+    j1 = ops.lookup_join(session, bridge_a,
+                         [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
+                          ops.TG_BIGINT], [0], [0, 1, 2, 3])
+    j1.add_input(latemm)
+    j1.finish()
+    both = _take_device_page(session, j1)  # (ok, minL, maxL, nlate, minA, maxA)
     # F-status orders
     opage = ops.page_from_device(session, ([(obufs["ok"].value, ops.TG_BIGINT),
                                             (obufs["status"].value, ops.TG_TINYINT)],
@@ -2326,9 +2318,10 @@ def q21_gpu(session, sf, order_count=None, limit=100):
     top.add_input(sc)
     pages = top.drain()
     elapsed = time.time() - t0
-    for op in (proj, a_all, fl, fo, bo, sj, fq, agg, fs, bs, sj2, ff, top):
+    for op in (a_all, fl, a_late, ba, j1, fo, bo, sj, fq, agg, fs, bs, sj2,
+               ff, top):
         op.close()
-    for br in (bridge_o, bridge_s):
+    for br in (bridge_a, bridge_o, bridge_s):
         br.close()
     session.tpch_lineitem_free(li)
     _free_bufs(session, obufs)
