@@ -1231,8 +1231,8 @@ class TestDynamicFilter2:
         f2.close()
         assert np.array_equal(pages2[0][0]["values"],
                               probe_keys[np.isin(probe_keys, build_keys)])
-        bridge.close()
         b.close()
+        bridge.close()
 
 
 class TestMemoryAccounting:
