@@ -63,6 +63,11 @@ tg_status tg_pool_alloc(tg_session* s, void** out, size_t bytes)
         e = hipMalloc(out, bytes);
     }
     if (e != hipSuccess) {
+        /* clear the STICKY per-thread last error: a failed hipMalloc would
+         * otherwise surface from the next hipGetLastError() after an
+         * unrelated kernel launch (found via the OOM test poisoning the
+         * dictionary-filter test that ran after it) */
+        (void)hipGetLastError();
         TG_SET_ERR("device OOM allocating %zu bytes", bytes);
         return TG_ERR_OOM;
     }

@@ -2097,26 +2097,6 @@ def q9_gpu(session, sf, order_count=None, part_count=None):
                            [0, 1], [2])
     bps.add_input(ps_green)
     bps.drain()
-    # orders: (ok -> year) build
-    opage = ops.page_from_device(session, ([(obufs["ok"].value, ops.TG_BIGINT),
-                                            (obufs["od"].value, ops.TG_INTEGER)],
-                                           order_count))
-    # year = 1992 + bin(od): bins at Jan 1 epoch days
-    jan = [_D(y, 1, 1) for y in range(1993, 2000)]
-    yexpr = [("col", 1), ("i64", jan[0]), "ge"]
-    for j_ in jan[1:]:
-        yexpr += [("col", 1), ("i64", j_), "ge", "add"]
-    fo = ops.filter_project(session, None, [ops.expr(("col", 0)),
-                                            ops.expr(*yexpr)],
-                            [ops.TG_BIGINT, ops.TG_BIGINT])
-    fo.add_input(opage)
-    fo.finish()
-    oyear = _take_device_page(session, fo)         # (ok, yearbin 0..7)
-    bridge_o = ops.JoinBridge(session)
-    bo = ops.hash_builder(session, bridge_o, [ops.TG_BIGINT, ops.TG_BIGINT],
-                          [0], [1])
-    bo.add_input(oyear)
-    bo.drain()
     # supplier: (sk -> nationkey) build
     sn = ops.page_from_device(session, ([(s_sk.value, ops.TG_BIGINT),
                                          (s_nk.value, ops.TG_TINYINT)],
@@ -2155,27 +2135,49 @@ def q9_gpu(session, sf, order_count=None, part_count=None):
     jps.add_input(lg)
     jps.finish()
     wcost = _take_device_page(session, jps)  # (sk, ok, qty, rev, cost_cents)
-    jo = ops.lookup_join(session, bridge_o,
+    # join orders for the year: BUILD the (smaller) green-lineitem side and
+    # PROBE orders, with a fused dynamic filter pruning non-green orders
+    # before the year projection (was: build 150M orders rows)
+    bridge_w = ops.JoinBridge(session)
+    ops.request_bitmap(bridge_w)
+    bw = ops.hash_builder(session, bridge_w,
+                          [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE,
+                           ops.TG_DOUBLE, ops.TG_BIGINT], [1], [0, 2, 3, 4])
+    bw.add_input(wcost)
+    bw.drain()
+    opage = ops.page_from_device(session, ([(obufs["ok"].value, ops.TG_BIGINT),
+                                            (obufs["od"].value, ops.TG_INTEGER)],
+                                           order_count))
+    jan = [_D(y, 1, 1) for y in range(1993, 2000)]
+    yexpr = [("col", 1), ("i64", jan[0]), "ge"]
+    for j_ in jan[1:]:
+        yexpr += [("col", 1), ("i64", j_), "ge", "add"]
+    fo = ops.filter_project_df(session, None,
+                               [ops.expr(("col", 0)), ops.expr(*yexpr)],
+                               [ops.TG_BIGINT, ops.TG_BIGINT], bridge_w, 0)
+    fo.add_input(opage)
+    fo.finish()
+    oyear = _take_device_page(session, fo)    # (ok, yearbin) green orders only
+    jo = ops.lookup_join(session, bridge_w,
+                         [ops.TG_BIGINT, ops.TG_BIGINT], [0],
+                         [1])                  # emit yearbin + (sk,qty,rev,cost)
+    jo.add_input(oyear)
+    jo.finish()
+    wyear = _take_device_page(session, jo)   # (yearbin, sk, qty, rev, cost)
+    jn = ops.lookup_join(session, bridge_n,
                          [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE,
                           ops.TG_DOUBLE, ops.TG_BIGINT], [1],
                          [0, 2, 3, 4])
-    jo.add_input(wcost)
-    jo.finish()
-    wyear = _take_device_page(session, jo)   # (sk, qty, rev, cost, yearbin)
-    jn = ops.lookup_join(session, bridge_n,
-                         [ops.TG_BIGINT, ops.TG_DOUBLE, ops.TG_DOUBLE,
-                          ops.TG_BIGINT, ops.TG_BIGINT], [0],
-                         [1, 2, 3, 4])
     jn.add_input(wyear)
     jn.finish()
-    wn = _take_device_page(session, jn)      # (qty, rev, cost, yearbin, nk)
+    wn = _take_device_page(session, jn)      # (yearbin, qty, rev, cost, nk)
     # amount = rev - (cost/100.0) * qty  (same IEEE ops as the reference:
     # supplycost DOUBLE times quantity DOUBLE)
     fa = ops.filter_project(session, None,
-                            [ops.expr(("col", 3)), ops.expr(("col", 4)),
-                             ops.expr(("col", 1),
-                                      ("col", 2), ("f64", 100.0), "div",
-                                      ("col", 0), "mul", "sub")],
+                            [ops.expr(("col", 0)), ops.expr(("col", 4)),
+                             ops.expr(("col", 2),
+                                      ("col", 3), ("f64", 100.0), "div",
+                                      ("col", 1), "mul", "sub")],
                             [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_DOUBLE])
     fa.add_input(wn)
     fa.finish()
@@ -2185,9 +2187,9 @@ def q9_gpu(session, sf, order_count=None, part_count=None):
     agg.add_input(amounts)
     pages = agg.drain()
     elapsed = time.time() - t0
-    for op in (fp, bg, sjp, fps, bps, fo, bo, bn, sjl, fl, jps, jo, jn, fa, agg):
+    for op in (fp, bg, sjp, fps, bps, bw, fo, bn, sjl, fl, jps, jo, jn, fa, agg):
         op.close()
-    for br in (bridge_g, bridge_ps, bridge_o, bridge_n):
+    for br in (bridge_g, bridge_ps, bridge_w, bridge_n):
         br.close()
     session.tpch_lineitem_free(li)
     _free_bufs(session, obufs)
