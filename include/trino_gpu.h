@@ -167,6 +167,12 @@ typedef struct tg_agg_spec {
     int32_t fn;
     int32_t input_channel;
     int32_t scale_pow;      /* TG_AGG_SUM_F64_EXACT only; else 0 */
+    /* masked aggregation (AggregationMask analog, predicate form): the row
+     * contributes to THIS aggregate only when col[mask_gt_a] > col[mask_gt_b]
+     * (filter clause lowered into the accumulator; -1 = unmasked). Lets one
+     * pass compute filtered and unfiltered aggregates side by side. */
+    int32_t mask_gt_a;
+    int32_t mask_gt_b;
     int32_t _pad;
 } tg_agg_spec;
 

@@ -1335,3 +1335,74 @@ class TestDenseAggregation:
         assert np.all(np.diff(gk) > 0)
         for k, c in zip(gk.tolist(), gc.tolist()):
             assert exp[k] == c
+
+
+class TestMaskedAggregates:
+    def test_streaming_masked_min_max_count(self, sess, ops):
+        """masked aggregation (AggregationMask predicate form): one pass
+        computes unmasked and col_a>col_b-masked aggregates side by side —
+        the Q21 shape (late-line min/max/count per order)."""
+        r = rng(31)
+        runs = 4000
+        reps = r.integers(1, 8, runs)
+        keys = np.repeat(np.arange(runs, dtype=np.int64) * 2 + 1, reps)
+        n = len(keys)
+        sk = r.integers(1, 1000, n).astype(np.int64)
+        a = r.integers(0, 100, n).astype(np.int32)
+        b = r.integers(0, 100, n).astype(np.int32)
+        page = ops.page_from_numpy([keys, sk, b, a])   # mask: col3 > col2
+        agg = ops.streaming_aggregation(
+            sess, 0,
+            [(ops.AGG_MIN_I64, 1), (ops.AGG_MAX_I64, 1),
+             (ops.AGG_MIN_I64, 1, 0, 3, 2), (ops.AGG_MAX_I64, 1, 0, 3, 2),
+             (ops.AGG_COUNT_STAR, -1, 0, 3, 2)])
+        agg.add_input(page)
+        agg.finish()
+        pages = agg.drain()
+        agg.close()
+        out = pages[0]
+        starts = np.concatenate([[0], np.cumsum(reps)[:-1]]).astype(np.int64)
+        late = a > b
+        skL = np.where(late, sk, np.int64(2**62))
+        skH = np.where(late, sk, np.int64(-2**62))
+        assert np.array_equal(out[1]["values"], np.minimum.reduceat(sk, starts))
+        assert np.array_equal(out[2]["values"], np.maximum.reduceat(sk, starts))
+        # groups with no late rows keep the min/max identities
+        expL = np.minimum.reduceat(skL, starts)
+        expH = np.maximum.reduceat(skH, starts)
+        nlate = np.add.reduceat(late.astype(np.int64), starts)
+        gotL = out[3]["values"]
+        gotH = out[4]["values"]
+        gotN = out[5]["values"]
+        assert np.array_equal(gotN, nlate)
+        m = nlate > 0
+        assert np.array_equal(gotL[m], expL[m])
+        assert np.array_equal(gotH[m], expH[m])
+
+    def test_hash_masked_sum(self, sess, ops):
+        r = rng(32)
+        n = 200_000
+        keys = r.integers(0, 500, n).astype(np.int64)
+        v = r.integers(0, 1000, n).astype(np.int64)
+        a = r.integers(0, 10, n).astype(np.int32)
+        b = r.integers(0, 10, n).astype(np.int32)
+        page = ops.page_from_numpy([keys, v, a, b])
+        agg = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                   [(ops.AGG_SUM_I64, 1),
+                                    (ops.AGG_SUM_I64, 1, 0, 2, 3)])
+        agg.add_input(page)
+        agg.finish()
+        pages = agg.drain()
+        agg.close()
+        out = pages[0]
+        import collections
+        e_all = collections.defaultdict(int)
+        e_m = collections.defaultdict(int)
+        for k, vv, aa, bb in zip(keys.tolist(), v.tolist(), a.tolist(), b.tolist()):
+            e_all[k] += vv
+            if aa > bb:
+                e_m[k] += vv
+        for k, s_all, s_m in zip(out[0]["values"].tolist(),
+                                 out[1]["values"].tolist(),
+                                 out[2]["values"].tolist()):
+            assert e_all[k] == s_all and e_m[k] == s_m

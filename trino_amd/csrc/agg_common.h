@@ -11,6 +11,8 @@ struct KAgg {
     double* sum;          /* f64 state (or null); SUM_F64_EXACT: i128 HI words */
     long long* cnt;       /* i64 state (count / int sum); SUM_F64_EXACT: LO words */
     double scale;         /* SUM_F64_EXACT: 2^scale_pow */
+    int32_t mask_a = -1;  /* row gate col[mask_a] > col[mask_b]; a==b => off */
+    int32_t mask_b = -1;
 };
 #define MAX_AGGS 12
 

@@ -67,6 +67,12 @@ __device__ static inline uint64_t kcol_word(const KColH& c, int64_t i)
     }
 }
 
+/* signed integer view of a fixed-width channel (masked-aggregate gates) */
+__device__ static inline int64_t kcol_sval(const KColH& c, int64_t i)
+{
+    return (int64_t)kcol_word(c, i);
+}
+
 /* per-channel HASH_CODE of the canonical word (types map as the reference's
  * long-based HASH_CODE operators; DOUBLE uses raw (normalized) bits) */
 __device__ static inline uint64_t kcol_hash(const KColH& c, int64_t i)

@@ -216,12 +216,19 @@ __global__ void k_agg_update(const int32_t* __restrict__ gids, int64_t n,
             KAgg ag = aggs[a];
             int mmode = (ag.fn == TG_AGG_MIN_I64) ? 1
                       : (ag.fn == TG_AGG_MAX_I64) ? 2 : 0;
+            bool row_ok = active;
+            if (active && ag.mask_a >= 0 && ag.mask_a != ag.mask_b) {
+                /* masked aggregate: col[mask_a] > col[mask_b] gates the row
+                 * (typed: both int32 or int64 supported via kcol_word) */
+                row_ok = kcol_sval(cols[ag.mask_a], i) >
+                         kcol_sval(cols[ag.mask_b], i);
+            }
             /* per-lane addend for this agg */
             long long ci = mmode == 1 ? INT64_MAX
                          : mmode == 2 ? INT64_MIN : 0;   /* integer addend */
             double cf = 0.0;        /* f64 addend */
             unsigned long long lo = 0, hi = 0;   /* exact i128 addend */
-            if (active) {
+            if (row_ok) {
                 switch (ag.fn) {
                     case TG_AGG_COUNT_STAR:
                         ci = (step == 0) ? 1 : ((const int64_t*)cols[ag.in_ch].data)[i];
@@ -452,11 +459,16 @@ __global__ void k_agg_update_sorted(const int32_t* __restrict__ gids, int64_t n,
             KAgg ag = aggs[a];
             int mmode = (ag.fn == TG_AGG_MIN_I64) ? 1
                       : (ag.fn == TG_AGG_MAX_I64) ? 2 : 0;
+            bool row_ok = active;
+            if (active && ag.mask_a >= 0 && ag.mask_a != ag.mask_b) {
+                row_ok = kcol_sval(cols[ag.mask_a], i) >
+                         kcol_sval(cols[ag.mask_b], i);
+            }
             long long ci = mmode == 1 ? INT64_MAX
                          : mmode == 2 ? INT64_MIN : 0;
             double cf = 0.0;
             unsigned long long lo = 0, hi = 0;
-            if (active) {
+            if (row_ok) {
                 switch (ag.fn) {
                     case TG_AGG_COUNT_STAR:
                         ci = (step == 0) ? 1 : ((const int64_t*)cols[ag.in_ch].data)[i];
@@ -1091,6 +1103,8 @@ extern "C" tg_status tg_hash_aggregation_create(tg_session* s,
         KAgg k{};
         k.fn = aggs[a].fn;
         k.in_ch = aggs[a].input_channel;
+        k.mask_a = aggs[a].mask_gt_a;
+        k.mask_b = aggs[a].mask_gt_b;
         bool needs_sum = (k.fn == TG_AGG_SUM_F64 || k.fn == TG_AGG_AVG_F64 ||
                           k.fn == TG_AGG_SUM_F64_EXACT);
         bool needs_cnt = (k.fn != TG_AGG_SUM_F64);
@@ -1159,6 +1173,36 @@ __global__ void k_dense_update(const KColH key, int64_t n, int64_t key_min,
     }
 }
 
+/* LDS-privatized variant for small ranges (<= 8192 entries): per-block
+ * shared histogram, one global atomic per present entry per block — fixes
+ * hot-entry contention (Q13's ~64-value count histogram over 13.5M rows
+ * measured 25.3 ms with direct global atomics). */
+__global__ void k_dense_update_lds(const KColH key, int64_t n, int64_t key_min,
+                                   int64_t range, long long* __restrict__ state,
+                                   const KColH val, int fn)
+{
+    extern __shared__ long long h[];
+    for (int64_t i = threadIdx.x; i < range; i += blockDim.x) h[i] = 0;
+    __syncthreads();
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        if (kcol_is_null(key, i)) continue;
+        int64_t k = (int64_t)kcol_word(key, i) - key_min;
+        if (k < 0 || k >= range) continue;
+        long long add = 1;
+        if (fn == TG_AGG_SUM_I64 || fn == TG_AGG_COUNT_COL)
+            add = (fn == TG_AGG_COUNT_COL)
+                ? (kcol_is_null(val, i) ? 0 : 1)
+                : (kcol_is_null(val, i) ? 0 : ((const long long*)val.data)[i]);
+        if (add) atomicAdd((unsigned long long*)&h[k], (unsigned long long)add);
+    }
+    __syncthreads();
+    for (int64_t k = threadIdx.x; k < range; k += blockDim.x)
+        if (h[k]) atomicAdd((unsigned long long*)&state[k],
+                            (unsigned long long)h[k]);
+}
+
 __global__ void k_dense_present(const long long* __restrict__ state, int64_t range,
                                 int32_t* __restrict__ chunk_counts, int64_t nchunks,
                                 int64_t chunk)
@@ -1219,9 +1263,16 @@ struct DenseAggOp : tg_operator {
                                hipMemcpyDeviceToHost));
         KColH kc = h[key_channel];
         KColH vc = (agg.input_channel >= 0) ? h[agg.input_channel] : kc;
-        hipLaunchKernelGGL(k_dense_update, dim3(tg_grid_for(in.n)), dim3(TG_BLOCK),
-                           0, s->stream, kc, in.n, key_min, range, state, vc,
-                           agg.fn);
+        if (range <= 8192) {
+            hipLaunchKernelGGL(k_dense_update_lds, dim3(tg_grid_for(in.n)),
+                               dim3(TG_BLOCK), range * 8, s->stream, kc, in.n,
+                               key_min, range, state, vc, agg.fn);
+        }
+        else {
+            hipLaunchKernelGGL(k_dense_update, dim3(tg_grid_for(in.n)),
+                               dim3(TG_BLOCK), 0, s->stream, kc, in.n,
+                               key_min, range, state, vc, agg.fn);
+        }
         TG_HIP_CHECK(hipGetLastError());
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
         tg_pool_free(s, d_all);

@@ -339,6 +339,8 @@ extern "C" tg_status tg_streaming_aggregation_create(tg_session* s,
         KAgg k{};
         k.fn = aggs[a].fn;
         k.in_ch = aggs[a].input_channel;
+        k.mask_a = aggs[a].mask_gt_a;
+        k.mask_b = aggs[a].mask_gt_b;
         k.scale = 1.0;
         for (int32_t sp_ = 0; sp_ < aggs[a].scale_pow; sp_++) k.scale *= 2.0;
         k.sum = nullptr;

@@ -580,6 +580,25 @@ struct LikePat {
     int n_segs, anchor_start, anchor_end;
 };
 
+/* SWAR first-byte scan: next index >= from with s[i] == c, else -1
+ * (byte-loop scanning measured 38.8 ms for Q13's 150M x ~49 B comments) */
+__device__ static inline int like_find_byte(const uint8_t* s, int from,
+                                            int len, uint8_t c)
+{
+    int i = from;
+    for (; i < len && ((uintptr_t)(s + i) & 7); i++)
+        if (s[i] == c) return i;
+    const uint64_t pat = 0x0101010101010101ull * c;
+    for (; i + 8 <= len; i += 8) {
+        uint64_t w = *(const uint64_t*)(s + i) ^ pat;
+        uint64_t m = (w - 0x0101010101010101ull) & ~w & 0x8080808080808080ull;
+        if (m) return i + (__ffsll((unsigned long long)m) - 1) / 8;
+    }
+    for (; i < len; i++)
+        if (s[i] == c) return i;
+    return -1;
+}
+
 __device__ static bool like_match(const uint8_t* txt, int len, const LikePat& p)
 {
     int pos = 0;
@@ -600,10 +619,14 @@ __device__ static bool like_match(const uint8_t* txt, int len, const LikePat& p)
             continue;
         }
         bool found = false;
-        for (int i = pos; i + sl <= len && !found; i++) {
+        int i = pos;
+        while (i + sl <= len) {
+            i = like_find_byte(txt, i, len - sl + 1, (uint8_t)seg[0]);
+            if (i < 0) break;
             bool eq = true;
-            for (int j = 0; j < sl && eq; j++) eq = txt[i + j] == (uint8_t)seg[j];
-            if (eq) { pos = i + sl; found = true; }
+            for (int j = 1; j < sl && eq; j++) eq = txt[i + j] == (uint8_t)seg[j];
+            if (eq) { pos = i + sl; found = true; break; }
+            i++;
         }
         if (!found) return false;
     }

@@ -59,7 +59,9 @@ class TgExpr(ctypes.Structure):
 
 class TgAggSpec(ctypes.Structure):
     _fields_ = [("fn", ctypes.c_int32), ("input_channel", ctypes.c_int32),
-                ("scale_pow", ctypes.c_int32), ("_pad", ctypes.c_int32)]
+                ("scale_pow", ctypes.c_int32),
+                ("mask_gt_a", ctypes.c_int32), ("mask_gt_b", ctypes.c_int32),
+                ("_pad", ctypes.c_int32)]
 
 
 (OP_COL, OP_CONST_F64, OP_CONST_I64, OP_ADD, OP_SUB, OP_MUL, OP_DIV,
@@ -280,6 +282,8 @@ def streaming_aggregation(session, key_channel, aggs, step=STEP_SINGLE):
         sp[i].fn = a[0]
         sp[i].input_channel = a[1]
         sp[i].scale_pow = a[2] if len(a) > 2 else 0
+        sp[i].mask_gt_a = a[3] if len(a) > 4 else -1
+        sp[i].mask_gt_b = a[4] if len(a) > 4 else -1
     _check(_lib.tg_streaming_aggregation_create(session._h, key_channel, sp,
                                                 len(aggs), step, ctypes.byref(h)))
     op = Operator(session, h)
@@ -297,6 +301,8 @@ def hash_aggregation(session, group_channels, group_types, aggs, step=STEP_SINGL
         sp[i].fn = a[0]
         sp[i].input_channel = a[1]
         sp[i].scale_pow = a[2] if len(a) > 2 else 0
+        sp[i].mask_gt_a = a[3] if len(a) > 4 else -1
+        sp[i].mask_gt_b = a[4] if len(a) > 4 else -1
     _check(_lib.tg_hash_aggregation_create(session._h, gc.ctypes.data, len(gc),
                                            gt.ctypes.data, sp, len(aggs), step,
                                            ctypes.byref(h)))
@@ -610,6 +616,8 @@ def dense_aggregation(session, key_channel, key_min, key_max, agg):
     sp.fn = agg[0]
     sp.input_channel = agg[1]
     sp.scale_pow = 0
+    sp.mask_gt_a = -1
+    sp.mask_gt_b = -1
     _check(_lib.tg_dense_aggregation_create(session._h, key_channel, key_min,
                                             key_max, ctypes.byref(sp),
                                             ctypes.byref(h)))
