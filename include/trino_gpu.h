@@ -374,6 +374,11 @@ tg_status tg_parquet_read_columns(tg_session*, tg_parquet_file*,
 tg_status tg_copy_dtod(tg_session*, void* dst_dev, const void* src_dev,
                        int64_t bytes);
 
+/* MarkDistinctOperator analog: appends a BOOLEAN channel marking each
+ * row's first occurrence over the key channels (streaming pass-through) */
+tg_status tg_mark_distinct_create(tg_session*, const int32_t* key_channels,
+    int32_t n_key_channels, const int32_t* key_types, tg_operator**);
+
 /* dense-range single-BIGINT-key aggregation (direct array state, one
  * atomic per row; groups emit in key order). For count-shaped aggregations
  * over keys with known dense statistics (e.g. generated custkeys). */

@@ -1406,3 +1406,34 @@ class TestMaskedAggregates:
                                  out[1]["values"].tolist(),
                                  out[2]["values"].tolist()):
             assert e_all[k] == s_all and e_m[k] == s_m
+
+
+class TestMarkDistinct:
+    def test_first_occurrence_flags(self, sess, ops):
+        """MarkDistinctOperator analog: BOOLEAN channel true exactly on each
+        key's first occurrence, across pages (global row order)."""
+        r = rng(41)
+        n = 50_000
+        k1 = r.integers(0, 3000, n).astype(np.int64)
+        k2 = r.integers(0, 4, n).astype(np.int32)
+        v = r.standard_normal(n)
+        op = ops.mark_distinct(sess, [0, 1], [ops.TG_BIGINT, ops.TG_INTEGER])
+        half = n // 2
+        p1 = ops.page_from_numpy([k1[:half], k2[:half], v[:half]])
+        p2 = ops.page_from_numpy([k1[half:], k2[half:], v[half:]])
+        op.add_input(p1)
+        op.add_input(p2)
+        op.finish()
+        pages = op.drain()
+        op.close()
+        flags = np.concatenate([np.asarray(p[3]["values"]) for p in pages])
+        seen = set()
+        exp = np.zeros(n, np.int8)
+        for i, key in enumerate(zip(k1.tolist(), k2.tolist())):
+            if key not in seen:
+                seen.add(key)
+                exp[i] = 1
+        assert np.array_equal(flags, exp)
+        # pass-through channels intact
+        vals = np.concatenate([np.asarray(p[2]["values"]) for p in pages])
+        assert np.array_equal(vals, v)

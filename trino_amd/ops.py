@@ -624,3 +624,22 @@ def dense_aggregation(session, key_channel, key_min, key_max, agg):
     op = Operator(session, h)
     op._keep = (sp,)
     return op
+
+
+_lib.tg_mark_distinct_create.restype = ctypes.c_int
+_lib.tg_mark_distinct_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                         ctypes.c_int32, ctypes.c_void_p,
+                                         ctypes.c_void_p]
+
+
+def mark_distinct(session, key_channels, key_types):
+    """MarkDistinctOperator analog: pass-through + BOOLEAN first-occurrence
+    channel appended after the input channels."""
+    h = ctypes.c_void_p()
+    kc = _i32arr(key_channels)
+    kt = _i32arr(key_types)
+    _check(_lib.tg_mark_distinct_create(session._h, kc.ctypes.data, len(kc),
+                                        kt.ctypes.data, ctypes.byref(h)))
+    op = Operator(session, h)
+    op._keep = (kc, kt)
+    return op
