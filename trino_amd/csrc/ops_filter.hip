@@ -898,6 +898,40 @@ __global__ void k_proj_mulsub(const double* __restrict__ a, const double* __rest
     }
 }
 
+__global__ void k_proj_cmp_flag(KCol col, int op, double cval, int64_t icval,
+                                int int_mode, const int32_t* __restrict__ pos,
+                                int32_t n, double* __restrict__ out)
+{
+    int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; k < n; k += stride) {
+        int64_t i = pos ? (int64_t)pos[k] : k;
+        TVal v = load_tcol(col, i);
+        bool r;
+        if (int_mode) {
+            switch (op) {
+                case TG_EXPR_LE: r = v.i <= icval; break;
+                case TG_EXPR_LT: r = v.i < icval; break;
+                case TG_EXPR_GE: r = v.i >= icval; break;
+                case TG_EXPR_GT: r = v.i > icval; break;
+                case TG_EXPR_EQ: r = v.i == icval; break;
+                default: r = v.i != icval; break;
+            }
+        }
+        else {
+            switch (op) {
+                case TG_EXPR_LE: r = v.f <= cval; break;
+                case TG_EXPR_LT: r = v.f < cval; break;
+                case TG_EXPR_GE: r = v.f >= cval; break;
+                case TG_EXPR_GT: r = v.f > cval; break;
+                case TG_EXPR_EQ: r = v.f == cval; break;
+                default: r = v.f != cval; break;
+            }
+        }
+        out[k] = r ? 1.0 : 0.0;
+    }
+}
+
 tg_status run_project(tg_session* s, const ExprProgram& proj, tg_type out_type,
                       const DevPage& page, const int32_t* d_positions, int32_t count,
                       DevBlock* out)
@@ -905,6 +939,33 @@ tg_status run_project(tg_session* s, const ExprProgram& proj, tg_type out_type,
     /* identity projection = typed gather */
     if (proj.count == 1 && proj.insts[0].op == TG_EXPR_COL)
         return run_gather(s, page.blocks[proj.insts[0].arg0], d_positions, count, out);
+
+    /* COL cmp CONST boolean projection (Q7/Q8's year flag): specialized
+     * write of 1.0/0.0 — the interpreter cost ~10.7 ms over 180M rows */
+    if (proj.count == 3 && proj.insts[0].op == TG_EXPR_COL &&
+        (proj.insts[1].op == TG_EXPR_CONST_F64 || proj.insts[1].op == TG_EXPR_CONST_I64) &&
+        proj.insts[2].op >= TG_EXPR_LE && proj.insts[2].op <= TG_EXPR_NE) {
+        const DevBlock& a = page.blocks[proj.insts[0].arg0];
+        if (!a.valid && a.type != TG_VARCHAR && out_type != TG_BIGINT) {
+            bool is_i64c = proj.insts[1].op == TG_EXPR_CONST_I64;
+            double cv = is_i64c ? (double)proj.insts[1].imm.i64
+                                : proj.insts[1].imm.f64;
+            int64_t icv = is_i64c ? proj.insts[1].imm.i64
+                                  : (int64_t)proj.insts[1].imm.f64;
+            KCol kc{a.data, nullptr, (int32_t)a.type, 0};
+            out->type = TG_DOUBLE;
+            out->n = count;
+            TG_POOL_ALLOC(s, &out->data, (int64_t)(count ? count : 1) * 8);
+            hipLaunchKernelGGL(k_proj_cmp_flag, dim3(tg_grid_for(count)),
+                               dim3(TG_BLOCK), 0, s->stream, kc,
+                               proj.insts[2].op, cv, icv,
+                               is_i64c && type_is_int(a.type) ? 1 : 0,
+                               d_positions, count, (double*)out->data);
+            TG_HIP_CHECK(hipGetLastError());
+            TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+            return TG_OK;
+        }
+    }
 
     /* a*(const-b) fast path (both DOUBLE, no nulls) */
     if (proj.count == 5 &&
