@@ -6,14 +6,17 @@
  * factories would make for the covered TPC-H plans.
  *
  *   trino_gpu_cli q1 [sf]      fused Q1 (scan/filter/agg), prints the 4 rows
- *   trino_gpu_cli q3 [sf]      3-way join pipeline, prints the top 10
+ *   trino_gpu_cli q3 [sf]      3-way join pipeline + fused dynamic filter
  *   trino_gpu_cli q6 [sf]      scan/filter/scalar agg
- * Requires a HIP device (no CPU fallback by design).
+ *   trino_gpu_cli q13 [sf]     pool LIKE + dense-range aggregation (round 2)
+ * (q4/q12/q14/q18 as well — see main.) Requires a HIP device (no CPU
+ * fallback by design).
  */
 #include "../include/trino_gpu.h"
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <algorithm>
 #include <string>
 #include <vector>
 
@@ -191,6 +194,10 @@ static int run_q3(tg_session* s, double sf)
 
     tg_join_bridge* br2 = nullptr;
     die(tg_join_bridge_create(s, &br2), "br2");
+    /* dynamic filter: ask the build for a key-membership bitmap so the
+     * lineitem scan below can drop non-matching orderkeys in-kernel
+     * (DynamicPageFilter analog; round-2 ABI) */
+    die(tg_join_bridge_request_bitmap(br2), "br2 df");
     int32_t bt2[2] = {TG_BIGINT, TG_INTEGER}, b2out = 1;
     tg_operator* b2 = nullptr;
     die(tg_hash_builder_create(s, br2, bt2, 2, &kc0, 1, &b2out, 1, &b2), "b2");
@@ -211,7 +218,7 @@ static int run_q3(tg_session* s, double sf)
     tg_expr lpr[2] = {{lp0.data(), 1}, {lp1.data(), (int32_t)lp1.size()}};
     int32_t lot[2] = {TG_BIGINT, TG_DOUBLE};
     tg_operator* f3 = nullptr;
-    die(tg_filter_project_create(s, &lfe, lpr, lot, 2, &f3), "f3");
+    die(tg_filter_project_create_df(s, &lfe, lpr, lot, 2, br2, 0, &f3), "f3");
     die(tg_operator_add_input(f3, &lpage), "f3 add");
     die(tg_operator_finish(f3), "f3 fin");
     tg_page li_sel{};
@@ -652,10 +659,85 @@ static int run_q18(tg_session* s, double sf)
     return 0;
 }
 
+static int run_q13(tg_session* s, double sf)
+{
+    /* Q13 (python mirror: q13_gpu, ref sql/planner tests TestTpchLocalQueries
+     * shape): o_comment NOT LIKE '%special%requests%' evaluated directly over
+     * the device text-pool slices, per-customer order counts and the final
+     * histogram both via dense-range aggregation — the round-2 ABI
+     * (tg_tpch_gen_orders3 / tg_pool_like_flags / tg_dense_aggregation_create)
+     * driven from pure C++. */
+    int64_t n_orders = (int64_t)(1500000 * sf), n_cust = (int64_t)(150000 * sf);
+    void *o_ck, *o_coff, *o_clen, *o_flag;
+    die(tg_device_malloc(s, &o_ck, n_orders * 8), "malloc");
+    die(tg_device_malloc(s, &o_coff, n_orders * 8), "malloc");
+    die(tg_device_malloc(s, &o_clen, n_orders * 4), "malloc");
+    die(tg_device_malloc(s, &o_flag, n_orders), "malloc");
+    die(tg_tpch_gen_orders3(s, sf, 1, n_orders, nullptr, (int64_t*)o_ck,
+                            nullptr, nullptr, nullptr, nullptr,
+                            (int64_t*)o_coff, (int32_t*)o_clen), "gen ord3");
+    die(tg_pool_like_flags(s, (const int64_t*)o_coff, (const int32_t*)o_clen,
+                           n_orders, "%special%requests%", (uint8_t*)o_flag),
+        "like");
+
+    std::vector<tg_block> ob = {dev_block(TG_BIGINT, o_ck, n_orders),
+                                dev_block(TG_TINYINT, o_flag, n_orders)};
+    tg_page opage = dev_page(ob, n_orders);
+    std::vector<tg_expr_inst> fe = {I_col(1), I_i64(0), I_op(TG_EXPR_EQ)};
+    std::vector<tg_expr_inst> p0 = {I_col(0)};
+    tg_expr fex{fe.data(), (int32_t)fe.size()};
+    tg_expr prj{p0.data(), (int32_t)p0.size()};
+    int32_t oty = TG_BIGINT;
+    tg_operator* f = nullptr;
+    die(tg_filter_project_create(s, &fex, &prj, &oty, 1, &f), "filter");
+    die(tg_operator_add_input(f, &opage), "f add");
+    die(tg_operator_finish(f), "f fin");
+    int fin = 0;
+    tg_page kept{};
+    die(tg_operator_get_output(f, &kept, &fin), "f out");
+
+    tg_agg_spec cnt{TG_AGG_COUNT_STAR, -1, 0, 0, 0, 0};
+    tg_operator* a1 = nullptr;
+    die(tg_dense_aggregation_create(s, 0, 1, n_cust, &cnt, &a1), "dense1");
+    die(tg_operator_add_input(a1, &kept), "a1 add");
+    die(tg_operator_finish(a1), "a1 fin");
+    tg_page percust{};
+    die(tg_operator_get_output(a1, &percust, &fin), "a1 out");
+    int64_t n_with = percust.position_count;
+
+    tg_operator* a2 = nullptr;
+    die(tg_dense_aggregation_create(s, 1, 1, 4096, &cnt, &a2), "dense2");
+    die(tg_operator_add_input(a2, &percust), "a2 add");
+    die(tg_operator_finish(a2), "a2 fin");
+    tg_page hist{};
+    die(tg_operator_get_output(a2, &hist, &fin), "a2 out");
+
+    int64_t m = hist.position_count;
+    std::vector<int64_t> ccnt(m + 1), cdist(m + 1);
+    if (m) {
+        die(tg_copy_dtoh(s, ccnt.data(), hist.blocks[0].data, m * 8), "dtoh");
+        die(tg_copy_dtoh(s, cdist.data(), hist.blocks[1].data, m * 8), "dtoh");
+    }
+    ccnt[m] = 0;
+    cdist[m] = n_cust - n_with;  /* customers with no qualifying orders */
+    std::vector<int64_t> idx(m + 1);
+    for (int64_t i = 0; i <= m; i++) idx[i] = i;
+    std::sort(idx.begin(), idx.end(), [&](int64_t a, int64_t b) {
+        if (cdist[a] != cdist[b]) return cdist[a] > cdist[b];
+        return ccnt[a] > ccnt[b];
+    });
+    printf("c_count|custdist\n");
+    for (int64_t i : idx)
+        printf("%lld|%lld\n", (long long)ccnt[i], (long long)cdist[i]);
+    for (tg_operator* op : {f, a1, a2}) tg_operator_close(op);
+    for (void* p : {o_ck, o_coff, o_clen, o_flag}) die(tg_device_free(s, p), "free");
+    return 0;
+}
+
 int main(int argc, char** argv)
 {
     if (argc < 2 || !strcmp(argv[1], "--help")) {
-        printf("usage: %s q1|q3|q4|q6|q12|q14|q18 [scale_factor]  (version: %s)\n",
+        printf("usage: %s q1|q3|q4|q6|q12|q13|q14|q18 [scale_factor]  (version: %s)\n",
                argv[0], tg_version());
         return argc < 2 ? 1 : 0;
     }
@@ -674,6 +756,7 @@ int main(int argc, char** argv)
     else if (!strcmp(argv[1], "q14")) rc = run_q14(s, sf);
     else if (!strcmp(argv[1], "q12")) rc = run_q12(s, sf);
     else if (!strcmp(argv[1], "q18")) rc = run_q18(s, sf);
+    else if (!strcmp(argv[1], "q13")) rc = run_q13(s, sf);
     else fprintf(stderr, "unknown query %s\n", argv[1]);
     tg_session_close(s);
     return rc;
