@@ -374,6 +374,27 @@ tg_status tg_parquet_read_columns(tg_session*, tg_parquet_file*,
 tg_status tg_copy_dtod(tg_session*, void* dst_dev, const void* src_dev,
                        int64_t bytes);
 
+/* adaptive partial aggregation (operator/aggregation/partial/
+ * PartialAggregationController.java:34-100 analog): one controller is
+ * shared across the PARTIAL-step hash aggregations of a plan node. After
+ * >= 1.5x max_partial_bytes of input has been sampled, if the ratio of
+ * unique output rows to input rows exceeds the threshold (reference
+ * session default 0.8 — adaptive_partial_aggregation_unique_rows_ratio_
+ * threshold), partial aggregation flips to pass-through: pages are
+ * re-shaped into the partial-state channel layout with no hash-table work
+ * and the FINAL stage does the grouping. Re-enabled after 200x more bytes
+ * (same constants as the reference). on_flush is exposed so host drivers
+ * (and CPU tests) can feed flush statistics directly. Restrictions:
+ * fixed-width group keys, unmasked aggregates. */
+typedef struct tg_pa_controller tg_pa_controller;
+tg_status tg_pa_controller_create(int64_t max_partial_bytes,
+    double unique_rows_ratio_threshold, tg_pa_controller** out);
+void tg_pa_controller_close(tg_pa_controller*);
+int32_t tg_pa_controller_disabled(tg_pa_controller*);
+tg_status tg_pa_controller_on_flush(tg_pa_controller*, int64_t bytes,
+    int64_t rows, int64_t unique_rows, int32_t have_unique);
+tg_status tg_hash_aggregation_set_controller(tg_operator*, tg_pa_controller*);
+
 /* MarkDistinctOperator analog: appends a BOOLEAN channel marking each
  * row's first occurrence over the key channels (streaming pass-through) */
 tg_status tg_mark_distinct_create(tg_session*, const int32_t* key_channels,

@@ -48,3 +48,27 @@ def test_no_gpu_fails_loudly():
     with pytest.raises(trino_amd.TrinoGpuError) as ei:
         trino_amd.Session(0)
     assert "no HIP device" in str(ei.value) or "status" in str(ei.value)
+
+
+def test_partial_agg_controller_state_machine():
+    """PartialAggregationController.java:66-96 semantics via the exposed
+    on_flush: disable after >=1.5x max_partial_bytes sampled with
+    unique/input ratio > threshold; re-enable after 200x more bytes;
+    disabled-mode stats are ignored while enabled (java:69-72)."""
+    _built()
+    from trino_amd.ops import PartialAggController
+    c = PartialAggController(max_partial_bytes=1000, threshold=0.8)
+    assert not c.disabled
+    c.on_flush(1000, 100, 95, True)          # 1000 < 1500: keep sampling
+    assert not c.disabled
+    c.on_flush(600, 60, 58, True)            # 1600 >= 1500, 153/160 > 0.8
+    assert c.disabled
+    c.on_flush(1500 * 200, 10, 0, False)     # 200x bytes -> re-enable, reset
+    assert not c.disabled
+    c.on_flush(5000, 1000, 5, True)          # low unique ratio: stays on
+    assert not c.disabled
+    c2 = PartialAggController(max_partial_bytes=1000, threshold=0.8)
+    c2.on_flush(10**6, 10**6, 0, False)      # no unique stats while enabled
+    assert not c2.disabled
+    c.close()
+    c2.close()

@@ -1437,3 +1437,142 @@ class TestMarkDistinct:
         # pass-through channels intact
         vals = np.concatenate([np.asarray(p[2]["values"]) for p in pages])
         assert np.array_equal(vals, v)
+
+
+class TestAdaptivePartialAgg:
+    """PartialAggregationController analog (adaptive partial aggregation):
+    a high-unique-ratio partial flush flips the shared controller; later
+    partial operators re-shape pages into partial-state layout with no
+    hash-table work; FINAL over the mixed partial outputs is exact."""
+
+    def test_disable_pass_through_and_final_parity(self, sess, ops):
+        r = rng(90)
+        n = 60_000
+        half = n // 2
+        # ~88% unique in each half: above the 0.8 reference threshold
+        keys = r.integers(1000, 1000 + 4 * half, n).astype(np.int64)
+        vals_f = r.uniform(0, 100, n)
+        vals_i = r.integers(-1000, 1000, n).astype(np.int64)
+        aggs_p = [(ops.AGG_COUNT_STAR, -1), (ops.AGG_SUM_F64, 1),
+                  (ops.AGG_SUM_I64, 2), (ops.AGG_MIN_I64, 2),
+                  (ops.AGG_MAX_I64, 2)]
+        ctrl = ops.PartialAggController(max_partial_bytes=1000, threshold=0.8)
+        partial_pages = []
+        op1 = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT], aggs_p,
+                                   step=ops.STEP_PARTIAL)
+        ctrl.attach(op1)
+        assert not ctrl.disabled
+        op1.add_input(ops.page_from_numpy([keys[:half], vals_f[:half],
+                                           vals_i[:half]]))
+        partial_pages += op1.drain()
+        op1.close()
+        assert ctrl.disabled          # flush saw ratio ~0.88 > 0.8
+
+        op2 = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT], aggs_p,
+                                   step=ops.STEP_PARTIAL)
+        ctrl.attach(op2)
+        op2.add_input(ops.page_from_numpy([keys[half:], vals_f[half:],
+                                           vals_i[half:]]))
+        outs2 = op2.drain()
+        op2.close()
+        # pass-through: one output row per input row, states are the
+        # per-row partial forms
+        assert len(outs2) == 1
+        p2 = outs2[0]
+        assert np.array_equal(p2[0]["values"], keys[half:])
+        assert np.all(np.asarray(p2[1]["values"]) == 1)          # COUNT
+        assert np.array_equal(p2[2]["values"], vals_f[half:])    # SUM_F64
+        assert np.array_equal(p2[3]["values"], vals_i[half:])    # SUM_I64
+        assert np.array_equal(p2[4]["values"], vals_i[half:])    # MIN
+        assert np.array_equal(p2[5]["values"], vals_i[half:])    # MAX
+        partial_pages += outs2
+
+        aggs_f = [(ops.AGG_COUNT_STAR, 1), (ops.AGG_SUM_F64, 2),
+                  (ops.AGG_SUM_I64, 3), (ops.AGG_MIN_I64, 4),
+                  (ops.AGG_MAX_I64, 5)]
+        fin = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT], aggs_f,
+                                   step=ops.STEP_FINAL)
+        for p in partial_pages:
+            fin.add_input(ops.page_from_numpy(
+                [np.asarray(b["values"]) for b in p]))
+        out = fin.drain()[0]
+        fin.close()
+        ctrl.close()
+
+        gids, ng, by_gid, _ = oracle.bigint_groupby(keys)
+        exp_cnt = oracle.grouped_count(gids, ng)
+        got_key = np.asarray(out[0]["values"])
+        order = {k: i for i, k in enumerate(got_key.tolist())}
+        perm = np.array([order[k] for k in by_gid.tolist()])
+        assert np.array_equal(got_key[perm], by_gid)
+        assert np.array_equal(np.asarray(out[1]["values"])[perm], exp_cnt)
+        exp_si = np.zeros(ng, np.int64)
+        np.add.at(exp_si, gids, vals_i)
+        assert np.array_equal(np.asarray(out[3]["values"])[perm], exp_si)
+        exp_min = np.full(ng, np.iinfo(np.int64).max)
+        np.minimum.at(exp_min, gids, vals_i)
+        exp_max = np.full(ng, np.iinfo(np.int64).min)
+        np.maximum.at(exp_max, gids, vals_i)
+        assert np.array_equal(np.asarray(out[4]["values"])[perm], exp_min)
+        assert np.array_equal(np.asarray(out[5]["values"])[perm], exp_max)
+        exp_sf = oracle.grouped_sum_f64(gids, vals_f, ng)
+        got_sf = np.asarray(out[2]["values"])[perm]
+        assert np.allclose(got_sf, exp_sf, rtol=1e-12, atol=1e-9)
+
+    def test_pass_through_avg_and_exact_states(self, sess, ops):
+        """Two-channel partial states (AVG count+sum, EXACT lo+hi) through
+        the pass-through path combine exactly in FINAL."""
+        r = rng(91)
+        n = 40_000
+        keys = r.integers(0, 97, n).astype(np.int64)
+        vals = r.integers(90100, 209900, n) / 100.0    # v*2^43 integer
+        aggs_p = [(ops.AGG_AVG_F64, 1), (ops.AGG_SUM_F64_EXACT, 1, 43)]
+        ctrl = ops.PartialAggController(max_partial_bytes=1000, threshold=0.8)
+        # force-disable via a synthetic flush (reference: another driver of
+        # the plan node already flipped it)
+        ctrl.on_flush(2000, 100, 100, True)
+        assert ctrl.disabled
+        op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT], aggs_p,
+                                  step=ops.STEP_PARTIAL)
+        ctrl.attach(op)
+        op.add_input(ops.page_from_numpy([keys, vals]))
+        pout = op.drain()[0]
+        op.close()
+        ctrl.close()
+        # layout: key, avg_cnt, avg_sum, exact_lo, exact_hi
+        assert np.array_equal(pout[0]["values"], keys)
+        assert np.all(np.asarray(pout[1]["values"]) == 1)
+        assert np.array_equal(pout[2]["values"], vals)
+        fin = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                   [(ops.AGG_AVG_F64, 1),
+                                    (ops.AGG_SUM_F64_EXACT, 3, 43)],
+                                   step=ops.STEP_FINAL)
+        fin.add_input(ops.page_from_numpy(
+            [np.asarray(b["values"]) for b in pout]))
+        out = fin.drain()[0]
+        fin.close()
+        gids, ng, by_gid, _ = oracle.bigint_groupby(keys)
+        exp = oracle.grouped_sum_f64_exact(gids, vals, ng, scale_pow=43)
+        assert np.array_equal(out[0]["values"], by_gid)
+        assert np.array_equal(np.asarray(out[2]["values"]), exp)  # bit-exact
+        cnt = oracle.grouped_count(gids, ng)
+        got_avg = np.asarray(out[1]["values"])
+        assert np.allclose(got_avg, exp / cnt, rtol=1e-12)
+
+    def test_unsupported_specs_rejected(self, sess, ops):
+        import trino_amd
+        ctrl = ops.PartialAggController()
+        # SINGLE step: rejected
+        op = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                  [(ops.AGG_COUNT_STAR, -1)])
+        with pytest.raises(trino_amd.TrinoGpuError):
+            ctrl.attach(op)
+        op.close()
+        # VARCHAR group key: rejected
+        op = ops.hash_aggregation(sess, [0], [ops.TG_VARCHAR],
+                                  [(ops.AGG_COUNT_STAR, -1)],
+                                  step=ops.STEP_PARTIAL)
+        with pytest.raises(trino_amd.TrinoGpuError):
+            ctrl.attach(op)
+        op.close()
+        ctrl.close()

@@ -709,6 +709,107 @@ __global__ void k_emit_avg(const double* __restrict__ sum, const long long* __re
 #include <algorithm>
 #include <numeric>
 
+#include <atomic>
+#include <mutex>
+
+/* ---- adaptive partial aggregation ----------------------------------------
+ * PartialAggregationController analog (operator/aggregation/partial/
+ * PartialAggregationController.java:34-100): shared across the PARTIAL-step
+ * HashAggOps of one plan node. After sampling >= 1.5x max_partial_bytes of
+ * input, if the unique(output)/input row ratio exceeds the threshold
+ * (reference session default 0.8), partial aggregation is disabled — pages
+ * are re-shaped into partial-state layout with no hash-table work and flow
+ * straight through (the FINAL stage does the real grouping). Re-enabled
+ * after 200x more input bytes in case efficiency improved (same constants
+ * as the reference). */
+struct tg_pa_controller {
+    std::mutex m;
+    std::atomic<bool> disabled{false};
+    int64_t max_partial_bytes = 0;
+    double unique_ratio_threshold = 0.8;
+    int64_t total_bytes = 0, total_rows = 0, total_unique = 0;
+
+    void on_flush(int64_t bytes, int64_t rows, int64_t unique, int have_unique)
+    {
+        std::lock_guard<std::mutex> g(m);
+        bool dis = disabled.load(std::memory_order_relaxed);
+        /* when PA was re-enabled, ignore stats from disabled-mode flushes
+         * (PartialAggregationController.java:69-72) */
+        if (!dis && !have_unique) return;
+        total_bytes += bytes;
+        total_rows += rows;
+        if (have_unique) total_unique += unique;
+        if (!dis && total_bytes >= (int64_t)(max_partial_bytes * 1.5) &&
+            total_rows > 0 &&
+            (double)total_unique / (double)total_rows > unique_ratio_threshold) {
+            disabled.store(true, std::memory_order_relaxed);
+            dis = true;
+        }
+        if (dis && total_bytes >= (int64_t)(max_partial_bytes * 1.5) * 200) {
+            total_bytes = total_rows = total_unique = 0;
+            disabled.store(false, std::memory_order_relaxed);
+        }
+    }
+};
+
+/* pass-through re-shape: one raw input row -> its partial-state form (the
+ * exact channel layout HashAggOp::emit produces for PARTIAL), null inputs
+ * mapped to the combine-neutral element of each aggregate. out1 only for
+ * the two-channel states (AVG count+sum, EXACT lo+hi). */
+__global__ void k_pa_state(const KColH* __restrict__ cols, int64_t n, KAgg ag,
+                           long long* __restrict__ out0,
+                           long long* __restrict__ out1)
+{
+    int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = i0; i < n; i += stride) {
+        switch (ag.fn) {
+            case TG_AGG_COUNT_STAR:
+                out0[i] = 1;
+                break;
+            case TG_AGG_COUNT_COL:
+                out0[i] = kcol_is_null(cols[ag.in_ch], i) ? 0 : 1;
+                break;
+            case TG_AGG_SUM_I64:
+                out0[i] = kcol_is_null(cols[ag.in_ch], i) ? 0
+                        : ((const int64_t*)cols[ag.in_ch].data)[i];
+                break;
+            case TG_AGG_SUM_F64:
+                ((double*)out0)[i] = kcol_is_null(cols[ag.in_ch], i) ? 0.0
+                        : ((const double*)cols[ag.in_ch].data)[i];
+                break;
+            case TG_AGG_SUM_F64_EXACT: {
+                __int128 yi = 0;
+                if (!kcol_is_null(cols[ag.in_ch], i))
+                    yi = (__int128)(long long)
+                         (((const double*)cols[ag.in_ch].data)[i] * ag.scale);
+                out0[i] = (long long)(unsigned long long)(unsigned __int128)yi;
+                out1[i] = (long long)(unsigned long long)
+                          ((unsigned __int128)yi >> 64);
+                break;
+            }
+            case TG_AGG_MIN_I64:
+                out0[i] = kcol_is_null(cols[ag.in_ch], i) ? INT64_MAX
+                        : ((const int64_t*)cols[ag.in_ch].data)[i];
+                break;
+            case TG_AGG_MAX_I64:
+                out0[i] = kcol_is_null(cols[ag.in_ch], i) ? INT64_MIN
+                        : ((const int64_t*)cols[ag.in_ch].data)[i];
+                break;
+            case TG_AGG_AVG_F64:
+                if (kcol_is_null(cols[ag.in_ch], i)) {
+                    out0[i] = 0;
+                    ((double*)out1)[i] = 0.0;
+                }
+                else {
+                    out0[i] = 1;
+                    ((double*)out1)[i] = ((const double*)cols[ag.in_ch].data)[i];
+                }
+                break;
+        }
+    }
+}
+
 struct HashAggOp : tg_operator {
     std::vector<int32_t> group_channels;
     std::vector<tg_type> group_types;
@@ -723,6 +824,8 @@ struct HashAggOp : tg_operator {
     int32_t* d_group_types = nullptr;  /* device copy for rehash/emit */
     int64_t rows_seen = 0;
     bool emitted = false;
+    tg_pa_controller* pa = nullptr;    /* adaptive partial agg (not owned) */
+    int64_t pa_bytes = 0, pa_rows = 0; /* input accounted since last flush */
 
     tg_status init_table(int64_t cap, int64_t max_groups)
     {
@@ -841,11 +944,81 @@ struct HashAggOp : tg_operator {
         return TG_OK;
     }
 
+    static int64_t page_bytes(const DevPage& p)
+    {
+        int64_t b = 0;
+        for (const DevBlock& blk : p.blocks) b += blk.n * blk.elem_size();
+        return b;
+    }
+
+    /* partial aggregation adaptively disabled: emit the page in partial-
+     * state layout with no hash-table work (HashAggregationOperator's
+     * spillToDiskAndClear-free adaptive path) */
+    tg_status pass_through(DevPage&& in)
+    {
+        DevPage outp;
+        outp.n = in.n;
+        int64_t n = in.n;
+        int64_t in_bytes = page_bytes(in);
+        for (int32_t gc : group_channels) {
+            const DevBlock& src = in.blocks[gc];
+            DevBlock b;
+            b.type = src.type;
+            b.n = n;
+            TG_POOL_ALLOC(s, &b.data, (n ? n : 1) * src.elem_size());
+            TG_HIP_CHECK(hipMemcpyAsync(b.data, src.data, n * src.elem_size(),
+                                        hipMemcpyDeviceToDevice, s->stream));
+            if (src.valid) {
+                int64_t words = (n + 63) / 64;
+                TG_POOL_ALLOC(s, &b.valid, (words ? words : 1) * 8);
+                TG_HIP_CHECK(hipMemcpyAsync(b.valid, src.valid, words * 8,
+                                            hipMemcpyDeviceToDevice, s->stream));
+            }
+            outp.blocks.push_back(b);
+        }
+        KColH* d_all = nullptr;
+        tg_status st = make_kcols(s, in, nullptr, (int)in.blocks.size(), &d_all);
+        if (st != TG_OK) { tg_free_page(s, &in); return st; }
+        for (const KAgg& ag : agg_state) {
+            bool two = (ag.fn == TG_AGG_AVG_F64 || ag.fn == TG_AGG_SUM_F64_EXACT);
+            DevBlock b0;
+            b0.type = (ag.fn == TG_AGG_SUM_F64) ? TG_DOUBLE : TG_BIGINT;
+            b0.n = n;
+            TG_POOL_ALLOC(s, &b0.data, (n ? n : 1) * 8);
+            DevBlock b1;
+            if (two) {
+                b1.type = (ag.fn == TG_AGG_AVG_F64) ? TG_DOUBLE : TG_BIGINT;
+                b1.n = n;
+                TG_POOL_ALLOC(s, &b1.data, (n ? n : 1) * 8);
+            }
+            hipLaunchKernelGGL(k_pa_state, dim3(tg_grid_for(n)), dim3(TG_BLOCK),
+                               0, s->stream, d_all, n, ag, (long long*)b0.data,
+                               two ? (long long*)b1.data : nullptr);
+            TG_HIP_CHECK(hipGetLastError());
+            outp.blocks.push_back(b0);
+            if (two) outp.blocks.push_back(b1);
+        }
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        tg_pool_free(s, d_all);
+        tg_free_page(s, &in);
+        /* disabled-mode flush: bytes/rows only (no unique count) — feeds
+         * the re-enable counter (PartialAggregationController.java:82-88) */
+        pa->on_flush(in_bytes, n, 0, 0);
+        stage_output(std::move(outp));
+        return TG_OK;
+    }
+
     tg_status add_input(const tg_page* page) override
     {
         DevPage in;
         tg_status st = tg_upload_page(s, page, &in);
         if (st != TG_OK) return st;
+        if (pa && step == TG_STEP_PARTIAL) {
+            if (pa->disabled.load(std::memory_order_relaxed))
+                return pass_through(std::move(in));
+            pa_bytes += page_bytes(in);
+            pa_rows += in.n;
+        }
         KColH* d_keys = nullptr;
         int32_t* d_gids = nullptr;
         if (group_channels.empty()) {
@@ -1094,6 +1267,11 @@ have_gids:;
         }
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
         tg_pool_free(s, d_obn);
+        if (pa && step == TG_STEP_PARTIAL) {
+            /* flush accounting: unique groups vs rows in (onFlush analog) */
+            pa->on_flush(pa_bytes, pa_rows, ng, 1);
+            pa_bytes = pa_rows = 0;
+        }
         stage_output(std::move(outp));
         return TG_OK;
     }
@@ -1424,5 +1602,68 @@ extern "C" tg_status tg_mark_distinct_create(tg_session* s,
                                               TG_STEP_SINGLE, out);
     if (st != TG_OK) return st;
     static_cast<HashAggOp*>(*out)->mark_distinct = true;
+    return TG_OK;
+}
+
+/* ---- adaptive partial aggregation ABI ---- */
+extern "C" tg_status tg_pa_controller_create(int64_t max_partial_bytes,
+    double unique_rows_ratio_threshold, tg_pa_controller** out)
+{
+    if (!out || max_partial_bytes <= 0 || unique_rows_ratio_threshold <= 0.0) {
+        TG_SET_ERR("invalid partial aggregation controller spec");
+        return TG_ERR_INVALID_ARG;
+    }
+    auto* c = new tg_pa_controller();
+    c->max_partial_bytes = max_partial_bytes;
+    c->unique_ratio_threshold = unique_rows_ratio_threshold;
+    *out = c;
+    return TG_OK;
+}
+
+extern "C" void tg_pa_controller_close(tg_pa_controller* c) { delete c; }
+
+extern "C" int32_t tg_pa_controller_disabled(tg_pa_controller* c)
+{
+    return (c && c->disabled.load(std::memory_order_relaxed)) ? 1 : 0;
+}
+
+extern "C" tg_status tg_pa_controller_on_flush(tg_pa_controller* c,
+    int64_t bytes, int64_t rows, int64_t unique_rows, int32_t have_unique)
+{
+    if (!c) {
+        TG_SET_ERR("null controller");
+        return TG_ERR_INVALID_ARG;
+    }
+    c->on_flush(bytes, rows, unique_rows, have_unique);
+    return TG_OK;
+}
+
+extern "C" tg_status tg_hash_aggregation_set_controller(tg_operator* op,
+    tg_pa_controller* c)
+{
+    auto* h = dynamic_cast<HashAggOp*>(op);
+    if (!h || !c) {
+        TG_SET_ERR("controller requires a hash aggregation operator");
+        return TG_ERR_INVALID_ARG;
+    }
+    if (h->step != TG_STEP_PARTIAL || h->mark_distinct) {
+        TG_SET_ERR("adaptive partial aggregation applies to the PARTIAL step");
+        return TG_ERR_INVALID_ARG;
+    }
+    for (tg_type ty : h->group_types) {
+        if (ty == TG_VARCHAR) {
+            TG_SET_ERR("adaptive partial aggregation: VARCHAR group keys "
+                       "unsupported (pass-through copies fixed-width keys)");
+            return TG_ERR_UNSUPPORTED;
+        }
+    }
+    for (const tg_agg_spec& a : h->aggs) {
+        if (a.mask_gt_a >= 0 && a.mask_gt_a != a.mask_gt_b) {
+            TG_SET_ERR("adaptive partial aggregation: masked aggregates "
+                       "unsupported");
+            return TG_ERR_UNSUPPORTED;
+        }
+    }
+    h->pa = c;
     return TG_OK;
 }

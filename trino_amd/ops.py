@@ -632,6 +632,51 @@ _lib.tg_mark_distinct_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                          ctypes.c_void_p]
 
 
+_lib.tg_pa_controller_create.restype = ctypes.c_int
+_lib.tg_pa_controller_create.argtypes = [ctypes.c_int64, ctypes.c_double,
+                                         ctypes.c_void_p]
+_lib.tg_pa_controller_close.restype = None
+_lib.tg_pa_controller_close.argtypes = [ctypes.c_void_p]
+_lib.tg_pa_controller_disabled.restype = ctypes.c_int32
+_lib.tg_pa_controller_disabled.argtypes = [ctypes.c_void_p]
+_lib.tg_pa_controller_on_flush.restype = ctypes.c_int
+_lib.tg_pa_controller_on_flush.argtypes = [ctypes.c_void_p, ctypes.c_int64,
+                                           ctypes.c_int64, ctypes.c_int64,
+                                           ctypes.c_int32]
+_lib.tg_hash_aggregation_set_controller.restype = ctypes.c_int
+_lib.tg_hash_aggregation_set_controller.argtypes = [ctypes.c_void_p,
+                                                    ctypes.c_void_p]
+
+
+class PartialAggController:
+    """PartialAggregationController analog (adaptive partial aggregation):
+    shared across the PARTIAL-step hash aggregations of one plan node.
+    Disables partial aggregation when unique/input rows > threshold after
+    1.5x max_partial_bytes of input; re-enables after 200x more bytes."""
+
+    def __init__(self, max_partial_bytes=16 << 20, threshold=0.8):
+        h = ctypes.c_void_p()
+        _check(_lib.tg_pa_controller_create(max_partial_bytes, threshold,
+                                            ctypes.byref(h)))
+        self._h = h
+
+    @property
+    def disabled(self):
+        return bool(_lib.tg_pa_controller_disabled(self._h))
+
+    def on_flush(self, bytes_, rows, unique_rows=0, have_unique=False):
+        _check(_lib.tg_pa_controller_on_flush(self._h, bytes_, rows,
+                                              unique_rows, 1 if have_unique else 0))
+
+    def attach(self, op):
+        _check(_lib.tg_hash_aggregation_set_controller(op._h, self._h))
+
+    def close(self):
+        if self._h:
+            _lib.tg_pa_controller_close(self._h)
+            self._h = None
+
+
 def mark_distinct(session, key_channels, key_types):
     """MarkDistinctOperator analog: pass-through + BOOLEAN first-occurrence
     channel appended after the input channels."""
