@@ -406,6 +406,13 @@ tg_status tg_mark_distinct_create(tg_session*, const int32_t* key_channels,
 tg_status tg_dense_aggregation_create(tg_session*, int32_t key_channel,
     int64_t key_min, int64_t key_max, const tg_agg_spec* agg, tg_operator**);
 
+/* sort-based DISTINCT of a non-negative BIGINT device column (radix sort +
+ * unique compaction; `bits` = key width to sort). For near-all-unique
+ * dedups (count(DISTINCT ...)) where hash aggregation pays random keystore
+ * probes per row. d_out sized n; *out_n gets the distinct count. */
+tg_status tg_dedup_i64(tg_session*, const int64_t* d_in, int64_t n,
+    int32_t bits, int64_t* d_out, int64_t* out_n);
+
 /* stream timer (HIP events on the session stream) for bench rooflines */
 tg_status tg_timer_start(tg_session*);
 tg_status tg_timer_stop(tg_session*, double* elapsed_ms);

@@ -1456,7 +1456,11 @@ class TestAdaptivePartialAgg:
         aggs_p = [(ops.AGG_COUNT_STAR, -1), (ops.AGG_SUM_F64, 1),
                   (ops.AGG_SUM_I64, 2), (ops.AGG_MIN_I64, 2),
                   (ops.AGG_MAX_I64, 2)]
-        ctrl = ops.PartialAggController(max_partial_bytes=1000, threshold=0.8)
+        # 1.5x threshold (150 KB) well under the first flush's 720 KB, and
+        # the 200x re-enable bound (30 MB) well above it — a realistic
+        # mid-stream disable (with tiny max_partial_bytes one big flush
+        # crosses BOTH bounds in a single onFlush, like the reference would)
+        ctrl = ops.PartialAggController(max_partial_bytes=100_000, threshold=0.8)
         partial_pages = []
         op1 = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT], aggs_p,
                                    step=ops.STEP_PARTIAL)
@@ -1576,3 +1580,35 @@ class TestAdaptivePartialAgg:
             ctrl.attach(op)
         op.close()
         ctrl.close()
+
+
+class TestDedupSort:
+    """tg_dedup_i64 (radix sort + unique compaction) == np.unique, including
+    the reduced-bit-width path used by Q16's packed keys."""
+
+    def test_matches_np_unique(self, sess, ops):
+        import ctypes
+        from trino_amd import copy_dtoh
+        from trino_amd.ops import _lib
+        _lib.tg_device_malloc.restype = ctypes.c_int
+        _lib.tg_device_malloc.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                          ctypes.c_int64]
+        _lib.tg_device_free.restype = ctypes.c_int
+        _lib.tg_device_free.argtypes = [ctypes.c_void_p, ctypes.c_void_p]
+        r = rng(95)
+        for n, hi, bits in ((1, 10, 64), (1000, 50, 64),
+                            (200_000, 1 << 40, 52), (500_000, 300_000, 64)):
+            keys = r.integers(0, hi, n).astype(np.int64)
+            d_in = ctypes.c_void_p()
+            d_out = ctypes.c_void_p()
+            assert _lib.tg_device_malloc(sess._h, ctypes.byref(d_in), n * 8) == 0
+            assert _lib.tg_device_malloc(sess._h, ctypes.byref(d_out), n * 8) == 0
+            assert _lib.tg_copy_htod(sess._h, d_in, keys.ctypes.data, n * 8) == 0
+            m = ops.dedup_i64(sess, d_in, n, d_out, bits=bits)
+            exp = np.unique(keys)
+            assert m == len(exp)
+            got = np.empty(m, np.int64)
+            copy_dtoh(sess, got, d_out)
+            assert np.array_equal(got, exp)        # sorted ascending
+            _lib.tg_device_free(sess._h, d_in)
+            _lib.tg_device_free(sess._h, d_out)

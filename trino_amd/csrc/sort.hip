@@ -101,3 +101,63 @@ tg_status run_argsort_i64(tg_session* s, int64_t* d_keys, int64_t n, int32_t* d_
     tg_pool_free(s, d_vals);
     return TG_OK;
 }
+
+/* sort-based DISTINCT of a non-negative BIGINT column: radix_sort_keys
+ * + rocprim::unique compaction, sequential-bandwidth only. Replaces the
+ * generic hash aggregation dedup when nearly every row is unique (Q16's
+ * ~12M distinct (combo,suppkey) pairs: the keystore insert + group-remap
+ * sort pay random HBM round-trips per row; the sorted path streams).
+ * d_out must hold n entries; *h_out_n gets the distinct count. `bits` is
+ * the key width to sort (<=64; fewer radix passes for packed keys). */
+extern "C" tg_status tg_dedup_i64(tg_session* s, const int64_t* d_in,
+    int64_t n, int32_t bits, int64_t* d_out, int64_t* h_out_n)
+{
+    if (!s || !d_in || !d_out || !h_out_n || n < 0 || bits < 1 || bits > 64) {
+        TG_SET_ERR("invalid dedup spec");
+        return TG_ERR_INVALID_ARG;
+    }
+    if (n <= 1) {
+        if (n == 1)
+            TG_HIP_CHECK(hipMemcpyAsync(d_out, d_in, 8,
+                                        hipMemcpyDeviceToDevice, s->stream));
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        *h_out_n = n;
+        return TG_OK;
+    }
+    uint64_t* d_a = nullptr;
+    uint64_t* d_b = nullptr;
+    TG_POOL_ALLOC(s, &d_a, n * 8);
+    TG_POOL_ALLOC(s, &d_b, n * 8);
+    TG_HIP_CHECK(hipMemcpyAsync(d_a, d_in, n * 8, hipMemcpyDeviceToDevice,
+                                s->stream));
+    size_t temp_bytes = 0;
+    hipError_t e0 = rocprim::radix_sort_keys(nullptr, temp_bytes, d_a, d_b,
+                                             (size_t)n, 0, bits, s->stream);
+    if (e0 != hipSuccess) { TG_SET_ERR("rocprim size query: %s", hipGetErrorName(e0)); return TG_ERR_HIP; }
+    void* d_temp = nullptr;
+    TG_POOL_ALLOC(s, &d_temp, (int64_t)temp_bytes);
+    hipError_t e = rocprim::radix_sort_keys(d_temp, temp_bytes, d_a, d_b,
+                                            (size_t)n, 0, bits, s->stream);
+    if (e != hipSuccess) { TG_SET_ERR("rocprim radix_sort_keys: %s", hipGetErrorName(e)); return TG_ERR_HIP; }
+    tg_pool_free(s, d_temp);
+    size_t u_bytes = 0;
+    int64_t* d_count = nullptr;
+    TG_POOL_ALLOC(s, &d_count, 8);
+    e0 = rocprim::unique(nullptr, u_bytes, d_b, (uint64_t*)d_out,
+                         (int64_t*)d_count, (size_t)n,
+                         rocprim::equal_to<uint64_t>(), s->stream);
+    if (e0 != hipSuccess) { TG_SET_ERR("rocprim unique size query: %s", hipGetErrorName(e0)); return TG_ERR_HIP; }
+    TG_POOL_ALLOC(s, &d_temp, (int64_t)u_bytes);
+    e = rocprim::unique(d_temp, u_bytes, d_b, (uint64_t*)d_out,
+                        (int64_t*)d_count, (size_t)n,
+                        rocprim::equal_to<uint64_t>(), s->stream);
+    if (e != hipSuccess) { TG_SET_ERR("rocprim unique: %s", hipGetErrorName(e)); return TG_ERR_HIP; }
+    TG_HIP_CHECK(hipMemcpyAsync(h_out_n, d_count, 8, hipMemcpyDeviceToHost,
+                                s->stream));
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    tg_pool_free(s, d_a);
+    tg_pool_free(s, d_b);
+    tg_pool_free(s, d_temp);
+    tg_pool_free(s, d_count);
+    return TG_OK;
+}

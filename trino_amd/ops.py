@@ -632,6 +632,21 @@ _lib.tg_mark_distinct_create.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                          ctypes.c_void_p]
 
 
+_lib.tg_dedup_i64.restype = ctypes.c_int
+_lib.tg_dedup_i64.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                              ctypes.c_int64, ctypes.c_int32,
+                              ctypes.c_void_p, ctypes.c_void_p]
+
+
+def dedup_i64(session, d_in, n, d_out, bits=64):
+    """Sort-based DISTINCT of a device BIGINT column; returns the distinct
+    count (results sorted ascending in d_out)."""
+    out_n = ctypes.c_int64(0)
+    _check(_lib.tg_dedup_i64(session._h, d_in, n, bits, d_out,
+                             ctypes.byref(out_n)))
+    return out_n.value
+
+
 _lib.tg_pa_controller_create.restype = ctypes.c_int
 _lib.tg_pa_controller_create.argtypes = [ctypes.c_int64, ctypes.c_double,
                                          ctypes.c_void_p]

@@ -1392,10 +1392,17 @@ def q16_gpu(session, sf, part_count=None):
     fpk.add_input(joined)
     fpk.finish()
     packed = _take_device_page(session, fpk)
-    d1 = ops.hash_aggregation(session, [0], [ops.TG_BIGINT], [])
-    d1.add_input(packed)
-    d1.finish()
-    dedup = _take_device_page(session, d1)
+    # DISTINCT via sort-based dedup (radix sort + unique compaction):
+    # ~11.9M of the 11.88M packed keys are unique, so the hash-agg dedup
+    # paid random keystore probes + a group-remap sort for nothing
+    # (profiled 31.0 ms of Q16's 51 ms at SF100; tg_dedup_i64 sorts
+    # 52 packed bits sequentially instead)
+    n_packed = packed.position_count
+    d_dedup = _device_buffer(session, max(n_packed, 1) * 8)
+    n_uniq = ops.dedup_i64(session, packed.blocks[0].data, n_packed,
+                           d_dedup, bits=52)
+    dedup = ops.page_from_device(session, ([(d_dedup.value, ops.TG_BIGINT)],
+                                           n_uniq))
     # group by packed >> 32 (brand/type/size combo) counting suppliers
     fg = ops.filter_project(session, None,
                             [ops.expr(("col", 0), ("i64", 1 << 32), "div")],
@@ -1408,11 +1415,11 @@ def q16_gpu(session, sf, part_count=None):
     d2.add_input(combos)
     pages = d2.drain()
     elapsed = time.time() - t0
-    for op in (bb, fp, bp, sj, fnb, j, fpk, d1, fg, d2):
+    for op in (bb, fp, bp, sj, fnb, j, fpk, fg, d2):
         op.close()
     bridge_bad.close()
     bridge_p.close()
-    for p in (p_pk, p_ty, p_br, p_sz, ps_pk, ps_sk, d_cflag):
+    for p in (p_pk, p_ty, p_br, p_sz, ps_pk, ps_sk, d_cflag, d_dedup):
         _device_free(session, p)
     out = pages[0]
     combo = np.asarray(out[0]["values"]).astype(np.int64)
