@@ -826,6 +826,7 @@ struct HashAggOp : tg_operator {
     bool emitted = false;
     tg_pa_controller* pa = nullptr;    /* adaptive partial agg (not owned) */
     int64_t pa_bytes = 0, pa_rows = 0; /* input accounted since last flush */
+    bool passed_through = false;       /* some pages bypassed the table */
 
     tg_status init_table(int64_t cap, int64_t max_groups)
     {
@@ -1004,6 +1005,7 @@ struct HashAggOp : tg_operator {
         /* disabled-mode flush: bytes/rows only (no unique count) — feeds
          * the re-enable counter (PartialAggregationController.java:82-88) */
         pa->on_flush(in_bytes, n, 0, 0);
+        passed_through = true;
         stage_output(std::move(outp));
         return TG_OK;
     }
@@ -1138,6 +1140,10 @@ have_gids:;
     tg_status emit() /* build the output page: groups in remapped id order */
     {
         int32_t ng = n_groups_host();
+        /* every page bypassed the table (adaptive pass-through): the
+         * partial states are already staged; no hash-table output page */
+        if (passed_through && !group_channels.empty() && ng == 0)
+            return TG_OK;
         /* remap: order groups by first-occurrence row (== the reference's
          * insertion-order ids, deterministically) */
         int32_t* d_obn = nullptr;
