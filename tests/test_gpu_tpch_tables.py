@@ -214,3 +214,32 @@ class TestMinMaxAgg:
         assert np.array_equal(out[1]["values"], emin)
         assert np.array_equal(out[2]["values"], emax)
         assert np.array_equal(out[3]["values"], reps.astype(np.int64))
+
+
+class TestIndexedPoolLike:
+    def test_indexed_vs_byte_scan(self, sess, ops):
+        """The indexed pool LIKE (segment-occurrence index over the 300 MiB
+        pool + per-row binary searches) must agree bitwise with the per-row
+        byte-scan kernel on the same slices (TG_LIKE_IDX forces each path)."""
+        import os
+        from trino_amd import _lib, _check
+        no = int(1_500_000 * SF)
+        bo = [None, None, None, None, None, None,
+              dbuf(sess, no * 8), dbuf(sess, no * 4)]
+        _check(_lib.tg_tpch_gen_orders3(sess._h, SF, 1, no, *bo))
+        for pat in ("%special%requests%", "%furiousl%", "%ironic%the%"):
+            d_a = dbuf(sess, no)
+            d_b = dbuf(sess, no)
+            os.environ["TG_LIKE_IDX"] = "1"       # force indexed
+            ops.pool_like_flags(sess, bo[6], bo[7], no, pat, d_a)
+            os.environ["TG_LIKE_IDX"] = "0"       # force byte scan
+            ops.pool_like_flags(sess, bo[6], bo[7], no, pat, d_b)
+            del os.environ["TG_LIKE_IDX"]
+            a = dl(sess, np.empty(no, np.uint8), d_a)
+            b = dl(sess, np.empty(no, np.uint8), d_b)
+            assert np.array_equal(a, b), pat
+            assert a.sum() > 0                    # pattern actually matches
+            dfree(sess, d_a)
+            dfree(sess, d_b)
+        for x in (bo[6], bo[7]):
+            dfree(sess, x)

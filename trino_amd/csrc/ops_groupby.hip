@@ -468,6 +468,17 @@ __global__ void k_agg_update_sorted(const int32_t* __restrict__ gids, int64_t n,
         int32_t g = active ? gids[i] : -1;
         int32_t gnext = __shfl_down(g, 1, 64);
         bool is_last = active && (lane == 63 || i + 1 >= n || gnext != g);
+        /* run topology is agg-invariant: hoist the per-step "stays in my
+         * run" masks out of the agg loop (was re-shuffling g for every
+         * aggregate — q21's 5-agg pass re-ran the whole 4-register scan
+         * 5x; with the masks hoisted each agg scans only the registers
+         * its class needs: 30.0 ms -> see profiles/) */
+        bool segok[6];
+        #pragma unroll
+        for (int off = 1, k = 0; off < 64; off <<= 1, k++) {
+            int32_t og = __shfl_up(g, off, 64);
+            segok[k] = (lane >= off) && (og == g);
+        }
         for (int a = 0; a < n_aggs; a++) {
             KAgg ag = aggs[a];
             int mmode = (ag.fn == TG_AGG_MIN_I64) ? 1
@@ -533,21 +544,43 @@ __global__ void k_agg_update_sorted(const int32_t* __restrict__ gids, int64_t n,
                         break;
                 }
             }
-            /* segmented inclusive scan on equal-gid prefixes */
-            #pragma unroll
-            for (int off = 1; off < 64; off <<= 1) {
-                long long oci = __shfl_up(ci, off, 64);
-                double ocf = __shfl_up(cf, off, 64);
-                unsigned long long olo = __shfl_up(lo, off, 64);
-                unsigned long long ohi = __shfl_up(hi, off, 64);
-                int32_t og = __shfl_up(g, off, 64);
-                if (lane >= off && og == g) {
-                    ci = mmode == 1 ? (oci < ci ? oci : ci)
-                       : mmode == 2 ? (oci > ci ? oci : ci) : ci + oci;
-                    cf += ocf;
-                    unsigned long long nlo = lo + olo;
-                    hi = hi + ohi + (nlo < lo ? 1ull : 0ull);
-                    lo = nlo;
+            /* segmented inclusive scan on equal-gid prefixes — only the
+             * registers this aggregate class carries */
+            if (ag.fn == TG_AGG_SUM_F64) {
+                #pragma unroll
+                for (int off = 1, k = 0; off < 64; off <<= 1, k++) {
+                    double ocf = __shfl_up(cf, off, 64);
+                    if (segok[k]) cf += ocf;
+                }
+            }
+            else if (ag.fn == TG_AGG_AVG_F64) {
+                #pragma unroll
+                for (int off = 1, k = 0; off < 64; off <<= 1, k++) {
+                    long long oci = __shfl_up(ci, off, 64);
+                    double ocf = __shfl_up(cf, off, 64);
+                    if (segok[k]) { ci += oci; cf += ocf; }
+                }
+            }
+            else if (ag.fn == TG_AGG_SUM_F64_EXACT) {
+                #pragma unroll
+                for (int off = 1, k = 0; off < 64; off <<= 1, k++) {
+                    unsigned long long olo = __shfl_up(lo, off, 64);
+                    unsigned long long ohi = __shfl_up(hi, off, 64);
+                    if (segok[k]) {
+                        unsigned long long nlo = lo + olo;
+                        hi = hi + ohi + (nlo < lo ? 1ull : 0ull);
+                        lo = nlo;
+                    }
+                }
+            }
+            else {   /* COUNT/SUM_I64/MIN/MAX: single i64 register */
+                #pragma unroll
+                for (int off = 1, k = 0; off < 64; off <<= 1, k++) {
+                    long long oci = __shfl_up(ci, off, 64);
+                    if (segok[k]) {
+                        ci = mmode == 1 ? (oci < ci ? oci : ci)
+                           : mmode == 2 ? (oci > ci ? oci : ci) : ci + oci;
+                    }
                 }
             }
             if (is_last) {

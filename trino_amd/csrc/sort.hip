@@ -161,3 +161,29 @@ extern "C" tg_status tg_dedup_i64(tg_session* s, const int64_t* d_in,
     tg_pool_free(s, d_count);
     return TG_OK;
 }
+
+/* keys-only ascending u32 sort (segment-occurrence index for the indexed
+ * pool LIKE; positions emitted by atomic append arrive unordered) */
+tg_status run_sort_keys_u32(tg_session* s, uint32_t* d_keys, int64_t n)
+{
+    if (n <= 1) return TG_OK;
+    uint32_t* d_keys2 = nullptr;
+    TG_POOL_ALLOC(s, &d_keys2, n * 4);
+    size_t temp_bytes = 0;
+    hipError_t e0 = rocprim::radix_sort_keys(nullptr, temp_bytes, d_keys,
+                                             d_keys2, (size_t)n, 0, 32,
+                                             s->stream);
+    if (e0 != hipSuccess) { TG_SET_ERR("rocprim size query: %s", hipGetErrorName(e0)); return TG_ERR_HIP; }
+    void* d_temp = nullptr;
+    TG_POOL_ALLOC(s, &d_temp, (int64_t)temp_bytes);
+    hipError_t e = rocprim::radix_sort_keys(d_temp, temp_bytes, d_keys,
+                                            d_keys2, (size_t)n, 0, 32,
+                                            s->stream);
+    if (e != hipSuccess) { TG_SET_ERR("rocprim radix_sort_keys: %s", hipGetErrorName(e)); return TG_ERR_HIP; }
+    TG_HIP_CHECK(hipMemcpyAsync(d_keys, d_keys2, n * 4,
+                                hipMemcpyDeviceToDevice, s->stream));
+    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    tg_pool_free(s, d_keys2);
+    tg_pool_free(s, d_temp);
+    return TG_OK;
+}
