@@ -2108,15 +2108,12 @@ def q9_gpu(session, sf, order_count=None, part_count=None):
     pspage = ops.page_from_device(session, ([(ps_pk.value, ops.TG_BIGINT),
                                              (ps_sk.value, ops.TG_BIGINT),
                                              (ps_sc.value, ops.TG_BIGINT)], n))
-    sjp = ops.semi_join(session, bridge_g, 0)
-    sjp.add_input(pspage)
-    sjp.finish()
-    psm = _take_device_page(session, sjp)
-    fps = ops.filter_project(session, ops.expr(("col", 3), ("i64", 1), "eq"),
-                             [ops.expr(("col", 0)), ops.expr(("col", 1)),
-                              ops.expr(("col", 2))],
-                             [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT])
-    fps.add_input(psm)
+    fps = ops.filter_project_df(session, None,
+                                [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                                 ops.expr(("col", 2))],
+                                [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT],
+                                bridge_g, 0)
+    fps.add_input(pspage)
     fps.finish()
     ps_green = _take_device_page(session, fps)
     bridge_ps = ops.JoinBridge(session)
@@ -2142,18 +2139,18 @@ def q9_gpu(session, sf, order_count=None, part_count=None):
                                             (li.extendedprice, ops.TG_DOUBLE),
                                             (li.discount, ops.TG_DOUBLE)],
                                            li.row_count))
-    sjl = ops.semi_join(session, bridge_g, 0)
-    sjl.add_input(lpage)
-    sjl.finish()
-    lm = _take_device_page(session, sjl)
-    fl = ops.filter_project(session, ops.expr(("col", 6), ("i64", 1), "eq"),
-                            [ops.expr(("col", 0)), ops.expr(("col", 1)),
-                             ops.expr(("col", 2)), ops.expr(("col", 3)),
-                             ops.expr(("col", 4), ("f64", 1.0), ("col", 5),
-                                      "sub", "mul")],
-                            [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
-                             ops.TG_DOUBLE, ops.TG_DOUBLE])
-    fl.add_input(lm)
+    # green-part membership fused into the scan (DF-only filter over the
+    # set-builder bitmap): one pass instead of semi-join + compacting
+    # filter, which each re-streamed the 600M-row page
+    fl = ops.filter_project_df(session, None,
+                               [ops.expr(("col", 0)), ops.expr(("col", 1)),
+                                ops.expr(("col", 2)), ops.expr(("col", 3)),
+                                ops.expr(("col", 4), ("f64", 1.0), ("col", 5),
+                                         "sub", "mul")],
+                               [ops.TG_BIGINT, ops.TG_BIGINT, ops.TG_BIGINT,
+                                ops.TG_DOUBLE, ops.TG_DOUBLE],
+                               bridge_g, 0)
+    fl.add_input(lpage)
     fl.finish()
     lg = _take_device_page(session, fl)     # (pk, sk, ok, qty, rev)
     jps = ops.lookup_join(session, bridge_ps,
@@ -2215,7 +2212,7 @@ def q9_gpu(session, sf, order_count=None, part_count=None):
     agg.add_input(amounts)
     pages = agg.drain()
     elapsed = time.time() - t0
-    for op in (fp, bg, sjp, fps, bps, bw, fo, bn, sjl, fl, jps, jo, jn, fa, agg):
+    for op in (fp, bg, fps, bps, bw, fo, bn, fl, jps, jo, jn, fa, agg):
         op.close()
     for br in (bridge_g, bridge_ps, bridge_w, bridge_n):
         br.close()
