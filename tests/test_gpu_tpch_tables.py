@@ -228,18 +228,16 @@ class TestIndexedPoolLike:
               dbuf(sess, no * 8), dbuf(sess, no * 4)]
         _check(_lib.tg_tpch_gen_orders3(sess._h, SF, 1, no, *bo))
         for pat in ("%special%requests%", "%furiousl%", "%ironic%the%"):
-            d_a = dbuf(sess, no)
-            d_b = dbuf(sess, no)
-            os.environ["TG_LIKE_IDX"] = "1"       # force indexed
-            ops.pool_like_flags(sess, bo[6], bo[7], no, pat, d_a)
-            os.environ["TG_LIKE_IDX"] = "0"       # force byte scan
-            ops.pool_like_flags(sess, bo[6], bo[7], no, pat, d_b)
+            outs = []
+            for mode in ("0", "1", "2"):    # byte scan / index / sorted scan
+                d = dbuf(sess, no)
+                os.environ["TG_LIKE_IDX"] = mode
+                ops.pool_like_flags(sess, bo[6], bo[7], no, pat, d)
+                outs.append(dl(sess, np.empty(no, np.uint8), d))
+                dfree(sess, d)
             del os.environ["TG_LIKE_IDX"]
-            a = dl(sess, np.empty(no, np.uint8), d_a)
-            b = dl(sess, np.empty(no, np.uint8), d_b)
-            assert np.array_equal(a, b), pat
-            assert a.sum() > 0                    # pattern actually matches
-            dfree(sess, d_a)
-            dfree(sess, d_b)
+            assert np.array_equal(outs[0], outs[1]), pat
+            assert np.array_equal(outs[0], outs[2]), pat
+            assert outs[0].sum() > 0              # pattern actually matches
         for x in (bo[6], bo[7]):
             dfree(sess, x)

@@ -115,5 +115,7 @@ tg_status run_hash_rows(tg_session* s, const DevPage& page,
 tg_status run_sort_pairs(tg_session* s, uint64_t* d_keys, int64_t* d_vals, int64_t n);
 tg_status run_argsort_i64(tg_session* s, int64_t* d_keys, int64_t n, int32_t* d_out_idx);
 tg_status run_sort_keys_u32(tg_session* s, uint32_t* d_keys, int64_t n);
+tg_status run_sort_pairs_bits(tg_session* s, uint64_t* d_keys, int64_t* d_vals,
+                              int64_t n, int bits);
 tg_status run_sort_pairs_u32(tg_session* s, uint32_t* d_keys, int32_t* d_vals, int64_t n);
 tg_status run_scan_i32(tg_session* s, int32_t* d_arr, int64_t n, int32_t* d_total);

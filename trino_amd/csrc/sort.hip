@@ -26,8 +26,10 @@ __global__ void k_i64_to_i32(const int64_t* __restrict__ in, int64_t n,
 }
 
 /* stable ascending sort of (u64 key, i64 value) pairs; keys/vals updated
- * in place (double-buffered internally) */
-tg_status run_sort_pairs(tg_session* s, uint64_t* d_keys, int64_t* d_vals, int64_t n)
+ * in place (double-buffered internally). `bits` limits the radix passes
+ * when the key range is known (e.g. pool offsets < 2^29). */
+tg_status run_sort_pairs_bits(tg_session* s, uint64_t* d_keys, int64_t* d_vals,
+                              int64_t n, int bits)
 {
     if (n <= 1) return TG_OK;
     uint64_t* d_keys2 = nullptr;
@@ -36,13 +38,13 @@ tg_status run_sort_pairs(tg_session* s, uint64_t* d_keys, int64_t* d_vals, int64
     TG_POOL_ALLOC(s, &d_vals2, n * 8);
     size_t temp_bytes = 0;
     hipError_t e0 = rocprim::radix_sort_pairs(nullptr, temp_bytes, d_keys, d_keys2,
-                                              d_vals, d_vals2, (size_t)n, 0, 64,
+                                              d_vals, d_vals2, (size_t)n, 0, bits,
                                               s->stream);
     if (e0 != hipSuccess) { TG_SET_ERR("rocprim size query: %s", hipGetErrorName(e0)); return TG_ERR_HIP; }
     void* d_temp = nullptr;
     TG_POOL_ALLOC(s, &d_temp, (int64_t)temp_bytes);
     hipError_t e = rocprim::radix_sort_pairs(d_temp, temp_bytes, d_keys, d_keys2,
-                                             d_vals, d_vals2, (size_t)n, 0, 64,
+                                             d_vals, d_vals2, (size_t)n, 0, bits,
                                              s->stream);
     if (e != hipSuccess) { TG_SET_ERR("rocprim radix_sort_pairs: %s", hipGetErrorName(e)); return TG_ERR_HIP; }
     TG_HIP_CHECK(hipMemcpyAsync(d_keys, d_keys2, n * 8, hipMemcpyDeviceToDevice, s->stream));
@@ -52,6 +54,11 @@ tg_status run_sort_pairs(tg_session* s, uint64_t* d_keys, int64_t* d_vals, int64
     tg_pool_free(s, d_vals2);
     tg_pool_free(s, d_temp);
     return TG_OK;
+}
+
+tg_status run_sort_pairs(tg_session* s, uint64_t* d_keys, int64_t* d_vals, int64_t n)
+{
+    return run_sort_pairs_bits(s, d_keys, d_vals, n, 64);
 }
 
 /* stable ascending sort of (u32 key, i32 value) pairs (join match

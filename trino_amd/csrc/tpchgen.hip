@@ -665,6 +665,8 @@ __global__ void k_var_like(const uint8_t* __restrict__ bytes,
  * per-row byte scan (like_match) measured 26.4 ms for Q13's 150M x ~49 B
  * comments at SF100; see profiles/ for the indexed number. */
 tg_status run_sort_keys_u32(tg_session* s, uint32_t* d_keys, int64_t n);
+tg_status run_sort_pairs_bits(tg_session* s, uint64_t* d_keys, int64_t* d_vals,
+                              int64_t n, int bits);
 
 struct LikeIdx {
     const uint32_t* occ[LIKE_MAX_SEGS];
@@ -704,6 +706,37 @@ __device__ static inline long long lb_u32(const uint32_t* __restrict__ a,
         else hi = m;
     }
     return lo;
+}
+
+/* byte-scan in pool-offset order: rows sorted by slice offset visit the
+ * pool sequentially (neighbouring rows read the same cache lines), then
+ * the verdict scatters back by row id. The offset-random per-row scan
+ * measured 26.4 ms for Q13's 150M x ~49 B slices; the binary-search
+ * occurrence index measured ~51 ms (each 4 B probe drags a 64 B LLC
+ * line); sorting first makes the scan itself cache-resident. */
+__global__ void k_pool_like_sorted(const uint8_t* __restrict__ pool,
+                                   const uint64_t* __restrict__ sorted_offs,
+                                   const int64_t* __restrict__ len_row,
+                                   int64_t n, LikePat p,
+                                   uint8_t* __restrict__ flags)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        int64_t lr = len_row[i];
+        int32_t len = (int32_t)(lr >> 32);
+        uint32_t row = (uint32_t)lr;
+        flags[row] = like_match(pool + sorted_offs[i], len, p) ? 1 : 0;
+    }
+}
+
+__global__ void k_pack_len_row(const int32_t* __restrict__ lens, int64_t n,
+                               int64_t* __restrict__ len_row)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride)
+        len_row[i] = ((int64_t)lens[i] << 32) | (uint32_t)i;
 }
 
 __global__ void k_pool_like_indexed(const int64_t* __restrict__ offs,
@@ -767,9 +800,33 @@ extern "C" tg_status tg_pool_like_flags(tg_session* s, const int64_t* d_offs,
     LikePat p;
     st = parse_like(pattern, &p);
     if (st != TG_OK) return st;
-    const char* ev = getenv("TG_LIKE_IDX");   /* 0 = force byte-scan (A/B) */
-    if (!(ev && ev[0] == '0') && !p.anchor_start && !p.anchor_end &&
-        p.n_segs >= 1 && (n >= (1 << 20) || (ev && ev[0] == '1'))) {
+    /* TG_LIKE_IDX: 0 = per-row byte scan, 1 = occurrence index,
+     * 2 = offset-sorted byte scan (default above 1M rows — A/B in
+     * profiles/: plain 26.4 ms / indexed ~51 / sorted see r02) */
+    const char* ev = getenv("TG_LIKE_IDX");
+    char mode = ev ? ev[0] : ((n >= (1 << 20)) ? '2' : '0');
+    if (mode == '2') {
+        uint64_t* d_so = nullptr;
+        int64_t* d_lr = nullptr;
+        TG_POOL_ALLOC(s, &d_so, n * 8);
+        TG_POOL_ALLOC(s, &d_lr, n * 8);
+        TG_HIP_CHECK(hipMemcpyAsync(d_so, d_offs, n * 8,
+                                    hipMemcpyDeviceToDevice, s->stream));
+        hipLaunchKernelGGL(k_pack_len_row, dim3(tg_grid_for(n)), dim3(TG_BLOCK),
+                           0, s->stream, d_lens, n, d_lr);
+        TG_HIP_CHECK(hipGetLastError());
+        tg_status sst = run_sort_pairs_bits(s, d_so, d_lr, n, 29);
+        if (sst != TG_OK) return sst;
+        hipLaunchKernelGGL(k_pool_like_sorted, dim3(tg_grid_for(n)),
+                           dim3(TG_BLOCK), 0, s->stream, d_pool, d_so, d_lr,
+                           n, p, d_flags);
+        TG_HIP_CHECK(hipGetLastError());
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        tg_pool_free(s, d_so);
+        tg_pool_free(s, d_lr);
+        return TG_OK;
+    }
+    if (mode == '1' && !p.anchor_start && !p.anchor_end && p.n_segs >= 1) {
         /* floating pattern over many pool slices: index the pool once */
         LikeIdx ix{};
         ix.n_segs = p.n_segs;
