@@ -401,8 +401,9 @@ tg_status tg_mark_distinct_create(tg_session*, const int32_t* key_channels,
     int32_t n_key_channels, const int32_t* key_types, tg_operator**);
 
 /* dense-range single-BIGINT-key aggregation (direct array state, one
- * atomic per row; groups emit in key order). For count-shaped aggregations
- * over keys with known dense statistics (e.g. generated custkeys). */
+ * atomic per row — or an atomic pair for the 128-bit exact sum; groups
+ * emit in key order). For COUNT/SUM_I64/SUM_F64_EXACT over keys with
+ * known dense statistics (generated custkeys/suppkeys). */
 tg_status tg_dense_aggregation_create(tg_session*, int32_t key_channel,
     int64_t key_min, int64_t key_max, const tg_agg_spec* agg, tg_operator**);
 

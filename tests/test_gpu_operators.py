@@ -1612,3 +1612,31 @@ class TestDedupSort:
             assert np.array_equal(got, exp)        # sorted ascending
             _lib.tg_device_free(sess._h, d_in)
             _lib.tg_device_free(sess._h, d_out)
+
+
+class TestDenseExactSum:
+    def test_dense_exact_matches_hash_exact(self, sess, ops):
+        """Dense-range SUM_F64_EXACT (2-word direct-array state) is
+        bit-equal to the hash-aggregation exact sum on the same input."""
+        r = rng(97)
+        n = 400_000
+        keys = r.integers(1, 5001, n).astype(np.int64)
+        vals = r.integers(90100, 209900, n) / 100.0    # v*2^43 integer
+        d = ops.dense_aggregation(sess, 0, 1, 5000,
+                                  (ops.AGG_SUM_F64_EXACT, 1, 43))
+        d.add_input(ops.page_from_numpy([keys, vals]))
+        dout = d.drain()[0]
+        d.close()
+        h = ops.hash_aggregation(sess, [0], [ops.TG_BIGINT],
+                                 [(ops.AGG_SUM_F64_EXACT, 1, 43)])
+        h.add_input(ops.page_from_numpy([keys, vals]))
+        hout = h.drain()[0]
+        h.close()
+        dk = np.asarray(dout[0]["values"])
+        dv = np.asarray(dout[1]["values"])
+        hk = np.asarray(hout[0]["values"])
+        hv = np.asarray(hout[1]["values"])
+        assert len(dk) == len(hk)
+        order = np.argsort(hk, kind="stable")
+        assert np.array_equal(dk, hk[order])           # key-ascending emit
+        assert np.array_equal(dv, hv[order])           # bit-equal sums

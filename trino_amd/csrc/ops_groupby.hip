@@ -1419,7 +1419,7 @@ extern "C" tg_status tg_hash_aggregation_create(tg_session* s,
 
 __global__ void k_dense_update(const KColH key, int64_t n, int64_t key_min,
                                int64_t range, long long* __restrict__ state,
-                               const KColH val, int fn)
+                               const KColH val, int fn, double scale)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -1427,6 +1427,21 @@ __global__ void k_dense_update(const KColH key, int64_t n, int64_t key_min,
         if (kcol_is_null(key, i)) continue;
         int64_t k = (int64_t)kcol_word(key, i) - key_min;
         if (k < 0 || k >= range) continue;   /* guarded (caller promises) */
+        if (fn == TG_AGG_SUM_F64_EXACT) {
+            /* 128-bit fixed-point direct-array state (2 words per key):
+             * same carry-propagating atomics as the hash/streaming paths */
+            if (kcol_is_null(val, i)) continue;
+            double y = ((const double*)val.data)[i] * scale;
+            __int128 yi = (__int128)(long long)y;
+            unsigned long long lov = (unsigned long long)(unsigned __int128)yi;
+            unsigned long long hiv =
+                (unsigned long long)((unsigned __int128)yi >> 64);
+            unsigned long long old =
+                atomicAdd((unsigned long long*)&state[2 * k], lov);
+            unsigned long long carry = (old + lov) < lov ? 1ull : 0ull;
+            atomicAdd((unsigned long long*)&state[2 * k + 1], hiv + carry);
+            continue;
+        }
         long long add = 1;
         if (fn == TG_AGG_SUM_I64 || fn == TG_AGG_COUNT_COL)
             add = (fn == TG_AGG_COUNT_COL)
@@ -1468,14 +1483,16 @@ __global__ void k_dense_update_lds(const KColH key, int64_t n, int64_t key_min,
 
 __global__ void k_dense_present(const long long* __restrict__ state, int64_t range,
                                 int32_t* __restrict__ chunk_counts, int64_t nchunks,
-                                int64_t chunk)
+                                int64_t chunk, int words)
 {
     int64_t c = (int64_t)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
     if (c >= nchunks) return;
     int lane = threadIdx.x % 64;
     int64_t lo = c * chunk, hi = min(lo + chunk, range);
     int32_t cnt = 0;
-    for (int64_t i = lo + lane; i < hi; i += 64) cnt += state[i] != 0;
+    for (int64_t i = lo + lane; i < hi; i += 64)
+        cnt += (state[i * words] != 0 ||
+                (words == 2 && state[i * words + 1] != 0));
     #pragma unroll
     for (int off = 32; off >= 1; off >>= 1) cnt += __shfl_xor(cnt, off, 64);
     if (lane == 0) chunk_counts[c] = cnt;
@@ -1506,11 +1523,45 @@ __global__ void k_dense_emit(const long long* __restrict__ state, int64_t range,
     }
 }
 
+/* exact-sum variant: 2-word state, emits the correctly-rounded double */
+__global__ void k_dense_emit_exact(const long long* __restrict__ state,
+                                   int64_t range, int64_t key_min,
+                                   const int32_t* __restrict__ offs,
+                                   int64_t nchunks, int64_t chunk,
+                                   double inv_scale,
+                                   int64_t* __restrict__ out_keys,
+                                   double* __restrict__ out_vals)
+{
+    int64_t c = blockIdx.x;
+    if (c >= nchunks) return;
+    if (threadIdx.x >= 64) return;
+    int lane = threadIdx.x;
+    int64_t lo = c * chunk, hi = min(lo + chunk, range);
+    int32_t run = offs[c];
+    for (int64_t g = lo; g < hi; g += 64) {
+        int64_t i = g + lane;
+        bool p = (i < hi) && (state[2 * i] != 0 || state[2 * i + 1] != 0);
+        unsigned long long b = __ballot(p);
+        int before = __popcll(b & ((1ull << lane) - 1ull));
+        if (p) {
+            unsigned __int128 v =
+                ((unsigned __int128)(unsigned long long)state[2 * i + 1] << 64) |
+                (unsigned long long)state[2 * i];
+            __int128 sv = (__int128)v;
+            out_keys[run + before] = key_min + i;
+            out_vals[run + before] = (double)sv * inv_scale;
+        }
+        run += __popcll(b);
+    }
+}
+
 struct DenseAggOp : tg_operator {
     int32_t key_channel = 0;
     int64_t key_min = 0, range = 0;
     tg_agg_spec agg{};
     long long* state = nullptr;
+    double scale = 1.0;     /* SUM_F64_EXACT fixed-point scale (2^scale_pow) */
+    int words = 1;          /* state words per key (2 for SUM_F64_EXACT) */
     bool emitted = false;
 
     tg_status add_input(const tg_page* page) override
@@ -1526,7 +1577,7 @@ struct DenseAggOp : tg_operator {
                                hipMemcpyDeviceToHost));
         KColH kc = h[key_channel];
         KColH vc = (agg.input_channel >= 0) ? h[agg.input_channel] : kc;
-        if (range <= 8192) {
+        if (range <= 8192 && words == 1) {
             hipLaunchKernelGGL(k_dense_update_lds, dim3(tg_grid_for(in.n)),
                                dim3(TG_BLOCK), range * 8, s->stream, kc, in.n,
                                key_min, range, state, vc, agg.fn);
@@ -1534,7 +1585,7 @@ struct DenseAggOp : tg_operator {
         else {
             hipLaunchKernelGGL(k_dense_update, dim3(tg_grid_for(in.n)),
                                dim3(TG_BLOCK), 0, s->stream, kc, in.n,
-                               key_min, range, state, vc, agg.fn);
+                               key_min, range, state, vc, agg.fn, scale);
         }
         TG_HIP_CHECK(hipGetLastError());
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
@@ -1555,7 +1606,7 @@ struct DenseAggOp : tg_operator {
         hipLaunchKernelGGL(k_dense_present,
                            dim3((uint32_t)((nchunks + wpb - 1) / wpb)),
                            dim3(TG_BLOCK), 0, s->stream, state, range, d_offs,
-                           nchunks, CH);
+                           nchunks, CH, words);
         TG_HIP_CHECK(hipGetLastError());
         tg_status st = run_scan_i32(s, d_offs, nchunks, d_total);
         if (st != TG_OK) return st;
@@ -1568,11 +1619,21 @@ struct DenseAggOp : tg_operator {
         bk.type = TG_BIGINT; bk.n = total;
         TG_POOL_ALLOC(s, &bk.data, (int64_t)(total ? total : 1) * 8);
         DevBlock& bv = outp.blocks[1];
-        bv.type = TG_BIGINT; bv.n = total;
+        bv.type = (agg.fn == TG_AGG_SUM_F64_EXACT) ? TG_DOUBLE : TG_BIGINT;
+        bv.n = total;
         TG_POOL_ALLOC(s, &bv.data, (int64_t)(total ? total : 1) * 8);
-        hipLaunchKernelGGL(k_dense_emit, dim3((uint32_t)nchunks), dim3(64), 0,
-                           s->stream, state, range, key_min, d_offs, nchunks,
-                           CH, (int64_t*)bk.data, (int64_t*)bv.data);
+        if (agg.fn == TG_AGG_SUM_F64_EXACT) {
+            hipLaunchKernelGGL(k_dense_emit_exact, dim3((uint32_t)nchunks),
+                               dim3(64), 0, s->stream, state, range, key_min,
+                               d_offs, nchunks, CH, 1.0 / scale,
+                               (int64_t*)bk.data, (double*)bv.data);
+        }
+        else {
+            hipLaunchKernelGGL(k_dense_emit, dim3((uint32_t)nchunks), dim3(64),
+                               0, s->stream, state, range, key_min, d_offs,
+                               nchunks, CH, (int64_t*)bk.data,
+                               (int64_t*)bv.data);
+        }
         TG_HIP_CHECK(hipGetLastError());
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
         tg_pool_free(s, d_offs);
@@ -1609,8 +1670,8 @@ extern "C" tg_status tg_dense_aggregation_create(tg_session* s,
         return TG_ERR_INVALID_ARG;
     }
     if (agg->fn != TG_AGG_COUNT_STAR && agg->fn != TG_AGG_COUNT_COL &&
-        agg->fn != TG_AGG_SUM_I64) {
-        TG_SET_ERR("dense aggregation supports COUNT/SUM_I64");
+        agg->fn != TG_AGG_SUM_I64 && agg->fn != TG_AGG_SUM_F64_EXACT) {
+        TG_SET_ERR("dense aggregation supports COUNT/SUM_I64/SUM_F64_EXACT");
         return TG_ERR_UNSUPPORTED;
     }
     auto* op = new DenseAggOp();
@@ -1619,11 +1680,15 @@ extern "C" tg_status tg_dense_aggregation_create(tg_session* s,
     op->key_min = key_min;
     op->range = key_max - key_min + 1;
     op->agg = *agg;
-    if (tg_pool_alloc(s, (void**)&op->state, op->range * 8) != TG_OK) {
+    if (agg->fn == TG_AGG_SUM_F64_EXACT) {
+        op->words = 2;
+        for (int32_t sp_ = 0; sp_ < agg->scale_pow; sp_++) op->scale *= 2.0;
+    }
+    if (tg_pool_alloc(s, (void**)&op->state, op->range * 8 * op->words) != TG_OK) {
         delete op;
         return TG_ERR_OOM;
     }
-    (void)hipMemsetAsync(op->state, 0, op->range * 8, s->stream);
+    (void)hipMemsetAsync(op->state, 0, op->range * 8 * op->words, s->stream);
     (void)hipStreamSynchronize(s->stream);
     *out = op;
     return TG_OK;

@@ -610,12 +610,12 @@ _lib.tg_dense_aggregation_create.argtypes = [ctypes.c_void_p, ctypes.c_int32,
 
 def dense_aggregation(session, key_channel, key_min, key_max, agg):
     """Direct-array aggregation over a dense BIGINT key range; one
-    (fn, input_channel) agg; output (key, value) in key order."""
+    (fn, input_channel[, scale_pow]) agg; output (key, value) in key order."""
     h = ctypes.c_void_p()
     sp = TgAggSpec()
     sp.fn = agg[0]
     sp.input_channel = agg[1]
-    sp.scale_pow = 0
+    sp.scale_pow = agg[2] if len(agg) > 2 else 0
     sp.mask_gt_a = -1
     sp.mask_gt_b = -1
     _check(_lib.tg_dense_aggregation_create(session._h, key_channel, key_min,

@@ -698,8 +698,12 @@ def q15_gpu(session, sf, order_start=1, order_count=None):
     f.add_input(lpage)
     f.finish()
     sel = _take_device_page(session, f)
-    agg = ops.hash_aggregation(session, [0], [ops.TG_BIGINT],
-                               [(ops.AGG_SUM_F64_EXACT, 1, 43)])
+    # suppkeys are dense 1..10k*SF: direct-array exact sums (the ~1M-group
+    # hash table sized its keystore for the 22.7M incoming rows — ~900 MB
+    # of allocations that cost ~58 ms whenever the pool was cold, which in
+    # the 22-query sweep was every step)
+    agg = ops.dense_aggregation(session, 0, 1, int(10_000 * sf),
+                                (ops.AGG_SUM_F64_EXACT, 1, 43))
     agg.add_input(sel)
     pages = agg.drain()
     elapsed = time.time() - t0
