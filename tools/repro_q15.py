@@ -50,15 +50,26 @@ def main(sf=100.0):
         fm, am, nsel, ng = timed_q15(s, sf)
         print(f"   q15: filter {fm:7.2f} ms  agg {am:7.2f} ms  "
               f"(sel={nsel} groups={ng})", flush=True)
-    for name in ("q12", "q13", "q14"):
+    names = [n for n in ("q2", "q3", "q4", "q5", "q7", "q8", "q6", "q9",
+                         "q10", "q11", "q12", "q13", "q14", "q16", "q17",
+                         "q18", "q19", "q20", "q21", "q22")]
+    for name in names:
         t0 = time.time()
         getattr(q, f"{name}_gpu")(s, sf)
-        print(f"-- {name}: {(time.time()-t0)*1e3:.1f} ms", flush=True)
-    print("-- after q12..q14, q15 x3:")
+        print(f"-- {name}: {(time.time()-t0)*1e3:7.1f} ms   pool {_mem(s)}",
+              flush=True)
+    print("-- after the full cycle, q15 x3:")
     for i in range(3):
         fm, am, nsel, ng = timed_q15(s, sf)
-        print(f"   q15: filter {fm:7.2f} ms  agg {am:7.2f} ms", flush=True)
+        print(f"   q15: filter {fm:7.2f} ms  agg {am:7.2f} ms  pool {_mem(s)}",
+              flush=True)
     s.close()
+
+
+def _mem(s):
+    import trino_amd
+    tot, cach = trino_amd.session_memory(s)
+    return f"total {tot/2**30:.1f}G cached {cach/2**30:.1f}G"
 
 
 if __name__ == "__main__":
