@@ -33,6 +33,11 @@ extern "C" tg_status tg_session_create(int device_ordinal, tg_session** out)
     TG_HIP_CHECK(hipSetDevice(device_ordinal));
     tg_session* s = new tg_session();
     s->device = device_ordinal;
+    {
+        size_t freeb = 0, totalb = 0;
+        if (hipMemGetInfo(&freeb, &totalb) == hipSuccess && totalb)
+            s->mem_cap = (size_t)(totalb * 0.92);
+    }
     if (hipStreamCreate(&s->stream) != hipSuccess ||
         hipEventCreate(&s->ev_start) != hipSuccess ||
         hipEventCreate(&s->ev_stop) != hipSuccess) {
@@ -42,6 +47,23 @@ extern "C" tg_status tg_session_create(int device_ordinal, tg_session** out)
     }
     *out = s;
     return TG_OK;
+}
+
+/* evict the LARGEST cached buffers until `need` more bytes fit under the
+ * soft cap. Largest-first keeps the many small latency-sensitive buffers
+ * (flag vectors, offsets, states) warm; the previous drop-the-whole-cache
+ * fallback fired at the same point of every 22-query sweep cycle and the
+ * next query (q15) re-paid tens of ms of fresh hipMallocs every step. */
+static void pool_evict_for(tg_session* s, size_t need)
+{
+    while (!s->pool_free.empty() && s->mem_cap &&
+           s->pool_bytes + need > s->mem_cap) {
+        auto last = std::prev(s->pool_free.end());
+        (void)hipFree(last->second);
+        s->pool_bytes -= last->first;
+        s->pool_sizes.erase(last->second);
+        s->pool_free.erase(last);
+    }
 }
 
 tg_status tg_pool_alloc(tg_session* s, void** out, size_t bytes)
@@ -54,12 +76,16 @@ tg_status tg_pool_alloc(tg_session* s, void** out, size_t bytes)
         s->pool_free.erase(it);
         return TG_OK;
     }
+    pool_evict_for(s, bytes);
     hipError_t e = hipMalloc(out, bytes);
-    if (e == hipErrorOutOfMemory) {
-        /* drop the cache and retry once */
-        for (auto& kv : s->pool_free) (void)hipFree(kv.second);
-        for (auto& kv : s->pool_free) s->pool_sizes.erase(kv.second);
-        s->pool_free.clear();
+    while (e == hipErrorOutOfMemory && !s->pool_free.empty()) {
+        /* cap estimate was optimistic: evict largest-first and retry */
+        (void)hipGetLastError();
+        auto last = std::prev(s->pool_free.end());
+        (void)hipFree(last->second);
+        s->pool_bytes -= last->first;
+        s->pool_sizes.erase(last->second);
+        s->pool_free.erase(last);
         e = hipMalloc(out, bytes);
     }
     if (e != hipSuccess) {

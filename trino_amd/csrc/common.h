@@ -35,7 +35,8 @@ struct tg_session {
      * Single-threaded per session (one driver thread per operator chain). */
     std::multimap<size_t, void*> pool_free;   /* size -> buffer */
     std::map<void*, size_t> pool_sizes;       /* live + cached buffer sizes */
-    size_t pool_bytes = 0;
+    size_t pool_bytes = 0;                    /* live + cached total */
+    size_t mem_cap = 0;                       /* soft cap (92% of VRAM) */
 };
 
 tg_status tg_pool_alloc(tg_session* s, void** out, size_t bytes);
