@@ -18,6 +18,7 @@ probe lineitem -> hash agg -> TopN). TopN is §8(f) next-row: the final
 ORDER BY ... LIMIT 10 here runs on the grouped output (tiny) on the host.
 """
 import ctypes
+import os
 import time
 
 import numpy as np
@@ -683,6 +684,7 @@ def q15_gpu(session, sf, order_start=1, order_count=None):
     if order_count is None:
         order_count = int(1_500_000 * sf)
     li = session.tpch_lineitem(sf, order_start, order_count, with_suppkey=True)
+    trace = os.environ.get("TG_Q15_TRACE")
     t0 = time.time()
     lpage = ops.page_from_device(session, ([(li.suppkey, ops.TG_BIGINT),
                                             (li.shipdate, ops.TG_INTEGER),
@@ -698,15 +700,31 @@ def q15_gpu(session, sf, order_start=1, order_count=None):
     f.add_input(lpage)
     f.finish()
     sel = _take_device_page(session, f)
+    if trace:
+        import sys
+        print(f"[q15] filter {(time.time()-t0)*1e3:.1f} ms", file=sys.stderr,
+              flush=True)
     # suppkeys are dense 1..10k*SF: direct-array exact sums (the ~1M-group
     # hash table sized its keystore for the 22.7M incoming rows — ~900 MB
     # of allocations that cost ~58 ms whenever the pool was cold, which in
     # the 22-query sweep was every step)
     agg = ops.dense_aggregation(session, 0, 1, int(10_000 * sf),
                                 (ops.AGG_SUM_F64_EXACT, 1, 43))
+    if trace:
+        import sys
+        tc = time.time()
+        print(f"[q15] agg create {(tc-t0)*1e3:.1f} ms cum", file=sys.stderr,
+              flush=True)
     agg.add_input(sel)
+    if trace:
+        tu = time.time()
+        print(f"[q15] agg update {(tu-t0)*1e3:.1f} ms cum", file=sys.stderr,
+              flush=True)
     pages = agg.drain()
     elapsed = time.time() - t0
+    if trace:
+        print(f"[q15] drain {(elapsed)*1e3:.1f} ms cum", file=sys.stderr,
+              flush=True)
     f.close()
     agg.close()
     session.tpch_lineitem_free(li)
