@@ -720,11 +720,19 @@ def q15_gpu(session, sf, order_start=1, order_count=None):
         tu = time.time()
         print(f"[q15] agg update {(tu-t0)*1e3:.1f} ms cum", file=sys.stderr,
               flush=True)
-    pages = agg.drain()
+        agg.finish()
+        print(f"[q15] finish {(time.time()-t0)*1e3:.1f} ms cum",
+              file=sys.stderr, flush=True)
+        devp = _take_device_page(session, agg)   # triggers emit, no download
+        print(f"[q15] emit {(time.time()-t0)*1e3:.1f} ms cum "
+              f"(groups={devp.position_count})", file=sys.stderr, flush=True)
+        from .ops import _download_page
+        pages = [_download_page(session, devp)]
+        print(f"[q15] download {(time.time()-t0)*1e3:.1f} ms cum",
+              file=sys.stderr, flush=True)
+    else:
+        pages = agg.drain()
     elapsed = time.time() - t0
-    if trace:
-        print(f"[q15] drain {(elapsed)*1e3:.1f} ms cum", file=sys.stderr,
-              flush=True)
     f.close()
     agg.close()
     session.tpch_lineitem_free(li)
