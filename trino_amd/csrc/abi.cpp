@@ -116,6 +116,7 @@ void tg_pool_free(tg_session* s, void* p)
 extern "C" void tg_session_close(tg_session* s)
 {
     if (!s) return;
+    if (s->pin_buf) (void)hipHostFree(s->pin_buf);
     for (auto& kv : s->pool_sizes) (void)hipFree(kv.first);
     (void)hipEventDestroy(s->ev_start);
     (void)hipEventDestroy(s->ev_stop);
@@ -123,11 +124,39 @@ extern "C" void tg_session_close(tg_session* s)
     delete s;
 }
 
-/* test helper: copy device memory to host (generator parity tests) */
+/* device-to-host copy staged through a cached pinned buffer: a direct
+ * DtoH into pageable numpy memory measured ~0.3 GB/s mid-sweep (50 ms for
+ * q15's 16 MB result download, HIP-trace-attributed); DMA into pinned +
+ * host memcpy is deterministic regardless of the destination's paging. */
 extern "C" tg_status tg_copy_dtoh(tg_session* s, void* dst, const void* src, int64_t bytes)
 {
-    TG_HIP_CHECK(hipMemcpyAsync(dst, src, (size_t)bytes, hipMemcpyDeviceToHost, s->stream));
-    TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+    constexpr size_t PIN_MAX = 64ull << 20;
+    size_t want = (size_t)bytes < PIN_MAX ? (size_t)bytes : PIN_MAX;
+    if (s->pin_cap < want) {
+        if (s->pin_buf) (void)hipHostFree(s->pin_buf);
+        s->pin_buf = nullptr;
+        s->pin_cap = 0;
+        size_t cap = 1 << 20;
+        while (cap < want) cap <<= 1;
+        if (hipHostMalloc(&s->pin_buf, cap) == hipSuccess) s->pin_cap = cap;
+        else (void)hipGetLastError();
+    }
+    if (!s->pin_buf) {   /* pinned alloc failed: direct copy fallback */
+        TG_HIP_CHECK(hipMemcpyAsync(dst, src, (size_t)bytes,
+                                    hipMemcpyDeviceToHost, s->stream));
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        return TG_OK;
+    }
+    size_t off = 0;
+    while (off < (size_t)bytes) {
+        size_t chunk = (size_t)bytes - off;
+        if (chunk > s->pin_cap) chunk = s->pin_cap;
+        TG_HIP_CHECK(hipMemcpyAsync(s->pin_buf, (const char*)src + off, chunk,
+                                    hipMemcpyDeviceToHost, s->stream));
+        TG_HIP_CHECK(hipStreamSynchronize(s->stream));
+        memcpy((char*)dst + off, s->pin_buf, chunk);
+        off += chunk;
+    }
     return TG_OK;
 }
 

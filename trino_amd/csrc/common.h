@@ -37,6 +37,8 @@ struct tg_session {
     std::map<void*, size_t> pool_sizes;       /* live + cached buffer sizes */
     size_t pool_bytes = 0;                    /* live + cached total */
     size_t mem_cap = 0;                       /* soft cap (92% of VRAM) */
+    void* pin_buf = nullptr;                  /* pinned DtoH staging buffer */
+    size_t pin_cap = 0;
 };
 
 tg_status tg_pool_alloc(tg_session* s, void** out, size_t bytes);
