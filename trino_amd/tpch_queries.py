@@ -726,8 +726,21 @@ def q15_gpu(session, sf, order_start=1, order_count=None):
         devp = _take_device_page(session, agg)   # triggers emit, no download
         print(f"[q15] emit {(time.time()-t0)*1e3:.1f} ms cum "
               f"(groups={devp.position_count})", file=sys.stderr, flush=True)
-        from .ops import _download_page
-        pages = [_download_page(session, devp)]
+        ng = devp.position_count
+        ta = time.time()
+        a0 = np.empty(ng, np.int64)
+        a1 = np.empty(ng, np.float64)
+        tb = time.time()
+        copy_dtoh(session, a0, devp.blocks[0].data)
+        tc = time.time()
+        copy_dtoh(session, a1, devp.blocks[1].data)
+        td = time.time()
+        copy_dtoh(session, a0, devp.blocks[0].data)   # repeat, warm
+        te = time.time()
+        print(f"[q15] npalloc {(tb-ta)*1e3:.1f} copy0 {(tc-tb)*1e3:.1f} "
+              f"copy1 {(td-tc)*1e3:.1f} copy0again {(te-td)*1e3:.1f} ms",
+              file=sys.stderr, flush=True)
+        pages = [[{"values": a0}, {"values": a1}]]
         print(f"[q15] download {(time.time()-t0)*1e3:.1f} ms cum",
               file=sys.stderr, flush=True)
     else:
