@@ -124,10 +124,22 @@ extern "C" void tg_session_close(tg_session* s)
     delete s;
 }
 
-/* device-to-host copy staged through a cached pinned buffer: a direct
- * DtoH into pageable numpy memory measured ~0.3 GB/s mid-sweep (50 ms for
- * q15's 16 MB result download, HIP-trace-attributed); DMA into pinned +
- * host memcpy is deterministic regardless of the destination's paging. */
+/* device-to-host copy by COMPUTE KERNEL into device-mapped pinned memory
+ * (then host memcpy). The SDMA copy engine pays a ~20-50 ms wake-up for
+ * the FIRST DtoH after a long kernel-only stretch (q15's 8 MB result
+ * download measured 23.9 ms mid-sweep, the repeat 0.4 ms — per-copy
+ * trace); the compute path stays warm because the query just used it. */
+__global__ void k_copy_out(const uint8_t* __restrict__ src,
+                           uint8_t* __restrict__ dst, size_t n)
+{
+    size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+    size_t stride = (size_t)gridDim.x * blockDim.x;
+    size_t nv = n / 16;
+    for (size_t k = i; k < nv; k += stride)
+        ((ulonglong2*)dst)[k] = ((const ulonglong2*)src)[k];
+    for (size_t k = nv * 16 + i; k < n; k += stride) dst[k] = src[k];
+}
+
 extern "C" tg_status tg_copy_dtoh(tg_session* s, void* dst, const void* src, int64_t bytes)
 {
     constexpr size_t PIN_MAX = 64ull << 20;
@@ -138,10 +150,17 @@ extern "C" tg_status tg_copy_dtoh(tg_session* s, void* dst, const void* src, int
         s->pin_cap = 0;
         size_t cap = 1 << 20;
         while (cap < want) cap <<= 1;
-        if (hipHostMalloc(&s->pin_buf, cap) == hipSuccess) s->pin_cap = cap;
+        if (hipHostMalloc(&s->pin_buf, cap, hipHostMallocMapped) == hipSuccess)
+            s->pin_cap = cap;
         else (void)hipGetLastError();
     }
-    if (!s->pin_buf) {   /* pinned alloc failed: direct copy fallback */
+    void* pin_dev = nullptr;
+    if (s->pin_buf &&
+        hipHostGetDevicePointer(&pin_dev, s->pin_buf, 0) != hipSuccess) {
+        (void)hipGetLastError();
+        pin_dev = nullptr;
+    }
+    if (!s->pin_buf || !pin_dev) {   /* fallback: plain DtoH */
         TG_HIP_CHECK(hipMemcpyAsync(dst, src, (size_t)bytes,
                                     hipMemcpyDeviceToHost, s->stream));
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
@@ -151,8 +170,12 @@ extern "C" tg_status tg_copy_dtoh(tg_session* s, void* dst, const void* src, int
     while (off < (size_t)bytes) {
         size_t chunk = (size_t)bytes - off;
         if (chunk > s->pin_cap) chunk = s->pin_cap;
-        TG_HIP_CHECK(hipMemcpyAsync(s->pin_buf, (const char*)src + off, chunk,
-                                    hipMemcpyDeviceToHost, s->stream));
+        int grid = (int)((chunk / 16 + 255) / 256);
+        if (grid < 1) grid = 1;
+        if (grid > 4096) grid = 4096;
+        hipLaunchKernelGGL(k_copy_out, dim3(grid), dim3(256), 0, s->stream,
+                           (const uint8_t*)src + off, (uint8_t*)pin_dev, chunk);
+        TG_HIP_CHECK(hipGetLastError());
         TG_HIP_CHECK(hipStreamSynchronize(s->stream));
         memcpy((char*)dst + off, s->pin_buf, chunk);
         off += chunk;
