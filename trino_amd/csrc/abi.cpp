@@ -154,7 +154,7 @@ __global__ void k_copy_out(const uint8_t* __restrict__ src,
  * overlaps the query's own kernels instead of its result download.
  * Rate-limited to one launch per 50 ms; writes its OWN 64 B pinned sink
  * (pin_buf may be mid-download on the main stream). */
-extern "C" void tg_host_link_warm(tg_session* s)
+void tg_host_link_warm(tg_session* s)
 {
     auto now = std::chrono::duration_cast<std::chrono::milliseconds>(
                    std::chrono::steady_clock::now().time_since_epoch())
