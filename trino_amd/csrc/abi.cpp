@@ -3,7 +3,6 @@
  * device and every operator entry point requires a session.
  */
 #include "common.h"
-#include <chrono>
 
 thread_local std::string tg_error_buf;
 
@@ -117,12 +116,6 @@ void tg_pool_free(tg_session* s, void* p)
 extern "C" void tg_session_close(tg_session* s)
 {
     if (!s) return;
-    if (s->warm_stream) {
-        (void)hipStreamSynchronize(s->warm_stream);
-        (void)hipStreamDestroy(s->warm_stream);
-    }
-    if (s->warm_src) (void)hipFree(s->warm_src);
-    if (s->warm_pin) (void)hipHostFree(s->warm_pin);
     if (s->pin_buf) (void)hipHostFree(s->pin_buf);
     for (auto& kv : s->pool_sizes) (void)hipFree(kv.first);
     (void)hipEventDestroy(s->ev_start);
@@ -145,48 +138,6 @@ __global__ void k_copy_out(const uint8_t* __restrict__ src,
     for (size_t k = i; k < nv; k += stride)
         ((ulonglong2*)dst)[k] = ((const ulonglong2*)src)[k];
     for (size_t k = nv * 16 + i; k < n; k += stride) dst[k] = src[k];
-}
-
-/* fire-and-forget device->pinned-host write on a side stream: the first
- * host transfer after a long kernel-only stretch pays an ~18 ms path
- * wake-up (measured; the repeat costs 0.4 ms). Called from
- * tg_upload_page, i.e. early in every operator pipeline, so the wake-up
- * overlaps the query's own kernels instead of its result download.
- * Rate-limited to one launch per 50 ms; writes its OWN 64 B pinned sink
- * (pin_buf may be mid-download on the main stream). */
-void tg_host_link_warm(tg_session* s)
-{
-    auto now = std::chrono::duration_cast<std::chrono::milliseconds>(
-                   std::chrono::steady_clock::now().time_since_epoch())
-                   .count();
-    if (s->warm_last_ms && now - s->warm_last_ms < 50) return;
-    s->warm_last_ms = now;
-    if (!s->warm_pin &&
-        hipHostMalloc(&s->warm_pin, 64, hipHostMallocMapped) != hipSuccess) {
-        (void)hipGetLastError();
-        s->warm_pin = nullptr;
-        return;
-    }
-    void* pin_dev = nullptr;
-    if (hipHostGetDevicePointer(&pin_dev, s->warm_pin, 0) != hipSuccess) {
-        (void)hipGetLastError();
-        return;
-    }
-    if (!s->warm_stream &&
-        hipStreamCreateWithFlags(&s->warm_stream, hipStreamNonBlocking) !=
-            hipSuccess) {
-        (void)hipGetLastError();
-        s->warm_stream = nullptr;
-        return;
-    }
-    if (!s->warm_src && hipMalloc(&s->warm_src, 16) != hipSuccess) {
-        (void)hipGetLastError();
-        s->warm_src = nullptr;
-        return;
-    }
-    hipLaunchKernelGGL(k_copy_out, dim3(1), dim3(64), 0, s->warm_stream,
-                       (const uint8_t*)s->warm_src, (uint8_t*)pin_dev, 16);
-    (void)hipGetLastError();
 }
 
 extern "C" tg_status tg_copy_dtoh(tg_session* s, void* dst, const void* src, int64_t bytes)

@@ -39,14 +39,7 @@ struct tg_session {
     size_t mem_cap = 0;                       /* soft cap (92% of VRAM) */
     void* pin_buf = nullptr;                  /* pinned DtoH staging buffer */
     size_t pin_cap = 0;
-    hipStream_t warm_stream = nullptr;        /* async host-link warm-up */
-    void* warm_src = nullptr;                 /* 16 B device source */
-    void* warm_pin = nullptr;                 /* 64 B pinned sink (own buf:
-                                                 pin_buf may be mid-copy) */
-    long long warm_last_ms = 0;
 };
-
-void tg_host_link_warm(tg_session* s);
 
 tg_status tg_pool_alloc(tg_session* s, void** out, size_t bytes);
 void tg_pool_free(tg_session* s, void* p);

@@ -67,7 +67,6 @@ static tg_status decode_to_flat(tg_session* s, const tg_block* b, DevBlock* out)
 
 tg_status tg_upload_page(tg_session* s, const tg_page* in, DevPage* out)
 {
-    tg_host_link_warm(s);   /* overlap the DtoH path wake-up (abi.cpp) */
     out->n = in->position_count;
     out->blocks.resize(in->channel_count);
     for (int c = 0; c < in->channel_count; c++) {
