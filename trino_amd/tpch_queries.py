@@ -710,42 +710,16 @@ def q15_gpu(session, sf, order_start=1, order_count=None):
     # the 22-query sweep was every step)
     agg = ops.dense_aggregation(session, 0, 1, int(10_000 * sf),
                                 (ops.AGG_SUM_F64_EXACT, 1, 43))
-    if trace:
-        import sys
-        tc = time.time()
-        print(f"[q15] agg create {(tc-t0)*1e3:.1f} ms cum", file=sys.stderr,
-              flush=True)
     agg.add_input(sel)
     if trace:
-        tu = time.time()
-        print(f"[q15] agg update {(tu-t0)*1e3:.1f} ms cum", file=sys.stderr,
-              flush=True)
-        agg.finish()
-        print(f"[q15] finish {(time.time()-t0)*1e3:.1f} ms cum",
+        import sys
+        print(f"[q15] agg update {(time.time()-t0)*1e3:.1f} ms cum",
               file=sys.stderr, flush=True)
-        devp = _take_device_page(session, agg)   # triggers emit, no download
-        print(f"[q15] emit {(time.time()-t0)*1e3:.1f} ms cum "
-              f"(groups={devp.position_count})", file=sys.stderr, flush=True)
-        ng = devp.position_count
-        ta = time.time()
-        a0 = np.empty(ng, np.int64)
-        a1 = np.empty(ng, np.float64)
-        tb = time.time()
-        copy_dtoh(session, a0, devp.blocks[0].data)
-        tc = time.time()
-        copy_dtoh(session, a1, devp.blocks[1].data)
-        td = time.time()
-        copy_dtoh(session, a0, devp.blocks[0].data)   # repeat, warm
-        te = time.time()
-        print(f"[q15] npalloc {(tb-ta)*1e3:.1f} copy0 {(tc-tb)*1e3:.1f} "
-              f"copy1 {(td-tc)*1e3:.1f} copy0again {(te-td)*1e3:.1f} ms",
-              file=sys.stderr, flush=True)
-        pages = [[{"values": a0}, {"values": a1}]]
-        print(f"[q15] download {(time.time()-t0)*1e3:.1f} ms cum",
-              file=sys.stderr, flush=True)
-    else:
-        pages = agg.drain()
+    pages = agg.drain()
     elapsed = time.time() - t0
+    if trace:
+        print(f"[q15] drain {elapsed*1e3:.1f} ms cum", file=sys.stderr,
+              flush=True)
     f.close()
     agg.close()
     session.tpch_lineitem_free(li)
