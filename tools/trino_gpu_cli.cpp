@@ -17,6 +17,7 @@
 #include <cstdlib>
 #include <cstring>
 #include <algorithm>
+#include <map>
 #include <string>
 #include <vector>
 
@@ -734,10 +735,185 @@ static int run_q13(tg_session* s, double sf)
     return 0;
 }
 
+static int run_q16(tg_session* s, double sf)
+{
+    /* Q16 (python mirror: q16_gpu): part/partsupp device generators, the
+     * BBB '%Customer%Complaints%' overlay via the generic VARCHAR LIKE
+     * kernel, semi-join exclusion, and count(DISTINCT ps_suppkey) per
+     * (brand, type, size) through the SORTED dedup (`tg_dedup_i64`) —
+     * driving the newest round-2 ABI from pure C++. */
+    int64_t np_ = (int64_t)(200000 * sf), ns = (int64_t)(10000 * sf);
+    void *p_pk, *p_ty, *p_br, *p_sz, *ps_pk, *ps_sk;
+    die(tg_device_malloc(s, &p_pk, np_ * 8), "malloc");
+    die(tg_device_malloc(s, &p_ty, np_ * 2), "malloc");
+    die(tg_device_malloc(s, &p_br, np_), "malloc");
+    die(tg_device_malloc(s, &p_sz, np_ * 4), "malloc");
+    die(tg_device_malloc(s, &ps_pk, np_ * 4 * 8), "malloc");
+    die(tg_device_malloc(s, &ps_sk, np_ * 4 * 8), "malloc");
+    die(tg_tpch_gen_part2(s, sf, 1, np_, (int64_t*)p_pk, (int16_t*)p_ty,
+                          (uint8_t*)p_br, (int32_t*)p_sz, nullptr, nullptr,
+                          nullptr), "gen part");
+    die(tg_tpch_gen_partsupp(s, sf, 1, np_, (int64_t*)ps_pk, (int64_t*)ps_sk,
+                             nullptr, nullptr), "gen ps");
+    /* complaint suppliers (BBB overlay) */
+    int32_t* d_soff = nullptr;
+    uint8_t* d_sbytes = nullptr;
+    die(tg_tpch_gen_supplier_comments(s, sf, 1, ns, &d_soff, &d_sbytes),
+        "gen scmnt");
+    void* d_cflag = nullptr;
+    die(tg_device_malloc(s, &d_cflag, ns), "malloc");
+    die(tg_varchar_like_flags(s, d_sbytes, d_soff, ns,
+                              "%Customer%Complaints%", (uint8_t*)d_cflag),
+        "like");
+    std::vector<uint8_t> cf(ns);
+    die(tg_copy_dtoh(s, cf.data(), d_cflag, ns), "dtoh");
+    std::vector<int64_t> bad;
+    for (int64_t i = 0; i < ns; i++)
+        if (cf[i]) bad.push_back(i + 1);
+    tg_join_bridge* brb = nullptr;
+    die(tg_join_bridge_create(s, &brb), "brb");
+    int32_t bt = TG_BIGINT;
+    tg_operator* bb = nullptr;
+    die(tg_set_builder_create(s, brb, &bt, 1, 0, &bb), "setb");
+    tg_block bblk{};
+    bblk.type = TG_BIGINT;
+    bblk.position_count = (int64_t)bad.size();
+    bblk.data = bad.data();
+    tg_page bpage{};
+    bpage.channel_count = 1;
+    bpage.position_count = (int64_t)bad.size();
+    bpage.blocks = &bblk;
+    die(tg_operator_add_input(bb, &bpage), "setb add");
+    die(tg_operator_finish(bb), "setb fin");
+
+    /* part filter: size IN (...) AND brand<>45 AND NOT type BETWEEN 65..69 */
+    std::vector<tg_block> pb = {dev_block(TG_BIGINT, p_pk, np_),
+                                dev_block(TG_TINYINT, p_br, np_),
+                                dev_block(TG_SMALLINT, p_ty, np_),
+                                dev_block(TG_INTEGER, p_sz, np_)};
+    tg_page ppage = dev_page(pb, np_);
+    static const int sizes[8] = {49, 14, 23, 45, 19, 3, 36, 9};
+    std::vector<tg_expr_inst> fe;
+    for (int i = 0; i < 8; i++) {
+        fe.push_back(I_col(3));
+        fe.push_back(I_i64(sizes[i]));
+        fe.push_back(I_op(TG_EXPR_EQ));
+        if (i) fe.push_back(I_op(TG_EXPR_OR));
+    }
+    fe.push_back(I_col(1)); fe.push_back(I_i64(45)); fe.push_back(I_op(TG_EXPR_NE));
+    fe.push_back(I_op(TG_EXPR_AND));
+    fe.push_back(I_col(2)); fe.push_back(I_i64(65)); fe.push_back(I_i64(69));
+    fe.push_back(I_op(TG_EXPR_BETWEEN)); fe.push_back(I_op(TG_EXPR_NOT));
+    fe.push_back(I_op(TG_EXPR_AND));
+    std::vector<tg_expr_inst> c0 = {I_col(0)}, c1 = {I_col(1)}, c2 = {I_col(2)},
+                              c3 = {I_col(3)};
+    tg_expr fex{fe.data(), (int32_t)fe.size()};
+    tg_expr prj[4] = {{c0.data(), 1}, {c1.data(), 1}, {c2.data(), 1}, {c3.data(), 1}};
+    int32_t pot[4] = {TG_BIGINT, TG_TINYINT, TG_SMALLINT, TG_INTEGER};
+    tg_operator* fp = nullptr;
+    die(tg_filter_project_create(s, &fex, prj, pot, 4, &fp), "fp");
+    die(tg_operator_add_input(fp, &ppage), "fp add");
+    die(tg_operator_finish(fp), "fp fin");
+    int fin = 0;
+    tg_page psel{};
+    die(tg_operator_get_output(fp, &psel, &fin), "fp out");
+    tg_join_bridge* brp = nullptr;
+    die(tg_join_bridge_create(s, &brp), "brp");
+    int32_t bty[4] = {TG_BIGINT, TG_TINYINT, TG_SMALLINT, TG_INTEGER};
+    int32_t kc0 = 0, bouts[3] = {1, 2, 3};
+    tg_operator* bp = nullptr;
+    die(tg_hash_builder_create(s, brp, bty, 4, &kc0, 1, bouts, 3, &bp), "bp");
+    die(tg_operator_add_input(bp, &psel), "bp add");
+    die(tg_operator_finish(bp), "bp fin");
+
+    /* partsupp minus complaint suppliers, join part attrs */
+    std::vector<tg_block> psb = {dev_block(TG_BIGINT, ps_pk, np_ * 4),
+                                 dev_block(TG_BIGINT, ps_sk, np_ * 4)};
+    tg_page pspage = dev_page(psb, np_ * 4);
+    tg_operator* sj = nullptr;
+    die(tg_semi_join_create(s, brb, 1, &sj), "semi");
+    die(tg_operator_add_input(sj, &pspage), "semi add");
+    die(tg_operator_finish(sj), "semi fin");
+    tg_page marked{};
+    die(tg_operator_get_output(sj, &marked, &fin), "semi out");
+    std::vector<tg_expr_inst> nb = {I_col(2), I_i64(0), I_op(TG_EXPR_EQ)};
+    std::vector<tg_expr_inst> n0 = {I_col(0)}, n1 = {I_col(1)};
+    tg_expr nbe{nb.data(), (int32_t)nb.size()};
+    tg_expr npr[2] = {{n0.data(), 1}, {n1.data(), 1}};
+    int32_t not_[2] = {TG_BIGINT, TG_BIGINT};
+    tg_operator* fnb = nullptr;
+    die(tg_filter_project_create(s, &nbe, npr, not_, 2, &fnb), "fnb");
+    die(tg_operator_add_input(fnb, &marked), "fnb add");
+    die(tg_operator_finish(fnb), "fnb fin");
+    tg_page ps_ok{};
+    die(tg_operator_get_output(fnb, &ps_ok, &fin), "fnb out");
+    int32_t jt[2] = {TG_BIGINT, TG_BIGINT}, jouts[1] = {1};
+    tg_operator* j = nullptr;
+    die(tg_lookup_join_create(s, brp, jt, 2, &kc0, 1, jouts, 1, &j), "j");
+    die(tg_operator_add_input(j, &ps_ok), "j add");
+    die(tg_operator_finish(j), "j fin");
+    tg_page joined{};
+    die(tg_operator_get_output(j, &joined, &fin), "j out");
+
+    /* pack ((brand*160+type)*64+size)<<32 | suppkey, sorted dedup, count */
+    std::vector<tg_expr_inst> pk = {I_col(1), I_i64(160), I_op(TG_EXPR_MUL),
+                                    I_col(2), I_op(TG_EXPR_ADD), I_i64(64),
+                                    I_op(TG_EXPR_MUL), I_col(3),
+                                    I_op(TG_EXPR_ADD), I_i64(1ll << 32),
+                                    I_op(TG_EXPR_MUL), I_col(0),
+                                    I_op(TG_EXPR_ADD)};
+    tg_expr pke{pk.data(), (int32_t)pk.size()};
+    int32_t pko = TG_BIGINT;
+    tg_operator* fpk = nullptr;
+    die(tg_filter_project_create(s, nullptr, &pke, &pko, 1, &fpk), "pack");
+    die(tg_operator_add_input(fpk, &joined), "pack add");
+    die(tg_operator_finish(fpk), "pack fin");
+    tg_page packed{};
+    die(tg_operator_get_output(fpk, &packed, &fin), "pack out");
+    void* d_dedup = nullptr;
+    die(tg_device_malloc(s, &d_dedup, (packed.position_count ? packed.position_count : 1) * 8),
+        "malloc");
+    int64_t nuniq = 0;
+    die(tg_dedup_i64(s, (const int64_t*)packed.blocks[0].data,
+                     packed.position_count, 52, (int64_t*)d_dedup, &nuniq),
+        "dedup");
+    fprintf(stderr, "[%lld packed pairs -> %lld distinct]\n",
+            (long long)packed.position_count, (long long)nuniq);
+
+    /* count suppliers per combo (packed >> 32) on host (combos ~28k) */
+    std::vector<int64_t> u(nuniq);
+    die(tg_copy_dtoh(s, u.data(), d_dedup, nuniq * 8), "dtoh");
+    std::map<int64_t, int> cnt;
+    for (int64_t v : u) cnt[v >> 32]++;
+    static const char* T1[6] = {"STANDARD", "SMALL", "MEDIUM", "LARGE", "ECONOMY", "PROMO"};
+    static const char* T2[5] = {"ANODIZED", "BURNISHED", "PLATED", "POLISHED", "BRUSHED"};
+    static const char* T3[5] = {"TIN", "NICKEL", "BRASS", "STEEL", "COPPER"};
+    std::vector<std::pair<int64_t, int>> rows(cnt.begin(), cnt.end());
+    std::sort(rows.begin(), rows.end(), [&](auto& a, auto& b) {
+        if (a.second != b.second) return a.second > b.second;
+        return a.first < b.first;
+    });
+    printf("p_brand|p_type|p_size|supplier_cnt\n");
+    int shown = 0;
+    for (auto& r : rows) {
+        long long combo = r.first, szv = combo % 64, ty = (combo / 64) % 160,
+                  br = combo / 64 / 160;
+        printf("Brand#%lld|%s %s %s|%lld|%d\n", br, T1[ty / 25],
+               T2[(ty / 5) % 5], T3[ty % 5], szv, r.second);
+        if (++shown >= 10) break;
+    }
+    for (tg_operator* o : {bb, fp, bp, sj, fnb, j, fpk}) tg_operator_close(o);
+    tg_join_bridge_close(brb);
+    tg_join_bridge_close(brp);
+    for (void* pp : {p_pk, p_ty, p_br, p_sz, ps_pk, ps_sk, d_cflag, d_dedup})
+        die(tg_device_free(s, pp), "free");
+    return 0;
+}
+
 int main(int argc, char** argv)
 {
     if (argc < 2 || !strcmp(argv[1], "--help")) {
-        printf("usage: %s q1|q3|q4|q6|q12|q13|q14|q18 [scale_factor]  (version: %s)\n",
+        printf("usage: %s q1|q3|q4|q6|q12|q13|q14|q16|q18 [scale_factor]  (version: %s)\n",
                argv[0], tg_version());
         return argc < 2 ? 1 : 0;
     }
@@ -757,6 +933,7 @@ int main(int argc, char** argv)
     else if (!strcmp(argv[1], "q12")) rc = run_q12(s, sf);
     else if (!strcmp(argv[1], "q18")) rc = run_q18(s, sf);
     else if (!strcmp(argv[1], "q13")) rc = run_q13(s, sf);
+    else if (!strcmp(argv[1], "q16")) rc = run_q16(s, sf);
     else fprintf(stderr, "unknown query %s\n", argv[1]);
     tg_session_close(s);
     return rc;
