@@ -888,18 +888,29 @@ static int run_q16(tg_session* s, double sf)
     static const char* T1[6] = {"STANDARD", "SMALL", "MEDIUM", "LARGE", "ECONOMY", "PROMO"};
     static const char* T2[5] = {"ANODIZED", "BURNISHED", "PLATED", "POLISHED", "BRUSHED"};
     static const char* T3[5] = {"TIN", "NICKEL", "BRASS", "STEEL", "COPPER"};
-    std::vector<std::pair<int64_t, int>> rows(cnt.begin(), cnt.end());
-    std::sort(rows.begin(), rows.end(), [&](auto& a, auto& b) {
-        if (a.second != b.second) return a.second > b.second;
-        return a.first < b.first;
+    /* ORDER BY supplier_cnt DESC, p_brand, p_type (string), p_size —
+     * the fixture's tie-break */
+    struct Row { std::string brand, type; int size, n; };
+    std::vector<Row> rows;
+    for (auto& kv : cnt) {
+        long long combo = kv.first, szv = combo % 64, ty = (combo / 64) % 160,
+                  br = combo / 64 / 160;
+        char bb[16], tb[48];
+        snprintf(bb, sizeof bb, "Brand#%lld", br);
+        snprintf(tb, sizeof tb, "%s %s %s", T1[ty / 25], T2[(ty / 5) % 5],
+                 T3[ty % 5]);
+        rows.push_back({bb, tb, (int)szv, kv.second});
+    }
+    std::sort(rows.begin(), rows.end(), [](const Row& a, const Row& b) {
+        if (a.n != b.n) return a.n > b.n;
+        if (a.brand != b.brand) return a.brand < b.brand;
+        if (a.type != b.type) return a.type < b.type;
+        return a.size < b.size;
     });
     printf("p_brand|p_type|p_size|supplier_cnt\n");
     int shown = 0;
     for (auto& r : rows) {
-        long long combo = r.first, szv = combo % 64, ty = (combo / 64) % 160,
-                  br = combo / 64 / 160;
-        printf("Brand#%lld|%s %s %s|%lld|%d\n", br, T1[ty / 25],
-               T2[(ty / 5) % 5], T3[ty % 5], szv, r.second);
+        printf("%s|%s|%d|%d\n", r.brand.c_str(), r.type.c_str(), r.size, r.n);
         if (++shown >= 10) break;
     }
     for (tg_operator* o : {bb, fp, bp, sj, fnb, j, fpk}) tg_operator_close(o);
