@@ -89,6 +89,11 @@ def combo_owner_rank(combo, world):
 
 
 def main():
+    # this libomp's first parallel region costs ~2.5 s when OMP_NUM_THREADS
+    # is unset (measured; 0.4 s with it set) — pin the default before the
+    # parquet decode / cpu_baseline legs initialize the runtime. Explicit
+    # omp_set_num_threads calls (the calibrated baseline) still override.
+    os.environ.setdefault("OMP_NUM_THREADS", str(os.cpu_count() or 8))
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)

@@ -788,15 +788,16 @@ extern "C" tg_status tg_parquet_read_columns(tg_session* s, tg_parquet_file* f,
     const int64_t* dict_caps, int32_t** out_dict_offsets,
     int32_t** out_dict_counts)
 {
+    /* columns SEQUENTIAL, row groups parallel (each read_column's inner
+     * `parallel for` uses the full team). The previous nested-team design
+     * (outer team of n_cols, inner teams of threads/n_cols) measured
+     * 0.5-3.7 s for 6 columns x 24M rows where the sequential-columns
+     * version takes ~0.1 s: with threads/n_cols == 1 every column decoded
+     * its row groups serially, and the per-call nested team setup
+     * thrashed. */
     tg_status st = TG_OK;
-    int total = omp_get_max_threads();
-    int inner = total / (n_cols > 0 ? n_cols : 1);
-    if (inner < 1) inner = 1;
-    omp_set_max_active_levels(2);
-    #pragma omp parallel for num_threads(n_cols) schedule(dynamic)
-    for (int32_t c = 0; c < n_cols; c++) {
-        omp_set_num_threads(inner);
-        tg_status cst = tg_parquet_read_column(
+    for (int32_t c = 0; c < n_cols && st == TG_OK; c++) {
+        st = tg_parquet_read_column(
             s, f, cols[c], out_values ? out_values[c] : nullptr,
             out_valid ? out_valid[c] : nullptr,
             out_ids ? out_ids[c] : nullptr,
@@ -804,7 +805,6 @@ extern "C" tg_status tg_parquet_read_columns(tg_session* s, tg_parquet_file* f,
             dict_caps ? dict_caps[c] : 0,
             out_dict_offsets ? out_dict_offsets[c] : nullptr,
             out_dict_counts ? out_dict_counts[c] : nullptr);
-        if (cst != TG_OK) st = cst;
     }
     return st;
 }
